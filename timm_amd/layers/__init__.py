@@ -1,0 +1,49 @@
+from .activations import *
+from .adaptive_avgmax_pool import (
+    AdaptiveAvgMaxPool2d, AdaptiveCatAvgMaxPool2d, SelectAdaptivePool2d,
+    adaptive_avgmax_pool2d, adaptive_catavgmax_pool2d, adaptive_pool_feat_mult, select_adaptive_pool2d,
+)
+from .attention import Attention, AttentionRope, maybe_add_mask
+from .attention_pool import AttentionPoolLatent
+from .classifier import ClassifierHead, NormMlpClassifierHead, create_classifier
+from .cond_conv2d import CondConv2d, get_condconv_initializer
+from .config import (
+    is_exportable, is_no_jit, is_scriptable, set_exportable, set_layer_config, set_no_jit, set_scriptable,
+    set_fused_attn, use_fused_attn, set_reentrant_ckpt, use_reentrant_ckpt,
+)
+from .conv2d_same import Conv2dSame, Conv2dSameExport, conv2d_same
+from .conv_bn_act import ConvBnAct, ConvNormAct, ConvNormActAa
+from .create_act import create_act_layer, get_act_fn, get_act_layer
+from .create_conv2d import create_conv2d
+from .create_norm import create_norm_layer, get_norm_layer
+from .drop import DropBlock2d, DropPath, calculate_drop_path_rates, drop_block_2d, drop_path
+from .format import Format, FormatT, get_channel_dim, get_spatial_dim, nchw_to, nhwc_to
+from .grn import GlobalResponseNorm
+from .helpers import extend_tuple, make_divisible, to_2tuple, to_3tuple, to_4tuple, to_ntuple
+from .layer_scale import LayerScale, LayerScale2d
+from .linear import Linear
+from .mixed_conv2d import MixedConv2d
+from .mlp import ConvMlp, GatedMlp, GlobalResponseNormMlp, GluMlp, Mlp, SwiGLU, SwiGLUPacked
+from .norm import (
+    GroupNorm, GroupNorm1, LayerNorm, LayerNorm2d, LayerNormFp32, RmsNorm, RmsNorm2d, RmsNormFp32,
+    SimpleNorm, SimpleNorm2d,
+)
+from .norm_act import (
+    BatchNormAct2d, FrozenBatchNormAct2d, GroupNormAct, LayerNormAct2d, SyncBatchNormAct,
+    convert_sync_batchnorm, freeze_batch_norm_2d, get_norm_act_layer, unfreeze_batch_norm_2d,
+)
+from .padding import get_padding, get_same_padding, pad_same
+from .patch_dropout import PatchDropout, patch_dropout_forward
+from .patch_embed import PatchEmbed, PatchEmbedWithSize, resample_patch_embed
+from .pool2d_same import AvgPool2dSame, MaxPool2dSame, create_pool2d
+from .pos_embed import resample_abs_pos_embed, resample_abs_pos_embed_nhwc
+from .pos_embed_sincos import (
+    FourierEmbed, RotaryEmbedding, RotaryEmbeddingCat, apply_keep_indices_nlc, apply_rot_embed,
+    apply_rot_embed_cat, apply_rot_embed_list, build_fourier_pos_embed, build_rotary_pos_embed,
+    build_sincos2d_pos_embed, create_rope_embed, freq_bands, pixel_freq_bands, rope_rotate_half, rot,
+)
+from .squeeze_excite import EffectiveSEModule, EffectiveSqueezeExcite, SEModule, SqueezeExcite, SqueezeExciteCl
+from .trace_utils import _assert
+from .weight_init import (
+    init_weight_jax, init_weight_vit, lecun_normal_, trunc_normal_, trunc_normal_tf_, variance_scaling_,
+)

@@ -1,0 +1,73 @@
+"""Conv + Norm + Act block (reference `timm/layers/conv_bn_act.py`)."""
+from typing import Any, Dict, Optional, Type
+
+from torch import nn as nn
+
+from .create_conv2d import create_conv2d
+from .norm_act import get_norm_act_layer
+
+
+class ConvNormAct(nn.Module):
+    def __init__(
+            self,
+            in_channels: int,
+            out_channels: int,
+            kernel_size: int = 1,
+            stride: int = 1,
+            padding: Any = '',
+            dilation: int = 1,
+            groups: int = 1,
+            bias: bool = False,
+            apply_norm: bool = True,
+            apply_act: bool = True,
+            norm_layer: Type[nn.Module] = nn.BatchNorm2d,
+            act_layer: Optional[Type[nn.Module]] = nn.ReLU,
+            drop_layer: Optional[Type[nn.Module]] = None,
+            conv_kwargs: Optional[Dict[str, Any]] = None,
+            norm_kwargs: Optional[Dict[str, Any]] = None,
+            act_kwargs: Optional[Dict[str, Any]] = None,
+    ):
+        super().__init__()
+        conv_kwargs = conv_kwargs or {}
+        norm_kwargs = norm_kwargs or {}
+        act_kwargs = act_kwargs or {}
+        use_aa = False
+
+        self.conv = create_conv2d(
+            in_channels, out_channels, kernel_size, stride=stride,
+            padding=padding, dilation=dilation, groups=groups, bias=bias, **conv_kwargs)
+
+        if apply_norm:
+            # NOTE for backwards compatibility with models that use separate norm and act layer definitions
+            norm_act_layer = get_norm_act_layer(norm_layer, act_layer)
+            # NOTE for backwards (weight) compatibility, norm layer name remains `.bn`
+            if drop_layer:
+                norm_kwargs['drop_layer'] = drop_layer
+            self.bn = norm_act_layer(
+                out_channels,
+                apply_act=apply_act,
+                act_kwargs=act_kwargs,
+                **norm_kwargs,
+            )
+        else:
+            self.bn = nn.Sequential()
+            if drop_layer:
+                norm_kwargs['drop_layer'] = drop_layer
+                self.bn.add_module('drop', drop_layer())
+
+    @property
+    def in_channels(self):
+        return self.conv.in_channels
+
+    @property
+    def out_channels(self):
+        return self.conv.out_channels
+
+    def forward(self, x):
+        x = self.conv(x)
+        x = self.bn(x)
+        return x
+
+
+ConvBnAct = ConvNormAct
+ConvNormActAa = ConvNormAct  # anti-aliasing variant collapse (no aa by default)

@@ -1,0 +1,33 @@
+from ._builder import (
+    build_model_with_cfg, load_custom_pretrained, load_pretrained, pretrained_cfg_for_features,
+    resolve_pretrained_cfg, set_pretrained_download_progress, set_pretrained_check_hash,
+)
+from ._factory import create_model, parse_model_name, safe_model_name
+from ._features import (
+    FeatureInfo, FeatureHooks, FeatureDictNet, FeatureListNet, FeatureHookNet, FeatureGetterNet,
+    feature_take_indices,
+)
+from ._features_fx import (
+    FeatureGraphNet, GraphExtractNet, create_feature_extractor, get_graph_node_names,
+    register_notrace_module, is_notrace_module, get_notrace_modules,
+    register_notrace_function, is_notrace_function, get_notrace_functions,
+)
+from ._helpers import clean_state_dict, load_state_dict, load_checkpoint, remap_state_dict, resume_checkpoint
+from ._hub import (
+    load_model_config_from_hf, load_state_dict_from_hf, push_to_hf_hub,
+)
+from ._manipulate import (
+    model_parameters, named_apply, named_modules, named_modules_with_params, group_modules,
+    group_parameters, checkpoint, checkpoint_seq, adapt_input_conv,
+)
+from ._pretrained import PretrainedCfg, DefaultCfg, filter_pretrained_cfg
+from ._prune import adapt_model_from_string
+from ._registry import (
+    split_model_name_tag, get_arch_name, generate_default_cfgs, register_model,
+    register_model_deprecations, model_entrypoint, list_models, list_pretrained, get_deprecated_models,
+    is_model, list_modules, is_model_in_modules, is_model_pretrained, get_pretrained_cfg,
+    get_pretrained_cfg_value, get_pretrained_cfgs_for_arch,
+)
+
+# architecture modules (registration happens at import time)
+from .vision_transformer import *

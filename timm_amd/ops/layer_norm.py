@@ -1,0 +1,104 @@
+"""Fused LayerNorm / RMSNorm dispatch.
+
+Replaces the reference's `F.layer_norm`/`F.rms_norm` hot path
+(`timm/layers/norm.py:70-290`, `timm/layers/fast_norm.py:119-160`) with a
+single-pass gfx950 HIP kernel (fp32 accumulators, vectorized bf16 loads).
+CPU path uses the plain PyTorch composition and serves as the numerics
+reference for GPU tests.
+"""
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import _load_extension
+
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        ext = _load_extension()
+        x = x.contiguous()
+        y, mean, rstd = ext.layer_norm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load_extension()
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ext.layer_norm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+class _RmsNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, eps):
+        ext = _load_extension()
+        x = x.contiguous()
+        y, rstd = ext.rms_norm_fwd(x, weight, eps)
+        ctx.save_for_backward(x, weight, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        ext = _load_extension()
+        x, weight, rstd = ctx.saved_tensors
+        dx, dw = ext.rms_norm_bwd(dy.contiguous(), x, weight, rstd)
+        return dx, dw, None
+
+
+def layer_norm(
+        x: torch.Tensor,
+        normalized_shape,
+        weight: Optional[torch.Tensor] = None,
+        bias: Optional[torch.Tensor] = None,
+        eps: float = 1e-6,
+) -> torch.Tensor:
+    """LayerNorm over the last dim. HIP fused kernel on device, F.layer_norm on CPU."""
+    if x.is_cuda and _load_extension() is not None and weight is not None \
+            and len(normalized_shape) == 1 and x.shape[-1] == normalized_shape[0]:
+        if bias is None:
+            bias = torch.zeros_like(weight)
+        return _LayerNormFn.apply(x, weight, bias, eps)
+    if x.is_cuda:
+        from . import use_hip
+        use_hip(x)  # raises if ext required but missing
+    return F.layer_norm(x, normalized_shape, weight, bias, eps)
+
+
+def rms_norm(
+        x: torch.Tensor,
+        normalized_shape,
+        weight: Optional[torch.Tensor] = None,
+        eps: float = 1e-6,
+) -> torch.Tensor:
+    if x.is_cuda and _load_extension() is not None and weight is not None \
+            and len(normalized_shape) == 1 and x.shape[-1] == normalized_shape[0]:
+        return _RmsNormFn.apply(x, weight, eps)
+    if x.is_cuda:
+        from . import use_hip
+        use_hip(x)
+    # fp32 reference path
+    dtype = x.dtype
+    v = x.float().pow(2).mean(dim=-1, keepdim=True)
+    y = x.float() * torch.rsqrt(v + eps)
+    if weight is not None:
+        y = y * weight.float()
+    return y.to(dtype)
+
+
+# aliases used by norm layer modules; `_act` variants exist so a future fused
+# norm+activation epilogue can slot in without touching callers.
+def layer_norm_act(x, normalized_shape, weight=None, bias=None, eps=1e-6, act=None):
+    y = layer_norm(x, normalized_shape, weight, bias, eps)
+    if act is not None:
+        y = act(y)
+    return y
+
+
+def rms_norm_act(x, normalized_shape, weight=None, eps=1e-6, act=None):
+    y = rms_norm(x, normalized_shape, weight, eps)
+    if act is not None:
+        y = act(y)
+    return y
