@@ -1,0 +1,150 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract: flagship training step throughput.
+
+Measures BASELINE.json's headline metric — train samples/sec for
+vit_base_patch16_224 at 224x224 bf16 — on N GPUs of one node (weak scaling,
+one rank per GPU over RCCL).  Protocol mirrors the reference's benchmark.py
+(synthetic randn inputs, random-init weights, warmup + timed steps with
+device sync around the timed region — /root/reference/benchmark.py:305-420).
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+For N>1 the driver launches via torch.distributed.run with one rank per GPU;
+rank/world info is read from the env.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=20)
+    p.add_argument('--warmup', type=int, default=5)
+    p.add_argument('--model', type=str, default='vit_base_patch16_224')
+    p.add_argument('--batch-size', type=int, default=256, help='per-GPU batch size')
+    p.add_argument('--img-size', type=int, default=224)
+    p.add_argument('--mode', type=str, default='train', choices=['train', 'infer'])
+    p.add_argument('--amp-dtype', type=str, default='bfloat16')
+    return p.parse_args()
+
+
+def main():
+    args = parse_args()
+
+    world_size = int(os.environ.get('WORLD_SIZE', 1))
+    rank = int(os.environ.get('RANK', 0))
+    local_rank = int(os.environ.get('LOCAL_RANK', 0))
+    distributed = world_size > 1
+
+    assert torch.cuda.is_available(), 'bench.py requires a GPU (MI355X)'
+    device = torch.device(f'cuda:{local_rank}')
+    torch.cuda.set_device(device)
+
+    if distributed:
+        torch.distributed.init_process_group(backend='nccl')
+
+    import timm_amd
+    from timm_amd import ops
+    assert ops.has_ext(), 'timm_amd._C HIP extension must be built for bench'
+
+    dtype = torch.bfloat16 if args.amp_dtype == 'bfloat16' else torch.float16
+    torch.manual_seed(42 + rank)
+
+    model = timm_amd.create_model(args.model, num_classes=1000)
+    model = model.to(device=device, dtype=dtype)
+
+    if args.mode == 'train':
+        model.train()
+        from timm_amd.optim import AdamW
+        if distributed:
+            from timm_amd.parallel import BucketedDataParallel
+            model = BucketedDataParallel(model, bucket_cap_mb=50.)
+        optimizer = AdamW(model.parameters(), lr=1e-4, weight_decay=0.05)
+    else:
+        model.eval()
+        optimizer = None
+
+    B = args.batch_size
+    x = torch.randn(B, 3, args.img_size, args.img_size, device=device, dtype=dtype)
+    target = torch.randint(0, 1000, (B,), device=device)
+
+    def train_step():
+        for bucket in getattr(model, '_buckets', []):
+            bucket.flat.zero_()
+        if not distributed:
+            optimizer.zero_grad(set_to_none=True)
+        out = model(x)
+        loss = torch.nn.functional.cross_entropy(out.float(), target)
+        loss.backward()
+        if distributed:
+            model.finish_gradient_sync()
+        optimizer.step()
+
+    @torch.no_grad()
+    def infer_step():
+        model(x)
+
+    step = train_step if args.mode == 'train' else infer_step
+
+    # warmup
+    for _ in range(args.warmup):
+        step()
+
+    # timed region: barrier + sync on both sides, exactly K steps
+    if distributed:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    torch.cuda.synchronize()
+    if distributed:
+        torch.distributed.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    # max over ranks (elapsed measured after barrier; use all_reduce MAX on time)
+    if distributed:
+        t = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = t.item()
+
+    ms_per_step = elapsed / args.steps * 1000.
+    n_gpus = world_size
+    samples_per_sec = B * n_gpus * args.steps / elapsed  # whole-job aggregate
+
+    baseline = 390.86 if args.mode == 'train' else 2992.79  # reference benchmark CSVs
+
+    if rank == 0:
+        result = {
+            'metric': f'{args.mode}_samples_per_sec',
+            'value': round(samples_per_sec, 2),
+            'unit': 'samples/sec',
+            'n_gpus': n_gpus,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': round(ms_per_step, 3),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': round(samples_per_sec / baseline, 3) if baseline else None,
+            'dtype': 'bf16' if dtype == torch.bfloat16 else 'fp16',
+            'data': 'synthetic',
+            'config': {
+                'model': args.model,
+                'global_batch': B * n_gpus,
+                'img_size': args.img_size,
+                'parallelism': f'dp{n_gpus}',
+            },
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -1,0 +1,175 @@
+"""Model tests in the reference's style (`/root/reference/tests/test_models.py`):
+parametrized forward/backward over the registry with tiny inputs, default-cfg
+invariants (forward_features unpooled shape, reset_classifier, global_pool='').
+CPU-only; GPU parity lives in test_ops_gpu.py.
+"""
+import fnmatch
+
+import pytest
+import torch
+import torch.nn as nn
+
+import timm_amd
+from timm_amd import list_models, create_model
+
+# tiny input sizes to keep CPU runtime sane
+TARGET_FWD_SIZE = MAX_FWD_SIZE = 384
+TARGET_BWD_SIZE = 128
+MAX_BWD_SIZE = 320
+
+# models too big for CPU CI
+EXCLUDE_FILTERS = ['*giant*', '*huge*', '*so400m*', '*_large*', 'vit_large*', 'eva02_large*', 'eva_giant*']
+
+
+def _get_input_size(model=None, model_name='', target=None):
+    if model is None:
+        assert model_name, "One of model or model_name must be provided"
+        input_size = timm_amd.get_pretrained_cfg_value(model_name, 'input_size')
+        fixed_input_size = timm_amd.get_pretrained_cfg_value(model_name, 'fixed_input_size')
+        min_input_size = timm_amd.get_pretrained_cfg_value(model_name, 'min_input_size')
+    else:
+        default_cfg = model.default_cfg
+        input_size = default_cfg['input_size']
+        fixed_input_size = default_cfg.get('fixed_input_size', None)
+        min_input_size = default_cfg.get('min_input_size', None)
+    assert input_size is not None
+
+    if fixed_input_size:
+        return input_size
+    if min_input_size:
+        if target and max(input_size) > target:
+            input_size = min_input_size
+    else:
+        if target and max(input_size) > target:
+            input_size = tuple([min(x, target) for x in input_size])
+    return input_size
+
+
+def _models(filters='', exclude=()):
+    return list_models(filter=filters, exclude_filters=list(EXCLUDE_FILTERS) + list(exclude))
+
+
+@pytest.mark.base
+@pytest.mark.parametrize('model_name', _models())
+@pytest.mark.parametrize('batch_size', [1])
+def test_model_forward(model_name, batch_size):
+    """Run a single forward pass with each model."""
+    model = create_model(model_name, pretrained=False, num_classes=10)
+    model.eval()
+
+    input_size = _get_input_size(model=model, target=TARGET_FWD_SIZE)
+    if max(input_size) > MAX_FWD_SIZE:
+        pytest.skip("Fixed input size model > limit.")
+    inputs = torch.randn((batch_size, *input_size))
+    outputs = model(inputs)
+
+    assert outputs.shape[0] == batch_size
+    assert not torch.isnan(outputs).any(), 'Output included NaNs'
+
+
+@pytest.mark.base
+@pytest.mark.parametrize('model_name', _models(exclude=['*_384*', '*_448*', '*_512*']))
+@pytest.mark.parametrize('batch_size', [2])
+def test_model_backward(model_name, batch_size):
+    """Run a single forward + backward pass with each model."""
+    input_size = _get_input_size(model_name=model_name, target=TARGET_BWD_SIZE)
+    if max(input_size) > MAX_BWD_SIZE:
+        pytest.skip("Fixed input size model > limit.")
+
+    model = create_model(model_name, pretrained=False, num_classes=10)
+    model.train()
+    num_params = sum([x.numel() for x in model.parameters()])
+
+    inputs = torch.randn((batch_size, *input_size))
+    outputs = model(inputs)
+    if isinstance(outputs, tuple):
+        outputs = torch.cat(outputs)
+    outputs.mean().backward()
+    for n, x in model.named_parameters():
+        assert x.grad is not None, f'No gradient for {n}'
+    num_grad = sum([x.grad.numel() for x in model.parameters() if x.grad is not None])
+
+    assert outputs.shape[-1] == 10
+    assert num_params == num_grad, 'Some parameters are missing gradients'
+    assert not torch.isnan(outputs).any(), 'Output included NaNs'
+
+
+@pytest.mark.base
+@pytest.mark.parametrize('model_name', ['vit_base_patch16_224', 'vit_tiny_patch16_224'])
+def test_model_default_cfgs(model_name):
+    """Run a single forward pass with each model in feature extraction mode."""
+    model = create_model(model_name, pretrained=False)
+    model.eval()
+    state_dict = model.state_dict()
+    cfg = model.default_cfg
+
+    pool_size = cfg.get('pool_size', None)
+    input_size = model.default_cfg['input_size']
+
+    input_tensor = torch.randn((1, *input_size))
+
+    # test forward_features (always unpooled)
+    outputs = model.forward_features(input_tensor)
+    assert outputs.shape[-1] == model.num_features
+
+    # test forward after deleting the classifier, output should be poolled, size(-1) == model.num_features
+    model.reset_classifier(0)
+    outputs = model.forward(input_tensor)
+    assert len(outputs.shape) == 2
+    assert outputs.shape[-1] == model.num_features
+
+    # test model forward without pooling and classifier
+    model.reset_classifier(0, '')  # reset classifier and disable global pooling
+    outputs = model.forward(input_tensor)
+    assert len(outputs.shape) == 3
+    assert outputs.shape[-1] == model.num_features
+
+    # check classifier name matches default_cfg
+    if cfg.get('num_classes', None):
+        classifier = cfg['classifier']
+        if not isinstance(classifier, (tuple, list)):
+            classifier = classifier,
+        for c in classifier:
+            assert c + ".weight" in state_dict.keys(), f'{c} not in model params'
+
+    # check first conv(s) names match default_cfg
+    first_conv = cfg.get('first_conv', None)
+    if isinstance(first_conv, str):
+        first_conv = (first_conv,)
+    if first_conv is not None:
+        assert isinstance(first_conv, (tuple, list))
+        for fc in first_conv:
+            assert fc + ".weight" in state_dict.keys(), f'{fc} not in model params'
+
+
+@pytest.mark.base
+def test_forward_intermediates():
+    model = create_model('vit_tiny_patch16_224', num_classes=10)
+    model.eval()
+    x = torch.randn(1, 3, 224, 224)
+    final, intermediates = model.forward_intermediates(x, indices=[2, 5, 11])
+    assert len(intermediates) == 3
+    for inter in intermediates:
+        assert inter.ndim == 4
+        assert inter.shape[1] == model.embed_dim
+        assert inter.shape[-2:] == (14, 14)
+
+
+@pytest.mark.base
+def test_features_only():
+    model = create_model('vit_tiny_patch16_224', features_only=True)
+    model.eval()
+    x = torch.randn(1, 3, 224, 224)
+    out = model(x)
+    assert isinstance(out, list)
+    assert all(o.ndim == 4 for o in out)
+
+
+@pytest.mark.base
+def test_registry_wildcards():
+    models = list_models('vit_base*')
+    assert 'vit_base_patch16_224' in models
+    assert timm_amd.is_model('vit_base_patch16_224')
+    assert not timm_amd.is_model('definitely_not_a_model')
+    pretrained_cfg = timm_amd.get_pretrained_cfg('vit_base_patch16_224')
+    assert pretrained_cfg.input_size == (3, 224, 224)

@@ -1,0 +1,252 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference.
+
+Run on the MI355X box: pytest tests/test_ops_gpu.py -m gpu -x -q
+"""
+import math
+
+import pytest
+import torch
+
+import timm_amd  # noqa
+from timm_amd import ops
+
+pytestmark = pytest.mark.gpu
+
+
+def _ext():
+    assert ops.has_ext(), "HIP extension must be built (python setup.py build_ext --inplace)"
+    return ops.require_ext()
+
+
+def rel_err(a, b):
+    a, b = a.float(), b.float()
+    return ((a - b).norm() / b.norm().clamp(min=1e-8)).item()
+
+
+@pytest.mark.parametrize('shape', [(8, 197, 768), (4, 50, 1024), (2, 7, 640)])
+@pytest.mark.parametrize('dtype', [torch.bfloat16, torch.float32])
+def test_layer_norm_fwd_bwd(shape, dtype):
+    _ext()
+    torch.manual_seed(0)
+    x = torch.randn(shape, device='cuda', dtype=dtype)
+    w = torch.randn(shape[-1], device='cuda', dtype=dtype) * 0.1 + 1.0
+    b = torch.randn(shape[-1], device='cuda', dtype=dtype) * 0.1
+
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True)
+    y = ops.layer_norm(x1, (shape[-1],), w1, b1, 1e-6)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    # fp32 reference
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    y_ref = torch.nn.functional.layer_norm(x2, (shape[-1],), w2, b2, 1e-6)
+    y_ref.backward(dy.float())
+
+    tol = 2e-2 if dtype == torch.bfloat16 else 2e-5
+    assert rel_err(y, y_ref) < tol
+    assert rel_err(x1.grad, x2.grad) < tol * 2
+    assert rel_err(w1.grad, w2.grad) < tol * 2
+    assert rel_err(b1.grad, b2.grad) < tol * 2
+
+
+@pytest.mark.parametrize('shape', [(8, 197, 768), (3, 33, 512)])
+def test_rms_norm_fwd_bwd(shape):
+    _ext()
+    torch.manual_seed(1)
+    dtype = torch.bfloat16
+    x = torch.randn(shape, device='cuda', dtype=dtype)
+    w = torch.randn(shape[-1], device='cuda', dtype=dtype) * 0.1 + 1.0
+
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    y = ops.rms_norm(x1, (shape[-1],), w1, 1e-6)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    v = x2.pow(2).mean(-1, keepdim=True)
+    y_ref = x2 * torch.rsqrt(v + 1e-6) * w2
+    y_ref.backward(dy.float())
+
+    assert rel_err(y, y_ref) < 2e-2
+    assert rel_err(x1.grad, x2.grad) < 4e-2
+    assert rel_err(w1.grad, w2.grad) < 4e-2
+
+
+@pytest.mark.parametrize('act', ['gelu', 'gelu_tanh', 'silu', 'relu', 'quick_gelu'])
+def test_bias_act(act):
+    _ext()
+    torch.manual_seed(2)
+    x = torch.randn(64, 3072, device='cuda', dtype=torch.bfloat16)
+    b = torch.randn(3072, device='cuda', dtype=torch.bfloat16)
+    x1 = x.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True)
+    y = ops.bias_act(x1, b1, act)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x2 = x.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    h = x2 + b2
+    if act == 'gelu':
+        y_ref = torch.nn.functional.gelu(h)
+    elif act == 'gelu_tanh':
+        y_ref = torch.nn.functional.gelu(h, approximate='tanh')
+    elif act == 'silu':
+        y_ref = torch.nn.functional.silu(h)
+    elif act == 'relu':
+        y_ref = torch.relu(h)
+    else:
+        y_ref = h * torch.sigmoid(1.702 * h)
+    y_ref.backward(dy.float())
+
+    assert rel_err(y, y_ref) < 2e-2
+    assert rel_err(x1.grad, x2.grad) < 3e-2
+    assert rel_err(b1.grad, b2.grad) < 3e-2
+
+
+def test_residual_scale_add():
+    _ext()
+    torch.manual_seed(3)
+    B, N, C = 8, 197, 768
+    x = torch.randn(B, N, C, device='cuda', dtype=torch.bfloat16)
+    y = torch.randn(B, N, C, device='cuda', dtype=torch.bfloat16)
+    g = torch.randn(C, device='cuda', dtype=torch.bfloat16) * 0.1
+
+    x1 = x.clone().requires_grad_(True)
+    y1 = y.clone().requires_grad_(True)
+    g1 = g.clone().requires_grad_(True)
+    out = ops.residual_scale_add(x1, y1, g1)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    x2 = x.detach().float().requires_grad_(True)
+    y2 = y.detach().float().requires_grad_(True)
+    g2 = g.detach().float().requires_grad_(True)
+    out_ref = x2 + y2 * g2
+    out_ref.backward(dout.float())
+
+    assert rel_err(out, out_ref) < 2e-2
+    assert rel_err(x1.grad, x2.grad) < 2e-2
+    assert rel_err(y1.grad, y2.grad) < 2e-2
+    assert rel_err(g1.grad, g2.grad) < 3e-2
+
+
+@pytest.mark.parametrize('shape', [
+    (2, 12, 197, 64),    # ViT-B/224
+    (1, 16, 577, 64),    # EVA-L/336
+    (2, 4, 100, 64),     # ragged N
+    (1, 2, 33, 128),     # D=128
+    (1, 2, 64, 96),      # D=96 (unswizzled path)
+    (2, 8, 1, 64),       # q_len=1 (attention pool uses Nq==Nk, but exercise small N)
+])
+def test_attention_fwd_bwd(shape):
+    _ext()
+    torch.manual_seed(4)
+    B, H, N, D = shape
+    q = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    k = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    v = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+
+    q1, k1, v1 = [t.clone().requires_grad_(True) for t in (q, k, v)]
+    o = ops.flash_attention(q1, k1, v1)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    q2, k2, v2 = [t.detach().float().requires_grad_(True) for t in (q, k, v)]
+    scale = 1.0 / math.sqrt(D)
+    attn = (q2 @ k2.transpose(-2, -1)) * scale
+    o_ref = attn.softmax(-1) @ v2
+    o_ref.backward(do.float())
+
+    assert rel_err(o, o_ref) < 2e-2, f'fwd err {rel_err(o, o_ref)}'
+    assert rel_err(q1.grad, q2.grad) < 4e-2
+    assert rel_err(k1.grad, k2.grad) < 4e-2
+    assert rel_err(v1.grad, v2.grad) < 4e-2
+
+
+def test_attention_mask():
+    _ext()
+    torch.manual_seed(5)
+    B, H, N, D = 2, 4, 70, 64
+    q = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    k = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    v = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    # NaFlex-style padding mask: last 10 keys of batch 1 are invalid
+    valid = torch.ones(B, N, device='cuda', dtype=torch.bool)
+    valid[1, -10:] = False
+    mask = torch.zeros(B, 1, N, N, device='cuda')
+    mask.masked_fill_(~valid.view(B, 1, 1, N), float('-inf'))
+
+    o = ops.flash_attention(q, k, v, attn_mask=mask)
+
+    q2, k2, v2 = q.float(), k.float(), v.float()
+    attn = (q2 @ k2.transpose(-2, -1)) / math.sqrt(D) + mask
+    o_ref = attn.softmax(-1) @ v2
+
+    assert rel_err(o, o_ref) < 2e-2
+
+
+def test_fused_adamw_matches_reference():
+    _ext()
+    torch.manual_seed(6)
+    shapes = [(768, 768), (3072,), (16, 3, 4, 4), (1000, 768)]
+    params = [torch.randn(s, device='cuda', dtype=torch.float32) for s in shapes]
+    grads = [torch.randn(s, device='cuda', dtype=torch.float32) for s in shapes]
+    ref_params = [p.clone() for p in params]
+    m = [torch.zeros_like(p) for p in params]
+    v = [torch.zeros_like(p) for p in params]
+    m_ref = [torch.zeros_like(p) for p in params]
+    v_ref = [torch.zeros_like(p) for p in params]
+
+    lr, b1, b2, eps, wd = 1e-3, 0.9, 0.999, 1e-8, 0.05
+    for step in range(1, 4):
+        ops.fused_adamw_step(params, grads, m, v, lr, b1, b2, eps, wd, step)
+        # reference math
+        for p, g, mm, vv in zip(ref_params, grads, m_ref, v_ref):
+            p.mul_(1 - lr * wd)
+            mm.mul_(b1).add_(g, alpha=1 - b1)
+            vv.mul_(b2).addcmul_(g, g, value=1 - b2)
+            bc1 = 1 - b1 ** step
+            bc2 = 1 - b2 ** step
+            denom = (vv.sqrt() / math.sqrt(bc2)).add_(eps)
+            p.addcdiv_(mm, denom, value=-lr / bc1)
+
+    for p, pr in zip(params, ref_params):
+        assert rel_err(p, pr) < 1e-5
+
+
+def test_fused_lerp_and_l2norm():
+    _ext()
+    torch.manual_seed(7)
+    a = [torch.randn(100, 50, device='cuda'), torch.randn(333, device='cuda')]
+    b = [torch.randn_like(t) for t in a]
+    a_ref = [t.clone() for t in a]
+    ops.fused_lerp_(a, b, 0.25)
+    torch._foreach_lerp_(a_ref, b, 0.25)
+    for x, y in zip(a, a_ref):
+        assert rel_err(x, y) < 1e-6
+
+    n = ops.fused_l2norm(a)
+    n_ref = torch.linalg.vector_norm(torch.cat([t.flatten() for t in a]))
+    assert abs(n.item() - n_ref.item()) / n_ref.item() < 1e-5
+
+
+def test_vit_block_gpu_vs_cpu():
+    """Full ViT block on GPU (all HIP kernels) vs CPU fp32 reference."""
+    _ext()
+    torch.manual_seed(8)
+    import timm_amd
+    model = timm_amd.create_model('vit_tiny_patch16_224', num_classes=10)
+    model.eval()
+    x = torch.randn(2, 3, 224, 224)
+    with torch.no_grad():
+        y_cpu = model(x.float())
+        m_gpu = model.to('cuda', torch.bfloat16)
+        y_gpu = m_gpu(x.to('cuda', torch.bfloat16))
+    assert rel_err(y_gpu.cpu(), y_cpu) < 5e-2, f'model output err {rel_err(y_gpu.cpu(), y_cpu)}'

@@ -27,7 +27,7 @@ def attention_available(q: torch.Tensor) -> bool:
     if not q.is_cuda or _load_extension() is None:
         return False
     D = q.shape[-1]
-    return D <= 128 and q.dtype in (torch.bfloat16, torch.float16)
+    return D <= 128 and D % 32 == 0 and q.dtype == torch.bfloat16
 
 
 def _math_sdpa(q, k, v, attn_mask=None, scale=None):

@@ -1,0 +1,256 @@
+// Flash-style fused multi-head attention forward for gfx950 (CDNA4).
+//
+// Replaces the reference's F.scaled_dot_product_attention hot path
+// (timm/layers/attention.py:124-129, timm/models/eva.py:246-251).
+//
+// Structure (correctness-first v1, per the CDNA4 guide's fused-attn anatomy):
+//  * 256-thread block = 4 waves; each block owns a 64-row Q tile of one (b,h)
+//  * K/V staged in LDS in 32-row tiles; Q staged once per block
+//  * QK^T and PV on MFMA v_mfma_f32_16x16x32_bf16 (per-wave 16x16 tiles)
+//  * online softmax with fp32 running max/sum state per q-row
+//    (C/D fragment layout: row=(lane>>4)*4+reg, col=lane&15 — all lanes of a
+//     16-lane group share 4 rows, row-reduce = shfl over 16 lanes)
+//  * optional additive fp32 mask [B,1,Nq,Nk] (NaFlex padding masks)
+//  * returns O (bf16) + logsumexp (fp32) for the exact GEMM-based backward
+//
+// Supported: dtype bf16, head_dim in {32,64,96,128}, any Nq/Nk.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include "common.h"
+
+namespace {
+
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+
+constexpr int kQTile = 64;     // q rows per block
+constexpr int kKvTile = 32;    // kv rows per LDS stage
+constexpr int kBlockThreads = 256;
+constexpr float kNegInf = -1e30f;
+
+// LDS XOR swizzle for stride-D row-major bf16 tiles (guide §6 Guideline 4):
+// byte_off ^= (row & 7) << 4 spreads the 16 same-column lanes over 8 banks.
+template <bool kSwz>
+__device__ __forceinline__ int swz(int row, int byte_off) {
+  return kSwz ? (byte_off ^ ((row & 7) << 4)) : byte_off;
+}
+
+template <int kMaxD, bool kHasMask, bool kSwizzle>
+__global__ __launch_bounds__(kBlockThreads)
+void attn_fwd_kernel(
+    const __bf16* __restrict__ q,     // [B,H,Nq,D]
+    const __bf16* __restrict__ k,     // [B,H,Nk,D]
+    const __bf16* __restrict__ v,     // [B,H,Nk,D]
+    const float* __restrict__ mask,   // [B,1,Nq,Nk] or null
+    __bf16* __restrict__ o,           // [B,H,Nq,D]
+    float* __restrict__ lse,          // [B,H,Nq]
+    int B, int H, int Nq, int Nk, int D, float scale) {
+  // LDS: Q[64][D] | K[32][D] | V[32][D] | P[4][16][32]
+  __shared__ __bf16 q_lds[kQTile * kMaxD];
+  __shared__ __bf16 k_lds[kKvTile * kMaxD];
+  __shared__ __bf16 v_lds[kKvTile * kMaxD];
+  __shared__ __bf16 p_lds[4 * 16 * kKvTile];
+
+  const int bh = blockIdx.y;
+  const int b = bh / H;
+  const int qbase = blockIdx.x * kQTile;
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int l16 = lane & 15;      // within-frag row/col index
+  const int g4 = lane >> 4;       // 16-lane group id (0..3)
+
+  const long qk_base = (long)bh * Nq * D;
+  const long kv_base = (long)bh * Nk * D;
+  const int d8 = D / 8;           // 16B chunks per row
+
+  // ---- stage Q tile (cooperative, 16B chunks) ----
+  {
+    const int chunks = kQTile * d8;
+    for (int c = threadIdx.x; c < chunks; c += kBlockThreads) {
+      int row = c / d8, col8 = c % d8;
+      bf16x8_t val = {};
+      if (qbase + row < Nq) {
+        val = *reinterpret_cast<const bf16x8_t*>(q + qk_base + (long)(qbase + row) * D + col8 * 8);
+      }
+      *reinterpret_cast<bf16x8_t*>(
+          reinterpret_cast<char*>(q_lds) + swz<kSwizzle>(row, row * D * 2 + col8 * 16)) = val;
+    }
+  }
+  __syncthreads();
+
+  // ---- per-row online softmax state (4 rows per lane-group, replicated over 16 lanes) ----
+  float m_run[4], l_run[4];
+  f32x4 acc_o[kMaxD / 16];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = kNegInf; l_run[r] = 0.f; }
+#pragma unroll
+  for (int f = 0; f < kMaxD / 16; ++f) acc_o[f] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int n_kv_tiles = (Nk + kKvTile - 1) / kKvTile;
+  for (int kt = 0; kt < n_kv_tiles; ++kt) {
+    const int kv0 = kt * kKvTile;
+    // ---- stage K/V tile ----
+    {
+      const int chunks = kKvTile * d8;
+      for (int c = threadIdx.x; c < chunks; c += kBlockThreads) {
+        int row = c / d8, col8 = c % d8;
+        bf16x8_t kval = {}, vval = {};
+        if (kv0 + row < Nk) {
+          kval = *reinterpret_cast<const bf16x8_t*>(k + kv_base + (long)(kv0 + row) * D + col8 * 8);
+          vval = *reinterpret_cast<const bf16x8_t*>(v + kv_base + (long)(kv0 + row) * D + col8 * 8);
+        }
+        *reinterpret_cast<bf16x8_t*>(
+            reinterpret_cast<char*>(k_lds) + swz<kSwizzle>(row, row * D * 2 + col8 * 16)) = kval;
+        // V is read column-wise in PV (no 16B-row reads) -> keep linear
+        *reinterpret_cast<bf16x8_t*>(reinterpret_cast<char*>(v_lds) + row * D * 2 + col8 * 16) = vval;
+      }
+    }
+    __syncthreads();
+
+    // ---- S = scale * Q K^T (+mask), two 16x16 fragments along kv ----
+    float p_frag[2][4];  // [n16][reg] exp'd probabilities
+    float m_tile[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) m_tile[r] = kNegInf;
+
+    float s_frag[2][4];
+#pragma unroll
+    for (int n16 = 0; n16 < 2; ++n16) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+      for (int d0 = 0; d0 < D; d0 += 32) {
+        bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(q_lds) +
+            swz<kSwizzle>(wave * 16 + l16, (wave * 16 + l16) * D * 2 + (d0 + g4 * 8) * 2));
+        bf16x8_t bfr = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(k_lds) +
+            swz<kSwizzle>(n16 * 16 + l16, (n16 * 16 + l16) * D * 2 + (d0 + g4 * 8) * 2));
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, acc, 0, 0, 0);
+      }
+      const int kvcol = kv0 + n16 * 16 + l16;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float s = acc[r] * scale;
+        const int qrow = qbase + wave * 16 + g4 * 4 + r;
+        if (kvcol >= Nk || qrow >= Nq) {
+          s = kNegInf;
+        } else if (kHasMask) {
+          float mv = mask[((long)b * Nq + qrow) * Nk + kvcol];
+          s += mv;
+          if (s < kNegInf) s = kNegInf;
+        }
+        s_frag[n16][r] = s;
+        m_tile[r] = fmaxf(m_tile[r], s);
+      }
+    }
+#pragma unroll
+    for (int r = 0; r < 4; ++r) m_tile[r] = group16_reduce_max(m_tile[r]);
+
+    // ---- online softmax update ----
+    float factor[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float m_new = fmaxf(m_run[r], m_tile[r]);
+      factor[r] = (m_run[r] == kNegInf) ? 0.f : __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      float psum = 0.f;
+#pragma unroll
+      for (int n16 = 0; n16 < 2; ++n16) {
+        float p = (s_frag[n16][r] <= kNegInf) ? 0.f : __expf(s_frag[n16][r] - m_new);
+        p_frag[n16][r] = p;
+        psum += p;
+      }
+      l_run[r] = l_run[r] * factor[r] + group16_reduce_sum(psum);
+    }
+
+    // ---- write P to this wave's LDS region (16 rows x 32 cols) ----
+#pragma unroll
+    for (int n16 = 0; n16 < 2; ++n16) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        p_lds[wave * (16 * kKvTile) + (g4 * 4 + r) * kKvTile + n16 * 16 + l16] =
+            (__bf16)p_frag[n16][r];
+      }
+    }
+    __asm__ volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V, rescaling old accumulator ----
+    bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
+        p_lds + wave * (16 * kKvTile) + l16 * kKvTile + g4 * 8);
+#pragma unroll
+    for (int f = 0; f < kMaxD / 16; ++f) {
+      if (f * 16 >= D) break;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc_o[f][r] *= factor[r];
+      // B operand: V[kv=(g4*8+j)][dcol=f*16+l16] — strided column gather
+      bf16x8_t bv;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        bv[j] = v_lds[(g4 * 8 + j) * D + f * 16 + l16];
+      }
+      acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, acc_o[f], 0, 0, 0);
+    }
+    __syncthreads();  // before next tile overwrites K/V
+  }
+
+  // ---- normalize + write O, LSE ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = qbase + wave * 16 + g4 * 4 + r;
+    if (qrow >= Nq) continue;
+    const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    for (int f = 0; f < kMaxD / 16; ++f) {
+      if (f * 16 >= D) break;
+      o[qk_base + (long)qrow * D + f * 16 + l16] = (__bf16)(acc_o[f][r] * inv_l);
+    }
+    if (l16 == 0) {
+      lse[(long)bh * Nq + qrow] =
+          (l_run[r] > 0.f) ? (m_run[r] + __logf(l_run[r])) : INFINITY;
+    }
+  }
+}
+
+template <int kMaxD>
+void launch_attn_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+                     const c10::optional<at::Tensor>& mask, at::Tensor& o, at::Tensor& lse,
+                     int B, int H, int Nq, int Nk, int D, float scale, hipStream_t stream) {
+  dim3 grid(cdiv(Nq, kQTile), B * H);
+  dim3 block(kBlockThreads);
+  const bool has_mask = mask.has_value();
+  const float* mp = has_mask ? mask->data_ptr<float>() : nullptr;
+  auto args = [&](auto kern) {
+    hipLaunchKernelGGL(kern, grid, block, 0, stream,
+        (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(), (const __bf16*)v.data_ptr(),
+        mp, (__bf16*)o.data_ptr(), lse.data_ptr<float>(), B, H, Nq, Nk, D, scale);
+  };
+  // XOR swizzle is only bijective within a row when the row stride (D*2 bytes)
+  // is a power of two >= 128B; D=32/96 run unswizzled.
+  const bool swizzle = (D == 64 || D == 128);
+  if (has_mask && swizzle) args(attn_fwd_kernel<kMaxD, true, true>);
+  else if (has_mask) args(attn_fwd_kernel<kMaxD, true, false>);
+  else if (swizzle) args(attn_fwd_kernel<kMaxD, false, true>);
+  else args(attn_fwd_kernel<kMaxD, false, false>);
+  HIP_CHECK_LAST();
+}
+
+}  // namespace
+
+std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                      c10::optional<at::Tensor> mask, double scale) {
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention_fwd: bf16 only");
+  TORCH_CHECK(q.dim() == 4);
+  int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
+  int Nk = k.size(2);
+  TORCH_CHECK(D % 32 == 0 && D <= 128, "attention_fwd: head_dim must be multiple of 32, <=128");
+  if (mask.has_value()) {
+    TORCH_CHECK(mask->is_contiguous() && mask->scalar_type() == at::kFloat);
+    TORCH_CHECK(mask->size(0) == B && mask->size(2) == Nq && mask->size(3) == Nk);
+  }
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, H, Nq}, q.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  if (D <= 32) launch_attn_fwd<32>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);
+  else if (D <= 64) launch_attn_fwd<64>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);
+  else if (D <= 96) launch_attn_fwd<96>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);
+  else launch_attn_fwd<128>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);
+  return {o, lse};
+}

@@ -1,0 +1,47 @@
+// Python bindings for the timm_amd gfx950 HIP kernel extension.
+#include <torch/extension.h>
+
+// layernorm.hip
+std::vector<at::Tensor> layer_norm_fwd(at::Tensor x, at::Tensor w, at::Tensor b, double eps);
+std::vector<at::Tensor> layer_norm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                       at::Tensor mean, at::Tensor rstd);
+std::vector<at::Tensor> rms_norm_fwd(at::Tensor x, at::Tensor w, double eps);
+std::vector<at::Tensor> rms_norm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w, at::Tensor rstd);
+
+// elementwise.hip
+at::Tensor bias_act_fwd(at::Tensor x, at::Tensor b, long act_id);
+std::vector<at::Tensor> bias_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor b, long act_id);
+at::Tensor residual_scale_add_fwd(at::Tensor x, at::Tensor y,
+                                  c10::optional<at::Tensor> gamma,
+                                  c10::optional<at::Tensor> keep);
+std::vector<at::Tensor> residual_scale_add_bwd(at::Tensor dout,
+                                               c10::optional<at::Tensor> y,
+                                               c10::optional<at::Tensor> gamma,
+                                               c10::optional<at::Tensor> keep);
+
+// attention.hip
+std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                      c10::optional<at::Tensor> mask, double scale);
+
+// multi_tensor.hip
+void multi_tensor_adamw(
+    std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+    double lr, double beta1, double beta2, double eps, double wd, double bc1, double bc2);
+void multi_tensor_lerp(std::vector<at::Tensor> dsts, std::vector<at::Tensor> srcs, double weight);
+at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layer_norm_fwd", &layer_norm_fwd, "fused LayerNorm fwd (gfx950)");
+  m.def("layer_norm_bwd", &layer_norm_bwd, "fused LayerNorm bwd (gfx950)");
+  m.def("rms_norm_fwd", &rms_norm_fwd, "fused RMSNorm fwd (gfx950)");
+  m.def("rms_norm_bwd", &rms_norm_bwd, "fused RMSNorm bwd (gfx950)");
+  m.def("bias_act_fwd", &bias_act_fwd, "fused bias+activation fwd (gfx950)");
+  m.def("bias_act_bwd", &bias_act_bwd, "fused bias+activation bwd (gfx950)");
+  m.def("residual_scale_add_fwd", &residual_scale_add_fwd, "residual+LayerScale+DropPath fwd");
+  m.def("residual_scale_add_bwd", &residual_scale_add_bwd, "residual+LayerScale+DropPath bwd");
+  m.def("attention_fwd", &attention_fwd, "flash attention fwd (MFMA, gfx950)");
+  m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
+  m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
+  m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");
+}

@@ -53,7 +53,8 @@ def _act_eager(x, act: str):
 
 def bias_act(x: torch.Tensor, bias: Optional[torch.Tensor], act: str = 'gelu') -> torch.Tensor:
     """act(x + bias) over last dim; act in {gelu, gelu_tanh, silu, relu, identity, quick_gelu}."""
-    if x.is_cuda and _load_extension() is not None and bias is not None and act in _ACT_IDS:
+    if x.is_cuda and _load_extension() is not None and bias is not None and act in _ACT_IDS \
+            and x.dtype in (torch.bfloat16, torch.float16) and bias.dtype == x.dtype:
         return _BiasActFn.apply(x, bias, _ACT_IDS[act])
     if x.is_cuda:
         from . import use_hip
@@ -76,10 +77,11 @@ class _ResidualScaleAddFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         ext = _load_extension()
-        y, gamma, keep_mask = ctx.saved_tensors if len(ctx.saved_tensors) == 3 else (None, None, None)
+        y, gamma, keep_mask = ctx.saved_tensors
         dout = dout.contiguous()
         dx = dout
-        dy, dgamma = ext.residual_scale_add_bwd(dout, y, gamma, keep_mask)
+        dy, dgamma = ext.residual_scale_add_bwd(
+            dout, y if gamma is not None else None, gamma, keep_mask)
         return dx, dy, dgamma, None
 
 
@@ -99,7 +101,8 @@ def residual_scale_add(
         if scale_by_keep and keep_prob > 0.:
             keep_mask = keep_mask / keep_prob
     if x.is_cuda and _load_extension() is not None and x.is_contiguous() and y.is_contiguous() \
-            and (gamma is None or gamma.ndim == 1):
+            and x.dtype in (torch.bfloat16, torch.float16) and y.dtype == x.dtype \
+            and (gamma is None or (gamma.ndim == 1 and gamma.dtype == x.dtype)):
         return _ResidualScaleAddFn.apply(x, y, gamma, keep_mask)
     if x.is_cuda:
         from . import use_hip

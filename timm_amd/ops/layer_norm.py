@@ -57,6 +57,7 @@ def layer_norm(
 ) -> torch.Tensor:
     """LayerNorm over the last dim. HIP fused kernel on device, F.layer_norm on CPU."""
     if x.is_cuda and _load_extension() is not None and weight is not None \
+            and weight.dtype == x.dtype and (bias is None or bias.dtype == x.dtype) \
             and len(normalized_shape) == 1 and x.shape[-1] == normalized_shape[0]:
         if bias is None:
             bias = torch.zeros_like(weight)
@@ -74,6 +75,7 @@ def rms_norm(
         eps: float = 1e-6,
 ) -> torch.Tensor:
     if x.is_cuda and _load_extension() is not None and weight is not None \
+            and weight.dtype == x.dtype \
             and len(normalized_shape) == 1 and x.shape[-1] == normalized_shape[0]:
         return _RmsNormFn.apply(x, weight, eps)
     if x.is_cuda:
