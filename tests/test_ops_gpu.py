@@ -20,7 +20,9 @@ def _ext():
 
 def rel_err(a, b):
     a, b = a.float(), b.float()
-    return ((a - b).norm() / b.norm().clamp(min=1e-8)).item()
+    # denominator floored at 1e-3 so a near-zero reference (e.g. dq when Nq==1,
+    # where softmax is constant) doesn't blow up numeric noise
+    return ((a - b).norm() / b.norm().clamp(min=1e-3)).item()
 
 
 @pytest.mark.parametrize('shape', [(8, 197, 768), (4, 50, 1024), (2, 7, 640)])

@@ -45,7 +45,16 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, attn_mask, scale):
         ext = _load_extension()
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        # kernel reads strided q/k/v (e.g. views into the packed qkv tensor)
+        # as long as head_dim is contiguous and 16B-alignment holds
+        def _ok(t):
+            return t.stride(-1) == 1 and all(s % 8 == 0 or s == 0 for s in t.stride()[:3])
+        if not _ok(q):
+            q = q.contiguous()
+        if not _ok(k):
+            k = k.contiguous()
+        if not _ok(v):
+            v = v.contiguous()
         if attn_mask is not None:
             attn_mask = attn_mask.contiguous().to(torch.float32)
         o, lse = ext.attention_fwd(q, k, v, attn_mask, scale)
@@ -58,19 +67,23 @@ class _FlashAttnFn(torch.autograd.Function):
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         scale = ctx.scale
-        # exact flash backward via GEMMs + saved LSE (recompute P)
-        qf, kf, vf, of, dof = q.float(), k.float(), v.float(), o.float(), do.float()
-        s = (qf @ kf.transpose(-2, -1)) * scale
+        do = do.contiguous()
+        # exact flash backward: recompute P from saved LSE; all GEMMs stay bf16
+        # so they run on the MFMA path (hipBLASLt accumulates fp32 internally);
+        # only the softmax recompute elementwise is fp32.
+        s = q @ k.transpose(-2, -1)  # bf16 GEMM, fp32 accum
+        s = s.float() * scale
         if ctx.attn_mask is not None:
-            s = s + ctx.attn_mask
-        p = torch.exp(s - lse.unsqueeze(-1))  # [B,H,Nq,Nk], rows sum to 1
-        dv = p.transpose(-2, -1) @ dof
-        dp = dof @ vf.transpose(-2, -1)
-        delta = (dof * of).sum(-1, keepdim=True)
-        ds = p * (dp - delta) * scale
-        dq = ds @ kf
-        dk = ds.transpose(-2, -1) @ qf
-        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None, None
+            s += ctx.attn_mask
+        p = torch.exp(s - lse.unsqueeze(-1))  # rows sum to 1 (or 0 for dead rows)
+        p16 = p.to(q.dtype)
+        dv = p16.transpose(-2, -1) @ do
+        dp = do @ v.transpose(-2, -1)  # bf16 GEMM
+        delta = (do * o).float().sum(-1, keepdim=True)
+        ds = (p * (dp.float() - delta)).mul_(scale).to(q.dtype)
+        dq = ds @ k
+        dk = ds.transpose(-2, -1) @ q
+        return dq, dk, dv, None, None
 
 
 def flash_attention(

@@ -37,13 +37,16 @@ __device__ __forceinline__ int swz(int row, int byte_off) {
 template <int kMaxD, bool kHasMask, bool kSwizzle>
 __global__ __launch_bounds__(kBlockThreads)
 void attn_fwd_kernel(
-    const __bf16* __restrict__ q,     // [B,H,Nq,D]
-    const __bf16* __restrict__ k,     // [B,H,Nk,D]
-    const __bf16* __restrict__ v,     // [B,H,Nk,D]
+    const __bf16* __restrict__ q,     // [B,H,Nq,D] via strides (last dim contiguous)
+    const __bf16* __restrict__ k,
+    const __bf16* __restrict__ v,
     const float* __restrict__ mask,   // [B,1,Nq,Nk] or null
-    __bf16* __restrict__ o,           // [B,H,Nq,D]
+    __bf16* __restrict__ o,           // written [B,Nq,H,D] (BNHD) so proj reshape is free
     float* __restrict__ lse,          // [B,H,Nq]
-    int B, int H, int Nq, int Nk, int D, float scale) {
+    int B, int H, int Nq, int Nk, int D, float scale,
+    long q_sb, long q_sh, long q_sn,
+    long k_sb, long k_sh, long k_sn,
+    long v_sb, long v_sh, long v_sn) {
   // LDS: Q[64][D] | K[32][D] | V[32][D] | P[4][16][32]
   __shared__ __bf16 q_lds[kQTile * kMaxD];
   __shared__ __bf16 k_lds[kKvTile * kMaxD];
@@ -52,14 +55,16 @@ void attn_fwd_kernel(
 
   const int bh = blockIdx.y;
   const int b = bh / H;
+  const int h = bh % H;
   const int qbase = blockIdx.x * kQTile;
   const int wave = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x % WAVE_SIZE;
   const int l16 = lane & 15;      // within-frag row/col index
   const int g4 = lane >> 4;       // 16-lane group id (0..3)
 
-  const long qk_base = (long)bh * Nq * D;
-  const long kv_base = (long)bh * Nk * D;
+  const __bf16* q_bh = q + (long)b * q_sb + (long)h * q_sh;
+  const __bf16* k_bh = k + (long)b * k_sb + (long)h * k_sh;
+  const __bf16* v_bh = v + (long)b * v_sb + (long)h * v_sh;
   const int d8 = D / 8;           // 16B chunks per row
 
   // ---- stage Q tile (cooperative, 16B chunks) ----
@@ -69,7 +74,7 @@ void attn_fwd_kernel(
       int row = c / d8, col8 = c % d8;
       bf16x8_t val = {};
       if (qbase + row < Nq) {
-        val = *reinterpret_cast<const bf16x8_t*>(q + qk_base + (long)(qbase + row) * D + col8 * 8);
+        val = *reinterpret_cast<const bf16x8_t*>(q_bh + (long)(qbase + row) * q_sn + col8 * 8);
       }
       *reinterpret_cast<bf16x8_t*>(
           reinterpret_cast<char*>(q_lds) + swz<kSwizzle>(row, row * D * 2 + col8 * 16)) = val;
@@ -95,8 +100,8 @@ void attn_fwd_kernel(
         int row = c / d8, col8 = c % d8;
         bf16x8_t kval = {}, vval = {};
         if (kv0 + row < Nk) {
-          kval = *reinterpret_cast<const bf16x8_t*>(k + kv_base + (long)(kv0 + row) * D + col8 * 8);
-          vval = *reinterpret_cast<const bf16x8_t*>(v + kv_base + (long)(kv0 + row) * D + col8 * 8);
+          kval = *reinterpret_cast<const bf16x8_t*>(k_bh + (long)(kv0 + row) * k_sn + col8 * 8);
+          vval = *reinterpret_cast<const bf16x8_t*>(v_bh + (long)(kv0 + row) * v_sn + col8 * 8);
         }
         *reinterpret_cast<bf16x8_t*>(
             reinterpret_cast<char*>(k_lds) + swz<kSwizzle>(row, row * D * 2 + col8 * 16)) = kval;
@@ -191,7 +196,8 @@ void attn_fwd_kernel(
     __syncthreads();  // before next tile overwrites K/V
   }
 
-  // ---- normalize + write O, LSE ----
+  // ---- normalize + write O (BNHD layout), LSE ----
+  __bf16* o_bh = o + ((long)b * Nq) * (H * D) + (long)h * D;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow = qbase + wave * 16 + g4 * 4 + r;
@@ -199,7 +205,7 @@ void attn_fwd_kernel(
     const float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
     for (int f = 0; f < kMaxD / 16; ++f) {
       if (f * 16 >= D) break;
-      o[qk_base + (long)qrow * D + f * 16 + l16] = (__bf16)(acc_o[f][r] * inv_l);
+      o_bh[(long)qrow * (H * D) + f * 16 + l16] = (__bf16)(acc_o[f][r] * inv_l);
     }
     if (l16 == 0) {
       lse[(long)bh * Nq + qrow] =
@@ -219,7 +225,10 @@ void launch_attn_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
   auto args = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, block, 0, stream,
         (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(), (const __bf16*)v.data_ptr(),
-        mp, (__bf16*)o.data_ptr(), lse.data_ptr<float>(), B, H, Nq, Nk, D, scale);
+        mp, (__bf16*)o.data_ptr(), lse.data_ptr<float>(), B, H, Nq, Nk, D, scale,
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2));
   };
   // XOR swizzle is only bijective within a row when the row stride (D*2 bytes)
   // is a power of two >= 128B; D=32/96 run unswizzled.
@@ -235,7 +244,8 @@ void launch_attn_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
 
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       c10::optional<at::Tensor> mask, double scale) {
-  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() && v.is_contiguous());
+  TORCH_CHECK(q.is_cuda() && q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,
+              "attention_fwd: innermost (head_dim) stride must be 1");
   TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attention_fwd: bf16 only");
   TORCH_CHECK(q.dim() == 4);
   int B = q.size(0), H = q.size(1), Nq = q.size(2), D = q.size(3);
@@ -245,7 +255,10 @@ std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
     TORCH_CHECK(mask->is_contiguous() && mask->scalar_type() == at::kFloat);
     TORCH_CHECK(mask->size(0) == B && mask->size(2) == Nq && mask->size(3) == Nk);
   }
-  auto o = at::empty_like(q);
+  // O allocated [B, Nq, H, D] and returned as a permuted [B,H,Nq,D] view so
+  // the caller's transpose(1,2).reshape(B,N,C) is a zero-copy reshape.
+  auto o_bnhd = at::empty({B, Nq, H, D}, q.options());
+  auto o = o_bnhd.permute({0, 2, 1, 3});
   auto lse = at::empty({B, H, Nq}, q.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   if (D <= 32) launch_attn_fwd<32>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);

@@ -85,6 +85,7 @@ __global__ void bias_act_fwd_kernel(
   }
 }
 
+// Slow path (cols not 16B-divisible): per-element LDS atomics.
 template <typename T>
 __global__ void bias_act_bwd_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ b,
@@ -103,6 +104,66 @@ __global__ void bias_act_bwd_kernel(
   }
   __syncthreads();
   for (int i = threadIdx.x; i < cols; i += blockDim.x) atomicAdd(&db[i], db_s[i]);
+}
+
+// Fast path: wave-per-row, 16B vector loads, per-lane register dB accumulation
+// (fixed lane->column mapping across rows), one global atomic/col/wave.
+template <typename T, int kChunks>
+__global__ __launch_bounds__(256)
+void bias_act_bwd_fast_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ b,
+    T* __restrict__ dx, float* __restrict__ db, long rows, int cols, int act) {
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int nvec = cols / 8;
+  constexpr int kWaves = 256 / WAVE_SIZE;
+
+  float db_acc[kChunks][8];
+  float b_reg[kChunks][8];
+#pragma unroll
+  for (int c = 0; c < kChunks; ++c) {
+    int i = lane + c * WAVE_SIZE;
+    if (i < nvec) {
+      short8 bv = reinterpret_cast<const short8*>(b)[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        b_reg[c][j] = Elem<T>::to_f32(reinterpret_cast<const T*>(&bv)[j]);
+        db_acc[c][j] = 0.f;
+      }
+    }
+  }
+
+  for (long row = blockIdx.x * kWaves + wave; row < rows; row += (long)gridDim.x * kWaves) {
+    const short8* dyv = reinterpret_cast<const short8*>(dy + row * cols);
+    const short8* xv = reinterpret_cast<const short8*>(x + row * cols);
+    short8* dxv = reinterpret_cast<short8*>(dx + row * cols);
+#pragma unroll
+    for (int c = 0; c < kChunks; ++c) {
+      int i = lane + c * WAVE_SIZE;
+      if (i < nvec) {
+        short8 dv = dyv[i];
+        short8 xv8 = xv[i];
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = Elem<T>::to_f32(reinterpret_cast<const T*>(&xv8)[j]) + b_reg[c][j];
+          float g = Elem<T>::to_f32(reinterpret_cast<const T*>(&dv)[j]) * act_bwd(f, act);
+          reinterpret_cast<T*>(&out)[j] = Elem<T>::from_f32(g);
+          db_acc[c][j] += g;
+        }
+        dxv[i] = out;
+      }
+    }
+  }
+
+#pragma unroll
+  for (int c = 0; c < kChunks; ++c) {
+    int i = lane + c * WAVE_SIZE;
+    if (i < nvec) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&db[i * 8 + j], db_acc[c][j]);
+    }
+  }
 }
 
 // ---- residual_scale_add ----
@@ -203,13 +264,30 @@ std::vector<at::Tensor> bias_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor b, 
   auto db = at::zeros({cols}, x.options().dtype(at::kFloat));
   long n = x.numel();
   auto stream = at::hip::getCurrentHIPStream();
-  int blocks = std::min((long)1024, (n + kBlock - 1) / kBlock);
   AT_DISPATCH_REDUCED_FLOATING_TYPES(x.scalar_type(), "bias_act_bwd", [&] {
     using T = typename ToHip<scalar_t>::type;
-    hipLaunchKernelGGL((bias_act_bwd_kernel<T>), dim3(blocks), dim3(kBlock),
-        cols * sizeof(float), stream,
-        (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)b.data_ptr(),
-        (T*)dx.data_ptr(), db.data_ptr<float>(), n, cols, (int)act_id);
+    long rows = n / cols;
+    int chunks = (cols / 8 + 63) / 64;
+    if (cols % 8 == 0 && chunks <= 8) {
+      int blocks = (int)std::min((long)2048, (rows + 3) / 4);
+      auto go = [&](auto tag) {
+        hipLaunchKernelGGL((bias_act_bwd_fast_kernel<T, decltype(tag)::value>),
+            dim3(blocks), dim3(256), 0, stream,
+            (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)b.data_ptr(),
+            (T*)dx.data_ptr(), db.data_ptr<float>(), rows, cols, (int)act_id);
+      };
+      if (chunks <= 1) go(std::integral_constant<int, 1>{});
+      else if (chunks <= 2) go(std::integral_constant<int, 2>{});
+      else if (chunks <= 4) go(std::integral_constant<int, 4>{});
+      else if (chunks <= 6) go(std::integral_constant<int, 6>{});
+      else go(std::integral_constant<int, 8>{});
+    } else {
+      int blocks = std::min((long)1024, (n + kBlock - 1) / kBlock);
+      hipLaunchKernelGGL((bias_act_bwd_kernel<T>), dim3(blocks), dim3(kBlock),
+          cols * sizeof(float), stream,
+          (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)b.data_ptr(),
+          (T*)dx.data_ptr(), db.data_ptr<float>(), n, cols, (int)act_id);
+    }
   });
   HIP_CHECK_LAST();
   return {dx, db.to(b.scalar_type())};

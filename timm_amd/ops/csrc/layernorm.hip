@@ -171,6 +171,113 @@ __global__ void norm_bwd_kernel(
   }
 }
 
+// Fast backward: vectorized 16B loads, per-lane register accumulation of
+// dW/dB (lane->chunk mapping is fixed across rows), one global fp32 atomic
+// per column per wave at the end.  Requires cols % 8 == 0 and
+// cols <= kChunks*512.
+template <typename T, bool kRms, int kChunks>
+__global__ __launch_bounds__(kBlock)
+void norm_bwd_fast_kernel(
+    const T* __restrict__ dy,
+    const T* __restrict__ x,
+    const T* __restrict__ w,
+    const float* __restrict__ mean_in,
+    const float* __restrict__ rstd_in,
+    T* __restrict__ dx,
+    float* __restrict__ dw,
+    float* __restrict__ db,
+    int rows, int cols) {
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  const int nvec = cols / 8;
+
+  float dw_acc[kChunks][8];
+  float db_acc[kChunks][8];  // folded away for kRms (all uses guarded by constexpr cond)
+#pragma unroll
+  for (int c = 0; c < kChunks; ++c)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      dw_acc[c][j] = 0.f;
+      if (!kRms) db_acc[c][j] = 0.f;
+    }
+
+  // preload w into registers (shared across all rows)
+  float w_reg[kChunks][8];
+#pragma unroll
+  for (int c = 0; c < kChunks; ++c) {
+    int i = lane + c * WAVE_SIZE;
+    if (i < nvec) {
+      short8 wv = reinterpret_cast<const short8*>(w)[i];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) w_reg[c][j] = Elem<T>::to_f32(reinterpret_cast<const T*>(&wv)[j]);
+    }
+  }
+
+  for (int row = blockIdx.x * kWavesPerBlock + wave; row < rows;
+       row += gridDim.x * kWavesPerBlock) {
+    const short8* dyv = reinterpret_cast<const short8*>(dy + (long)row * cols);
+    const short8* xv = reinterpret_cast<const short8*>(x + (long)row * cols);
+    const float mean = kRms ? 0.f : mean_in[row];
+    const float rstd = rstd_in[row];
+
+    float dy_reg[kChunks][8], xh_reg[kChunks][8];
+    float c1 = 0.f, c2 = 0.f;
+#pragma unroll
+    for (int c = 0; c < kChunks; ++c) {
+      int i = lane + c * WAVE_SIZE;
+      if (i < nvec) {
+        short8 dv = dyv[i];
+        short8 xv8 = xv[i];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = Elem<T>::to_f32(reinterpret_cast<const T*>(&dv)[j]);
+          float xhat = (Elem<T>::to_f32(reinterpret_cast<const T*>(&xv8)[j]) - mean) * rstd;
+          dy_reg[c][j] = g;
+          xh_reg[c][j] = xhat;
+          float gw = g * w_reg[c][j];
+          c1 += gw * xhat;
+          c2 += gw;
+        }
+      }
+    }
+    c1 = wave_reduce_sum(c1) / cols;
+    c2 = wave_reduce_sum(c2) / cols;
+
+    short8* dxv = reinterpret_cast<short8*>(dx + (long)row * cols);
+#pragma unroll
+    for (int c = 0; c < kChunks; ++c) {
+      int i = lane + c * WAVE_SIZE;
+      if (i < nvec) {
+        short8 out;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float g = dy_reg[c][j];
+          float xhat = xh_reg[c][j];
+          float gw = g * w_reg[c][j];
+          float val = kRms ? (gw - xhat * c1) : (gw - c2 - xhat * c1);
+          reinterpret_cast<T*>(&out)[j] = Elem<T>::from_f32(rstd * val);
+          dw_acc[c][j] += g * xhat;
+          if (!kRms) db_acc[c][j] += g;
+        }
+        dxv[i] = out;
+      }
+    }
+  }
+
+  // flush per-lane accumulators: one global atomic per column per wave
+#pragma unroll
+  for (int c = 0; c < kChunks; ++c) {
+    int i = lane + c * WAVE_SIZE;
+    if (i < nvec) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        atomicAdd(&dw[i * 8 + j], dw_acc[c][j]);
+        if (!kRms) atomicAdd(&db[i * 8 + j], db_acc[c][j]);
+      }
+    }
+  }
+}
+
 template <typename T>
 void launch_norm_fwd(const at::Tensor& x, const at::Tensor& w, const c10::optional<at::Tensor>& b,
                      at::Tensor& y, at::Tensor& mean, at::Tensor& rstd, double eps, bool rms,
@@ -190,11 +297,48 @@ void launch_norm_fwd(const at::Tensor& x, const at::Tensor& w, const c10::option
   HIP_CHECK_LAST();
 }
 
+template <typename T, bool kRms>
+bool try_launch_norm_bwd_fast(const T* dy, const T* x, const T* w,
+                              const float* mean, const float* rstd,
+                              T* dx, float* dw, float* db,
+                              int rows, int cols, hipStream_t stream) {
+  if (cols % 8 != 0 || sizeof(T) != 2) return false;
+  int blocks = std::min(cdiv(rows, kWavesPerBlock), 2048);
+  int chunks = cdiv(cols / 8, WAVE_SIZE);
+  auto go = [&](auto tag) {
+    constexpr int kChunks = decltype(tag)::value;
+    hipLaunchKernelGGL((norm_bwd_fast_kernel<T, kRms, kChunks>),
+        dim3(blocks), dim3(kBlock), 0, stream,
+        dy, x, w, mean, rstd, dx, dw, db, rows, cols);
+  };
+  if (chunks <= 1) go(std::integral_constant<int, 1>{});
+  else if (chunks <= 2) go(std::integral_constant<int, 2>{});
+  else if (chunks <= 4) go(std::integral_constant<int, 4>{});
+  else if (chunks <= 8 && kRms) go(std::integral_constant<int, 8>{});
+  else if (chunks <= 6) go(std::integral_constant<int, 6>{});
+  else return false;
+  HIP_CHECK_LAST();
+  return true;
+}
+
 template <typename T>
 void launch_norm_bwd(const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& mean, const at::Tensor& rstd,
                      at::Tensor& dx, at::Tensor& dw, at::Tensor& db, bool rms,
                      int rows, int cols, hipStream_t stream) {
+  if (rms) {
+    if (try_launch_norm_bwd_fast<T, true>(
+            (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+            nullptr, rstd.data_ptr<float>(), (T*)dx.data_ptr(),
+            dw.data_ptr<float>(), nullptr, rows, cols, stream))
+      return;
+  } else {
+    if (try_launch_norm_bwd_fast<T, false>(
+            (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)w.data_ptr(),
+            mean->data_ptr<float>(), rstd.data_ptr<float>(), (T*)dx.data_ptr(),
+            dw.data_ptr<float>(), db.data_ptr<float>(), rows, cols, stream))
+      return;
+  }
   int blocks = std::min(cdiv(rows, kWavesPerBlock), 1024);
   size_t smem = (rms ? 1 : 2) * cols * sizeof(float);
   if (rms) {
