@@ -167,8 +167,14 @@ def test_attention_fwd_bwd(shape):
     o_ref.backward(do.float())
 
     assert rel_err(o, o_ref) < 2e-2, f'fwd err {rel_err(o, o_ref)}'
-    assert rel_err(q1.grad, q2.grad) < 4e-2
-    assert rel_err(k1.grad, k2.grad) < 4e-2
+    if N == 1:
+        # softmax over one key is constant -> dq = dk = 0 analytically; only
+        # check bf16 rounding noise stays small in RMS terms
+        assert q1.grad.float().pow(2).mean().sqrt().item() < 0.05
+        assert k1.grad.float().pow(2).mean().sqrt().item() < 0.05
+    else:
+        assert rel_err(q1.grad, q2.grad) < 4e-2
+        assert rel_err(k1.grad, k2.grad) < 4e-2
     assert rel_err(v1.grad, v2.grad) < 4e-2
 
 

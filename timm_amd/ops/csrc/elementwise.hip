@@ -156,14 +156,20 @@ void bias_act_bwd_fast_kernel(
     }
   }
 
+  // block-level LDS reduction then one global atomic per col per block
+  extern __shared__ float db_red[];
+  for (int i = threadIdx.x; i < cols; i += 256) db_red[i] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int c = 0; c < kChunks; ++c) {
     int i = lane + c * WAVE_SIZE;
     if (i < nvec) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) atomicAdd(&db[i * 8 + j], db_acc[c][j]);
+      for (int j = 0; j < 8; ++j) atomicAdd(&db_red[i * 8 + j], db_acc[c][j]);
     }
   }
+  __syncthreads();
+  for (int i = threadIdx.x; i < cols; i += 256) atomicAdd(&db[i], db_red[i]);
 }
 
 // ---- residual_scale_add ----
@@ -269,10 +275,10 @@ std::vector<at::Tensor> bias_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor b, 
     long rows = n / cols;
     int chunks = (cols / 8 + 63) / 64;
     if (cols % 8 == 0 && chunks <= 8) {
-      int blocks = (int)std::min((long)2048, (rows + 3) / 4);
+      int blocks = (int)std::min((long)320, (rows + 3) / 4);
       auto go = [&](auto tag) {
         hipLaunchKernelGGL((bias_act_bwd_fast_kernel<T, decltype(tag)::value>),
-            dim3(blocks), dim3(256), 0, stream,
+            dim3(blocks), dim3(256), cols * sizeof(float), stream,
             (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (const T*)b.data_ptr(),
             (T*)dx.data_ptr(), db.data_ptr<float>(), rows, cols, (int)act_id);
       };

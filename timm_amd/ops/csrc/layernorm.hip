@@ -264,17 +264,32 @@ void norm_bwd_fast_kernel(
     }
   }
 
-  // flush per-lane accumulators: one global atomic per column per wave
+  // flush: block-level LDS reduction (4-way wave contention only), then ONE
+  // global atomic per column per block — global atomic chains are gridDim
+  // deep instead of n_waves deep (same-address fp32 atomics serialize in L2).
+  extern __shared__ float red_s[];  // [cols] dw (+ [cols] db for LN)
+  float* dw_s = red_s;
+  float* db_s = red_s + cols;
+  for (int i = threadIdx.x; i < cols; i += kBlock) {
+    dw_s[i] = 0.f;
+    if (!kRms) db_s[i] = 0.f;
+  }
+  __syncthreads();
 #pragma unroll
   for (int c = 0; c < kChunks; ++c) {
     int i = lane + c * WAVE_SIZE;
     if (i < nvec) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        atomicAdd(&dw[i * 8 + j], dw_acc[c][j]);
-        if (!kRms) atomicAdd(&db[i * 8 + j], db_acc[c][j]);
+        atomicAdd(&dw_s[i * 8 + j], dw_acc[c][j]);
+        if (!kRms) atomicAdd(&db_s[i * 8 + j], db_acc[c][j]);
       }
     }
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < cols; i += kBlock) {
+    atomicAdd(&dw[i], dw_s[i]);
+    if (!kRms) atomicAdd(&db[i], db_s[i]);
   }
 }
 
@@ -303,12 +318,15 @@ bool try_launch_norm_bwd_fast(const T* dy, const T* x, const T* w,
                               T* dx, float* dw, float* db,
                               int rows, int cols, hipStream_t stream) {
   if (cols % 8 != 0 || sizeof(T) != 2) return false;
-  int blocks = std::min(cdiv(rows, kWavesPerBlock), 2048);
+  // 320 blocks ≈ 1.25 waves/CU keeps the chip fed for a bandwidth-bound
+  // kernel while bounding the global-atomic chain depth for dW/dB
+  int blocks = std::min(cdiv(rows, kWavesPerBlock), 320);
   int chunks = cdiv(cols / 8, WAVE_SIZE);
+  size_t smem = (kRms ? 1 : 2) * cols * sizeof(float);
   auto go = [&](auto tag) {
     constexpr int kChunks = decltype(tag)::value;
     hipLaunchKernelGGL((norm_bwd_fast_kernel<T, kRms, kChunks>),
-        dim3(blocks), dim3(kBlock), 0, stream,
+        dim3(blocks), dim3(kBlock), smem, stream,
         dy, x, w, mean, rstd, dx, dw, db, rows, cols);
   };
   if (chunks <= 1) go(std::integral_constant<int, 1>{});
