@@ -66,21 +66,18 @@ class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
+        ext = _load_extension()
         scale = ctx.scale
         do = do.contiguous()
-        # exact flash backward: recompute P from saved LSE; all GEMMs stay bf16
-        # so they run on the MFMA path (hipBLASLt accumulates fp32 internally);
-        # only the softmax recompute elementwise is fp32.
-        s = q @ k.transpose(-2, -1)  # bf16 GEMM, fp32 accum
-        s = s.float() * scale
-        if ctx.attn_mask is not None:
-            s += ctx.attn_mask
-        p = torch.exp(s - lse.unsqueeze(-1))  # rows sum to 1 (or 0 for dead rows)
-        p16 = p.to(q.dtype)
-        dv = p16.transpose(-2, -1) @ do
-        dp = do @ v.transpose(-2, -1)  # bf16 GEMM
-        delta = (do * o).float().sum(-1, keepdim=True)
-        ds = (p * (dp.float() - delta)).mul_(scale).to(q.dtype)
+        # exact flash backward: recompute P from saved LSE. GEMMs stay bf16
+        # (MFMA via hipBLASLt, fp32 internal accum); the softmax-recompute
+        # elementwise runs in two fused HIP kernels (attn_bwd_softmax/_ds).
+        s = (q @ k.transpose(-2, -1)).contiguous()  # bf16 GEMM
+        p = ext.attn_bwd_softmax(s, lse, ctx.attn_mask, scale)  # exp(s*scale+mask-lse)
+        dv = p.transpose(-2, -1) @ do
+        dp = (do @ v.transpose(-2, -1)).contiguous()  # bf16 GEMM
+        delta = (do * o).float().sum(-1)  # [B,H,Nq] fp32
+        ds = ext.attn_bwd_ds(p, dp, delta, scale)
         dq = ds @ k
         dk = ds.transpose(-2, -1) @ q
         return dq, dk, dv, None, None

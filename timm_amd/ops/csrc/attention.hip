@@ -242,6 +242,125 @@ void launch_attn_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
 
 }  // namespace
 
+namespace {
+
+// ---- backward helper kernels: fuse the softmax recompute elementwise ----
+// One wave per row (lse/mask shared per row); vectorized when Nk % 8 == 0,
+// scalar tail otherwise.
+// p = exp(s * scale + mask - lse[row])   (s bf16 in, p bf16 out, fp32 math)
+template <bool kHasMask>
+__global__ __launch_bounds__(256)
+void attn_bwd_softmax_kernel(
+    const __bf16* __restrict__ s,    // [B,H,Nq,Nk]
+    const float* __restrict__ lse,   // [B,H,Nq]
+    const float* __restrict__ mask,  // [B,1,Nq,Nk] or null
+    __bf16* __restrict__ p,
+    long rows, int Nk, int Nq, int H, float scale) {
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  constexpr int kWaves = 256 / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * kWaves + wave; row < rows;
+       row += (long)gridDim.x * kWaves) {
+    const float l = lse[row];
+    const __bf16* sr = s + row * Nk;
+    __bf16* pr = p + row * Nk;
+    const float* mrow = nullptr;
+    if (kHasMask) {
+      long q = row % Nq;
+      long b = row / ((long)H * Nq);
+      mrow = mask + (b * Nq + q) * Nk;
+    }
+    int nvec = (Nk % 8 == 0) ? Nk / 8 : 0;
+    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+      bf16x8_t sv = *reinterpret_cast<const bf16x8_t*>(sr + i * 8);
+      bf16x8_t out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)sv[j] * scale;
+        if (kHasMask) f += mrow[i * 8 + j];
+        out[j] = (__bf16)__expf(f - l);
+      }
+      *reinterpret_cast<bf16x8_t*>(pr + i * 8) = out;
+    }
+    for (int i = nvec * 8 + lane; i < Nk; i += WAVE_SIZE) {
+      float f = (float)sr[i] * scale;
+      if (kHasMask) f += mrow[i];
+      pr[i] = (__bf16)__expf(f - l);
+    }
+  }
+}
+
+// ds = p * (dp - delta[row]) * scale   (all bf16 except delta fp32)
+__global__ __launch_bounds__(256)
+void attn_bwd_ds_kernel(
+    const __bf16* __restrict__ p,
+    const __bf16* __restrict__ dp,
+    const float* __restrict__ delta,  // [B,H,Nq]
+    __bf16* __restrict__ ds,
+    long rows, int Nk, float scale) {
+  const int wave = threadIdx.x / WAVE_SIZE;
+  const int lane = threadIdx.x % WAVE_SIZE;
+  constexpr int kWaves = 256 / WAVE_SIZE;
+  for (long row = (long)blockIdx.x * kWaves + wave; row < rows;
+       row += (long)gridDim.x * kWaves) {
+    const float d = delta[row];
+    const __bf16* pr = p + row * Nk;
+    const __bf16* dpr = dp + row * Nk;
+    __bf16* dsr = ds + row * Nk;
+    int nvec = (Nk % 8 == 0) ? Nk / 8 : 0;
+    for (int i = lane; i < nvec; i += WAVE_SIZE) {
+      bf16x8_t pv = *reinterpret_cast<const bf16x8_t*>(pr + i * 8);
+      bf16x8_t dpv = *reinterpret_cast<const bf16x8_t*>(dpr + i * 8);
+      bf16x8_t out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        out[j] = (__bf16)(((float)pv[j] * ((float)dpv[j] - d)) * scale);
+      }
+      *reinterpret_cast<bf16x8_t*>(dsr + i * 8) = out;
+    }
+    for (int i = nvec * 8 + lane; i < Nk; i += WAVE_SIZE) {
+      dsr[i] = (__bf16)(((float)pr[i] * ((float)dpr[i] - d)) * scale);
+    }
+  }
+}
+
+}  // namespace
+
+at::Tensor attn_bwd_softmax(at::Tensor s, at::Tensor lse, c10::optional<at::Tensor> mask,
+                            double scale) {
+  TORCH_CHECK(s.is_cuda() && s.is_contiguous() && s.scalar_type() == at::kBFloat16);
+  int H = s.size(1), Nq = s.size(2), Nk = s.size(3);
+  auto p = at::empty_like(s);
+  long rows = s.numel() / Nk;
+  auto stream = at::hip::getCurrentHIPStream();
+  int blocks = (int)std::min((long)4096, (rows + 3) / 4);
+  if (mask.has_value()) {
+    hipLaunchKernelGGL((attn_bwd_softmax_kernel<true>), dim3(blocks), dim3(256), 0, stream,
+        (const __bf16*)s.data_ptr(), lse.data_ptr<float>(), mask->data_ptr<float>(),
+        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale);
+  } else {
+    hipLaunchKernelGGL((attn_bwd_softmax_kernel<false>), dim3(blocks), dim3(256), 0, stream,
+        (const __bf16*)s.data_ptr(), lse.data_ptr<float>(), nullptr,
+        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale);
+  }
+  HIP_CHECK_LAST();
+  return p;
+}
+
+at::Tensor attn_bwd_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale) {
+  TORCH_CHECK(p.is_cuda() && p.is_contiguous() && dp.is_contiguous());
+  int Nk = p.size(3);
+  auto ds = at::empty_like(p);
+  long rows = p.numel() / Nk;
+  auto stream = at::hip::getCurrentHIPStream();
+  int blocks = (int)std::min((long)4096, (rows + 3) / 4);
+  hipLaunchKernelGGL(attn_bwd_ds_kernel, dim3(blocks), dim3(256), 0, stream,
+      (const __bf16*)p.data_ptr(), (const __bf16*)dp.data_ptr(), delta.data_ptr<float>(),
+      (__bf16*)ds.data_ptr(), rows, Nk, (float)scale);
+  HIP_CHECK_LAST();
+  return ds;
+}
+
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       c10::optional<at::Tensor> mask, double scale) {
   TORCH_CHECK(q.is_cuda() && q.stride(3) == 1 && k.stride(3) == 1 && v.stride(3) == 1,

@@ -22,6 +22,9 @@ std::vector<at::Tensor> residual_scale_add_bwd(at::Tensor dout,
 // attention.hip
 std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
                                       c10::optional<at::Tensor> mask, double scale);
+at::Tensor attn_bwd_softmax(at::Tensor s, at::Tensor lse, c10::optional<at::Tensor> mask,
+                            double scale);
+at::Tensor attn_bwd_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale);
 
 // multi_tensor.hip
 void multi_tensor_adamw(
@@ -41,6 +44,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("residual_scale_add_fwd", &residual_scale_add_fwd, "residual+LayerScale+DropPath fwd");
   m.def("residual_scale_add_bwd", &residual_scale_add_bwd, "residual+LayerScale+DropPath bwd");
   m.def("attention_fwd", &attention_fwd, "flash attention fwd (MFMA, gfx950)");
+  m.def("attn_bwd_softmax", &attn_bwd_softmax, "fused softmax recompute for attention bwd");
+  m.def("attn_bwd_ds", &attn_bwd_ds, "fused dS epilogue for attention bwd");
   m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
   m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");

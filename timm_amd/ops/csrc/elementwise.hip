@@ -275,7 +275,7 @@ std::vector<at::Tensor> bias_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor b, 
     long rows = n / cols;
     int chunks = (cols / 8 + 63) / 64;
     if (cols % 8 == 0 && chunks <= 8) {
-      int blocks = (int)std::min((long)320, (rows + 3) / 4);
+      int blocks = (int)std::min((long)640, (rows + 3) / 4);
       auto go = [&](auto tag) {
         hipLaunchKernelGGL((bias_act_bwd_fast_kernel<T, decltype(tag)::value>),
             dim3(blocks), dim3(256), cols * sizeof(float), stream,
