@@ -14,7 +14,11 @@ from .config import (
 from .conv2d_same import Conv2dSame, Conv2dSameExport, conv2d_same
 from .conv_bn_act import ConvBnAct, ConvNormAct, ConvNormActAa
 from .create_act import create_act_layer, get_act_fn, get_act_layer
+from .create_attn import create_attn, create_attn_layer, get_attn
 from .create_conv2d import create_conv2d
+from .eca import CecaModule, EcaModule, EfficientChannelAttn
+from .gather_excite import GatherExcite
+from .global_context import GlobalContext
 from .create_norm import create_norm_layer, get_norm_layer
 from .drop import DropBlock2d, DropPath, calculate_drop_path_rates, drop_block_2d, drop_path
 from .format import Format, FormatT, get_channel_dim, get_spatial_dim, nchw_to, nhwc_to

@@ -30,4 +30,6 @@ from ._registry import (
 )
 
 # architecture modules (registration happens at import time)
+from .convnext import *
+from .resnet import *
 from .vision_transformer import *
