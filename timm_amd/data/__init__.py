@@ -1,1 +1,21 @@
+from .auto_augment import (
+    RandAugment, AutoAugment, rand_augment_ops, auto_augment_policy,
+    rand_augment_transform, auto_augment_transform, augment_and_mix_transform, AugMixAugment,
+)
+from .config import resolve_data_config, resolve_model_data_config
 from .constants import *
+from .dataset import ImageDataset, IterableImageDataset, AugMixDataset
+from .dataset_factory import create_dataset
+from .distributed_sampler import OrderedDistributedSampler, RepeatAugSampler
+from .loader import create_loader, fast_collate, PrefetchLoader, MultiEpochsDataLoader
+from .mixup import Mixup, FastCollateMixup, mixup_target, rand_bbox, rand_bbox_minmax
+from .random_erasing import RandomErasing
+from .real_labels import RealLabelsImagenet
+from .transforms import (
+    ToNumpy, ToTensor, MaybeToTensor, MaybePILToTensor, str_to_interp_mode, str_to_pil_interp,
+    interp_mode_to_str, RandomResizedCropAndInterpolation, CenterCropOrPad, RandomCropOrPad,
+    RandomPad, ResizeKeepRatio, TrimBorder,
+)
+from .transforms_factory import (
+    create_transform, transforms_imagenet_eval, transforms_imagenet_train, transforms_noaug_train,
+)
