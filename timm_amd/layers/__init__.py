@@ -47,6 +47,7 @@ from .pos_embed_sincos import (
     build_sincos2d_pos_embed, create_rope_embed, freq_bands, pixel_freq_bands, rope_rotate_half, rot,
 )
 from .squeeze_excite import EffectiveSEModule, EffectiveSqueezeExcite, SEModule, SqueezeExcite, SqueezeExciteCl
+from .test_time_pool import TestTimePoolHead, apply_test_time_pool
 from .trace_utils import _assert
 from .weight_init import (
     init_weight_jax, init_weight_vit, lecun_normal_, trunc_normal_, trunc_normal_tf_, variance_scaling_,

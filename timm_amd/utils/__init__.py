@@ -7,6 +7,7 @@ from .distributed import (
     distribute_bn, init_distributed_device, is_distributed_env, is_primary, reduce_tensor,
     world_info_from_env,
 )
+from .jit import set_jit_fuser, set_jit_legacy
 from .log import setup_default_logging, FormatterNoInfo
 from .metrics import AverageMeter, accuracy
 from .misc import natural_key, add_bool_arg, ParseKwargs
