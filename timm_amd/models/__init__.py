@@ -31,5 +31,6 @@ from ._registry import (
 
 # architecture modules (registration happens at import time)
 from .convnext import *
+from .eva import *
 from .resnet import *
 from .vision_transformer import *
