@@ -258,3 +258,38 @@ def test_vit_block_gpu_vs_cpu():
         m_gpu = model.to('cuda', torch.bfloat16)
         y_gpu = m_gpu(x.to('cuda', torch.bfloat16))
     assert rel_err(y_gpu.cpu(), y_cpu) < 5e-2, f'model output err {rel_err(y_gpu.cpu(), y_cpu)}'
+
+
+@pytest.mark.parametrize('shape,k,stride', [
+    ((2, 128, 56, 56), 7, 1),   # ConvNeXt-B stage 0
+    ((2, 512, 14, 14), 7, 1),
+    ((2, 64, 32, 32), 3, 2),    # EfficientNet-style dw s2
+    ((2, 96, 28, 28), 5, 1),
+])
+def test_depthwise_conv_nhwc(shape, k, stride):
+    _ext()
+    torch.manual_seed(9)
+    B, C, H, W = shape
+    pad = k // 2
+    x = torch.randn(B, C, H, W, device='cuda', dtype=torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    w = torch.randn(C, 1, k, k, device='cuda', dtype=torch.bfloat16) * 0.2
+    b = torch.randn(C, device='cuda', dtype=torch.bfloat16) * 0.1
+
+    x1 = x.clone().requires_grad_(True)
+    w1 = w.clone().requires_grad_(True)
+    b1 = b.clone().requires_grad_(True)
+    y = ops.depthwise_conv2d(x1, w1, b1, stride=stride, padding=pad)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    x2 = x.detach().float().requires_grad_(True)
+    w2 = w.detach().float().requires_grad_(True)
+    b2 = b.detach().float().requires_grad_(True)
+    y_ref = torch.nn.functional.conv2d(x2, w2, b2, stride=stride, padding=pad, groups=C)
+    y_ref.backward(dy.float())
+
+    assert rel_err(y, y_ref) < 2e-2, f'fwd err {rel_err(y, y_ref)}'
+    assert rel_err(x1.grad, x2.grad) < 3e-2
+    assert rel_err(w1.grad, w2.grad) < 3e-2
+    assert rel_err(b1.grad, b2.grad) < 3e-2

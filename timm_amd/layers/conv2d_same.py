@@ -49,6 +49,28 @@ class Conv2dSameExport(nn.Conv2d):
         return F.conv2d(x, self.weight, self.bias, self.stride, self.padding, self.dilation, self.groups)
 
 
+class DepthwiseConv2d(nn.Conv2d):
+    """Depthwise nn.Conv2d routed to the gfx950 NHWC HIP kernels when the
+    input is channels-last bf16 on device; identical params/state-dict to
+    nn.Conv2d."""
+
+    def forward(self, x):
+        from .. import ops
+        if (
+                x.is_cuda
+                and x.dtype == torch.bfloat16
+                and self.dilation[0] == 1 and self.dilation[1] == 1
+                and self.kernel_size[0] == self.kernel_size[1]
+                and self.stride[0] == self.stride[1]
+                and self.padding[0] == self.padding[1]
+                and x.is_contiguous(memory_format=torch.channels_last)
+        ):
+            return ops.depthwise_conv2d(
+                x, self.weight, self.bias,
+                stride=self.stride[0], padding=self.padding[0], dilation=1)
+        return super().forward(x)
+
+
 def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
     from .padding import get_padding_value
     padding = kwargs.pop('padding', '')
@@ -56,4 +78,6 @@ def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
     padding, is_dynamic = get_padding_value(padding, kernel_size, **kwargs)
     if is_dynamic:
         return Conv2dSame(in_chs, out_chs, kernel_size, **kwargs)
+    if kwargs.get('groups', 1) == in_chs and in_chs == out_chs and kwargs.get('dilation', 1) == 1:
+        return DepthwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
     return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)

@@ -26,6 +26,14 @@ at::Tensor attn_bwd_softmax(at::Tensor s, at::Tensor lse, c10::optional<at::Tens
                             double scale);
 at::Tensor attn_bwd_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale);
 
+// depthwise_conv.hip
+at::Tensor dwconv_fwd(at::Tensor x, at::Tensor w_t, c10::optional<at::Tensor> bias,
+                      long stride, long pad, long K, long Ho, long Wo);
+at::Tensor dwconv_bwd_data(at::Tensor dy, at::Tensor w_t, long stride, long pad, long K,
+                           long H, long W);
+std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stride, long pad,
+                                          long K, bool need_bias);
+
 // multi_tensor.hip
 void multi_tensor_adamw(
     std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
@@ -46,6 +54,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_fwd", &attention_fwd, "flash attention fwd (MFMA, gfx950)");
   m.def("attn_bwd_softmax", &attn_bwd_softmax, "fused softmax recompute for attention bwd");
   m.def("attn_bwd_ds", &attn_bwd_ds, "fused dS epilogue for attention bwd");
+  m.def("dwconv_fwd", &dwconv_fwd, "NHWC depthwise conv fwd (gfx950)");
+  m.def("dwconv_bwd_data", &dwconv_bwd_data, "NHWC depthwise conv bwd-data");
+  m.def("dwconv_bwd_weight", &dwconv_bwd_weight, "NHWC depthwise conv bwd-weight");
   m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
   m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");
