@@ -31,6 +31,7 @@ from ._registry import (
 
 # architecture modules (registration happens at import time)
 from .convnext import *
+from .efficientnet import *
 from .eva import *
 from .resnet import *
 from .vision_transformer import *
