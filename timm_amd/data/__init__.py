@@ -19,3 +19,8 @@ from .transforms import (
 from .transforms_factory import (
     create_transform, transforms_imagenet_eval, transforms_imagenet_train, transforms_noaug_train,
 )
+from .naflex_dataset import NaFlexCollator, NaFlexMapDatasetWrapper, calculate_naflex_batch_size
+from .naflex_loader import NaFlexPrefetchLoader, create_naflex_loader
+from .naflex_transforms import (
+    Patchify, RandomResizedCropToSequence, ResizeToSequence, get_image_size_for_seq, patchify_image,
+)

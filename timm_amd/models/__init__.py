@@ -33,5 +33,7 @@ from ._registry import (
 from .convnext import *
 from .efficientnet import *
 from .eva import *
+from .mobilenetv3 import *
+from .naflexvit import *
 from .resnet import *
 from .vision_transformer import *

@@ -1,0 +1,519 @@
+"""MobileNetV3 / MobileNetV4 / HardCoReNAS — MI355X-native.
+
+Capability parity with reference `timm/models/mobilenetv3.py` (1,526 LoC):
+`MobileNetV3` w/ efficient head (global pool BEFORE conv_head), builder-DSL
+arch defs, V4 universal-inverted-residual variants.
+"""
+from functools import partial
+from typing import Callable, List, Optional, Tuple, Union
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD, IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD
+from ..layers import SelectAdaptivePool2d, create_conv2d, get_norm_act_layer
+from ._builder import build_model_with_cfg, pretrained_cfg_for_features
+from ._efficientnet_blocks import SqueezeExcite
+from ._efficientnet_builder import (
+    BlockArgs, EfficientNetBuilder, decode_arch_def, efficientnet_init_weights,
+    round_channels, resolve_bn_args, resolve_act_layer,
+)
+from ._features import FeatureInfo, FeatureHooks, feature_take_indices
+from ._manipulate import checkpoint_seq
+from ._registry import generate_default_cfgs, register_model
+
+__all__ = ['MobileNetV3', 'MobileNetV3Features']
+
+
+class MobileNetV3(nn.Module):
+    """MobileNetV3 w/ efficient final head: features pooled before the final
+    1x1 conv (unlike EfficientNet where conv_head runs at full resolution)."""
+
+    def __init__(
+            self,
+            block_args: BlockArgs,
+            num_classes: int = 1000,
+            in_chans: int = 3,
+            stem_size: int = 16,
+            fix_stem: bool = False,
+            num_features: int = 1280,
+            head_bias: bool = True,
+            head_norm: bool = False,
+            pad_type: str = '',
+            act_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            aa_layer: Optional[Callable] = None,
+            se_layer: Optional[Callable] = None,
+            se_from_exp: bool = True,
+            round_chs_fn: Callable = round_channels,
+            drop_rate: float = 0.,
+            drop_path_rate: float = 0.,
+            layer_scale_init_value: Optional[float] = None,
+            global_pool: str = 'avg',
+    ):
+        super().__init__()
+        act_layer = act_layer or nn.ReLU
+        norm_layer = norm_layer or nn.BatchNorm2d
+        norm_act_layer = get_norm_act_layer(norm_layer, act_layer)
+        se_layer = se_layer or SqueezeExcite
+        self.num_classes = num_classes
+        self.drop_rate = drop_rate
+        self.grad_checkpointing = False
+
+        # Stem
+        if not fix_stem:
+            stem_size = round_chs_fn(stem_size)
+        self.conv_stem = create_conv2d(in_chans, stem_size, 3, stride=2, padding=pad_type)
+        self.bn1 = norm_act_layer(stem_size, inplace=True)
+
+        # Middle stages (IR/ER/DS Blocks)
+        builder = EfficientNetBuilder(
+            output_stride=32,
+            pad_type=pad_type,
+            round_chs_fn=round_chs_fn,
+            se_from_exp=se_from_exp,
+            act_layer=act_layer,
+            norm_layer=norm_layer,
+            aa_layer=aa_layer,
+            se_layer=se_layer,
+            drop_path_rate=drop_path_rate,
+            layer_scale_init_value=layer_scale_init_value,
+        )
+        self.blocks = nn.Sequential(*builder(stem_size, block_args))
+        self.feature_info = builder.features
+        head_chs = builder.in_chs
+
+        # Head + Pooling
+        self.num_features = head_chs
+        self.head_hidden_size = num_features
+        self.global_pool = SelectAdaptivePool2d(pool_type=global_pool)
+        num_pooled_chs = head_chs * self.global_pool.feat_mult()
+        if head_norm:
+            # mobilenet-v4 post-pooling PW conv is followed by a norm+act layer
+            self.conv_head = create_conv2d(num_pooled_chs, self.head_hidden_size, 1, padding=pad_type)  # never bias
+            self.norm_head = norm_act_layer(self.head_hidden_size)
+            self.act2 = nn.Identity()
+        else:
+            # mobilenet-v3 and others only have an activation after final PW conv
+            self.conv_head = create_conv2d(num_pooled_chs, self.head_hidden_size, 1, padding=pad_type, bias=head_bias)
+            self.norm_head = nn.Identity()
+            self.act2 = act_layer(inplace=True)
+        self.flatten = nn.Flatten(1) if global_pool else nn.Identity()  # don't flatten if pooling disabled
+        self.classifier = nn.Linear(self.head_hidden_size, num_classes) if num_classes > 0 else nn.Identity()
+
+        efficientnet_init_weights(self)
+
+    def as_sequential(self):
+        layers = [self.conv_stem, self.bn1]
+        layers.extend(self.blocks)
+        layers.extend([self.global_pool, self.conv_head, self.norm_head, self.act2])
+        layers.extend([nn.Flatten(), nn.Dropout(self.drop_rate), self.classifier])
+        return nn.Sequential(*layers)
+
+    @torch.jit.ignore
+    def group_matcher(self, coarse: bool = False):
+        return dict(
+            stem=r'^conv_stem|bn1',
+            blocks=r'^blocks\.(\d+)' if coarse else r'^blocks\.(\d+)\.(\d+)'
+        )
+
+    @torch.jit.ignore
+    def set_grad_checkpointing(self, enable: bool = True):
+        self.grad_checkpointing = enable
+
+    @torch.jit.ignore
+    def get_classifier(self) -> nn.Module:
+        return self.classifier
+
+    def reset_classifier(self, num_classes: int, global_pool: str = 'avg'):
+        self.num_classes = num_classes
+        # NOTE: cannot meaningfully change pooling of efficient head after creation
+        self.global_pool = SelectAdaptivePool2d(pool_type=global_pool)
+        self.flatten = nn.Flatten(1) if global_pool else nn.Identity()  # don't flatten if pooling disabled
+        self.classifier = nn.Linear(self.head_hidden_size, num_classes) if num_classes > 0 else nn.Identity()
+
+    def forward_intermediates(
+            self,
+            x: torch.Tensor,
+            indices: Optional[Union[int, List[int]]] = None,
+            norm: bool = False,
+            stop_early: bool = False,
+            output_fmt: str = 'NCHW',
+            intermediates_only: bool = False,
+            extra_blocks: bool = False,
+    ) -> Union[List[torch.Tensor], Tuple[torch.Tensor, List[torch.Tensor]]]:
+        assert output_fmt in ('NCHW',), 'Output shape must be NCHW.'
+        if extra_blocks:
+            take_indices, max_index = feature_take_indices(len(self.blocks) + 1, indices)
+        else:
+            take_indices, max_index = feature_take_indices(len(self.feature_info), indices)
+            take_indices = [self.feature_info[i]['stage'] for i in take_indices]
+            max_index = self.feature_info[max_index]['stage']
+        intermediates = []
+
+        # forward pass
+        feat_idx = 0  # stem is index 0
+        x = self.conv_stem(x)
+        x = self.bn1(x)
+        if feat_idx in take_indices:
+            intermediates.append(x)
+
+        if torch.jit.is_scripting() or not stop_early:  # can't slice blocks in torchscript
+            blocks = self.blocks
+        else:
+            blocks = self.blocks[:max_index]
+        for blk in blocks:
+            feat_idx += 1
+            x = blk(x)
+            if feat_idx in take_indices:
+                intermediates.append(x)
+
+        if intermediates_only:
+            return intermediates
+
+        return x, intermediates
+
+    def prune_intermediate_layers(
+            self,
+            indices: Union[int, List[int]] = 1,
+            prune_norm: bool = False,
+            prune_head: bool = True,
+            extra_blocks: bool = False,
+    ):
+        if extra_blocks:
+            take_indices, max_index = feature_take_indices(len(self.blocks) + 1, indices)
+        else:
+            take_indices, max_index = feature_take_indices(len(self.feature_info), indices)
+            max_index = self.feature_info[max_index]['stage']
+        self.blocks = self.blocks[:max_index]  # truncate blocks w/ stem as idx 0
+        if prune_head:
+            self.conv_head = nn.Identity()
+            self.norm_head = nn.Identity()
+            self.reset_classifier(0, '')
+        return take_indices
+
+    def forward_features(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.conv_stem(x)
+        x = self.bn1(x)
+        if self.grad_checkpointing and not torch.jit.is_scripting():
+            x = checkpoint_seq(self.blocks, x, flatten=True)
+        else:
+            x = self.blocks(x)
+        return x
+
+    def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
+        x = self.global_pool(x)
+        x = self.conv_head(x)
+        x = self.norm_head(x)
+        x = self.act2(x)
+        x = self.flatten(x)
+        if pre_logits:
+            return x
+        if self.drop_rate > 0.:
+            x = F.dropout(x, p=self.drop_rate, training=self.training)
+        return self.classifier(x)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.forward_features(x)
+        x = self.forward_head(x)
+        return x
+
+
+class MobileNetV3Features(nn.Module):
+    """MobileNetV3 feature extractor."""
+
+    def __init__(
+            self,
+            block_args: BlockArgs,
+            out_indices: Tuple[int, ...] = (0, 1, 2, 3, 4),
+            feature_location: str = 'bottleneck',
+            in_chans: int = 3,
+            stem_size: int = 16,
+            fix_stem: bool = False,
+            output_stride: int = 32,
+            pad_type: str = '',
+            round_chs_fn: Callable = round_channels,
+            se_from_exp: bool = True,
+            act_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            aa_layer: Optional[Callable] = None,
+            se_layer: Optional[Callable] = None,
+            drop_rate: float = 0.,
+            drop_path_rate: float = 0.,
+            layer_scale_init_value: Optional[float] = None,
+    ):
+        super().__init__()
+        act_layer = act_layer or nn.ReLU
+        norm_layer = norm_layer or nn.BatchNorm2d
+        norm_act_layer = get_norm_act_layer(norm_layer, act_layer)
+        se_layer = se_layer or SqueezeExcite
+        self.drop_rate = drop_rate
+        self.grad_checkpointing = False
+
+        # Stem
+        if not fix_stem:
+            stem_size = round_chs_fn(stem_size)
+        self.conv_stem = create_conv2d(in_chans, stem_size, 3, stride=2, padding=pad_type)
+        self.bn1 = norm_act_layer(stem_size, inplace=True)
+
+        # Middle stages (IR/ER/DS Blocks)
+        builder = EfficientNetBuilder(
+            output_stride=output_stride,
+            pad_type=pad_type,
+            round_chs_fn=round_chs_fn,
+            se_from_exp=se_from_exp,
+            act_layer=act_layer,
+            norm_layer=norm_layer,
+            aa_layer=aa_layer,
+            se_layer=se_layer,
+            drop_path_rate=drop_path_rate,
+            layer_scale_init_value=layer_scale_init_value,
+            feature_location=feature_location,
+        )
+        self.blocks = nn.Sequential(*builder(stem_size, block_args))
+        self.feature_info = FeatureInfo(builder.features, out_indices)
+        self._stage_out_idx = {f['stage']: f['index'] for f in self.feature_info.get_dicts()}
+
+        efficientnet_init_weights(self)
+
+        # Register feature extraction hooks with FeatureHooks helper
+        self.feature_hooks = None
+        if feature_location != 'bottleneck':
+            hooks = self.feature_info.get_dicts(keys=('module', 'hook_type'))
+            self.feature_hooks = FeatureHooks(hooks, self.named_modules())
+
+    @torch.jit.ignore
+    def set_grad_checkpointing(self, enable: bool = True):
+        self.grad_checkpointing = enable
+
+    def forward(self, x) -> List[torch.Tensor]:
+        x = self.conv_stem(x)
+        x = self.bn1(x)
+        if self.feature_hooks is None:
+            features = []
+            if 0 in self._stage_out_idx:
+                features.append(x)  # add stem out
+            for i, b in enumerate(self.blocks):
+                x = b(x)
+                if i + 1 in self._stage_out_idx:
+                    features.append(x)
+            return features
+        else:
+            self.blocks(x)
+            out = self.feature_hooks.get_output(x.device)
+            return list(out.values())
+
+
+def _create_mnv3(variant, pretrained=False, **kwargs):
+    features_mode = ''
+    model_cls = MobileNetV3
+    kwargs_filter = None
+    if kwargs.pop('features_only', False):
+        if 'feature_cfg' in kwargs or 'feature_cls' in kwargs:
+            features_mode = 'cfg'
+        else:
+            kwargs_filter = ('num_classes', 'num_features', 'head_conv', 'head_bias', 'head_norm', 'global_pool')
+            model_cls = MobileNetV3Features
+            features_mode = 'cls'
+
+    model = build_model_with_cfg(
+        model_cls,
+        variant,
+        pretrained,
+        features_only=features_mode == 'cfg',
+        pretrained_strict=features_mode != 'cls',
+        kwargs_filter=kwargs_filter,
+        **kwargs,
+    )
+    if features_mode == 'cls':
+        model.default_cfg = model.pretrained_cfg = pretrained_cfg_for_features(model.default_cfg)
+    return model
+
+
+def _gen_mobilenet_v3(variant, channel_multiplier=1.0, depth_multiplier=1.0, pretrained=False, **kwargs):
+    """Creates a MobileNet-V3 model."""
+    if 'small' in variant:
+        num_features = 1024
+        arch_def = [
+            # stage 0, 112x112 in
+            ['ds_r1_k3_s2_e1_c16_se0.25_nre'],  # relu
+            # stage 1, 56x56 in
+            ['ir_r1_k3_s2_e4.5_c24_nre', 'ir_r1_k3_s1_e3.67_c24_nre'],  # relu
+            # stage 2, 28x28 in
+            ['ir_r1_k5_s2_e4_c40_se0.25', 'ir_r2_k5_s1_e6_c40_se0.25'],  # hard-swish
+            # stage 3, 14x14 in
+            ['ir_r2_k5_s1_e3_c48_se0.25'],  # hard-swish
+            # stage 4, 14x14in
+            ['ir_r3_k5_s2_e6_c96_se0.25'],  # hard-swish
+            # stage 6, 7x7 in
+            ['cn_r1_k1_s1_c576'],  # hard-swish
+        ]
+    else:
+        num_features = 1280
+        arch_def = [
+            # stage 0, 112x112 in
+            ['ds_r1_k3_s1_e1_c16_nre'],  # relu
+            # stage 1, 112x112 in
+            ['ir_r1_k3_s2_e4_c24_nre', 'ir_r1_k3_s1_e3_c24_nre'],  # relu
+            # stage 2, 56x56 in
+            ['ir_r3_k5_s2_e3_c40_se0.25_nre'],  # relu
+            # stage 3, 28x28 in
+            ['ir_r1_k3_s2_e6_c80', 'ir_r1_k3_s1_e2.5_c80', 'ir_r2_k3_s1_e2.3_c80'],  # hard-swish
+            # stage 4, 14x14in
+            ['ir_r2_k3_s1_e6_c112_se0.25'],  # hard-swish
+            # stage 5, 14x14in
+            ['ir_r3_k5_s2_e6_c160_se0.25'],  # hard-swish
+            # stage 6, 7x7 in
+            ['cn_r1_k1_s1_c960'],  # hard-swish
+        ]
+    se_layer = partial(SqueezeExcite, gate_layer='hard_sigmoid', force_act_layer=nn.ReLU, rd_round_fn=round_channels)
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def, depth_multiplier),
+        num_features=num_features,
+        stem_size=16,
+        fix_stem=channel_multiplier < 0.75,
+        round_chs_fn=partial(round_channels, multiplier=channel_multiplier),
+        norm_layer=kwargs.pop('norm_layer', None) or partial(nn.BatchNorm2d, **resolve_bn_args(kwargs)),
+        act_layer=resolve_act_layer(kwargs, 'hard_swish'),
+        se_layer=se_layer,
+        **kwargs,
+    )
+    model = _create_mnv3(variant, pretrained, **model_kwargs)
+    return model
+
+
+def _gen_mobilenet_v4(variant, channel_multiplier=1.0, group_size=None, pretrained=False, **kwargs):
+    """Creates a MobileNet-V4 model (conv-only variants)."""
+    num_features = 1280
+    if 'medium' in variant:
+        stem_size = 32
+        act_layer = resolve_act_layer(kwargs, 'relu')
+        arch_def = [
+            # stage 0, 112x112 in
+            ['er_r1_k3_s2_e4_c48'],
+            # stage 1, 56x56 in
+            ['uir_r1_a3_k5_s2_e4_c80', 'uir_r1_a3_k3_s1_e2_c80'],
+            # stage 2, 28x28 in
+            [
+                'uir_r1_a3_k5_s2_e6_c160',
+                'uir_r2_a3_k3_s1_e4_c160',
+                'uir_r1_a3_k5_s1_e4_c160',
+                'uir_r1_a3_k3_s1_e4_c160',
+                'uir_r1_a3_k0_s1_e4_c160',
+                'uir_r1_a0_k0_s1_e2_c160',
+                'uir_r1_a3_k0_s1_e4_c160',
+            ],
+            # stage 3, 14x14in
+            [
+                'uir_r1_a5_k5_s2_e6_c256',
+                'uir_r1_a5_k5_s1_e4_c256',
+                'uir_r2_a3_k5_s1_e4_c256',
+                'uir_r1_a0_k0_s1_e4_c256',
+                'uir_r1_a3_k0_s1_e4_c256',
+                'uir_r1_a3_k5_s1_e2_c256',
+                'uir_r1_a5_k5_s1_e4_c256',
+                'uir_r2_a0_k0_s1_e4_c256',
+                'uir_r1_a5_k0_s1_e2_c256',
+            ],
+            # stage 4, 7x7
+            ['cn_r1_k1_s1_c960'],
+        ]
+    else:  # small
+        stem_size = 32
+        act_layer = resolve_act_layer(kwargs, 'relu')
+        arch_def = [
+            # stage 0, 112x112 in
+            ['cn_r1_k3_s2_e1_c32', 'cn_r1_k1_s1_e1_c32'],
+            # stage 1, 56x56 in
+            ['cn_r1_k3_s2_e1_c96', 'cn_r1_k1_s1_e1_c64'],
+            # stage 2, 28x28 in
+            [
+                'uir_r1_a5_k5_s2_e3_c96',
+                'uir_r4_a0_k3_s1_e2_c96',
+                'uir_r1_a3_k0_s1_e4_c96',
+            ],
+            # stage 3, 14x14 in
+            [
+                'uir_r1_a3_k3_s2_e6_c128',
+                'uir_r1_a5_k5_s1_e4_c128',
+                'uir_r1_a0_k5_s1_e4_c128',
+                'uir_r1_a0_k5_s1_e3_c128',
+                'uir_r2_a0_k3_s1_e4_c128',
+            ],
+            # stage 4, 7x7
+            ['cn_r1_k1_s1_c960'],
+        ]
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def, group_size=group_size),
+        head_bias=False,
+        head_norm=True,
+        num_features=num_features,
+        stem_size=stem_size,
+        fix_stem=channel_multiplier < 0.75,
+        round_chs_fn=partial(round_channels, multiplier=channel_multiplier),
+        norm_layer=kwargs.pop('norm_layer', None) or partial(nn.BatchNorm2d, **resolve_bn_args(kwargs)),
+        act_layer=act_layer,
+        layer_scale_init_value=1e-5,
+        **kwargs,
+    )
+    model = _create_mnv3(variant, pretrained, **model_kwargs)
+    return model
+
+
+def _cfg(url: str = '', **kwargs):
+    return {
+        'url': url,
+        'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': (7, 7),
+        'crop_pct': 0.875, 'interpolation': 'bilinear',
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'conv_stem', 'classifier': 'classifier',
+        **kwargs,
+    }
+
+
+default_cfgs = generate_default_cfgs({
+    'mobilenetv3_large_100.ra_in1k': _cfg(interpolation='bicubic'),
+    'mobilenetv3_large_075.untrained': _cfg(),
+    'mobilenetv3_small_100.lamb_in1k': _cfg(interpolation='bicubic'),
+    'mobilenetv3_small_075.lamb_in1k': _cfg(interpolation='bicubic'),
+    'mobilenetv4_conv_small.e2400_r224_in1k': _cfg(interpolation='bicubic', test_input_size=(3, 256, 256), test_crop_pct=0.95),
+    'mobilenetv4_conv_medium.e500_r256_in1k': _cfg(
+        interpolation='bicubic', input_size=(3, 256, 256), pool_size=(8, 8), test_input_size=(3, 320, 320), test_crop_pct=1.0),
+})
+
+
+@register_model
+def mobilenetv3_large_100(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_large_100', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv3_large_075(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_large_075', 0.75, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv3_small_100(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_small_100', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv3_small_075(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_small_075', 0.75, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_small(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_small', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_medium(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_medium', 1.0, pretrained=pretrained, **kwargs)
+    return model
