@@ -293,3 +293,24 @@ def test_depthwise_conv_nhwc(shape, k, stride):
     assert rel_err(x1.grad, x2.grad) < 3e-2
     assert rel_err(w1.grad, w2.grad) < 3e-2
     assert rel_err(b1.grad, b2.grad) < 3e-2
+
+
+@pytest.mark.parametrize('model_name', ['eva02_tiny_patch14_224', 'naflexvit_base_patch16_gap', 'convnext_atto', 'efficientnet_b0'])
+def test_model_gpu_vs_cpu(model_name):
+    """Model forward on GPU (HIP kernels) vs CPU fp32 reference."""
+    _ext()
+    torch.manual_seed(11)
+    import timm_amd
+    model = timm_amd.create_model(model_name, num_classes=10)
+    model.eval()
+    x = torch.randn(2, 3, 224, 224)
+    with torch.no_grad():
+        y_cpu = model(x.float())
+        m_gpu = model.to('cuda', torch.bfloat16)
+        xg = x.to('cuda', torch.bfloat16)
+        if 'convnext' in model_name or 'efficientnet' in model_name:
+            m_gpu = m_gpu.to(memory_format=torch.channels_last)
+            xg = xg.contiguous(memory_format=torch.channels_last)
+        y_gpu = m_gpu(xg)
+    err = rel_err(y_gpu.cpu(), y_cpu)
+    assert err < 0.1, f'{model_name} output err {err}'
