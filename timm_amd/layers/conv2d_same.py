@@ -49,6 +49,31 @@ class Conv2dSameExport(nn.Conv2d):
         return F.conv2d(x, self.weight, self.bias, self.stride, self.padding, self.dilation, self.groups)
 
 
+class PatchifyConv2d(nn.Conv2d):
+    """Non-overlapping (stride == kernel, pad 0) conv lowered to
+    reshape + hipBLASLt GEMM — MIOpen's bf16 NHWC path falls back to naive
+    double-accum kernels for these shapes (ConvNeXt stem 4x4/s4 and 2x2/s2
+    downsamples were 88% of remaining conv time, see gpurun_out/prof_cnx2).
+    Identical params/state-dict to nn.Conv2d."""
+
+    def forward(self, x):
+        kh, kw = self.kernel_size
+        B, C, H, W = x.shape
+        if (
+                x.is_cuda
+                and self.stride[0] == kh and self.stride[1] == kw
+                and self.padding[0] == 0 and self.padding[1] == 0
+                and H % kh == 0 and W % kw == 0
+        ):
+            nh, nw = H // kh, W // kw
+            patches = x.view(B, C, nh, kh, nw, kw).permute(0, 2, 4, 1, 3, 5).reshape(B, nh * nw, C * kh * kw)
+            w = self.weight.reshape(self.out_channels, -1)
+            y = torch.nn.functional.linear(patches, w, self.bias)
+            # [B, nh, nw, O] -> NCHW view with channels-last strides
+            return y.view(B, nh, nw, self.out_channels).permute(0, 3, 1, 2)
+        return super().forward(x)
+
+
 class DepthwiseConv2d(nn.Conv2d):
     """Depthwise nn.Conv2d routed to the gfx950 NHWC HIP kernels when the
     input is channels-last bf16 on device; identical params/state-dict to
@@ -80,4 +105,10 @@ def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
         return Conv2dSame(in_chs, out_chs, kernel_size, **kwargs)
     if kwargs.get('groups', 1) == in_chs and in_chs == out_chs and kwargs.get('dilation', 1) == 1:
         return DepthwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
+    ks = kernel_size if isinstance(kernel_size, int) else None
+    if (
+            ks is not None and padding == 0 and kwargs.get('groups', 1) == 1
+            and kwargs.get('dilation', 1) == 1 and kwargs.get('stride', 1) == ks
+    ):
+        return PatchifyConv2d(in_chs, out_chs, kernel_size, padding=0, **kwargs)
     return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)

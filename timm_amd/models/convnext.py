@@ -235,9 +235,11 @@ class ConvNeXt(nn.Module):
 
         assert stem_type in ('patch', 'overlap', 'overlap_tiered', 'overlap_act')
         if stem_type == 'patch':
-            # NOTE: this stem is a minimal form of ViT PatchEmbed, as used in SwinTransformer w/ patch_size = 4
+            # NOTE: this stem is a minimal form of ViT PatchEmbed; stride==kernel so it
+            # lowers to reshape + hipBLASLt GEMM (PatchifyConv2d) instead of MIOpen
+            from ..layers.conv2d_same import PatchifyConv2d
             self.stem = nn.Sequential(
-                nn.Conv2d(in_chans, dims[0], kernel_size=patch_size, stride=patch_size, bias=conv_bias),
+                PatchifyConv2d(in_chans, dims[0], kernel_size=patch_size, stride=patch_size, bias=conv_bias),
                 norm_layer(dims[0]),
             )
             stem_stride = patch_size
