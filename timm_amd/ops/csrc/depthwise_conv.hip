@@ -127,10 +127,12 @@ void dwconv_bwd_data_kernel(
 }
 
 // ---- backward weight: dw[kh,kw,c] = sum_{b,ho,wo} dy[b,ho,wo,c] * x[...] ----
-// Grid: (spatial-chunk, tap, channel-vec). Per-thread register accumulation
-// over its spatial slice, block-level LDS reduce, one global fp32 atomic
-// per (tap, channel) per block.
-template <typename T>
+// Grid: (spatial-chunk, kh, channel-vec).  One dy read serves a full kernel
+// ROW (all kw taps accumulate in registers, kMaxK*8 floats); x reads for
+// neighbouring kw share cache lines.  Read amplification K vs the naive
+// per-tap scheme's K*K.  Block-level LDS reduce, one fp32 atomic per
+// (tap,channel) per block.
+template <typename T, int kMaxK>
 __global__ __launch_bounds__(kThreads)
 void dwconv_bwd_weight_kernel(
     const T* __restrict__ dy,     // [B,Ho,Wo,C]
@@ -140,18 +142,21 @@ void dwconv_bwd_weight_kernel(
     int B, int H, int W, int C,
     int Ho, int Wo, int K, int stride, int pad) {
   const int c8 = C / 8;
-  const int tap = blockIdx.y;          // kh*K+kw
-  const int kh = tap / K, kw = tap % K;
+  const int kh = blockIdx.y;
   const int cv = blockIdx.z;
   const int c0 = cv * 8;
 
   const long spatial = (long)B * Ho * Wo;
-  float acc[8];
+  float acc[kMaxK][8];
   float bacc[8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) { acc[j] = 0.f; bacc[j] = 0.f; }
+  for (int kw = 0; kw < kMaxK; ++kw)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[kw][j] = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) bacc[j] = 0.f;
 
-  const bool do_bias = (dbias != nullptr) && (tap == 0);
+  const bool do_bias = (dbias != nullptr) && (kh == 0);
   for (long s = (long)blockIdx.x * kThreads + threadIdx.x; s < spatial;
        s += (long)gridDim.x * kThreads) {
     const int wo = s % Wo;
@@ -159,38 +164,52 @@ void dwconv_bwd_weight_kernel(
     const int ho = p % Ho;
     const int b = p / Ho;
     const int hi = ho * stride - pad + kh;
-    const int wi = wo * stride - pad + kw;
     bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(
         dy + (((long)b * Ho + ho) * Wo + wo) * C + c0);
     if (do_bias) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) bacc[j] += (float)gv[j];
     }
-    if (hi < 0 || hi >= H || wi < 0 || wi >= W) continue;
-    bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(
-        x + (((long)b * H + hi) * W + wi) * C + c0);
+    if (hi < 0 || hi >= H) continue;
+    const T* xrow = x + (((long)b * H + hi) * W) * C + c0;
+    const int wi0 = wo * stride - pad;
+    for (int kw = 0; kw < K; ++kw) {
+      const int wi = wi0 + kw;
+      if (wi < 0 || wi >= W) continue;
+      bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi * C);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc[j] += (float)gv[j] * (float)xv[j];
+      for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)xv[j];
+    }
   }
 
-  // block reduce: 8 floats per thread -> LDS [kThreads][8] tree or atomic into [8]
-  __shared__ float red[8];
+  // block reduce via LDS (wave-level shuffle first), then one global atomic
+  // per (tap, channel) per block
+  __shared__ float red[kMaxK][8];
   __shared__ float redb[8];
-  if (threadIdx.x < 8) { red[threadIdx.x] = 0.f; redb[threadIdx.x] = 0.f; }
+  if (threadIdx.x < 8) {
+    redb[threadIdx.x] = 0.f;
+    for (int kw = 0; kw < kMaxK; ++kw) red[kw][threadIdx.x] = 0.f;
+  }
   __syncthreads();
-  // wave-level reduce first to cut LDS atomic contention to 4 per slot
+  for (int kw = 0; kw < K; ++kw) {
 #pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    float v = wave_reduce_sum(acc[j]);
-    float vb = do_bias ? wave_reduce_sum(bacc[j]) : 0.f;
-    if ((threadIdx.x % WAVE_SIZE) == 0) {
-      atomicAdd(&red[j], v);
-      if (do_bias) atomicAdd(&redb[j], vb);
+    for (int j = 0; j < 8; ++j) {
+      float v = wave_reduce_sum(acc[kw][j]);
+      if ((threadIdx.x % WAVE_SIZE) == 0) atomicAdd(&red[kw][j], v);
+    }
+  }
+  if (do_bias) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float vb = wave_reduce_sum(bacc[j]);
+      if ((threadIdx.x % WAVE_SIZE) == 0) atomicAdd(&redb[j], vb);
     }
   }
   __syncthreads();
   if (threadIdx.x < 8) {
-    atomicAdd(&dw[(long)tap * C + c0 + threadIdx.x], red[threadIdx.x]);
+    for (int kw = 0; kw < K; ++kw) {
+      atomicAdd(&dw[((long)kh * K + kw) * C + c0 + threadIdx.x], red[kw][threadIdx.x]);
+    }
     if (do_bias) atomicAdd(&dbias[c0 + threadIdx.x], redb[threadIdx.x]);
   }
 }
@@ -247,11 +266,21 @@ std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stri
   }
   auto stream = at::hip::getCurrentHIPStream();
   long spatial = (long)B * Ho * Wo;
-  int sblocks = (int)std::min((long)512, (spatial + kThreads - 1) / kThreads);
-  dim3 grid(sblocks, K * K, C / 8);
-  hipLaunchKernelGGL((dwconv_bwd_weight_kernel<__bf16>), grid, dim3(kThreads), 0, stream,
-      (const __bf16*)dy.data_ptr(), (const __bf16*)x.data_ptr(),
-      dw.data_ptr<float>(), dbias_ptr, B, H, W, C, Ho, Wo, (int)K, (int)stride, (int)pad);
+  // size spatial chunks so total blocks ≈ several per CU without huge atomic depth
+  long per_kc = (spatial + kThreads - 1) / kThreads;
+  int sblocks = (int)std::min(std::max<long>(2048 / (K * (C / 8)) + 1, 8), per_kc);
+  dim3 grid(sblocks, K, C / 8);
+  TORCH_CHECK(K <= 9, "dwconv_bwd_weight: kernel size <= 9 supported");
+  auto launch = [&](auto tag) {
+    hipLaunchKernelGGL((dwconv_bwd_weight_kernel<__bf16, decltype(tag)::value>),
+        grid, dim3(kThreads), 0, stream,
+        (const __bf16*)dy.data_ptr(), (const __bf16*)x.data_ptr(),
+        dw.data_ptr<float>(), dbias_ptr, B, H, W, C, Ho, Wo, (int)K, (int)stride, (int)pad);
+  };
+  if (K <= 3) launch(std::integral_constant<int, 3>{});
+  else if (K <= 5) launch(std::integral_constant<int, 5>{});
+  else if (K <= 7) launch(std::integral_constant<int, 7>{});
+  else launch(std::integral_constant<int, 9>{});
   HIP_CHECK_LAST();
   if (need_bias) return {dw, dbias};
   return {dw};
