@@ -33,6 +33,7 @@ from ._registry import (
 from .convnext import *
 from .efficientnet import *
 from .eva import *
+from .hiera import *
 from .mobilenetv3 import *
 from .naflexvit import *
 from .resnet import *

@@ -1,0 +1,831 @@
+"""Hiera — hierarchical ViT (reference `timm/models/hiera.py`, 1,049 LoC).
+
+`Unroll` (:112) / `Reroll` (:174) mask-unit shuffling, `MaskUnitAttention`
+(:255 — windowed + pooled-q attention), `HieraBlock` (:346), `Hiera` (:461).
+Attention runs on the fused HIP kernel (head dims 24-96 fall back to the
+math path when not a multiple of 32; main dims are 96/… so kernel covers
+the large stages).
+"""
+import math
+from functools import partial
+from typing import Callable, Dict, List, Optional, Tuple, Type, Union
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
+from .. import ops
+from ..layers import DropPath, Mlp, LayerNorm, use_fused_attn, _assert, get_norm_layer, to_2tuple
+from ._builder import build_model_with_cfg
+from ._features import feature_take_indices
+from ._manipulate import checkpoint
+from ._registry import generate_default_cfgs, register_model
+
+__all__ = ['Hiera']
+
+
+def conv_nd(n: int) -> Type[nn.Module]:
+    """Returns a conv with nd (e.g., Conv2d for n=2). Work up to n=3."""
+    return [nn.Identity, nn.Conv1d, nn.Conv2d, nn.Conv3d][n]
+
+
+def get_resized_mask(target_size: List[int], mask: torch.Tensor) -> torch.Tensor:
+    # target_size: [(T), (H), W]
+    # (spatial) mask: [B, C, (t), (h), w]
+    if mask is None:
+        return mask
+
+    _assert(len(mask.shape[2:]) == len(target_size), "mask spatial shape and target_size must match.")
+    if mask.shape[2:] != target_size:
+        return F.interpolate(mask.float(), size=target_size)
+    return mask
+
+
+def undo_windowing(
+        x: torch.Tensor,
+        shape: List[int],
+        mu_shape: List[int],
+) -> torch.Tensor:
+    """Restore spatial organization by undoing windowed organization of mask units."""
+    D = len(shape)
+    B, C = x.shape[0], x.shape[-1]
+    # [B, #MUy*#MUx, MUy, MUx, C] -> [B, #MUy, #MUx, MUy, MUx, C]
+    num_MUs = [s // mu for s, mu in zip(shape, mu_shape)]
+    x = x.view(B, *num_MUs, *mu_shape, C)
+
+    # [B, #MUy, #MUx, MUy, MUx, C] -> [B, #MUy*MUy, #MUx*MUx, C]
+    permute = (
+        [0]
+        + sum([list(p) for p in zip(range(1, 1 + D), range(1 + D, 1 + 2 * D))], [])
+        + [len(x.shape) - 1]
+    )
+    x = x.permute(permute).reshape(B, *shape, C)
+
+    return x
+
+
+class Unroll(nn.Module):
+    """Reorders tokens so mask units (and pooled tokens) are contiguous in
+    memory (reference `hiera.py:112`)."""
+
+    def __init__(
+            self,
+            input_size: Tuple[int, ...],
+            patch_stride: Tuple[int, ...],
+            unroll_schedule: List[Tuple[int, ...]],
+    ):
+        super().__init__()
+        self.size = [i // s for i, s in zip(input_size, patch_stride)]
+        self.schedule = unroll_schedule
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """Input: Flattened patch embeddings [B, N, C]
+        Output: Patch embeddings [B, N, C] permuted such that [B, 4, N//4, C].max(1)
+        etc. performs MaxPoolNd."""
+        B, _, C = x.shape
+
+        cur_size = self.size
+        x = x.view(*([B] + cur_size + [C]))
+
+        for strides in self.schedule:
+            # Move patches with the given strides to the batch dimension
+
+            # Create the new shape (e.g. [B, H//stride, stride, W//stride, stride, C])
+            cur_size = [i // s for i, s in zip(cur_size, strides)]
+            new_shape = [B] + sum([[i, s] for i, s in zip(cur_size, strides)], []) + [C]
+            x = x.view(new_shape)
+
+            # Move the patches to the batch dimension (e.g. [B, S, H//stride, W//stride, C])
+            L = len(new_shape)
+            permute = [0] + list(range(2, L - 1, 2)) + list(range(1, L - 1, 2)) + [L - 1]
+            x = x.permute(permute)
+
+            # Now finally flatten the relevant dims into the batch dimension
+            x = x.flatten(0, len(strides))
+            B *= math.prod(strides)
+
+        x = x.reshape(-1, math.prod(self.size), C)
+        return x
+
+
+class Reroll(nn.Module):
+    """Undoes the "unroll" operation so that you can use intermediate features
+    (reference `hiera.py:174`)."""
+
+    def __init__(
+            self,
+            input_size: Tuple[int, ...],
+            patch_stride: Tuple[int, ...],
+            unroll_schedule: List[Tuple[int, ...]],
+            stage_ends: List[int],
+            q_pool: int,
+    ):
+        super().__init__()
+        self.size = [i // s for i, s in zip(input_size, patch_stride)]
+
+        # The first stage has to reverse everything
+        # The next stage has to reverse all but the first unroll, etc.
+        self.schedule = {}
+        size = self.size
+        for i in range(stage_ends[-1] + 1):
+            self.schedule[i] = unroll_schedule, size
+            # schedule unchanged if no pooling at a stage end
+            if i in stage_ends[:q_pool]:
+                if len(unroll_schedule) > 0:
+                    size = [n // s for n, s in zip(size, unroll_schedule[0])]
+                unroll_schedule = unroll_schedule[1:]
+
+    def forward(
+            self,
+            x: torch.Tensor,
+            block_idx: int,
+            mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        """Roll the given tensor back up to spatial order assuming it's from the given block."""
+        schedule, size = self.schedule[block_idx]
+        B, N, C = x.shape
+
+        D = len(size)
+        cur_mu_shape = [1] * D
+
+        for strides in schedule:
+            # Extract the current patch from N
+            x = x.view(B, *strides, N // math.prod(strides), *cur_mu_shape, C)
+
+            # Move that patch into the current MU
+            # Example in 2d: [B, Sy, Sx, N//(Sy*Sx), MUy, MUx, C] -> [B, N//(Sy*Sx), Sy, MUy, Sx, MUx, C]
+            L = len(x.shape)
+            permute = (
+                [0, 1 + D]
+                + sum([list(p) for p in zip(range(1, 1 + D), range(1 + D + 1, L - 1))], [])
+                + [L - 1]
+            )
+            x = x.permute(permute)
+
+            # Reshape to [B, N//(Sy*Sx), *MU, C]
+            for i in range(D):
+                cur_mu_shape[i] *= strides[i]
+            x = x.reshape(B, -1, *cur_mu_shape, C)
+            N = x.shape[1]
+
+        # Current shape (e.g., 2d: [B, #MUy*#MUx, MUy, MUx, C])
+        x = x.view(B, N, *cur_mu_shape, C)
+
+        # If masked, return [B, #MUs, MUy, MUx, C]
+        if mask is not None:
+            return x
+
+        # If not masked, we can return [B, H, W, C]
+        x = undo_windowing(x, size, cur_mu_shape)
+
+        return x
+
+
+class MaskUnitAttention(nn.Module):
+    """The attention used in Hiera: local attention within mask units with
+    optional query pooling (reference `hiera.py:255`)."""
+    fused_attn: torch.jit.Final[bool]
+
+    def __init__(
+            self,
+            dim: int,
+            dim_out: int,
+            heads: int,
+            q_stride: int = 1,
+            window_size: int = 0,
+            use_mask_unit_attn: bool = False,
+    ):
+        super().__init__()
+
+        self.dim = dim
+        self.dim_out = dim_out
+        self.heads = heads
+        self.q_stride = q_stride
+        self.head_dim = dim_out // heads
+        self.scale = self.head_dim ** -0.5
+        self.fused_attn = use_fused_attn()
+
+        self.qkv = nn.Linear(dim, 3 * dim_out)
+        self.proj = nn.Linear(dim_out, dim_out)
+
+        self.window_size = window_size
+        self.use_mask_unit_attn = use_mask_unit_attn
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        """Input should be of shape [batch, tokens, channels]."""
+        B, N, _ = x.shape
+        num_windows = (N // (self.q_stride * self.window_size)) if self.use_mask_unit_attn else 1
+        qkv = self.qkv(x).reshape(B, -1, num_windows, 3, self.heads, self.head_dim).permute(3, 0, 4, 2, 1, 5)
+        q, k, v = qkv.unbind(0)
+
+        if self.q_stride > 1:
+            # Refer to Unroll to see how this performs a maxpool-Nd
+            q = q.view(B, self.heads, num_windows, self.q_stride, -1, self.head_dim).amax(dim=3)
+
+        if self.fused_attn and self.head_dim % 32 == 0 and q.dtype == torch.bfloat16:
+            # flatten (windows) into batch for the fused kernel
+            qq = q.permute(0, 2, 1, 3, 4).reshape(-1, self.heads, q.shape[-2], self.head_dim)
+            kk = k.permute(0, 2, 1, 3, 4).reshape(-1, self.heads, k.shape[-2], self.head_dim)
+            vv = v.permute(0, 2, 1, 3, 4).reshape(-1, self.heads, v.shape[-2], self.head_dim)
+            xo = ops.flash_attention(qq, kk, vv)
+            x = xo.reshape(B, num_windows, self.heads, -1, self.head_dim).permute(0, 2, 1, 3, 4)
+        else:
+            attn = (q * self.scale) @ k.transpose(-1, -2)
+            attn = attn.softmax(dim=-1)
+            x = attn @ v
+
+        x = x.transpose(1, 3).reshape(B, -1, self.dim_out)
+        x = self.proj(x)
+        return x
+
+
+class HieraBlock(nn.Module):
+    """Hiera block: MU-attention + MLP, w/ dim expansion proj (reference `hiera.py:346`)."""
+
+    def __init__(
+            self,
+            dim: int,
+            dim_out: int,
+            heads: int,
+            mlp_ratio: float = 4.0,
+            drop_path: float = 0.0,
+            init_values: Optional[float] = None,
+            norm_layer: Type[nn.Module] = nn.LayerNorm,
+            act_layer: Type[nn.Module] = nn.GELU,
+            q_stride: int = 1,
+            window_size: int = 0,
+            use_expand_proj: bool = True,
+            use_mask_unit_attn: bool = False,
+    ):
+        super().__init__()
+        self.dim = dim
+        self.dim_out = dim_out
+
+        self.norm1 = norm_layer(dim)
+        if dim != dim_out:
+            self.do_expand = True
+            if use_expand_proj:
+                self.proj = nn.Linear(dim, dim_out)
+            else:
+                assert dim_out == dim * 2
+                self.proj = None
+        else:
+            self.do_expand = False
+            self.proj = None
+        self.attn = MaskUnitAttention(
+            dim,
+            dim_out,
+            heads,
+            q_stride,
+            window_size,
+            use_mask_unit_attn,
+        )
+        self.ls1 = nn.Identity() if init_values is None else LayerScaleH(dim_out, init_values)
+        self.drop_path1 = DropPath(drop_path) if drop_path > 0 else nn.Identity()
+
+        self.norm2 = norm_layer(dim_out)
+        self.mlp = Mlp(dim_out, int(dim_out * mlp_ratio), act_layer=act_layer)
+        self.ls2 = nn.Identity() if init_values is None else LayerScaleH(dim_out, init_values)
+        self.drop_path2 = DropPath(drop_path) if drop_path > 0 else nn.Identity()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        # Attention + Q Pooling
+        x_norm = self.norm1(x)
+        if self.do_expand:
+            if self.proj is not None:
+                x = self.proj(x_norm)
+                # max-pool over the unrolled q_stride tokens
+                x = x.view(x.shape[0], self.attn.q_stride, -1, x.shape[-1]).amax(dim=1)
+            else:
+                x = torch.cat([
+                    x.view(x.shape[0], self.attn.q_stride, -1, x.shape[-1]).amax(dim=1),
+                    x.view(x.shape[0], self.attn.q_stride, -1, x.shape[-1]).mean(dim=1),
+                ], dim=-1)
+        x = x + self.drop_path1(self.ls1(self.attn(x_norm)))
+
+        # MLP
+        x = x + self.drop_path2(self.ls2(self.mlp(self.norm2(x))))
+        return x
+
+
+class LayerScaleH(nn.Module):
+    def __init__(self, dim: int, init_values: float = 1e-5, inplace: bool = False):
+        super().__init__()
+        self.inplace = inplace
+        self.gamma = nn.Parameter(init_values * torch.ones(dim))
+
+    def forward(self, x):
+        return x.mul_(self.gamma) if self.inplace else x * self.gamma
+
+
+class PatchEmbedH(nn.Module):
+    """Hiera patch embed: overlapping conv (7x7 s4)."""
+
+    def __init__(
+            self,
+            dim_in: int,
+            dim_out: int,
+            kernel: Tuple[int, ...],
+            stride: Tuple[int, ...],
+            padding: Tuple[int, ...],
+            reshape: bool = True,
+    ):
+        super().__init__()
+
+        # Support any number of spatial dimensions
+        self.spatial_dims = len(kernel)
+        self.reshape = reshape
+        self.proj = conv_nd(self.spatial_dims)(
+            dim_in,
+            dim_out,
+            kernel_size=kernel,
+            stride=stride,
+            padding=padding,
+        )
+
+    def forward(
+            self,
+            x: torch.Tensor,
+            mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        if mask is not None:
+            mask = get_resized_mask(target_size=x.shape[2:], mask=mask)
+            x = self.proj(x * mask.to(torch.bool))
+        else:
+            x = self.proj(x)
+        if self.reshape:
+            x = x.reshape(x.shape[0], x.shape[1], -1).transpose(2, 1)
+        return x
+
+
+class Hiera(nn.Module):
+    """Hiera model (reference `hiera.py:461`)."""
+
+    def __init__(
+            self,
+            img_size: Tuple[int, ...] = (224, 224),
+            in_chans: int = 3,
+            embed_dim: int = 96,  # initial embed dim
+            num_heads: int = 1,  # initial number of heads
+            num_classes: int = 1000,
+            global_pool: str = 'avg',
+            stages: Tuple[int, ...] = (2, 3, 16, 3),
+            q_pool: int = 3,  # number of q_pool stages
+            q_stride: Tuple[int, ...] = (2, 2),
+            mask_unit_size: Tuple[int, ...] = (8, 8),  # must divide q_stride ** (#stages-1)
+            # mask_unit_attn: which stages use mask unit attention?
+            mask_unit_attn: Tuple[bool, ...] = (True, True, False, False),
+            use_expand_proj: bool = True,
+            dim_mul: float = 2.0,
+            head_mul: float = 2.0,
+            patch_kernel: Tuple[int, ...] = (7, 7),
+            patch_stride: Tuple[int, ...] = (4, 4),
+            patch_padding: Tuple[int, ...] = (3, 3),
+            mlp_ratio: float = 4.0,
+            drop_path_rate: float = 0.0,
+            init_values: Optional[float] = None,
+            fix_init: bool = True,
+            weight_init: str = '',
+            norm_layer: Union[str, nn.Module] = "LayerNorm",
+            drop_rate: float = 0.0,
+            patch_drop_rate: float = 0.0,
+            head_init_scale: float = 0.001,
+            sep_pos_embed: bool = False,
+            abs_win_pos_embed: bool = False,
+            global_pos_size: Tuple[int, int] = (14, 14),
+    ):
+        super().__init__()
+        self.num_classes = num_classes
+        self.grad_checkpointing = False
+        norm_layer = get_norm_layer(norm_layer)
+        if isinstance(img_size, int):
+            img_size = to_2tuple(img_size)
+
+        self.patch_stride = patch_stride
+        self.tokens_spatial_shape = [i // s for i, s in zip(img_size, patch_stride)]
+        num_tokens = math.prod(self.tokens_spatial_shape)
+        flat_mu_size = math.prod(mask_unit_size)
+        flat_q_stride = math.prod(q_stride)
+        assert q_pool < len(stages)
+        self.q_pool, self.q_stride = q_pool, q_stride
+        self.mu_size, self.mask_unit_size = flat_mu_size, mask_unit_size
+        self.mask_spatial_shape = [i // s for i, s in zip(self.tokens_spatial_shape, self.mask_unit_size)]
+        self.stage_ends = [sum(stages[:i]) - 1 for i in range(1, len(stages) + 1)]
+        self.patch_drop_rate = patch_drop_rate
+
+        self.patch_embed = PatchEmbedH(
+            in_chans,
+            embed_dim,
+            patch_kernel,
+            patch_stride,
+            patch_padding,
+        )
+
+        self.pos_embed: Optional[nn.Parameter] = None
+        self.pos_embed_win: Optional[nn.Parameter] = None
+        if sep_pos_embed:
+            self.pos_embed_spatial = nn.Parameter(
+                torch.zeros(1, self.tokens_spatial_shape[1] * self.tokens_spatial_shape[2], embed_dim))
+            self.pos_embed_temporal = nn.Parameter(
+                torch.zeros(1, self.tokens_spatial_shape[0], embed_dim))
+        else:
+            self.pos_embed_spatial = self.pos_embed_temporal = None
+            if abs_win_pos_embed:
+                # absolute win, params NCHW to make tile + interpolate more natural before add & reshape
+                self.pos_embed = nn.Parameter(torch.zeros(1, embed_dim, *global_pos_size))
+                self.pos_embed_win = nn.Parameter(torch.zeros(1, embed_dim, *mask_unit_size))
+            else:
+                self.pos_embed = nn.Parameter(torch.zeros(1, num_tokens, embed_dim))
+
+        # Setup roll and reroll modules
+        self.unroll = Unroll(
+            img_size,
+            patch_stride,
+            [q_stride] * len(self.stage_ends[:-1])
+        )
+        self.reroll = Reroll(
+            img_size,
+            patch_stride,
+            [q_stride] * len(self.stage_ends[:-1]),
+            self.stage_ends,
+            q_pool,
+        )
+        # q_pool locations
+        q_pool_blocks = [x + 1 for x in self.stage_ends[:q_pool]]
+
+        # Transformer blocks
+        cur_stage = 0
+        depth = sum(stages)
+        dpr = [x.item() for x in torch.linspace(0, drop_path_rate, depth)]  # stochastic depth decay rule
+        self.blocks = nn.ModuleList()
+        self.feature_info = []
+        for i in range(depth):
+            dim_out = embed_dim
+            # Mask unit or global attention.
+            # Lag by 1 block, so that global attention,
+            # applied post pooling on lower resolution
+            use_mask_unit_attn = mask_unit_attn[cur_stage]
+
+            if i - 1 in self.stage_ends:
+                dim_out = int(embed_dim * dim_mul)
+                num_heads = int(num_heads * head_mul)
+                cur_stage += 1
+                if i in q_pool_blocks:
+                    flat_mu_size //= flat_q_stride
+
+            block = HieraBlock(
+                dim=embed_dim,
+                dim_out=dim_out,
+                heads=num_heads,
+                mlp_ratio=mlp_ratio,
+                drop_path=dpr[i],
+                init_values=init_values,
+                norm_layer=norm_layer,
+                q_stride=(flat_q_stride if i in q_pool_blocks else 1),
+                window_size=flat_mu_size,
+                use_expand_proj=use_expand_proj,
+                use_mask_unit_attn=use_mask_unit_attn,
+            )
+            embed_dim = dim_out
+            if i in self.stage_ends:
+                self.feature_info += [
+                    dict(num_chs=dim_out, reduction=2 ** (cur_stage + 2), module=f'blocks.{self.stage_ends[cur_stage]}')]
+            self.blocks.append(block)
+
+        self.num_features = self.head_hidden_size = embed_dim
+        self.norm = norm_layer(embed_dim)
+        self.head_drop = nn.Dropout(drop_rate)
+        self.head = nn.Linear(embed_dim, num_classes) if num_classes > 0 else nn.Identity()
+        self.global_pool = global_pool
+
+        # Initialize everything
+        if sep_pos_embed:
+            nn.init.trunc_normal_(self.pos_embed_spatial, std=0.02)
+            nn.init.trunc_normal_(self.pos_embed_temporal, std=0.02)
+        else:
+            if self.pos_embed is not None:
+                nn.init.trunc_normal_(self.pos_embed, std=0.02)
+            if self.pos_embed_win is not None:
+                nn.init.trunc_normal_(self.pos_embed_win, std=0.02)
+
+        if weight_init != 'skip':
+            init_fn = partial(self._init_weights)
+            self.apply(init_fn)
+        if isinstance(self.head, nn.Linear):
+            self.head.weight.data.mul_(head_init_scale)
+            if self.head.bias is not None:
+                self.head.bias.data.mul_(head_init_scale)
+
+    def _init_weights(self, m, init_bias=0.02):
+        if isinstance(m, (nn.Linear, nn.Conv1d, nn.Conv2d, nn.Conv3d)):
+            nn.init.trunc_normal_(m.weight, std=0.02)
+            if isinstance(m, nn.Linear) and m.bias is not None:
+                nn.init.constant_(m.bias, init_bias)
+        elif isinstance(m, nn.LayerNorm):
+            nn.init.constant_(m.bias, init_bias)
+            nn.init.constant_(m.weight, 1.0)
+
+    @torch.jit.ignore
+    def no_weight_decay(self):
+        if self.pos_embed_spatial is not None:
+            return ["pos_embed_spatial", "pos_embed_temporal"]
+        else:
+            return ["pos_embed", "pos_embed_win"]
+
+    @torch.jit.ignore
+    def group_matcher(self, coarse: bool = False) -> Dict:
+        return dict(
+            stem=r'^pos_embed|pos_embed_spatial|pos_embed_temporal|pos_embed_win|patch_embed',
+            blocks=[(r'^blocks\.(\d+)', None), (r'^norm', (99999,))]
+        )
+
+    @torch.jit.ignore
+    def set_grad_checkpointing(self, enable: bool = True) -> None:
+        self.grad_checkpointing = enable
+
+    @torch.jit.ignore
+    def get_classifier(self):
+        return self.head
+
+    def reset_classifier(self, num_classes: int, global_pool: Optional[str] = None):
+        self.num_classes = num_classes
+        if global_pool is not None:
+            self.global_pool = global_pool
+        self.head = nn.Linear(self.num_features, num_classes) if num_classes > 0 else nn.Identity()
+
+    def get_random_mask(self, x: torch.Tensor, mask_ratio: float) -> torch.Tensor:
+        """Generates a random mask, mask_ratio fraction of mask units masked out.
+
+        Shape: [B, #MUs_all]; 1 is *keep*, 0 is *remove*.
+        """
+        B = x.shape[0]
+        num_windows = math.prod(self.mask_spatial_shape)
+        len_keep = int(num_windows * (1 - mask_ratio))
+        noise = torch.rand(B, num_windows, device=x.device)
+
+        # Sort noise for each sample
+        ids_shuffle = torch.argsort(noise, dim=1)  # ascend: small is keep, large is remove
+        ids_restore = torch.argsort(ids_shuffle, dim=1)
+
+        # Generate the binary mask: 1 is *keep*, 0 is *remove*
+        # Note this is opposite to original MAE
+        mask = torch.zeros([B, num_windows], device=x.device)
+        mask[:, :len_keep] = 1
+        # Unshuffle to get the binary mask
+        mask = torch.gather(mask, dim=1, index=ids_restore)
+
+        return mask.bool()
+
+    def _pos_embed(self, x) -> torch.Tensor:
+        if self.pos_embed_win is not None:
+            # absolute win position embedding, from
+            # Window Attention is Bugged: How not to Interpolate Position Embeddings (https://arxiv.org/abs/2311.05613)
+            pos_embed_win = self.pos_embed_win.tile(self.mask_spatial_shape)
+            pos_embed = F.interpolate(
+                self.pos_embed,
+                size=pos_embed_win.shape[-2:],
+                mode='bicubic',
+                antialias=True,
+            )
+            pos_embed = pos_embed + pos_embed_win
+            pos_embed = pos_embed.flatten(2).transpose(1, 2)
+        elif self.pos_embed is not None:
+            pos_embed = self.pos_embed
+        else:
+            pos_embed = (
+                self.pos_embed_spatial.repeat(1, self.tokens_spatial_shape[0], 1)
+                +
+                torch.repeat_interleave(
+                    self.pos_embed_temporal,
+                    self.tokens_spatial_shape[1] * self.tokens_spatial_shape[2],
+                    dim=1,
+                )
+            )
+        x = x + pos_embed
+        return x
+
+    def forward_intermediates(
+            self,
+            x: torch.Tensor,
+            mask: Optional[torch.Tensor] = None,
+            indices: Optional[Union[int, List[int]]] = None,
+            norm: bool = False,
+            stop_early: bool = True,
+            output_fmt: str = 'NCHW',
+            intermediates_only: bool = False,
+            coarse: bool = True,
+    ) -> Union[List[torch.Tensor], Tuple[torch.Tensor, List[torch.Tensor]]]:
+        assert not norm, 'normalization of features not supported'
+        assert output_fmt in ('NCHW', 'NHWC'), 'Output format must be one of NCHW, NHWC.'
+        if coarse:
+            take_indices, max_index = feature_take_indices(len(self.stage_ends), indices)
+            take_indices = [self.stage_ends[i] for i in take_indices]
+            max_index = self.stage_ends[max_index]
+        else:
+            take_indices, max_index = feature_take_indices(len(self.blocks), indices)
+
+        if mask is not None:
+            patch_mask = mask.view(x.shape[0], 1, *self.mask_spatial_shape)  # B, C, *mask_spatial_shape
+        else:
+            patch_mask = None
+        x = self.patch_embed(x, mask=patch_mask)
+        x = self._pos_embed(x)
+        x = self.unroll(x)
+
+        # Discard masked tokens
+        if mask is not None:
+            x = x[mask[..., None].tile(1, self.mu_size, x.shape[2])].view(x.shape[0], -1, x.shape[-1])
+
+        intermediates = []
+        if torch.jit.is_scripting() or not stop_early:  # can't slice blocks in torchscript
+            blocks = self.blocks
+        else:
+            blocks = self.blocks[:max_index + 1]
+        for i, blk in enumerate(blocks):
+            x = blk(x)
+            if i in take_indices:
+                x_int = self.reroll(x, i, mask=mask)
+                intermediates.append(x_int.permute(0, 3, 1, 2) if output_fmt == 'NCHW' else x_int)
+
+        if intermediates_only:
+            return intermediates
+
+        return x, intermediates
+
+    def prune_intermediate_layers(
+            self,
+            indices: Union[int, List[int]] = 1,
+            prune_norm: bool = False,
+            prune_head: bool = True,
+            coarse: bool = True,
+    ):
+        if coarse:
+            take_indices, max_index = feature_take_indices(len(self.stage_ends), indices)
+            max_index = self.stage_ends[max_index]
+        else:
+            take_indices, max_index = feature_take_indices(len(self.blocks), indices)
+        self.blocks = self.blocks[:max_index + 1]  # truncate blocks
+        if prune_head:
+            self.head = nn.Identity()
+        return take_indices
+
+    def forward_features(
+            self,
+            x: torch.Tensor,
+            mask: Optional[torch.Tensor] = None,
+            return_intermediates: bool = False,
+    ) -> torch.Tensor:
+        """mask should be a boolean tensor of shape [B, #MUt*#MUy*#MUx] where
+        #MU are the number of mask units in that dim."""
+        if self.training and self.patch_drop_rate > 0:
+            # using mask for something like 'patch dropout' via mask-units in supervised train / fine-tune
+            assert mask is None
+            mask = self.get_random_mask(x, mask_ratio=self.patch_drop_rate)
+
+        if mask is not None:
+            patch_mask = mask.view(x.shape[0], 1, *self.mask_spatial_shape)  # B, C, *mask_spatial_shape
+        else:
+            patch_mask = None
+        x = self.patch_embed(x, mask=patch_mask)
+        x = self._pos_embed(x)
+        x = self.unroll(x)
+
+        # Discard masked tokens
+        if mask is not None:
+            x = x[mask[..., None].tile(1, self.mu_size, x.shape[2])].view(x.shape[0], -1, x.shape[-1])
+
+        intermediates = []
+        for i, blk in enumerate(self.blocks):
+            if self.grad_checkpointing and not torch.jit.is_scripting():
+                x = checkpoint(blk, x)
+            else:
+                x = blk(x)
+            if return_intermediates and i in self.stage_ends:
+                intermediates.append(self.reroll(x, i, mask=mask))
+
+        # x may not always be in spatial order here.
+        # e.g. if q_pool = 2, mask_unit_size = (8, 8), and
+        # q_stride = (2, 2), not all unrolls were consumed,
+        # intermediates[-1] is x in spatial order
+        if return_intermediates:
+            return x, intermediates
+
+        return x
+
+    def forward_head(self, x, pre_logits: bool = False) -> torch.Tensor:
+        if self.global_pool == 'avg':
+            x = x.mean(dim=1)
+        x = self.norm(x)
+        x = self.head_drop(x)
+        x = x if pre_logits else self.head(x)
+        return x
+
+    def forward(
+            self,
+            x: torch.Tensor,
+            mask: Optional[torch.Tensor] = None,
+    ) -> torch.Tensor:
+        x = self.forward_features(x, mask=mask)
+        if mask is None:
+            x = self.forward_head(x)
+        return x
+
+
+def _cfg(url='', **kwargs):
+    return {
+        'url': url,
+        'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': None,
+        'crop_pct': 0.9, 'interpolation': 'bicubic', 'fixed_input_size': True,
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'patch_embed.proj', 'classifier': 'head',
+        **kwargs,
+    }
+
+
+default_cfgs = generate_default_cfgs({
+    'hiera_tiny_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_small_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_base_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_base_plus_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_large_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_huge_224.mae_in1k_ft_in1k': _cfg(),
+    'hiera_small_abswin_256.sbb2_e200_in12k_ft_in1k': _cfg(input_size=(3, 256, 256), crop_pct=0.95),
+})
+
+
+def checkpoint_filter_fn(state_dict, model=None):
+    state_dict = state_dict.get('model_state', state_dict)
+    state_dict = state_dict.get('model', state_dict)
+    output = {}
+    for k, v in state_dict.items():
+        if k == 'pos_embed' and v.ndim == 3 and model.pos_embed is not None and model.pos_embed.ndim == 3 \
+                and v.shape[1] != model.pos_embed.shape[1]:
+            # positional embedding resize via 2d interp
+            num_new = model.pos_embed.shape[1]
+            g_old = int(math.sqrt(v.shape[1]))
+            g_new = int(math.sqrt(num_new))
+            v = F.interpolate(
+                v.transpose(1, 2).reshape(1, -1, g_old, g_old),
+                (g_new, g_new), mode='bicubic', antialias=True,
+            ).flatten(2).transpose(1, 2)
+        k = k.replace('encoder_norm', 'norm')
+        output[k] = v
+    return output
+
+
+def _create_hiera(variant: str, pretrained: bool = False, **kwargs) -> Hiera:
+    out_indices = kwargs.pop('out_indices', 4)
+    return build_model_with_cfg(
+        Hiera,
+        variant,
+        pretrained,
+        pretrained_filter_fn=checkpoint_filter_fn,
+        feature_cfg=dict(out_indices=out_indices, feature_cls='getter'),
+        **kwargs,
+    )
+
+
+@register_model
+def hiera_tiny_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=96, num_heads=1, stages=(1, 2, 7, 2))
+    return _create_hiera('hiera_tiny_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_small_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=96, num_heads=1, stages=(1, 2, 11, 2))
+    return _create_hiera('hiera_small_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_base_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=96, num_heads=1, stages=(2, 3, 16, 3))
+    return _create_hiera('hiera_base_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_base_plus_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=112, num_heads=2, stages=(2, 3, 16, 3))
+    return _create_hiera('hiera_base_plus_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_large_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=144, num_heads=2, stages=(2, 6, 36, 4))
+    return _create_hiera('hiera_large_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_huge_224(pretrained=False, **kwargs):
+    model_args = dict(embed_dim=256, num_heads=4, stages=(2, 6, 36, 4))
+    return _create_hiera('hiera_huge_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_small_abswin_256(pretrained=False, **kwargs):
+    model_args = dict(
+        embed_dim=96, num_heads=1, stages=(1, 2, 11, 2), abs_win_pos_embed=True,
+        init_values=1e-5, weight_init='jax', use_expand_proj=False,
+    )
+    return _create_hiera('hiera_small_abswin_256', pretrained=pretrained, **dict(model_args, **kwargs))
