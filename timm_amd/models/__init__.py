@@ -34,7 +34,9 @@ from .convnext import *
 from .efficientnet import *
 from .eva import *
 from .hiera import *
+from .maxxvit import *
 from .mobilenetv3 import *
 from .naflexvit import *
+from .swin_transformer import *
 from .resnet import *
 from .vision_transformer import *
