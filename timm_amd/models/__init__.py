@@ -31,10 +31,12 @@ from ._registry import (
 
 # architecture modules (registration happens at import time)
 from .convnext import *
+from .deit import *
 from .efficientnet import *
 from .eva import *
 from .hiera import *
 from .maxxvit import *
+from .mlp_mixer import *
 from .mobilenetv3 import *
 from .naflexvit import *
 from .swin_transformer import *
