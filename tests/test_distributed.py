@@ -131,3 +131,31 @@ def test_bucketed_ddp_no_sync_accumulation():
 
 def test_reduce_tensor():
     _spawn(_run_utils, 29513)
+
+
+def test_train_script_distributed(tmp_path):
+    """Full train.py end-to-end under torch.distributed.run (gloo, ws=2, CPU):
+    synthetic data, 1 epoch train + validate + checkpoint save + summary."""
+    import json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cmd = [
+        sys.executable, '-m', 'torch.distributed.run',
+        '--nnodes=1', '--nproc-per-node', '2',
+        '--master-addr', '127.0.0.1', '--master-port', '29521',
+        os.path.join(repo, 'train.py'),
+        '--model', 'resnet18', '--synthetic', '--synthetic-len', '32', '-b', '8',
+        '--epochs', '1', '--opt', 'sgd', '--lr', '0.1', '--sched', 'none',
+        '--no-prefetcher', '--workers', '0', '--device', 'cpu',
+        '--output', str(tmp_path), '--experiment', 'smoke',
+    ]
+    r = subprocess.run(cmd, capture_output=True, text=True, timeout=300, cwd=repo)
+    assert r.returncode == 0, f'train.py failed:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}'
+    out = tmp_path / 'smoke'
+    assert (out / 'last.pth.tar').exists()
+    summary = (out / 'summary.csv')
+    assert summary.exists()
+    ckpt = torch.load(out / 'last.pth.tar', map_location='cpu', weights_only=False)
+    assert ckpt['arch'] == 'resnet18'
+    assert 'state_dict' in ckpt and 'optimizer' in ckpt
