@@ -200,6 +200,47 @@ def test_attention_mask():
     assert rel_err(o, o_ref) < 2e-2
 
 
+def test_attention_perhead_bias():
+    """Per-head rel-pos bias [1,H,N,N] and window-cyclic mask [mB,H,N,N] with
+    mB dividing B (Swin shift masks) both run through the fused kernel."""
+    _ext()
+    torch.manual_seed(7)
+    B, H, N, D = 4, 4, 49, 32
+    q = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    k = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    v = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+
+    for mB in (1, 2, B):  # broadcast / cyclic (b % mB) / per-batch
+        bias = torch.randn(mB, H, N, N, device='cuda') * 2.0
+        q1, k1, v1 = [t.clone().requires_grad_(True) for t in (q, k, v)]
+        o = ops.flash_attention(q1, k1, v1, attn_mask=bias)
+        do = torch.randn_like(o)
+        o.backward(do)
+
+        q2, k2, v2 = [t.detach().float().requires_grad_(True) for t in (q, k, v)]
+        full = bias.repeat(B // mB, 1, 1, 1)
+        attn = (q2 @ k2.transpose(-2, -1)) / math.sqrt(D) + full
+        o_ref = attn.softmax(-1) @ v2
+        o_ref.backward(do.float())
+
+        assert rel_err(o, o_ref) < 2e-2, f'mB={mB} fwd err {rel_err(o, o_ref)}'
+        assert rel_err(q1.grad, q2.grad) < 5e-2, f'mB={mB}'
+        assert rel_err(k1.grad, k2.grad) < 5e-2, f'mB={mB}'
+        assert rel_err(v1.grad, v2.grad) < 5e-2, f'mB={mB}'
+
+
+def test_swin_gpu_matches_cpu():
+    torch.manual_seed(8)
+    m = timm_amd.create_model('swin_tiny_patch4_window7_224', num_classes=10)
+    m.eval()
+    x = torch.randn(2, 3, 224, 224)
+    with torch.no_grad():
+        ref = m(x)
+        out = m.to('cuda', torch.bfloat16)(x.to('cuda', torch.bfloat16))
+    err = (out.float().cpu() - ref).abs().max().item()
+    assert err < 0.35, f'swin gpu/cpu mismatch {err}'
+
+
 def test_fused_adamw_matches_reference():
     _ext()
     torch.manual_seed(6)

@@ -13,6 +13,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
+from ..ops import flash_attention
+from ..ops.attention import attention_available
 from ..layers import (
     ClassifierHead, DropPath, Mlp, PatchEmbed, _assert, calculate_drop_path_rates, get_norm_layer,
     to_2tuple, to_ntuple, trunc_normal_, use_fused_attn,
@@ -108,19 +110,26 @@ class WindowAttention(nn.Module):
         qkv = self.qkv(x).reshape(B_, N, 3, self.num_heads, -1).permute(2, 0, 3, 1, 4)
         q, k, v = qkv.unbind(0)
 
-        # always the unfused/math composition here: the additive bias differs per
-        # window and per head ([1,nH,N,N] + shift mask) and N is small (<=64),
-        # so the batched GEMMs via hipBLASLt are already efficient
-        q = q * self.scale
-        attn = q @ k.transpose(-2, -1)
-        attn = attn + self._get_rel_pos_bias()
-        if mask is not None:
-            num_win = mask.shape[0]
-            attn = attn.view(-1, num_win, self.num_heads, N, N) + mask.unsqueeze(1).unsqueeze(0)
-            attn = attn.view(-1, self.num_heads, N, N)
-        attn = self.softmax(attn)
-        attn = self.attn_drop(attn)
-        x = attn @ v
+        if self.fused_attn and attention_available(q) and self.attn_drop.p == 0.:
+            # fused flash path: the kernel's mask batch index is (b % mask_B),
+            # so the per-window shift mask combines with the per-head rel-pos
+            # bias into one small [num_win, nH, N, N] tensor — no expansion to
+            # the full B*num_win batch.
+            bias = self._get_rel_pos_bias()  # [1, nH, N, N]
+            if mask is not None:
+                bias = bias + mask.unsqueeze(1)  # [num_win, nH, N, N]
+            x = flash_attention(q, k, v, attn_mask=bias, scale=self.scale)
+        else:
+            q = q * self.scale
+            attn = q @ k.transpose(-2, -1)
+            attn = attn + self._get_rel_pos_bias()
+            if mask is not None:
+                num_win = mask.shape[0]
+                attn = attn.view(-1, num_win, self.num_heads, N, N) + mask.unsqueeze(1).unsqueeze(0)
+                attn = attn.view(-1, self.num_heads, N, N)
+            attn = self.softmax(attn)
+            attn = self.attn_drop(attn)
+            x = attn @ v
 
         x = x.transpose(1, 2).reshape(B_, N, -1)
         x = self.proj(x)

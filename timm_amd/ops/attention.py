@@ -99,7 +99,13 @@ def flash_attention(
         return _dropout_sdpa(q, k, v, attn_mask, dropout_p, scale)
     if attention_available(q):
         if attn_mask is not None:
-            attn_mask = attn_mask.expand(q.shape[0], 1, q.shape[2], k.shape[2])
+            # kernel accepts [B|1, H|1, Nq, Nk]: keep batch/head dims unexpanded
+            # (rel-pos bias is [1,H,N,N], padding masks [B,1,N,N]) so the
+            # contiguous copy below stays small.
+            while attn_mask.dim() < 4:
+                attn_mask = attn_mask.unsqueeze(0)
+            attn_mask = attn_mask.expand(
+                attn_mask.shape[0], attn_mask.shape[1], q.shape[2], k.shape[2])
         return _FlashAttnFn.apply(q, k, v, attn_mask, scale)
     if q.is_cuda:
         from . import use_hip

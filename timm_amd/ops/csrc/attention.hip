@@ -40,10 +40,13 @@ void attn_fwd_kernel(
     const __bf16* __restrict__ q,     // [B,H,Nq,D] via strides (last dim contiguous)
     const __bf16* __restrict__ k,
     const __bf16* __restrict__ v,
-    const float* __restrict__ mask,   // [B,1,Nq,Nk] or null
+    const float* __restrict__ mask,   // [B|1,H|1,Nq,Nk] via m_sb/m_sh, or null
     __bf16* __restrict__ o,           // written [B,Nq,H,D] (BNHD) so proj reshape is free
     float* __restrict__ lse,          // [B,H,Nq]
     int B, int H, int Nq, int Nk, int D, float scale,
+    // mask batch index is (b % mB): mB==1 broadcasts, mB==B is per-batch, and
+    // mB==num_windows serves window-cyclic masks (Swin shift) for free.
+    int mB, long m_sb, long m_sh,
     long q_sb, long q_sh, long q_sn,
     long k_sb, long k_sh, long k_sn,
     long v_sb, long v_sh, long v_sn) {
@@ -138,7 +141,7 @@ void attn_fwd_kernel(
         if (kvcol >= Nk || qrow >= Nq) {
           s = kNegInf;
         } else if (kHasMask) {
-          float mv = mask[((long)b * Nq + qrow) * Nk + kvcol];
+          float mv = mask[(long)(b % mB) * m_sb + (long)h * m_sh + (long)qrow * Nk + kvcol];
           s += mv;
           if (s < kNegInf) s = kNegInf;
         }
@@ -222,10 +225,18 @@ void launch_attn_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor&
   dim3 block(kBlockThreads);
   const bool has_mask = mask.has_value();
   const float* mp = has_mask ? mask->data_ptr<float>() : nullptr;
+  int mB = 1;
+  long m_sb = 0, m_sh = 0;
+  if (has_mask) {
+    mB = (int)mask->size(0);
+    m_sb = mB == 1 ? 0 : mask->size(1) * (long)Nq * Nk;
+    m_sh = mask->size(1) == 1 ? 0 : (long)Nq * Nk;
+  }
   auto args = [&](auto kern) {
     hipLaunchKernelGGL(kern, grid, block, 0, stream,
         (const __bf16*)q.data_ptr(), (const __bf16*)k.data_ptr(), (const __bf16*)v.data_ptr(),
         mp, (__bf16*)o.data_ptr(), lse.data_ptr<float>(), B, H, Nq, Nk, D, scale,
+        mB, m_sb, m_sh,
         q.stride(0), q.stride(1), q.stride(2),
         k.stride(0), k.stride(1), k.stride(2),
         v.stride(0), v.stride(1), v.stride(2));
@@ -253,9 +264,10 @@ __global__ __launch_bounds__(256)
 void attn_bwd_softmax_kernel(
     const __bf16* __restrict__ s,    // [B,H,Nq,Nk]
     const float* __restrict__ lse,   // [B,H,Nq]
-    const float* __restrict__ mask,  // [B,1,Nq,Nk] or null
+    const float* __restrict__ mask,  // [B|1,H|1,Nq,Nk] via m_sb/m_sh, or null
     __bf16* __restrict__ p,
-    long rows, int Nk, int Nq, int H, float scale) {
+    long rows, int Nk, int Nq, int H, float scale,
+    int mB, long m_sb, long m_sh) {
   const int wave = threadIdx.x / WAVE_SIZE;
   const int lane = threadIdx.x % WAVE_SIZE;
   constexpr int kWaves = 256 / WAVE_SIZE;
@@ -267,8 +279,9 @@ void attn_bwd_softmax_kernel(
     const float* mrow = nullptr;
     if (kHasMask) {
       long q = row % Nq;
+      long h = (row / Nq) % H;
       long b = row / ((long)H * Nq);
-      mrow = mask + (b * Nq + q) * Nk;
+      mrow = mask + (b % mB) * m_sb + h * m_sh + q * Nk;
     }
     int nvec = (Nk % 8 == 0) ? Nk / 8 : 0;
     for (int i = lane; i < nvec; i += WAVE_SIZE) {
@@ -335,13 +348,16 @@ at::Tensor attn_bwd_softmax(at::Tensor s, at::Tensor lse, c10::optional<at::Tens
   auto stream = at::hip::getCurrentHIPStream();
   int blocks = (int)std::min((long)4096, (rows + 3) / 4);
   if (mask.has_value()) {
+    int mB = (int)mask->size(0);
+    long m_sb = mB == 1 ? 0 : mask->size(1) * (long)Nq * Nk;
+    long m_sh = mask->size(1) == 1 ? 0 : (long)Nq * Nk;
     hipLaunchKernelGGL((attn_bwd_softmax_kernel<true>), dim3(blocks), dim3(256), 0, stream,
         (const __bf16*)s.data_ptr(), lse.data_ptr<float>(), mask->data_ptr<float>(),
-        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale);
+        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale, mB, m_sb, m_sh);
   } else {
     hipLaunchKernelGGL((attn_bwd_softmax_kernel<false>), dim3(blocks), dim3(256), 0, stream,
         (const __bf16*)s.data_ptr(), lse.data_ptr<float>(), nullptr,
-        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale);
+        (__bf16*)p.data_ptr(), rows, Nk, Nq, H, (float)scale, 1, 0, 0);
   }
   HIP_CHECK_LAST();
   return p;
@@ -372,7 +388,10 @@ std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   TORCH_CHECK(D % 32 == 0 && D <= 128, "attention_fwd: head_dim must be multiple of 32, <=128");
   if (mask.has_value()) {
     TORCH_CHECK(mask->is_contiguous() && mask->scalar_type() == at::kFloat);
-    TORCH_CHECK(mask->size(0) == B && mask->size(2) == Nq && mask->size(3) == Nk);
+    TORCH_CHECK(mask->size(0) >= 1 && B % mask->size(0) == 0 &&
+                (mask->size(1) == H || mask->size(1) == 1) &&
+                mask->size(2) == Nq && mask->size(3) == Nk,
+                "attention_fwd: mask must be [mB|1,H|1,Nq,Nk] with mB dividing B");
   }
   // O allocated [B, Nq, H, D] and returned as a permuted [B,H,Nq,D] view so
   // the caller's transpose(1,2).reshape(B,N,C) is a zero-copy reshape.
