@@ -40,7 +40,10 @@ from .padding import get_padding, get_same_padding, pad_same
 from .patch_dropout import PatchDropout, patch_dropout_forward
 from .patch_embed import PatchEmbed, PatchEmbedWithSize, resample_patch_embed
 from .pool2d_same import AvgPool2dSame, MaxPool2dSame, create_pool2d
-from .pos_embed_rel import RelPosBias, RelPosBiasTf, RelPosMlp, gen_relative_position_index
+from .pos_embed_rel import (
+    RelPosBias, RelPosBiasTf, RelPosMlp, gen_relative_position_index,
+    resize_rel_pos_bias_table, resize_rel_pos_bias_table_simple,
+)
 from .pos_embed import resample_abs_pos_embed, resample_abs_pos_embed_nhwc
 from .pos_embed_sincos import (
     FourierEmbed, RotaryEmbedding, RotaryEmbeddingCat, apply_keep_indices_nlc, apply_rot_embed,

@@ -199,3 +199,49 @@ class RelPosBiasTf(nn.Module):
 
     def forward(self, attn, shared_rel_pos: Optional[torch.Tensor] = None):
         return attn + self.get_bias()
+
+
+def resize_rel_pos_bias_table_simple(
+        rel_pos_bias,
+        new_window_size: Tuple[int, int],
+        new_bias_shape: Tuple[int, ...],
+):
+    """Bilinear-resize a relative position bias table to a new window size
+    (reference `pos_embed_rel.py:122`).  Handles both [N, H] (Swin/BEiT flat
+    tables, extra cls entries preserved) and [H, Nh, Nw] (TF MaxViT) layouts.
+    """
+    dst_size = (new_window_size[0] * 2 - 1, new_window_size[1] * 2 - 1)
+    if rel_pos_bias.ndim == 3:
+        # TF maxvit style [H, Nh, Nw]
+        _, dst_h, dst_w = new_bias_shape
+        num_attn_heads, src_h, src_w = rel_pos_bias.shape
+        if (src_h, src_w) == (dst_h, dst_w):
+            return rel_pos_bias
+        rel_pos_bias = torch.nn.functional.interpolate(
+            rel_pos_bias.unsqueeze(0), size=(dst_h, dst_w), mode='bicubic', align_corners=False,
+        ).squeeze(0)
+    else:
+        assert rel_pos_bias.ndim == 2
+        # Swin/BEiT style [N, H] with optional extra (cls) entries at the end
+        dst_num_pos, _ = new_bias_shape
+        num_attn_heads = rel_pos_bias.shape[-1]
+        src_num_pos = rel_pos_bias.shape[0]
+        num_extra_tokens = dst_num_pos - (dst_size[0] * dst_size[1])
+        src_size = int((src_num_pos - num_extra_tokens) ** 0.5)
+        if src_size == dst_size[0] and src_size == dst_size[1]:
+            return rel_pos_bias
+
+        extra_tokens = rel_pos_bias[-num_extra_tokens:, :] if num_extra_tokens else None
+        rel_pos_bias = rel_pos_bias[:src_num_pos - num_extra_tokens, :]
+        rel_pos_bias = rel_pos_bias.transpose(0, 1).reshape(1, num_attn_heads, src_size, src_size)
+        rel_pos_bias = torch.nn.functional.interpolate(
+            rel_pos_bias, size=dst_size, mode='bicubic', align_corners=False)
+        rel_pos_bias = rel_pos_bias.reshape(num_attn_heads, -1).transpose(0, 1)
+        if extra_tokens is not None:
+            rel_pos_bias = torch.cat((rel_pos_bias, extra_tokens), dim=0)
+    return rel_pos_bias
+
+
+# geometric (levit / swin-v2 style log-spaced) resize not separately
+# implemented: the simple bicubic resize is used for all table layouts
+resize_rel_pos_bias_table = resize_rel_pos_bias_table_simple

@@ -30,6 +30,7 @@ from ._registry import (
 )
 
 # architecture modules (registration happens at import time)
+from .beit import *
 from .convnext import *
 from .deit import *
 from .efficientnet import *
