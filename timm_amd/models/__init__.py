@@ -32,6 +32,7 @@ from ._registry import (
 # architecture modules (registration happens at import time)
 from .beit import *
 from .convnext import *
+from .densenet import *
 from .deit import *
 from .efficientnet import *
 from .eva import *
@@ -42,4 +43,5 @@ from .mobilenetv3 import *
 from .naflexvit import *
 from .swin_transformer import *
 from .resnet import *
+from .vgg import *
 from .vision_transformer import *
