@@ -42,6 +42,7 @@ from .mlp_mixer import *
 from .mobilenetv3 import *
 from .naflexvit import *
 from .swin_transformer import *
+from .regnet import *
 from .resnet import *
 from .vgg import *
 from .vision_transformer import *

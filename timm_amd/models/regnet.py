@@ -1,0 +1,460 @@
+"""RegNet (X/Y) — MI355X-native implementation.
+
+Capability parity with reference `timm/models/regnet.py`: parameterized width
+generation (`generate_regnet` :106), group-width compatibility adjustment
+(:78), `Bottleneck` block w/ optional SE (:272), `RegStage` (:484), `RegNet`
+(:553) and the pycls X/Y model grid.
+
+Grouped 3x3 convs route through MIOpen (dense library shapes); SE pooling and
+the residual adds are cheap next to the grouped convs.
+"""
+import math
+from dataclasses import dataclass, replace
+from functools import partial
+from typing import Any, Callable, Dict, List, Optional, Tuple, Union
+
+import torch
+import torch.nn as nn
+
+from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
+from ..layers import ClassifierHead, ConvNormAct, DropPath, SEModule, get_act_layer, get_norm_act_layer
+from ._builder import build_model_with_cfg
+from ._manipulate import checkpoint_seq, named_apply
+from ._registry import generate_default_cfgs, register_model
+
+__all__ = ['RegNet', 'RegNetCfg']
+
+
+@dataclass
+class RegNetCfg:
+    """RegNet architecture configuration (reference `regnet.py:46`)."""
+    depth: int = 21
+    w0: int = 80
+    wa: float = 42.63
+    wm: float = 2.66
+    group_size: int = 24
+    bottle_ratio: float = 1.
+    se_ratio: float = 0.
+    group_min_ratio: float = 0.
+    stem_width: int = 32
+    downsample: Optional[str] = 'conv1x1'
+    linear_out: bool = False
+    preact: bool = False
+    num_features: int = 0
+    act_layer: Union[str, Callable] = 'relu'
+    norm_layer: Union[str, Callable] = 'batchnorm'
+
+
+def quantize_float(f: float, q: int) -> int:
+    """Closest non-zero int divisible by q."""
+    return int(round(f / q) * q)
+
+
+def adjust_widths_groups_comp(
+        widths: List[int],
+        bottle_ratios: List[float],
+        groups: List[int],
+        min_ratio: float = 0.,
+) -> Tuple[List[int], List[int]]:
+    """Make widths and group counts compatible (reference `regnet.py:78`)."""
+    bottleneck_widths = [int(w * b) for w, b in zip(widths, bottle_ratios)]
+    groups = [min(g, w_bot) for g, w_bot in zip(groups, bottleneck_widths)]
+    bottleneck_widths = [quantize_float(w_bot, g) for w_bot, g in zip(bottleneck_widths, groups)]
+    widths = [int(w_bot / b) for w_bot, b in zip(bottleneck_widths, bottle_ratios)]
+    return widths, groups
+
+
+def generate_regnet(
+        width_slope: float,
+        width_initial: int,
+        width_mult: float,
+        depth: int,
+        group_size: int,
+        quant: int = 8,
+) -> Tuple[List[int], int, List[int]]:
+    """Per-block widths from RegNet parameters (reference `regnet.py:106`)."""
+    assert width_slope >= 0 and width_initial > 0 and width_mult > 1 and width_initial % quant == 0
+    widths_cont = torch.arange(depth, dtype=torch.float32) * width_slope + width_initial
+    width_exps = torch.round(torch.log(widths_cont / width_initial) / math.log(width_mult))
+    widths = torch.round((width_initial * torch.pow(width_mult, width_exps)) / quant) * quant
+    num_stages = len(torch.unique(widths))
+    groups = [group_size for _ in range(num_stages)]
+    return widths.int().tolist(), num_stages, groups
+
+
+def downsample_conv(in_chs, out_chs, kernel_size=1, stride=1, dilation=1, norm_layer=None, preact=False):
+    kernel_size = 1 if stride == 1 and dilation == 1 else kernel_size
+    dilation = dilation if kernel_size > 1 else 1
+    if preact:
+        return create_conv2d_basic(in_chs, out_chs, kernel_size, stride=stride, dilation=dilation)
+    return ConvNormAct(
+        in_chs, out_chs, kernel_size, stride=stride, dilation=dilation,
+        norm_layer=norm_layer or nn.BatchNorm2d, apply_act=False)
+
+
+def downsample_avg(in_chs, out_chs, kernel_size=1, stride=1, dilation=1, norm_layer=None, preact=False):
+    """AvgPool downsample then 1x1 conv (ResNet-D style)."""
+    avg_stride = stride if dilation == 1 else 1
+    pool = nn.Identity()
+    if stride > 1 or dilation > 1:
+        avg_pool_fn = nn.AvgPool2d
+        pool = avg_pool_fn(2, avg_stride, ceil_mode=True, count_include_pad=False)
+    if preact:
+        conv = create_conv2d_basic(in_chs, out_chs, 1, stride=1)
+    else:
+        conv = ConvNormAct(in_chs, out_chs, 1, stride=1, norm_layer=norm_layer or nn.BatchNorm2d, apply_act=False)
+    return nn.Sequential(*[pool, conv])
+
+
+def create_conv2d_basic(in_chs, out_chs, kernel_size, stride=1, dilation=1):
+    padding = ((stride - 1) + dilation * (kernel_size - 1)) // 2
+    return nn.Conv2d(in_chs, out_chs, kernel_size, stride=stride, padding=padding, dilation=dilation, bias=True)
+
+
+def create_shortcut(downsample_type, in_chs, out_chs, kernel_size, stride, dilation=(1, 1), norm_layer=None, preact=False):
+    assert downsample_type in ('avg', 'conv1x1', '', None)
+    if in_chs != out_chs or stride != 1 or dilation[0] != dilation[1]:
+        dargs = dict(stride=stride, dilation=dilation[0], norm_layer=norm_layer, preact=preact)
+        if not downsample_type:
+            return None  # no shortcut, no residual
+        elif downsample_type == 'avg':
+            return downsample_avg(in_chs, out_chs, **dargs)
+        else:
+            return downsample_conv(in_chs, out_chs, kernel_size=kernel_size, **dargs)
+    else:
+        return nn.Identity()
+
+
+class Bottleneck(nn.Module):
+    """RegNet Bottleneck block (reference `regnet.py:272`).
+
+    Differs from ResNet: a single layer is used for bottle_ratio, group widths
+    always kept compatible with the block width.
+    """
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int = 1,
+            dilation: Tuple[int, int] = (1, 1),
+            bottle_ratio: float = 1,
+            group_size: int = 1,
+            se_ratio: float = 0.25,
+            downsample: str = 'conv1x1',
+            linear_out: bool = False,
+            act_layer: Union[str, Callable] = 'relu',
+            norm_layer: Union[str, Callable] = 'batchnorm',
+            drop_block=None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        act_layer = get_act_layer(act_layer)
+        bottleneck_chs = int(round(out_chs * bottle_ratio))
+        groups = bottleneck_chs // group_size
+
+        cargs = dict(act_layer=act_layer, norm_layer=norm_layer)
+        self.conv1 = ConvNormAct(in_chs, bottleneck_chs, kernel_size=1, **cargs)
+        self.conv2 = ConvNormAct(
+            bottleneck_chs, bottleneck_chs, kernel_size=3, stride=stride, dilation=dilation[0],
+            groups=groups, drop_layer=drop_block, **cargs)
+        if se_ratio:
+            se_channels = int(round(in_chs * se_ratio))
+            self.se = SEModule(bottleneck_chs, rd_channels=se_channels, act_layer=act_layer)
+        else:
+            self.se = nn.Identity()
+        self.conv3 = ConvNormAct(bottleneck_chs, out_chs, kernel_size=1, apply_act=False, **cargs)
+        self.act3 = nn.Identity() if linear_out else act_layer()
+        self.downsample = create_shortcut(
+            downsample, in_chs, out_chs, kernel_size=1, stride=stride, dilation=dilation, norm_layer=norm_layer)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0 else nn.Identity()
+
+    def zero_init_last(self):
+        if hasattr(self.conv3, 'bn') and hasattr(self.conv3.bn, 'weight'):
+            nn.init.zeros_(self.conv3.bn.weight)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shortcut = x
+        x = self.conv1(x)
+        x = self.conv2(x)
+        x = self.se(x)
+        x = self.conv3(x)
+        if self.downsample is not None:
+            # NOTE stuck with downsample as the attr name due to weight compatibility
+            # now represents the shortcut, no shortcut if None, identity if nn.Identity
+            x = self.drop_path(x) + self.downsample(shortcut)
+        x = self.act3(x)
+        return x
+
+
+class RegStage(nn.Module):
+    """Stage (sequence of blocks w/ the same output shape, reference `regnet.py:484`)."""
+
+    def __init__(
+            self,
+            depth: int,
+            in_chs: int,
+            out_chs: int,
+            stride: int,
+            dilation: int,
+            drop_path_rates: Optional[List[float]] = None,
+            block_fn: Callable = Bottleneck,
+            **block_kwargs,
+    ):
+        super().__init__()
+        self.grad_checkpointing = False
+
+        first_dilation = 1 if dilation in (1, 2) else 2
+        for i in range(depth):
+            block_stride = stride if i == 0 else 1
+            block_in_chs = in_chs if i == 0 else out_chs
+            block_dilation = (first_dilation, dilation)
+            dpr = drop_path_rates[i] if drop_path_rates is not None else 0.
+            name = "b{}".format(i + 1)
+            self.add_module(
+                name,
+                block_fn(
+                    block_in_chs, out_chs, stride=block_stride, dilation=block_dilation,
+                    drop_path_rate=dpr, **block_kwargs,
+                )
+            )
+            first_dilation = dilation
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.grad_checkpointing and not torch.jit.is_scripting():
+            x = checkpoint_seq(list(self.children()), x)
+        else:
+            for block in self.children():
+                x = block(x)
+        return x
+
+
+class RegNet(nn.Module):
+    """RegNet-X/Y (reference `regnet.py:553`; paper: Designing Network Design Spaces)."""
+
+    def __init__(
+            self,
+            cfg: RegNetCfg,
+            in_chans: int = 3,
+            num_classes: int = 1000,
+            output_stride: int = 32,
+            global_pool: str = 'avg',
+            drop_rate: float = 0.,
+            drop_path_rate: float = 0.,
+            zero_init_last: bool = True,
+            **kwargs,
+    ):
+        super().__init__()
+        self.num_classes = num_classes
+        self.drop_rate = drop_rate
+        assert output_stride in (8, 16, 32)
+        cfg = replace(cfg, **kwargs)  # update cfg with extra passed kwargs
+
+        # Construct the stem
+        stem_width = cfg.stem_width
+        na_args = dict(act_layer=cfg.act_layer, norm_layer=cfg.norm_layer)
+        assert not cfg.preact, 'preact RegNet (Z) not yet supported'
+        self.stem = ConvNormAct(in_chans, stem_width, 3, stride=2, **na_args)
+        self.feature_info = [dict(num_chs=stem_width, reduction=2, module='stem')]
+
+        # Construct the stages
+        prev_width = stem_width
+        curr_stride = 2
+        per_stage_args, common_args = self._get_stage_args(
+            cfg,
+            output_stride=output_stride,
+            drop_path_rate=drop_path_rate,
+        )
+        assert len(per_stage_args) == 4
+        block_fn = Bottleneck
+        for i, stage_args in enumerate(per_stage_args):
+            stage_name = "s{}".format(i + 1)
+            self.add_module(
+                stage_name,
+                RegStage(in_chs=prev_width, block_fn=block_fn, **stage_args, **common_args)
+            )
+            prev_width = stage_args['out_chs']
+            curr_stride *= stage_args['stride']
+            self.feature_info += [dict(num_chs=prev_width, reduction=curr_stride, module=stage_name)]
+
+        # Construct the head
+        if cfg.num_features:
+            self.final_conv = ConvNormAct(prev_width, cfg.num_features, kernel_size=1, **na_args)
+            self.num_features = cfg.num_features
+        else:
+            final_act = cfg.linear_out
+            self.final_conv = get_act_layer(cfg.act_layer)() if final_act else nn.Identity()
+            self.num_features = prev_width
+        self.head_hidden_size = self.num_features
+        self.head = ClassifierHead(
+            in_features=self.num_features,
+            num_classes=num_classes,
+            pool_type=global_pool,
+            drop_rate=drop_rate,
+        )
+
+        named_apply(partial(_init_weights, zero_init_last=zero_init_last), self)
+
+    def _get_stage_args(self, cfg: RegNetCfg, default_stride: int = 2, output_stride: int = 32, drop_path_rate: float = 0.):
+        # Generate RegNet ws per block
+        widths, num_stages, stage_gs = generate_regnet(cfg.wa, cfg.w0, cfg.wm, cfg.depth, cfg.group_size)
+
+        # Convert to per stage format
+        stage_widths, stage_depths = torch.unique(torch.tensor(widths), return_counts=True)
+        stage_widths, stage_depths = stage_widths.tolist(), stage_depths.tolist()
+        stage_br = [cfg.bottle_ratio for _ in range(num_stages)]
+        stage_strides = []
+        stage_dilations = []
+        net_stride = 2
+        dilation = 1
+        for _ in range(num_stages):
+            if net_stride >= output_stride:
+                dilation *= default_stride
+                stride = 1
+            else:
+                stride = default_stride
+                net_stride *= stride
+            stage_strides.append(stride)
+            stage_dilations.append(dilation)
+        stage_dpr = torch.split(torch.linspace(0, drop_path_rate, sum(stage_depths)), stage_depths)
+        stage_dpr = [x.tolist() for x in stage_dpr]
+
+        # Adjust the compatibility of ws and gws
+        stage_widths, stage_gs = adjust_widths_groups_comp(
+            stage_widths, stage_br, stage_gs, min_ratio=cfg.group_min_ratio)
+        arg_names = ['out_chs', 'stride', 'dilation', 'depth', 'bottle_ratio', 'group_size', 'drop_path_rates']
+        per_stage_args = [
+            dict(zip(arg_names, params)) for params in
+            zip(stage_widths, stage_strides, stage_dilations, stage_depths, stage_br, stage_gs, stage_dpr)
+        ]
+        common_args = dict(
+            downsample=cfg.downsample,
+            se_ratio=cfg.se_ratio,
+            linear_out=cfg.linear_out,
+            act_layer=cfg.act_layer,
+            norm_layer=cfg.norm_layer,
+        )
+        return per_stage_args, common_args
+
+    @torch.jit.ignore
+    def group_matcher(self, coarse: bool = False) -> Dict:
+        return dict(
+            stem=r'^stem',
+            blocks=r'^s(\d+)' if coarse else r'^s(\d+)\.b(\d+)',
+        )
+
+    @torch.jit.ignore
+    def set_grad_checkpointing(self, enable: bool = True):
+        for s in list(self.children())[1:-1]:
+            if isinstance(s, RegStage):
+                s.grad_checkpointing = enable
+
+    @torch.jit.ignore
+    def get_classifier(self) -> nn.Module:
+        return self.head.fc
+
+    def reset_classifier(self, num_classes: int, global_pool: Optional[str] = None):
+        self.num_classes = num_classes
+        self.head.reset(num_classes, pool_type=global_pool)
+
+    def forward_features(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.stem(x)
+        x = self.s1(x)
+        x = self.s2(x)
+        x = self.s3(x)
+        x = self.s4(x)
+        x = self.final_conv(x)
+        return x
+
+    def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
+        return self.head(x, pre_logits=pre_logits) if pre_logits else self.head(x)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.forward_features(x)
+        x = self.forward_head(x)
+        return x
+
+
+def _init_weights(module: nn.Module, name: str = '', zero_init_last: bool = False) -> None:
+    if isinstance(module, nn.Conv2d):
+        fan_out = module.kernel_size[0] * module.kernel_size[1] * module.out_channels
+        fan_out //= module.groups
+        module.weight.data.normal_(0, math.sqrt(2.0 / fan_out))
+        if module.bias is not None:
+            nn.init.zeros_(module.bias)
+    elif isinstance(module, nn.Linear):
+        nn.init.normal_(module.weight, mean=0.0, std=0.01)
+        if module.bias is not None:
+            nn.init.zeros_(module.bias)
+    elif zero_init_last and hasattr(module, 'zero_init_last'):
+        module.zero_init_last()
+
+
+model_cfgs = dict(
+    # RegNet-X
+    regnetx_002=RegNetCfg(w0=24, wa=36.44, wm=2.49, group_size=8, depth=13),
+    regnetx_004=RegNetCfg(w0=24, wa=24.48, wm=2.54, group_size=16, depth=22),
+    regnetx_006=RegNetCfg(w0=48, wa=36.97, wm=2.24, group_size=24, depth=16),
+    regnetx_008=RegNetCfg(w0=56, wa=35.73, wm=2.28, group_size=16, depth=16),
+    regnetx_016=RegNetCfg(w0=80, wa=34.01, wm=2.25, group_size=24, depth=18),
+    regnetx_032=RegNetCfg(w0=88, wa=26.31, wm=2.25, group_size=48, depth=25),
+    regnetx_040=RegNetCfg(w0=96, wa=38.65, wm=2.43, group_size=40, depth=23),
+    regnetx_064=RegNetCfg(w0=184, wa=60.83, wm=2.07, group_size=56, depth=17),
+    regnetx_080=RegNetCfg(w0=80, wa=49.56, wm=2.88, group_size=120, depth=23),
+    regnetx_120=RegNetCfg(w0=168, wa=73.36, wm=2.37, group_size=112, depth=19),
+    regnetx_160=RegNetCfg(w0=216, wa=55.59, wm=2.1, group_size=128, depth=22),
+    regnetx_320=RegNetCfg(w0=320, wa=69.86, wm=2.0, group_size=168, depth=23),
+    # RegNet-Y (w/ SE)
+    regnety_002=RegNetCfg(w0=24, wa=36.44, wm=2.49, group_size=8, depth=13, se_ratio=0.25),
+    regnety_004=RegNetCfg(w0=48, wa=27.89, wm=2.09, group_size=8, depth=16, se_ratio=0.25),
+    regnety_006=RegNetCfg(w0=48, wa=32.54, wm=2.32, group_size=16, depth=15, se_ratio=0.25),
+    regnety_008=RegNetCfg(w0=56, wa=38.84, wm=2.4, group_size=16, depth=14, se_ratio=0.25),
+    regnety_016=RegNetCfg(w0=48, wa=20.71, wm=2.65, group_size=24, depth=27, se_ratio=0.25),
+    regnety_032=RegNetCfg(w0=80, wa=42.63, wm=2.66, group_size=24, depth=21, se_ratio=0.25),
+    regnety_040=RegNetCfg(w0=96, wa=31.41, wm=2.24, group_size=64, depth=22, se_ratio=0.25),
+    regnety_064=RegNetCfg(w0=112, wa=33.22, wm=2.27, group_size=72, depth=25, se_ratio=0.25),
+    regnety_080=RegNetCfg(w0=192, wa=76.82, wm=2.19, group_size=56, depth=17, se_ratio=0.25),
+    regnety_120=RegNetCfg(w0=168, wa=73.36, wm=2.37, group_size=112, depth=19, se_ratio=0.25),
+    regnety_160=RegNetCfg(w0=200, wa=106.23, wm=2.48, group_size=112, depth=18, se_ratio=0.25),
+    regnety_320=RegNetCfg(w0=232, wa=115.89, wm=2.53, group_size=232, depth=20, se_ratio=0.25),
+)
+
+
+def _create_regnet(variant: str, pretrained: bool, **kwargs) -> RegNet:
+    return build_model_with_cfg(
+        RegNet, variant, pretrained,
+        model_cfg=model_cfgs[variant],
+        feature_cfg=dict(flatten_sequential=True),
+        **kwargs,
+    )
+
+
+def _cfg(url: str = '', **kwargs) -> Dict[str, Any]:
+    return {
+        'url': url, 'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': (7, 7),
+        'crop_pct': 0.875, 'interpolation': 'bicubic',
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'stem.conv', 'classifier': 'head.fc',
+        **kwargs,
+    }
+
+
+default_cfgs = generate_default_cfgs({
+    **{f'{name}.pycls_in1k': _cfg() for name in model_cfgs},
+    'regnety_032.ra_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
+    'regnety_040.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
+    'regnety_064.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
+    'regnety_080.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
+    'regnety_160.deit_in1k': _cfg(),
+})
+
+
+def _register(name):
+    def fn(pretrained: bool = False, **kwargs) -> RegNet:
+        return _create_regnet(name, pretrained, **kwargs)
+    fn.__name__ = name
+    return register_model(fn)
+
+
+for _name in list(model_cfgs):
+    globals()[_name] = _register(_name)

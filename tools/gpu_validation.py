@@ -3,8 +3,11 @@ bench.py's default run.  Each check prints one line; run via gpurun.
 
 Usage: python tools/gpu_validation.py [muon_eva02] [naflex_infer] [swin_train]
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
