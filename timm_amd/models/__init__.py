@@ -36,6 +36,7 @@ from .densenet import *
 from .deit import *
 from .efficientnet import *
 from .eva import *
+from .ghostnet import *
 from .hiera import *
 from .maxxvit import *
 from .mlp_mixer import *
