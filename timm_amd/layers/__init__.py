@@ -51,6 +51,7 @@ from .pos_embed_sincos import (
     apply_rot_embed_cat, apply_rot_embed_list, build_fourier_pos_embed, build_rotary_pos_embed,
     build_sincos2d_pos_embed, create_rope_embed, freq_bands, pixel_freq_bands, rope_rotate_half, rot,
 )
+from .std_conv import ScaledStdConv2d, ScaledStdConv2dSame, StdConv2d, StdConv2dSame
 from .squeeze_excite import EffectiveSEModule, EffectiveSqueezeExcite, SEModule, SqueezeExcite, SqueezeExciteCl
 from .test_time_pool import TestTimePoolHead, apply_test_time_pool
 from .trace_utils import _assert

@@ -41,6 +41,7 @@ from .hiera import *
 from .maxxvit import *
 from .mlp_mixer import *
 from .mobilenetv3 import *
+from .nfnet import *
 from .naflexvit import *
 from .swin_transformer import *
 from .regnet import *
