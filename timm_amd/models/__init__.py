@@ -31,6 +31,7 @@ from ._registry import (
 
 # architecture modules (registration happens at import time)
 from .beit import *
+from .cait import *
 from .convnext import *
 from .densenet import *
 from .deit import *
@@ -47,4 +48,5 @@ from .swin_transformer import *
 from .regnet import *
 from .resnet import *
 from .vgg import *
+from .xcit import *
 from .vision_transformer import *
