@@ -46,6 +46,7 @@ from .nfnet import *
 from .naflexvit import *
 from .swin_transformer import *
 from .regnet import *
+from .pvt_v2 import *
 from .resnet import *
 from .vgg import *
 from .xcit import *
