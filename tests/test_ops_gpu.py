@@ -241,6 +241,22 @@ def test_swin_gpu_matches_cpu():
     assert err < 0.35, f'swin gpu/cpu mismatch {err}'
 
 
+def test_attn_bwd_preprocess():
+    """Fused dO-copy + delta kernel vs composed torch ops, incl. strided BNHD input."""
+    from timm_amd.ops import _load_extension
+    ext = _load_extension()
+    torch.manual_seed(9)
+    for B, H, N, D in [(2, 4, 197, 64), (1, 3, 50, 96), (2, 2, 33, 128), (3, 5, 17, 32)]:
+        # BNHD storage viewed as [B,H,N,D] — the fwd's output layout
+        do_s = torch.randn(B, N, H, D, device='cuda', dtype=torch.bfloat16).permute(0, 2, 1, 3)
+        o_s = torch.randn(B, N, H, D, device='cuda', dtype=torch.bfloat16).permute(0, 2, 1, 3)
+        do_c, delta = ext.attn_bwd_preprocess(do_s, o_s)
+        assert do_c.is_contiguous()
+        assert torch.equal(do_c, do_s.contiguous())
+        ref = (do_s.float() * o_s.float()).sum(-1)
+        assert (delta - ref).abs().max().item() < ref.abs().max().item() * 1e-2 + 1e-3, (B, H, N, D)
+
+
 def test_fused_adamw_matches_reference():
     _ext()
     torch.manual_seed(6)

@@ -68,15 +68,19 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         ext = _load_extension()
         scale = ctx.scale
-        do = do.contiguous()
         # exact flash backward: recompute P from saved LSE. GEMMs stay bf16
         # (MFMA via hipBLASLt, fp32 internal accum); the softmax-recompute
-        # elementwise runs in two fused HIP kernels (attn_bwd_softmax/_ds).
+        # elementwise runs in fused HIP kernels (attn_bwd_preprocess/_softmax/_ds).
+        if do.dim() == 4 and do.stride(-1) == 1 and do.shape[-1] % 32 == 0 and do.shape[-1] <= 128:
+            # fused: contiguous dO + delta = rowsum(dO*O) in one pass
+            do, delta = ext.attn_bwd_preprocess(do, o)
+        else:
+            do = do.contiguous()
+            delta = (do * o).float().sum(-1)  # [B,H,Nq] fp32
         s = (q @ k.transpose(-2, -1)).contiguous()  # bf16 GEMM
         p = ext.attn_bwd_softmax(s, lse, ctx.attn_mask, scale)  # exp(s*scale+mask-lse)
         dv = p.transpose(-2, -1) @ do
         dp = (do @ v.transpose(-2, -1)).contiguous()  # bf16 GEMM
-        delta = (do * o).float().sum(-1)  # [B,H,Nq] fp32
         ds = ext.attn_bwd_ds(p, dp, delta, scale)
         dq = ds @ k
         dk = ds.transpose(-2, -1) @ q

@@ -405,3 +405,88 @@ std::vector<at::Tensor> attention_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   else launch_attn_fwd<128>(q, k, v, mask, o, lse, B, H, Nq, Nk, D, (float)scale, stream);
   return {o, lse};
 }
+
+namespace {
+
+// ---- bwd preprocess: fuse the do-contiguous copy with delta = rowsum(do*o) ----
+// do/o arrive as strided [B,H,N,D] views of BNHD storage (the fwd's output
+// layout); one pass reads both, emits contiguous bf16 dO plus fp32 delta.
+// Replaces three torch kernels (direct_copy + mul + fp32 reduce) that moved
+// ~5x the bytes.  Layout: lanes_per_row = D/8 lanes cooperate on one row
+// with bf16x8 vector loads; a 256-thread block covers 256*8/D rows.
+__global__ __launch_bounds__(256)
+void attn_bwd_preprocess_kernel(
+    const __bf16* __restrict__ dov,  // strided [B,H,N,D]
+    const __bf16* __restrict__ ov,   // strided [B,H,N,D]
+    __bf16* __restrict__ do_c,       // contiguous [B,H,N,D]
+    float* __restrict__ delta,       // [B,H,N]
+    long rows, int D,
+    long do_sb, long do_sh, long do_sn,
+    long o_sb, long o_sh, long o_sn,
+    int H, int N) {
+  const int lpr = D / 8;                       // lanes per row (D multiple of 32 -> 4/8/12/16)
+  const int rows_per_block = 256 / lpr;
+  const int local_row = threadIdx.x / lpr;
+  const int lane_in_row = threadIdx.x % lpr;
+
+  for (long row = (long)blockIdx.x * rows_per_block + local_row; row < rows;
+       row += (long)gridDim.x * rows_per_block) {
+    const long n = row % N;
+    const long h = (row / N) % H;
+    const long b = row / ((long)H * N);
+    const __bf16* do_row = dov + b * do_sb + h * do_sh + n * do_sn;
+    const __bf16* o_row = ov + b * o_sb + h * o_sh + n * o_sn;
+
+    bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(do_row + lane_in_row * 8);
+    bf16x8_t oval = *reinterpret_cast<const bf16x8_t*>(o_row + lane_in_row * 8);
+    float acc = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc += (float)dv[j] * (float)oval[j];
+
+    // reduce across the lpr lanes of this row: xor ladder for pow2 lpr
+    // (D=32/64/128), LDS fallback for lpr=12 (D=96)
+    if ((lpr & (lpr - 1)) == 0) {
+      for (int off = lpr / 2; off > 0; off /= 2) {
+        acc += __shfl_xor(acc, off, 64);
+      }
+    } else {
+      __shared__ float red[256];
+      red[threadIdx.x] = acc;
+      __syncthreads();
+      if (lane_in_row == 0) {
+        float s = 0.f;
+        for (int j = 0; j < lpr; ++j) s += red[local_row * lpr + j];
+        acc = s;
+      }
+      __syncthreads();
+    }
+
+    *reinterpret_cast<bf16x8_t*>(do_c + row * D + lane_in_row * 8) = dv;
+    if (lane_in_row == 0) delta[row] = acc;
+  }
+}
+
+}  // namespace
+
+std::vector<at::Tensor> attn_bwd_preprocess(at::Tensor dout, at::Tensor o) {
+  TORCH_CHECK(dout.is_cuda() && dout.dim() == 4 && dout.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dout.stride(3) == 1 && o.stride(3) == 1, "head_dim must be contiguous");
+  int B = dout.size(0), H = dout.size(1), N = dout.size(2), D = dout.size(3);
+  TORCH_CHECK(D % 32 == 0 && D <= 128);
+  auto do_c = at::empty({B, H, N, D}, dout.options());
+  auto delta = at::empty({B, H, N}, dout.options().dtype(at::kFloat));
+  long rows = (long)B * H * N;
+  int lpr = D / 8;
+  int rows_per_block = 256 / lpr;
+  int blocks = (int)std::min((rows + rows_per_block - 1) / rows_per_block, (long)8192);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(attn_bwd_preprocess_kernel, dim3(blocks), dim3(256), 0, stream,
+      (const __bf16*)dout.data_ptr(), (const __bf16*)o.data_ptr(),
+      (__bf16*)do_c.data_ptr(), delta.data_ptr<float>(),
+      rows, D,
+      dout.stride(0), dout.stride(1), dout.stride(2),
+      o.stride(0), o.stride(1), o.stride(2),
+      H, N);
+  HIP_CHECK_LAST();
+  return {do_c, delta};
+}
