@@ -52,6 +52,11 @@ from .pos_embed_sincos import (
     build_sincos2d_pos_embed, create_rope_embed, freq_bands, pixel_freq_bands, rope_rotate_half, rot,
 )
 from .std_conv import ScaledStdConv2d, ScaledStdConv2dSame, StdConv2d, StdConv2dSame
+from .filter_response_norm import FilterResponseNormAct2d, FilterResponseNormTlu2d
+from .selective_kernel import SelectiveKernel, SelectiveKernelAttn
+from .space_to_depth import DepthToSpace, SpaceToDepth
+from .split_attn import RadixSoftmax, SplitAttn
+from .split_batchnorm import SplitBatchNorm2d, convert_splitbn_model
 from .squeeze_excite import EffectiveSEModule, EffectiveSqueezeExcite, SEModule, SqueezeExcite, SqueezeExciteCl
 from .test_time_pool import TestTimePoolHead, apply_test_time_pool
 from .trace_utils import _assert
