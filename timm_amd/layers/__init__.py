@@ -52,6 +52,10 @@ from .pos_embed_sincos import (
     build_sincos2d_pos_embed, create_rope_embed, freq_bands, pixel_freq_bands, rope_rotate_half, rot,
 )
 from .std_conv import ScaledStdConv2d, ScaledStdConv2dSame, StdConv2d, StdConv2dSame
+from .evo_norm import (
+    EvoNorm2dB0, EvoNorm2dB1, EvoNorm2dB2, EvoNorm2dS0, EvoNorm2dS0a, EvoNorm2dS1, EvoNorm2dS1a,
+    EvoNorm2dS2, EvoNorm2dS2a,
+)
 from .filter_response_norm import FilterResponseNormAct2d, FilterResponseNormTlu2d
 from .selective_kernel import SelectiveKernel, SelectiveKernelAttn
 from .space_to_depth import DepthToSpace, SpaceToDepth

@@ -420,11 +420,11 @@ void attn_bwd_preprocess_kernel(
     const __bf16* __restrict__ ov,   // strided [B,H,N,D]
     __bf16* __restrict__ do_c,       // contiguous [B,H,N,D]
     float* __restrict__ delta,       // [B,H,N]
-    long rows, int D,
+    long rows, int D, int lpr,       // lpr: pow2 lanes per row (host picks)
     long do_sb, long do_sh, long do_sn,
     long o_sb, long o_sh, long o_sn,
     int H, int N) {
-  const int lpr = D / 8;                       // lanes per row (D multiple of 32 -> 4/8/12/16)
+  const int vecs = (D / 8) / lpr;              // bf16x8 chunks per lane
   const int rows_per_block = 256 / lpr;
   const int local_row = threadIdx.x / lpr;
   const int lane_in_row = threadIdx.x % lpr;
@@ -437,31 +437,20 @@ void attn_bwd_preprocess_kernel(
     const __bf16* do_row = dov + b * do_sb + h * do_sh + n * do_sn;
     const __bf16* o_row = ov + b * o_sb + h * o_sh + n * o_sn;
 
-    bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(do_row + lane_in_row * 8);
-    bf16x8_t oval = *reinterpret_cast<const bf16x8_t*>(o_row + lane_in_row * 8);
     float acc = 0.f;
+    for (int v = 0; v < vecs; ++v) {
+      const int c8 = v * lpr + lane_in_row;
+      bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(do_row + c8 * 8);
+      bf16x8_t oval = *reinterpret_cast<const bf16x8_t*>(o_row + c8 * 8);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc += (float)dv[j] * (float)oval[j];
-
-    // reduce across the lpr lanes of this row: xor ladder for pow2 lpr
-    // (D=32/64/128), LDS fallback for lpr=12 (D=96)
-    if ((lpr & (lpr - 1)) == 0) {
-      for (int off = lpr / 2; off > 0; off /= 2) {
-        acc += __shfl_xor(acc, off, 64);
-      }
-    } else {
-      __shared__ float red[256];
-      red[threadIdx.x] = acc;
-      __syncthreads();
-      if (lane_in_row == 0) {
-        float s = 0.f;
-        for (int j = 0; j < lpr; ++j) s += red[local_row * lpr + j];
-        acc = s;
-      }
-      __syncthreads();
+      for (int j = 0; j < 8; ++j) acc += (float)dv[j] * (float)oval[j];
+      *reinterpret_cast<bf16x8_t*>(do_c + row * D + c8 * 8) = dv;
     }
 
-    *reinterpret_cast<bf16x8_t*>(do_c + row * D + lane_in_row * 8) = dv;
+    // xor ladder over the lpr (always pow2) lanes of this row
+    for (int off = lpr / 2; off > 0; off /= 2) {
+      acc += __shfl_xor(acc, off, 64);
+    }
     if (lane_in_row == 0) delta[row] = acc;
   }
 }
@@ -477,13 +466,14 @@ std::vector<at::Tensor> attn_bwd_preprocess(at::Tensor dout, at::Tensor o) {
   auto delta = at::empty({B, H, N}, dout.options().dtype(at::kFloat));
   long rows = (long)B * H * N;
   int lpr = D / 8;
+  if (lpr & (lpr - 1)) lpr = 4;  // non-pow2 chunk count (D=96): 4 lanes x 3 chunks
   int rows_per_block = 256 / lpr;
   int blocks = (int)std::min((rows + rows_per_block - 1) / rows_per_block, (long)8192);
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(attn_bwd_preprocess_kernel, dim3(blocks), dim3(256), 0, stream,
       (const __bf16*)dout.data_ptr(), (const __bf16*)o.data_ptr(),
       (__bf16*)do_c.data_ptr(), delta.data_ptr<float>(),
-      rows, D,
+      rows, D, lpr,
       dout.stride(0), dout.stride(1), dout.stride(2),
       o.stride(0), o.stride(1), o.stride(2),
       H, N);
