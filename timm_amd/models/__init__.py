@@ -48,7 +48,10 @@ from .naflexvit import *
 from .swin_transformer import *
 from .regnet import *
 from .pvt_v2 import *
+from .res2net import *
+from .resnest import *
 from .resnet import *
+from .sknet import *
 from .vgg import *
 from .xcit import *
 from .vision_transformer import *
