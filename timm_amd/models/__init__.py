@@ -40,6 +40,7 @@ from .efficientnet import *
 from .eva import *
 from .ghostnet import *
 from .hiera import *
+from .inception_v3 import *
 from .maxxvit import *
 from .mlp_mixer import *
 from .mobilenetv3 import *
