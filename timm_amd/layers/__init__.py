@@ -14,6 +14,7 @@ from .config import (
 )
 from .conv2d_same import Conv2dSame, Conv2dSameExport, conv2d_same
 from .conv_bn_act import ConvBnAct, ConvNormAct, ConvNormActAa
+from .conv_helpers_extra import SeparableConv2d, SeparableConvBnAct, SeparableConvNormAct
 from .create_act import create_act_layer, get_act_fn, get_act_layer
 from .create_attn import create_attn, create_attn_layer, get_attn
 from .create_conv2d import create_conv2d

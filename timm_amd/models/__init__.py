@@ -56,3 +56,4 @@ from .sknet import *
 from .vgg import *
 from .xcit import *
 from .vision_transformer import *
+from .vovnet import *
