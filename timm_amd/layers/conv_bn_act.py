@@ -3,6 +3,7 @@ from typing import Any, Dict, Optional, Type
 
 from torch import nn as nn
 
+from .blur_pool import create_aa
 from .create_conv2d import create_conv2d
 from .norm_act import get_norm_act_layer
 
@@ -22,6 +23,7 @@ class ConvNormAct(nn.Module):
             apply_act: bool = True,
             norm_layer: Type[nn.Module] = nn.BatchNorm2d,
             act_layer: Optional[Type[nn.Module]] = nn.ReLU,
+            aa_layer: Optional[Type[nn.Module]] = None,
             drop_layer: Optional[Type[nn.Module]] = None,
             conv_kwargs: Optional[Dict[str, Any]] = None,
             norm_kwargs: Optional[Dict[str, Any]] = None,
@@ -31,10 +33,10 @@ class ConvNormAct(nn.Module):
         conv_kwargs = conv_kwargs or {}
         norm_kwargs = norm_kwargs or {}
         act_kwargs = act_kwargs or {}
-        use_aa = False
+        use_aa = aa_layer is not None and stride > 1
 
         self.conv = create_conv2d(
-            in_channels, out_channels, kernel_size, stride=stride,
+            in_channels, out_channels, kernel_size, stride=1 if use_aa else stride,
             padding=padding, dilation=dilation, groups=groups, bias=bias, **conv_kwargs)
 
         if apply_norm:
@@ -55,6 +57,8 @@ class ConvNormAct(nn.Module):
                 norm_kwargs['drop_layer'] = drop_layer
                 self.bn.add_module('drop', drop_layer())
 
+        self.aa = create_aa(aa_layer, out_channels, stride=stride, enable=use_aa, noop=None)
+
     @property
     def in_channels(self):
         return self.conv.in_channels
@@ -66,6 +70,8 @@ class ConvNormAct(nn.Module):
     def forward(self, x):
         x = self.conv(x)
         x = self.bn(x)
+        if self.aa is not None:
+            x = self.aa(x)
         return x
 
 

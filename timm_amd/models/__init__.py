@@ -53,6 +53,7 @@ from .res2net import *
 from .resnest import *
 from .resnet import *
 from .sknet import *
+from .tresnet import *
 from .vgg import *
 from .xcit import *
 from .vision_transformer import *
