@@ -42,6 +42,7 @@ from .ghostnet import *
 from .hiera import *
 from .inception_v3 import *
 from .maxxvit import *
+from .metaformer import *
 from .mlp_mixer import *
 from .mobilenetv3 import *
 from .nfnet import *
