@@ -38,6 +38,7 @@ from .densenet import *
 from .deit import *
 from .efficientnet import *
 from .eva import *
+from .focalnet import *
 from .ghostnet import *
 from .hiera import *
 from .inception_v3 import *
