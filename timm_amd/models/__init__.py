@@ -36,6 +36,7 @@ from .cait import *
 from .convnext import *
 from .densenet import *
 from .deit import *
+from .dpn import *
 from .efficientnet import *
 from .eva import *
 from .focalnet import *
