@@ -42,6 +42,7 @@ from .eva import *
 from .focalnet import *
 from .ghostnet import *
 from .hiera import *
+from .inception_next import *
 from .inception_v3 import *
 from .maxxvit import *
 from .metaformer import *
