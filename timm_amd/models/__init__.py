@@ -37,6 +37,7 @@ from .convnext import *
 from .densenet import *
 from .deit import *
 from .dpn import *
+from .edgenext import *
 from .efficientnet import *
 from .eva import *
 from .focalnet import *
