@@ -352,7 +352,15 @@ def test_depthwise_conv_nhwc(shape, k, stride):
     assert rel_err(b1.grad, b2.grad) < 3e-2
 
 
-@pytest.mark.parametrize('model_name', ['eva02_tiny_patch14_224', 'naflexvit_base_patch16_gap', 'convnext_atto', 'efficientnet_b0'])
+@pytest.mark.parametrize('model_name', [
+    'eva02_tiny_patch14_224', 'naflexvit_base_patch16_gap', 'convnext_atto', 'efficientnet_b0',
+    # round-1 breadth additions — one representative per new family
+    'beit_base_patch16_224', 'cait_xxs24_224', 'xcit_nano_12_p16_224', 'pvt_v2_b0',
+    'caformer_s18', 'poolformer_s12', 'dm_nfnet_f0', 'regnety_032', 'repvgg_a0',
+    'resnest26d', 'res2net50_26w_4s', 'tresnet_m', 'ese_vovnet19b_dw', 'ghostnet_100',
+    'densenet121', 'vgg11_bn', 'inception_v3', 'inception_next_atto', 'edgenext_xx_small',
+    'focalnet_tiny_srf', 'dpn68', 'skresnet18', 'gernet_s',
+])
 def test_model_gpu_vs_cpu(model_name):
     """Model forward on GPU (HIP kernels) vs CPU fp32 reference."""
     _ext()
