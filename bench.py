@@ -38,6 +38,8 @@ def parse_args():
     p.add_argument('--img-size', type=int, default=224)
     p.add_argument('--mode', type=str, default='train', choices=['train', 'infer'])
     p.add_argument('--amp-dtype', type=str, default='bfloat16')
+    p.add_argument('--channels-last', action='store_true', default=None,
+                   help='NHWC memory format (default: auto-on for conv nets)')
     return p.parse_args()
 
 
@@ -66,6 +68,15 @@ def main():
     model = timm_amd.create_model(args.model, num_classes=1000)
     model = model.to(device=device, dtype=dtype)
 
+    # NHWC routes depthwise/dense convs to our gfx950 kernels / MIOpen's fast
+    # paths; transformer models stay NCHW (patchify GEMM is layout-free)
+    channels_last = args.channels_last
+    if channels_last is None:
+        conv_families = ('convnext', 'resnet', 'efficientnet', 'mobilenet', 'regnet', 'nfnet', 'densenet')
+        channels_last = any(f in args.model for f in conv_families)
+    if channels_last:
+        model = model.to(memory_format=torch.channels_last)
+
     if args.mode == 'train':
         model.train()
         from timm_amd.optim import AdamW
@@ -79,6 +90,8 @@ def main():
 
     B = args.batch_size
     x = torch.randn(B, 3, args.img_size, args.img_size, device=device, dtype=dtype)
+    if channels_last:
+        x = x.contiguous(memory_format=torch.channels_last)
     target = torch.randint(0, 1000, (B,), device=device)
 
     def train_step():
