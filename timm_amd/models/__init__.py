@@ -59,6 +59,7 @@ from .resnest import *
 from .resnet import *
 from .sknet import *
 from .tresnet import *
+from .twins import *
 from .vgg import *
 from .xcit import *
 from .vision_transformer import *
