@@ -38,6 +38,7 @@ from .densenet import *
 from .deit import *
 from .dpn import *
 from .edgenext import *
+from .efficientformer import *
 from .efficientnet import *
 from .eva import *
 from .focalnet import *
