@@ -33,6 +33,7 @@ from ._registry import (
 from .beit import *
 from .byobnet import *
 from .cait import *
+from .convmixer import *
 from .convnext import *
 from .densenet import *
 from .deit import *
