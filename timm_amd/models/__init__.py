@@ -54,6 +54,7 @@ from .mobilenetv3 import *
 from .nfnet import *
 from .naflexvit import *
 from .swin_transformer import *
+from .swin_transformer_v2 import *
 from .regnet import *
 from .pvt_v2 import *
 from .res2net import *
