@@ -60,6 +60,7 @@ from .pvt_v2 import *
 from .res2net import *
 from .resnest import *
 from .resnet import *
+from .rexnet import *
 from .sknet import *
 from .tresnet import *
 from .twins import *
