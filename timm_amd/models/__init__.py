@@ -56,6 +56,7 @@ from .naflexvit import *
 from .swin_transformer import *
 from .swin_transformer_v2 import *
 from .regnet import *
+from .pit import *
 from .pvt_v2 import *
 from .res2net import *
 from .resnest import *
