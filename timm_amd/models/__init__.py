@@ -36,6 +36,7 @@ from .cait import *
 from .convmixer import *
 from .convnext import *
 from .densenet import *
+from .davit import *
 from .deit import *
 from .dpn import *
 from .edgenext import *
