@@ -52,6 +52,7 @@ from .maxxvit import *
 from .metaformer import *
 from .mlp_mixer import *
 from .mobilenetv3 import *
+from .mobilevit import *
 from .nfnet import *
 from .naflexvit import *
 from .swin_transformer import *
