@@ -127,11 +127,13 @@ void dwconv_bwd_data_kernel(
 }
 
 // ---- backward weight: dw[kh,kw,c] = sum_{b,ho,wo} dy[b,ho,wo,c] * x[...] ----
-// Grid: (spatial-chunk, kh, channel-vec).  One dy read serves a full kernel
-// ROW (all kw taps accumulate in registers, kMaxK*8 floats); x reads for
-// neighbouring kw share cache lines.  Read amplification K vs the naive
-// per-tap scheme's K*K.  Block-level LDS reduce, one fp32 atomic per
-// (tap,channel) per block.
+// Grid: (spatial-chunk, kh, channel-superchunk).  Lane layout is
+// channel-ACROSS-lanes: lane%16 picks an 8-wide channel chunk, lane/16 picks
+// one of 4 pixels, so a wave's dy/x loads are 4 fully-coalesced 256B pixel
+// rows instead of 64 scattered 16B strided reads (the v1 layout's ~8x read
+// amplification made this kernel 48% of a ConvNeXt train step).  All kw taps
+// of one kernel row accumulate in registers; 4-pixel xor-shfl reduce, then
+// LDS across waves, then one fp32 atomic per (tap,channel) per block.
 template <typename T, int kMaxK>
 __global__ __launch_bounds__(kThreads)
 void dwconv_bwd_weight_kernel(
@@ -141,10 +143,14 @@ void dwconv_bwd_weight_kernel(
     float* __restrict__ dbias,    // [C] fp32 (pre-zeroed) or null
     int B, int H, int W, int C,
     int Ho, int Wo, int K, int stride, int pad) {
+  constexpr int kLanesPerPix = 16;              // channel chunks covered per pixel
+  constexpr int kPixPerBlock = kThreads / kLanesPerPix;  // 16
   const int c8 = C / 8;
   const int kh = blockIdx.y;
-  const int cv = blockIdx.z;
-  const int c0 = cv * 8;
+  const int chunk = blockIdx.z * kLanesPerPix + (threadIdx.x % kLanesPerPix);
+  const int pix = threadIdx.x / kLanesPerPix;   // 0..15 within block
+  const bool active = chunk < c8;
+  const int c0 = chunk * 8;
 
   const long spatial = (long)B * Ho * Wo;
   float acc[kMaxK][8];
@@ -157,60 +163,88 @@ void dwconv_bwd_weight_kernel(
   for (int j = 0; j < 8; ++j) bacc[j] = 0.f;
 
   const bool do_bias = (dbias != nullptr) && (kh == 0);
-  for (long s = (long)blockIdx.x * kThreads + threadIdx.x; s < spatial;
-       s += (long)gridDim.x * kThreads) {
-    const int wo = s % Wo;
-    long p = s / Wo;
-    const int ho = p % Ho;
-    const int b = p / Ho;
-    const int hi = ho * stride - pad + kh;
-    bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(
-        dy + (((long)b * Ho + ho) * Wo + wo) * C + c0);
-    if (do_bias) {
+  if (active) {
+    for (long s = (long)blockIdx.x * kPixPerBlock + pix; s < spatial;
+         s += (long)gridDim.x * kPixPerBlock) {
+      const int wo = s % Wo;
+      long p = s / Wo;
+      const int ho = p % Ho;
+      const int b = p / Ho;
+      const int hi = ho * stride - pad + kh;
+      bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(
+          dy + (((long)b * Ho + ho) * Wo + wo) * C + c0);
+      if (do_bias) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) bacc[j] += (float)gv[j];
-    }
-    if (hi < 0 || hi >= H) continue;
-    const T* xrow = x + (((long)b * H + hi) * W) * C + c0;
-    const int wi0 = wo * stride - pad;
-    for (int kw = 0; kw < K; ++kw) {
-      const int wi = wi0 + kw;
-      if (wi < 0 || wi >= W) continue;
-      bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi * C);
+        for (int j = 0; j < 8; ++j) bacc[j] += (float)gv[j];
+      }
+      if (hi < 0 || hi >= H) continue;
+      const T* xrow = x + (((long)b * H + hi) * W) * C + c0;
+      const int wi0 = wo * stride - pad;
+      for (int kw = 0; kw < K; ++kw) {
+        const int wi = wi0 + kw;
+        if (wi < 0 || wi >= W) continue;
+        bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi * C);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)xv[j];
+        for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)xv[j];
+      }
     }
   }
 
-  // block reduce via LDS (wave-level shuffle first), then one global atomic
-  // per (tap, channel) per block
-  __shared__ float red[kMaxK][8];
-  __shared__ float redb[8];
-  if (threadIdx.x < 8) {
-    redb[threadIdx.x] = 0.f;
-    for (int kw = 0; kw < kMaxK; ++kw) red[kw][threadIdx.x] = 0.f;
-  }
-  __syncthreads();
-  for (int kw = 0; kw < K; ++kw) {
+  // reduce the 4 pixels of each wave (lanes xor 16, 32 share a channel chunk)
+#pragma unroll
+  for (int kw = 0; kw < kMaxK; ++kw) {
+    if (kw >= K) break;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float v = wave_reduce_sum(acc[kw][j]);
-      if ((threadIdx.x % WAVE_SIZE) == 0) atomicAdd(&red[kw][j], v);
+      acc[kw][j] += __shfl_xor(acc[kw][j], 16, 64);
+      acc[kw][j] += __shfl_xor(acc[kw][j], 32, 64);
     }
   }
   if (do_bias) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      float vb = wave_reduce_sum(bacc[j]);
-      if ((threadIdx.x % WAVE_SIZE) == 0) atomicAdd(&redb[j], vb);
+      bacc[j] += __shfl_xor(bacc[j], 16, 64);
+      bacc[j] += __shfl_xor(bacc[j], 32, 64);
+    }
+  }
+
+  // cross-wave reduce in LDS: [chunk-in-block][tap][8]
+  __shared__ float red[kLanesPerPix][kMaxK][8];
+  __shared__ float redb[kLanesPerPix][8];
+  if (threadIdx.x < kLanesPerPix) {
+#pragma unroll
+    for (int kw = 0; kw < kMaxK; ++kw)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[threadIdx.x][kw][j] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) redb[threadIdx.x][j] = 0.f;
+  }
+  __syncthreads();
+  const int lane = threadIdx.x % WAVE_SIZE;
+  if (lane < kLanesPerPix && active) {
+    const int cib = threadIdx.x % kLanesPerPix;
+    for (int kw = 0; kw < K; ++kw) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&red[cib][kw][j], acc[kw][j]);
+    }
+    if (do_bias) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&redb[cib][j], bacc[j]);
     }
   }
   __syncthreads();
-  if (threadIdx.x < 8) {
-    for (int kw = 0; kw < K; ++kw) {
-      atomicAdd(&dw[((long)kh * K + kw) * C + c0 + threadIdx.x], red[kw][threadIdx.x]);
+
+  // one global atomic per (tap, channel) per block
+  if (threadIdx.x < kLanesPerPix * 8) {
+    const int cib = threadIdx.x / 8;
+    const int j = threadIdx.x % 8;
+    const int c = (blockIdx.z * kLanesPerPix + cib) * 8 + j;
+    if (c < C) {
+      for (int kw = 0; kw < K; ++kw) {
+        atomicAdd(&dw[((long)kh * K + kw) * C + c], red[cib][kw][j]);
+      }
+      if (do_bias) atomicAdd(&dbias[c], redb[cib][j]);
     }
-    if (do_bias) atomicAdd(&dbias[c0 + threadIdx.x], redb[threadIdx.x]);
   }
 }
 
@@ -266,10 +300,12 @@ std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stri
   }
   auto stream = at::hip::getCurrentHIPStream();
   long spatial = (long)B * Ho * Wo;
-  // size spatial chunks so total blocks ≈ several per CU without huge atomic depth
-  long per_kc = (spatial + kThreads - 1) / kThreads;
-  int sblocks = (int)std::min(std::max<long>(2048 / (K * (C / 8)) + 1, 8), per_kc);
-  dim3 grid(sblocks, K, C / 8);
+  // spatial chunks sized for ~2-4 blocks/CU; atomic depth per dw element is
+  // bounded by sblocks (LDS pre-reduction leaves one atomic per block)
+  int zc = (int)((C / 8 + 15) / 16);
+  long per_pix = (spatial + 15) / 16;
+  int sblocks = (int)std::min(std::max<long>(1024 / (K * zc) + 1, 16), per_pix);
+  dim3 grid(sblocks, K, zc);
   TORCH_CHECK(K <= 9, "dwconv_bwd_weight: kernel size <= 9 supported");
   auto launch = [&](auto tag) {
     hipLaunchKernelGGL((dwconv_bwd_weight_kernel<__bf16, decltype(tag)::value>),
