@@ -44,6 +44,7 @@ from .efficientformer import *
 from .efficientnet import *
 from .eva import *
 from .focalnet import *
+from .gcvit import *
 from .ghostnet import *
 from .hiera import *
 from .inception_next import *

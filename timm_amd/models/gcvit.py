@@ -535,9 +535,10 @@ class GlobalContextVit(nn.Module):
 
 
 def _create_gcvit(variant, pretrained=False, **kwargs):
+    out_indices = kwargs.pop('out_indices', (0, 1, 2, 3))
     model = build_model_with_cfg(
         GlobalContextVit, variant, pretrained,
-        feature_cfg=dict(flatten_sequential=True),
+        feature_cfg=dict(flatten_sequential=True, out_indices=out_indices),
         **kwargs,
     )
     return model
