@@ -66,6 +66,7 @@ from .resnest import *
 from .resnet import *
 from .rexnet import *
 from .sknet import *
+from .tiny_vit import *
 from .tresnet import *
 from .twins import *
 from .vgg import *
