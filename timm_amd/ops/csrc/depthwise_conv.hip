@@ -248,6 +248,142 @@ void dwconv_bwd_weight_kernel(
   }
 }
 
+// ---- stride-1 row-sliding variant: each lane walks one output row with a
+// K-deep ring of x vectors in registers, so x is loaded ONCE per (row, kh)
+// instead of K times (NHWC neighbours are C*2 bytes apart -> no cache-line
+// sharing across kw; the generic kernel pays K^2 total read amplification,
+// this one pays K).
+template <typename T, int kMaxK>
+__global__ __launch_bounds__(kThreads)
+void dwconv_bwd_weight_s1_kernel(
+    const T* __restrict__ dy,     // [B,Ho,Wo,C]
+    const T* __restrict__ x,      // [B,H,W,C]
+    float* __restrict__ dw,       // [K*K, C] fp32 (pre-zeroed)
+    float* __restrict__ dbias,    // [C] fp32 (pre-zeroed) or null
+    int B, int H, int W, int C,
+    int Ho, int Wo, int K, int pad) {
+  constexpr int kLanesPerPix = 16;
+  constexpr int kRowsPerBlock = kThreads / kLanesPerPix;  // 16
+  const int c8 = C / 8;
+  const int kh = blockIdx.y;
+  const int chunk = blockIdx.z * kLanesPerPix + (threadIdx.x % kLanesPerPix);
+  const int rix = threadIdx.x / kLanesPerPix;
+  const bool active = chunk < c8;
+  const int c0 = chunk * 8;
+
+  const long rows = (long)B * Ho;
+  float acc[kMaxK][8];
+  float bacc[8];
+#pragma unroll
+  for (int kw = 0; kw < kMaxK; ++kw)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[kw][j] = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) bacc[j] = 0.f;
+
+  const bool do_bias = (dbias != nullptr) && (kh == 0);
+  if (active) {
+    for (long r = (long)blockIdx.x * kRowsPerBlock + rix; r < rows;
+         r += (long)gridDim.x * kRowsPerBlock) {
+      const int ho = r % Ho;
+      const int b = r / Ho;
+      const int hi = ho + kh - pad;   // stride == 1
+      const bool row_ok = (hi >= 0 && hi < H);
+      const T* dyrow = dy + (((long)b * Ho + ho) * Wo) * C + c0;
+      const T* xrow = x + (((long)b * H + (row_ok ? hi : 0)) * W) * C + c0;
+
+      // ring of x vectors covering [wo - pad, wo - pad + K - 1]
+      bf16x8_t ring[kMaxK];
+#pragma unroll
+      for (int i = 0; i < kMaxK; ++i) ring[i] = bf16x8_t{};
+      // preload taps for wo = 0: wi = -pad .. K-1-pad; slot i holds wi = wo - pad + i
+      for (int i = 0; i < K - 1; ++i) {
+        const int wi = i - pad;
+        if (row_ok && wi >= 0 && wi < W)
+          ring[i] = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi * C);
+      }
+      int head = K - 1;  // next slot to fill
+      for (int wo = 0; wo < Wo; ++wo) {
+        const int wi_new = wo - pad + K - 1;
+        bf16x8_t nv = bf16x8_t{};
+        if (row_ok && wi_new >= 0 && wi_new < W)
+          nv = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi_new * C);
+        ring[head] = nv;
+        head = (head + 1 < K) ? head + 1 : 0;  // now head == slot of oldest (kw=0)
+
+        bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(dyrow + (long)wo * C);
+        if (do_bias) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bacc[j] += (float)gv[j];
+        }
+        if (!row_ok) continue;
+        int slot = head;
+        for (int kw = 0; kw < K; ++kw) {
+          bf16x8_t xv = ring[slot];
+          slot = (slot + 1 < K) ? slot + 1 : 0;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)xv[j];
+        }
+      }
+    }
+  }
+
+  // reduce the 4 row-lanes of each wave (lanes xor 16, 32 share a channel chunk)
+#pragma unroll
+  for (int kw = 0; kw < kMaxK; ++kw) {
+    if (kw >= K) break;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[kw][j] += __shfl_xor(acc[kw][j], 16, 64);
+      acc[kw][j] += __shfl_xor(acc[kw][j], 32, 64);
+    }
+  }
+  if (do_bias) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      bacc[j] += __shfl_xor(bacc[j], 16, 64);
+      bacc[j] += __shfl_xor(bacc[j], 32, 64);
+    }
+  }
+
+  __shared__ float red[kLanesPerPix][kMaxK][8];
+  __shared__ float redb[kLanesPerPix][8];
+  if (threadIdx.x < kLanesPerPix) {
+#pragma unroll
+    for (int kw = 0; kw < kMaxK; ++kw)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) red[threadIdx.x][kw][j] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) redb[threadIdx.x][j] = 0.f;
+  }
+  __syncthreads();
+  const int lane = threadIdx.x % WAVE_SIZE;
+  if (lane < kLanesPerPix && active) {
+    const int cib = threadIdx.x % kLanesPerPix;
+    for (int kw = 0; kw < K; ++kw) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&red[cib][kw][j], acc[kw][j]);
+    }
+    if (do_bias) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) atomicAdd(&redb[cib][j], bacc[j]);
+    }
+  }
+  __syncthreads();
+
+  if (threadIdx.x < kLanesPerPix * 8) {
+    const int cib = threadIdx.x / 8;
+    const int j = threadIdx.x % 8;
+    const int c = (blockIdx.z * kLanesPerPix + cib) * 8 + j;
+    if (c < C) {
+      for (int kw = 0; kw < K; ++kw) {
+        atomicAdd(&dw[((long)kh * K + kw) * C + c], red[cib][kw][j]);
+      }
+      if (do_bias) atomicAdd(&dbias[c], redb[cib][j]);
+    }
+  }
+}
+
 template <typename scalar_t> struct ToHipD { using type = float; };
 template <> struct ToHipD<at::BFloat16> { using type = __hip_bfloat16; };
 
@@ -308,6 +444,17 @@ std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stri
   dim3 grid(sblocks, K, zc);
   TORCH_CHECK(K <= 9, "dwconv_bwd_weight: kernel size <= 9 supported");
   auto launch = [&](auto tag) {
+    if (stride == 1) {
+      // row-sliding variant: grid.x sized over rows (B*Ho)
+      long per_row = ((long)B * Ho + 15) / 16;
+      int rb = (int)std::min(std::max<long>(1024 / (K * zc) + 1, 16), per_row);
+      dim3 grid_s1(rb, K, zc);
+      hipLaunchKernelGGL((dwconv_bwd_weight_s1_kernel<__bf16, decltype(tag)::value>),
+          grid_s1, dim3(kThreads), 0, stream,
+          (const __bf16*)dy.data_ptr(), (const __bf16*)x.data_ptr(),
+          dw.data_ptr<float>(), dbias_ptr, B, H, W, C, Ho, Wo, (int)K, (int)pad);
+      return;
+    }
     hipLaunchKernelGGL((dwconv_bwd_weight_kernel<__bf16, decltype(tag)::value>),
         grid, dim3(kThreads), 0, stream,
         (const __bf16*)dy.data_ptr(), (const __bf16*)x.data_ptr(),
