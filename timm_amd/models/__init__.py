@@ -46,6 +46,7 @@ from .eva import *
 from .focalnet import *
 from .gcvit import *
 from .ghostnet import *
+from .hardcorenas import *
 from .hiera import *
 from .inception_next import *
 from .inception_v3 import *
