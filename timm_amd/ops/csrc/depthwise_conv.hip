@@ -261,7 +261,8 @@ void dwconv_bwd_weight_s1_kernel(
     float* __restrict__ dw,       // [K*K, C] fp32 (pre-zeroed)
     float* __restrict__ dbias,    // [C] fp32 (pre-zeroed) or null
     int B, int H, int W, int C,
-    int Ho, int Wo, int K, int pad) {
+    int Ho, int Wo, int pad) {
+  constexpr int K = kMaxK;  // compile-time K keeps the ring fully in registers
   constexpr int kLanesPerPix = 16;
   constexpr int kRowsPerBlock = kThreads / kLanesPerPix;  // 16
   const int c8 = C / 8;
@@ -292,38 +293,42 @@ void dwconv_bwd_weight_s1_kernel(
       const T* dyrow = dy + (((long)b * Ho + ho) * Wo) * C + c0;
       const T* xrow = x + (((long)b * H + (row_ok ? hi : 0)) * W) * C + c0;
 
-      // ring of x vectors covering [wo - pad, wo - pad + K - 1]
+      // K-deep window of x vectors; STATIC shifting only — runtime-indexed
+      // register arrays spill to scratch and serialize the whole loop
       bf16x8_t ring[kMaxK];
 #pragma unroll
       for (int i = 0; i < kMaxK; ++i) ring[i] = bf16x8_t{};
-      // preload taps for wo = 0: wi = -pad .. K-1-pad; slot i holds wi = wo - pad + i
-      for (int i = 0; i < K - 1; ++i) {
+      // preload taps for wo = 0 into slots 0..K-2 (slot i == tap kw=i)
+#pragma unroll
+      for (int i = 0; i < kMaxK - 1; ++i) {
+        if (i >= K - 1) break;
         const int wi = i - pad;
         if (row_ok && wi >= 0 && wi < W)
           ring[i] = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi * C);
       }
-      int head = K - 1;  // next slot to fill
       for (int wo = 0; wo < Wo; ++wo) {
         const int wi_new = wo - pad + K - 1;
         bf16x8_t nv = bf16x8_t{};
         if (row_ok && wi_new >= 0 && wi_new < W)
           nv = *reinterpret_cast<const bf16x8_t*>(xrow + (long)wi_new * C);
-        ring[head] = nv;
-        head = (head + 1 < K) ? head + 1 : 0;  // now head == slot of oldest (kw=0)
+        ring[K - 1] = nv;
 
         bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(dyrow + (long)wo * C);
         if (do_bias) {
 #pragma unroll
           for (int j = 0; j < 8; ++j) bacc[j] += (float)gv[j];
         }
-        if (!row_ok) continue;
-        int slot = head;
-        for (int kw = 0; kw < K; ++kw) {
-          bf16x8_t xv = ring[slot];
-          slot = (slot + 1 < K) ? slot + 1 : 0;
+        if (row_ok) {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)xv[j];
+          for (int kw = 0; kw < kMaxK; ++kw) {
+            if (kw >= K) break;
+#pragma unroll
+            for (int j = 0; j < 8; ++j) acc[kw][j] += (float)gv[j] * (float)ring[kw][j];
+          }
         }
+        // static shift: slot i <- slot i+1 (tap kw at next wo)
+#pragma unroll
+        for (int i = 0; i < kMaxK - 1; ++i) ring[i] = ring[i + 1];
       }
     }
   }
@@ -444,7 +449,7 @@ std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stri
   dim3 grid(sblocks, K, zc);
   TORCH_CHECK(K <= 9, "dwconv_bwd_weight: kernel size <= 9 supported");
   auto launch = [&](auto tag) {
-    if (stride == 1) {
+    if (stride == 1 && K == decltype(tag)::value) {
       // row-sliding variant: grid.x sized over rows (B*Ho)
       long per_row = ((long)B * Ho + 15) / 16;
       int rb = (int)std::min(std::max<long>(1024 / (K * zc) + 1, 16), per_row);
@@ -452,7 +457,7 @@ std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stri
       hipLaunchKernelGGL((dwconv_bwd_weight_s1_kernel<__bf16, decltype(tag)::value>),
           grid_s1, dim3(kThreads), 0, stream,
           (const __bf16*)dy.data_ptr(), (const __bf16*)x.data_ptr(),
-          dw.data_ptr<float>(), dbias_ptr, B, H, W, C, Ho, Wo, (int)K, (int)pad);
+          dw.data_ptr<float>(), dbias_ptr, B, H, W, C, Ho, Wo, (int)pad);
       return;
     }
     hipLaunchKernelGGL((dwconv_bwd_weight_kernel<__bf16, decltype(tag)::value>),
