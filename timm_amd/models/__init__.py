@@ -66,6 +66,7 @@ from .res2net import *
 from .resnest import *
 from .resnet import *
 from .rexnet import *
+from .selecsls import *
 from .sknet import *
 from .tiny_vit import *
 from .tresnet import *
