@@ -222,8 +222,12 @@ def get_norm_act_layer(norm_layer, act_layer=None):
         type_name = norm_layer.__name__.lower()
         if type_name.startswith('batchnorm'):
             norm_act_layer = BatchNormAct2d
+        elif type_name.startswith('groupnormact') or type_name.startswith('layernormact'):
+            norm_act_layer = norm_layer  # already a norm+act type
         elif type_name.startswith('groupnorm'):
             norm_act_layer = _group_norm_act_factory
+        elif type_name.startswith('evonorm') or type_name.startswith('filterresponsenorm'):
+            norm_act_layer = norm_layer  # fused norm+act by construction
         else:
             raise AssertionError(f"No equivalent norm_act layer for {type_name}")
 

@@ -64,6 +64,7 @@ from .pit import *
 from .pvt_v2 import *
 from .res2net import *
 from .resnest import *
+from .resnetv2 import *
 from .resnet import *
 from .rexnet import *
 from .selecsls import *

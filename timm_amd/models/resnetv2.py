@@ -1,0 +1,698 @@
+"""ResNet-V2 (pre-activation / Big Transfer) — MI355X-native implementation.
+
+Capability parity with reference `timm/models/resnetv2.py`: `PreActBasic`
+(:50), `PreActBottleneck` (:142), post-act `Bottleneck` (:243), conv/avg
+downsamples (:326/:359), `ResNetStage` (:398), BiT stems incl. 'fixed'
+SAME-pad approximation (:473), `ResNetV2` (:521) and BiT x1/x3 + timm d/t
+variants.  BiT models use weight-standardized convs (`layers/std_conv.py`) +
+GroupNormAct.
+"""
+from collections import OrderedDict
+from functools import partial
+from typing import Any, Callable, Dict, List, Optional, Tuple, Union
+
+import torch
+import torch.nn as nn
+
+from ..data.constants import IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD
+from ..layers import (
+    AvgPool2dSame, BatchNormAct2d, ClassifierHead, DropPath, GroupNormAct, StdConv2d, create_conv2d,
+    create_pool2d, get_act_layer, get_norm_act_layer, make_divisible,
+)
+from ._builder import build_model_with_cfg
+from ._features import feature_take_indices
+from ._manipulate import checkpoint_seq, named_apply
+from ._registry import generate_default_cfgs, register_model
+
+__all__ = ['ResNetV2']
+
+
+class PreActBasic(nn.Module):
+    """Pre-activation basic block."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: Optional[int] = None,
+            bottle_ratio: float = 1.0,
+            stride: int = 1,
+            dilation: int = 1,
+            first_dilation: Optional[int] = None,
+            groups: int = 1,
+            act_layer: Optional[Callable] = None,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            proj_layer: Optional[Callable] = None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        first_dilation = first_dilation or dilation
+        conv_layer = conv_layer or StdConv2d
+        norm_layer = norm_layer or partial(GroupNormAct, num_groups=32)
+        out_chs = out_chs or in_chs
+        mid_chs = make_divisible(out_chs * bottle_ratio)
+
+        if proj_layer is not None and (stride != 1 or first_dilation != dilation or in_chs != out_chs):
+            self.downsample = proj_layer(
+                in_chs, out_chs, stride=stride, dilation=dilation, first_dilation=first_dilation,
+                preact=True, conv_layer=conv_layer, norm_layer=norm_layer)
+        else:
+            self.downsample = None
+
+        self.norm1 = norm_layer(in_chs)
+        self.conv1 = conv_layer(in_chs, mid_chs, 3, stride=stride, dilation=first_dilation, groups=groups)
+        self.norm2 = norm_layer(mid_chs)
+        self.conv2 = conv_layer(mid_chs, out_chs, 3, dilation=dilation, groups=groups)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0 else nn.Identity()
+
+    def zero_init_last(self) -> None:
+        nn.init.zeros_(self.conv2.weight)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x_preact = self.norm1(x)
+
+        shortcut = x
+        if self.downsample is not None:
+            shortcut = self.downsample(x_preact)
+
+        x = self.conv1(x_preact)
+        x = self.conv2(self.norm2(x))
+        x = self.drop_path(x)
+        return x + shortcut
+
+
+class PreActBottleneck(nn.Module):
+    """Pre-activation bottleneck (Identity Mappings paper), stride on the 3x3."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: Optional[int] = None,
+            bottle_ratio: float = 0.25,
+            stride: int = 1,
+            dilation: int = 1,
+            first_dilation: Optional[int] = None,
+            groups: int = 1,
+            act_layer: Optional[Callable] = None,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            proj_layer: Optional[Callable] = None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        first_dilation = first_dilation or dilation
+        conv_layer = conv_layer or StdConv2d
+        norm_layer = norm_layer or partial(GroupNormAct, num_groups=32)
+        out_chs = out_chs or in_chs
+        mid_chs = make_divisible(out_chs * bottle_ratio)
+
+        if proj_layer is not None:
+            self.downsample = proj_layer(
+                in_chs, out_chs, stride=stride, dilation=dilation, first_dilation=first_dilation,
+                preact=True, conv_layer=conv_layer, norm_layer=norm_layer)
+        else:
+            self.downsample = None
+
+        self.norm1 = norm_layer(in_chs)
+        self.conv1 = conv_layer(in_chs, mid_chs, 1)
+        self.norm2 = norm_layer(mid_chs)
+        self.conv2 = conv_layer(mid_chs, mid_chs, 3, stride=stride, dilation=first_dilation, groups=groups)
+        self.norm3 = norm_layer(mid_chs)
+        self.conv3 = conv_layer(mid_chs, out_chs, 1)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0 else nn.Identity()
+
+    def zero_init_last(self) -> None:
+        nn.init.zeros_(self.conv3.weight)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x_preact = self.norm1(x)
+
+        shortcut = x
+        if self.downsample is not None:
+            shortcut = self.downsample(x_preact)
+
+        x = self.conv1(x_preact)
+        x = self.conv2(self.norm2(x))
+        x = self.conv3(self.norm3(x))
+        x = self.drop_path(x)
+        return x + shortcut
+
+
+class Bottleneck(nn.Module):
+    """Post-activation bottleneck for V2-style weight compat w/ V1 ordering."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: Optional[int] = None,
+            bottle_ratio: float = 0.25,
+            stride: int = 1,
+            dilation: int = 1,
+            first_dilation: Optional[int] = None,
+            groups: int = 1,
+            act_layer: Optional[Callable] = None,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            proj_layer: Optional[Callable] = None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        first_dilation = first_dilation or dilation
+        act_layer = act_layer or nn.ReLU
+        conv_layer = conv_layer or StdConv2d
+        norm_layer = norm_layer or partial(GroupNormAct, num_groups=32)
+        out_chs = out_chs or in_chs
+        mid_chs = make_divisible(out_chs * bottle_ratio)
+
+        if proj_layer is not None:
+            self.downsample = proj_layer(
+                in_chs, out_chs, stride=stride, dilation=dilation,
+                preact=False, conv_layer=conv_layer, norm_layer=norm_layer)
+        else:
+            self.downsample = None
+
+        self.conv1 = conv_layer(in_chs, mid_chs, 1)
+        self.norm1 = norm_layer(mid_chs)
+        self.conv2 = conv_layer(mid_chs, mid_chs, 3, stride=stride, dilation=first_dilation, groups=groups)
+        self.norm2 = norm_layer(mid_chs)
+        self.conv3 = conv_layer(mid_chs, out_chs, 1)
+        self.norm3 = norm_layer(out_chs, apply_act=False)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0 else nn.Identity()
+        self.act3 = act_layer(inplace=True)
+
+    def zero_init_last(self) -> None:
+        if getattr(self.norm3, 'weight', None) is not None:
+            nn.init.zeros_(self.norm3.weight)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shortcut = x
+        if self.downsample is not None:
+            shortcut = self.downsample(x)
+
+        x = self.conv1(x)
+        x = self.norm1(x)
+        x = self.conv2(x)
+        x = self.norm2(x)
+        x = self.conv3(x)
+        x = self.norm3(x)
+        x = self.drop_path(x)
+        x = self.act3(x + shortcut)
+        return x
+
+
+class DownsampleConv(nn.Module):
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int = 1,
+            dilation: int = 1,
+            first_dilation: Optional[int] = None,
+            preact: bool = True,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+    ):
+        super().__init__()
+        self.conv = conv_layer(in_chs, out_chs, 1, stride=stride)
+        self.norm = nn.Identity() if preact else norm_layer(out_chs, apply_act=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.norm(self.conv(x))
+
+
+class DownsampleAvg(nn.Module):
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int = 1,
+            dilation: int = 1,
+            first_dilation: Optional[int] = None,
+            preact: bool = True,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+    ):
+        super().__init__()
+        avg_stride = stride if dilation == 1 else 1
+        if stride > 1 or dilation > 1:
+            avg_pool_fn = AvgPool2dSame if avg_stride == 1 and dilation > 1 else nn.AvgPool2d
+            self.pool = avg_pool_fn(2, avg_stride, ceil_mode=True, count_include_pad=False)
+        else:
+            self.pool = nn.Identity()
+        self.conv = conv_layer(in_chs, out_chs, 1, stride=1)
+        self.norm = nn.Identity() if preact else norm_layer(out_chs, apply_act=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.norm(self.conv(self.pool(x)))
+
+
+class ResNetStage(nn.Module):
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int,
+            dilation: int,
+            depth: int,
+            bottle_ratio: float = 0.25,
+            groups: int = 1,
+            avg_down: bool = False,
+            block_dpr: Optional[List[float]] = None,
+            block_fn: Callable = PreActBottleneck,
+            act_layer: Optional[Callable] = None,
+            conv_layer: Optional[Callable] = None,
+            norm_layer: Optional[Callable] = None,
+            **block_kwargs: Any,
+    ):
+        super().__init__()
+        self.grad_checkpointing = False
+
+        first_dilation = 1 if dilation in (1, 2) else 2
+        layer_kwargs = dict(act_layer=act_layer, conv_layer=conv_layer, norm_layer=norm_layer)
+        proj_layer = DownsampleAvg if avg_down else DownsampleConv
+        prev_chs = in_chs
+        self.blocks = nn.Sequential()
+        for block_idx in range(depth):
+            drop_path_rate = block_dpr[block_idx] if block_dpr else 0.
+            stride = stride if block_idx == 0 else 1
+            self.blocks.add_module(str(block_idx), block_fn(
+                prev_chs,
+                out_chs,
+                stride=stride,
+                dilation=dilation,
+                bottle_ratio=bottle_ratio,
+                groups=groups,
+                first_dilation=first_dilation,
+                proj_layer=proj_layer,
+                drop_path_rate=drop_path_rate,
+                **layer_kwargs,
+                **block_kwargs,
+            ))
+            prev_chs = out_chs
+            first_dilation = dilation
+            proj_layer = None
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.grad_checkpointing and not torch.jit.is_scripting():
+            x = checkpoint_seq(self.blocks, x)
+        else:
+            x = self.blocks(x)
+        return x
+
+
+def is_stem_deep(stem_type: str) -> bool:
+    return any([s in stem_type for s in ('deep', 'tiered')])
+
+
+def create_resnetv2_stem(
+        in_chs: int,
+        out_chs: int = 64,
+        stem_type: str = '',
+        preact: bool = True,
+        conv_layer: Callable = StdConv2d,
+        norm_layer: Callable = partial(GroupNormAct, num_groups=32),
+) -> nn.Sequential:
+    stem = OrderedDict()
+    assert stem_type in ('', 'fixed', 'same', 'deep', 'deep_fixed', 'deep_same', 'tiered')
+
+    if is_stem_deep(stem_type):
+        if 'tiered' in stem_type:
+            stem_chs = (3 * out_chs // 8, out_chs // 2)
+        else:
+            stem_chs = (out_chs // 2, out_chs // 2)
+        stem['conv1'] = conv_layer(in_chs, stem_chs[0], kernel_size=3, stride=2)
+        stem['norm1'] = norm_layer(stem_chs[0])
+        stem['conv2'] = conv_layer(stem_chs[0], stem_chs[1], kernel_size=3, stride=1)
+        stem['norm2'] = norm_layer(stem_chs[1])
+        stem['conv3'] = conv_layer(stem_chs[1], out_chs, kernel_size=3, stride=1)
+        if not preact:
+            stem['norm3'] = norm_layer(out_chs)
+    else:
+        stem['conv'] = conv_layer(in_chs, out_chs, kernel_size=7, stride=2)
+        if not preact:
+            stem['norm'] = norm_layer(out_chs)
+
+    if 'fixed' in stem_type:
+        # 'fixed' SAME-padding approximation used by BiT models
+        stem['pad'] = nn.ConstantPad2d(1, 0.)
+        stem['pool'] = nn.MaxPool2d(kernel_size=3, stride=2, padding=0)
+    elif 'same' in stem_type:
+        stem['pool'] = create_pool2d('max', kernel_size=3, stride=2, padding='same')
+    else:
+        stem['pool'] = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+
+    return nn.Sequential(stem)
+
+
+class ResNetV2(nn.Module):
+    """Pre-activation ResNet / BiT (reference `resnetv2.py:521`)."""
+
+    def __init__(
+            self,
+            layers: List[int],
+            channels: Tuple[int, ...] = (256, 512, 1024, 2048),
+            num_classes: int = 1000,
+            in_chans: int = 3,
+            global_pool: str = 'avg',
+            output_stride: int = 32,
+            width_factor: int = 1,
+            stem_chs: int = 64,
+            stem_type: str = '',
+            avg_down: bool = False,
+            preact: bool = True,
+            basic: bool = False,
+            bottle_ratio: float = 0.25,
+            act_layer: Callable = nn.ReLU,
+            norm_layer: Callable = partial(GroupNormAct, num_groups=32),
+            conv_layer: Callable = StdConv2d,
+            drop_rate: float = 0.,
+            drop_path_rate: float = 0.,
+            zero_init_last: bool = False,
+    ):
+        super().__init__()
+        self.num_classes = num_classes
+        self.drop_rate = drop_rate
+        wf = width_factor
+        norm_layer = get_norm_act_layer(norm_layer, act_layer=act_layer)
+        act_layer = get_act_layer(act_layer)
+
+        self.feature_info = []
+        stem_chs = make_divisible(stem_chs * wf)
+        self.stem = create_resnetv2_stem(
+            in_chans, stem_chs, stem_type, preact, conv_layer=conv_layer, norm_layer=norm_layer)
+        stem_feat = ('stem.conv3' if is_stem_deep(stem_type) else 'stem.conv') if preact else 'stem.norm'
+        self.feature_info.append(dict(num_chs=stem_chs, reduction=2, module=stem_feat))
+
+        prev_chs = stem_chs
+        curr_stride = 4
+        dilation = 1
+        block_dprs = [x.tolist() for x in torch.linspace(0, drop_path_rate, sum(layers)).split(layers)]
+        if preact:
+            block_fn = PreActBasic if basic else PreActBottleneck
+        else:
+            assert not basic
+            block_fn = Bottleneck
+        self.stages = nn.Sequential()
+        for stage_idx, (d, c, bdpr) in enumerate(zip(layers, channels, block_dprs)):
+            out_chs = make_divisible(c * wf)
+            stride = 1 if stage_idx == 0 else 2
+            if curr_stride >= output_stride:
+                dilation *= stride
+                stride = 1
+            stage = ResNetStage(
+                prev_chs,
+                out_chs,
+                stride=stride,
+                dilation=dilation,
+                depth=d,
+                bottle_ratio=bottle_ratio,
+                avg_down=avg_down,
+                act_layer=act_layer,
+                conv_layer=conv_layer,
+                norm_layer=norm_layer,
+                block_dpr=bdpr,
+                block_fn=block_fn,
+            )
+            prev_chs = out_chs
+            curr_stride *= stride
+            self.feature_info += [dict(num_chs=prev_chs, reduction=curr_stride, module=f'stages.{stage_idx}')]
+            self.stages.add_module(str(stage_idx), stage)
+
+        self.num_features = self.head_hidden_size = prev_chs
+        self.norm = norm_layer(self.num_features) if preact else nn.Identity()
+        self.head = ClassifierHead(
+            self.num_features,
+            num_classes,
+            pool_type=global_pool,
+            drop_rate=self.drop_rate,
+            use_conv=True,
+        )
+
+        self.init_weights(zero_init_last=zero_init_last)
+
+    @torch.jit.ignore
+    def init_weights(self, zero_init_last: bool = True) -> None:
+        named_apply(partial(_init_weights, zero_init_last=zero_init_last), self)
+
+    @torch.jit.ignore
+    def group_matcher(self, coarse: bool = False) -> Dict[str, Any]:
+        return dict(
+            stem=r'^stem',
+            blocks=r'^stages\.(\d+)' if coarse else [
+                (r'^stages\.(\d+)\.blocks\.(\d+)', None),
+                (r'^norm', (99999,)),
+            ]
+        )
+
+    @torch.jit.ignore
+    def set_grad_checkpointing(self, enable: bool = True) -> None:
+        for s in self.stages:
+            s.grad_checkpointing = enable
+
+    @torch.jit.ignore
+    def get_classifier(self) -> nn.Module:
+        return self.head.fc
+
+    def reset_classifier(self, num_classes: int, global_pool: Optional[str] = None) -> None:
+        self.num_classes = num_classes
+        self.head.reset(num_classes, global_pool)
+
+    def forward_intermediates(
+            self,
+            x: torch.Tensor,
+            indices: Optional[Union[int, List[int]]] = None,
+            norm: bool = False,
+            stop_early: bool = False,
+            output_fmt: str = 'NCHW',
+            intermediates_only: bool = False,
+    ) -> Union[List[torch.Tensor], Tuple[torch.Tensor, List[torch.Tensor]]]:
+        assert output_fmt in ('NCHW',), 'Output shape must be NCHW.'
+        intermediates = []
+        take_indices, max_index = feature_take_indices(len(self.stages) + 1, indices)
+
+        feat_idx = 0
+        H, W = x.shape[-2:]
+        x_down = None
+        for stem in self.stem:
+            x = stem(x)
+            if x.shape[-2:] == (H // 2, W // 2):
+                x_down = x
+        if feat_idx in take_indices:
+            intermediates.append(x_down)
+        last_idx = len(self.stages)
+        if torch.jit.is_scripting() or not stop_early:
+            stages = self.stages
+        else:
+            stages = self.stages[:max_index]
+
+        for feat_idx, stage in enumerate(stages, start=1):
+            x = stage(x)
+            if feat_idx in take_indices:
+                if feat_idx == last_idx:
+                    x_inter = self.norm(x) if norm else x
+                    intermediates.append(x_inter)
+                else:
+                    intermediates.append(x)
+
+        if intermediates_only:
+            return intermediates
+
+        if feat_idx == last_idx:
+            x = self.norm(x)
+        return x, intermediates
+
+    def prune_intermediate_layers(
+            self,
+            indices: Union[int, List[int]] = 1,
+            prune_norm: bool = False,
+            prune_head: bool = True,
+    ):
+        take_indices, max_index = feature_take_indices(len(self.stages) + 1, indices)
+        self.stages = self.stages[:max_index]
+        if prune_norm:
+            self.norm = nn.Identity()
+        if prune_head:
+            self.reset_classifier(0, '')
+        return take_indices
+
+    def forward_features(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.stem(x)
+        x = self.stages(x)
+        x = self.norm(x)
+        return x
+
+    def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
+        return self.head(x, pre_logits=pre_logits) if pre_logits else self.head(x)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.forward_features(x)
+        x = self.forward_head(x)
+        return x
+
+
+def _init_weights(module: nn.Module, name: str = '', zero_init_last: bool = True) -> None:
+    if isinstance(module, nn.Linear) or ('head.fc' in name and isinstance(module, nn.Conv2d)):
+        nn.init.normal_(module.weight, mean=0.0, std=0.01)
+        nn.init.zeros_(module.bias)
+    elif isinstance(module, nn.Conv2d):
+        nn.init.kaiming_normal_(module.weight, mode='fan_out', nonlinearity='relu')
+        if module.bias is not None:
+            nn.init.zeros_(module.bias)
+    elif isinstance(module, (nn.BatchNorm2d, nn.LayerNorm, nn.GroupNorm)):
+        nn.init.ones_(module.weight)
+        nn.init.zeros_(module.bias)
+    elif zero_init_last and hasattr(module, 'zero_init_last'):
+        module.zero_init_last()
+
+
+def _create_resnetv2(variant: str, pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return build_model_with_cfg(
+        ResNetV2, variant, pretrained,
+        feature_cfg=dict(flatten_sequential=True),
+        **kwargs,
+    )
+
+
+def _create_resnetv2_bit(variant: str, pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2(
+        variant,
+        pretrained=pretrained,
+        stem_type='fixed',
+        conv_layer=partial(StdConv2d, eps=1e-8),
+        **kwargs,
+    )
+
+
+def _cfg(url: str = '', **kwargs: Any) -> Dict[str, Any]:
+    return {
+        'url': url, 'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': (7, 7),
+        'crop_pct': 0.875, 'interpolation': 'bilinear',
+        'mean': IMAGENET_INCEPTION_MEAN, 'std': IMAGENET_INCEPTION_STD,
+        'first_conv': 'stem.conv', 'classifier': 'head.fc',
+        **kwargs,
+    }
+
+
+default_cfgs = generate_default_cfgs({
+    'resnetv2_50x1_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
+    'resnetv2_50x1_bit.goog_in21k': _cfg(num_classes=21843),
+    'resnetv2_50x3_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
+    'resnetv2_101x1_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
+    'resnetv2_101x3_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
+    'resnetv2_152x2_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
+    'resnetv2_152x4_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 480, 480), pool_size=(15, 15), crop_pct=1.0),
+    'resnetv2_50.a1h_in1k': _cfg(interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288)),
+    'resnetv2_50d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_50t.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_101.a1h_in1k': _cfg(interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288)),
+    'resnetv2_101d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_152.untrained': _cfg(interpolation='bicubic'),
+    'resnetv2_152d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_50d_gn.ah_in1k': _cfg(interpolation='bicubic', first_conv='stem.conv1',
+                                    crop_pct=0.95, test_input_size=(3, 288, 288)),
+    'resnetv2_50d_evos.ah_in1k': _cfg(interpolation='bicubic', first_conv='stem.conv1',
+                                      crop_pct=0.95, test_input_size=(3, 288, 288)),
+})
+
+
+@register_model
+def resnetv2_50x1_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_50x1_bit', pretrained=pretrained, layers=[3, 4, 6, 3], width_factor=1, **kwargs)
+
+
+@register_model
+def resnetv2_50x3_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_50x3_bit', pretrained=pretrained, layers=[3, 4, 6, 3], width_factor=3, **kwargs)
+
+
+@register_model
+def resnetv2_101x1_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_101x1_bit', pretrained=pretrained, layers=[3, 4, 23, 3], width_factor=1, **kwargs)
+
+
+@register_model
+def resnetv2_101x3_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_101x3_bit', pretrained=pretrained, layers=[3, 4, 23, 3], width_factor=3, **kwargs)
+
+
+@register_model
+def resnetv2_152x2_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_152x2_bit', pretrained=pretrained, layers=[3, 8, 36, 3], width_factor=2, **kwargs)
+
+
+@register_model
+def resnetv2_152x4_bit(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    return _create_resnetv2_bit(
+        'resnetv2_152x4_bit', pretrained=pretrained, layers=[3, 8, 36, 3], width_factor=4, **kwargs)
+
+
+@register_model
+def resnetv2_50(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(layers=[3, 4, 6, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d)
+    return _create_resnetv2('resnetv2_50', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_50d(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(
+        layers=[3, 4, 6, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d,
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_50d', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_50t(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(
+        layers=[3, 4, 6, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d,
+        stem_type='tiered', avg_down=True)
+    return _create_resnetv2('resnetv2_50t', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_101(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(layers=[3, 4, 23, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d)
+    return _create_resnetv2('resnetv2_101', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_101d(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(
+        layers=[3, 4, 23, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d,
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_101d', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_152(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(layers=[3, 8, 36, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d)
+    return _create_resnetv2('resnetv2_152', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_152d(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(
+        layers=[3, 8, 36, 3], conv_layer=create_conv2d, norm_layer=BatchNormAct2d,
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_152d', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_50d_gn(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    model_args = dict(
+        layers=[3, 4, 6, 3], conv_layer=StdConv2d, norm_layer=partial(GroupNormAct, num_groups=32),
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_50d_gn', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_50d_evos(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    from ..layers import EvoNorm2dS0
+    model_args = dict(
+        layers=[3, 4, 6, 3], conv_layer=StdConv2d, norm_layer=EvoNorm2dS0,
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_50d_evos', pretrained=pretrained, **dict(model_args, **kwargs))
