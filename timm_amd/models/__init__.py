@@ -68,6 +68,7 @@ from .resnetv2 import *
 from .resnet import *
 from .rexnet import *
 from .selecsls import *
+from .starnet import *
 from .sknet import *
 from .tiny_vit import *
 from .tresnet import *
@@ -76,3 +77,5 @@ from .vgg import *
 from .xcit import *
 from .vision_transformer import *
 from .vovnet import *
+from .xception import *
+from .xception_aligned import *
