@@ -79,3 +79,5 @@ from .vision_transformer import *
 from .vovnet import *
 from .xception import *
 from .xception_aligned import *
+from .convit import *
+from .senet import *
