@@ -81,3 +81,4 @@ from .xception import *
 from .xception_aligned import *
 from .convit import *
 from .senet import *
+from .visformer import *
