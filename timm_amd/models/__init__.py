@@ -82,3 +82,4 @@ from .xception_aligned import *
 from .convit import *
 from .senet import *
 from .visformer import *
+from .fasternet import *
