@@ -83,3 +83,4 @@ from .convit import *
 from .senet import *
 from .visformer import *
 from .fasternet import *
+from .shvit import *
