@@ -84,3 +84,4 @@ from .senet import *
 from .visformer import *
 from .fasternet import *
 from .shvit import *
+from .dla import *
