@@ -85,3 +85,4 @@ from .visformer import *
 from .fasternet import *
 from .shvit import *
 from .dla import *
+from .cspnet import *
