@@ -86,3 +86,4 @@ from .fasternet import *
 from .shvit import *
 from .dla import *
 from .cspnet import *
+from .repvit import *
