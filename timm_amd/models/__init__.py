@@ -87,3 +87,4 @@ from .shvit import *
 from .dla import *
 from .cspnet import *
 from .repvit import *
+from .swiftformer import *
