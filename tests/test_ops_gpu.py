@@ -360,6 +360,10 @@ def test_depthwise_conv_nhwc(shape, k, stride):
     'resnest26d', 'res2net50_26w_4s', 'tresnet_m', 'ese_vovnet19b_dw', 'ghostnet_100',
     'densenet121', 'vgg11_bn', 'inception_v3', 'inception_next_atto', 'edgenext_xx_small',
     'focalnet_tiny_srf', 'dpn68', 'skresnet18', 'gernet_s',
+    # batch 2 family additions
+    'resnetv2_50x1_bit', 'cspresnet50', 'dla34', 'repvit_m0_9', 'swiftformer_xs',
+    'fasternet_t0', 'shvit_s1', 'visformer_tiny', 'convit_tiny', 'starnet_s1',
+    'xception41', 'legacy_seresnet18', 'selecsls42b',
 ])
 def test_model_gpu_vs_cpu(model_name):
     """Model forward on GPU (HIP kernels) vs CPU fp32 reference."""
