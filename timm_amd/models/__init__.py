@@ -88,3 +88,4 @@ from .dla import *
 from .cspnet import *
 from .repvit import *
 from .swiftformer import *
+from .sequencer import *
