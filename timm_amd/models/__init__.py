@@ -89,3 +89,4 @@ from .cspnet import *
 from .repvit import *
 from .swiftformer import *
 from .sequencer import *
+from .repghost import *
