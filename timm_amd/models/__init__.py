@@ -90,3 +90,4 @@ from .repvit import *
 from .swiftformer import *
 from .sequencer import *
 from .repghost import *
+from .rdnet import *
