@@ -6,7 +6,7 @@ from .adaptive_avgmax_pool import (
 from .attention import Attention, AttentionRope, maybe_add_mask
 from .attention_pool import AttentionPoolLatent
 from .blur_pool import BlurPool2d, create_aa
-from .classifier import ClassifierHead, NormMlpClassifierHead, create_classifier
+from .classifier import ClassifierHead, ClNormMlpClassifierHead, NormMlpClassifierHead, create_classifier
 from .cond_conv2d import CondConv2d, get_condconv_initializer
 from .config import (
     is_exportable, is_no_jit, is_scriptable, set_exportable, set_layer_config, set_no_jit, set_scriptable,

@@ -91,3 +91,4 @@ from .swiftformer import *
 from .sequencer import *
 from .repghost import *
 from .rdnet import *
+from .mambaout import *
