@@ -92,3 +92,4 @@ from .sequencer import *
 from .repghost import *
 from .rdnet import *
 from .mambaout import *
+from .crossvit import *
