@@ -93,3 +93,4 @@ from .repghost import *
 from .rdnet import *
 from .mambaout import *
 from .crossvit import *
+from .tnt import *

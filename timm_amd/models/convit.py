@@ -42,7 +42,8 @@ class GPSA(nn.Module):
 
     def forward(self, x):
         B, N, C = x.shape
-        if self.rel_indices is None or self.rel_indices.shape[1] != N:
+        if (self.rel_indices is None or self.rel_indices.shape[1] != N
+                or self.rel_indices.device != x.device or self.rel_indices.dtype != x.dtype):
             self.rel_indices = self.get_rel_indices(N)
         attn = self.get_attention(x)
         v = self.v(x).reshape(B, N, self.num_heads, C // self.num_heads).permute(0, 2, 1, 3)
