@@ -94,3 +94,4 @@ from .rdnet import *
 from .mambaout import *
 from .crossvit import *
 from .tnt import *
+from .nest import *
