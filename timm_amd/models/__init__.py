@@ -95,3 +95,4 @@ from .mambaout import *
 from .crossvit import *
 from .tnt import *
 from .nest import *
+from .levit import *
