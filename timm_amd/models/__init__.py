@@ -96,3 +96,4 @@ from .crossvit import *
 from .tnt import *
 from .nest import *
 from .levit import *
+from .efficientformer_v2 import *
