@@ -97,3 +97,4 @@ from .tnt import *
 from .nest import *
 from .levit import *
 from .efficientformer_v2 import *
+from .volo import *
