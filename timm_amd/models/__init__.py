@@ -98,3 +98,4 @@ from .nest import *
 from .levit import *
 from .efficientformer_v2 import *
 from .volo import *
+from .mvitv2 import *
