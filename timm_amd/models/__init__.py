@@ -99,3 +99,4 @@ from .levit import *
 from .efficientformer_v2 import *
 from .volo import *
 from .mvitv2 import *
+from .hgnet import *
