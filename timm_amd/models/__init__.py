@@ -100,3 +100,4 @@ from .efficientformer_v2 import *
 from .volo import *
 from .mvitv2 import *
 from .hgnet import *
+from .nextvit import *
