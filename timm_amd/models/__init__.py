@@ -101,3 +101,4 @@ from .volo import *
 from .mvitv2 import *
 from .hgnet import *
 from .nextvit import *
+from .coat import *
