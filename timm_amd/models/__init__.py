@@ -102,3 +102,4 @@ from .mvitv2 import *
 from .hgnet import *
 from .nextvit import *
 from .coat import *
+from .vision_transformer_relpos import *
