@@ -103,3 +103,4 @@ from .hgnet import *
 from .nextvit import *
 from .coat import *
 from .vision_transformer_relpos import *
+from .efficientvit_msra import *
