@@ -104,3 +104,4 @@ from .nextvit import *
 from .coat import *
 from .vision_transformer_relpos import *
 from .efficientvit_msra import *
+from .efficientvit_mit import *
