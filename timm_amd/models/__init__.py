@@ -105,3 +105,4 @@ from .coat import *
 from .vision_transformer_relpos import *
 from .efficientvit_msra import *
 from .efficientvit_mit import *
+from .hrnet import *
