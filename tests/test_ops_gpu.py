@@ -364,6 +364,11 @@ def test_depthwise_conv_nhwc(shape, k, stride):
     'resnetv2_50x1_bit', 'cspresnet50', 'dla34', 'repvit_m0_9', 'swiftformer_xs',
     'fasternet_t0', 'shvit_s1', 'visformer_tiny', 'convit_tiny', 'starnet_s1',
     'xception41', 'legacy_seresnet18', 'selecsls42b',
+    # batch 3 family additions
+    'levit_conv_128s', 'mvitv2_tiny', 'volo_d1_224', 'coat_lite_tiny', 'nest_tiny',
+    'efficientformerv2_s0', 'crossvit_tiny_240', 'tnt_s_patch16_224', 'sequencer2d_s',
+    'mambaout_femto', 'rdnet_tiny', 'repghostnet_100', 'hgnetv2_b0', 'nextvit_small',
+    'efficientvit_m0', 'efficientvit_b0', 'hrnet_w18_small', 'vit_relpos_small_patch16_224',
 ])
 def test_model_gpu_vs_cpu(model_name):
     """Model forward on GPU (HIP kernels) vs CPU fp32 reference."""
@@ -372,7 +377,8 @@ def test_model_gpu_vs_cpu(model_name):
     import timm_amd
     model = timm_amd.create_model(model_name, num_classes=10)
     model.eval()
-    x = torch.randn(2, 3, 224, 224)
+    in_sz = model.pretrained_cfg.get('input_size', (3, 224, 224))[-1] if hasattr(model, 'pretrained_cfg') else 224
+    x = torch.randn(2, 3, in_sz, in_sz)
     with torch.no_grad():
         y_cpu = model(x.float())
         m_gpu = model.to('cuda', torch.bfloat16)
