@@ -106,3 +106,4 @@ from .vision_transformer_relpos import *
 from .efficientvit_msra import *
 from .efficientvit_mit import *
 from .hrnet import *
+from .pnasnet import *
