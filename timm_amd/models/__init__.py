@@ -107,3 +107,4 @@ from .efficientvit_msra import *
 from .efficientvit_mit import *
 from .hrnet import *
 from .pnasnet import *
+from .nasnet import *
