@@ -108,3 +108,4 @@ from .efficientvit_mit import *
 from .hrnet import *
 from .pnasnet import *
 from .nasnet import *
+from .inception_v4 import *
