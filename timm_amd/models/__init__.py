@@ -109,3 +109,4 @@ from .hrnet import *
 from .pnasnet import *
 from .nasnet import *
 from .inception_v4 import *
+from .inception_resnet_v2 import *
