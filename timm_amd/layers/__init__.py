@@ -41,6 +41,7 @@ from .norm_act import (
 from .padding import get_padding, get_same_padding, pad_same
 from .patch_dropout import PatchDropout, patch_dropout_forward
 from .patch_embed import PatchEmbed, PatchEmbedWithSize, resample_patch_embed
+from .hybrid_embed import HybridEmbed, HybridEmbedWithSize
 from .pool2d_same import AvgPool2dSame, MaxPool2dSame, create_pool2d
 from .pos_embed_rel import (
     RelPosBias, RelPosBiasTf, RelPosMlp, gen_relative_position_index,

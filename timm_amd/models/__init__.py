@@ -110,3 +110,4 @@ from .pnasnet import *
 from .nasnet import *
 from .inception_v4 import *
 from .inception_resnet_v2 import *
+from .vision_transformer_hybrid import *
