@@ -111,3 +111,4 @@ from .nasnet import *
 from .inception_v4 import *
 from .inception_resnet_v2 import *
 from .vision_transformer_hybrid import *
+from .vision_transformer_sam import *
