@@ -112,3 +112,4 @@ from .inception_v4 import *
 from .inception_resnet_v2 import *
 from .vision_transformer_hybrid import *
 from .vision_transformer_sam import *
+from .swin_transformer_v2_cr import *
