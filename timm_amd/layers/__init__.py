@@ -59,6 +59,9 @@ from .evo_norm import (
     EvoNorm2dS2, EvoNorm2dS2a,
 )
 from .filter_response_norm import FilterResponseNormAct2d, FilterResponseNormTlu2d
+from .conv_self_attn import BottleneckAttn, HaloAttn, LambdaLayer, RelPos2d
+from .grid import ndgrid, meshgrid
+from .non_local_attn import NonLocalAttn, BatNonLocalAttn
 from .selective_kernel import SelectiveKernel, SelectiveKernelAttn
 from .space_to_depth import DepthToSpace, SpaceToDepth
 from .split_attn import RadixSoftmax, SplitAttn

@@ -3,9 +3,13 @@ import functools
 
 import torch
 
+from .conv_self_attn import BottleneckAttn, HaloAttn, LambdaLayer
 from .eca import EcaModule, CecaModule
 from .gather_excite import GatherExcite
 from .global_context import GlobalContext
+from .non_local_attn import NonLocalAttn, BatNonLocalAttn
+from .selective_kernel import SelectiveKernel
+from .split_attn import SplitAttn
 from .squeeze_excite import SEModule, EffectiveSEModule
 
 
@@ -33,6 +37,22 @@ def get_attn(attn_type):
                 module_cls = GlobalContext
             elif attn_type == 'gca':
                 module_cls = functools.partial(GlobalContext, fuse_add=True, fuse_scale=False)
+            # Attention / attention-like modules w/ significant compute
+            elif attn_type == 'sk':
+                module_cls = SelectiveKernel
+            elif attn_type == 'splat':
+                module_cls = SplitAttn
+            elif attn_type == 'nl':
+                module_cls = NonLocalAttn
+            elif attn_type == 'bat':
+                module_cls = BatNonLocalAttn
+            # Self-attention / attention-like modules w/ significant compute
+            elif attn_type == 'lambda':
+                module_cls = LambdaLayer
+            elif attn_type == 'bottleneck':
+                module_cls = BottleneckAttn
+            elif attn_type == 'halo':
+                module_cls = HaloAttn
             else:
                 assert False, "Invalid attn module (%s)" % attn_type
         elif isinstance(attn_type, bool):

@@ -31,6 +31,7 @@ from ._registry import (
 
 # architecture modules (registration happens at import time)
 from .beit import *
+from .byoanet import *
 from .byobnet import *
 from .cait import *
 from .convmixer import *

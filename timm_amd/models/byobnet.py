@@ -30,7 +30,7 @@ from ._features import feature_take_indices
 from ._manipulate import checkpoint_seq, named_apply
 from ._registry import generate_default_cfgs, register_model
 
-__all__ = ['ByobNet', 'ByoModelCfg', 'ByoBlockCfg']
+__all__ = ['ByobNet', 'ByoModelCfg', 'ByoBlockCfg', 'interleave_blocks']
 
 
 @dataclass
@@ -46,6 +46,8 @@ class ByoBlockCfg:
     # these override the model-level cfg for this block/stage
     attn_layer: Optional[str] = None
     attn_kwargs: Optional[Dict[str, Any]] = None
+    self_attn_layer: Optional[str] = None
+    self_attn_kwargs: Optional[Dict[str, Any]] = None
     block_kwargs: Optional[Dict[str, Any]] = None
 
 
@@ -67,6 +69,8 @@ class ByoModelCfg:
 
     attn_layer: Optional[str] = None
     attn_kwargs: dict = field(default_factory=lambda: dict())
+    self_attn_layer: Optional[str] = None
+    self_attn_kwargs: dict = field(default_factory=lambda: dict())
     block_kwargs: Dict[str, Any] = field(default_factory=lambda: dict())
 
 
@@ -82,6 +86,31 @@ def _rep_vgg_bcfg(
             return width // groups if block_idx % 2 else 0
     bcfg = tuple([ByoBlockCfg(type='rep', d=d, c=c * wf, gs=group_size) for d, c, wf in zip(d, c, wf)])
     return bcfg
+
+
+def interleave_blocks(
+        types: Tuple[str, str],
+        d: int,
+        every: Union[int, List[int]] = 1,
+        first: bool = False,
+        **kwargs,
+) -> Tuple[ByoBlockCfg, ...]:
+    """Alternate two block types through a stage (reference `byobnet.py`).
+
+    ``every`` as int -> one `types[1]` block every `every + 1` blocks
+    (starting at index `every`, or 0 if `first`); as a list -> explicit
+    indices for `types[1]`.
+    """
+    assert len(types) == 2
+    if isinstance(every, int):
+        every = list(range(0 if first else every, d, every + 1))
+        if not every:
+            every = [d - 1]
+    every = set(every)
+    return tuple(
+        ByoBlockCfg(type=types[1] if i in every else types[0], d=1, **kwargs)
+        for i in range(d)
+    )
 
 
 def expand_blocks_cfg(stage_blocks_cfg: Union[ByoBlockCfg, Sequence[ByoBlockCfg]]) -> List[ByoBlockCfg]:
@@ -109,6 +138,7 @@ class LayerFn:
     norm_act: Callable = BatchNormAct2d
     act: Callable = nn.ReLU
     attn: Optional[Callable] = None
+    self_attn: Optional[Callable] = None
 
 
 class DownsampleAvg(nn.Module):
@@ -525,12 +555,79 @@ class RepVggBlock(nn.Module):
         return kernel * t, beta - running_mean * gamma / std
 
 
+class SelfAttnBlock(nn.Module):
+    """ResNet-like self-attention block: 1x1 - [kxk] - self-attn - 1x1
+    (reference `byobnet.py:1040`).  Used for BoTNet / HaloNet / LambdaNet
+    stages via ``layers.self_attn`` (see layers/conv_self_attn.py)."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            kernel_size: int = 3,
+            stride: int = 1,
+            dilation: Tuple[int, int] = (1, 1),
+            bottle_ratio: float = 1.,
+            group_size: Optional[int] = None,
+            downsample: str = 'avg',
+            extra_conv: bool = False,
+            linear_out: bool = False,
+            bottle_in: bool = False,
+            post_attn_na: bool = True,
+            feat_size: Optional[Tuple[int, int]] = None,
+            layers: LayerFn = None,
+            drop_block: Callable = None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        assert layers is not None
+        mid_chs = make_divisible((in_chs if bottle_in else out_chs) * bottle_ratio)
+        groups = num_groups(group_size, mid_chs)
+
+        self.shortcut = create_shortcut(
+            downsample, in_chs, out_chs, stride=stride, dilation=dilation, apply_act=False, layers=layers)
+
+        self.conv1_1x1 = layers.conv_norm_act(in_chs, mid_chs, 1)
+        if extra_conv:
+            self.conv2_kxk = layers.conv_norm_act(
+                mid_chs, mid_chs, kernel_size, stride=stride, dilation=dilation[0],
+                groups=groups, drop_layer=drop_block)
+            stride = 1  # striding done via conv if enabled
+        else:
+            self.conv2_kxk = nn.Identity()
+        opt_kwargs = {} if feat_size is None else dict(feat_size=feat_size)
+        self.self_attn = layers.self_attn(mid_chs, stride=stride, **opt_kwargs)
+        self.post_attn = layers.norm_act(mid_chs) if post_attn_na else nn.Identity()
+        self.conv3_1x1 = layers.conv_norm_act(mid_chs, out_chs, 1, apply_act=False)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0. else nn.Identity()
+        self.act = nn.Identity() if linear_out else layers.act(inplace=True)
+
+    def init_weights(self, zero_init_last: bool = False):
+        if zero_init_last and self.shortcut is not None and getattr(self.conv3_1x1.bn, 'weight', None) is not None:
+            nn.init.zeros_(self.conv3_1x1.bn.weight)
+        if hasattr(self.self_attn, 'reset_parameters'):
+            self.self_attn.reset_parameters()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shortcut = x
+        x = self.conv1_1x1(x)
+        x = self.conv2_kxk(x)
+        x = self.self_attn(x)
+        x = self.post_attn(x)
+        x = self.conv3_1x1(x)
+        x = self.drop_path(x)
+        if self.shortcut is not None:
+            x = x + self.shortcut(shortcut)
+        return self.act(x)
+
+
 _block_registry = dict(
     basic=BasicBlock,
     bottle=BottleneckBlock,
     dark=DarkBlock,
     edge=EdgeBlock,
     rep=RepVggBlock,
+    self_attn=SelfAttnBlock,
 )
 
 
@@ -672,6 +769,17 @@ def update_block_kwargs(block_kwargs: Dict[str, Any], block_cfg: ByoBlockCfg, mo
             attn_layer = partial(get_attn(attn_layer), **attn_kwargs) if attn_layer is not None else None
         layer_fns = replace(layer_fns, attn=attn_layer)
 
+    self_attn_set = block_cfg.self_attn_layer is not None
+    if self_attn_set or block_cfg.self_attn_kwargs is not None:
+        if self_attn_set and not block_cfg.self_attn_layer:  # '' disables
+            self_attn_layer = None
+        else:
+            self_attn_kwargs = override_kwargs(block_cfg.self_attn_kwargs, model_cfg.self_attn_kwargs)
+            self_attn_layer = block_cfg.self_attn_layer or model_cfg.self_attn_layer
+            self_attn_layer = partial(get_attn(self_attn_layer), **self_attn_kwargs) \
+                if self_attn_layer is not None else None
+        layer_fns = replace(layer_fns, self_attn=self_attn_layer)
+
     block_kwargs['layers'] = layer_fns
     block_kwargs.update(override_kwargs(block_cfg.block_kwargs, model_cfg.block_kwargs))
 
@@ -685,11 +793,16 @@ def drop_blocks(drop_prob: float = 0., block_size: int = 3, num_stages: int = 4)
     return dbs
 
 
+def reduce_feat_size(feat_size, stride=2):
+    return None if feat_size is None else tuple([s // stride for s in feat_size])
+
+
 def create_byob_stages(
         cfg: ByoModelCfg,
         drop_path_rate: float,
         output_stride: int,
         stem_feat: Dict[str, Any],
+        feat_size: Optional[Tuple[int, int]] = None,
         drop_block_rate: float = 0.,
         drop_block_size: int = 3,
         layers: Optional[LayerFn] = None,
@@ -735,10 +848,15 @@ def create_byob_stages(
                 drop_path_rate=dpr[stage_idx][block_idx],
                 layers=layers,
             )
+            if block_cfg.type in ('self_attn',):
+                # blocks that need a concrete feature size (e.g. BottleneckAttn rel-pos)
+                block_kwargs['feat_size'] = feat_size
             block_kwargs_fn(block_kwargs, block_cfg=block_cfg, model_cfg=cfg)
             blocks += [create_block(block_cfg.type, **block_kwargs)]
             first_dilation = dilation
             prev_chs = out_chs
+            if stride > 1 and block_idx == 0:
+                feat_size = reduce_feat_size(feat_size, stride)
 
         stages += [nn.Sequential(*blocks)]
         prev_feat = dict(num_chs=prev_chs, reduction=net_stride, module=f'stages.{stage_idx}', stage=stage_idx + 1)
@@ -752,7 +870,8 @@ def get_layer_fns(cfg: ByoModelCfg) -> LayerFn:
     norm_act = get_norm_act_layer(norm_layer=cfg.norm_layer, act_layer=act)
     conv_norm_act = partial(ConvNormAct, norm_layer=cfg.norm_layer, act_layer=act)
     attn = partial(get_attn(cfg.attn_layer), **cfg.attn_kwargs) if cfg.attn_layer else None
-    return LayerFn(conv_norm_act=conv_norm_act, norm_act=norm_act, act=act, attn=attn)
+    self_attn = partial(get_attn(cfg.self_attn_layer), **cfg.self_attn_kwargs) if cfg.self_attn_layer else None
+    return LayerFn(conv_norm_act=conv_norm_act, norm_act=norm_act, act=act, attn=attn, self_attn=self_attn)
 
 
 class ByobNet(nn.Module):
@@ -765,6 +884,7 @@ class ByobNet(nn.Module):
             in_chans: int = 3,
             global_pool: Optional[str] = None,
             output_stride: int = 32,
+            img_size: Optional[Union[int, Tuple[int, int]]] = None,
             drop_rate: float = 0.,
             drop_block_rate: float = 0.,
             drop_block_size: int = 3,
@@ -779,6 +899,9 @@ class ByobNet(nn.Module):
 
         cfg = replace(cfg, **kwargs)
         layers = get_layer_fns(cfg)
+        if cfg.fixed_input_size:
+            assert img_size is not None, 'img_size argument is required for fixed input size model'
+        feat_size = to_2tuple(img_size) if img_size is not None else None
 
         self.feature_info = []
         if isinstance(cfg.stem_chs, (list, tuple)):
@@ -794,11 +917,13 @@ class ByobNet(nn.Module):
         )
         self.feature_info.extend(stem_feat[:-1])
 
+        feat_size = reduce_feat_size(feat_size, stride=stem_feat[-1]['reduction'])
         self.stages, stage_feat = create_byob_stages(
             cfg,
             drop_path_rate,
             output_stride,
             stem_feat[-1],
+            feat_size=feat_size,
             drop_block_rate=drop_block_rate,
             drop_block_size=drop_block_size,
             layers=layers,
