@@ -83,6 +83,7 @@ from .xception_aligned import *
 from .convit import *
 from .senet import *
 from .visformer import *
+from .fastvit import *
 from .fasternet import *
 from .shvit import *
 from .dla import *
