@@ -6,6 +6,8 @@ Includes `convert_sync_batchnorm` (reference `:167`) and FrozenBatchNormAct2d.
 """
 from typing import Optional, Type, Union
 
+import functools
+
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
@@ -190,10 +192,16 @@ def unfreeze_batch_norm_2d(module):
     return res
 
 
-_NORM_ACT_MAP = dict(
-    batchnorm=BatchNormAct2d,
-    batchnorm2d=BatchNormAct2d,
-)
+def _norm_act_map():
+    # lazy: GroupNormAct/LayerNormAct* defined below in this module
+    return dict(
+        batchnorm=BatchNormAct2d,
+        batchnorm2d=BatchNormAct2d,
+        groupnorm=GroupNormAct,
+        groupnorm1=functools.partial(GroupNormAct, num_groups=1),
+        layernorm=LayerNormAct,
+        layernorm2d=LayerNormAct2d,
+    )
 _NORM_ACT_TYPES = {BatchNormAct2d, SyncBatchNormAct, FrozenBatchNormAct2d}
 # has act_layer arg to define act type
 _NORM_ACT_REQUIRES_ARG = {BatchNormAct2d}
@@ -213,7 +221,7 @@ def get_norm_act_layer(norm_layer, act_layer=None):
         if not norm_layer:
             return None
         layer_name = norm_layer.replace('_', '').lower().split('-')[0]
-        norm_act_layer = _NORM_ACT_MAP[layer_name]
+        norm_act_layer = _norm_act_map()[layer_name]
     elif norm_layer in _NORM_ACT_TYPES:
         norm_act_layer = norm_layer
     elif isinstance(norm_layer, types.FunctionType):
@@ -231,7 +239,8 @@ def get_norm_act_layer(norm_layer, act_layer=None):
         else:
             raise AssertionError(f"No equivalent norm_act layer for {type_name}")
 
-    if norm_act_layer in _NORM_ACT_REQUIRES_ARG:
+    base = norm_act_layer.func if isinstance(norm_act_layer, functools.partial) else norm_act_layer
+    if base in _NORM_ACT_REQUIRES_ARG or base in (GroupNormAct, LayerNormAct, LayerNormAct2d):
         norm_act_kwargs.setdefault('act_layer', act_layer)
     if norm_act_kwargs:
         norm_act_layer = functools.partial(norm_act_layer, **norm_act_kwargs)
@@ -267,6 +276,22 @@ class LayerNormAct2d(LayerNorm2d):
 
     def forward(self, x):
         x = super().forward(x)
+        x = self.drop(x)
+        x = self.act(x)
+        return x
+
+
+class LayerNormAct(nn.LayerNorm):
+    """LayerNorm (last-dim) + act, for NLC tensors."""
+    def __init__(
+            self, normalization_shape, eps=1e-6, affine=True,
+            apply_act=True, act_layer=nn.ReLU, act_kwargs=None, inplace=True, drop_layer=None):
+        super().__init__(normalization_shape, eps=eps, elementwise_affine=affine)
+        self.drop = drop_layer() if drop_layer is not None else nn.Identity()
+        self.act = _create_act(act_layer, act_kwargs=act_kwargs, inplace=inplace, apply_act=apply_act)
+
+    def forward(self, x):
+        x = F.layer_norm(x, self.normalized_shape, self.weight, self.bias, self.eps)
         x = self.drop(x)
         x = self.act(x)
         return x

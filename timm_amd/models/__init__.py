@@ -99,6 +99,7 @@ from .tnt import *
 from .nest import *
 from .levit import *
 from .efficientformer_v2 import *
+from .vitamin import *
 from .volo import *
 from .mvitv2 import *
 from .hgnet import *
