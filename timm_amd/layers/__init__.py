@@ -4,6 +4,7 @@ from .adaptive_avgmax_pool import (
     adaptive_avgmax_pool2d, adaptive_catavgmax_pool2d, adaptive_pool_feat_mult, select_adaptive_pool2d,
 )
 from .attention import Attention, AttentionRope, maybe_add_mask
+from .attention2d import Attention2d, MultiQueryAttention2d, MultiQueryAttentionV2
 from .attention_pool import AttentionPoolLatent
 from .blur_pool import BlurPool2d, create_aa
 from .classifier import ClassifierHead, ClNormMlpClassifierHead, NormMlpClassifierHead, create_classifier

@@ -12,12 +12,13 @@ import torch.nn as nn
 from torch.nn import functional as F
 
 from ..layers import (
-    create_conv2d, DropPath, LayerScale2d, get_norm_act_layer, create_act_layer, make_divisible,
+    Attention2d, MultiQueryAttention2d, create_conv2d, DropPath, LayerScale2d, get_norm_act_layer,
+    create_act_layer, make_divisible, to_2tuple,
 )
 
 __all__ = [
     'SqueezeExcite', 'ConvBnAct', 'DepthwiseSeparableConv', 'InvertedResidual',
-    'CondConvResidual', 'EdgeResidual', 'UniversalInvertedResidual',
+    'CondConvResidual', 'EdgeResidual', 'UniversalInvertedResidual', 'MobileAttention',
 ]
 
 ModuleType = Type[nn.Module]
@@ -491,6 +492,112 @@ class EdgeResidual(nn.Module):
         x = self.se(x)
         x = self.conv_pwl(x)
         x = self.bn2(x)
+        if self.has_skip:
+            x = self.drop_path(x) + shortcut
+        return x
+
+class MobileAttention(nn.Module):
+    """Mobile attention block (MobileNetV4/V5): optional CPE dw-conv,
+    pre-norm, MQA or MHSA on the NCHW map, layer-scale + residual.
+
+    Reference `_efficientnet_blocks.py` MobileAttention.
+    """
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int = 1,
+            dw_kernel_size: int = 3,
+            dilation: int = 1,
+            group_size: int = 1,
+            pad_type: str = '',
+            num_heads: int = 8,
+            key_dim: int = 64,
+            value_dim: int = 64,
+            use_multi_query: bool = False,
+            query_strides: int = (1, 1),
+            kv_stride: int = 1,
+            cpe_dw_kernel_size: int = 3,
+            noskip: bool = False,
+            act_layer: ModuleType = nn.ReLU,
+            norm_layer: ModuleType = nn.BatchNorm2d,
+            aa_layer: Optional[ModuleType] = None,
+            drop_path_rate: float = 0.,
+            attn_drop: float = 0.0,
+            proj_drop: float = 0.0,
+            layer_scale_init_value: Optional[float] = 1e-5,
+            use_bias: bool = False,
+            use_cpe: bool = False,
+    ):
+        super().__init__()
+        norm_act_layer = get_norm_act_layer(norm_layer, act_layer)
+        self.has_skip = (stride == 1 and in_chs == out_chs) and not noskip
+        self.query_strides = to_2tuple(query_strides)
+        self.kv_stride = kv_stride
+        self.has_query_stride = any([s > 1 for s in self.query_strides])
+
+        # per-block conditional position encoding as a separable dw conv
+        if use_cpe:
+            self.conv_cpe_dw = create_conv2d(
+                in_chs, in_chs, kernel_size=cpe_dw_kernel_size,
+                dilation=dilation, depthwise=True, bias=True)
+        else:
+            self.conv_cpe_dw = None
+
+        self.norm = norm_act_layer(in_chs, apply_act=False)
+
+        if num_heads is None:
+            assert in_chs % key_dim == 0
+            num_heads = in_chs // key_dim
+
+        if use_multi_query:
+            self.attn = MultiQueryAttention2d(
+                in_chs,
+                dim_out=out_chs,
+                num_heads=num_heads,
+                key_dim=key_dim,
+                value_dim=value_dim,
+                query_strides=query_strides,
+                kv_stride=kv_stride,
+                dw_kernel_size=dw_kernel_size,
+                dilation=dilation,
+                padding=pad_type,
+                attn_drop=attn_drop,
+                proj_drop=proj_drop,
+                norm_layer=norm_layer,
+            )
+        else:
+            self.attn = Attention2d(
+                in_chs,
+                dim_out=out_chs,
+                num_heads=num_heads,
+                attn_drop=attn_drop,
+                proj_drop=proj_drop,
+                bias=use_bias,
+            )
+
+        if layer_scale_init_value is not None:
+            self.layer_scale = LayerScale2d(out_chs, layer_scale_init_value)
+        else:
+            self.layer_scale = nn.Identity()
+
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate else nn.Identity()
+
+    def feature_info(self, location):
+        if location == 'expansion':
+            return dict(module='conv_pw', hook_type='forward_pre', num_chs=self.conv_pw.in_channels)
+        else:
+            return dict(module='', num_chs=self.conv_pw.out_channels)
+
+    def forward(self, x):
+        if self.conv_cpe_dw is not None:
+            x = x + self.conv_cpe_dw(x)
+
+        shortcut = x
+        x = self.norm(x)
+        x = self.attn(x)
+        x = self.layer_scale(x)
         if self.has_skip:
             x = self.drop_path(x) + shortcut
         return x

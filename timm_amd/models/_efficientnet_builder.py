@@ -18,7 +18,7 @@ import torch.nn as nn
 from ..layers import get_act_layer, get_norm_act_layer, make_divisible
 from ._efficientnet_blocks import (
     SqueezeExcite, ConvBnAct, DepthwiseSeparableConv, InvertedResidual, CondConvResidual,
-    EdgeResidual, UniversalInvertedResidual,
+    EdgeResidual, UniversalInvertedResidual, MobileAttention,
 )
 from ._manipulate import named_modules
 
@@ -193,6 +193,26 @@ def _decode_block_str(block_str):
             dw_kernel_size_end=end_kernel_size,  # overload pw ks arg for dw end
             exp_ratio=float(options['e']),
             se_ratio=float(options.get('se', 0.)),
+            noskip=skip is False,
+        ))
+    elif block_type == 'mha':
+        kv_dim = int(options['d'])
+        block_args.update(dict(
+            dw_kernel_size=_parse_ksize(options['k']),
+            num_heads=int(options['h']),
+            key_dim=kv_dim,
+            value_dim=kv_dim,
+            kv_stride=int(options.get('v', 1)),
+            noskip=skip is False,
+        ))
+    elif block_type == 'mqa':
+        kv_dim = int(options['d'])
+        block_args.update(dict(
+            dw_kernel_size=_parse_ksize(options['k']),
+            num_heads=int(options['h']),
+            key_dim=kv_dim,
+            value_dim=kv_dim,
+            kv_stride=int(options.get('v', 1)),
             noskip=skip is False,
         ))
     else:
@@ -380,6 +400,12 @@ class EfficientNetBuilder:
         elif bt == 'uir':
             _log_info_if('  UniversalInvertedResidual {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
             block = UniversalInvertedResidual(**ba, layer_scale_init_value=self.layer_scale_init_value)
+        elif bt == 'mqa':
+            _log_info_if('  MobileMultiQueryAttention {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
+            block = MobileAttention(**ba, use_multi_query=True, layer_scale_init_value=self.layer_scale_init_value)
+        elif bt == 'mha':
+            _log_info_if('  MobileMultiHeadAttention {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
+            block = MobileAttention(**ba, layer_scale_init_value=self.layer_scale_init_value)
         else:
             assert False, 'Unknown block type (%s) while building model.' % bt
 

@@ -384,77 +384,190 @@ def _gen_mobilenet_v3(variant, channel_multiplier=1.0, depth_multiplier=1.0, pre
 
 
 def _gen_mobilenet_v4(variant, channel_multiplier=1.0, group_size=None, pretrained=False, **kwargs):
-    """Creates a MobileNet-V4 model (conv-only variants)."""
+    """Creates a MobileNet-V4 model (conv and hybrid variants, reference `mobilenetv3.py:_gen_mobilenet_v4`).
+
+    Hybrid variants interleave UIR blocks with MQA attention ('mqa_*' block
+    strings -> MobileAttention w/ MultiQueryAttention2d)."""
     num_features = 1280
-    if 'medium' in variant:
-        stem_size = 32
-        act_layer = resolve_act_layer(kwargs, 'relu')
-        arch_def = [
-            # stage 0, 112x112 in
-            ['er_r1_k3_s2_e4_c48'],
-            # stage 1, 56x56 in
-            ['uir_r1_a3_k5_s2_e4_c80', 'uir_r1_a3_k3_s1_e2_c80'],
-            # stage 2, 28x28 in
-            [
-                'uir_r1_a3_k5_s2_e6_c160',
-                'uir_r2_a3_k3_s1_e4_c160',
-                'uir_r1_a3_k5_s1_e4_c160',
-                'uir_r1_a3_k3_s1_e4_c160',
-                'uir_r1_a3_k0_s1_e4_c160',
-                'uir_r1_a0_k0_s1_e2_c160',
-                'uir_r1_a3_k0_s1_e4_c160',
-            ],
-            # stage 3, 14x14in
-            [
-                'uir_r1_a5_k5_s2_e6_c256',
-                'uir_r1_a5_k5_s1_e4_c256',
-                'uir_r2_a3_k5_s1_e4_c256',
-                'uir_r1_a0_k0_s1_e4_c256',
-                'uir_r1_a3_k0_s1_e4_c256',
-                'uir_r1_a3_k5_s1_e2_c256',
-                'uir_r1_a5_k5_s1_e4_c256',
-                'uir_r2_a0_k0_s1_e4_c256',
-                'uir_r1_a5_k0_s1_e2_c256',
-            ],
-            # stage 4, 7x7
-            ['cn_r1_k1_s1_c960'],
-        ]
-    else:  # small
-        stem_size = 32
-        act_layer = resolve_act_layer(kwargs, 'relu')
-        arch_def = [
-            # stage 0, 112x112 in
-            ['cn_r1_k3_s2_e1_c32', 'cn_r1_k1_s1_e1_c32'],
-            # stage 1, 56x56 in
-            ['cn_r1_k3_s2_e1_c96', 'cn_r1_k1_s1_e1_c64'],
-            # stage 2, 28x28 in
-            [
-                'uir_r1_a5_k5_s2_e3_c96',
-                'uir_r4_a0_k3_s1_e2_c96',
-                'uir_r1_a3_k0_s1_e4_c96',
-            ],
-            # stage 3, 14x14 in
-            [
-                'uir_r1_a3_k3_s2_e6_c128',
-                'uir_r1_a5_k5_s1_e4_c128',
-                'uir_r1_a0_k5_s1_e4_c128',
-                'uir_r1_a0_k5_s1_e3_c128',
-                'uir_r2_a0_k3_s1_e4_c128',
-            ],
-            # stage 4, 7x7
-            ['cn_r1_k1_s1_c960'],
-        ]
+    if 'hybrid' in variant:
+        layer_scale_init_value = 1e-5
+        if 'medium' in variant:
+            stem_size = 32
+            act_layer = resolve_act_layer(kwargs, 'relu')
+            arch_def = [
+                # stage 0, 112x112 in
+                ['er_r1_k3_s2_e4_c48'],
+                # stage 1, 56x56 in
+                ['uir_r1_a3_k5_s2_e4_c80', 'uir_r1_a3_k3_s1_e2_c80'],
+                # stage 2, 28x28 in
+                [
+                    'uir_r1_a3_k5_s2_e6_c160',
+                    'uir_r1_a0_k0_s1_e2_c160',
+                    'uir_r1_a3_k3_s1_e4_c160',
+                    'uir_r1_a3_k5_s1_e4_c160',
+                    'mqa_r1_k3_h4_s1_v2_d64_c160',
+                    'uir_r1_a3_k3_s1_e4_c160',
+                    'mqa_r1_k3_h4_s1_v2_d64_c160',
+                    'uir_r1_a3_k0_s1_e4_c160',
+                    'mqa_r1_k3_h4_s1_v2_d64_c160',
+                    'uir_r1_a3_k3_s1_e4_c160',
+                    'mqa_r1_k3_h4_s1_v2_d64_c160',
+                    'uir_r1_a3_k0_s1_e4_c160',
+                ],
+                # stage 3, 14x14 in
+                [
+                    'uir_r1_a5_k5_s2_e6_c256',
+                    'uir_r1_a5_k5_s1_e4_c256',
+                    'uir_r2_a3_k5_s1_e4_c256',
+                    'uir_r1_a0_k0_s1_e2_c256',
+                    'uir_r1_a3_k5_s1_e2_c256',
+                    'uir_r1_a0_k0_s1_e2_c256',
+                    'uir_r1_a0_k0_s1_e4_c256',
+                    'mqa_r1_k3_h4_s1_d64_c256',
+                    'uir_r1_a3_k0_s1_e4_c256',
+                    'mqa_r1_k3_h4_s1_d64_c256',
+                    'uir_r1_a5_k5_s1_e4_c256',
+                    'mqa_r1_k3_h4_s1_d64_c256',
+                    'uir_r1_a5_k0_s1_e4_c256',
+                    'mqa_r1_k3_h4_s1_d64_c256',
+                    'uir_r1_a5_k0_s1_e4_c256',
+                ],
+                # stage 4, 7x7
+                ['cn_r1_k1_s1_c960'],
+            ]
+        elif 'large' in variant:
+            stem_size = 24
+            act_layer = resolve_act_layer(kwargs, 'gelu')
+            arch_def = [
+                ['er_r1_k3_s2_e4_c48'],
+                ['uir_r1_a3_k5_s2_e4_c96', 'uir_r1_a3_k3_s1_e4_c96'],
+                [
+                    'uir_r1_a3_k5_s2_e4_c192',
+                    'uir_r3_a3_k3_s1_e4_c192',
+                    'uir_r1_a3_k5_s1_e4_c192',
+                    'uir_r2_a5_k3_s1_e4_c192',
+                    'mqa_r1_k3_h8_s1_v2_d48_c192',
+                    'uir_r1_a5_k3_s1_e4_c192',
+                    'mqa_r1_k3_h8_s1_v2_d48_c192',
+                    'uir_r1_a5_k3_s1_e4_c192',
+                    'mqa_r1_k3_h8_s1_v2_d48_c192',
+                    'uir_r1_a5_k3_s1_e4_c192',
+                    'mqa_r1_k3_h8_s1_v2_d48_c192',
+                    'uir_r1_a3_k0_s1_e4_c192',
+                ],
+                [
+                    'uir_r4_a5_k5_s2_e4_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                    'uir_r1_a5_k3_s1_e4_c512',
+                    'uir_r2_a5_k0_s1_e4_c512',
+                    'uir_r1_a5_k3_s1_e4_c512',
+                    'uir_r1_a5_k5_s1_e4_c512',
+                    'mqa_r1_k3_h8_s1_d64_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                    'mqa_r1_k3_h8_s1_d64_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                    'mqa_r1_k3_h8_s1_d64_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                    'mqa_r1_k3_h8_s1_d64_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                ],
+                ['cn_r1_k1_s1_c960'],
+            ]
+        else:
+            assert False, f'Unknown variant {variant}.'
+    else:
+        layer_scale_init_value = None
+        if 'small' in variant:
+            stem_size = 32
+            act_layer = resolve_act_layer(kwargs, 'relu')
+            arch_def = [
+                # stage 0, 112x112 in
+                ['cn_r1_k3_s2_e1_c32', 'cn_r1_k1_s1_e1_c32'],
+                # stage 1, 56x56 in
+                ['cn_r1_k3_s2_e1_c96', 'cn_r1_k1_s1_e1_c64'],
+                # stage 2, 28x28 in
+                [
+                    'uir_r1_a5_k5_s2_e3_c96',
+                    'uir_r4_a0_k3_s1_e2_c96',
+                    'uir_r1_a3_k0_s1_e4_c96',
+                ],
+                # stage 3, 14x14 in
+                [
+                    'uir_r1_a3_k3_s2_e6_c128',
+                    'uir_r1_a5_k5_s1_e4_c128',
+                    'uir_r1_a0_k5_s1_e4_c128',
+                    'uir_r1_a0_k5_s1_e3_c128',
+                    'uir_r2_a0_k3_s1_e4_c128',
+                ],
+                # stage 4, 7x7
+                ['cn_r1_k1_s1_c960'],
+            ]
+        elif 'medium' in variant:
+            stem_size = 32
+            act_layer = resolve_act_layer(kwargs, 'relu')
+            arch_def = [
+                ['er_r1_k3_s2_e4_c48'],
+                ['uir_r1_a3_k5_s2_e4_c80', 'uir_r1_a3_k3_s1_e2_c80'],
+                [
+                    'uir_r1_a3_k5_s2_e6_c160',
+                    'uir_r2_a3_k3_s1_e4_c160',
+                    'uir_r1_a3_k5_s1_e4_c160',
+                    'uir_r1_a3_k3_s1_e4_c160',
+                    'uir_r1_a3_k0_s1_e4_c160',
+                    'uir_r1_a0_k0_s1_e2_c160',
+                    'uir_r1_a3_k0_s1_e4_c160',
+                ],
+                [
+                    'uir_r1_a5_k5_s2_e6_c256',
+                    'uir_r1_a5_k5_s1_e4_c256',
+                    'uir_r2_a3_k5_s1_e4_c256',
+                    'uir_r1_a0_k0_s1_e4_c256',
+                    'uir_r1_a3_k0_s1_e4_c256',
+                    'uir_r1_a3_k5_s1_e2_c256',
+                    'uir_r1_a5_k5_s1_e4_c256',
+                    'uir_r2_a0_k0_s1_e4_c256',
+                    'uir_r1_a5_k0_s1_e2_c256',
+                ],
+                ['cn_r1_k1_s1_c960'],
+            ]
+        elif 'large' in variant:
+            stem_size = 24
+            act_layer = resolve_act_layer(kwargs, 'relu')
+            arch_def = [
+                ['er_r1_k3_s2_e4_c48'],
+                ['uir_r1_a3_k5_s2_e4_c96', 'uir_r1_a3_k3_s1_e4_c96'],
+                [
+                    'uir_r1_a3_k5_s2_e4_c192',
+                    'uir_r3_a3_k3_s1_e4_c192',
+                    'uir_r1_a3_k5_s1_e4_c192',
+                    'uir_r5_a5_k3_s1_e4_c192',
+                    'uir_r1_a3_k0_s1_e4_c192',
+                ],
+                [
+                    'uir_r4_a5_k5_s2_e4_c512',
+                    'uir_r1_a5_k0_s1_e4_c512',
+                    'uir_r1_a5_k3_s1_e4_c512',
+                    'uir_r2_a5_k0_s1_e4_c512',
+                    'uir_r1_a5_k3_s1_e4_c512',
+                    'uir_r1_a5_k5_s1_e4_c512',
+                    'uir_r3_a5_k0_s1_e4_c512',
+                ],
+                ['cn_r1_k1_s1_c960'],
+            ]
+        else:
+            assert False, f'Unknown variant {variant}.'
+
     model_kwargs = dict(
         block_args=decode_arch_def(arch_def, group_size=group_size),
         head_bias=False,
         head_norm=True,
         num_features=num_features,
         stem_size=stem_size,
-        fix_stem=channel_multiplier < 0.75,
+        fix_stem=channel_multiplier < 1.0,
         round_chs_fn=partial(round_channels, multiplier=channel_multiplier),
         norm_layer=kwargs.pop('norm_layer', None) or partial(nn.BatchNorm2d, **resolve_bn_args(kwargs)),
         act_layer=act_layer,
-        layer_scale_init_value=1e-5,
+        layer_scale_init_value=layer_scale_init_value,
         **kwargs,
     )
     model = _create_mnv3(variant, pretrained, **model_kwargs)
@@ -477,7 +590,21 @@ default_cfgs = generate_default_cfgs({
     'mobilenetv3_large_075.untrained': _cfg(),
     'mobilenetv3_small_100.lamb_in1k': _cfg(interpolation='bicubic'),
     'mobilenetv3_small_075.lamb_in1k': _cfg(interpolation='bicubic'),
+    'mobilenetv4_conv_small_035.untrained': _cfg(interpolation='bicubic', test_input_size=(3, 256, 256), test_crop_pct=0.95),
+    'mobilenetv4_conv_small_050.e3000_r224_in1k': _cfg(interpolation='bicubic', test_input_size=(3, 256, 256), test_crop_pct=0.95),
     'mobilenetv4_conv_small.e2400_r224_in1k': _cfg(interpolation='bicubic', test_input_size=(3, 256, 256), test_crop_pct=0.95),
+    'mobilenetv4_conv_large.e600_r384_in1k': _cfg(
+        input_size=(3, 384, 384), pool_size=(12, 12), test_input_size=(3, 448, 448), test_crop_pct=1.0, interpolation='bicubic'),
+    'mobilenetv4_conv_aa_medium.untrained': _cfg(input_size=(3, 256, 256), pool_size=(8, 8), crop_pct=0.95),
+    'mobilenetv4_conv_blur_medium.e500_r224_in1k': _cfg(test_input_size=(3, 256, 256), test_crop_pct=0.95, interpolation='bicubic'),
+    'mobilenetv4_conv_aa_large.e600_r384_in12k_ft_in1k': _cfg(
+        input_size=(3, 384, 384), pool_size=(12, 12), test_input_size=(3, 480, 480), test_crop_pct=1.0, interpolation='bicubic'),
+    'mobilenetv4_hybrid_medium.ix_e550_r256_in1k': _cfg(
+        input_size=(3, 256, 256), pool_size=(8, 8), test_input_size=(3, 320, 320), test_crop_pct=1.0, interpolation='bicubic'),
+    'mobilenetv4_hybrid_large.ix_e600_r384_in1k': _cfg(
+        input_size=(3, 384, 384), pool_size=(12, 12), test_input_size=(3, 448, 448), test_crop_pct=1.0, interpolation='bicubic'),
+    'mobilenetv4_hybrid_medium_075.untrained': _cfg(input_size=(3, 256, 256), pool_size=(8, 8), crop_pct=0.95),
+    'mobilenetv4_hybrid_large_075.untrained': _cfg(input_size=(3, 256, 256), pool_size=(8, 8), crop_pct=0.95),
     'mobilenetv4_conv_medium.e500_r256_in1k': _cfg(
         interpolation='bicubic', input_size=(3, 256, 256), pool_size=(8, 8), test_input_size=(3, 320, 320), test_crop_pct=1.0),
 })
@@ -516,4 +643,64 @@ def mobilenetv4_conv_small(pretrained=False, **kwargs) -> MobileNetV3:
 @register_model
 def mobilenetv4_conv_medium(pretrained=False, **kwargs) -> MobileNetV3:
     model = _gen_mobilenet_v4('mobilenetv4_conv_medium', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_small_035(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_small_035', 0.35, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_small_050(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_small_050', 0.50, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_large(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_large', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_hybrid_medium(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_hybrid_medium', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_hybrid_large(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_hybrid_large', 1.0, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_aa_medium(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_aa_medium', 1.0, pretrained=pretrained, aa_layer='avg', **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_blur_medium(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_blur_medium', 1.0, pretrained=pretrained, aa_layer='blurpc', **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_conv_aa_large(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_conv_aa_large', 1.0, pretrained=pretrained, aa_layer='avg', **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_hybrid_medium_075(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_hybrid_medium_075', 0.75, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv4_hybrid_large_075(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v4('mobilenetv4_hybrid_large_075', 0.75, pretrained=pretrained, **kwargs)
     return model
