@@ -36,7 +36,7 @@ from .norm import (
     SimpleNorm, SimpleNorm2d,
 )
 from .norm_act import (
-    BatchNormAct2d, FrozenBatchNormAct2d, GroupNormAct, LayerNormAct, LayerNormAct2d, SyncBatchNormAct,
+    BatchNormAct2d, FrozenBatchNormAct2d, GroupNormAct, LayerNormAct, LayerNormAct2d, RmsNormAct, RmsNormAct2d, SyncBatchNormAct,
     convert_sync_batchnorm, freeze_batch_norm_2d, get_norm_act_layer, unfreeze_batch_norm_2d,
 )
 from .padding import get_padding, get_same_padding, pad_same

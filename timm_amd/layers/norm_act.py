@@ -201,6 +201,8 @@ def _norm_act_map():
         groupnorm1=functools.partial(GroupNormAct, num_groups=1),
         layernorm=LayerNormAct,
         layernorm2d=LayerNormAct2d,
+        rmsnorm=RmsNormAct,
+        rmsnorm2d=RmsNormAct2d,
     )
 _NORM_ACT_TYPES = {BatchNormAct2d, SyncBatchNormAct, FrozenBatchNormAct2d}
 # has act_layer arg to define act type
@@ -230,8 +232,17 @@ def get_norm_act_layer(norm_layer, act_layer=None):
         type_name = norm_layer.__name__.lower()
         if type_name.startswith('batchnorm'):
             norm_act_layer = BatchNormAct2d
-        elif type_name.startswith('groupnormact') or type_name.startswith('layernormact'):
+        elif type_name.startswith('groupnormact') or type_name.startswith('layernormact') \
+                or type_name.startswith('rmsnormact'):
             norm_act_layer = norm_layer  # already a norm+act type
+        elif type_name == 'rmsnorm2d':
+            norm_act_layer = RmsNormAct2d
+        elif type_name == 'rmsnorm':
+            norm_act_layer = RmsNormAct
+        elif type_name == 'layernorm2d':
+            norm_act_layer = LayerNormAct2d
+        elif type_name == 'layernorm':
+            norm_act_layer = LayerNormAct
         elif type_name.startswith('groupnorm'):
             norm_act_layer = _group_norm_act_factory
         elif type_name.startswith('evonorm') or type_name.startswith('filterresponsenorm'):
@@ -240,7 +251,7 @@ def get_norm_act_layer(norm_layer, act_layer=None):
             raise AssertionError(f"No equivalent norm_act layer for {type_name}")
 
     base = norm_act_layer.func if isinstance(norm_act_layer, functools.partial) else norm_act_layer
-    if base in _NORM_ACT_REQUIRES_ARG or base in (GroupNormAct, LayerNormAct, LayerNormAct2d):
+    if base in _NORM_ACT_REQUIRES_ARG or base in (GroupNormAct, LayerNormAct, LayerNormAct2d, RmsNormAct, RmsNormAct2d):
         norm_act_kwargs.setdefault('act_layer', act_layer)
     if norm_act_kwargs:
         norm_act_layer = functools.partial(norm_act_layer, **norm_act_kwargs)
@@ -292,6 +303,38 @@ class LayerNormAct(nn.LayerNorm):
 
     def forward(self, x):
         x = F.layer_norm(x, self.normalized_shape, self.weight, self.bias, self.eps)
+        x = self.drop(x)
+        x = self.act(x)
+        return x
+
+
+class RmsNormAct(RmsNorm):
+    """RMSNorm + act for NLC tensors."""
+    def __init__(
+            self, num_channels, eps=1e-6, affine=True,
+            apply_act=True, act_layer=nn.ReLU, act_kwargs=None, inplace=True, drop_layer=None):
+        super().__init__(num_channels, eps=eps, affine=affine)
+        self.drop = drop_layer() if drop_layer is not None else nn.Identity()
+        self.act = _create_act(act_layer, act_kwargs=act_kwargs, inplace=inplace, apply_act=apply_act)
+
+    def forward(self, x):
+        x = super().forward(x)
+        x = self.drop(x)
+        x = self.act(x)
+        return x
+
+
+class RmsNormAct2d(RmsNorm2d):
+    """RMSNorm (over C of NCHW) + act."""
+    def __init__(
+            self, num_channels, eps=1e-6, affine=True,
+            apply_act=True, act_layer=nn.ReLU, act_kwargs=None, inplace=True, drop_layer=None):
+        super().__init__(num_channels, eps=eps, affine=affine)
+        self.drop = drop_layer() if drop_layer is not None else nn.Identity()
+        self.act = _create_act(act_layer, act_kwargs=act_kwargs, inplace=inplace, apply_act=apply_act)
+
+    def forward(self, x):
+        x = super().forward(x)
         x = self.drop(x)
         x = self.act(x)
         return x

@@ -55,6 +55,7 @@ from .maxxvit import *
 from .metaformer import *
 from .mlp_mixer import *
 from .mobilenetv3 import *
+from .mobilenetv5 import *
 from .mobilevit import *
 from .nfnet import *
 from .naflexvit import *
