@@ -73,10 +73,10 @@ def lecun_normal_(tensor):
     variance_scaling_(tensor, mode='fan_in', distribution='truncated_normal')
 
 
-def init_weight_jax(module: nn.Module, name: str = '', head_bias: float = 0.):
+def init_weight_jax(module: nn.Module, name: str = '', head_bias: float = 0., classifier_name: str = 'head'):
     """JAX-style init for ViT (reference `vision_transformer.py:1305-1393` init modes)."""
     if isinstance(module, nn.Linear):
-        if name.startswith('head'):
+        if name.startswith(classifier_name):
             nn.init.zeros_(module.weight)
             nn.init.constant_(module.bias, head_bias)
         else:
@@ -94,10 +94,20 @@ def init_weight_jax(module: nn.Module, name: str = '', head_bias: float = 0.):
         module.init_weights()
 
 
-def init_weight_vit(module: nn.Module, name: str = ''):
+def init_weight_vit(
+        module: nn.Module,
+        name: str = '',
+        init_bias: float = 0.,
+        head_bias: float = 0.,
+        classifier_name: str = 'head',
+):
     if isinstance(module, nn.Linear):
-        trunc_normal_(module.weight, std=.02)
-        if module.bias is not None:
-            nn.init.zeros_(module.bias)
+        if name.startswith(classifier_name) and module.bias is not None:
+            nn.init.zeros_(module.weight)
+            nn.init.constant_(module.bias, head_bias)
+        else:
+            trunc_normal_(module.weight, std=.02)
+            if module.bias is not None:
+                nn.init.constant_(module.bias, init_bias)
     elif hasattr(module, 'init_weights'):
         module.init_weights()

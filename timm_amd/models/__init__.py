@@ -103,6 +103,7 @@ from .efficientformer_v2 import *
 from .vitamin import *
 from .volo import *
 from .mvitv2 import *
+from .hieradet_sam2 import *
 from .hgnet import *
 from .nextvit import *
 from .coat import *
