@@ -46,6 +46,7 @@ from .efficientnet import *
 from .eva import *
 from .focalnet import *
 from .gcvit import *
+from .gemma4_vit import *
 from .ghostnet import *
 from .hardcorenas import *
 from .hiera import *

@@ -71,16 +71,24 @@ def batch_patchify(
         x: torch.Tensor,
         patch_size: Tuple[int, int],
         pad: bool = True,
+        channels_last: bool = True,
 ) -> Tuple[torch.Tensor, Tuple[int, int]]:
-    """[B,C,H,W] -> ([B, N, P*P*C], (nH, nW)) patchification (reference `:163`)."""
+    """[B,C,H,W] -> ([B, N, P*P*C], (nH, nW)) patchification (reference `:163`).
+
+    channels_last=True gives P-P-C flat patches (NaFlex default); False gives
+    C-P-P flat (HF / Gemma4 native layout).
+    """
     B, C, H, W = x.shape
     ph, pw = patch_size
     if pad and (H % ph or W % pw):
         x = F.pad(x, (0, (pw - W % pw) % pw, 0, (ph - H % ph) % ph))
         B, C, H, W = x.shape
     nh, nw = H // ph, W // pw
-    # [B, C, nh, ph, nw, pw] -> [B, nh*nw, ph*pw*C]  (P-P-C patch layout)
-    patches = x.view(B, C, nh, ph, nw, pw).permute(0, 2, 4, 3, 5, 1).reshape(B, nh * nw, ph * pw * C)
+    if channels_last:
+        patches = x.view(B, C, nh, ph, nw, pw).permute(0, 2, 4, 3, 5, 1)  # (B, nh, nw, ph, pw, C)
+    else:
+        patches = x.view(B, C, nh, ph, nw, pw).permute(0, 2, 4, 1, 3, 5)  # (B, nh, nw, C, ph, pw)
+    patches = patches.reshape(B, nh * nw, ph * pw * C)
     return patches, (nh, nw)
 
 
