@@ -82,6 +82,7 @@ from .vision_transformer import *
 from .vovnet import *
 from .xception import *
 from .xception_aligned import *
+from .cpubone import *
 from .convit import *
 from .senet import *
 from .visformer import *
