@@ -90,6 +90,7 @@ from .fastvit import *
 from .fasternet import *
 from .shvit import *
 from .dla import *
+from .csatv2 import *
 from .cspnet import *
 from .repvit import *
 from .swiftformer import *
