@@ -25,6 +25,8 @@ EXCLUDE_FILTERS = [
     'convformer_b36', 'caformer_b36', 'poolformer_m48', 'poolformerv2_m48', 'densenet264d',
     'regnetx_320', 'regnety_320', 'ese_vovnet99b', 'tresnet_xl', 'vgg19*', 'twins_svt_large',
     'twins_pcpvt_large', 'xcit_medium*',
+    # encoder-only towers (non-classifier output) and >250M-param towers
+    '*_enc', 'mobilenetv5_300m*', 'gemma4_vit_570m*', 'fastvit_mci4',
 ]
 
 
