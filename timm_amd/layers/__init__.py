@@ -60,6 +60,8 @@ from .evo_norm import (
     EvoNorm2dS2, EvoNorm2dS2a,
 )
 from .filter_response_norm import FilterResponseNormAct2d, FilterResponseNormTlu2d
+from .cbam import CbamModule, LightCbamModule, ChannelAttn, LightChannelAttn, SpatialAttn, LightSpatialAttn
+from .coord_attn import CoordAttn, EfficientLocalAttn, SimpleCoordAttn, StripAttn
 from .conv_self_attn import BottleneckAttn, HaloAttn, LambdaLayer, RelPos2d
 from .grid import ndgrid, meshgrid
 from .non_local_attn import NonLocalAttn, BatNonLocalAttn

@@ -3,7 +3,9 @@ import functools
 
 import torch
 
+from .cbam import CbamModule, LightCbamModule
 from .conv_self_attn import BottleneckAttn, HaloAttn, LambdaLayer
+from .coord_attn import CoordAttn, EfficientLocalAttn, SimpleCoordAttn, StripAttn
 from .eca import EcaModule, CecaModule
 from .gather_excite import GatherExcite
 from .global_context import GlobalContext
@@ -37,6 +39,18 @@ def get_attn(attn_type):
                 module_cls = GlobalContext
             elif attn_type == 'gca':
                 module_cls = functools.partial(GlobalContext, fuse_add=True, fuse_scale=False)
+            elif attn_type == 'cbam':
+                module_cls = CbamModule
+            elif attn_type == 'lcbam':
+                module_cls = LightCbamModule
+            elif attn_type == 'coord':
+                module_cls = CoordAttn
+            elif attn_type == 'scoord':
+                module_cls = SimpleCoordAttn
+            elif attn_type == 'ela':
+                module_cls = EfficientLocalAttn
+            elif attn_type == 'strip':
+                module_cls = StripAttn
             # Attention / attention-like modules w/ significant compute
             elif attn_type == 'sk':
                 module_cls = SelectiveKernel
