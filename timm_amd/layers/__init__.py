@@ -5,6 +5,7 @@ from .adaptive_avgmax_pool import (
 )
 from .attention import Attention, AttentionRope, maybe_add_mask
 from .attention2d import Attention2d, MultiQueryAttention2d, MultiQueryAttentionV2
+from .attention_pool2d import AttentionPool2d, RotAttentionPool2d
 from .attention_pool import AttentionPoolLatent
 from .blur_pool import BlurPool2d, create_aa
 from .classifier import ClassifierHead, ClNormMlpClassifierHead, NormMlpClassifierHead, create_classifier

@@ -432,12 +432,18 @@ class RotaryEmbeddingCat(nn.Module):
 
 def create_rope_embed(
         rope_type: str = 'cat',
-        dim: int = 64,
+        dim: int = 768,
+        num_heads: int = 12,
         **kwargs,
 ):
-    """RoPE factory (reference `pos_embed_sincos.py:1315`)."""
+    """RoPE factory (reference `pos_embed_sincos.py:1315`).
+
+    ``dim`` is the TOTAL embedding dim; per-head dim is derived via num_heads.
+    """
     if rope_type in ('', 'cat', 'regular'):
-        return RotaryEmbeddingCat(dim, **kwargs)
+        kwargs.pop('rotate_half', None)  # not supported by cat variant
+        return RotaryEmbeddingCat(dim // num_heads, **kwargs)
     if rope_type == 'base':
-        return RotaryEmbedding(dim, **kwargs)
+        kwargs.pop('rotate_half', None)
+        return RotaryEmbedding(dim // num_heads, **kwargs)
     raise ValueError(f'Unknown rope type {rope_type}')
