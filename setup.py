@@ -18,6 +18,7 @@ sources = [
     os.path.join(CSRC, 'layernorm.hip'),
     os.path.join(CSRC, 'elementwise.hip'),
     os.path.join(CSRC, 'attention.hip'),
+    os.path.join(CSRC, 'attention_bwd.hip'),
     os.path.join(CSRC, 'depthwise_conv.hip'),
     os.path.join(CSRC, 'multi_tensor.hip'),
 ]

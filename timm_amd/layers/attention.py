@@ -64,7 +64,21 @@ class Attention(nn.Module):
             attn_mask: Optional[torch.Tensor] = None,
     ) -> torch.Tensor:
         B, N, C = x.shape
-        qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim).permute(2, 0, 3, 1, 4)
+        qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim)
+        if self.fused_attn and isinstance(self.q_norm, nn.Identity) and isinstance(self.k_norm, nn.Identity):
+            # packed path: no unbind in the autograd graph — backward writes
+            # dq/dk/dv into one packed dqkv buffer (ops/attention.py)
+            x = ops.flash_attention_qkv(
+                qkv,
+                attn_mask=attn_mask,
+                dropout_p=self.attn_drop.p if self.training else 0.,
+            )
+            x = x.transpose(1, 2).reshape(B, N, self.attn_dim)
+            x = self.norm(x)
+            x = self.proj(x)
+            x = self.proj_drop(x)
+            return x
+        qkv = qkv.permute(2, 0, 3, 1, 4)
         q, k, v = qkv.unbind(0)
         q, k = self.q_norm(q), self.k_norm(k)
 

@@ -441,9 +441,13 @@ def create_rope_embed(
     ``dim`` is the TOTAL embedding dim; per-head dim is derived via num_heads.
     """
     if rope_type in ('', 'cat', 'regular'):
-        kwargs.pop('rotate_half', None)  # not supported by cat variant
+        if kwargs.pop('rotate_half', False):
+            raise NotImplementedError(
+                'rotate_half RoPE layout is not implemented for the cat variant; '
+                'interleaved rotation would give wrong outputs silently')
         return RotaryEmbeddingCat(dim // num_heads, **kwargs)
     if rope_type == 'base':
-        kwargs.pop('rotate_half', None)
+        if kwargs.pop('rotate_half', False):
+            raise NotImplementedError('rotate_half RoPE layout is not implemented')
         return RotaryEmbedding(dim // num_heads, **kwargs)
     raise ValueError(f'Unknown rope type {rope_type}')

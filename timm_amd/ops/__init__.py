@@ -69,7 +69,7 @@ def use_hip(x: torch.Tensor) -> bool:
 
 from .conv import depthwise_conv2d  # noqa: E402
 from .layer_norm import layer_norm_act, rms_norm_act, layer_norm, rms_norm  # noqa: E402
-from .attention import flash_attention, attention_available  # noqa: E402
+from .attention import flash_attention, flash_attention_qkv, attention_available  # noqa: E402
 from .elementwise import bias_act, residual_scale_add  # noqa: E402
 from .fused_optim import fused_adamw_step, fused_lerp_, fused_l2norm  # noqa: E402
 
@@ -77,7 +77,7 @@ __all__ = [
     'has_ext', 'require_ext', 'use_hip',
     'depthwise_conv2d',
     'layer_norm', 'layer_norm_act', 'rms_norm', 'rms_norm_act',
-    'flash_attention', 'attention_available',
+    'flash_attention', 'flash_attention_qkv', 'attention_available',
     'bias_act', 'residual_scale_add',
     'fused_adamw_step', 'fused_lerp_', 'fused_l2norm',
 ]

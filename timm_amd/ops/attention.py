@@ -1,26 +1,34 @@
 """Fused multi-head attention dispatch (SDPA replacement).
 
 The reference leans on `F.scaled_dot_product_attention`
-(`timm/layers/attention.py:124-129`, `timm/models/eva.py:246-251`).  Here the
-device path is a hand-written flash-style gfx950 HIP kernel:
+(`timm/layers/attention.py:124-129`, `timm/models/eva.py:239-251`).  Here both
+directions are hand-written flash-style gfx950 HIP kernels:
 
  * forward: MFMA 16x16x32 bf16 QK^T + online softmax + PV, returns O and
    logsumexp (LSE) per row.  Supports optional additive mask (NaFlex padding
    masks, rel-pos bias broadcast over batch).
- * backward: exact recompute using hipBLASLt GEMMs (torch.matmul) with the
-   saved LSE — the GEMM-shaped work of the backward runs on the MFMA library
-   path; a fully-fused bwd kernel is a later optimization.
+ * backward: fused two-pass flash backward (attn_bwd_dq / attn_bwd_dkdv
+   kernels) — softmax recomputed from LSE inside the tile loop, dQ/dK/dV
+   accumulated in fp32 MFMA registers, no [B,H,Nq,Nk] tensor ever touches
+   HBM.  (The round-1 hipBLASLt GEMM-recompute chain remains available via
+   TIMM_AMD_ATTN_BWD=gemm for A/B comparison.)
+ * flash_attention_qkv: packed-qkv entry — takes the [B,N,3,H,D] projection
+   output directly and writes gradients into one packed dqkv buffer, so the
+   qkv-unbind `aten::copy_`/stack grad chain never appears in the graph.
 
 CPU path = reference math composition (fp32 softmax), which doubles as the
 numerics oracle for the GPU tests.
 """
 import math
+import os
 from typing import Optional
 
 import torch
 import torch.nn.functional as F
 
 from . import _load_extension
+
+_BWD_MODE = os.environ.get('TIMM_AMD_ATTN_BWD', 'fused')
 
 
 def attention_available(q: torch.Tensor) -> bool:
@@ -41,19 +49,59 @@ def _math_sdpa(q, k, v, attn_mask=None, scale=None):
     return out.to(q.dtype)
 
 
+def _strides_ok(t):
+    return t.stride(-1) == 1 and all(s % 8 == 0 or s == 0 for s in t.stride()[:3])
+
+
+def _prep_mask(attn_mask, q, k):
+    # kernel accepts [B|1, H|1, Nq, Nk]: keep batch/head dims unexpanded
+    # (rel-pos bias is [1,H,N,N], padding masks [B,1,N,N]) so the contiguous
+    # copy stays small.
+    while attn_mask.dim() < 4:
+        attn_mask = attn_mask.unsqueeze(0)
+    attn_mask = attn_mask.expand(
+        attn_mask.shape[0], attn_mask.shape[1], q.shape[2], k.shape[2])
+    return attn_mask.contiguous().to(torch.float32)
+
+
+def _fused_backward(ext, q, k, v, o, lse, do, attn_mask, scale, dq, dk, dv):
+    """Shared fused-backward driver writing into preallocated dq/dk/dv."""
+    if do.dim() == 4 and do.stride(-1) == 1 and do.shape[-1] % 32 == 0 and do.shape[-1] <= 128:
+        do_c, delta = ext.attn_bwd_preprocess(do, o)
+    else:
+        do_c = do.contiguous()
+        delta = (do_c * o).float().sum(-1)
+    ext.attention_bwd(q, k, v, do_c, lse, delta, attn_mask, dq, dk, dv, scale)
+
+
+def _gemm_backward(ext, q, k, v, o, lse, do, attn_mask, scale):
+    """Round-1 exact recompute via hipBLASLt GEMMs (A/B reference path)."""
+    if do.dim() == 4 and do.stride(-1) == 1 and do.shape[-1] % 32 == 0 and do.shape[-1] <= 128:
+        do, delta = ext.attn_bwd_preprocess(do, o)
+    else:
+        do = do.contiguous()
+        delta = (do * o).float().sum(-1)
+    s = (q @ k.transpose(-2, -1)).contiguous()
+    p = ext.attn_bwd_softmax(s, lse, attn_mask, scale)
+    dv = p.transpose(-2, -1) @ do
+    dp = (do @ v.transpose(-2, -1)).contiguous()
+    ds = ext.attn_bwd_ds(p, dp, delta, scale)
+    dq = ds @ k
+    dk = ds.transpose(-2, -1) @ q
+    return dq, dk, dv
+
+
 class _FlashAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, attn_mask, scale):
         ext = _load_extension()
         # kernel reads strided q/k/v (e.g. views into the packed qkv tensor)
         # as long as head_dim is contiguous and 16B-alignment holds
-        def _ok(t):
-            return t.stride(-1) == 1 and all(s % 8 == 0 or s == 0 for s in t.stride()[:3])
-        if not _ok(q):
+        if not _strides_ok(q):
             q = q.contiguous()
-        if not _ok(k):
+        if not _strides_ok(k):
             k = k.contiguous()
-        if not _ok(v):
+        if not _strides_ok(v):
             v = v.contiguous()
         if attn_mask is not None:
             attn_mask = attn_mask.contiguous().to(torch.float32)
@@ -68,23 +116,48 @@ class _FlashAttnFn(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         ext = _load_extension()
         scale = ctx.scale
-        # exact flash backward: recompute P from saved LSE. GEMMs stay bf16
-        # (MFMA via hipBLASLt, fp32 internal accum); the softmax-recompute
-        # elementwise runs in fused HIP kernels (attn_bwd_preprocess/_softmax/_ds).
-        if do.dim() == 4 and do.stride(-1) == 1 and do.shape[-1] % 32 == 0 and do.shape[-1] <= 128:
-            # fused: contiguous dO + delta = rowsum(dO*O) in one pass
-            do, delta = ext.attn_bwd_preprocess(do, o)
+        if _BWD_MODE == 'gemm':
+            dq, dk, dv = _gemm_backward(ext, q, k, v, o, lse, do, ctx.attn_mask, scale)
         else:
-            do = do.contiguous()
-            delta = (do * o).float().sum(-1)  # [B,H,Nq] fp32
-        s = (q @ k.transpose(-2, -1)).contiguous()  # bf16 GEMM
-        p = ext.attn_bwd_softmax(s, lse, ctx.attn_mask, scale)  # exp(s*scale+mask-lse)
-        dv = p.transpose(-2, -1) @ do
-        dp = (do @ v.transpose(-2, -1)).contiguous()  # bf16 GEMM
-        ds = ext.attn_bwd_ds(p, dp, delta, scale)
-        dq = ds @ k
-        dk = ds.transpose(-2, -1) @ q
+            dq = torch.empty_like(q, memory_format=torch.contiguous_format)
+            dk = torch.empty_like(k, memory_format=torch.contiguous_format)
+            dv = torch.empty_like(v, memory_format=torch.contiguous_format)
+            _fused_backward(ext, q, k, v, o, lse, do, ctx.attn_mask, scale, dq, dk, dv)
         return dq, dk, dv, None, None
+
+
+class _FlashAttnQkvFn(torch.autograd.Function):
+    """Packed-qkv attention: input [B,N,3,H,D], grads written into one packed
+    dqkv buffer (kills the unbind/stack grad copies)."""
+
+    @staticmethod
+    def forward(ctx, qkv, attn_mask, scale):
+        ext = _load_extension()
+        B, N, three, H, D = qkv.shape
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)  # [B,H,N,D] strided view
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        if attn_mask is not None:
+            attn_mask = attn_mask.contiguous().to(torch.float32)
+        o, lse = ext.attention_fwd(q, k, v, attn_mask, scale)
+        ctx.save_for_backward(qkv, o, lse)
+        ctx.attn_mask = attn_mask
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        qkv, o, lse = ctx.saved_tensors
+        ext = _load_extension()
+        q = qkv[:, :, 0].permute(0, 2, 1, 3)
+        k = qkv[:, :, 1].permute(0, 2, 1, 3)
+        v = qkv[:, :, 2].permute(0, 2, 1, 3)
+        dqkv = torch.empty_like(qkv)
+        dq = dqkv[:, :, 0].permute(0, 2, 1, 3)
+        dk = dqkv[:, :, 1].permute(0, 2, 1, 3)
+        dv = dqkv[:, :, 2].permute(0, 2, 1, 3)
+        _fused_backward(ext, q, k, v, o, lse, do, ctx.attn_mask, ctx.scale, dq, dk, dv)
+        return dqkv, None, None
 
 
 def flash_attention(
@@ -103,13 +176,7 @@ def flash_attention(
         return _dropout_sdpa(q, k, v, attn_mask, dropout_p, scale)
     if attention_available(q):
         if attn_mask is not None:
-            # kernel accepts [B|1, H|1, Nq, Nk]: keep batch/head dims unexpanded
-            # (rel-pos bias is [1,H,N,N], padding masks [B,1,N,N]) so the
-            # contiguous copy below stays small.
-            while attn_mask.dim() < 4:
-                attn_mask = attn_mask.unsqueeze(0)
-            attn_mask = attn_mask.expand(
-                attn_mask.shape[0], attn_mask.shape[1], q.shape[2], k.shape[2])
+            attn_mask = _prep_mask(attn_mask, q, k)
         return _FlashAttnFn.apply(q, k, v, attn_mask, scale)
     if q.is_cuda:
         from . import use_hip
@@ -117,6 +184,33 @@ def flash_attention(
             # ext present but shape/dtype unsupported (e.g. fp32, D>128): exact composition
             return _math_sdpa(q, k, v, attn_mask, scale)
     return _math_sdpa(q, k, v, attn_mask, scale)
+
+
+def flash_attention_qkv(
+        qkv: torch.Tensor,
+        attn_mask: Optional[torch.Tensor] = None,
+        dropout_p: float = 0.,
+        scale: Optional[float] = None,
+) -> torch.Tensor:
+    """Packed attention on the qkv projection output.
+
+    qkv: [B, N, 3, H, D] (the reshaped nn.Linear output — no permute/unbind
+    copies).  Returns O as a [B,H,N,D] view of BNHD storage, so the usual
+    `transpose(1,2).reshape(B,N,C)` is free.
+    """
+    B, N, three, H, D = qkv.shape
+    assert three == 3
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    q = qkv[:, :, 0].permute(0, 2, 1, 3)
+    if dropout_p == 0. and attention_available(q) and qkv.stride(-1) == 1:
+        if attn_mask is not None:
+            k = qkv[:, :, 1].permute(0, 2, 1, 3)
+            attn_mask = _prep_mask(attn_mask, q, k)
+        return _FlashAttnQkvFn.apply(qkv, attn_mask, scale)
+    k = qkv[:, :, 1].permute(0, 2, 1, 3)
+    v = qkv[:, :, 2].permute(0, 2, 1, 3)
+    return flash_attention(q, k, v, attn_mask=attn_mask, dropout_p=dropout_p, scale=scale)
 
 
 def _dropout_sdpa(q, k, v, attn_mask, dropout_p, scale):

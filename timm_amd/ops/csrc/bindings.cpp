@@ -27,6 +27,13 @@ at::Tensor attn_bwd_softmax(at::Tensor s, at::Tensor lse, c10::optional<at::Tens
 at::Tensor attn_bwd_ds(at::Tensor p, at::Tensor dp, at::Tensor delta, double scale);
 std::vector<at::Tensor> attn_bwd_preprocess(at::Tensor dout, at::Tensor o);
 
+// attention_bwd.hip
+void attention_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                   at::Tensor dov, at::Tensor lse, at::Tensor delta,
+                   c10::optional<at::Tensor> mask,
+                   at::Tensor dq, at::Tensor dk, at::Tensor dv,
+                   double scale);
+
 // depthwise_conv.hip
 at::Tensor dwconv_fwd(at::Tensor x, at::Tensor w_t, c10::optional<at::Tensor> bias,
                       long stride, long pad, long K, long Ho, long Wo);
@@ -56,6 +63,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_bwd_softmax", &attn_bwd_softmax, "fused softmax recompute for attention bwd");
   m.def("attn_bwd_ds", &attn_bwd_ds, "fused dS epilogue for attention bwd");
   m.def("attn_bwd_preprocess", &attn_bwd_preprocess, "fused dO copy + delta for attention bwd");
+  m.def("attention_bwd", &attention_bwd, "fused flash attention bwd (MFMA, gfx950)");
   m.def("dwconv_fwd", &dwconv_fwd, "NHWC depthwise conv fwd (gfx950)");
   m.def("dwconv_bwd_data", &dwconv_bwd_data, "NHWC depthwise conv bwd-data");
   m.def("dwconv_bwd_weight", &dwconv_bwd_weight, "NHWC depthwise conv bwd-weight");

@@ -140,8 +140,19 @@ class BucketedDataParallel(nn.Module):
                 bucket.work = None
                 bucket.flat.mul_(inv)
             elif bucket.ready_count:
-                # partially-ready bucket (shouldn't happen in a full backward)
+                # A partially-ready bucket means some params received no grad
+                # this backward (unused branch / aux head).  Silently skipping
+                # the reduce would let ranks apply different gradients, and
+                # force-reducing here can reorder collectives across ranks
+                # (NCCL deadlock hazard), so fail loudly like torch DDP does
+                # without find_unused_parameters.
                 bucket.ready_count = 0
+                raise RuntimeError(
+                    'BucketedDataParallel: a gradient bucket is only partially ready at '
+                    'sync time — some parameters received no gradient in this backward '
+                    '(unused model branch?). All parameters must contribute a grad every '
+                    'step, or the unused branch must be detached from the reducer.'
+                )
 
     def zero_grad_buckets(self, set_to_none: bool = False):
         """Zero the flat gradient buffers (p.grad views stay attached)."""

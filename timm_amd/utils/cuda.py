@@ -26,9 +26,14 @@ class NativeScaler:
             parameters=None,
             create_graph=False,
             need_update=True,
+            pre_step_fn=None,
     ):
         self._scaler.scale(loss).backward(create_graph=create_graph)
         if need_update:
+            if pre_step_fn is not None:
+                # e.g. task.finish_gradient_sync: wait on in-flight bucket
+                # all-reduces + apply 1/world averaging BEFORE unscale/step
+                pre_step_fn()
             if clip_grad is not None:
                 assert parameters is not None
                 self._scaler.unscale_(optimizer)  # unscale the gradients of optimizer's assigned params in-place

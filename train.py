@@ -718,6 +718,10 @@ def train_one_epoch(
 
     data_start_time = update_start_time = time.time()
     optimizer.zero_grad()
+    if task._distributed_model is not None:
+        # zero_grad(set_to_none=True) detaches p.grad from the flat bucket
+        # views; re-attach so the first backward accumulates into the buckets
+        task._distributed_model.zero_grad_buckets()
     update_sample_count = 0
     for batch_idx, (input, target) in enumerate(loader):
         if args.max_steps_per_epoch and batch_idx >= args.max_steps_per_epoch:
@@ -756,6 +760,7 @@ def train_one_epoch(
                     parameters=task.model.parameters(),
                     create_graph=second_order,
                     need_update=need_update,
+                    pre_step_fn=task.finish_gradient_sync if need_update else None,
                 )
             else:
                 _loss.backward(create_graph=second_order)
