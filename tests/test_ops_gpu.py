@@ -178,6 +178,69 @@ def test_attention_fwd_bwd(shape):
     assert rel_err(v1.grad, v2.grad) < 4e-2
 
 
+@pytest.mark.parametrize('shape', [
+    (2, 12, 197, 64),    # ViT-B/224
+    (1, 8, 1024, 64),    # NaFlex max bucket
+    (1, 4, 333, 128),    # ragged + D=128
+])
+def test_attention_qkv_packed(shape):
+    """Packed [B,N,3,H,D] path: fwd + fused bwd writing one dqkv buffer."""
+    _ext()
+    torch.manual_seed(11)
+    B, H, N, D = shape
+    qkv = torch.randn(B, N, 3, H, D, device='cuda', dtype=torch.bfloat16)
+    qkv1 = qkv.clone().requires_grad_(True)
+    o = ops.flash_attention_qkv(qkv1)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    qkv2 = qkv.detach().float().requires_grad_(True)
+    q2, k2, v2 = qkv2.permute(2, 0, 3, 1, 4).unbind(0)
+    attn = (q2 @ k2.transpose(-2, -1)) / math.sqrt(D)
+    o_ref = attn.softmax(-1) @ v2
+    o_ref.backward(do.float())
+
+    assert rel_err(o, o_ref) < 2e-2
+    assert rel_err(qkv1.grad, qkv2.grad) < 4e-2
+
+
+def test_attention_padding_mask_bwd():
+    """NaFlex-style -inf key-padding mask through the fused backward: masked
+    keys must receive zero dk/dv and no NaN may leak from exp(-inf) paths."""
+    _ext()
+    torch.manual_seed(12)
+    B, H, N, D = 2, 4, 300, 64
+    q = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    k = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    v = torch.randn(B, H, N, D, device='cuda', dtype=torch.bfloat16)
+    valid = torch.ones(B, N, device='cuda', dtype=torch.bool)
+    valid[1, -77:] = False
+    mask = torch.zeros(B, 1, N, N, device='cuda')
+    mask.masked_fill_(~valid.view(B, 1, 1, N), float('-inf'))
+
+    q1, k1, v1 = [t.clone().requires_grad_(True) for t in (q, k, v)]
+    o = ops.flash_attention(q1, k1, v1, attn_mask=mask)
+    do = torch.randn_like(o)
+    o.backward(do)
+    for g in (q1.grad, k1.grad, v1.grad):
+        assert torch.isfinite(g.float()).all()
+
+    q2, k2, v2 = [t.detach().float().requires_grad_(True) for t in (q, k, v)]
+    attn = (q2 @ k2.transpose(-2, -1)) / math.sqrt(D) + mask
+    o_ref = attn.softmax(-1) @ v2
+    o_ref.backward(do.float())
+
+    assert rel_err(o, o_ref) < 2e-2
+    assert rel_err(q1.grad, q2.grad) < 4e-2
+    assert rel_err(k1.grad, k2.grad) < 4e-2
+    assert rel_err(v1.grad, v2.grad) < 4e-2
+    # masked keys attract zero gradient
+    kv_pad = k1.grad[1, :, -77:].float()
+    vv_pad = v1.grad[1, :, -77:].float()
+    assert kv_pad.abs().max().item() == 0.0
+    assert vv_pad.abs().max().item() == 0.0
+
+
 def test_attention_mask():
     _ext()
     torch.manual_seed(5)
