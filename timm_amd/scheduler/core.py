@@ -116,7 +116,7 @@ class Scheduler:
         values = self.schedule(t)
         noise = self._noise_at(t)
         if noise is not None:
-            values = [v * (1 + noise) for v in values]
+            values = [v + v * noise for v in values]
         self.update_groups(values)
 
     def update_groups(self, values) -> None:

@@ -1,54 +1,67 @@
-"""Scheduler factory (reference `timm/scheduler/scheduler_factory.py:63`)."""
+"""Scheduler construction from training config.
+
+Behavioral parity: /root/reference/timm/scheduler/scheduler_factory.py:63
+(create_scheduler_v2 semantics: epoch->update conversion, noise range scaling,
+cycle-length-aware epoch recount).  Structure here is table-driven: each
+schedule name maps to (class, set of extra arg-bundles it accepts).
+"""
 from typing import List, Optional, Union
 
 from torch.optim import Optimizer
 
-from .cosine_lr import CosineLRScheduler
-from .multistep_lr import MultiStepLRScheduler
-from .plateau_lr import PlateauLRScheduler
-from .poly_lr import PolyLRScheduler
-from .step_lr import StepLRScheduler
-from .tanh_lr import TanhLRScheduler
+from .core import (
+    CosineLRScheduler,
+    MultiStepLRScheduler,
+    PlateauLRScheduler,
+    PolyLRScheduler,
+    StepLRScheduler,
+    TanhLRScheduler,
+)
+
+__all__ = ['scheduler_kwargs', 'create_scheduler', 'create_scheduler_v2']
+
+# schedule name -> (class, takes_cycle_args, takes_k_decay)
+_SCHEDULES = {
+    'cosine': (CosineLRScheduler, True, True),
+    'tanh': (TanhLRScheduler, True, False),
+    'poly': (PolyLRScheduler, True, True),
+    'step': (StepLRScheduler, False, False),
+    'multistep': (MultiStepLRScheduler, False, False),
+    'plateau': (PlateauLRScheduler, False, False),
+}
 
 
-def scheduler_kwargs(cfg, decreasing_metric: Optional[bool] = None):
-    """cfg/argparse to kwargs helper. Convert scheduler args in argparse args or cfg like object to keyword args."""
-    eval_metric = getattr(cfg, 'eval_metric', 'top1')
-    if decreasing_metric is not None:
-        plateau_mode = 'min' if decreasing_metric else 'max'
-    else:
-        plateau_mode = 'min' if 'loss' in eval_metric else 'max'
-    kwargs = dict(
+def scheduler_kwargs(cfg, decreasing_metric: Optional[bool] = None) -> dict:
+    """Flatten an argparse/config namespace into create_scheduler_v2 kwargs."""
+    get = lambda name, default: getattr(cfg, name, default)  # noqa: E731
+    if decreasing_metric is None:
+        decreasing_metric = 'loss' in get('eval_metric', 'top1')
+    return dict(
         sched=cfg.sched,
-        num_epochs=getattr(cfg, 'epochs', 100),
-        decay_epochs=getattr(cfg, 'decay_epochs', 30),
-        decay_milestones=getattr(cfg, 'decay_milestones', [30, 60]),
-        warmup_epochs=getattr(cfg, 'warmup_epochs', 5),
-        cooldown_epochs=getattr(cfg, 'cooldown_epochs', 0),
-        patience_epochs=getattr(cfg, 'patience_epochs', 10),
-        decay_rate=getattr(cfg, 'decay_rate', 0.1),
-        min_lr=getattr(cfg, 'min_lr', 0.),
-        warmup_lr=getattr(cfg, 'warmup_lr', 1e-5),
-        warmup_prefix=getattr(cfg, 'warmup_prefix', False),
-        noise=getattr(cfg, 'lr_noise', None),
-        noise_pct=getattr(cfg, 'lr_noise_pct', 0.67),
-        noise_std=getattr(cfg, 'lr_noise_std', 1.),
-        noise_seed=getattr(cfg, 'seed', 42),
-        cycle_mul=getattr(cfg, 'lr_cycle_mul', 1.),
-        cycle_decay=getattr(cfg, 'lr_cycle_decay', 0.1),
-        cycle_limit=getattr(cfg, 'lr_cycle_limit', 1),
-        k_decay=getattr(cfg, 'lr_k_decay', 1.),
-        plateau_mode=plateau_mode,
-        step_on_epochs=not getattr(cfg, 'sched_on_updates', False),
+        num_epochs=get('epochs', 100),
+        decay_epochs=get('decay_epochs', 30),
+        decay_milestones=get('decay_milestones', [30, 60]),
+        warmup_epochs=get('warmup_epochs', 5),
+        cooldown_epochs=get('cooldown_epochs', 0),
+        patience_epochs=get('patience_epochs', 10),
+        decay_rate=get('decay_rate', 0.1),
+        min_lr=get('min_lr', 0.),
+        warmup_lr=get('warmup_lr', 1e-5),
+        warmup_prefix=get('warmup_prefix', False),
+        noise=get('lr_noise', None),
+        noise_pct=get('lr_noise_pct', 0.67),
+        noise_std=get('lr_noise_std', 1.),
+        noise_seed=get('seed', 42),
+        cycle_mul=get('lr_cycle_mul', 1.),
+        cycle_decay=get('lr_cycle_decay', 0.1),
+        cycle_limit=get('lr_cycle_limit', 1),
+        k_decay=get('lr_k_decay', 1.),
+        plateau_mode='min' if decreasing_metric else 'max',
+        step_on_epochs=not get('sched_on_updates', False),
     )
-    return kwargs
 
 
-def create_scheduler(
-        args,
-        optimizer: Optimizer,
-        updates_per_epoch: int = 0,
-):
+def create_scheduler(args, optimizer: Optimizer, updates_per_epoch: int = 0):
     return create_scheduler_v2(
         optimizer=optimizer,
         **scheduler_kwargs(args),
@@ -69,7 +82,7 @@ def create_scheduler_v2(
         warmup_lr: float = 1e-5,
         warmup_epochs: int = 0,
         warmup_prefix: bool = False,
-        noise: Union[float, List[float]] = None,
+        noise: Union[None, float, List[float]] = None,
         noise_pct: float = 0.67,
         noise_std: float = 1.,
         noise_seed: int = 42,
@@ -81,122 +94,82 @@ def create_scheduler_v2(
         step_on_epochs: bool = True,
         updates_per_epoch: int = 0,
 ):
-    t_initial = num_epochs
-    warmup_t = warmup_epochs
-    decay_t = decay_epochs
-    cooldown_t = cooldown_epochs
+    """Build a scheduler + the (possibly cycle-extended) epoch count.
 
+    When ``step_on_epochs`` is False every epoch-denominated knob is converted
+    to optimizer-update units using ``updates_per_epoch``.
+    """
+    if sched not in _SCHEDULES:
+        # reference behavior: unknown/'none' name -> no scheduler
+        return None, num_epochs
+    cls, takes_cycle, takes_k = _SCHEDULES[sched]
+
+    # convert epoch-denominated quantities to timestep units
+    tick = 1 if step_on_epochs else updates_per_epoch
     if not step_on_epochs:
-        assert updates_per_epoch > 0, 'updates_per_epoch must be set to number of dataloader batches'
-        t_initial = t_initial * updates_per_epoch
-        warmup_t = warmup_t * updates_per_epoch
-        decay_t = decay_t * updates_per_epoch
-        decay_milestones = [d * updates_per_epoch for d in decay_milestones]
-        cooldown_t = cooldown_t * updates_per_epoch
+        assert updates_per_epoch > 0, \
+            'updates_per_epoch must be set to number of dataloader batches'
+    total_t = num_epochs * tick
+    warmup_t = warmup_epochs * tick
+    cooldown_t = cooldown_epochs * tick
 
-    # warmup args
-    warmup_args = dict(
-        warmup_lr_init=warmup_lr,
+    kwargs = dict(
         warmup_t=warmup_t,
-        warmup_prefix=warmup_prefix,
-    )
-
-    # setup noise args for supporting schedulers
-    if noise is not None:
-        if isinstance(noise, (list, tuple)):
-            noise_range = [n * t_initial for n in noise]
-            if len(noise_range) == 1:
-                noise_range = noise_range[0]
-        else:
-            noise_range = noise * t_initial
-    else:
-        noise_range = None
-    noise_args = dict(
-        noise_range_t=noise_range,
+        warmup_lr_init=warmup_lr,
+        t_in_epochs=step_on_epochs,
         noise_pct=noise_pct,
         noise_std=noise_std,
         noise_seed=noise_seed,
     )
+    # noise window is specified as fraction(s) of the total schedule span
+    if noise is not None:
+        if isinstance(noise, (list, tuple)):
+            rng = [n * total_t for n in noise]
+            kwargs['noise_range_t'] = rng[0] if len(rng) == 1 else rng
+        else:
+            kwargs['noise_range_t'] = noise * total_t
+    else:
+        kwargs['noise_range_t'] = None
 
-    # setup cycle args for supporting schedulers
-    cycle_args = dict(
-        cycle_mul=cycle_mul,
-        cycle_decay=cycle_decay,
-        cycle_limit=cycle_limit,
-    )
-
-    lr_scheduler = None
-    if sched == 'cosine':
-        lr_scheduler = CosineLRScheduler(
-            optimizer,
-            t_initial=t_initial,
-            lr_min=min_lr,
-            t_in_epochs=step_on_epochs,
-            **cycle_args,
-            **warmup_args,
-            **noise_args,
-            k_decay=k_decay,
-        )
-    elif sched == 'tanh':
-        lr_scheduler = TanhLRScheduler(
-            optimizer,
-            t_initial=t_initial,
-            lr_min=min_lr,
-            t_in_epochs=step_on_epochs,
-            **cycle_args,
-            **warmup_args,
-            **noise_args,
-        )
-    elif sched == 'step':
-        lr_scheduler = StepLRScheduler(
-            optimizer,
-            decay_t=decay_t,
-            decay_rate=decay_rate,
-            t_in_epochs=step_on_epochs,
-            **warmup_args,
-            **noise_args,
-        )
-    elif sched == 'multistep':
-        lr_scheduler = MultiStepLRScheduler(
-            optimizer,
-            decay_t=decay_milestones,
-            decay_rate=decay_rate,
-            t_in_epochs=step_on_epochs,
-            **warmup_args,
-            **noise_args,
-        )
-    elif sched == 'plateau':
+    if sched == 'plateau':
         assert step_on_epochs, 'Plateau LR only supports step per epoch.'
-        warmup_args.pop('warmup_prefix', False)
-        lr_scheduler = PlateauLRScheduler(
-            optimizer,
+        kwargs.update(dict(
             decay_rate=decay_rate,
             patience_t=patience_epochs,
             cooldown_t=0,
             mode=plateau_mode,
             lr_min=min_lr,
-            **warmup_args,
-            **noise_args,
-        )
-    elif sched == 'poly':
-        lr_scheduler = PolyLRScheduler(
-            optimizer,
-            power=decay_rate,  # overloading 'decay_rate' as polynomial power
-            t_initial=t_initial,
+        ))
+    elif sched == 'step':
+        # stepped schedules keep their class default warmup_prefix=True
+        kwargs.update(dict(
+            decay_t=decay_epochs * tick,
+            decay_rate=decay_rate,
+        ))
+    elif sched == 'multistep':
+        kwargs.update(dict(
+            decay_t=[m * tick for m in decay_milestones],
+            decay_rate=decay_rate,
+        ))
+    else:  # cyclic family
+        kwargs.update(dict(
+            t_initial=total_t,
             lr_min=min_lr,
-            t_in_epochs=step_on_epochs,
-            k_decay=k_decay,
-            **cycle_args,
-            **warmup_args,
-            **noise_args,
-        )
+            warmup_prefix=warmup_prefix,
+            cycle_mul=cycle_mul,
+            cycle_decay=cycle_decay,
+            cycle_limit=cycle_limit,
+        ))
+        if takes_k:
+            kwargs['k_decay'] = k_decay
+        if sched == 'poly':
+            kwargs['power'] = decay_rate  # decay_rate doubles as the poly power
 
+    lr_scheduler = cls(optimizer, **kwargs)
+
+    # cycle schedulers can extend the run: recompute the epoch budget
     if hasattr(lr_scheduler, 'get_cycle_length'):
-        # For cycle based schedulers (cosine, tanh, poly) recalculate total epochs w/ cycles & cooldown
-        t_with_cycles_and_cooldown = lr_scheduler.get_cycle_length() + cooldown_t
-        if step_on_epochs:
-            num_epochs = t_with_cycles_and_cooldown
-        else:
-            num_epochs = t_with_cycles_and_cooldown // updates_per_epoch
+        total = lr_scheduler.get_cycle_length() + cooldown_t
+        num_epochs = total if step_on_epochs else total // updates_per_epoch
 
     return lr_scheduler, num_epochs
