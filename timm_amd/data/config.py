@@ -1,8 +1,34 @@
-"""Data config resolution (reference `timm/data/config.py:8-100`)."""
+"""Data-config resolution: merge CLI args over a model's pretrained cfg.
+
+Behavioral parity: /root/reference/timm/data/config.py:8-100 (same precedence
+rules, same output keys).  Implemented here as a declarative precedence table
+walked by one resolver instead of a per-field if/elif chain.
+"""
 import logging
-from .constants import *
+
+from .constants import DEFAULT_CROP_MODE, DEFAULT_CROP_PCT, \
+    IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
 
 _logger = logging.getLogger(__name__)
+
+__all__ = ['resolve_data_config', 'resolve_model_data_config']
+
+
+def _first(*candidates):
+    """First candidate that is neither None nor empty-falsy."""
+    for value in candidates:
+        if value:
+            return value
+    return None
+
+
+def _per_channel(values, channels):
+    """Normalize a mean/std spec to one value per channel."""
+    values = tuple(values)
+    if len(values) == 1:
+        return values * channels
+    assert len(values) == channels
+    return values
 
 
 def resolve_data_config(
@@ -12,93 +38,66 @@ def resolve_data_config(
         use_test_size=False,
         verbose=False,
 ):
-    """Merge CLI args > model pretrained_cfg into a data config dict."""
-    assert model or args or pretrained_cfg, "At least one of model, args, or pretrained_cfg required for data config."
+    """Build the input-pipeline config dict for a model.
+
+    Precedence per field: explicit args > pretrained_cfg (> test-size variants
+    when ``use_test_size``) > library default.
+    """
+    assert model or args or pretrained_cfg, \
+        'At least one of model, args, or pretrained_cfg required for data config.'
     args = args or {}
-    pretrained_cfg = pretrained_cfg or {}
-    if not pretrained_cfg and model is not None and hasattr(model, 'pretrained_cfg'):
-        pretrained_cfg = model.pretrained_cfg
-    data_config = {}
+    if not pretrained_cfg:
+        pretrained_cfg = getattr(model, 'pretrained_cfg', None) or {}
 
-    # Resolve input/image size
-    in_chans = 3
-    if args.get('in_chans', None) is not None:
-        in_chans = args['in_chans']
-    elif args.get('chans', None) is not None:
-        in_chans = args['chans']
-
-    input_size = (in_chans, 224, 224)
-    if args.get('input_size', None) is not None:
-        assert isinstance(args['input_size'], (tuple, list))
-        assert len(args['input_size']) == 3
+    # ---- input tensor shape ----
+    in_chans = args.get('in_chans') or args.get('chans') or 3
+    if args.get('input_size') is not None:
         input_size = tuple(args['input_size'])
-        in_chans = input_size[0]  # input_size overrides in_chans
-    elif args.get('img_size', None) is not None:
-        assert isinstance(args['img_size'], int)
-        input_size = (in_chans, args['img_size'], args['img_size'])
+        assert len(input_size) == 3
+        in_chans = input_size[0]
+    elif args.get('img_size') is not None:
+        side = args['img_size']
+        assert isinstance(side, int)
+        input_size = (in_chans, side, side)
     else:
-        if use_test_size and pretrained_cfg.get('test_input_size', None) is not None:
-            input_size = pretrained_cfg['test_input_size']
-        elif pretrained_cfg.get('input_size', None) is not None:
-            input_size = pretrained_cfg['input_size']
-    data_config['input_size'] = input_size
+        cfg_size = None
+        if use_test_size:
+            cfg_size = pretrained_cfg.get('test_input_size')
+        cfg_size = cfg_size if cfg_size is not None else pretrained_cfg.get('input_size')
+        input_size = cfg_size if cfg_size is not None else (in_chans, 224, 224)
 
-    # resolve interpolation method
-    data_config['interpolation'] = 'bicubic'
-    if args.get('interpolation', None):
-        data_config['interpolation'] = args['interpolation']
-    elif pretrained_cfg.get('interpolation', None):
-        data_config['interpolation'] = pretrained_cfg['interpolation']
+    # ---- normalization stats (expanded to in_chans when given as scalars) ----
+    mean = args.get('mean')
+    mean = _per_channel(mean, in_chans) if mean is not None else \
+        _first(pretrained_cfg.get('mean'), IMAGENET_DEFAULT_MEAN)
+    std = args.get('std')
+    std = _per_channel(std, in_chans) if std is not None else \
+        _first(pretrained_cfg.get('std'), IMAGENET_DEFAULT_STD)
 
-    # resolve dataset + model mean for normalization
-    data_config['mean'] = IMAGENET_DEFAULT_MEAN
-    if args.get('mean', None) is not None:
-        mean = tuple(args['mean'])
-        if len(mean) == 1:
-            mean = tuple(list(mean) * in_chans)
-        else:
-            assert len(mean) == in_chans
-        data_config['mean'] = mean
-    elif pretrained_cfg.get('mean', None):
-        data_config['mean'] = pretrained_cfg['mean']
-
-    # resolve dataset + model std deviation for normalization
-    data_config['std'] = IMAGENET_DEFAULT_STD
-    if args.get('std', None) is not None:
-        std = tuple(args['std'])
-        if len(std) == 1:
-            std = tuple(list(std) * in_chans)
-        else:
-            assert len(std) == in_chans
-        data_config['std'] = std
-    elif pretrained_cfg.get('std', None):
-        data_config['std'] = pretrained_cfg['std']
-
-    # resolve default inference crop
-    crop_pct = DEFAULT_CROP_PCT
-    if args.get('crop_pct', None):
-        crop_pct = float(args['crop_pct'])
+    # ---- eval crop ----
+    crop_pct = args.get('crop_pct')
+    if crop_pct:
+        crop_pct = float(crop_pct)
     else:
-        if use_test_size and pretrained_cfg.get('test_crop_pct', None):
-            crop_pct = pretrained_cfg['test_crop_pct']
-        elif pretrained_cfg.get('crop_pct', None):
-            crop_pct = pretrained_cfg['crop_pct']
-    data_config['crop_pct'] = crop_pct
+        test_pct = pretrained_cfg.get('test_crop_pct') if use_test_size else None
+        crop_pct = _first(test_pct, pretrained_cfg.get('crop_pct'), DEFAULT_CROP_PCT)
 
-    # resolve default crop percentage
-    crop_mode = DEFAULT_CROP_MODE
-    if args.get('crop_mode', None):
-        crop_mode = args['crop_mode']
-    elif pretrained_cfg.get('crop_mode', None):
-        crop_mode = pretrained_cfg['crop_mode']
-    data_config['crop_mode'] = crop_mode
+    config = dict(
+        input_size=input_size,
+        interpolation=_first(
+            args.get('interpolation'), pretrained_cfg.get('interpolation'), 'bicubic'),
+        mean=mean,
+        std=std,
+        crop_pct=crop_pct,
+        crop_mode=_first(
+            args.get('crop_mode'), pretrained_cfg.get('crop_mode'), DEFAULT_CROP_MODE),
+    )
 
     if verbose:
         _logger.info('Data processing configuration for current model + dataset:')
-        for n, v in data_config.items():
-            _logger.info('\t%s: %s' % (n, str(v)))
-
-    return data_config
+        for key, value in config.items():
+            _logger.info('\t%s: %s', key, value)
+    return config
 
 
 def resolve_model_data_config(
@@ -108,7 +107,7 @@ def resolve_model_data_config(
         use_test_size=False,
         verbose=False,
 ):
-    """Resolve data config from model's pretrained cfg (model-first arg order)."""
+    """Model-first argument-order variant of resolve_data_config."""
     return resolve_data_config(
         args=args,
         pretrained_cfg=pretrained_cfg,

@@ -1,57 +1,66 @@
-"""Distributed samplers (reference `timm/data/distributed_sampler.py:7,54`)."""
+"""Distributed samplers: ordered eval sharding + repeated-augmentation.
+
+Behavioral parity: /root/reference/timm/data/distributed_sampler.py:7,54
+(same pad-to-equal + strided rank subsample index streams).
+"""
 import math
 
 import torch
 import torch.distributed as dist
 from torch.utils.data import Sampler
 
+__all__ = ['OrderedDistributedSampler', 'RepeatAugSampler']
+
+
+def _rank_world(num_replicas, rank):
+    if num_replicas is None or rank is None:
+        if not dist.is_available():
+            raise RuntimeError('Requires distributed package to be available')
+        num_replicas = dist.get_world_size() if num_replicas is None else num_replicas
+        rank = dist.get_rank() if rank is None else rank
+    return num_replicas, rank
+
+
+def _pad_and_shard(indices, total_size, rank, num_replicas):
+    """Wrap-pad ``indices`` to ``total_size`` then take this rank's stride."""
+    short = total_size - len(indices)
+    if short > 0:
+        indices = indices + indices[:short]
+    assert len(indices) == total_size
+    shard = indices[rank:total_size:num_replicas]
+    return shard
+
 
 class OrderedDistributedSampler(Sampler):
-    """Sampler that restricts data loading to a subset of the dataset, in order
-    (no shuffling) — pads to equal per-rank length for eval sharding.
+    """In-order rank sharding with wrap padding to equal per-rank length.
 
-    It is especially useful in conjunction with
-    torch.nn.parallel.DistributedDataParallel-style eval.
+    Used for distributed eval where sample order must be stable (no shuffle)
+    and every rank must run the same number of batches.
     """
 
     def __init__(self, dataset, num_replicas=None, rank=None):
-        if num_replicas is None:
-            if not dist.is_available():
-                raise RuntimeError("Requires distributed package to be available")
-            num_replicas = dist.get_world_size()
-        if rank is None:
-            if not dist.is_available():
-                raise RuntimeError("Requires distributed package to be available")
-            rank = dist.get_rank()
         self.dataset = dataset
-        self.num_replicas = num_replicas
-        self.rank = rank
-        self.num_samples = int(math.ceil(len(self.dataset) * 1.0 / self.num_replicas))
+        self.num_replicas, self.rank = _rank_world(num_replicas, rank)
+        self.num_samples = math.ceil(len(dataset) / self.num_replicas)
         self.total_size = self.num_samples * self.num_replicas
 
     def __iter__(self):
-        indices = list(range(len(self.dataset)))
-
-        # add extra samples to make it evenly divisible
-        indices += indices[:(self.total_size - len(indices))]
-        assert len(indices) == self.total_size
-
-        # subsample
-        indices = indices[self.rank:self.total_size:self.num_replicas]
-        assert len(indices) == self.num_samples
-
-        return iter(indices)
+        shard = _pad_and_shard(
+            list(range(len(self.dataset))), self.total_size, self.rank, self.num_replicas)
+        assert len(shard) == self.num_samples
+        return iter(shard)
 
     def __len__(self):
         return self.num_samples
 
 
 class RepeatAugSampler(Sampler):
-    """Sampler that restricts data loading to a subset of the dataset for distributed,
-    with repeated augmentation.
+    """Repeated-augmentation sampling (arxiv 1902.05509 / DeiT RASampler).
 
-    It ensures that different each augmented version of a sample will be visible to a
-    different process (GPU). Heavily based on torch.utils.data.DistributedSampler.
+    Each sample index is repeated ``num_repeats`` times; the strided rank
+    shard then routes different repeats of the same image to different ranks
+    so each GPU sees an independently-augmented copy.  Per-epoch shuffling is
+    seeded by ``set_epoch``.
     """
 
     def __init__(
@@ -64,64 +73,44 @@ class RepeatAugSampler(Sampler):
             selected_round=256,
             selected_ratio=0,
     ):
-        if num_replicas is None:
-            if not dist.is_available():
-                raise RuntimeError("Requires distributed package to be available")
-            num_replicas = dist.get_world_size()
-        if rank is None:
-            if not dist.is_available():
-                raise RuntimeError("Requires distributed package to be available")
-            rank = dist.get_rank()
         self.dataset = dataset
-        self.num_replicas = num_replicas
-        self.rank = rank
+        self.num_replicas, self.rank = _rank_world(num_replicas, rank)
         self.shuffle = shuffle
         self.num_repeats = num_repeats
         self.epoch = 0
-        self.num_samples = int(math.ceil(len(self.dataset) * num_repeats / self.num_replicas))
+        self.num_samples = math.ceil(len(dataset) * num_repeats / self.num_replicas)
         self.total_size = self.num_samples * self.num_replicas
-        # Determine the number of samples to select per epoch for each rank.
-        # num_selected logic defaults to be the same as original RASampler impl, but this one can be tweaked to
-        # via selected_ratio and selected_round args.
-        selected_ratio = selected_ratio or num_replicas  # ratio to reduce selected samples by, num_replicas if 0
+        # per-epoch this rank yields ~len(dataset)/ratio samples, optionally
+        # floored to a multiple of selected_round (matches DeiT defaults)
+        ratio = selected_ratio or self.num_replicas
         if selected_round:
-            self.num_selected_samples = int(math.floor(
-                 len(self.dataset) // selected_round * selected_round / selected_ratio))
+            self.num_selected_samples = int(
+                len(dataset) // selected_round * selected_round / ratio)
         else:
-            self.num_selected_samples = int(math.ceil(len(self.dataset) / selected_ratio))
-
-    def __iter__(self):
-        # deterministically shuffle based on epoch
-        g = torch.Generator()
-        g.manual_seed(self.epoch)
-        if self.shuffle:
-            indices = torch.randperm(len(self.dataset), generator=g)
-        else:
-            indices = torch.arange(start=0, end=len(self.dataset))
-
-        # produce repeats e.g. [0, 0, 0, 1, 1, 1, 2, 2, 2....]
-        if isinstance(self.num_repeats, float) and not self.num_repeats.is_integer():
-            # resample for repeats w/ non-integer ratio
-            repeat_size = math.ceil(self.num_repeats * len(self.dataset))
-            indices = indices[torch.tensor([int(i // self.num_repeats) for i in range(repeat_size)])]
-        else:
-            indices = torch.repeat_interleave(indices, repeats=int(self.num_repeats), dim=0)
-        indices = indices.tolist()  # leaving as tensor thrashes dataloader memory
-        # add extra samples to make it evenly divisible
-        padding_size = self.total_size - len(indices)
-        if padding_size > 0:
-            indices += indices[:padding_size]
-        assert len(indices) == self.total_size
-
-        # subsample per rank
-        indices = indices[self.rank:self.total_size:self.num_replicas]
-        assert len(indices) == self.num_samples
-
-        # return up to num selected samples
-        return iter(indices[:self.num_selected_samples])
-
-    def __len__(self):
-        return self.num_selected_samples
+            self.num_selected_samples = math.ceil(len(dataset) / ratio)
 
     def set_epoch(self, epoch):
         self.epoch = epoch
+
+    def _repeat(self, indices: torch.Tensor) -> list:
+        reps = self.num_repeats
+        if isinstance(reps, float) and not reps.is_integer():
+            # fractional repeat factor: index-map resample
+            out_len = math.ceil(reps * len(self.dataset))
+            picks = torch.tensor([int(i // reps) for i in range(out_len)])
+            return indices[picks].tolist()
+        return torch.repeat_interleave(indices, repeats=int(reps), dim=0).tolist()
+
+    def __iter__(self):
+        g = torch.Generator()
+        g.manual_seed(self.epoch)
+        if self.shuffle:
+            order = torch.randperm(len(self.dataset), generator=g)
+        else:
+            order = torch.arange(len(self.dataset))
+        shard = _pad_and_shard(self._repeat(order), self.total_size, self.rank, self.num_replicas)
+        assert len(shard) == self.num_samples
+        return iter(shard[:self.num_selected_samples])
+
+    def __len__(self):
+        return self.num_selected_samples

@@ -1,93 +1,97 @@
-"""Selectable global pooling (reference `timm/layers/adaptive_avgmax_pool.py`)."""
-from typing import Optional, Tuple, Union
+"""Selectable global pooling.
+
+Behavioral parity: /root/reference/timm/layers/adaptive_avgmax_pool.py
+(pool_type strings, feat-mult, fast/NHWC constraints).  Redesigned around a
+single mode-parameterized reduction instead of one class per mode: the four
+Fast* class names survive as thin subclasses of ``FastGlobalPool`` so module
+reprs and isinstance checks stay compatible.
+"""
+from typing import Tuple, Union
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from .format import get_spatial_dim, get_channel_dim
+from .format import get_channel_dim, get_spatial_dim
 
 _int_tuple_2_t = Union[int, Tuple[int, int]]
 
+_CAT_MODES = ('catavgmax',)
 
-def adaptive_pool_feat_mult(pool_type='avg'):
-    if pool_type.endswith('catavgmax'):
-        return 2
-    return 1
+
+def adaptive_pool_feat_mult(pool_type: str = 'avg') -> int:
+    """Channel multiplier of the pooled output (2 for concat modes)."""
+    return 2 if any(pool_type.endswith(m) for m in _CAT_MODES) else 1
 
 
 def adaptive_avgmax_pool2d(x, output_size: _int_tuple_2_t = 1):
-    x_avg = F.adaptive_avg_pool2d(x, output_size)
-    x_max = F.adaptive_max_pool2d(x, output_size)
-    return 0.5 * (x_avg + x_max)
+    return 0.5 * (F.adaptive_avg_pool2d(x, output_size) +
+                  F.adaptive_max_pool2d(x, output_size))
 
 
 def adaptive_catavgmax_pool2d(x, output_size: _int_tuple_2_t = 1):
-    x_avg = F.adaptive_avg_pool2d(x, output_size)
-    x_max = F.adaptive_max_pool2d(x, output_size)
-    return torch.cat((x_avg, x_max), 1)
+    return torch.cat((
+        F.adaptive_avg_pool2d(x, output_size),
+        F.adaptive_max_pool2d(x, output_size),
+    ), 1)
 
 
-def select_adaptive_pool2d(x, pool_type='avg', output_size: _int_tuple_2_t = 1):
-    if pool_type == 'avg':
-        x = F.adaptive_avg_pool2d(x, output_size)
-    elif pool_type == 'avgmax':
-        x = adaptive_avgmax_pool2d(x, output_size)
-    elif pool_type == 'catavgmax':
-        x = adaptive_catavgmax_pool2d(x, output_size)
-    elif pool_type == 'max':
-        x = F.adaptive_max_pool2d(x, output_size)
-    else:
-        assert False, 'Invalid pool type: %s' % pool_type
-    return x
+def select_adaptive_pool2d(x, pool_type: str = 'avg', output_size: _int_tuple_2_t = 1):
+    """Functional NCHW adaptive pool selected by name."""
+    fns = {
+        'avg': F.adaptive_avg_pool2d,
+        'max': F.adaptive_max_pool2d,
+        'avgmax': adaptive_avgmax_pool2d,
+        'catavgmax': adaptive_catavgmax_pool2d,
+    }
+    assert pool_type in fns, f'Invalid pool type: {pool_type}'
+    return fns[pool_type](x, output_size)
 
 
-class FastAdaptiveAvgPool(nn.Module):
-    def __init__(self, flatten: bool = False, input_fmt: str = 'NCHW'):
+class FastGlobalPool(nn.Module):
+    """Global (output_size=1) pooling over the spatial dims of NCHW/NHWC maps.
+
+    ``mode``: 'avg' | 'max' | 'avgmax' (mean of both) | 'catavgmax' (concat).
+    """
+
+    def __init__(self, mode: str = 'avg', flatten: bool = False, input_fmt: str = 'NCHW'):
         super().__init__()
+        self.mode = mode
         self.flatten = flatten
         self.dim = get_spatial_dim(input_fmt)
+        self.dim_cat = 1 if flatten else get_channel_dim(input_fmt)
 
     def forward(self, x):
-        return x.mean(self.dim, keepdim=not self.flatten)
+        keep = not self.flatten
+        if self.mode == 'avg':
+            return x.mean(self.dim, keepdim=keep)
+        if self.mode == 'max':
+            return x.amax(self.dim, keepdim=keep)
+        avg = x.mean(self.dim, keepdim=keep)
+        mx = x.amax(self.dim, keepdim=keep)
+        if self.mode == 'catavgmax':
+            return torch.cat((avg, mx), self.dim_cat)
+        return 0.5 * avg + 0.5 * mx
 
 
-class FastAdaptiveMaxPool(nn.Module):
+class FastAdaptiveAvgPool(FastGlobalPool):
     def __init__(self, flatten: bool = False, input_fmt: str = 'NCHW'):
-        super().__init__()
-        self.flatten = flatten
-        self.dim = get_spatial_dim(input_fmt)
-
-    def forward(self, x):
-        return x.amax(self.dim, keepdim=not self.flatten)
+        super().__init__('avg', flatten, input_fmt)
 
 
-class FastAdaptiveAvgMaxPool(nn.Module):
+class FastAdaptiveMaxPool(FastGlobalPool):
     def __init__(self, flatten: bool = False, input_fmt: str = 'NCHW'):
-        super().__init__()
-        self.flatten = flatten
-        self.dim = get_spatial_dim(input_fmt)
-
-    def forward(self, x):
-        x_avg = x.mean(self.dim, keepdim=not self.flatten)
-        x_max = x.amax(self.dim, keepdim=not self.flatten)
-        return 0.5 * x_avg + 0.5 * x_max
+        super().__init__('max', flatten, input_fmt)
 
 
-class FastAdaptiveCatAvgMaxPool(nn.Module):
+class FastAdaptiveAvgMaxPool(FastGlobalPool):
     def __init__(self, flatten: bool = False, input_fmt: str = 'NCHW'):
-        super().__init__()
-        self.flatten = flatten
-        self.dim_reduce = get_spatial_dim(input_fmt)
-        if flatten:
-            self.dim_cat = 1
-        else:
-            self.dim_cat = get_channel_dim(input_fmt)
+        super().__init__('avgmax', flatten, input_fmt)
 
-    def forward(self, x):
-        x_avg = x.mean(self.dim_reduce, keepdim=not self.flatten)
-        x_max = x.amax(self.dim_reduce, keepdim=not self.flatten)
-        return torch.cat((x_avg, x_max), self.dim_cat)
+
+class FastAdaptiveCatAvgMaxPool(FastGlobalPool):
+    def __init__(self, flatten: bool = False, input_fmt: str = 'NCHW'):
+        super().__init__('catavgmax', flatten, input_fmt)
 
 
 class AdaptiveAvgMaxPool2d(nn.Module):
@@ -109,7 +113,12 @@ class AdaptiveCatAvgMaxPool2d(nn.Module):
 
 
 class SelectAdaptivePool2d(nn.Module):
-    """Selectable global pooling layer with dynamic input kernel size."""
+    """Global pooling layer selected by pool_type string.
+
+    'fast*' variants (and any non-NCHW input) reduce with tensor ops and
+    require output_size == 1; classic variants use torch adaptive pools.
+    Empty pool_type passes through.
+    """
 
     def __init__(
             self,
@@ -121,49 +130,43 @@ class SelectAdaptivePool2d(nn.Module):
         super().__init__()
         assert input_fmt in ('NCHW', 'NHWC')
         self.pool_type = pool_type or ''
-        pool_type = pool_type.lower()
-        if not pool_type:
-            self.pool = nn.Identity()  # pass through
+        key = self.pool_type.lower()
+
+        if not key:
+            self.pool = nn.Identity()
             self.flatten = nn.Flatten(1) if flatten else nn.Identity()
-        elif pool_type.startswith('fast') or input_fmt != 'NCHW':
-            assert output_size == 1, 'Fast pooling and non NCHW input formats require output_size == 1.'
-            if pool_type.endswith('catavgmax'):
-                self.pool = FastAdaptiveCatAvgMaxPool(flatten, input_fmt=input_fmt)
-            elif pool_type.endswith('avgmax'):
-                self.pool = FastAdaptiveAvgMaxPool(flatten, input_fmt=input_fmt)
-            elif pool_type.endswith('max'):
-                self.pool = FastAdaptiveMaxPool(flatten, input_fmt=input_fmt)
-            elif pool_type == 'fast' or pool_type.endswith('avg'):
-                self.pool = FastAdaptiveAvgPool(flatten, input_fmt=input_fmt)
-            else:
-                assert False, 'Invalid pool type: %s' % pool_type
+            return
+
+        if key.startswith('fast') or input_fmt != 'NCHW':
+            assert output_size == 1, \
+                'Fast pooling and non NCHW input formats require output_size == 1.'
+            mode = key[4:].lstrip('_') if key.startswith('fast') else key
+            mode = mode or 'avg'
+            assert mode in ('avg', 'max', 'avgmax', 'catavgmax'), \
+                f'Invalid pool type: {pool_type}'
+            self.pool = FastGlobalPool(mode, flatten, input_fmt=input_fmt)
             self.flatten = nn.Identity()
-        else:
-            assert input_fmt == 'NCHW'
-            if pool_type == 'avgmax':
-                self.pool = AdaptiveAvgMaxPool2d(output_size)
-            elif pool_type == 'catavgmax':
-                self.pool = AdaptiveCatAvgMaxPool2d(output_size)
-            elif pool_type == 'max':
-                self.pool = nn.AdaptiveMaxPool2d(output_size)
-            elif pool_type == 'avg':
-                self.pool = nn.AdaptiveAvgPool2d(output_size)
-            else:
-                assert False, 'Invalid pool type: %s' % pool_type
-            self.flatten = nn.Flatten(1) if flatten else nn.Identity()
+            return
+
+        classic = {
+            'avg': lambda: nn.AdaptiveAvgPool2d(output_size),
+            'max': lambda: nn.AdaptiveMaxPool2d(output_size),
+            'avgmax': lambda: AdaptiveAvgMaxPool2d(output_size),
+            'catavgmax': lambda: AdaptiveCatAvgMaxPool2d(output_size),
+        }
+        assert key in classic, f'Invalid pool type: {pool_type}'
+        self.pool = classic[key]()
+        self.flatten = nn.Flatten(1) if flatten else nn.Identity()
 
     def is_identity(self):
         return not self.pool_type
 
     def forward(self, x):
-        x = self.pool(x)
-        x = self.flatten(x)
-        return x
+        return self.flatten(self.pool(x))
 
     def feat_mult(self):
         return adaptive_pool_feat_mult(self.pool_type)
 
     def __repr__(self):
-        return self.__class__.__name__ + '(' \
-               + 'pool_type=' + self.pool_type \
-               + ', flatten=' + str(self.flatten) + ')'
+        return (f'{self.__class__.__name__}(pool_type={self.pool_type}'
+                f', flatten={self.flatten})')

@@ -1,47 +1,64 @@
-"""Test-time pooling head (reference `timm/layers/test_time_pool.py`)."""
+"""Test-time pooling: evaluate a classifier at a resolution above its training
+size by sliding the original pool window and avg+max pooling the logit map.
+
+Behavioral parity: /root/reference/timm/layers/test_time_pool.py
+(TestTimePoolHead wrapping + the larger-than-default-size activation rule).
+"""
 import logging
-from torch import nn
+
 import torch.nn.functional as F
+from torch import nn
 
 from .adaptive_avgmax_pool import adaptive_avgmax_pool2d
 
 _logger = logging.getLogger(__name__)
 
+__all__ = ['TestTimePoolHead', 'apply_test_time_pool']
+
 
 class TestTimePoolHead(nn.Module):
+    """Replaces a model's pooled classifier with a convolutional one applied
+    to the unpooled feature map, then avg+max pools the class logits."""
+
     def __init__(self, base, original_pool=7):
         super().__init__()
         self.base = base
         self.original_pool = original_pool
-        base_fc = self.base.get_classifier()
-        if isinstance(base_fc, nn.Conv2d):
-            self.fc = base_fc
-        else:
-            self.fc = nn.Conv2d(
-                self.base.num_features, self.base.num_classes, kernel_size=1, bias=True)
-            self.fc.weight.data.copy_(base_fc.weight.data.view(self.fc.weight.size()))
-            self.fc.bias.data.copy_(base_fc.bias.data.view(self.fc.bias.size()))
-        self.base.reset_classifier(0)  # delete original fc layer
+        self.fc = self._as_conv_classifier(base)
+        base.reset_classifier(0)  # head now lives in self.fc
+
+    @staticmethod
+    def _as_conv_classifier(base):
+        head = base.get_classifier()
+        if isinstance(head, nn.Conv2d):
+            return head
+        conv = nn.Conv2d(base.num_features, base.num_classes, kernel_size=1, bias=True)
+        conv.weight.data.copy_(head.weight.data.view(conv.weight.shape))
+        conv.bias.data.copy_(head.bias.data.view(conv.bias.shape))
+        return conv
 
     def forward(self, x):
-        x = self.base.forward_features(x)
-        x = F.avg_pool2d(x, kernel_size=self.original_pool, stride=1)
-        x = self.fc(x)
-        x = adaptive_avgmax_pool2d(x, 1)
-        return x.flatten(1)
+        feats = self.base.forward_features(x)
+        # stride-1 window of the train-time pool size keeps per-position logits
+        feats = F.avg_pool2d(feats, kernel_size=self.original_pool, stride=1)
+        logits = self.fc(feats)
+        return adaptive_avgmax_pool2d(logits, 1).flatten(1)
 
 
 def apply_test_time_pool(model, config, use_test_size=False):
-    test_time_pool = False
-    if not hasattr(model, 'default_cfg') or not model.default_cfg:
+    """Wrap ``model`` in TestTimePoolHead when the eval input size exceeds the
+    pretrained default in both spatial dims.  Returns (model, enabled)."""
+    cfg = getattr(model, 'default_cfg', None)
+    if not cfg:
         return model, False
-    if use_test_size and 'test_input_size' in model.default_cfg:
-        df_input_size = model.default_cfg['test_input_size']
+    if use_test_size and 'test_input_size' in cfg:
+        trained_size = cfg['test_input_size']
     else:
-        df_input_size = model.default_cfg['input_size']
-    if config['input_size'][-1] > df_input_size[-1] and config['input_size'][-2] > df_input_size[-2]:
-        _logger.info('Target input size %s > pretrained default %s, using test time pooling' %
-                     (str(config['input_size'][-2:]), str(df_input_size[-2:])))
-        model = TestTimePoolHead(model, original_pool=model.default_cfg['pool_size'])
-        test_time_pool = True
-    return model, test_time_pool
+        trained_size = cfg['input_size']
+    target = config['input_size']
+    if not (target[-1] > trained_size[-1] and target[-2] > trained_size[-2]):
+        return model, False
+    _logger.info(
+        'Target input size %s > pretrained default %s, using test time pooling',
+        str(target[-2:]), str(trained_size[-2:]))
+    return TestTimePoolHead(model, original_pool=cfg['pool_size']), True

@@ -1,34 +1,41 @@
-"""Jensen-Shannon Divergence + CE loss for AugMix (reference `timm/loss/jsd.py`)."""
+"""AugMix consistency loss: clean-split CE + Jensen-Shannon divergence across
+augmentation splits.  Behavioral parity: /root/reference/timm/loss/jsd.py.
+"""
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
 from .cross_entropy import LabelSmoothingCrossEntropy
 
+__all__ = ['JsdCrossEntropy']
+
 
 class JsdCrossEntropy(nn.Module):
-    """Jensen-Shannon Divergence + Cross-Entropy Loss (AugMix consistency)."""
+    """CE on the clean split + alpha * JSD(clean, aug1, ..., augN-1).
+
+    Expects the batch stacked as num_splits equal chunks (clean first), the
+    AugMixDataset layout.
+    """
 
     def __init__(self, num_splits=3, alpha=12, smoothing=0.1):
         super().__init__()
         self.num_splits = num_splits
         self.alpha = alpha
-        if smoothing is not None and smoothing > 0:
-            self.cross_entropy_loss = LabelSmoothingCrossEntropy(smoothing)
-        else:
-            self.cross_entropy_loss = torch.nn.CrossEntropyLoss()
+        self.ce = (
+            LabelSmoothingCrossEntropy(smoothing)
+            if smoothing else nn.CrossEntropyLoss()
+        )
 
-    def __call__(self, output, target):
-        split_size = output.shape[0] // self.num_splits
-        assert split_size * self.num_splits == output.shape[0]
-        logits_split = torch.split(output, split_size)
+    def forward(self, output, target):
+        n = output.shape[0] // self.num_splits
+        assert n * self.num_splits == output.shape[0]
+        chunks = output.split(n)
 
-        # Cross-entropy is only computed on clean images
-        loss = self.cross_entropy_loss(logits_split[0], target[:split_size])
-        probs = [F.softmax(logits, dim=1) for logits in logits_split]
+        # supervised term sees only the unaugmented first chunk
+        loss = self.ce(chunks[0], target[:n])
 
-        # Clamp mixture distribution to avoid exploding KL divergence
-        logp_mixture = torch.clamp(torch.stack(probs).mean(axis=0), 1e-7, 1).log()
-        loss += self.alpha * sum([F.kl_div(
-            logp_mixture, p_split, reduction='batchmean') for p_split in probs]) / len(probs)
-        return loss
+        # JSD term: mean KL of each split distribution to the mixture
+        dists = [F.softmax(c, dim=1) for c in chunks]
+        log_mix = torch.stack(dists).mean(dim=0).clamp(1e-7, 1).log()
+        jsd = sum(F.kl_div(log_mix, d, reduction='batchmean') for d in dists)
+        return loss + self.alpha * jsd / len(dists)

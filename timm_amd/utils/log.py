@@ -1,25 +1,33 @@
-"""Logging helpers (reference `timm/utils/log.py:19`)."""
+"""Logging setup.  Behavioral parity: /root/reference/timm/utils/log.py:19
+(bare INFO lines on console, timestamped rotating file log when requested).
+"""
 import logging
 import logging.handlers
 
+__all__ = ['FormatterNoInfo', 'setup_default_logging']
+
 
 class FormatterNoInfo(logging.Formatter):
+    """INFO records print bare; other levels keep the 'LEVEL: msg' prefix."""
+
     def __init__(self, fmt='%(levelname)s: %(message)s'):
-        logging.Formatter.__init__(self, fmt)
+        super().__init__(fmt)
 
     def format(self, record):
         if record.levelno == logging.INFO:
             return str(record.getMessage())
-        return logging.Formatter.format(self, record)
+        return super().format(record)
 
 
 def setup_default_logging(default_level=logging.INFO, log_path=''):
-    console_handler = logging.StreamHandler()
-    console_handler.setFormatter(FormatterNoInfo())
-    logging.root.addHandler(console_handler)
-    logging.root.setLevel(default_level)
+    console = logging.StreamHandler()
+    console.setFormatter(FormatterNoInfo())
+    root = logging.root
+    root.addHandler(console)
+    root.setLevel(default_level)
     if log_path:
-        file_handler = logging.handlers.RotatingFileHandler(log_path, maxBytes=(2 ** 20) * 10, backupCount=3)
-        file_formatter = logging.Formatter("%(asctime)s - %(name)20s: [%(levelname)8s] - %(message)s")
-        file_handler.setFormatter(file_formatter)
-        logging.root.addHandler(file_handler)
+        rotating = logging.handlers.RotatingFileHandler(
+            log_path, maxBytes=10 * (1 << 20), backupCount=3)
+        rotating.setFormatter(logging.Formatter(
+            '%(asctime)s - %(name)20s: [%(levelname)8s] - %(message)s'))
+        root.addHandler(rotating)

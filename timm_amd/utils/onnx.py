@@ -1,18 +1,30 @@
-"""ONNX export helper (reference `timm/utils/onnx.py:17`)."""
-from typing import Optional, Tuple, List
+"""ONNX export + runtime-forward helpers.
+
+Behavioral parity: /root/reference/timm/utils/onnx.py:17 (same export knobs:
+dynamic batch/size axes, aten fallback, dynamo path, checker + numeric
+round-trip verification).
+"""
+from typing import List, Optional, Tuple
 
 import torch
 
+__all__ = ['onnx_forward', 'onnx_export']
+
 
 def onnx_forward(onnx_file, example_input):
+    """Run one forward through onnxruntime, returning the first output."""
     import onnxruntime
 
-    sess_options = onnxruntime.SessionOptions()
-    session = onnxruntime.InferenceSession(onnx_file, sess_options)
-    input_name = session.get_inputs()[0].name
-    output = session.run([], {input_name: example_input.numpy()})
-    output = output[0]
-    return output
+    session = onnxruntime.InferenceSession(onnx_file, onnxruntime.SessionOptions())
+    feed = {session.get_inputs()[0].name: example_input.numpy()}
+    return session.run([], feed)[0]
+
+
+def _resolve_example_input(model, input_size, batch_size, requires_grad):
+    if input_size is None:
+        assert hasattr(model, 'default_cfg')
+        input_size = model.default_cfg.get('input_size')
+    return torch.randn((batch_size,) + tuple(input_size), requires_grad=requires_grad)
 
 
 def onnx_export(
@@ -35,51 +47,33 @@ def onnx_export(
 ):
     import onnx
 
-    if training:
-        training_mode = torch.onnx.TrainingMode.TRAINING
-        model.train()
-    else:
-        training_mode = torch.onnx.TrainingMode.EVAL
-        model.eval()
-
+    model.train(training)
     if example_input is None:
-        if input_size is None:
-            assert hasattr(model, 'default_cfg')
-            input_size = model.default_cfg.get('input_size')
-        example_input = torch.randn((batch_size,) + input_size, requires_grad=training)
+        example_input = _resolve_example_input(model, input_size, batch_size, training)
 
-    # Run model once before export trace: sets padding for models with
-    # 'same'-style padding and ensures better tracing.
-    original_out = model(example_input)
-
-    input_names = input_names or ["input0"]
-    output_names = output_names or ["output0"]
-
-    dynamic_axes = {'input0': {0: 'batch'}, 'output0': {0: 'batch'}}
-    if dynamic_size:
-        dynamic_axes['input0'][2] = 'height'
-        dynamic_axes['input0'][3] = 'width'
-
-    if aten_fallback:
-        export_type = torch.onnx.OperatorExportTypes.ONNX_ATEN_FALLBACK
-    else:
-        export_type = torch.onnx.OperatorExportTypes.ONNX
+    # prime the model once: 'same'-pad layers latch their padding, and the
+    # eager output doubles as the dynamo-path verification reference
+    eager_out = model(example_input)
 
     if use_dynamo:
-        export_options = torch.onnx.ExportOptions(dynamic_shapes=dynamic_size)
         export_output = torch.onnx.dynamo_export(
-            model,
-            example_input,
-            export_options=export_options,
+            model, example_input,
+            export_options=torch.onnx.ExportOptions(dynamic_shapes=dynamic_size),
         )
         export_output.save(output_file)
         torch_out = None
     else:
+        input_names = input_names or ['input0']
+        output_names = output_names or ['output0']
+        dynamic_axes = {'input0': {0: 'batch'}, 'output0': {0: 'batch'}}
+        if dynamic_size:
+            dynamic_axes['input0'].update({2: 'height', 3: 'width'})
         torch_out = torch.onnx._export(
             model,
             example_input,
             output_file,
-            training=training_mode,
+            training=(torch.onnx.TrainingMode.TRAINING if training
+                      else torch.onnx.TrainingMode.EVAL),
             export_params=True,
             verbose=verbose,
             input_names=input_names,
@@ -87,16 +81,15 @@ def onnx_export(
             keep_initializers_as_inputs=keep_initializers,
             dynamic_axes=dynamic_axes,
             opset_version=opset,
-            operator_export_type=export_type,
+            operator_export_type=(
+                torch.onnx.OperatorExportTypes.ONNX_ATEN_FALLBACK if aten_fallback
+                else torch.onnx.OperatorExportTypes.ONNX),
         )
 
     if check:
-        onnx_model = onnx.load(output_file)
-        onnx.checker.check_model(onnx_model, full_check=True)  # assuming throw on error
+        onnx.checker.check_model(onnx.load(output_file), full_check=True)
         if check_forward and not training:
             import numpy as np
-            onnx_out = onnx_forward(output_file, example_input)
-            if torch_out is not None:
-                np.testing.assert_almost_equal(torch_out.numpy(), onnx_out, decimal=3)
-            else:
-                np.testing.assert_almost_equal(original_out.detach().numpy(), onnx_out, decimal=3)
+            ort_out = onnx_forward(output_file, example_input)
+            reference = torch_out if torch_out is not None else eager_out.detach()
+            np.testing.assert_almost_equal(reference.numpy(), ort_out, decimal=3)

@@ -1,28 +1,33 @@
-"""Epoch summary CSV / wandb logging (reference `timm/utils/summary.py:14,30`)."""
+"""Experiment output dirs + per-epoch metric rows (CSV and optional wandb).
+
+Behavioral parity: /root/reference/timm/utils/summary.py:14,30 (same CSV
+column naming: epoch, train_*, eval_*, lr).
+"""
 import csv
 import os
-from collections import OrderedDict
 
 try:
     import wandb
 except ImportError:
     wandb = None
 
+__all__ = ['get_outdir', 'update_summary']
+
 
 def get_outdir(path, *paths, inc=False):
-    outdir = os.path.join(path, *paths)
-    if not os.path.exists(outdir):
-        os.makedirs(outdir)
-    elif inc:
-        count = 1
-        outdir_inc = outdir + '-' + str(count)
-        while os.path.exists(outdir_inc):
-            count = count + 1
-            outdir_inc = outdir + '-' + str(count)
-            assert count < 100
-        outdir = outdir_inc
-        os.makedirs(outdir)
-    return outdir
+    """mkdir -p the joined path; with ``inc``, suffix -1, -2, ... if taken."""
+    base = os.path.join(path, *paths)
+    if not os.path.exists(base):
+        os.makedirs(base)
+        return base
+    if not inc:
+        return base
+    for n in range(1, 100):
+        candidate = f'{base}-{n}'
+        if not os.path.exists(candidate):
+            os.makedirs(candidate)
+            return candidate
+    raise AssertionError(f'could not find a free increment of {base}')
 
 
 def update_summary(
@@ -34,16 +39,17 @@ def update_summary(
         write_header=False,
         log_wandb=False,
 ):
-    rowd = OrderedDict(epoch=epoch)
-    rowd.update([('train_' + k, v) for k, v in train_metrics.items()])
+    """Append one epoch row to the summary CSV (and wandb when enabled)."""
+    row = {'epoch': epoch}
+    row.update({f'train_{k}': v for k, v in train_metrics.items()})
     if eval_metrics:
-        rowd.update([('eval_' + k, v) for k, v in eval_metrics.items()])
+        row.update({f'eval_{k}': v for k, v in eval_metrics.items()})
     if lr is not None:
-        rowd['lr'] = lr
+        row['lr'] = lr
     if log_wandb and wandb is not None:
-        wandb.log(rowd)
-    with open(filename, mode='a') as cf:
-        dw = csv.DictWriter(cf, fieldnames=rowd.keys())
-        if write_header:  # first iteration (epoch == 1 can't be used)
-            dw.writeheader()
-        dw.writerow(rowd)
+        wandb.log(row)
+    with open(filename, mode='a') as f:
+        writer = csv.DictWriter(f, fieldnames=row.keys())
+        if write_header:  # caller tracks first write (resume may append)
+            writer.writeheader()
+        writer.writerow(row)
