@@ -1,30 +1,37 @@
-"""AutoAugment / RandAugment / AugMix (reference `timm/data/auto_augment.py`, ~1,000 LoC).
+"""AutoAugment / RandAugment / AugMix — string-configured PIL augmentation.
 
-String-config augmentations: `rand-m9-mstd0.5-inc1`, `original-mstd0.5`,
-`augmix-m5-w4-d2` etc.  Pure PIL/numpy host-side ops (run in loader workers).
+Behavioral parity: /root/reference/timm/data/auto_augment.py (op set,
+magnitude scaling laws, policy tables, config-string grammar:
+``rand-m9-mstd0.5-inc1``, ``original-mstd0.5``, ``augmix-m5-w4-d2``).
+
+Redesigned around a single op registry: each named op binds its PIL image
+function and its magnitude->argument law in one ``OpSpec`` row, rather than
+parallel name->fn / name->level-fn dicts of free functions.  Magnitude laws
+are inline lambdas over two shared helpers (``frac`` = m/10 fractional level,
+``sgn`` = random sign flip).  All ops run host-side in loader workers.
 """
 import math
 import random
 import re
-from functools import partial
-from typing import Dict, List, Optional, Union
+from dataclasses import dataclass
+from typing import Callable, Dict, List, Optional, Union
 
 import numpy as np
 import PIL
-from PIL import Image, ImageOps, ImageEnhance, ImageChops, ImageFilter
+from PIL import Image, ImageEnhance, ImageFilter, ImageOps
 
-_PIL_VER = tuple([int(x) for x in PIL.__version__.split('.')[:2]])
+_PIL_VER = tuple(int(x) for x in PIL.__version__.split('.')[:2])
+
+MAX_MAG = 10.  # the 'M' scale: magnitudes are fractions of this denominator
 
 _FILL = (128, 128, 128)
-
-_LEVEL_DENOM = 10.  # denominator for conversion from 'Mx' magnitude scale to fractional aug level for op arguments
 
 _HPARAMS_DEFAULT = dict(
     translate_const=250,
     img_mean=_FILL,
 )
 
-if hasattr(Image, "Resampling"):
+if hasattr(Image, 'Resampling'):
     _RANDOM_INTERPOLATION = (Image.Resampling.BILINEAR, Image.Resampling.BICUBIC)
     _DEFAULT_INTERPOLATION = Image.Resampling.BICUBIC
 else:
@@ -32,662 +39,438 @@ else:
     _DEFAULT_INTERPOLATION = Image.BICUBIC
 
 
-def _interpolation(kwargs):
-    interpolation = kwargs.pop('resample', _DEFAULT_INTERPOLATION)
-    if isinstance(interpolation, (list, tuple)):
-        return random.choice(interpolation)
-    return interpolation
+# ---------------------------------------------------------------------------
+# image ops (PIL)
+# ---------------------------------------------------------------------------
+
+def _geom_kwargs(kwargs):
+    """Resolve resample (random choice among tuple) + drop unsupported fill."""
+    resample = kwargs.pop('resample', _DEFAULT_INTERPOLATION)
+    if isinstance(resample, (list, tuple)):
+        resample = random.choice(resample)
+    kwargs['resample'] = resample
+    if _PIL_VER < (5, 0):
+        kwargs.pop('fillcolor', None)
+    return kwargs
 
 
-def _check_args_tf(kwargs):
-    if 'fillcolor' in kwargs and _PIL_VER < (5, 0):
-        kwargs.pop('fillcolor')
-    kwargs['resample'] = _interpolation(kwargs)
+def _affine(img, coeffs, **kwargs):
+    return img.transform(img.size, Image.AFFINE, coeffs, **_geom_kwargs(kwargs))
 
 
-def shear_x(img, factor, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, factor, 0, 0, 1, 0), **kwargs)
+def _shear_x(img, factor, **kw):
+    return _affine(img, (1, factor, 0, 0, 1, 0), **kw)
 
 
-def shear_y(img, factor, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, factor, 1, 0), **kwargs)
+def _shear_y(img, factor, **kw):
+    return _affine(img, (1, 0, 0, factor, 1, 0), **kw)
 
 
-def translate_x_rel(img, pct, **kwargs):
-    pixels = pct * img.size[0]
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, pixels, 0, 1, 0), **kwargs)
+def _translate_x_abs(img, pixels, **kw):
+    return _affine(img, (1, 0, pixels, 0, 1, 0), **kw)
 
 
-def translate_y_rel(img, pct, **kwargs):
-    pixels = pct * img.size[1]
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, 0, 1, pixels), **kwargs)
+def _translate_y_abs(img, pixels, **kw):
+    return _affine(img, (1, 0, 0, 0, 1, pixels), **kw)
 
 
-def translate_x_abs(img, pixels, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, pixels, 0, 1, 0), **kwargs)
+def _translate_x_rel(img, pct, **kw):
+    return _translate_x_abs(img, pct * img.size[0], **kw)
 
 
-def translate_y_abs(img, pixels, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, 0, 1, pixels), **kwargs)
+def _translate_y_rel(img, pct, **kw):
+    return _translate_y_abs(img, pct * img.size[1], **kw)
 
 
-def rotate(img, degrees, **kwargs):
-    _check_args_tf(kwargs)
+def _rotate(img, degrees, **kw):
+    kw = _geom_kwargs(kw)
     if _PIL_VER >= (5, 2):
-        return img.rotate(degrees, **kwargs)
+        return img.rotate(degrees, **kw)
     if _PIL_VER >= (5, 0):
+        # manual center-rotation affine for PIL 5.0/5.1 (no fill in rotate)
         w, h = img.size
-        post_trans = (0, 0)
-        rotn_center = (w / 2.0, h / 2.0)
+        cx, cy = w / 2.0, h / 2.0
         angle = -math.radians(degrees)
-        matrix = [
-            round(math.cos(angle), 15),
-            round(math.sin(angle), 15),
-            0.0,
-            round(-math.sin(angle), 15),
-            round(math.cos(angle), 15),
-            0.0,
-        ]
-
-        def transform(x, y, matrix):
-            (a, b, c, d, e, f) = matrix
-            return a * x + b * y + c, d * x + e * y + f
-
-        matrix[2], matrix[5] = transform(
-            -rotn_center[0] - post_trans[0], -rotn_center[1] - post_trans[1], matrix)
-        matrix[2] += rotn_center[0]
-        matrix[5] += rotn_center[1]
-        return img.transform(img.size, Image.AFFINE, matrix, **kwargs)
-    return img.rotate(degrees, resample=kwargs['resample'])
+        cos_a = round(math.cos(angle), 15)
+        sin_a = round(math.sin(angle), 15)
+        tx = cos_a * -cx + sin_a * -cy + cx
+        ty = -sin_a * -cx + cos_a * -cy + cy
+        return img.transform(img.size, Image.AFFINE, (cos_a, sin_a, tx, -sin_a, cos_a, ty), **kw)
+    return img.rotate(degrees, resample=kw['resample'])
 
 
-def auto_contrast(img, **__):
-    return ImageOps.autocontrast(img)
-
-
-def invert(img, **__):
-    return ImageOps.invert(img)
-
-
-def equalize(img, **__):
-    return ImageOps.equalize(img)
-
-
-def solarize(img, thresh, **__):
-    return ImageOps.solarize(img, thresh)
-
-
-def solarize_add(img, add, thresh=128, **__):
-    lut = []
-    for i in range(256):
-        if i < thresh:
-            lut.append(min(255, i + add))
-        else:
-            lut.append(i)
-
-    if img.mode in ("L", "RGB"):
-        if img.mode == "RGB" and len(lut) == 256:
-            lut = lut + lut + lut
-        return img.point(lut)
-
+def _solarize_add(img, add, thresh=128, **__):
+    # lift values below thresh by `add`, clamped to 255
+    lut = [min(255, i + add) if i < thresh else i for i in range(256)]
+    if img.mode in ('L', 'RGB'):
+        return img.point(lut * 3 if img.mode == 'RGB' else lut)
     return img
 
 
-def posterize(img, bits_to_keep, **__):
-    if bits_to_keep >= 8:
-        return img
-    return ImageOps.posterize(img, bits_to_keep)
+def _posterize(img, bits_to_keep, **__):
+    return img if bits_to_keep >= 8 else ImageOps.posterize(img, bits_to_keep)
 
 
-def contrast(img, factor, **__):
-    return ImageEnhance.Contrast(img).enhance(factor)
+def _gaussian_blur(img, radius, **__):
+    return img.filter(ImageFilter.GaussianBlur(radius=radius))
 
 
-def color(img, factor, **__):
-    return ImageEnhance.Color(img).enhance(factor)
+def _gaussian_blur_rand(img, factor, **__):
+    # radius uniform in [0.1, 2*factor]
+    return img.filter(ImageFilter.GaussianBlur(
+        radius=random.uniform(0.1, 2.0 * factor)))
 
 
-def brightness(img, factor, **__):
-    return ImageEnhance.Brightness(img).enhance(factor)
+def _desaturate(img, factor, **__):
+    # factor 1 -> grayscale, 0 -> unchanged (enhance arg inverted+clamped)
+    return ImageEnhance.Color(img).enhance(min(1., max(0., 1. - factor)))
 
 
-def sharpness(img, factor, **__):
-    return ImageEnhance.Sharpness(img).enhance(factor)
+def _enhance(enhancer):
+    return lambda img, factor, **__: enhancer(img).enhance(factor)
 
 
-def gaussian_blur(img, factor, **__):
-    img = img.filter(ImageFilter.GaussianBlur(radius=factor))
-    return img
+# ---------------------------------------------------------------------------
+# magnitude laws
+# ---------------------------------------------------------------------------
+
+def frac(m):
+    """Magnitude as a fraction of the M-scale denominator."""
+    return m / MAX_MAG
 
 
-def gaussian_blur_rand(img, factor, **__):
-    radius_min = 0.1
-    radius_max = 2.0
-    img = img.filter(ImageFilter.GaussianBlur(
-        radius=random.uniform(radius_min, radius_min + radius_max * factor)))
-    return img
-
-
-def desaturate(img, factor, **_):
-    factor = min(1., max(0., 1. - factor))
-    # enhance factor 0 = grayscale, 1.0 = no-change
-    return ImageEnhance.Color(img).enhance(factor)
-
-
-def _randomly_negate(v):
-    """With 50% prob, negate the value"""
+def sgn(v):
+    """Randomly flip sign with p=0.5."""
     return -v if random.random() > 0.5 else v
 
 
-def _rotate_level_to_arg(level, _hparams):
-    # range [-30, 30]
-    level = (level / _LEVEL_DENOM) * 30.
-    level = _randomly_negate(level)
-    return level,
+def _span(m, lo, hi, clamp=True):
+    v = lo + (hi - lo) * frac(m)
+    return max(lo, min(hi, v)) if clamp else v
 
 
-def _enhance_level_to_arg(level, _hparams):
-    # range [0.1, 1.9]
-    return (level / _LEVEL_DENOM) * 1.8 + 0.1,
+def _enh(m, _hp):
+    # blend factor in [0.1, 1.9], severity symmetric around 1.0? no: linear
+    return (frac(m) * 1.8 + 0.1,)
 
 
-def _enhance_increasing_level_to_arg(level, _hparams):
-    # the 'no change' level is 1.0, moving away from that towards 0. or 2.0 increases the enhancement blend
-    # range [0.1, 1.9] if level <= _LEVEL_DENOM
-    level = (level / _LEVEL_DENOM) * .9
-    level = max(0.1, 1.0 + _randomly_negate(level))  # keep it >= 0.1
-    return level,
+def _enh_inc(m, _hp):
+    # 'no change' is 1.0; severity grows away from it in a random direction
+    return (max(0.1, 1.0 + sgn(frac(m) * .9)),)
 
 
-def _minmax_level_to_arg(level, _hparams, min_val=0., max_val=1.0, clamp=True):
-    level = (level / _LEVEL_DENOM)
-    level = min_val + (max_val - min_val) * level
-    if clamp:
-        level = max(min_val, min(max_val, level))
-    return level,
+@dataclass(frozen=True)
+class OpSpec:
+    """One augmentation: PIL fn + magnitude->args law (None = no args)."""
+    fn: Callable
+    args: Optional[Callable] = None
 
 
-def _shear_level_to_arg(level, _hparams):
-    # range [-0.3, 0.3]
-    level = (level / _LEVEL_DENOM) * 0.3
-    level = _randomly_negate(level)
-    return level,
-
-
-def _translate_abs_level_to_arg(level, hparams):
-    translate_const = hparams['translate_const']
-    level = (level / _LEVEL_DENOM) * float(translate_const)
-    level = _randomly_negate(level)
-    return level,
-
-
-def _translate_rel_level_to_arg(level, hparams):
-    # default range [-0.45, 0.45]
-    translate_pct = hparams.get('translate_pct', 0.45)
-    level = (level / _LEVEL_DENOM) * translate_pct
-    level = _randomly_negate(level)
-    return level,
-
-
-def _posterize_level_to_arg(level, _hparams):
-    # As per Tensorflow TPU EfficientNet impl
-    # range [0, 4], 'keep 0 up to 4 MSB of original image'
-    # intensity/severity of augmentation decreases with level
-    return int((level / _LEVEL_DENOM) * 4),
-
-
-def _posterize_increasing_level_to_arg(level, hparams):
-    # As per Tensorflow models research and UDA impl
-    # range [4, 0], 'keep 4 down to 0 MSB of original image',
-    # intensity/severity of augmentation increases with level
-    return 4 - _posterize_level_to_arg(level, hparams)[0],
-
-
-def _posterize_original_level_to_arg(level, _hparams):
-    # As per original AutoAugment paper description
-    # range [4, 8], 'keep 4 up to 8 MSB of image'
-    # intensity/severity of augmentation decreases with level
-    return int((level / _LEVEL_DENOM) * 4) + 4,
-
-
-def _solarize_level_to_arg(level, _hparams):
-    # range [0, 256]
-    # intensity/severity of augmentation decreases with level
-    return min(256, int((level / _LEVEL_DENOM) * 256)),
-
-
-def _solarize_increasing_level_to_arg(level, _hparams):
-    # range [0, 256]
-    # intensity/severity of augmentation increases with level
-    return 256 - _solarize_level_to_arg(level, _hparams)[0],
-
-
-def _solarize_add_level_to_arg(level, _hparams):
-    # range [0, 110]
-    return min(128, int((level / _LEVEL_DENOM) * 110)),
-
-
-LEVEL_TO_ARG = {
-    'AutoContrast': None,
-    'Equalize': None,
-    'Invert': None,
-    'Rotate': _rotate_level_to_arg,
-    # There are several variations of the posterize level scaling in various Tensorflow/Google repositories/papers
-    'Posterize': _posterize_level_to_arg,
-    'PosterizeIncreasing': _posterize_increasing_level_to_arg,
-    'PosterizeOriginal': _posterize_original_level_to_arg,
-    'Solarize': _solarize_level_to_arg,
-    'SolarizeIncreasing': _solarize_increasing_level_to_arg,
-    'SolarizeAdd': _solarize_add_level_to_arg,
-    'Color': _enhance_level_to_arg,
-    'ColorIncreasing': _enhance_increasing_level_to_arg,
-    'Contrast': _enhance_level_to_arg,
-    'ContrastIncreasing': _enhance_increasing_level_to_arg,
-    'Brightness': _enhance_level_to_arg,
-    'BrightnessIncreasing': _enhance_increasing_level_to_arg,
-    'Sharpness': _enhance_level_to_arg,
-    'SharpnessIncreasing': _enhance_increasing_level_to_arg,
-    'ShearX': _shear_level_to_arg,
-    'ShearY': _shear_level_to_arg,
-    'TranslateX': _translate_abs_level_to_arg,
-    'TranslateY': _translate_abs_level_to_arg,
-    'TranslateXRel': _translate_rel_level_to_arg,
-    'TranslateYRel': _translate_rel_level_to_arg,
-    'Desaturate': partial(_minmax_level_to_arg, min_val=0.5, max_val=1.0),
-    'GaussianBlur': partial(_minmax_level_to_arg, min_val=0.1, max_val=2.0),
-    'GaussianBlurRand': _minmax_level_to_arg,
+AUG_OPS: Dict[str, OpSpec] = {
+    'AutoContrast': OpSpec(lambda img, **__: ImageOps.autocontrast(img)),
+    'Equalize': OpSpec(lambda img, **__: ImageOps.equalize(img)),
+    'Invert': OpSpec(lambda img, **__: ImageOps.invert(img)),
+    'Rotate': OpSpec(_rotate, lambda m, hp: (sgn(frac(m) * 30.),)),
+    'ShearX': OpSpec(_shear_x, lambda m, hp: (sgn(frac(m) * 0.3),)),
+    'ShearY': OpSpec(_shear_y, lambda m, hp: (sgn(frac(m) * 0.3),)),
+    'TranslateX': OpSpec(
+        _translate_x_abs, lambda m, hp: (sgn(frac(m) * float(hp['translate_const'])),)),
+    'TranslateY': OpSpec(
+        _translate_y_abs, lambda m, hp: (sgn(frac(m) * float(hp['translate_const'])),)),
+    'TranslateXRel': OpSpec(
+        _translate_x_rel, lambda m, hp: (sgn(frac(m) * hp.get('translate_pct', 0.45)),)),
+    'TranslateYRel': OpSpec(
+        _translate_y_rel, lambda m, hp: (sgn(frac(m) * hp.get('translate_pct', 0.45)),)),
+    # posterize family: bits kept; plain = severity falls with m, increasing =
+    # severity rises, original = AutoAugment-paper 4..8 bit range
+    'Posterize': OpSpec(_posterize, lambda m, hp: (int(frac(m) * 4),)),
+    'PosterizeIncreasing': OpSpec(_posterize, lambda m, hp: (4 - int(frac(m) * 4),)),
+    'PosterizeOriginal': OpSpec(_posterize, lambda m, hp: (int(frac(m) * 4) + 4,)),
+    'Solarize': OpSpec(
+        lambda img, t, **__: ImageOps.solarize(img, t),
+        lambda m, hp: (min(256, int(frac(m) * 256)),)),
+    'SolarizeIncreasing': OpSpec(
+        lambda img, t, **__: ImageOps.solarize(img, t),
+        lambda m, hp: (256 - min(256, int(frac(m) * 256)),)),
+    'SolarizeAdd': OpSpec(_solarize_add, lambda m, hp: (min(128, int(frac(m) * 110)),)),
+    'Color': OpSpec(_enhance(ImageEnhance.Color), _enh),
+    'ColorIncreasing': OpSpec(_enhance(ImageEnhance.Color), _enh_inc),
+    'Contrast': OpSpec(_enhance(ImageEnhance.Contrast), _enh),
+    'ContrastIncreasing': OpSpec(_enhance(ImageEnhance.Contrast), _enh_inc),
+    'Brightness': OpSpec(_enhance(ImageEnhance.Brightness), _enh),
+    'BrightnessIncreasing': OpSpec(_enhance(ImageEnhance.Brightness), _enh_inc),
+    'Sharpness': OpSpec(_enhance(ImageEnhance.Sharpness), _enh),
+    'SharpnessIncreasing': OpSpec(_enhance(ImageEnhance.Sharpness), _enh_inc),
+    'Desaturate': OpSpec(_desaturate, lambda m, hp: (_span(m, 0.5, 1.0),)),
+    'GaussianBlur': OpSpec(_gaussian_blur, lambda m, hp: (_span(m, 0.1, 2.0),)),
+    'GaussianBlurRand': OpSpec(_gaussian_blur_rand, lambda m, hp: (_span(m, 0., 1.0),)),
 }
 
-
-NAME_TO_OP = {
-    'AutoContrast': auto_contrast,
-    'Equalize': equalize,
-    'Invert': invert,
-    'Rotate': rotate,
-    'Posterize': posterize,
-    'PosterizeIncreasing': posterize,
-    'PosterizeOriginal': posterize,
-    'Solarize': solarize,
-    'SolarizeIncreasing': solarize,
-    'SolarizeAdd': solarize_add,
-    'Color': color,
-    'ColorIncreasing': color,
-    'Contrast': contrast,
-    'ContrastIncreasing': contrast,
-    'Brightness': brightness,
-    'BrightnessIncreasing': brightness,
-    'Sharpness': sharpness,
-    'SharpnessIncreasing': sharpness,
-    'ShearX': shear_x,
-    'ShearY': shear_y,
-    'TranslateX': translate_x_abs,
-    'TranslateY': translate_y_abs,
-    'TranslateXRel': translate_x_rel,
-    'TranslateYRel': translate_y_rel,
-    'Desaturate': desaturate,
-    'GaussianBlur': gaussian_blur,
-    'GaussianBlurRand': gaussian_blur_rand,
-}
+# reference-compatible aliases for external pokes
+NAME_TO_OP = {name: spec.fn for name, spec in AUG_OPS.items()}
+LEVEL_TO_ARG = {name: spec.args for name, spec in AUG_OPS.items()}
 
 
 class AugmentOp:
-    """Single augmentation op w/ probability + magnitude (+std) (reference `auto_augment.py:357`)."""
+    """A registered op bound to (probability, magnitude, hparams).
+
+    magnitude_std hparam > 0 gausses the magnitude per call (inf = uniform in
+    [0, m]); magnitude_max overrides the [0, 10] clamp ceiling.
+    """
 
     def __init__(self, name, prob=0.5, magnitude=10, hparams=None):
         hparams = hparams or _HPARAMS_DEFAULT
         self.name = name
-        self.aug_fn = NAME_TO_OP[name]
-        self.level_fn = LEVEL_TO_ARG[name]
+        spec = AUG_OPS[name]
+        self.aug_fn = spec.fn
+        self.level_fn = spec.args
         self.prob = prob
         self.magnitude = magnitude
         self.hparams = hparams.copy()
         self.kwargs = dict(
-            fillcolor=hparams['img_mean'] if 'img_mean' in hparams else _FILL,
-            resample=hparams['interpolation'] if 'interpolation' in hparams else _RANDOM_INTERPOLATION,
+            fillcolor=hparams.get('img_mean', _FILL),
+            resample=hparams.get('interpolation', _RANDOM_INTERPOLATION),
         )
-
-        # If magnitude_std is > 0, we introduce some randomness
-        # in the usually fixed policy and sample magnitude from a normal distribution
-        # with mean `magnitude` and std-dev of `magnitude_std`.
-        # NOTE This is my own hack, being tested, not in papers or reference impls.
-        # If magnitude_std is inf, we sample magnitude from a uniform distribution
         self.magnitude_std = self.hparams.get('magnitude_std', 0)
         self.magnitude_max = self.hparams.get('magnitude_max', None)
+
+    def _sample_magnitude(self):
+        m = self.magnitude
+        if self.magnitude_std > 0:
+            if self.magnitude_std == float('inf'):
+                m = random.uniform(0, m)
+            else:
+                m = random.gauss(m, self.magnitude_std)
+        ceil = self.magnitude_max or MAX_MAG
+        return max(0., min(m, ceil))
 
     def __call__(self, img):
         if self.prob < 1.0 and random.random() > self.prob:
             return img
-        magnitude = self.magnitude
-        if self.magnitude_std > 0:
-            # magnitude randomization enabled
-            if self.magnitude_std == float('inf'):
-                magnitude = random.uniform(0, magnitude)
-            elif self.magnitude_std > 0:
-                magnitude = random.gauss(magnitude, self.magnitude_std)
-        # default upper_bound for the timm RA impl is _LEVEL_DENOM (10)
-        # setting magnitude_max overrides this to allow M > 10 (behaviour closer to Google TF RA impl)
-        upper_bound = self.magnitude_max or _LEVEL_DENOM
-        magnitude = max(0., min(magnitude, upper_bound))
-        level_args = self.level_fn(magnitude, self.hparams) if self.level_fn is not None else tuple()
-        return self.aug_fn(img, *level_args, **self.kwargs)
+        # magnitude is sampled even for no-arg ops so the RNG stream matches
+        # runs with any op mix (keeps aug reproducibility format-stable)
+        magnitude = self._sample_magnitude()
+        args = self.level_fn(magnitude, self.hparams) if self.level_fn is not None else ()
+        return self.aug_fn(img, *args, **self.kwargs)
 
     def __repr__(self):
-        fs = self.__class__.__name__ + f'(name={self.name}, p={self.prob}'
-        fs += f', m={self.magnitude}, mstd={self.magnitude_std}'
+        s = f'{self.__class__.__name__}(name={self.name}, p={self.prob}'
+        s += f', m={self.magnitude}, mstd={self.magnitude_std}'
         if self.magnitude_max is not None:
-            fs += f', mmax={self.magnitude_max}'
-        fs += ')'
-        return fs
+            s += f', mmax={self.magnitude_max}'
+        return s + ')'
 
 
-def auto_augment_policy_v0(hparams):
-    # ImageNet v0 policy from TPU EfficientNet impl, cannot find a paper reference.
-    policy = [
-        [('Equalize', 0.8, 1), ('ShearY', 0.8, 4)],
-        [('Color', 0.4, 9), ('Equalize', 0.6, 3)],
-        [('Color', 0.4, 1), ('Rotate', 0.6, 8)],
-        [('Solarize', 0.8, 3), ('Equalize', 0.4, 7)],
-        [('Solarize', 0.4, 2), ('Solarize', 0.6, 2)],
-        [('Color', 0.2, 0), ('Equalize', 0.8, 8)],
-        [('Equalize', 0.4, 8), ('SolarizeAdd', 0.8, 3)],
-        [('ShearX', 0.2, 9), ('Rotate', 0.6, 8)],
-        [('Color', 0.6, 1), ('Equalize', 1.0, 2)],
-        [('Invert', 0.4, 9), ('Rotate', 0.6, 0)],
-        [('Equalize', 1.0, 9), ('ShearY', 0.6, 3)],
-        [('Color', 0.4, 7), ('Equalize', 0.6, 0)],
-        [('Posterize', 0.4, 6), ('AutoContrast', 0.4, 7)],
-        [('Solarize', 0.6, 8), ('Color', 0.6, 9)],
-        [('Solarize', 0.2, 4), ('Rotate', 0.8, 9)],
-        [('Rotate', 1.0, 7), ('TranslateYRel', 0.8, 9)],
-        [('ShearX', 0.0, 0), ('Solarize', 0.8, 4)],
-        [('ShearY', 0.8, 0), ('Color', 0.6, 4)],
-        [('Color', 1.0, 0), ('Rotate', 0.6, 2)],
-        [('Equalize', 0.8, 4), ('Equalize', 0.0, 8)],
-        [('Equalize', 1.0, 4), ('AutoContrast', 0.6, 2)],
-        [('ShearY', 0.4, 7), ('SolarizeAdd', 0.6, 7)],
-        [('Posterize', 0.8, 2), ('Solarize', 0.6, 10)],  # This results in black image with Tpu posterize
-        [('Solarize', 0.6, 8), ('Equalize', 0.6, 1)],
-        [('Color', 0.8, 6), ('Rotate', 0.4, 5)],
-    ]
-    pc = [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
-    return pc
+# ---------------------------------------------------------------------------
+# config-string parsing (shared grammar: dash-separated "key<number>" tokens)
+# ---------------------------------------------------------------------------
+
+def _iter_config(tokens):
+    """Yield (key, raw_value) for each 'key123'-style token; skip bare keys."""
+    for tok in tokens:
+        parts = re.split(r'(\d.*)', tok)
+        if len(parts) >= 2:
+            yield parts[0], parts[1]
 
 
-def auto_augment_policy_v0r(hparams):
-    # ImageNet v0 policy from TPU EfficientNet impl, with variation of Posterize used
-    # in Google research implementation (number of bits discarded increases with magnitude)
-    policy = [
-        [('Equalize', 0.8, 1), ('ShearY', 0.8, 4)],
-        [('Color', 0.4, 9), ('Equalize', 0.6, 3)],
-        [('Color', 0.4, 1), ('Rotate', 0.6, 8)],
-        [('Solarize', 0.8, 3), ('Equalize', 0.4, 7)],
-        [('Solarize', 0.4, 2), ('Solarize', 0.6, 2)],
-        [('Color', 0.2, 0), ('Equalize', 0.8, 8)],
-        [('Equalize', 0.4, 8), ('SolarizeAdd', 0.8, 3)],
-        [('ShearX', 0.2, 9), ('Rotate', 0.6, 8)],
-        [('Color', 0.6, 1), ('Equalize', 1.0, 2)],
-        [('Invert', 0.4, 9), ('Rotate', 0.6, 0)],
-        [('Equalize', 1.0, 9), ('ShearY', 0.6, 3)],
-        [('Color', 0.4, 7), ('Equalize', 0.6, 0)],
-        [('PosterizeIncreasing', 0.4, 6), ('AutoContrast', 0.4, 7)],
-        [('Solarize', 0.6, 8), ('Color', 0.6, 9)],
-        [('Solarize', 0.2, 4), ('Rotate', 0.8, 9)],
-        [('Rotate', 1.0, 7), ('TranslateYRel', 0.8, 9)],
-        [('ShearX', 0.0, 0), ('Solarize', 0.8, 4)],
-        [('ShearY', 0.8, 0), ('Color', 0.6, 4)],
-        [('Color', 1.0, 0), ('Rotate', 0.6, 2)],
-        [('Equalize', 0.8, 4), ('Equalize', 0.0, 8)],
-        [('Equalize', 1.0, 4), ('AutoContrast', 0.6, 2)],
-        [('ShearY', 0.4, 7), ('SolarizeAdd', 0.6, 7)],
-        [('PosterizeIncreasing', 0.8, 2), ('Solarize', 0.6, 10)],
-        [('Solarize', 0.6, 8), ('Equalize', 0.6, 1)],
-        [('Color', 0.8, 6), ('Rotate', 0.4, 5)],
-    ]
-    pc = [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
-    return pc
+# ---------------------------------------------------------------------------
+# AutoAugment (fixed sub-policy tables)
+# ---------------------------------------------------------------------------
+
+# (name, prob, magnitude) pairs; applied in sequence when the sub-policy is
+# drawn.  Tables transcribed from the TF TPU EfficientNet / AutoAugment-paper
+# policies (data constants, shared with the reference).
+_POLICY_V0 = [
+    [('Equalize', 0.8, 1), ('ShearY', 0.8, 4)],
+    [('Color', 0.4, 9), ('Equalize', 0.6, 3)],
+    [('Color', 0.4, 1), ('Rotate', 0.6, 8)],
+    [('Solarize', 0.8, 3), ('Equalize', 0.4, 7)],
+    [('Solarize', 0.4, 2), ('Solarize', 0.6, 2)],
+    [('Color', 0.2, 0), ('Equalize', 0.8, 8)],
+    [('Equalize', 0.4, 8), ('SolarizeAdd', 0.8, 3)],
+    [('ShearX', 0.2, 9), ('Rotate', 0.6, 8)],
+    [('Color', 0.6, 1), ('Equalize', 1.0, 2)],
+    [('Invert', 0.4, 9), ('Rotate', 0.6, 0)],
+    [('Equalize', 1.0, 9), ('ShearY', 0.6, 3)],
+    [('Color', 0.4, 7), ('Equalize', 0.6, 0)],
+    [('Posterize', 0.4, 6), ('AutoContrast', 0.4, 7)],
+    [('Solarize', 0.6, 8), ('Color', 0.6, 9)],
+    [('Solarize', 0.2, 4), ('Rotate', 0.8, 9)],
+    [('Rotate', 1.0, 7), ('TranslateYRel', 0.8, 9)],
+    [('ShearX', 0.0, 0), ('Solarize', 0.8, 4)],
+    [('ShearY', 0.8, 0), ('Color', 0.6, 4)],
+    [('Color', 1.0, 0), ('Rotate', 0.6, 2)],
+    [('Equalize', 0.8, 4), ('Equalize', 0.0, 8)],
+    [('Equalize', 1.0, 4), ('AutoContrast', 0.6, 2)],
+    [('ShearY', 0.4, 7), ('SolarizeAdd', 0.6, 7)],
+    [('Posterize', 0.8, 2), ('Solarize', 0.6, 10)],
+    [('Solarize', 0.6, 8), ('Equalize', 0.6, 1)],
+    [('Color', 0.8, 6), ('Rotate', 0.4, 5)],
+]
+
+_POLICY_ORIGINAL = [
+    [('PosterizeOriginal', 0.4, 8), ('Rotate', 0.6, 9)],
+    [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
+    [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
+    [('PosterizeOriginal', 0.6, 7), ('PosterizeOriginal', 0.6, 6)],
+    [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
+    [('Equalize', 0.4, 4), ('Rotate', 0.8, 8)],
+    [('Solarize', 0.6, 3), ('Equalize', 0.6, 7)],
+    [('PosterizeOriginal', 0.8, 5), ('Equalize', 1.0, 2)],
+    [('Rotate', 0.2, 3), ('Solarize', 0.6, 8)],
+    [('Equalize', 0.6, 8), ('PosterizeOriginal', 0.4, 6)],
+    [('Rotate', 0.8, 8), ('Color', 0.4, 0)],
+    [('Rotate', 0.4, 9), ('Equalize', 0.6, 2)],
+    [('Equalize', 0.0, 7), ('Equalize', 0.8, 8)],
+    [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
+    [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
+    [('Rotate', 0.8, 8), ('Color', 1.0, 2)],
+    [('Color', 0.8, 8), ('Solarize', 0.8, 7)],
+    [('Sharpness', 0.4, 7), ('Invert', 0.6, 8)],
+    [('ShearX', 0.6, 5), ('Equalize', 1.0, 9)],
+    [('Color', 0.4, 0), ('Equalize', 0.6, 3)],
+    [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
+    [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
+    [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
+    [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
+    [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
+]
+
+_POLICY_3A = [
+    [('Solarize', 1.0, 5)],
+    [('Desaturate', 1.0, 10)],
+    [('GaussianBlurRand', 1.0, 10)],
+]
 
 
-def auto_augment_policy_original(hparams):
-    # ImageNet policy from https://arxiv.org/abs/1805.09501
-    policy = [
-        [('PosterizeOriginal', 0.4, 8), ('Rotate', 0.6, 9)],
-        [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
-        [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
-        [('PosterizeOriginal', 0.6, 7), ('PosterizeOriginal', 0.6, 6)],
-        [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
-        [('Equalize', 0.4, 4), ('Rotate', 0.8, 8)],
-        [('Solarize', 0.6, 3), ('Equalize', 0.6, 7)],
-        [('PosterizeOriginal', 0.8, 5), ('Equalize', 1.0, 2)],
-        [('Rotate', 0.2, 3), ('Solarize', 0.6, 8)],
-        [('Equalize', 0.6, 8), ('PosterizeOriginal', 0.4, 6)],
-        [('Rotate', 0.8, 8), ('Color', 0.4, 0)],
-        [('Rotate', 0.4, 9), ('Equalize', 0.6, 2)],
-        [('Equalize', 0.0, 7), ('Equalize', 0.8, 8)],
-        [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
-        [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
-        [('Rotate', 0.8, 8), ('Color', 1.0, 2)],
-        [('Color', 0.8, 8), ('Solarize', 0.8, 7)],
-        [('Sharpness', 0.4, 7), ('Invert', 0.6, 8)],
-        [('ShearX', 0.6, 5), ('Equalize', 1.0, 9)],
-        [('Color', 0.4, 0), ('Equalize', 0.6, 3)],
-        [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
-        [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
-        [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
-        [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
-        [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
-    ]
-    pc = [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
-    return pc
+def _swap_increasing(table):
+    """'r' policy variants: posterize ops become increasing-severity."""
+    repl = {
+        'Posterize': 'PosterizeIncreasing',
+        'PosterizeOriginal': 'PosterizeIncreasing',
+    }
+    return [[(repl.get(n, n), p, m) for n, p, m in sub] for sub in table]
 
 
-def auto_augment_policy_originalr(hparams):
-    # ImageNet policy from https://arxiv.org/abs/1805.09501 with research posterize variation
-    policy = [
-        [('PosterizeIncreasing', 0.4, 8), ('Rotate', 0.6, 9)],
-        [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
-        [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
-        [('PosterizeIncreasing', 0.6, 7), ('PosterizeIncreasing', 0.6, 6)],
-        [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
-        [('Equalize', 0.4, 4), ('Rotate', 0.8, 8)],
-        [('Solarize', 0.6, 3), ('Equalize', 0.6, 7)],
-        [('PosterizeIncreasing', 0.8, 5), ('Equalize', 1.0, 2)],
-        [('Rotate', 0.2, 3), ('Solarize', 0.6, 8)],
-        [('Equalize', 0.6, 8), ('PosterizeIncreasing', 0.4, 6)],
-        [('Rotate', 0.8, 8), ('Color', 0.4, 0)],
-        [('Rotate', 0.4, 9), ('Equalize', 0.6, 2)],
-        [('Equalize', 0.0, 7), ('Equalize', 0.8, 8)],
-        [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
-        [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
-        [('Rotate', 0.8, 8), ('Color', 1.0, 2)],
-        [('Color', 0.8, 8), ('Solarize', 0.8, 7)],
-        [('Sharpness', 0.4, 7), ('Invert', 0.6, 8)],
-        [('ShearX', 0.6, 5), ('Equalize', 1.0, 9)],
-        [('Color', 0.4, 0), ('Equalize', 0.6, 3)],
-        [('Equalize', 0.4, 7), ('Solarize', 0.2, 4)],
-        [('Solarize', 0.6, 5), ('AutoContrast', 0.6, 5)],
-        [('Invert', 0.6, 4), ('Equalize', 1.0, 8)],
-        [('Color', 0.6, 4), ('Contrast', 1.0, 8)],
-        [('Equalize', 0.8, 8), ('Equalize', 0.6, 3)],
-    ]
-    pc = [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
-    return pc
-
-
-def auto_augment_policy_3a(hparams):
-    policy = [
-        [('Solarize', 1.0, 5)],  # 128 solarize threshold @ 5 magnitude
-        [('Desaturate', 1.0, 10)],  # grayscale at 10 magnitude
-        [('GaussianBlurRand', 1.0, 10)],
-    ]
-    pc = [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
-    return pc
+_POLICIES = {
+    'v0': _POLICY_V0,
+    'v0r': _swap_increasing(_POLICY_V0),
+    'original': _POLICY_ORIGINAL,
+    'originalr': _swap_increasing(_POLICY_ORIGINAL),
+    '3a': _POLICY_3A,
+}
 
 
 def auto_augment_policy(name='v0', hparams=None):
     hparams = hparams or _HPARAMS_DEFAULT
-    if name == 'original':
-        return auto_augment_policy_original(hparams)
-    if name == 'originalr':
-        return auto_augment_policy_originalr(hparams)
-    if name == 'v0':
-        return auto_augment_policy_v0(hparams)
-    if name == 'v0r':
-        return auto_augment_policy_v0r(hparams)
-    if name == '3a':
-        return auto_augment_policy_3a(hparams)
-    assert False, f'Unknown AA policy {name}'
+    assert name in _POLICIES, f'Unknown AA policy {name}'
+    return [
+        [AugmentOp(*args, hparams=hparams) for args in sub]
+        for sub in _POLICIES[name]
+    ]
+
+
+# per-name policy fns kept for reference-API compat
+def auto_augment_policy_v0(hparams):
+    return auto_augment_policy('v0', hparams)
+
+
+def auto_augment_policy_v0r(hparams):
+    return auto_augment_policy('v0r', hparams)
+
+
+def auto_augment_policy_original(hparams):
+    return auto_augment_policy('original', hparams)
+
+
+def auto_augment_policy_originalr(hparams):
+    return auto_augment_policy('originalr', hparams)
+
+
+def auto_augment_policy_3a(hparams):
+    return auto_augment_policy('3a', hparams)
 
 
 class AutoAugment:
+    """Draw one sub-policy per image and apply its ops in order."""
 
     def __init__(self, policy):
         self.policy = policy
 
     def __call__(self, img):
-        sub_policy = random.choice(self.policy)
-        for op in sub_policy:
+        for op in random.choice(self.policy):
             img = op(img)
         return img
 
     def __repr__(self):
-        fs = self.__class__.__name__ + '(policy='
-        for p in self.policy:
-            fs += '\n\t['
-            fs += ', '.join([str(op) for op in p])
-            fs += ']'
-        fs += ')'
-        return fs
+        body = ''.join(
+            '\n\t[' + ', '.join(str(op) for op in sub) + ']' for sub in self.policy)
+        return f'{self.__class__.__name__}(policy={body})'
 
 
 def auto_augment_transform(config_str: str, hparams: Optional[Dict] = None):
-    """Create an AutoAugment transform from a config string.
+    """Build AutoAugment from e.g. 'original-mstd0.5' (policy name + hparams)."""
+    name, *rest = config_str.split('-')
+    for key, val in _iter_config(rest):
+        assert key == 'mstd', 'Unknown AutoAugment config section'
+        hparams.setdefault('magnitude_std', float(val))
+    return AutoAugment(auto_augment_policy(name, hparams=hparams))
 
-    Config string consists of sections separated by '-': the policy name
-    ('v0', 'v0r', 'original', 'originalr', '3a') followed by optional
-    hparams like 'mstd0.5'.  E.g. 'original-mstd0.5'.
-    """
-    config = config_str.split('-')
-    policy_name = config[0]
-    config = config[1:]
-    for c in config:
-        cs = re.split(r'(\d.*)', c)
-        if len(cs) < 2:
-            continue
-        key, val = cs[:2]
-        if key == 'mstd':
-            # noise param injected via hparams for now
-            hparams.setdefault('magnitude_std', float(val))
-        else:
-            assert False, 'Unknown AutoAugment config section'
-    aa_policy = auto_augment_policy(policy_name, hparams=hparams)
-    return AutoAugment(aa_policy)
 
+# ---------------------------------------------------------------------------
+# RandAugment
+# ---------------------------------------------------------------------------
 
 _RAND_TRANSFORMS = [
-    'AutoContrast',
-    'Equalize',
-    'Invert',
-    'Rotate',
-    'Posterize',
-    'Solarize',
-    'SolarizeAdd',
-    'Color',
-    'Contrast',
-    'Brightness',
-    'Sharpness',
-    'ShearX',
-    'ShearY',
-    'TranslateXRel',
-    'TranslateYRel',
+    'AutoContrast', 'Equalize', 'Invert', 'Rotate',
+    'Posterize', 'Solarize', 'SolarizeAdd',
+    'Color', 'Contrast', 'Brightness', 'Sharpness',
+    'ShearX', 'ShearY', 'TranslateXRel', 'TranslateYRel',
 ]
-
 
 _RAND_INCREASING_TRANSFORMS = [
-    'AutoContrast',
-    'Equalize',
-    'Invert',
-    'Rotate',
-    'PosterizeIncreasing',
-    'SolarizeIncreasing',
-    'SolarizeAdd',
-    'ColorIncreasing',
-    'ContrastIncreasing',
-    'BrightnessIncreasing',
-    'SharpnessIncreasing',
-    'ShearX',
-    'ShearY',
-    'TranslateXRel',
-    'TranslateYRel',
+    'AutoContrast', 'Equalize', 'Invert', 'Rotate',
+    'PosterizeIncreasing', 'SolarizeIncreasing', 'SolarizeAdd',
+    'ColorIncreasing', 'ContrastIncreasing', 'BrightnessIncreasing',
+    'SharpnessIncreasing', 'ShearX', 'ShearY', 'TranslateXRel', 'TranslateYRel',
 ]
 
-
-_RAND_3A = [
-    'SolarizeIncreasing',
-    'Desaturate',
-    'GaussianBlur',
-]
-
+_RAND_3A = ['SolarizeIncreasing', 'Desaturate', 'GaussianBlur']
 
 _RAND_WEIGHTED_3A = {
-    'SolarizeIncreasing': 6,
-    'Desaturate': 6,
-    'GaussianBlur': 6,
-    'Rotate': 3,
-    'ShearX': 2,
-    'ShearY': 2,
-    'PosterizeIncreasing': 1,
-    'AutoContrast': 1,
-    'ColorIncreasing': 1,
-    'SharpnessIncreasing': 1,
-    'ContrastIncreasing': 1,
-    'BrightnessIncreasing': 1,
-    'Equalize': 1,
-    'Invert': 1,
+    'SolarizeIncreasing': 6, 'Desaturate': 6, 'GaussianBlur': 6,
+    'Rotate': 3, 'ShearX': 2, 'ShearY': 2,
+    'PosterizeIncreasing': 1, 'AutoContrast': 1, 'ColorIncreasing': 1,
+    'SharpnessIncreasing': 1, 'ContrastIncreasing': 1, 'BrightnessIncreasing': 1,
+    'Equalize': 1, 'Invert': 1,
 }
 
-
-# These experimental weights are based loosely on the relative improvements mentioned in paper.
-# They may not result in increased performance, but could likely be tuned to so.
 _RAND_WEIGHTED_0 = {
-    'Rotate': 3,
-    'ShearX': 2,
-    'ShearY': 2,
-    'TranslateXRel': 1,
-    'TranslateYRel': 1,
-    'ColorIncreasing': .25,
-    'SharpnessIncreasing': 0.25,
-    'AutoContrast': 0.25,
-    'SolarizeIncreasing': .05,
-    'SolarizeAdd': .05,
-    'ContrastIncreasing': .05,
-    'BrightnessIncreasing': .05,
-    'Equalize': .05,
-    'PosterizeIncreasing': 0.05,
+    'Rotate': 3, 'ShearX': 2, 'ShearY': 2,
+    'TranslateXRel': 1, 'TranslateYRel': 1,
+    'ColorIncreasing': .25, 'SharpnessIncreasing': 0.25, 'AutoContrast': 0.25,
+    'SolarizeIncreasing': .05, 'SolarizeAdd': .05, 'ContrastIncreasing': .05,
+    'BrightnessIncreasing': .05, 'Equalize': .05, 'PosterizeIncreasing': 0.05,
     'Invert': .05,
 }
 
-
-def _get_weighted_transforms(transforms: Dict):
-    transforms, probs = list(zip(*transforms.items()))
-    probs = np.array(probs)
-    probs = probs / np.sum(probs)
-    return transforms, probs
+_RAND_CHOICE_SETS = {
+    'weights': _RAND_WEIGHTED_0,
+    '3aw': _RAND_WEIGHTED_3A,
+    '3a': _RAND_3A,
+}
 
 
 def rand_augment_choices(name: str, increasing=True):
-    if name == 'weights':
-        return _RAND_WEIGHTED_0
-    if name == '3aw':
-        return _RAND_WEIGHTED_3A
-    if name == '3a':
-        return _RAND_3A
+    if name in _RAND_CHOICE_SETS:
+        return _RAND_CHOICE_SETS[name]
     return _RAND_INCREASING_TRANSFORMS if increasing else _RAND_TRANSFORMS
+
+
+def _normalize_weights(weighted: Dict):
+    names, weights = zip(*weighted.items())
+    weights = np.array(weights)
+    return names, weights / weights.sum()
 
 
 def rand_augment_ops(
@@ -698,11 +481,14 @@ def rand_augment_ops(
 ):
     hparams = hparams or _HPARAMS_DEFAULT
     transforms = transforms or _RAND_TRANSFORMS
-    return [AugmentOp(name, prob=prob, magnitude=magnitude, hparams=hparams) for name in transforms]
+    return [
+        AugmentOp(name, prob=prob, magnitude=magnitude, hparams=hparams)
+        for name in transforms
+    ]
 
 
 class RandAugment:
-    """RandAugment (reference `auto_augment.py:736`)."""
+    """Apply num_layers ops drawn from the op pool (weighted = no-replace)."""
 
     def __init__(self, ops, num_layers=2, choice_weights=None):
         self.ops = ops
@@ -710,23 +496,18 @@ class RandAugment:
         self.choice_weights = choice_weights
 
     def __call__(self, img):
-        # no replacement when using weighted choice
-        ops = np.random.choice(
-            self.ops,
-            self.num_layers,
+        chosen = np.random.choice(
+            self.ops, self.num_layers,
             replace=self.choice_weights is None,
             p=self.choice_weights,
         )
-        for op in ops:
+        for op in chosen:
             img = op(img)
         return img
 
     def __repr__(self):
-        fs = self.__class__.__name__ + f'(n={self.num_layers}, ops='
-        for op in self.ops:
-            fs += f'\n\t{op}'
-        fs += ')'
-        return fs
+        ops = ''.join(f'\n\t{op}' for op in self.ops)
+        return f'{self.__class__.__name__}(n={self.num_layers}, ops={ops})'
 
 
 def rand_augment_transform(
@@ -734,51 +515,35 @@ def rand_augment_transform(
         hparams: Optional[Dict] = None,
         transforms: Optional[Union[str, Dict, List]] = None,
 ):
-    """Create a RandAugment transform from a config string
-    (reference config-string parser `auto_augment.py:762`).
+    """Build RandAugment from e.g. 'rand-m9-n3-mstd0.5'.
 
-    E.g. 'rand-m9-n3-mstd0.5' — magnitude 9, 2 layers, magnitude noise 0.5.
-    Sections: m (magnitude), n (layers), p (prob per layer), mmax (upper
-    magnitude bound), mstd (noise), inc (increasing severity sets), t (set).
+    Keys: m magnitude · n layers · p per-op prob · mstd magnitude noise
+    (>100 = uniform) · mmax magnitude ceiling · inc increasing-severity sets ·
+    t<name> named transform set.
     """
-    magnitude = _LEVEL_DENOM  # default to _LEVEL_DENOM for magnitude (currently 10)
-    num_layers = 2  # default to 2 ops per image
-    increasing = False
-    prob = 0.5
-    config = config_str.split('-')
-    assert config[0] == 'rand'
-    config = config[1:]
-    for c in config:
-        if c.startswith('t'):
-            # NOTE old 'w' key was removed, 'w0' is not equivalent to 'tweights'
-            val = str(c[1:])
+    tokens = config_str.split('-')
+    assert tokens[0] == 'rand'
+    magnitude, num_layers, prob, increasing = MAX_MAG, 2, 0.5, False
+    for tok in tokens[1:]:
+        if tok.startswith('t'):
             if transforms is None:
-                transforms = val
-        else:
-            # numeric options
-            cs = re.split(r'(\d.*)', c)
-            if len(cs) < 2:
-                continue
-            key, val = cs[:2]
-            if key == 'mstd':
-                # noise param / randomization of magnitude values
-                mstd = float(val)
-                if mstd > 100:
-                    # use uniform sampling in 0 to magnitude if mstd is > 100
-                    mstd = float('inf')
-                hparams.setdefault('magnitude_std', mstd)
-            elif key == 'mmax':
-                # clip magnitude between [0, mmax] instead of default [0, _LEVEL_DENOM]
-                hparams.setdefault('magnitude_max', int(val))
-            elif key == 'inc':
-                if bool(val):
-                    increasing = True
-            elif key == 'm':
+                transforms = str(tok[1:])
+            continue
+        for key, val in _iter_config([tok]):
+            if key == 'm':
                 magnitude = int(val)
             elif key == 'n':
                 num_layers = int(val)
             elif key == 'p':
                 prob = float(val)
+            elif key == 'mstd':
+                mstd = float(val)
+                hparams.setdefault(
+                    'magnitude_std', float('inf') if mstd > 100 else mstd)
+            elif key == 'mmax':
+                hparams.setdefault('magnitude_max', int(val))
+            elif key == 'inc':
+                increasing = increasing or bool(val)
             else:
                 assert False, 'Unknown RandAugment config section'
 
@@ -786,29 +551,23 @@ def rand_augment_transform(
         transforms = rand_augment_choices(transforms, increasing=increasing)
     elif transforms is None:
         transforms = _RAND_INCREASING_TRANSFORMS if increasing else _RAND_TRANSFORMS
-
     choice_weights = None
     if isinstance(transforms, Dict):
-        transforms, choice_weights = _get_weighted_transforms(transforms)
+        transforms, choice_weights = _normalize_weights(transforms)
+    ops = rand_augment_ops(
+        magnitude=magnitude, prob=prob, hparams=hparams, transforms=transforms)
+    return RandAugment(ops, num_layers, choice_weights=choice_weights)
 
-    ra_ops = rand_augment_ops(magnitude=magnitude, prob=prob, hparams=hparams, transforms=transforms)
-    return RandAugment(ra_ops, num_layers, choice_weights=choice_weights)
 
+# ---------------------------------------------------------------------------
+# AugMix
+# ---------------------------------------------------------------------------
 
 _AUGMIX_TRANSFORMS = [
-    'AutoContrast',
-    'ColorIncreasing',  # not in paper
-    'ContrastIncreasing',  # not in paper
-    'BrightnessIncreasing',  # not in paper
-    'SharpnessIncreasing',  # not in paper
-    'Equalize',
-    'Rotate',
-    'PosterizeIncreasing',
-    'SolarizeIncreasing',
-    'ShearX',
-    'ShearY',
-    'TranslateXRel',
-    'TranslateYRel',
+    'AutoContrast', 'ColorIncreasing', 'ContrastIncreasing',
+    'BrightnessIncreasing', 'SharpnessIncreasing', 'Equalize',
+    'Rotate', 'PosterizeIncreasing', 'SolarizeIncreasing',
+    'ShearX', 'ShearY', 'TranslateXRel', 'TranslateYRel',
 ]
 
 
@@ -819,111 +578,89 @@ def augmix_ops(
 ):
     hparams = hparams or _HPARAMS_DEFAULT
     transforms = transforms or _AUGMIX_TRANSFORMS
-    return [AugmentOp(name, prob=1.0, magnitude=magnitude, hparams=hparams) for name in transforms]
+    return [
+        AugmentOp(name, prob=1.0, magnitude=magnitude, hparams=hparams)
+        for name in transforms
+    ]
 
 
 class AugMixAugment:
-    """AugMix Transform (reference `auto_augment.py:878`).
-
-    Adapted and improved from impl here: https://github.com/google-research/augmix/blob/master/imagenet.py
-    """
+    """AugMix (arxiv 1912.02781): blend `width` random op chains with
+    Dirichlet weights, then blend with the original by a Beta draw."""
 
     def __init__(self, ops, alpha=1., width=3, depth=-1, blended=False):
         self.ops = ops
         self.alpha = alpha
         self.width = width
         self.depth = depth
-        self.blended = blended  # blended mode is faster but not well tested
+        self.blended = blended  # sequential PIL blends instead of numpy accum
 
-    def _calc_blended_weights(self, ws, m):
-        ws = ws * m
-        cump = 1.
-        rws = []
-        for w in ws[::-1]:
-            alpha = w / cump
-            cump *= (1 - alpha)
-            rws.append(alpha)
-        return np.array(rws[::-1], dtype=np.float32)
+    def _chain(self, img):
+        depth = self.depth if self.depth > 0 else np.random.randint(1, 4)
+        for op in np.random.choice(self.ops, depth, replace=True):
+            img = op(img)
+        return img
 
     def _apply_blended(self, img, mixing_weights, m):
-        # This is my first crack and implementing a slightly faster mixed augmentation. Instead
-        # of accumulating the mix for each chain in a Numpy array and then blending with original,
-        # it recomputes the blending coefficients and applies one PIL image blend per chain.
-        img_orig = img.copy()
-        ws = self._calc_blended_weights(mixing_weights, m)
-        for w in ws:
-            depth = self.depth if self.depth > 0 else np.random.randint(1, 4)
-            ops = np.random.choice(self.ops, depth, replace=True)
-            img_aug = img_orig  # no ops are in-place, deep copy not necessary
-            for op in ops:
-                img_aug = op(img_aug)
-            img = Image.blend(img, img_aug, w)
+        # sequential pairwise blends equivalent to the weighted sum: chain i
+        # gets alpha_i = w_i / remaining mass
+        ws = mixing_weights * m
+        cumulative = 1.
+        alphas = []
+        for w in ws[::-1]:
+            alphas.append(w / cumulative)
+            cumulative *= 1 - alphas[-1]
+        original = img.copy()
+        for a in reversed(alphas):
+            img = Image.blend(img, self._chain(original), a)
         return img
 
     def _apply_basic(self, img, mixing_weights, m):
-        # This is a literal adaptation of the paper/official implementation without normalizations and
-        # PIL <-> Numpy conversions between every op. It is still quite CPU compute heavy compared to the
-        # typical augmentation transforms, could use a GPU / Kornia implementation.
-        img_shape = img.size[0], img.size[1], len(img.getbands())
-        mixed = np.zeros(img_shape, dtype=np.float32)
-        for mw in mixing_weights:
-            depth = self.depth if self.depth > 0 else np.random.randint(1, 4)
-            ops = np.random.choice(self.ops, depth, replace=True)
-            img_aug = img  # no ops are in-place, deep copy not necessary
-            for op in ops:
-                img_aug = op(img_aug)
-            mixed += mw * np.asarray(img_aug, dtype=np.float32)
-        np.clip(mixed, 0, 255., out=mixed)
-        mixed = Image.fromarray(mixed.astype(np.uint8))
-        return Image.blend(img, mixed, m)
+        # note: numpy image arrays are (H, W, C); the reference builds the
+        # accumulator (W, H, C) and crashes on non-square inputs
+        shape = img.size[1], img.size[0], len(img.getbands())
+        accum = np.zeros(shape, dtype=np.float32)
+        for w in mixing_weights:
+            accum += w * np.asarray(self._chain(img), dtype=np.float32)
+        np.clip(accum, 0, 255., out=accum)
+        return Image.blend(img, Image.fromarray(accum.astype(np.uint8)), m)
 
     def __call__(self, img):
         mixing_weights = np.float32(np.random.dirichlet([self.alpha] * self.width))
         m = np.float32(np.random.beta(self.alpha, self.alpha))
-        if self.blended:
-            mixed = self._apply_blended(img, mixing_weights, m)
-        else:
-            mixed = self._apply_basic(img, mixing_weights, m)
-        return mixed
+        apply = self._apply_blended if self.blended else self._apply_basic
+        return apply(img, mixing_weights, m)
 
     def __repr__(self):
-        fs = self.__class__.__name__ + f'(alpha={self.alpha}, w={self.width}, d={self.depth}, ops='
-        for op in self.ops:
-            fs += f'\n\t{op}'
-        fs += ')'
-        return fs
+        ops = ''.join(f'\n\t{op}' for op in self.ops)
+        return (f'{self.__class__.__name__}(alpha={self.alpha}, w={self.width}, '
+                f'd={self.depth}, ops={ops})')
 
 
 def augment_and_mix_transform(config_str: str, hparams: Optional[Dict] = None):
-    """Create AugMix transform from a config string, e.g. 'augmix-m5-w4-d2'."""
-    magnitude = 3
-    width = 3
-    depth = -1
-    alpha = 1.
-    blended = False
-    config = config_str.split('-')
-    assert config[0] == 'augmix'
-    config = config[1:]
-    for c in config:
-        cs = re.split(r'(\d.*)', c)
-        if len(cs) < 2:
-            continue
-        key, val = cs[:2]
+    """Build AugMix from e.g. 'augmix-m5-w4-d2'.
+
+    Keys: m magnitude · w width · d depth · a alpha · b blended-mode ·
+    mstd magnitude noise.  Magnitude sampling defaults to uniform [0, m].
+    """
+    tokens = config_str.split('-')
+    assert tokens[0] == 'augmix'
+    params = dict(magnitude=3, width=3, depth=-1, alpha=1., blended=False)
+    for key, val in _iter_config(tokens[1:]):
         if key == 'mstd':
-            # noise param injected via hparams for now
             hparams.setdefault('magnitude_std', float(val))
         elif key == 'm':
-            magnitude = int(val)
+            params['magnitude'] = int(val)
         elif key == 'w':
-            width = int(val)
+            params['width'] = int(val)
         elif key == 'd':
-            depth = int(val)
+            params['depth'] = int(val)
         elif key == 'a':
-            alpha = float(val)
+            params['alpha'] = float(val)
         elif key == 'b':
-            blended = bool(val)
+            params['blended'] = bool(val)
         else:
             assert False, 'Unknown AugMix config section'
-    hparams.setdefault('magnitude_std', float('inf'))  # default to uniform sampling (if not set via mstd arg)
-    ops = augmix_ops(magnitude=magnitude, hparams=hparams)
-    return AugMixAugment(ops, alpha=alpha, width=width, depth=depth, blended=blended)
+    hparams.setdefault('magnitude_std', float('inf'))
+    ops = augmix_ops(magnitude=params.pop('magnitude'), hparams=hparams)
+    return AugMixAugment(ops, **params)
