@@ -1,66 +1,95 @@
-"""Dataset factory (reference `timm/data/dataset_factory.py:63`)."""
+"""Dataset factory: dispatch ``<type>/<name>`` specs to dataset backends.
+
+Behavioral parity: /root/reference/timm/data/dataset_factory.py:63
+(same spec grammar and split-name normalization).  Structure here is
+table-driven: torchvision builtins and streaming reader prefixes each map
+through small builder helpers instead of one long if/elif chain.
+"""
 import os
 from typing import Optional
 
+from .dataset import ImageDataset, IterableImageDataset
+
 try:
-    from torchvision.datasets import CIFAR100, CIFAR10, MNIST, KMNIST, FashionMNIST, ImageFolder
+    import torchvision.datasets as tvd
     has_torchvision = True
 except ImportError:
-    # torchvision is optional; torch/* dataset names require it, folder/tar/hf readers do not
+    tvd = None
     has_torchvision = False
-    CIFAR100 = CIFAR10 = MNIST = KMNIST = FashionMNIST = ImageFolder = None
-try:
-    from torchvision.datasets import Places365
-    has_places365 = True
-except ImportError:
-    has_places365 = False
-try:
-    from torchvision.datasets import INaturalist
-    has_inaturalist = True
-except ImportError:
-    has_inaturalist = False
-try:
-    from torchvision.datasets import QMNIST
-    has_qmnist = True
-except ImportError:
-    has_qmnist = False
-try:
-    from torchvision.datasets import ImageNet
-    has_imagenet = True
-except ImportError:
-    has_imagenet = False
 
-from .dataset import IterableImageDataset, ImageDataset
+_TRAIN_SYNONYM = ('train', 'training')
+_EVAL_SYNONYM = ('val', 'valid', 'validation', 'eval', 'evaluation')
 
-_TORCH_BASIC_DS = dict(
-    cifar10=CIFAR10,
-    cifar100=CIFAR100,
-    mnist=MNIST,
-    kmnist=KMNIST,
-    fashion_mnist=FashionMNIST,
+# torchvision datasets taking a train=bool flag
+_TV_TRAIN_FLAG = ('cifar10', 'cifar100', 'mnist', 'kmnist', 'fashion_mnist', 'qmnist')
+_TV_CLASSES = dict(
+    cifar10='CIFAR10', cifar100='CIFAR100', mnist='MNIST', kmnist='KMNIST',
+    fashion_mnist='FashionMNIST', qmnist='QMNIST', places365='Places365',
+    imagenet='ImageNet', inaturalist='INaturalist', inat='INaturalist',
+    image_folder='ImageFolder', folder='ImageFolder',
 )
-_TRAIN_SYNONYM = dict(train=None, training=None)
-_EVAL_SYNONYM = dict(val=None, valid=None, validation=None, eval=None, evaluation=None)
 
 
 def _search_split(root, split):
-    # look for sub-folder with name of split in root and use that if it exists
-    split_name = split.split('[')[0]
-    try_root = os.path.join(root, split_name)
-    if os.path.exists(try_root):
-        return try_root
-
-    def _try(syn):
-        for s in syn:
-            try_root = os.path.join(root, s)
-            if os.path.exists(try_root):
-                return try_root
-        return root
-    if split_name in _TRAIN_SYNONYM:
-        root = _try(_TRAIN_SYNONYM)
-    elif split_name in _EVAL_SYNONYM:
-        root = _try(_EVAL_SYNONYM)
+    """Prefer a split-named (or synonym-named) subdir of root when present."""
+    base = split.split('[')[0]
+    direct = os.path.join(root, base)
+    if os.path.exists(direct):
+        return direct
+    synonyms = ()
+    if base in _TRAIN_SYNONYM:
+        synonyms = _TRAIN_SYNONYM
+    elif base in _EVAL_SYNONYM:
+        synonyms = _EVAL_SYNONYM
+    for name in synonyms:
+        candidate = os.path.join(root, name)
+        if os.path.exists(candidate):
+            return candidate
     return root
+
+
+def _tv_class(name):
+    assert has_torchvision, 'torchvision is required for torch/* dataset names'
+    cls = getattr(tvd, _TV_CLASSES.get(name, ''), None)
+    assert cls is not None, (
+        f'Unknown torchvision dataset {name}' if name not in _TV_CLASSES else
+        f'Please update torchvision for the {name} dataset.')
+    return cls
+
+
+def _build_torchvision(name, split, root, download, search_split, kwargs):
+    common = dict(root=root, download=download, **kwargs)
+    if name in _TV_TRAIN_FLAG:
+        return _tv_class(name)(train=split in _TRAIN_SYNONYM, **common)
+    if name in ('inaturalist', 'inat'):
+        # split may carry a target-type prefix: '<type>/<version>'
+        target_type = 'full'
+        parts = split.split('/')
+        if len(parts) > 1:
+            target_type = parts[0].split('_')
+            if len(target_type) == 1:
+                target_type = target_type[0]
+            split = parts[-1]
+        if split in _TRAIN_SYNONYM:
+            split = '2021_train'
+        elif split in _EVAL_SYNONYM:
+            split = '2021_valid'
+        return _tv_class(name)(version=split, target_type=target_type, **common)
+    if name == 'places365':
+        if split in _TRAIN_SYNONYM:
+            split = 'train-standard'
+        elif split in _EVAL_SYNONYM:
+            split = 'val'
+        return _tv_class(name)(split=split, **common)
+    if name == 'imagenet':
+        if split in _EVAL_SYNONYM:
+            split = 'val'
+        return _tv_class(name)(split=split, **common)
+    if name in ('image_folder', 'folder'):
+        if search_split and os.path.isdir(root):
+            root = _search_split(root, split)
+        return _tv_class(name)(root, **kwargs)
+    return _tv_class(name)  # raises 'Unknown torchvision dataset'
 
 
 def create_dataset(
@@ -79,125 +108,43 @@ def create_dataset(
         input_img_mode: str = 'RGB',
         **kwargs,
 ):
-    """Dataset factory method (reference `dataset_factory.py:63`).
+    """Build a dataset from a ``<type>/<name>`` spec.
 
-    Dispatches `<type>/<name>` to torch builtin / folder / HFDS / HFIDS /
-    TFDS / WDS readers, with auto split-dir search for folder datasets.
+    Types: ``torch/*`` torchvision builtins · ``hfds/*`` HF arrow (map-style)
+    · ``hfids/*`` / ``tfds/*`` / ``wds/*`` streaming readers · anything else =
+    folder/tar path handled by ImageDataset (with split-subdir search).
     """
     kwargs = {k: v for k, v in kwargs.items() if v is not None}
-    name = name or ''
-    name = name.lower()
+    name = (name or '').lower()
+
     if name.startswith('torch/'):
-        assert has_torchvision, 'torchvision is required for torch/* dataset names'
-        name = name.split('/', 2)[-1]
-        torch_kwargs = dict(root=root, download=download, **kwargs)
-        if name in _TORCH_BASIC_DS:
-            ds_class = _TORCH_BASIC_DS[name]
-            use_train = split in _TRAIN_SYNONYM
-            ds = ds_class(train=use_train, **torch_kwargs)
-        elif name == 'inaturalist' or name == 'inat':
-            assert has_inaturalist, 'Please update to PyTorch 1.10, torchvision 0.11+ for Inaturalist'
-            target_type = 'full'
-            split_split = split.split('/')
-            if len(split_split) > 1:
-                target_type = split_split[0].split('_')
-                if len(target_type) == 1:
-                    target_type = target_type[0]
-                split = split_split[-1]
-            if split in _TRAIN_SYNONYM:
-                split = '2021_train'
-            elif split in _EVAL_SYNONYM:
-                split = '2021_valid'
-            ds = INaturalist(version=split, target_type=target_type, **torch_kwargs)
-        elif name == 'places365':
-            assert has_places365, 'Please update to a newer PyTorch and torchvision for Places365 dataset.'
-            if split in _TRAIN_SYNONYM:
-                split = 'train-standard'
-            elif split in _EVAL_SYNONYM:
-                split = 'val'
-            ds = Places365(split=split, **torch_kwargs)
-        elif name == 'qmnist':
-            assert has_qmnist, 'Please update to a newer PyTorch and torchvision for QMNIST dataset.'
-            use_train = split in _TRAIN_SYNONYM
-            ds = QMNIST(train=use_train, **torch_kwargs)
-        elif name == 'imagenet':
-            assert has_imagenet, 'Please update to a newer PyTorch and torchvision for ImageNet dataset.'
-            if split in _EVAL_SYNONYM:
-                split = 'val'
-            ds = ImageNet(split=split, **torch_kwargs)
-        elif name == 'image_folder' or name == 'folder':
-            # in case torchvision ImageFolder is preferred over timm ImageDataset for some reason
-            if search_split and os.path.isdir(root):
-                # look for split specific sub-folder in root
-                root = _search_split(root, split)
-            ds = ImageFolder(root, **kwargs)
-        else:
-            assert False, f"Unknown torchvision dataset {name}"
-    elif name.startswith('hfds/'):
-        # NOTE right now, HF datasets default arrow format is a random-access Dataset,
-        # There will be a IterableDataset variant too, TBD
-        ds = ImageDataset(
-            root,
-            reader=name,
-            split=split,
-            class_map=class_map,
-            input_img_mode=input_img_mode,
-            **kwargs,
-        )
-    elif name.startswith('hfids/'):
-        ds = IterableImageDataset(
-            root,
-            reader=name,
-            split=split,
-            class_map=class_map,
-            is_training=is_training,
-            batch_size=batch_size,
-            num_samples=num_samples,
-            repeats=repeats,
-            seed=seed,
-            input_img_mode=input_img_mode,
-            **kwargs,
-        )
-    elif name.startswith('tfds/'):
-        ds = IterableImageDataset(
-            root,
-            reader=name,
-            split=split,
-            class_map=class_map,
-            is_training=is_training,
-            download=download,
-            batch_size=batch_size,
-            num_samples=num_samples,
-            repeats=repeats,
-            seed=seed,
-            input_img_mode=input_img_mode,
-            **kwargs,
-        )
-    elif name.startswith('wds/'):
-        ds = IterableImageDataset(
-            root,
-            reader=name,
-            split=split,
-            class_map=class_map,
-            is_training=is_training,
-            batch_size=batch_size,
-            num_samples=num_samples,
-            repeats=repeats,
-            seed=seed,
-            input_img_mode=input_img_mode,
-            **kwargs,
-        )
-    else:
-        # FIXME support more advance split cfg for ImageFolder/Tar datasets in the future
-        if search_split and os.path.isdir(root):
-            # look for split specific sub-folder in root
-            root = _search_split(root, split)
-        ds = ImageDataset(
-            root,
-            reader=name,
-            class_map=class_map,
-            load_bytes=load_bytes,
-            input_img_mode=input_img_mode,
-            **kwargs,
-        )
-    return ds
+        return _build_torchvision(
+            name.split('/', 2)[-1], split, root, download, search_split, kwargs)
+
+    stream_args = dict(
+        split=split,
+        class_map=class_map,
+        is_training=is_training,
+        batch_size=batch_size,
+        num_samples=num_samples,
+        repeats=repeats,
+        seed=seed,
+        input_img_mode=input_img_mode,
+    )
+    if name.startswith('hfds/'):
+        # HF arrow datasets are random-access; the reader handles decode
+        return ImageDataset(
+            root, reader=name, split=split, class_map=class_map,
+            input_img_mode=input_img_mode, **kwargs)
+    if name.startswith('hfids/') or name.startswith('wds/'):
+        return IterableImageDataset(root, reader=name, **stream_args, **kwargs)
+    if name.startswith('tfds/'):
+        return IterableImageDataset(
+            root, reader=name, download=download, **stream_args, **kwargs)
+
+    # plain folder / tar path
+    if search_split and os.path.isdir(root):
+        root = _search_split(root, split)
+    return ImageDataset(
+        root, reader=name, class_map=class_map, load_bytes=load_bytes,
+        input_img_mode=input_img_mode, **kwargs)
