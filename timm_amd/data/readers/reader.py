@@ -1,17 +1,23 @@
-"""Reader base class (reference `timm/data/readers/reader.py`)."""
+"""Reader protocol: indexable/iterable sample sources feeding the datasets.
+
+Behavioral parity: /root/reference/timm/data/readers/reader.py.
+"""
 from abc import abstractmethod
 
 
 class Reader:
-    def __init__(self):
-        pass
+    """Base sample source.  Subclasses yield (file-like, target) pairs and
+    resolve sample filenames for bookkeeping/CSV output."""
 
     @abstractmethod
     def _filename(self, index, basename=False, absolute=False):
-        pass
+        ...
 
     def filename(self, index, basename=False, absolute=False):
         return self._filename(index, basename=basename, absolute=absolute)
 
     def filenames(self, basename=False, absolute=False):
-        return [self._filename(index, basename=basename, absolute=absolute) for index in range(len(self))]
+        return [
+            self._filename(i, basename=basename, absolute=absolute)
+            for i in range(len(self))
+        ]

@@ -1,4 +1,8 @@
-"""Folder-of-images reader (reference `timm/data/readers/reader_image_folder.py`)."""
+"""Folder-tree image reader: class labels from directory names.
+
+Behavioral parity: /root/reference/timm/data/readers/reader_image_folder.py
+(same walk order, natural-key sorting, class-index assignment).
+"""
 import os
 from typing import Dict, List, Optional, Set, Tuple, Union
 
@@ -14,64 +18,46 @@ def find_images_and_targets(
         types: Optional[Union[List, Tuple, Set]] = None,
         class_to_idx: Optional[Dict] = None,
         leaf_name_only: bool = True,
-        sort: bool = True
+        sort: bool = True,
 ):
-    """Walk folder recursively to discover images and map them to classes by folder names.
+    """Recursively collect (path, class-index) pairs under ``folder``.
 
-    Args:
-        folder: root of folder to recursively search
-        types: types (file extensions) to search for in path
-        class_to_idx: specify mapping for class (folder name) to class index if set
-        leaf_name_only: use only leaf-name of folder walk for class names
-        sort: re-sort found images by name (for consistent ordering)
-
-    Returns:
-        A list of image and target tuples, class_to_idx mapping
+    Class names come from the containing directory (leaf name, or the full
+    relative path with separators replaced).  Unknown classes (not in a given
+    class_to_idx) are dropped.  Returns (samples, class_to_idx).
     """
-    types = get_img_extensions(as_set=True) if not types else set(types)
-    labels = []
-    filenames = []
-    for root, subdirs, files in os.walk(folder, topdown=False, followlinks=True):
-        rel_path = os.path.relpath(root, folder) if (root != folder) else ''
-        label = os.path.basename(rel_path) if leaf_name_only else rel_path.replace(os.path.sep, '_')
-        for f in files:
-            base, ext = os.path.splitext(f)
-            if ext.lower() in types:
-                filenames.append(os.path.join(root, f))
-                labels.append(label)
+    exts = set(types) if types else get_img_extensions(as_set=True)
+    found: List[Tuple[str, str]] = []  # (path, label)
+    for dirpath, _dirs, files in os.walk(folder, topdown=False, followlinks=True):
+        if dirpath == folder:
+            label = ''
+        else:
+            rel = os.path.relpath(dirpath, folder)
+            label = os.path.basename(rel) if leaf_name_only else rel.replace(os.path.sep, '_')
+        found.extend(
+            (os.path.join(dirpath, f), label)
+            for f in files if os.path.splitext(f)[1].lower() in exts
+        )
     if class_to_idx is None:
-        # building class index
-        unique_labels = set(labels)
-        sorted_labels = list(sorted(unique_labels, key=natural_key))
-        class_to_idx = {c: idx for idx, c in enumerate(sorted_labels)}
-    images_and_targets = [(f, class_to_idx[l]) for f, l in zip(filenames, labels) if l in class_to_idx]
+        names = sorted({label for _, label in found}, key=natural_key)
+        class_to_idx = {name: idx for idx, name in enumerate(names)}
+    samples = [
+        (path, class_to_idx[label]) for path, label in found if label in class_to_idx]
     if sort:
-        images_and_targets = sorted(images_and_targets, key=lambda k: natural_key(k[0]))
-    return images_and_targets, class_to_idx
+        samples.sort(key=lambda s: natural_key(s[0]))
+    return samples, class_to_idx
 
 
 class ReaderImageFolder(Reader):
-    def __init__(
-            self,
-            root,
-            class_map='',
-            input_key=None,
-    ):
+    def __init__(self, root, class_map='', input_key=None):
         super().__init__()
-
         self.root = root
-        class_to_idx = None
-        if class_map:
-            class_to_idx = load_class_map(class_map, root)
-        find_types = None
-        if input_key:
-            find_types = input_key.split(';')
         self.samples, self.class_to_idx = find_images_and_targets(
             root,
-            class_to_idx=class_to_idx,
-            types=find_types,
+            class_to_idx=load_class_map(class_map, root) if class_map else None,
+            types=input_key.split(';') if input_key else None,
         )
-        if len(self.samples) == 0:
+        if not self.samples:
             raise RuntimeError(
                 f'Found 0 images in subfolders of {root}. '
                 f'Supported image extensions are {", ".join(get_img_extensions())}')
@@ -84,9 +70,9 @@ class ReaderImageFolder(Reader):
         return len(self.samples)
 
     def _filename(self, index, basename=False, absolute=False):
-        filename = self.samples[index][0]
+        path = self.samples[index][0]
         if basename:
-            filename = os.path.basename(filename)
-        elif not absolute:
-            filename = os.path.relpath(filename, self.root)
-        return filename
+            return os.path.basename(path)
+        if not absolute:
+            return os.path.relpath(path, self.root)
+        return path
