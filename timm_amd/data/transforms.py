@@ -1,19 +1,25 @@
-"""Geometry transforms (reference `timm/data/transforms.py`):
-`RandomResizedCropAndInterpolation` (:166), `CenterCropOrPad` (:314),
-`RandomCropOrPad` (:379), `ResizeKeepRatio` (:448), `TrimBorder` (:567),
-`ToNumpy`, str<->InterpolationMode helpers."""
+"""Geometry transforms for the input pipeline.
+
+Behavioral parity: /root/reference/timm/data/transforms.py
+(RandomResizedCropAndInterpolation :166, CenterCropOrPad :314,
+RandomCropOrPad :379, ResizeKeepRatio :448, TrimBorder :567, tensor/numpy
+converters, str<->interpolation maps).  All image math routes through the
+in-repo torchvision-free functional layer (`image_ops`).
+"""
 import math
 import numbers
 import random
 import warnings
-from typing import List, Optional, Sequence, Tuple, Union
+from typing import List, Sequence, Tuple, Union
 
+import numpy as np
 import torch
+from PIL import Image
+
 from . import image_ops as F
 from .image_ops import InterpolationMode
+
 has_interpolation_mode = True
-from PIL import Image
-import numpy as np
 
 __all__ = [
     "ToNumpy", "ToTensor", "str_to_interp_mode", "str_to_pil_interp", "interp_mode_to_str",
@@ -22,18 +28,23 @@ __all__ = [
 ]
 
 
+# ---------------------------------------------------------------------------
+# converters
+# ---------------------------------------------------------------------------
+
 class ToNumpy:
+    """PIL -> CHW uint8 ndarray (no scaling); used by the fast-collate path."""
 
     def __call__(self, pil_img):
-        np_img = np.array(pil_img, dtype=np.uint8)
-        if np_img.ndim < 3:
-            np_img = np.expand_dims(np_img, axis=-1)
-        np_img = np.rollaxis(np_img, 2)  # HWC to CHW
-        return np_img
+        arr = np.array(pil_img, dtype=np.uint8)
+        if arr.ndim < 3:
+            arr = arr[:, :, None]
+        return np.rollaxis(arr, 2)
 
 
 class ToTensor:
-    """ToTensor with no rescaling of values"""
+    """PIL -> tensor of the requested dtype, values NOT rescaled to [0,1]."""
+
     def __init__(self, dtype=torch.float32):
         self.dtype = dtype
 
@@ -42,73 +53,43 @@ class ToTensor:
 
 
 class MaybeToTensor(torch.nn.Module):
-    """Convert a PIL Image or ndarray to tensor if it's not already one."""
-
-    def __init__(self) -> None:
-        super().__init__()
+    """to_tensor (with [0,1] scaling) unless the input is already a tensor."""
 
     def forward(self, pic) -> torch.Tensor:
-        if isinstance(pic, torch.Tensor):
-            return pic
-        return F.to_tensor(pic)
+        return pic if isinstance(pic, torch.Tensor) else F.to_tensor(pic)
 
     def __repr__(self) -> str:
         return f"{self.__class__.__name__}()"
 
 
 class MaybePILToTensor(torch.nn.Module):
-    """Convert a PIL Image to a tensor of the same type - this does not scale values."""
-
-    def __init__(self) -> None:
-        super().__init__()
+    """pil_to_tensor (no value scaling) unless the input is already a tensor."""
 
     def forward(self, pic):
-        if isinstance(pic, torch.Tensor):
-            return pic
-        return F.pil_to_tensor(pic)
+        return pic if isinstance(pic, torch.Tensor) else F.pil_to_tensor(pic)
 
     def __repr__(self) -> str:
         return f"{self.__class__.__name__}()"
 
 
-# Pillow is deprecating the top-level resampling attributes (e.g., Image.BILINEAR) in
-# favor of the Image.Resampling enum. The top-level resampling attributes will be
-# removed in Pillow 10.
-if hasattr(Image, "Resampling"):
-    _pil_interpolation_to_str = {
-        Image.Resampling.NEAREST: 'nearest',
-        Image.Resampling.BILINEAR: 'bilinear',
-        Image.Resampling.BICUBIC: 'bicubic',
-        Image.Resampling.BOX: 'box',
-        Image.Resampling.HAMMING: 'hamming',
-        Image.Resampling.LANCZOS: 'lanczos',
-    }
-else:
-    _pil_interpolation_to_str = {
-        Image.NEAREST: 'nearest',
-        Image.BILINEAR: 'bilinear',
-        Image.BICUBIC: 'bicubic',
-        Image.BOX: 'box',
-        Image.HAMMING: 'hamming',
-        Image.LANCZOS: 'lanczos',
-    }
+# ---------------------------------------------------------------------------
+# interpolation-mode naming (one triple table -> all direction maps)
+# ---------------------------------------------------------------------------
 
-_str_to_pil_interpolation = {b: a for a, b in _pil_interpolation_to_str.items()}
-
-
-if has_interpolation_mode:
-    _torch_interpolation_to_str = {
-        InterpolationMode.NEAREST: 'nearest',
-        InterpolationMode.BILINEAR: 'bilinear',
-        InterpolationMode.BICUBIC: 'bicubic',
-        InterpolationMode.BOX: 'box',
-        InterpolationMode.HAMMING: 'hamming',
-        InterpolationMode.LANCZOS: 'lanczos',
-    }
-    _str_to_torch_interpolation = {b: a for a, b in _torch_interpolation_to_str.items()}
-else:
-    _pil_interpolation_to_torch = {}
-    _torch_interpolation_to_str = {}
+_PIL_MODES = Image.Resampling if hasattr(Image, 'Resampling') else Image
+_INTERP_TRIPLES = [
+    # (name, PIL resample, torch InterpolationMode)
+    ('nearest', _PIL_MODES.NEAREST, InterpolationMode.NEAREST),
+    ('bilinear', _PIL_MODES.BILINEAR, InterpolationMode.BILINEAR),
+    ('bicubic', _PIL_MODES.BICUBIC, InterpolationMode.BICUBIC),
+    ('box', _PIL_MODES.BOX, InterpolationMode.BOX),
+    ('hamming', _PIL_MODES.HAMMING, InterpolationMode.HAMMING),
+    ('lanczos', _PIL_MODES.LANCZOS, InterpolationMode.LANCZOS),
+]
+_pil_interpolation_to_str = {pil: name for name, pil, _t in _INTERP_TRIPLES}
+_str_to_pil_interpolation = {name: pil for name, pil, _t in _INTERP_TRIPLES}
+_torch_interpolation_to_str = {t: name for name, _pil, t in _INTERP_TRIPLES}
+_str_to_torch_interpolation = {name: t for name, _pil, t in _INTERP_TRIPLES}
 
 
 def str_to_pil_interp(mode_str):
@@ -116,42 +97,52 @@ def str_to_pil_interp(mode_str):
 
 
 def str_to_interp_mode(mode_str):
-    if has_interpolation_mode:
-        return _str_to_torch_interpolation[mode_str]
-    else:
-        return _str_to_pil_interpolation[mode_str]
+    return _str_to_torch_interpolation[mode_str]
 
 
 def interp_mode_to_str(mode):
-    if has_interpolation_mode:
-        return _torch_interpolation_to_str[mode]
-    else:
-        return _pil_interpolation_to_str[mode]
+    return _torch_interpolation_to_str[mode]
 
 
 _RANDOM_INTERPOLATION = (str_to_interp_mode('bilinear'), str_to_interp_mode('bicubic'))
 
 
+def _resolve_interp(interpolation):
+    """'random' -> the (bilinear, bicubic) pair; else the named mode."""
+    if interpolation == 'random':
+        return _RANDOM_INTERPOLATION
+    return str_to_interp_mode(interpolation)
+
+
+def _pick_interp(interpolation):
+    if isinstance(interpolation, (tuple, list)):
+        return random.choice(interpolation)
+    return interpolation
+
+
+def _interp_repr(interpolation):
+    if isinstance(interpolation, (tuple, list)):
+        return ' '.join(interp_mode_to_str(m) for m in interpolation)
+    return interp_mode_to_str(interpolation)
+
+
 def _setup_size(size, error_msg="Please provide only two dimensions (h, w) for size."):
     if isinstance(size, numbers.Number):
         return int(size), int(size)
-
     if isinstance(size, Sequence) and len(size) == 1:
         return size[0], size[0]
-
     if len(size) != 2:
         raise ValueError(error_msg)
-
     return size
 
 
-class RandomResizedCropAndInterpolation:
-    """Crop the given PIL Image to random size and aspect ratio with random interpolation.
+# ---------------------------------------------------------------------------
+# crops
+# ---------------------------------------------------------------------------
 
-    A crop of random size (default: of 0.08 to 1.0) of the original size and a random
-    aspect ratio (default: of 3/4 to 4/3) of the original aspect ratio is made. This crop
-    is finally resized to given size.  This is popularly used to train Inception networks.
-    """
+class RandomResizedCropAndInterpolation:
+    """Inception-style random-area/aspect crop resized to a fixed size, with
+    per-call random interpolation when requested."""
 
     def __init__(
             self,
@@ -160,71 +151,52 @@ class RandomResizedCropAndInterpolation:
             ratio=(3. / 4., 4. / 3.),
             interpolation='bilinear',
     ):
-        if isinstance(size, (list, tuple)):
-            self.size = tuple(size)
-        else:
-            self.size = (size, size)
-        if (scale[0] > scale[1]) or (ratio[0] > ratio[1]):
+        self.size = tuple(size) if isinstance(size, (list, tuple)) else (size, size)
+        if scale[0] > scale[1] or ratio[0] > ratio[1]:
             warnings.warn("range should be of kind (min, max)")
-
-        if interpolation == 'random':
-            self.interpolation = _RANDOM_INTERPOLATION
-        else:
-            self.interpolation = str_to_interp_mode(interpolation)
+        self.interpolation = _resolve_interp(interpolation)
         self.scale = scale
         self.ratio = ratio
 
     @staticmethod
     def get_params(img, scale, ratio):
-        """Get parameters for ``crop`` for a random sized crop."""
+        """Sample a (top, left, h, w) crop box; central-crop fallback after
+        10 rejected draws."""
         img_w, img_h = F.get_image_size(img)
         area = img_w * img_h
+        log_ratio = (math.log(ratio[0]), math.log(ratio[1]))
 
-        for attempt in range(10):
+        for _ in range(10):
             target_area = random.uniform(*scale) * area
-            log_ratio = (math.log(ratio[0]), math.log(ratio[1]))
-            aspect_ratio = math.exp(random.uniform(*log_ratio))
+            aspect = math.exp(random.uniform(*log_ratio))
+            w = int(round(math.sqrt(target_area * aspect)))
+            h = int(round(math.sqrt(target_area / aspect)))
+            if w <= img_w and h <= img_h:
+                top = random.randint(0, img_h - h)
+                left = random.randint(0, img_w - w)
+                return top, left, h, w
 
-            target_w = int(round(math.sqrt(target_area * aspect_ratio)))
-            target_h = int(round(math.sqrt(target_area / aspect_ratio)))
-            if target_w <= img_w and target_h <= img_h:
-                i = random.randint(0, img_h - target_h)
-                j = random.randint(0, img_w - target_w)
-                return i, j, target_h, target_w
-
-        # Fallback to central crop
+        # fallback: largest central crop within the aspect limits
         in_ratio = img_w / img_h
         if in_ratio < min(ratio):
-            target_w = img_w
-            target_h = int(round(target_w / min(ratio)))
+            w = img_w
+            h = int(round(w / min(ratio)))
         elif in_ratio > max(ratio):
-            target_h = img_h
-            target_w = int(round(target_h * max(ratio)))
-        else:  # whole image
-            target_w = img_w
-            target_h = img_h
-        i = (img_h - target_h) // 2
-        j = (img_w - target_w) // 2
-        return i, j, target_h, target_w
+            h = img_h
+            w = int(round(h * max(ratio)))
+        else:
+            w, h = img_w, img_h
+        return (img_h - h) // 2, (img_w - w) // 2, h, w
 
     def __call__(self, img):
-        i, j, h, w = self.get_params(img, self.scale, self.ratio)
-        if isinstance(self.interpolation, (tuple, list)):
-            interpolation = random.choice(self.interpolation)
-        else:
-            interpolation = self.interpolation
-        return F.resized_crop(img, i, j, h, w, self.size, interpolation)
+        top, left, h, w = self.get_params(img, self.scale, self.ratio)
+        return F.resized_crop(img, top, left, h, w, self.size, _pick_interp(self.interpolation))
 
     def __repr__(self):
-        if isinstance(self.interpolation, (tuple, list)):
-            interpolate_str = ' '.join([interp_mode_to_str(x) for x in self.interpolation])
-        else:
-            interpolate_str = interp_mode_to_str(self.interpolation)
-        format_string = self.__class__.__name__ + '(size={0}'.format(self.size)
-        format_string += ', scale={0}'.format(tuple(round(s, 4) for s in self.scale))
-        format_string += ', ratio={0}'.format(tuple(round(r, 4) for r in self.ratio))
-        format_string += ', interpolation={0})'.format(interpolate_str)
-        return format_string
+        return (f'{self.__class__.__name__}(size={self.size}'
+                f', scale={tuple(round(s, 4) for s in self.scale)}'
+                f', ratio={tuple(round(r, 4) for r in self.ratio)}'
+                f', interpolation={_interp_repr(self.interpolation)})')
 
 
 def center_crop_or_pad(
@@ -233,30 +205,28 @@ def center_crop_or_pad(
         fill: Union[int, Tuple[int, int, int]] = 0,
         padding_mode: str = 'constant',
 ) -> torch.Tensor:
-    """Center crops and/or pads the given image."""
-    output_size = _setup_size(output_size)
-    crop_height, crop_width = output_size
-    _, image_height, image_width = F.get_dimensions(img)
+    """Center crop, symmetrically padding first when the target is larger."""
+    crop_h, crop_w = _setup_size(output_size)
+    _, img_h, img_w = F.get_dimensions(img)
 
-    if crop_width > image_width or crop_height > image_height:
-        padding_ltrb = [
-            (crop_width - image_width) // 2 if crop_width > image_width else 0,
-            (crop_height - image_height) // 2 if crop_height > image_height else 0,
-            (crop_width - image_width + 1) // 2 if crop_width > image_width else 0,
-            (crop_height - image_height + 1) // 2 if crop_height > image_height else 0,
-        ]
-        img = F.pad(img, padding_ltrb, fill=fill, padding_mode=padding_mode)
-        _, image_height, image_width = F.get_dimensions(img)
-        if crop_width == image_width and crop_height == image_height:
+    if crop_w > img_w or crop_h > img_h:
+        extra_w = max(crop_w - img_w, 0)
+        extra_h = max(crop_h - img_h, 0)
+        img = F.pad(
+            img,
+            [extra_w // 2, extra_h // 2, (extra_w + 1) // 2, (extra_h + 1) // 2],
+            fill=fill, padding_mode=padding_mode)
+        _, img_h, img_w = F.get_dimensions(img)
+        if (crop_w, crop_h) == (img_w, img_h):
             return img
 
-    crop_top = int(round((image_height - crop_height) / 2.0))
-    crop_left = int(round((image_width - crop_width) / 2.0))
-    return F.crop(img, crop_top, crop_left, crop_height, crop_width)
+    top = int(round((img_h - crop_h) / 2.0))
+    left = int(round((img_w - crop_w) / 2.0))
+    return F.crop(img, top, left, crop_h, crop_w)
 
 
 class CenterCropOrPad(torch.nn.Module):
-    """Crops the given image at the center, padding if smaller than crop size."""
+    """Module wrapper over center_crop_or_pad."""
 
     def __init__(
             self,
@@ -285,26 +255,25 @@ def crop_or_pad(
         fill: Union[int, Tuple[int, int, int]] = 0,
         padding_mode: str = 'constant',
 ) -> torch.Tensor:
-    """Crops and/or pads image to meet target size, with control over fill and padding_mode."""
-    _, image_height, image_width = F.get_dimensions(img)
-    right = left + width
-    bottom = top + height
-    if left < 0 or top < 0 or right > image_width or bottom > image_height:
-        padding_ltrb = [
-            max(-left + min(0, right), 0),
-            max(-top + min(0, bottom), 0),
-            max(right - max(image_width, left), 0),
-            max(bottom - max(image_height, top), 0),
-        ]
-        img = F.pad(img, padding_ltrb, fill=fill, padding_mode=padding_mode)
-
-    top = max(top, 0)
-    left = max(left, 0)
-    return F.crop(img, top, left, height, width)
+    """Crop a (possibly out-of-bounds) box, padding whatever falls outside."""
+    _, img_h, img_w = F.get_dimensions(img)
+    right, bottom = left + width, top + height
+    if left < 0 or top < 0 or right > img_w or bottom > img_h:
+        img = F.pad(
+            img,
+            [
+                max(-left + min(0, right), 0),
+                max(-top + min(0, bottom), 0),
+                max(right - max(img_w, left), 0),
+                max(bottom - max(img_h, top), 0),
+            ],
+            fill=fill, padding_mode=padding_mode)
+    return F.crop(img, max(top, 0), max(left, 0), height, width)
 
 
 class RandomCropOrPad(torch.nn.Module):
-    """Crop and/or pad image with random placement within the crop or pad margin."""
+    """Random placement within the available crop margin (or pad margin when
+    the image is smaller than the target)."""
 
     def __init__(
             self,
@@ -319,53 +288,48 @@ class RandomCropOrPad(torch.nn.Module):
 
     @staticmethod
     def get_params(img, size):
-        _, image_height, image_width = F.get_dimensions(img)
-        delta_height = image_height - size[0]
-        delta_width = image_width - size[1]
-        top = int(math.copysign(random.randint(0, abs(delta_height)), delta_height))
-        left = int(math.copysign(random.randint(0, abs(delta_width)), delta_width))
+        _, img_h, img_w = F.get_dimensions(img)
+        dh, dw = img_h - size[0], img_w - size[1]
+        top = int(math.copysign(random.randint(0, abs(dh)), dh))
+        left = int(math.copysign(random.randint(0, abs(dw)), dw))
         return top, left
 
     def forward(self, img):
         top, left = self.get_params(img, self.size)
         return crop_or_pad(
-            img,
-            top=top,
-            left=left,
-            height=self.size[0],
-            width=self.size[1],
-            fill=self.fill,
-            padding_mode=self.padding_mode,
-        )
+            img, top=top, left=left, height=self.size[0], width=self.size[1],
+            fill=self.fill, padding_mode=self.padding_mode)
 
     def __repr__(self) -> str:
         return f"{self.__class__.__name__}(size={self.size})"
 
 
 class RandomPad:
+    """Pad up to input_size with a random left/top split of the slack."""
+
     def __init__(self, input_size, fill=0):
         self.input_size = input_size
         self.fill = fill
 
     @staticmethod
     def get_params(img, input_size):
-        width, height = F.get_image_size(img)
-        delta_width = max(input_size[1] - width, 0)
-        delta_height = max(input_size[0] - height, 0)
-        pad_left = random.randint(0, delta_width)
-        pad_top = random.randint(0, delta_height)
-        pad_right = delta_width - pad_left
-        pad_bottom = delta_height - pad_top
-        return pad_left, pad_top, pad_right, pad_bottom
+        w, h = F.get_image_size(img)
+        slack_w = max(input_size[1] - w, 0)
+        slack_h = max(input_size[0] - h, 0)
+        pad_left = random.randint(0, slack_w)
+        pad_top = random.randint(0, slack_h)
+        return pad_left, pad_top, slack_w - pad_left, slack_h - pad_top
 
     def __call__(self, img):
-        padding = self.get_params(img, self.input_size)
-        img = F.pad(img, padding, self.fill)
-        return img
+        return F.pad(img, self.get_params(img, self.input_size), self.fill)
 
 
 class ResizeKeepRatio:
-    """Resize while keeping ratio. Result is not guaranteed to match target size."""
+    """Aspect-preserving resize toward a target box.
+
+    ``longest`` interpolates between fit-shortest (0, may overshoot target)
+    and fit-longest (1, always inside target); optional random scale/aspect
+    jitter for train-time use."""
 
     def __init__(
             self,
@@ -378,15 +342,8 @@ class ResizeKeepRatio:
             random_aspect_prob=0.,
             random_aspect_range=(0.9, 1.11),
     ):
-        """"""
-        if isinstance(size, (list, tuple)):
-            self.size = tuple(size)
-        else:
-            self.size = (size, size)
-        if interpolation == 'random':
-            self.interpolation = _RANDOM_INTERPOLATION
-        else:
-            self.interpolation = str_to_interp_mode(interpolation)
+        self.size = tuple(size) if isinstance(size, (list, tuple)) else (size, size)
+        self.interpolation = _resolve_interp(interpolation)
         self.longest = float(longest)
         self.random_scale_prob = random_scale_prob
         self.random_scale_range = random_scale_range
@@ -405,72 +362,55 @@ class ResizeKeepRatio:
             random_aspect_prob=0.,
             random_aspect_range=(0.9, 1.11),
     ):
-        """Get parameters."""
-        img_w, img_h = img_size = F.get_image_size(img)
+        """Compute the output (h, w)."""
+        img_w, img_h = F.get_image_size(img)
         target_h, target_w = target_size
-        ratio_h = img_h / target_h
-        ratio_w = img_w / target_w
-        ratio = max(ratio_h, ratio_w) * longest + min(ratio_h, ratio_w) * (1. - longest)
+        # blend of shortest-fit and longest-fit downscale ratios
+        rh, rw = img_h / target_h, img_w / target_w
+        ratio = max(rh, rw) * longest + min(rh, rw) * (1. - longest)
 
+        scale_w = scale_h = 1.
         if random_scale_prob > 0 and random.random() < random_scale_prob:
-            ratio_factor = random.uniform(random_scale_range[0], random_scale_range[1])
+            jitter = random.uniform(random_scale_range[0], random_scale_range[1])
             if random_scale_area:
-                # make ratio factor equivalent to RRC area crop where < 1.0 = area zoom,
-                # otherwise like affine scale where < 1.0 = linear zoom out
-                ratio_factor = 1. / math.sqrt(ratio_factor)
-            ratio_factor = (ratio_factor, ratio_factor)
-        else:
-            ratio_factor = (1., 1.)
-
+                # treat the draw as an area factor (RRC-like): <1 zooms in
+                jitter = 1. / math.sqrt(jitter)
+            scale_w = scale_h = jitter
         if random_aspect_prob > 0 and random.random() < random_aspect_prob:
-            log_aspect = (math.log(random_aspect_range[0]), math.log(random_aspect_range[1]))
-            aspect_factor = math.exp(random.uniform(*log_aspect))
-            aspect_factor = math.sqrt(aspect_factor)
-            # currently applying random aspect adjustment equally to both dims,
-            # could change to keep output sizes above their target where possible
-            ratio_factor = (ratio_factor[0] / aspect_factor, ratio_factor[1] * aspect_factor)
+            log_rng = (math.log(random_aspect_range[0]), math.log(random_aspect_range[1]))
+            aspect = math.sqrt(math.exp(random.uniform(*log_rng)))
+            # split the aspect jitter evenly across both dims
+            scale_w, scale_h = scale_w / aspect, scale_h * aspect
 
-        size = [round(x * f / ratio) for x, f in zip(img_size, ratio_factor)][::-1]
-        return size
+        return [
+            round(img_h * scale_h / ratio),
+            round(img_w * scale_w / ratio),
+        ]
 
     def __call__(self, img):
         size = self.get_params(
             img, self.size, self.longest,
             self.random_scale_prob, self.random_scale_range, self.random_scale_area,
-            self.random_aspect_prob, self.random_aspect_range
-        )
-        if isinstance(self.interpolation, (tuple, list)):
-            interpolation = random.choice(self.interpolation)
-        else:
-            interpolation = self.interpolation
-        img = F.resize(img, size, interpolation)
-        return img
+            self.random_aspect_prob, self.random_aspect_range)
+        return F.resize(img, size, _pick_interp(self.interpolation))
 
     def __repr__(self):
-        if isinstance(self.interpolation, (tuple, list)):
-            interpolate_str = ' '.join([interp_mode_to_str(x) for x in self.interpolation])
-        else:
-            interpolate_str = interp_mode_to_str(self.interpolation)
-        format_string = self.__class__.__name__ + '(size={0}'.format(self.size)
-        format_string += f', interpolation={interpolate_str}'
-        format_string += f', longest={self.longest:.3f})'
-        return format_string
+        return (f'{self.__class__.__name__}(size={self.size}'
+                f', interpolation={_interp_repr(self.interpolation)}'
+                f', longest={self.longest:.3f})')
 
 
 class TrimBorder(torch.nn.Module):
+    """Crop a fixed border from every edge."""
 
-    def __init__(
-            self,
-            border_size: int,
-    ):
+    def __init__(self, border_size: int):
         super().__init__()
         self.border_size = border_size
 
     def forward(self, img):
         w, h = F.get_image_size(img)
-        top = left = self.border_size
-        top = min(top, h)
-        left = min(left, h)
-        height = max(0, h - 2 * self.border_size)
-        width = max(0, w - 2 * self.border_size)
-        return F.crop(img, top, left, height, width)
+        trim = min(self.border_size, h)
+        return F.crop(
+            img, trim, trim,
+            max(0, h - 2 * self.border_size),
+            max(0, w - 2 * self.border_size))
