@@ -1,7 +1,12 @@
-"""Model registry (reference `timm/models/_registry.py`, 352 LoC).
+"""Model registry: name -> entrypoint / pretrained-cfg bookkeeping.
 
-Decorator-driven registration at import time; wildcard listing; pretrained
-tags; deprecation shims.
+Behavioral parity: /root/reference/timm/models/_registry.py (decorator
+registration at import time, arch.tag naming, wildcard listing with tag
+expansion, deprecation shims).
+
+Redesigned as a single encapsulated ``ModelRegistry`` object (the reference
+keeps eight parallel module-global dicts); module-level functions delegate to
+the default instance so the public API is unchanged.
 """
 import fnmatch
 import re
@@ -10,7 +15,7 @@ import warnings
 from collections import defaultdict, deque
 from copy import deepcopy
 from dataclasses import replace
-from typing import Any, Callable, Dict, Iterable, List, Optional, Sequence, Set, Tuple, Union
+from typing import Any, Callable, Dict, List, Optional, Sequence, Set, Tuple, Union
 
 from ._pretrained import DefaultCfg, PretrainedCfg
 
@@ -21,167 +26,247 @@ __all__ = [
     'get_pretrained_cfg', 'get_pretrained_cfgs_for_arch', 'register_model_deprecations',
 ]
 
-_module_to_models: Dict[str, Set[str]] = defaultdict(set)  # dict of sets to check membership of model in module
-_model_to_module: Dict[str, str] = {}  # mapping of model names to module names
-_model_entrypoints: Dict[str, Callable[..., Any]] = {}  # mapping of model names to architecture entrypoint fns
-_model_has_pretrained: Set[str] = set()  # set of model names that have pretrained weight url present
-_model_default_cfgs: Dict[str, PretrainedCfg] = {}  # central repo for model arch.tag default cfgs
-_model_pretrained_cfgs: Dict[str, PretrainedCfg] = {}  # central repo for model arch.tag pretrained cfgs
-_model_with_tags: Dict[str, List[str]] = defaultdict(list)  # shortcut to map each model arch to all model + tag names
-_module_to_deprecated_models: Dict[str, Dict[str, Optional[str]]] = defaultdict(dict)
-_deprecated_models: Dict[str, Optional[str]] = {}
-
 
 def split_model_name_tag(model_name: str, no_tag: str = '') -> Tuple[str, str]:
-    model_name, *tag_list = model_name.split('.', 1)
-    tag = tag_list[0] if tag_list else no_tag
-    return model_name, tag
+    """'arch.tag' -> (arch, tag); tag defaults to ``no_tag`` when absent."""
+    arch, _, tag = model_name.partition('.')
+    return arch, tag or no_tag
 
 
 def get_arch_name(model_name: str) -> str:
     return split_model_name_tag(model_name)[0]
 
 
+def _natural_key(s: str) -> List[Union[int, str]]:
+    return [int(p) if p.isdigit() else p for p in re.split(r'(\d+)', s.lower())]
+
+
+class ModelRegistry:
+    """All model-name state in one place."""
+
+    def __init__(self):
+        self.entrypoints: Dict[str, Callable[..., Any]] = {}
+        self.arch_module: Dict[str, str] = {}
+        self.module_archs: Dict[str, Set[str]] = defaultdict(set)
+        self.pretrained_names: Set[str] = set()  # names (arch or arch.tag) with weights
+        self.default_cfgs: Dict[str, DefaultCfg] = {}
+        self.pretrained_cfgs: Dict[str, PretrainedCfg] = {}
+        self.arch_tagged_names: Dict[str, List[str]] = defaultdict(list)
+        self.deprecated: Dict[str, Optional[str]] = {}
+        self.module_deprecated: Dict[str, Dict[str, Optional[str]]] = defaultdict(dict)
+
+    # -- registration -------------------------------------------------------
+    def add(self, fn: Callable[..., Any]) -> Callable[..., Any]:
+        module = sys.modules[fn.__module__]
+        module_name = fn.__module__.rsplit('.', 1)[-1]
+        arch = fn.__name__
+
+        # expose through the defining module's __all__
+        if hasattr(module, '__all__'):
+            module.__all__.append(arch)
+        else:
+            module.__all__ = [arch]
+
+        if arch in self.entrypoints:
+            warnings.warn(
+                f'Overwriting {arch} in registry with {fn.__module__}.{arch}. This is '
+                'because the name being registered conflicts with an existing name. '
+                'Please check if this is not expected.',
+                stacklevel=2)
+        self.entrypoints[arch] = fn
+        self.arch_module[arch] = module_name
+        self.module_archs[module_name].add(arch)
+
+        cfg = getattr(module, 'default_cfgs', {}).get(arch, None)
+        if cfg is not None:
+            self._add_cfgs(arch, cfg)
+        return fn
+
+    def _add_cfgs(self, arch: str, default_cfg):
+        if not isinstance(default_cfg, DefaultCfg):
+            # legacy single-dict cfg -> one untagged entry
+            assert isinstance(default_cfg, dict)
+            default_cfg = DefaultCfg(
+                tags=deque(['']), cfgs={'': PretrainedCfg(**default_cfg)})
+
+        for idx, tag in enumerate(default_cfg.tags):
+            cfg = default_cfg.cfgs[tag]
+            full_name = f'{arch}.{tag}' if tag else arch
+            fixups = dict(architecture=arch, tag=tag or None)
+            if cfg.hf_hub_id == 'timm/':
+                fixups['hf_hub_id'] = 'timm/' + full_name  # hub repo named by arch.tag
+            cfg = replace(cfg, **fixups)
+
+            if idx == 0:
+                # first tag is the default resolution for the bare arch name
+                self.pretrained_cfgs[arch] = cfg
+                if cfg.has_weights:
+                    self.pretrained_names.add(arch)
+            if tag:
+                self.pretrained_cfgs[full_name] = cfg
+                if cfg.has_weights:
+                    self.pretrained_names.add(full_name)
+            self.arch_tagged_names[arch].append(full_name)
+        self.default_cfgs[arch] = default_cfg
+
+    def add_deprecations(self, module_path: str, mapping: Dict[str, Optional[str]]):
+        module = sys.modules[module_path]
+        module_name = module_path.rsplit('.', 1)[-1]
+        for old_name, new_name in mapping.items():
+            if hasattr(module, '__all__'):
+                module.__all__.append(old_name)
+            target_fn, target_tag = None, ''
+            if new_name:
+                target_arch, target_tag = split_model_name_tag(new_name)
+                target_fn = getattr(module, target_arch)
+            shim = _deprecation_shim(old_name, target_fn, target_tag)
+            setattr(module, old_name, shim)
+            self.entrypoints[old_name] = shim
+            self.arch_module[old_name] = module_name
+            self.module_archs[module_name].add(old_name)
+            self.deprecated[old_name] = new_name
+            self.module_deprecated[module_name][old_name] = new_name
+
+    # -- queries ------------------------------------------------------------
+    def names(
+            self,
+            filter: Union[str, List[str]] = '',
+            module: Union[str, List[str]] = '',
+            pretrained: bool = False,
+            exclude_filters: Union[str, List[str]] = '',
+            name_matches_cfg: bool = False,
+            include_tags: Optional[bool] = None,
+    ) -> List[str]:
+        if include_tags is None:
+            include_tags = pretrained  # tag expansion implied for weight listings
+
+        if not module:
+            pool: Set[str] = set(self.entrypoints)
+        elif isinstance(module, str):
+            pool = set(self.module_archs[module])
+        else:
+            assert isinstance(module, Sequence)
+            pool = set()
+            for m in module:
+                pool |= self.module_archs[m]
+        pool -= self.deprecated.keys()
+
+        include = [filter] if isinstance(filter, str) and filter else list(filter or [])
+        exclude = [exclude_filters] if isinstance(exclude_filters, str) and exclude_filters \
+            else list(exclude_filters or [])
+
+        if include_tags:
+            expanded: Set[str] = set()
+            for arch in pool:
+                expanded.update(self.arch_tagged_names[arch])
+            pool = expanded
+            include = [x for f in include for x in _with_tag_wildcard(f)]
+            exclude = [x for f in exclude for x in _with_tag_wildcard(f)]
+
+        if include:
+            selected: Set[str] = set()
+            for pattern in include:
+                selected |= set(fnmatch.filter(pool, pattern))
+        else:
+            selected = pool
+        for pattern in exclude:
+            selected -= set(fnmatch.filter(selected, pattern))
+
+        if pretrained:
+            selected &= self.pretrained_names
+        if name_matches_cfg:
+            selected &= set(self.pretrained_cfgs)
+        return sorted(selected, key=_natural_key)
+
+    def entrypoint(self, model_name: str, module_filter: Optional[str] = None):
+        arch = get_arch_name(model_name)
+        if module_filter and arch not in self.module_archs.get(module_filter, {}):
+            raise RuntimeError(f'Model ({model_name} not found in module {module_filter}.')
+        return self.entrypoints[arch]
+
+    def cfg_for(self, model_name: str, allow_unregistered: bool = True):
+        if model_name in self.pretrained_cfgs:
+            return deepcopy(self.pretrained_cfgs[model_name])
+        arch, tag = split_model_name_tag(model_name)
+        if arch in self.default_cfgs:
+            raise RuntimeError(f'Invalid pretrained tag ({tag}) for {arch}.')
+        if allow_unregistered:
+            return None
+        raise RuntimeError(f'Model architecture ({arch}) has no pretrained cfg registered.')
+
+
+def _with_tag_wildcard(pattern: str) -> List[str]:
+    """Untagged filter 'x' also matches tagged names via 'x.*'."""
+    base, tag = split_model_name_tag(pattern)
+    if tag:
+        return [pattern]
+    return [f'{base}.*', pattern]
+
+
+def _deprecation_shim(old_name: str, target_fn: Optional[Callable], target_tag: str):
+    def _shim(pretrained=False, **kwargs):
+        assert target_fn is not None, \
+            f'Model {old_name} has been removed with no replacement.'
+        new_name = f'{target_fn.__name__}.{target_tag}' if target_tag else target_fn.__name__
+        warnings.warn(
+            f'Mapping deprecated model name {old_name} to current {new_name}.',
+            stacklevel=2)
+        pretrained_cfg = kwargs.pop('pretrained_cfg', None)
+        return target_fn(
+            pretrained=pretrained, pretrained_cfg=pretrained_cfg or target_tag, **kwargs)
+    return _shim
+
+
+# the default (and only) registry instance + reference-compatible aliases of
+# its internal tables for external pokes
+_registry = ModelRegistry()
+_model_entrypoints = _registry.entrypoints
+_model_to_module = _registry.arch_module
+_module_to_models = _registry.module_archs
+_model_has_pretrained = _registry.pretrained_names
+_model_default_cfgs = _registry.default_cfgs
+_model_pretrained_cfgs = _registry.pretrained_cfgs
+_model_with_tags = _registry.arch_tagged_names
+_deprecated_models = _registry.deprecated
+_module_to_deprecated_models = _registry.module_deprecated
+
+
 def generate_default_cfgs(cfgs: Dict[str, Union[Dict[str, Any], PretrainedCfg]]):
+    """Group 'arch.tag' -> cfg mappings into per-arch DefaultCfg objects.
+
+    Default-tag priority: explicit untagged-with-weights first, then
+    '*'-suffixed tags, then the first tag with weights.
+    """
     out = defaultdict(DefaultCfg)
-    default_set = set()  # no tag and tags ending with * are prioritized as default
-
-    for k, v in cfgs.items():
-        if isinstance(v, dict):
-            v = PretrainedCfg(**v)
-        has_weights = v.has_weights
-
-        model, tag = split_model_name_tag(k)
-        is_default_set = model in default_set
-        priority = (has_weights and not tag) or (tag.endswith('*') and not is_default_set)
+    explicit_default: Set[str] = set()
+    for name, cfg in cfgs.items():
+        if isinstance(cfg, dict):
+            cfg = PretrainedCfg(**cfg)
+        arch, tag = split_model_name_tag(name)
+        promote = (cfg.has_weights and not tag) or \
+            (tag.endswith('*') and arch not in explicit_default)
         tag = tag.strip('*')
 
-        default_cfg = out[model]
-
-        if priority:
-            default_cfg.tags.appendleft(tag)
-            default_set.add(model)
-        elif has_weights and not default_cfg.is_pretrained:
-            default_cfg.tags.appendleft(tag)
+        entry = out[arch]
+        if promote:
+            entry.tags.appendleft(tag)
+            explicit_default.add(arch)
+        elif cfg.has_weights and not entry.is_pretrained:
+            entry.tags.appendleft(tag)
         else:
-            default_cfg.tags.append(tag)
-
-        if has_weights:
-            default_cfg.is_pretrained = True
-
-        default_cfg.cfgs[tag] = v
-
+            entry.tags.append(tag)
+        if cfg.has_weights:
+            entry.is_pretrained = True
+        entry.cfgs[tag] = cfg
     return out
 
 
 def register_model(fn: Callable[..., Any]) -> Callable[..., Any]:
-    # lookup containing module
-    mod = sys.modules[fn.__module__]
-    module_name_split = fn.__module__.split('.')
-    module_name = module_name_split[-1] if len(module_name_split) else ''
-
-    # add model to __all__ in module
-    model_name = fn.__name__
-    if hasattr(mod, '__all__'):
-        mod.__all__.append(model_name)
-    else:
-        mod.__all__ = [model_name]  # type: ignore
-
-    # add entries to registry dict/sets
-    if model_name in _model_entrypoints:
-        warnings.warn(
-            f'Overwriting {model_name} in registry with {fn.__module__}.{model_name}. This is because the name being '
-            'registered conflicts with an existing name. Please check if this is not expected.',
-            stacklevel=2,
-        )
-    _model_entrypoints[model_name] = fn
-    _model_to_module[model_name] = module_name
-    _module_to_models[module_name].add(model_name)
-    if hasattr(mod, 'default_cfgs') and model_name in mod.default_cfgs:
-        # this will catch all models that have entrypoint matching cfg key, but miss any aliasing
-        # entrypoints or non-matching combos
-        default_cfg = mod.default_cfgs[model_name]
-        if not isinstance(default_cfg, DefaultCfg):
-            # new style default cfg dataclass w/ multiple entries per model-arch
-            assert isinstance(default_cfg, dict)
-            # old style cfg dict per model-arch
-            pretrained_cfg = PretrainedCfg(**default_cfg)
-            default_cfg = DefaultCfg(tags=deque(['']), cfgs={'': pretrained_cfg})
-
-        for tag_idx, tag in enumerate(default_cfg.tags):
-            is_default = tag_idx == 0
-            pretrained_cfg = default_cfg.cfgs[tag]
-            model_name_tag = '.'.join([model_name, tag]) if tag else model_name
-            replace_items = dict(architecture=model_name, tag=tag if tag else None)
-            if pretrained_cfg.hf_hub_id and pretrained_cfg.hf_hub_id == 'timm/':
-                # auto-complete hub name w/ architecture.tag
-                replace_items['hf_hub_id'] = pretrained_cfg.hf_hub_id + model_name_tag
-            pretrained_cfg = replace(pretrained_cfg, **replace_items)
-
-            if is_default:
-                _model_pretrained_cfgs[model_name] = pretrained_cfg
-                if pretrained_cfg.has_weights:
-                    # add tagless entry if it's default and has weights
-                    _model_has_pretrained.add(model_name)
-
-            if tag:
-                _model_pretrained_cfgs[model_name_tag] = pretrained_cfg
-                if pretrained_cfg.has_weights:
-                    # add model w/ tag if tag is valid
-                    _model_has_pretrained.add(model_name_tag)
-                _model_with_tags[model_name].append(model_name_tag)
-            else:
-                _model_with_tags[model_name].append(model_name)  # has empty tag (to slowly remove these instances)
-
-        _model_default_cfgs[model_name] = default_cfg
-
-    return fn
-
-
-def _deprecated_model_shim(deprecated_name: str, current_fn: Optional[Callable] = None, current_tag: str = ''):
-    def _fn(pretrained=False, **kwargs):
-        assert current_fn is not None, f'Model {deprecated_name} has been removed with no replacement.'
-        current_name = '.'.join([current_fn.__name__, current_tag]) if current_tag else current_fn.__name__
-        warnings.warn(f'Mapping deprecated model name {deprecated_name} to current {current_name}.', stacklevel=2)
-        pretrained_cfg = kwargs.pop('pretrained_cfg', None)
-        return current_fn(pretrained=pretrained, pretrained_cfg=pretrained_cfg or current_tag, **kwargs)
-    return _fn
+    return _registry.add(fn)
 
 
 def register_model_deprecations(module_name: str, deprecation_map: Dict[str, Optional[str]]):
-    mod = sys.modules[module_name]
-    module_name_split = module_name.split('.')
-    module_name = module_name_split[-1] if len(module_name_split) else ''
-
-    for deprecated, current in deprecation_map.items():
-        if hasattr(mod, '__all__'):
-            mod.__all__.append(deprecated)
-        current_fn = None
-        current_tag = ''
-        if current:
-            current_name, current_tag = split_model_name_tag(current)
-            current_fn = getattr(mod, current_name)
-        deprecated_entrypoint_fn = _deprecated_model_shim(deprecated, current_fn, current_tag)
-        setattr(mod, deprecated, deprecated_entrypoint_fn)
-        _model_entrypoints[deprecated] = deprecated_entrypoint_fn
-        _model_to_module[deprecated] = module_name
-        _module_to_models[module_name].add(deprecated)
-        _deprecated_models[deprecated] = current
-        _module_to_deprecated_models[module_name][deprecated] = current
-
-
-def _natural_key(string_: str) -> List[Union[int, str]]:
-    """See https://blog.codinghorror.com/sorting-for-humans-natural-sort-order/"""
-    return [int(s) if s.isdigit() else s for s in re.split(r'(\d+)', string_.lower())]
-
-
-def _expand_filter(filter_: str):
-    """expand a 'base_filter' to 'base_filter.*' if no tag portion"""
-    filter_base, filter_tag = split_model_name_tag(filter_)
-    if not filter_tag:
-        return ['.'.join([filter_base, '*']), filter_]
-    else:
-        return [filter_]
+    _registry.add_deprecations(module_name, deprecation_map)
 
 
 def list_models(
@@ -192,137 +277,61 @@ def list_models(
         name_matches_cfg: bool = False,
         include_tags: Optional[bool] = None,
 ) -> List[str]:
-    """Return list of available model names, sorted alphabetically (reference `_registry.py:185-265`)."""
-    if filter:
-        include_filters = filter if isinstance(filter, (tuple, list)) else [filter]
-    else:
-        include_filters = []
-
-    if include_tags is None:
-        # FIXME should this be default behaviour? or default to include_tags=True?
-        include_tags = pretrained
-
-    if not module:
-        all_models: Set[str] = set(_model_entrypoints.keys())
-    else:
-        if isinstance(module, str):
-            all_models: Set[str] = _module_to_models[module]
-        else:
-            assert isinstance(module, Sequence)
-            all_models: Set[str] = set()
-            for m in module:
-                all_models.update(_module_to_models[m])
-    all_models = all_models - set(_deprecated_models.keys())  # remove deprecated models from listings
-
-    if include_tags:
-        # expand model names to include names w/ pretrained tags
-        models_with_tags: Set[str] = set()
-        for m in all_models:
-            models_with_tags.update(_model_with_tags[m])
-        all_models = models_with_tags
-        # expand include and exclude filters to include a wildcard on tag portion
-        include_filters = [ef for f in include_filters for ef in _expand_filter(f)]
-        exclude_filters = [ef for f in exclude_filters for ef in _expand_filter(f)]
-
-    if include_filters:
-        models: Set[str] = set()
-        for f in include_filters:
-            include_models = fnmatch.filter(all_models, f)  # include these models
-            if len(include_models):
-                models = models.union(include_models)
-    else:
-        models = all_models
-
-    if exclude_filters:
-        if not isinstance(exclude_filters, (tuple, list)):
-            exclude_filters = [exclude_filters]
-        for xf in exclude_filters:
-            exclude_models = fnmatch.filter(models, xf)  # exclude these models
-            if len(exclude_models):
-                models = models.difference(exclude_models)
-
-    if pretrained:
-        models = _model_has_pretrained.intersection(models)
-
-    if name_matches_cfg:
-        models = set(_model_pretrained_cfgs).intersection(models)
-
-    return sorted(models, key=_natural_key)
-
-
-def list_pretrained(
-        filter: Union[str, List[str]] = '',
-        exclude_filters: str = '',
-) -> List[str]:
-    return list_models(
+    """List registered model names matching the wildcard filters."""
+    return _registry.names(
         filter=filter,
-        pretrained=True,
+        module=module,
+        pretrained=pretrained,
         exclude_filters=exclude_filters,
-        include_tags=True,
+        name_matches_cfg=name_matches_cfg,
+        include_tags=include_tags,
     )
 
 
+def list_pretrained(filter: Union[str, List[str]] = '', exclude_filters: str = '') -> List[str]:
+    return _registry.names(
+        filter=filter, pretrained=True, exclude_filters=exclude_filters, include_tags=True)
+
+
 def get_deprecated_models(module: str = '') -> Dict[str, str]:
-    all_deprecated = _module_to_deprecated_models[module] if module else _deprecated_models
-    return deepcopy(all_deprecated)
+    source = _registry.module_deprecated[module] if module else _registry.deprecated
+    return deepcopy(source)
 
 
 def is_model(model_name: str) -> bool:
-    """Check if a model name exists."""
-    arch_name = get_arch_name(model_name)
-    return arch_name in _model_entrypoints
+    return get_arch_name(model_name) in _registry.entrypoints
 
 
 def model_entrypoint(model_name: str, module_filter: Optional[str] = None) -> Callable[..., Any]:
-    """Fetch a model entrypoint for specified model name."""
-    arch_name = get_arch_name(model_name)
-    if module_filter and arch_name not in _module_to_models.get(module_filter, {}):
-        raise RuntimeError(f'Model ({model_name} not found in module {module_filter}.')
-    return _model_entrypoints[arch_name]
+    return _registry.entrypoint(model_name, module_filter)
 
 
 def list_modules() -> List[str]:
-    """Return list of module names that contain models / model entrypoints."""
-    modules = _module_to_models.keys()
-    return sorted(modules)
+    return sorted(_registry.module_archs.keys())
 
 
-def is_model_in_modules(
-        model_name: str, module_names: Union[Tuple, List, Set]) -> bool:
-    arch_name = get_arch_name(model_name)
+def is_model_in_modules(model_name: str, module_names: Union[Tuple, List, Set]) -> bool:
     assert isinstance(module_names, (tuple, list, set))
-    return any(arch_name in _module_to_models[n] for n in module_names)
+    arch = get_arch_name(model_name)
+    return any(arch in _registry.module_archs[m] for m in module_names)
 
 
 def is_model_pretrained(model_name: str) -> bool:
-    return model_name in _model_has_pretrained
+    return model_name in _registry.pretrained_names
 
 
 def get_pretrained_cfg(model_name: str, allow_unregistered: bool = True) -> Optional[PretrainedCfg]:
-    if model_name in _model_pretrained_cfgs:
-        return deepcopy(_model_pretrained_cfgs[model_name])
-    arch_name, tag = split_model_name_tag(model_name)
-    if arch_name in _model_default_cfgs:
-        # if model arch exists, but the tag is wrong, error out
-        raise RuntimeError(f'Invalid pretrained tag ({tag}) for {arch_name}.')
-    if allow_unregistered:
-        # if model arch doesn't exist, it has no pretrained_cfg registered, allow a non-standard args
-        return None
-    raise RuntimeError(f'Model architecture ({arch_name}) has no pretrained cfg registered.')
+    return _registry.cfg_for(model_name, allow_unregistered=allow_unregistered)
 
 
 def get_pretrained_cfg_value(model_name: str, cfg_key: str) -> Optional[Any]:
-    """Get a specific model default_cfg value by key."""
-    cfg = get_pretrained_cfg(model_name, allow_unregistered=False)
+    cfg = _registry.cfg_for(model_name, allow_unregistered=False)
     return getattr(cfg, cfg_key, None)
 
 
 def get_pretrained_cfgs_for_arch(model_name: str) -> Dict[str, PretrainedCfg]:
-    arch_name = get_arch_name(model_name)
-    if arch_name in _model_default_cfgs:
-        out = {}
-        dcfg = _model_default_cfgs[arch_name]
-        for tag in dcfg.tags:
-            out[tag] = deepcopy(dcfg.cfgs[tag])
-        return out
-    return {}
+    arch = get_arch_name(model_name)
+    if arch not in _registry.default_cfgs:
+        return {}
+    dcfg = _registry.default_cfgs[arch]
+    return {tag: deepcopy(dcfg.cfgs[tag]) for tag in dcfg.tags}
