@@ -309,53 +309,51 @@ class UniversalInvertedResidual(nn.Module):
         conv_kwargs = conv_kwargs or {}
         self.has_skip = (in_chs == out_chs and stride == 1) and not noskip
 
+        # ConvNormAct units keep reference checkpoint key layout (.conv/.bn)
+        from ..layers import ConvNormAct
         if dw_kernel_size_start:
             dw_start_stride = stride if not dw_kernel_size_mid else 1
-            dw_start_groups = num_groups(group_size, in_chs)
-            self.dw_start = nn.Sequential(
-                create_conv2d(
-                    in_chs, in_chs, dw_kernel_size_start,
-                    stride=dw_start_stride, dilation=dilation, groups=dw_start_groups,
-                    padding=pad_type, **conv_kwargs),
-                norm_act_layer(in_chs, apply_act=False),
+            self.dw_start = ConvNormAct(
+                in_chs, in_chs, dw_kernel_size_start,
+                stride=dw_start_stride, dilation=dilation,
+                groups=num_groups(group_size, in_chs), padding=pad_type,
+                apply_act=False, act_layer=act_layer, norm_layer=norm_layer,
+                conv_kwargs=conv_kwargs,
             )
         else:
             self.dw_start = nn.Identity()
 
         mid_chs = make_divisible(in_chs * exp_ratio)
-        self.pw_exp = nn.Sequential(
-            create_conv2d(in_chs, mid_chs, 1, padding=pad_type, **conv_kwargs),
-            norm_act_layer(mid_chs, inplace=True),
+        self.pw_exp = ConvNormAct(
+            in_chs, mid_chs, 1, padding=pad_type,
+            act_layer=act_layer, norm_layer=norm_layer, conv_kwargs=conv_kwargs,
         )
 
         if dw_kernel_size_mid:
-            groups = num_groups(group_size, mid_chs)
-            self.dw_mid = nn.Sequential(
-                create_conv2d(
-                    mid_chs, mid_chs, dw_kernel_size_mid,
-                    stride=stride, dilation=dilation, groups=groups,
-                    padding=pad_type, **conv_kwargs),
-                norm_act_layer(mid_chs, inplace=True),
+            self.dw_mid = ConvNormAct(
+                mid_chs, mid_chs, dw_kernel_size_mid,
+                stride=stride, dilation=dilation,
+                groups=num_groups(group_size, mid_chs), padding=pad_type,
+                act_layer=act_layer, norm_layer=norm_layer, conv_kwargs=conv_kwargs,
             )
         else:
             self.dw_mid = nn.Identity()
 
         self.se = se_layer(mid_chs, act_layer=act_layer) if se_layer else nn.Identity()
 
-        self.pw_proj = nn.Sequential(
-            create_conv2d(mid_chs, out_chs, 1, padding=pad_type, **conv_kwargs),
-            norm_act_layer(out_chs, apply_act=False),
+        self.pw_proj = ConvNormAct(
+            mid_chs, out_chs, 1, padding=pad_type, apply_act=False,
+            act_layer=act_layer, norm_layer=norm_layer, conv_kwargs=conv_kwargs,
         )
 
         if dw_kernel_size_end:
             dw_end_stride = stride if not dw_kernel_size_start and not dw_kernel_size_mid else 1
-            dw_end_groups = num_groups(group_size, out_chs)
-            self.dw_end = nn.Sequential(
-                create_conv2d(
-                    out_chs, out_chs, dw_kernel_size_end,
-                    stride=dw_end_stride, dilation=dilation, groups=dw_end_groups,
-                    padding=pad_type, **conv_kwargs),
-                norm_act_layer(out_chs, apply_act=False),
+            self.dw_end = ConvNormAct(
+                out_chs, out_chs, dw_kernel_size_end,
+                stride=dw_end_stride, dilation=dilation,
+                groups=num_groups(group_size, out_chs), padding=pad_type,
+                apply_act=False, act_layer=act_layer, norm_layer=norm_layer,
+                conv_kwargs=conv_kwargs,
             )
         else:
             self.dw_end = nn.Identity()
@@ -368,9 +366,9 @@ class UniversalInvertedResidual(nn.Module):
 
     def feature_info(self, location):
         if location == 'expansion':
-            return dict(module='pw_proj.0', hook_type='forward_pre', num_chs=self.pw_proj[0].in_channels)
+            return dict(module='pw_proj.conv', hook_type='forward_pre', num_chs=self.pw_proj.conv.in_channels)
         else:
-            return dict(module='', num_chs=self.pw_proj[0].out_channels)
+            return dict(module='', num_chs=self.pw_proj.conv.out_channels)
 
     def forward(self, x):
         shortcut = x

@@ -1,10 +1,12 @@
-"""EfficientNet-family arch builder + block-spec string DSL
-(reference `timm/models/_efficientnet_builder.py`, 581 LoC).
+"""EfficientNet-family trunk builder + block-spec string DSL.
 
-Arch definitions are lists of per-stage block-spec strings like
-"ir_r2_k3_s2_e6_c24_se0.25": block type + repeat/kernel/stride/expansion/
-channels/SE options; `decode_arch_def` expands them, `EfficientNetBuilder`
-instantiates stages with channel-multiplier rounding and feature hooks.
+Behavioral parity: /root/reference/timm/models/_efficientnet_builder.py
+(the "ir_r2_k3_s2_e6_c24_se0.25" grammar, depth scaling, space2depth
+regions, output-stride dilation, feature-info bookkeeping, TF weight init).
+
+Redesign: block decoding and construction are table-driven — one
+``_BLOCK_KWARGS`` decoder map and one ``_BLOCK_FACTORY`` constructor map per
+block type — rather than parallel if/elif ladders.
 """
 import logging
 import math
@@ -24,35 +26,34 @@ from ._manipulate import named_modules
 
 __all__ = [
     'EfficientNetBuilder', 'BlockArgs', 'decode_arch_def', 'efficientnet_init_weights',
-    'resolve_bn_args', 'resolve_act_layer', 'round_channels', 'BN_MOMENTUM_TF_DEFAULT', 'BN_EPS_TF_DEFAULT',
+    'resolve_bn_args', 'resolve_act_layer', 'round_channels', 'BN_MOMENTUM_TF_DEFAULT',
+    'BN_EPS_TF_DEFAULT',
 ]
 
 _logger = logging.getLogger(__name__)
 
 _DEBUG_BUILDER = False
 
-# Defaults used for Google/Tensorflow training of mobile networks /w RMSprop as per
-# papers and TF reference implementations. PT momentum equiv for TF decay is (1 - TF decay)
+# TF-style BN defaults (PT momentum = 1 - TF decay)
 BN_MOMENTUM_TF_DEFAULT = 1 - 0.99
 BN_EPS_TF_DEFAULT = 1e-3
-_BN_ARGS_TF = dict(momentum=BN_MOMENTUM_TF_DEFAULT, eps=BN_EPS_TF_DEFAULT)
 
 BlockArgs = List[List[Dict[str, Any]]]
 
 
 def get_bn_args_tf():
-    return _BN_ARGS_TF.copy()
+    return dict(momentum=BN_MOMENTUM_TF_DEFAULT, eps=BN_EPS_TF_DEFAULT)
 
 
 def resolve_bn_args(kwargs):
-    bn_args = {}
-    bn_momentum = kwargs.pop('bn_momentum', None)
-    if bn_momentum is not None:
-        bn_args['momentum'] = bn_momentum
-    bn_eps = kwargs.pop('bn_eps', None)
-    if bn_eps is not None:
-        bn_args['eps'] = bn_eps
-    return bn_args
+    out = {}
+    momentum = kwargs.pop('bn_momentum', None)
+    if momentum is not None:
+        out['momentum'] = momentum
+    eps = kwargs.pop('bn_eps', None)
+    if eps is not None:
+        out['eps'] = eps
+    return out
 
 
 def resolve_act_layer(kwargs, default='relu'):
@@ -60,7 +61,7 @@ def resolve_act_layer(kwargs, default='relu'):
 
 
 def round_channels(channels, multiplier=1.0, divisor=8, channel_min=None, round_limit=0.9):
-    """Round number of filters based on depth multiplier."""
+    """Width-multiplier channel rounding (divisor-aligned)."""
     if not multiplier:
         return channels
     return make_divisible(channels * multiplier, divisor, channel_min, round_limit=round_limit)
@@ -71,196 +72,171 @@ def _log_info_if(msg, condition):
         _logger.info(msg)
 
 
-def _parse_ksize(ss):
-    if ss.isdigit():
-        return int(ss)
-    else:
-        return [int(k) for k in ss.split('.')]
+# ---------------------------------------------------------------------------
+# block-spec DSL: "<type>_<key><val>_..." tokens
+# ---------------------------------------------------------------------------
+
+# activation shorthand used by the 'n' option
+_ACT_SHORTHAND = dict(re='relu', r6='relu6', hs='hard_swish', sw='swish', mi='mish')
+
+
+def _parse_ksize(v):
+    """Kernel spec: '3' -> 3, '3.5.7' -> [3, 5, 7] (MixedConv)."""
+    return int(v) if v.isdigit() else [int(k) for k in v.split('.')]
+
+
+def _tokenize(ops):
+    """Parse option tokens -> (options dict, skip override)."""
+    options, skip = {}, None
+    for op in ops:
+        if op == 'noskip':
+            skip = False
+        elif op == 'skip':
+            skip = True
+        elif op.startswith('n'):
+            act_name = _ACT_SHORTHAND.get(op[1:])
+            if act_name is not None:
+                options['n'] = get_act_layer(act_name)
+        else:
+            parts = re.split(r'(\d.*)', op)
+            if len(parts) >= 2:
+                options[parts[0]] = parts[1]
+    return options, skip
+
+
+def _kwargs_ir(o, skip, start_k, end_k):
+    out = dict(
+        dw_kernel_size=_parse_ksize(o['k']),
+        exp_kernel_size=start_k,
+        pw_kernel_size=end_k,
+        exp_ratio=float(o['e']),
+        se_ratio=float(o.get('se', 0.)),
+        noskip=skip is False,
+        s2d=int(o.get('d', 0)) > 0,
+    )
+    if 'cc' in o:
+        out['num_experts'] = int(o['cc'])
+    return out
+
+
+def _kwargs_ds(o, skip, start_k, end_k, block_type='ds'):
+    return dict(
+        dw_kernel_size=_parse_ksize(o['k']),
+        pw_kernel_size=end_k,
+        se_ratio=float(o.get('se', 0.)),
+        pw_act=block_type == 'dsa',
+        noskip=block_type == 'dsa' or skip is False,
+        s2d=int(o.get('d', 0)) > 0,
+    )
+
+
+def _kwargs_er(o, skip, start_k, end_k):
+    return dict(
+        exp_kernel_size=_parse_ksize(o['k']),
+        pw_kernel_size=end_k,
+        exp_ratio=float(o['e']),
+        force_in_chs=int(o.get('fc', 0)),
+        se_ratio=float(o.get('se', 0.)),
+        noskip=skip is False,
+    )
+
+
+def _kwargs_cn(o, skip, start_k, end_k):
+    return dict(kernel_size=int(o['k']), skip=skip is True)
+
+
+def _kwargs_uir(o, skip, start_k, end_k):
+    # uir reinterprets the a/p kernels as start/end dw kernels (0 = absent)
+    return dict(
+        dw_kernel_size_start=_parse_ksize(o['a']) if 'a' in o else 0,
+        dw_kernel_size_mid=_parse_ksize(o['k']),
+        dw_kernel_size_end=_parse_ksize(o['p']) if 'p' in o else 0,
+        exp_ratio=float(o['e']),
+        se_ratio=float(o.get('se', 0.)),
+        noskip=skip is False,
+    )
+
+
+def _kwargs_attn(o, skip, start_k, end_k):
+    kv_dim = int(o['d'])
+    return dict(
+        dw_kernel_size=_parse_ksize(o['k']),
+        num_heads=int(o['h']),
+        key_dim=kv_dim,
+        value_dim=kv_dim,
+        kv_stride=int(o.get('v', 1)),
+        noskip=skip is False,
+    )
+
+
+_BLOCK_KWARGS = {
+    'ir': _kwargs_ir,
+    'ds': partial(_kwargs_ds, block_type='ds'),
+    'dsa': partial(_kwargs_ds, block_type='dsa'),
+    'er': _kwargs_er,
+    'cn': _kwargs_cn,
+    'uir': _kwargs_uir,
+    'mha': _kwargs_attn,
+    'mqa': _kwargs_attn,
+}
 
 
 def _decode_block_str(block_str):
-    """Decode block definition string.
+    """One spec string -> (block kwargs, repeat count).
 
-    Gets a list of block arg (dicts) through a string notation of arguments.
-    E.g. ir_r2_k3_s2_e1_i32_o16_se0.25_noskip
-
-    All args can exist in any order with the exception of the leading string which
-    is assumed to indicate the block type.
-
-    leading string - block type (
-      ir = InvertedResidual, ds = DepthwiseSep Conv, dsa = DeptwhiseSep Conv with pw act,
-      cn = ConvBnAct, uir = UniversalInvertedResidual, er = EdgeResidual)
-    r - number of repeat blocks,
-    k - kernel size,
-    s - strides (1-9),
-    e - expansion ratio,
-    c - output channels,
-    se - squeeze/excitation ratio
-    n - activation fn ('re', 'r6', 'hs', 'sw', 'mi')
+    Grammar: leading token = block type (ir/ds/dsa/er/cn/uir/mha/mqa), then
+    r repeats · k kernel · s stride · e expansion · c out-chans · se ratio ·
+    n act ('re','r6','hs','sw','mi') · a/p start/end kernels · d (s2d or
+    attn kv-dim) · h heads · v kv-stride · cc experts · gs group size ·
+    skip/noskip.
     """
     assert isinstance(block_str, str)
-    ops = block_str.split('_')
-    block_type = ops[0]  # take the block type off the front
-    ops = ops[1:]
-    options = {}
-    skip = None
-    for op in ops:
-        # string options being checked on individual basis, combine if they grow
-        if op == 'noskip':
-            skip = False  # force no skip connection
-        elif op == 'skip':
-            skip = True  # force a skip connection
-        elif op.startswith('n'):
-            # activation fn
-            key = op[0]
-            v = op[1:]
-            if v == 're':
-                value = get_act_layer('relu')
-            elif v == 'r6':
-                value = get_act_layer('relu6')
-            elif v == 'hs':
-                value = get_act_layer('hard_swish')
-            elif v == 'sw':
-                value = get_act_layer('swish')  # aka SiLU
-            elif v == 'mi':
-                value = get_act_layer('mish')
-            else:
-                continue
-            options[key] = value
-        else:
-            # all numeric options
-            splits = re.split(r'(\d.*)', op)
-            if len(splits) >= 2:
-                key, value = splits[:2]
-                options[key] = value
+    block_type, *ops = block_str.split('_')
+    assert block_type in _BLOCK_KWARGS, f'Unknown block type ({block_type})'
+    options, skip = _tokenize(ops)
 
-    # if act_layer is None, the model default (passed to model init) will be used
-    act_layer = options['n'] if 'n' in options else None
-    start_kernel_size = _parse_ksize(options['a']) if 'a' in options else 1
-    end_kernel_size = _parse_ksize(options['p']) if 'p' in options else 1
-    force_in_chs = int(options['fc']) if 'fc' in options else 0  # FIXME hack to deal with in_chs issue in TPU def
-    num_repeat = int(options['r'])
+    start_k = _parse_ksize(options['a']) if 'a' in options else 1
+    end_k = _parse_ksize(options['p']) if 'p' in options else 1
 
-    # each type of block has different valid arguments, fill accordingly
-    block_args = dict(
+    args = dict(
         block_type=block_type,
         out_chs=int(options['c']),
         stride=int(options['s']),
-        act_layer=act_layer,
+        act_layer=options.get('n'),  # None -> model default at build time
     )
-    if block_type == 'ir':
-        block_args.update(dict(
-            dw_kernel_size=_parse_ksize(options['k']),
-            exp_kernel_size=start_kernel_size,
-            pw_kernel_size=end_kernel_size,
-            exp_ratio=float(options['e']),
-            se_ratio=float(options.get('se', 0.)),
-            noskip=skip is False,
-            s2d=int(options.get('d', 0)) > 0,
-        ))
-        if 'cc' in options:
-            block_args['num_experts'] = int(options['cc'])
-    elif block_type == 'ds' or block_type == 'dsa':
-        block_args.update(dict(
-            dw_kernel_size=_parse_ksize(options['k']),
-            pw_kernel_size=end_kernel_size,
-            se_ratio=float(options.get('se', 0.)),
-            pw_act=block_type == 'dsa',
-            noskip=block_type == 'dsa' or skip is False,
-            s2d=int(options.get('d', 0)) > 0,
-        ))
-    elif block_type == 'er':
-        block_args.update(dict(
-            exp_kernel_size=_parse_ksize(options['k']),
-            pw_kernel_size=end_kernel_size,
-            exp_ratio=float(options['e']),
-            force_in_chs=force_in_chs,
-            se_ratio=float(options.get('se', 0.)),
-            noskip=skip is False,
-        ))
-    elif block_type == 'cn':
-        block_args.update(dict(
-            kernel_size=int(options['k']),
-            skip=skip is True,
-        ))
-    elif block_type == 'uir':
-        # override exp / proj kernels for start/end in uir block
-        start_kernel_size = _parse_ksize(options['a']) if 'a' in options else 0
-        end_kernel_size = _parse_ksize(options['p']) if 'p' in options else 0
-        block_args.update(dict(
-            dw_kernel_size_start=start_kernel_size,  # overload exp ks arg for dw start
-            dw_kernel_size_mid=_parse_ksize(options['k']),
-            dw_kernel_size_end=end_kernel_size,  # overload pw ks arg for dw end
-            exp_ratio=float(options['e']),
-            se_ratio=float(options.get('se', 0.)),
-            noskip=skip is False,
-        ))
-    elif block_type == 'mha':
-        kv_dim = int(options['d'])
-        block_args.update(dict(
-            dw_kernel_size=_parse_ksize(options['k']),
-            num_heads=int(options['h']),
-            key_dim=kv_dim,
-            value_dim=kv_dim,
-            kv_stride=int(options.get('v', 1)),
-            noskip=skip is False,
-        ))
-    elif block_type == 'mqa':
-        kv_dim = int(options['d'])
-        block_args.update(dict(
-            dw_kernel_size=_parse_ksize(options['k']),
-            num_heads=int(options['h']),
-            key_dim=kv_dim,
-            value_dim=kv_dim,
-            kv_stride=int(options.get('v', 1)),
-            noskip=skip is False,
-        ))
-    else:
-        assert False, 'Unknown block type (%s)' % block_type
-
+    args.update(_BLOCK_KWARGS[block_type](options, skip, start_k, end_k))
     if 'gs' in options:
-        block_args['group_size'] = int(options['gs'])
-
-    return block_args, num_repeat
+        args['group_size'] = int(options['gs'])
+    return args, int(options['r'])
 
 
 def _scale_stage_depth(stack_args, repeats, depth_multiplier=1.0, depth_trunc='ceil'):
-    """Per-stage depth scaling: scales the block repeats in each stage.
+    """Scale a stage's total repeat count, distributing over its block defs.
 
-    This depth scaling impl maintains compatibility with the EfficientNet scaling method,
-    while allowing sensible scaling for other models that may have multiple block arg definitions
-    in each stage.
+    Allocation runs in reverse so the first (often unique) block of a stage
+    is the last to gain repeats — matches EfficientNet compound scaling.
     """
-
-    # We scale the total repeat count for each stage, there may be multiple
-    # block arg defs per stage so we need to sum.
-    num_repeat = sum(repeats)
+    total = sum(repeats)
     if depth_trunc == 'round':
-        # Truncating to int by rounding allows stages with few repeats to remain
-        # proportionally smaller for longer. This is a good choice when stage definitions
-        # include single repeat stages that we'd prefer to keep that way as long as possible
-        num_repeat_scaled = max(1, round(num_repeat * depth_multiplier))
+        # rounding keeps small stages proportionally small for longer
+        total_scaled = max(1, round(total * depth_multiplier))
     else:
-        # The default for EfficientNet truncates repeats to int via 'ceil'.
-        # Any multiplier > 1.0 will result in an increased depth for every stage.
-        num_repeat_scaled = int(math.ceil(num_repeat * depth_multiplier))
+        # EfficientNet default: ceil, so any multiplier > 1 deepens all stages
+        total_scaled = int(math.ceil(total * depth_multiplier))
 
-    # Proportionally distribute repeat count scaling to each block definition in the stage.
-    # Allocation is done in reverse as it results in the first block being less likely to be scaled.
-    # The first block makes less sense to repeat in most of the arch definitions.
-    repeats_scaled = []
-    for r in repeats[::-1]:
-        rs = max(1, round((r / num_repeat * num_repeat_scaled)))
-        repeats_scaled.append(rs)
-        num_repeat -= r
-        num_repeat_scaled -= rs
-    repeats_scaled = repeats_scaled[::-1]
+    scaled = []
+    remaining, remaining_scaled = total, total_scaled
+    for r in reversed(repeats):
+        take = max(1, round(r / remaining * remaining_scaled))
+        scaled.append(take)
+        remaining -= r
+        remaining_scaled -= take
+    scaled.reverse()
 
-    # Apply the calculated scaling to each block arg in the stage
-    sa_scaled = []
-    for ba, rep in zip(stack_args, repeats_scaled):
-        sa_scaled.extend([deepcopy(ba) for _ in range(rep)])
-    return sa_scaled
+    out = []
+    for args, rep in zip(stack_args, scaled):
+        out.extend(deepcopy(args) for _ in range(rep))
+    return out
 
 
 def decode_arch_def(
@@ -271,46 +247,53 @@ def decode_arch_def(
         fix_first_last=False,
         group_size=None,
 ):
-    """Decode block architecture definition strings -> block kwargs.
-
-    Args:
-        arch_def: architecture definition strings, list of list of strings
-        depth_multiplier: network depth multiplier
-        depth_trunc: networ depth truncation mode when applying multiplier
-        experts_multiplier: CondConv experts multiplier
-        fix_first_last: fix first and last block depths when multiplier is applied
-        group_size: group size override for all blocks that weren't explicitly set in arch string
-
-    Returns:
-        list of list of block kwargs
-    """
-    arch_args = []
-    if isinstance(depth_multiplier, tuple):
-        assert len(depth_multiplier) == len(arch_def)
-    else:
+    """Expand per-stage spec-string lists into per-block kwargs lists."""
+    if not isinstance(depth_multiplier, tuple):
         depth_multiplier = (depth_multiplier,) * len(arch_def)
-    for stack_idx, (block_strings, multiplier) in enumerate(zip(arch_def, depth_multiplier)):
-        assert isinstance(block_strings, list)
-        stack_args = []
-        repeats = []
-        for block_str in block_strings:
-            assert isinstance(block_str, str)
-            ba, rep = _decode_block_str(block_str)
-            if ba.get('num_experts', 0) > 0 and experts_multiplier > 1:
-                ba['num_experts'] *= experts_multiplier
+    else:
+        assert len(depth_multiplier) == len(arch_def)
+
+    stages = []
+    for stage_idx, (specs, mult) in enumerate(zip(arch_def, depth_multiplier)):
+        assert isinstance(specs, list)
+        stage_args, repeats = [], []
+        for spec in specs:
+            assert isinstance(spec, str)
+            args, rep = _decode_block_str(spec)
+            if args.get('num_experts', 0) > 0 and experts_multiplier > 1:
+                args['num_experts'] *= experts_multiplier
             if group_size is not None:
-                ba.setdefault('group_size', group_size)
-            stack_args.append(ba)
+                args.setdefault('group_size', group_size)
+            stage_args.append(args)
             repeats.append(rep)
-        if fix_first_last and (stack_idx == 0 or stack_idx == len(arch_def) - 1):
-            arch_args.append(_scale_stage_depth(stack_args, repeats, 1.0, depth_trunc))
-        else:
-            arch_args.append(_scale_stage_depth(stack_args, repeats, multiplier, depth_trunc))
-    return arch_args
+        edge_stage = stage_idx in (0, len(arch_def) - 1)
+        stage_mult = 1.0 if (fix_first_last and edge_stage) else mult
+        stages.append(_scale_stage_depth(stage_args, repeats, stage_mult, depth_trunc))
+    return stages
+
+
+# ---------------------------------------------------------------------------
+# trunk builder
+# ---------------------------------------------------------------------------
+
+def get_attn_or_se(se_layer):
+    if se_layer is None:
+        return SqueezeExcite
+    if isinstance(se_layer, str):
+        if se_layer == 'se':
+            return SqueezeExcite
+        from ..layers import get_attn
+        return get_attn(se_layer)
+    return se_layer
 
 
 class EfficientNetBuilder:
-    """Build Trunk Blocks (reference `_efficientnet_builder.py:316`)."""
+    """Instantiate trunk stages from decoded block args.
+
+    Tracks running in_chs / stride / dilation across blocks, manages
+    space2depth regions, and records feature_info entries at stage
+    boundaries for feature-extraction wrappers.
+    """
 
     def __init__(
             self,
@@ -329,206 +312,180 @@ class EfficientNetBuilder:
         self.output_stride = output_stride
         self.pad_type = pad_type
         self.round_chs_fn = round_chs_fn
-        self.se_from_exp = se_from_exp  # calculate se channel reduction from expanded (mid) chs
+        self.se_from_exp = se_from_exp  # SE channels from expanded (mid) chs vs input
         self.act_layer = act_layer
         self.norm_layer = norm_layer
         self.aa_layer = aa_layer
         self.se_layer = get_attn_or_se(se_layer)
         try:
-            self.se_layer(8, rd_ratio=1.0)  # test if attn layer accepts rd_ratio arg
+            self.se_layer(8, rd_ratio=1.0)  # probe rd_ratio support
             self.se_has_ratio = True
         except TypeError:
             self.se_has_ratio = False
         self.drop_path_rate = drop_path_rate
         self.layer_scale_init_value = layer_scale_init_value
         if feature_location == 'depthwise':
-            # old 'depthwise' mode renamed 'expansion' to match TF impl, old expansion mode didn't make sense
             _logger.warning("feature_location=='depthwise' is deprecated, using 'expansion'")
             feature_location = 'expansion'
-        self.feature_location = feature_location
         assert feature_location in ('bottleneck', 'expansion', '')
+        self.feature_location = feature_location
         self.verbose = _DEBUG_BUILDER
 
-        # state updated during build, consumed by model
+        # running build state
         self.in_chs = None
         self.features = []
 
+    # per-type constructors; ls = layer_scale_init_value
+    def _factory(self, bt):
+        ls = self.layer_scale_init_value
+        table = {
+            'ir': lambda ba: CondConvResidual(**ba) if ba.get('num_experts', 0)
+                else InvertedResidual(**ba),
+            'ds': DepthwiseSeparableConv,
+            'dsa': DepthwiseSeparableConv,
+            'er': EdgeResidual,
+            'cn': ConvBnAct,
+            'uir': lambda ba: UniversalInvertedResidual(**ba, layer_scale_init_value=ls),
+            'mqa': lambda ba: MobileAttention(**ba, use_multi_query=True, layer_scale_init_value=ls),
+            'mha': lambda ba: MobileAttention(**ba, layer_scale_init_value=ls),
+        }
+        ctor = table[bt]
+        if isinstance(ctor, type):
+            return lambda ba: ctor(**ba)
+        return ctor
+
+    def _resolve_se(self, ba, se_ratio, s2d):
+        if not (se_ratio and self.se_layer is not None):
+            return
+        if not self.se_from_exp:
+            # SE sized from block input: fold out the expansion
+            se_ratio /= ba.get('exp_ratio', 1.0)
+        if s2d == 1:
+            se_ratio /= 4  # start of a space2depth region
+        if self.se_has_ratio:
+            ba['se_layer'] = partial(self.se_layer, rd_ratio=se_ratio)
+        else:
+            ba['se_layer'] = self.se_layer
+
     def _make_block(self, ba, block_idx, block_count):
-        drop_path_rate = self.drop_path_rate * block_idx / block_count
         bt = ba.pop('block_type')
         ba['in_chs'] = self.in_chs
         ba['out_chs'] = self.round_chs_fn(ba['out_chs'])
         s2d = ba.get('s2d', 0)
         if s2d > 0:
-            # adjust while space2depth active
-            ba['out_chs'] *= 4
-        if 'force_in_chs' in ba and ba['force_in_chs']:
-            # NOTE this is a hack to work around mismatch in TF EdgeEffNet impl
+            ba['out_chs'] *= 4  # channels quadruple while space2depth is active
+        if ba.get('force_in_chs'):
             ba['force_in_chs'] = self.round_chs_fn(ba['force_in_chs'])
         ba['pad_type'] = self.pad_type
-        # block act fn overrides the model default
-        ba['act_layer'] = ba['act_layer'] if ba['act_layer'] is not None else self.act_layer
+        ba['act_layer'] = ba['act_layer'] or self.act_layer
         assert ba['act_layer'] is not None
         ba['norm_layer'] = self.norm_layer
-        ba['drop_path_rate'] = drop_path_rate
+        ba['drop_path_rate'] = self.drop_path_rate * block_idx / block_count
+        self._resolve_se(ba, ba.pop('se_ratio', None), s2d)
 
-        se_ratio = ba.pop('se_ratio', None)
-        if se_ratio and self.se_layer is not None:
-            if not self.se_from_exp:
-                # adjust se_ratio by expansion ratio if calculating se channels from block input
-                se_ratio /= ba.get('exp_ratio', 1.0)
-            if s2d == 1:
-                # adjust for start of space2depth
-                se_ratio /= 4
-            if self.se_has_ratio:
-                ba['se_layer'] = partial(self.se_layer, rd_ratio=se_ratio)
-            else:
-                ba['se_layer'] = self.se_layer
-
-        if bt == 'ir':
-            _log_info_if('  InvertedResidual {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = CondConvResidual(**ba) if ba.get('num_experts', 0) else InvertedResidual(**ba)
-        elif bt == 'ds' or bt == 'dsa':
-            _log_info_if('  DepthwiseSeparable {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = DepthwiseSeparableConv(**ba)
-        elif bt == 'er':
-            _log_info_if('  EdgeResidual {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = EdgeResidual(**ba)
-        elif bt == 'cn':
-            _log_info_if('  ConvBnAct {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = ConvBnAct(**ba)
-        elif bt == 'uir':
-            _log_info_if('  UniversalInvertedResidual {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = UniversalInvertedResidual(**ba, layer_scale_init_value=self.layer_scale_init_value)
-        elif bt == 'mqa':
-            _log_info_if('  MobileMultiQueryAttention {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = MobileAttention(**ba, use_multi_query=True, layer_scale_init_value=self.layer_scale_init_value)
-        elif bt == 'mha':
-            _log_info_if('  MobileMultiHeadAttention {}, Args: {}'.format(block_idx, str(ba)), self.verbose)
-            block = MobileAttention(**ba, layer_scale_init_value=self.layer_scale_init_value)
-        else:
-            assert False, 'Unknown block type (%s) while building model.' % bt
-
-        self.in_chs = ba['out_chs']  # update in_chs for arg of next block
+        _log_info_if(f'  Block type {bt} {block_idx}, Args: {ba}', self.verbose)
+        block = self._factory(bt)(ba)
+        self.in_chs = ba['out_chs']
         return block
 
     def __call__(self, in_chs, model_block_args):
-        """Build the entire trunk of blocks from the architecture args."""
-        _log_info_if('Building model trunk with %d stages...' % len(model_block_args), self.verbose)
+        """Build all trunk stages; returns a list of nn.Sequential stages."""
+        _log_info_if(f'Building model trunk with {len(model_block_args)} stages...', self.verbose)
         self.in_chs = in_chs
-        total_block_count = sum([len(x) for x in model_block_args])
-        total_block_idx = 0
-        current_stride = 2
-        current_dilation = 1
+        block_count = sum(len(stage) for stage in model_block_args)
+        block_idx = 0
+        stride = 2
+        dilation = 1
+        s2d_phase = 0  # 0 inactive, 1 starting, 2 active
         stages = []
+
         if model_block_args[0][0]['stride'] > 1:
-            # if the first block starts with a stride, we need to extract first level feat from stem
-            feature_info = dict(module='bn1', num_chs=in_chs, stage=0, reduction=current_stride)
-            self.features.append(feature_info)
+            # stem output is itself a feature level when stage 0 downsamples
+            self.features.append(
+                dict(module='bn1', num_chs=in_chs, stage=0, reduction=stride))
 
-        # outer list of block_args defines the stacks
-        space2depth = 0
-        for stack_idx, stack_args in enumerate(model_block_args):
-            last_stack = stack_idx + 1 == len(model_block_args)
-            _log_info_if('Stack: {}'.format(stack_idx), self.verbose)
-            assert isinstance(stack_args, list)
-
+        for stage_idx, stage_args in enumerate(model_block_args):
+            assert isinstance(stage_args, list)
+            _log_info_if(f'Stack: {stage_idx}', self.verbose)
             blocks = []
-            # each stack (stage of blocks) contains a list of block arguments
-            for block_idx, block_args in enumerate(stack_args):
-                last_block = block_idx + 1 == len(stack_args)
-                _log_info_if(' Block: {}'.format(block_idx), self.verbose)
+            for i, ba in enumerate(stage_args):
+                last_in_stage = i + 1 == len(stage_args)
+                assert ba['stride'] in (1, 2)
+                if i >= 1:
+                    ba['stride'] = 1  # only a stage's first block may downsample
 
-                assert block_args['stride'] in (1, 2)
-                if block_idx >= 1:  # only the first block in any stack can have a stride > 1
-                    block_args['stride'] = 1
-
-                if not space2depth and block_args.pop('s2d', False):
-                    assert block_args['stride'] == 1
-                    space2depth = 1
-
-                if space2depth > 0:
-                    if space2depth == 2 and block_args['stride'] == 2:
-                        block_args['stride'] = 1
-                        # to end s2d region, need to correct expansion and se ratio relative to input
-                        block_args['exp_ratio'] /= 4
-                        space2depth = 0
+                # space2depth region tracking
+                if not s2d_phase and ba.pop('s2d', False):
+                    assert ba['stride'] == 1
+                    s2d_phase = 1
+                if s2d_phase > 0:
+                    if s2d_phase == 2 and ba['stride'] == 2:
+                        # region ends at the next downsample; undo the 4x
+                        ba['stride'] = 1
+                        ba['exp_ratio'] /= 4
+                        s2d_phase = 0
                     else:
-                        block_args['s2d'] = space2depth
+                        ba['s2d'] = s2d_phase
 
-                extract_features = False
-                if last_block:
-                    next_stack_idx = stack_idx + 1
-                    extract_features = next_stack_idx >= len(model_block_args) or \
-                        model_block_args[next_stack_idx][0]['stride'] > 1
+                # a stage boundary (or network end) right after this block
+                # means its output is a feature level
+                extract = last_in_stage and (
+                    stage_idx + 1 >= len(model_block_args) or
+                    model_block_args[stage_idx + 1][0]['stride'] > 1)
 
-                next_dilation = current_dilation
-                if block_args['stride'] > 1:
-                    next_output_stride = current_stride * block_args['stride']
-                    if next_output_stride > self.output_stride:
-                        next_dilation = current_dilation * block_args['stride']
-                        block_args['stride'] = 1
-                        _log_info_if('  Converting stride to dilation to maintain output_stride=={}'.format(
-                            self.output_stride), self.verbose)
+                # convert stride to dilation past the output_stride budget
+                next_dilation = dilation
+                if ba['stride'] > 1:
+                    if stride * ba['stride'] > self.output_stride:
+                        next_dilation = dilation * ba['stride']
+                        ba['stride'] = 1
+                        _log_info_if(
+                            f'  Converting stride to dilation to maintain '
+                            f'output_stride=={self.output_stride}', self.verbose)
                     else:
-                        current_stride = next_output_stride
-                block_args['dilation'] = current_dilation
-                if next_dilation != current_dilation:
-                    current_dilation = next_dilation
+                        stride *= ba['stride']
+                ba['dilation'] = dilation
+                dilation = next_dilation
 
-                # create the block
-                block = self._make_block(block_args, total_block_idx, total_block_count)
-                blocks.append(block)
+                blocks.append(self._make_block(ba, block_idx, block_count))
+                if s2d_phase == 1:
+                    s2d_phase = 2
 
-                if space2depth == 1:
-                    space2depth = 2
-
-                # stash feature module name and channel info for model feature extraction
-                if extract_features:
-                    feature_info = dict(
-                        stage=stack_idx + 1,
-                        reduction=current_stride,
-                        **block.feature_info(self.feature_location),
+                if extract:
+                    info = dict(
+                        stage=stage_idx + 1,
+                        reduction=stride,
+                        **blocks[-1].feature_info(self.feature_location),
                     )
-                    leaf_name = feature_info.get('module', '')
-                    if leaf_name:
-                        feature_info['module'] = '.'.join([f'blocks.{stack_idx}.{block_idx}', leaf_name])
+                    leaf = info.get('module', '')
+                    if leaf:
+                        info['module'] = f'blocks.{stage_idx}.{i}.{leaf}'
                     else:
-                        assert last_block
-                        feature_info['module'] = f'blocks.{stack_idx}'
-                    self.features.append(feature_info)
-
-                total_block_idx += 1  # incr global block idx (across all stacks)
+                        assert last_in_stage
+                        info['module'] = f'blocks.{stage_idx}'
+                    self.features.append(info)
+                block_idx += 1
             stages.append(nn.Sequential(*blocks))
         return stages
 
 
-def get_attn_or_se(se_layer):
-    if se_layer is None:
-        return SqueezeExcite
-    if isinstance(se_layer, str):
-        if se_layer == 'se':
-            return SqueezeExcite
-        from ..layers import get_attn
-        return get_attn(se_layer)
-    return se_layer
-
+# ---------------------------------------------------------------------------
+# TF-style weight init
+# ---------------------------------------------------------------------------
 
 def _init_weight_goog(m, n='', fix_group_fanout=True):
-    """Weight initialization as per Tensorflow official implementations."""
-    if isinstance(m, nn.Conv2d) or (hasattr(m, 'weight') and hasattr(m, 'weight_shape')):
-        from ..layers.cond_conv2d import CondConv2d, get_condconv_initializer
-        if isinstance(m, CondConv2d):
-            fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels
-            if fix_group_fanout:
-                fan_out //= m.groups
-            init_weight_fn = get_condconv_initializer(
-                lambda w: nn.init.normal_(w, 0, math.sqrt(2.0 / fan_out)), m.num_experts, m.weight_shape)
-            init_weight_fn(m.weight)
-            if m.bias is not None:
-                nn.init.zeros_(m.bias)
-            return
-    if isinstance(m, nn.Conv2d):
+    """Conv/BN/Linear init matching the TF reference implementations."""
+    from ..layers.cond_conv2d import CondConv2d, get_condconv_initializer
+    if isinstance(m, CondConv2d):
+        fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels
+        if fix_group_fanout:
+            fan_out //= m.groups
+        get_condconv_initializer(
+            lambda w: nn.init.normal_(w, 0, math.sqrt(2.0 / fan_out)),
+            m.num_experts, m.weight_shape)(m.weight)
+        if m.bias is not None:
+            nn.init.zeros_(m.bias)
+    elif isinstance(m, nn.Conv2d):
         fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels
         if fix_group_fanout:
             fan_out //= m.groups
@@ -539,12 +496,10 @@ def _init_weight_goog(m, n='', fix_group_fanout=True):
         nn.init.ones_(m.weight)
         nn.init.zeros_(m.bias)
     elif isinstance(m, nn.Linear):
-        fan_out = m.weight.size(0)  # fan-out
-        fan_in = 0
-        if 'routing_fn' in n:
-            fan_in = m.weight.size(1)
-        init_range = 1.0 / math.sqrt(fan_in + fan_out)
-        nn.init.uniform_(m.weight, -init_range, init_range)
+        fan_out = m.weight.size(0)
+        fan_in = m.weight.size(1) if 'routing_fn' in n else 0
+        bound = 1.0 / math.sqrt(fan_in + fan_out)
+        nn.init.uniform_(m.weight, -bound, bound)
         if m.bias is not None:
             nn.init.zeros_(m.bias)
 

@@ -13,7 +13,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
-from ..layers import DropPath, Mlp, PatchEmbed, _assert, to_2tuple, trunc_normal_
+from ..layers import DropPath, Mlp, PatchEmbed, _assert, to_2tuple, trunc_normal_, LayerNorm
 from ._builder import build_model_with_cfg
 from ._registry import generate_default_cfgs, register_model
 
@@ -254,7 +254,7 @@ class CoaT(nn.Module):
             proj_drop_rate: float = 0.,
             attn_drop_rate: float = 0.,
             drop_path_rate: float = 0.,
-            norm_layer: Type[nn.Module] = nn.LayerNorm,
+            norm_layer: Type[nn.Module] = LayerNorm,
             return_interm_layers: bool = False,
             out_features: Optional[List[str]] = None,
             crpe_window: Optional[dict] = None,

@@ -11,7 +11,7 @@ import torch
 import torch.nn as nn
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
-from ..layers import DropPath, Mlp, PatchEmbed, trunc_normal_
+from ..layers import DropPath, Mlp, PatchEmbed, trunc_normal_, LayerNorm
 from ._builder import build_model_with_cfg
 from ._registry import generate_default_cfgs, register_model
 
@@ -124,7 +124,7 @@ class MHSA(nn.Module):
 class Block(nn.Module):
     def __init__(
             self, dim, num_heads, mlp_ratio=4., qkv_bias=False, proj_drop=0.,
-            attn_drop=0., drop_path=0., act_layer=nn.GELU, norm_layer=nn.LayerNorm,
+            attn_drop=0., drop_path=0., act_layer=nn.GELU, norm_layer=LayerNorm,
             use_gpsa=True, locality_strength=1.):
         super().__init__()
         self.norm1 = norm_layer(dim)
@@ -166,7 +166,7 @@ class ConVit(nn.Module):
             proj_drop_rate: float = 0.,
             attn_drop_rate: float = 0.,
             drop_path_rate: float = 0.,
-            norm_layer: Type[nn.Module] = nn.LayerNorm,
+            norm_layer: Type[nn.Module] = LayerNorm,
             local_up_to_layer: int = 3,
             locality_strength: float = 1.,
             use_pos_embed: bool = True,

@@ -19,7 +19,8 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
-from ..layers import ConvBnAct, Linear, SelectAdaptivePool2d, make_divisible
+from ..layers import Linear, SelectAdaptivePool2d, make_divisible
+from ._efficientnet_blocks import ConvBnAct  # bn named `bn1` (reference checkpoint layout)
 from ._builder import build_model_with_cfg
 from ._efficientnet_blocks import SqueezeExcite
 from ._manipulate import checkpoint_seq

@@ -16,7 +16,7 @@ import torch.nn.functional as F
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
 from .. import ops
-from ..layers import DropPath, Mlp, LayerNorm, use_fused_attn, _assert, get_norm_layer, to_2tuple
+from ..layers import ClNormMlpClassifierHead, DropPath, Mlp, LayerNorm, use_fused_attn, _assert, get_norm_layer, to_2tuple
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
 from ._manipulate import checkpoint
@@ -494,9 +494,16 @@ class Hiera(nn.Module):
             self.blocks.append(block)
 
         self.num_features = self.head_hidden_size = embed_dim
-        self.norm = norm_layer(embed_dim)
-        self.head_drop = nn.Dropout(drop_rate)
-        self.head = nn.Linear(embed_dim, num_classes) if num_classes > 0 else nn.Identity()
+        # reference head structure (hiera.py:604): norm + fc live inside the
+        # head module so checkpoint keys are head.norm.* / head.fc.*
+        self.head = ClNormMlpClassifierHead(
+            embed_dim,
+            num_classes,
+            pool_type=global_pool,
+            drop_rate=drop_rate,
+            norm_layer=norm_layer,
+            input_fmt='NLC',
+        )
         self.global_pool = global_pool
 
         # Initialize everything
@@ -512,10 +519,10 @@ class Hiera(nn.Module):
         if weight_init != 'skip':
             init_fn = partial(self._init_weights)
             self.apply(init_fn)
-        if isinstance(self.head, nn.Linear):
-            self.head.weight.data.mul_(head_init_scale)
-            if self.head.bias is not None:
-                self.head.bias.data.mul_(head_init_scale)
+        if isinstance(self.head.fc, nn.Linear):
+            self.head.fc.weight.data.mul_(head_init_scale)
+            if self.head.fc.bias is not None:
+                self.head.fc.bias.data.mul_(head_init_scale)
 
     def _init_weights(self, m, init_bias=0.02):
         if isinstance(m, (nn.Linear, nn.Conv1d, nn.Conv2d, nn.Conv3d)):
@@ -546,13 +553,13 @@ class Hiera(nn.Module):
 
     @torch.jit.ignore
     def get_classifier(self):
-        return self.head
+        return self.head.fc
 
-    def reset_classifier(self, num_classes: int, global_pool: Optional[str] = None):
+    def reset_classifier(self, num_classes: int, global_pool: Optional[str] = None, reset_other: bool = False):
         self.num_classes = num_classes
         if global_pool is not None:
             self.global_pool = global_pool
-        self.head = nn.Linear(self.num_features, num_classes) if num_classes > 0 else nn.Identity()
+        self.head.reset(num_classes, global_pool, reset_other=reset_other)
 
     def get_random_mask(self, x: torch.Tensor, mask_ratio: float) -> torch.Tensor:
         """Generates a random mask, mask_ratio fraction of mask units masked out.
@@ -714,12 +721,7 @@ class Hiera(nn.Module):
         return x
 
     def forward_head(self, x, pre_logits: bool = False) -> torch.Tensor:
-        if self.global_pool == 'avg':
-            x = x.mean(dim=1)
-        x = self.norm(x)
-        x = self.head_drop(x)
-        x = x if pre_logits else self.head(x)
-        return x
+        return self.head(x, pre_logits=pre_logits) if pre_logits else self.head(x)
 
     def forward(
             self,

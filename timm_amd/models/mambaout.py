@@ -12,7 +12,7 @@ import torch
 from torch import nn
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
-from ..layers import ClNormMlpClassifierHead, DropPath, LayerScale, get_act_layer, trunc_normal_
+from ..layers import ClNormMlpClassifierHead, DropPath, LayerScale, get_act_layer, trunc_normal_, LayerNorm
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
 from ._manipulate import checkpoint_seq
@@ -23,7 +23,7 @@ __all__ = ['MambaOut']
 
 class Stem(nn.Module):
     def __init__(
-            self, in_chs=3, out_chs=96, mid_norm=True, act_layer=nn.GELU, norm_layer=nn.LayerNorm):
+            self, in_chs=3, out_chs=96, mid_norm=True, act_layer=nn.GELU, norm_layer=LayerNorm):
         super().__init__()
         self.conv1 = nn.Conv2d(in_chs, out_chs // 2, kernel_size=3, stride=2, padding=1)
         self.norm1 = norm_layer(out_chs // 2) if mid_norm else None
@@ -45,7 +45,7 @@ class Stem(nn.Module):
 
 
 class DownsampleNormFirst(nn.Module):
-    def __init__(self, in_chs=96, out_chs=198, norm_layer=nn.LayerNorm):
+    def __init__(self, in_chs=96, out_chs=198, norm_layer=LayerNorm):
         super().__init__()
         self.norm = norm_layer(in_chs)
         self.conv = nn.Conv2d(in_chs, out_chs, kernel_size=3, stride=2, padding=1)
@@ -59,7 +59,7 @@ class DownsampleNormFirst(nn.Module):
 
 
 class Downsample(nn.Module):
-    def __init__(self, in_chs=96, out_chs=198, norm_layer=nn.LayerNorm):
+    def __init__(self, in_chs=96, out_chs=198, norm_layer=LayerNorm):
         super().__init__()
         self.conv = nn.Conv2d(in_chs, out_chs, kernel_size=3, stride=2, padding=1)
         self.norm = norm_layer(out_chs)
@@ -77,7 +77,7 @@ class MlpHead(nn.Module):
 
     def __init__(
             self, in_features, num_classes=1000, pool_type='avg', act_layer=nn.GELU,
-            mlp_ratio=4, norm_layer=nn.LayerNorm, drop_rate=0., bias=True):
+            mlp_ratio=4, norm_layer=LayerNorm, drop_rate=0., bias=True):
         super().__init__()
         hidden_size = int(mlp_ratio * in_features) if mlp_ratio is not None else None
         self.pool_type = pool_type
@@ -124,7 +124,7 @@ class GatedConvBlock(nn.Module):
 
     def __init__(
             self, dim, expansion_ratio=8 / 3, kernel_size=7, conv_ratio=1.0,
-            ls_init_value=None, norm_layer=nn.LayerNorm, act_layer=nn.GELU,
+            ls_init_value=None, norm_layer=LayerNorm, act_layer=nn.GELU,
             drop_path=0., **kwargs):
         super().__init__()
         self.norm = norm_layer(dim)
@@ -158,7 +158,7 @@ class MambaOutStage(nn.Module):
     def __init__(
             self, dim, dim_out=None, depth=4, expansion_ratio=8 / 3, kernel_size=7,
             conv_ratio=1.0, downsample='', ls_init_value=None,
-            norm_layer=nn.LayerNorm, act_layer=nn.GELU, drop_path=0.):
+            norm_layer=LayerNorm, act_layer=nn.GELU, drop_path=0.):
         super().__init__()
         dim_out = dim_out or dim
         self.grad_checkpointing = False
