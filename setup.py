@@ -21,6 +21,7 @@ sources = [
     os.path.join(CSRC, 'attention_bwd.hip'),
     os.path.join(CSRC, 'depthwise_conv.hip'),
     os.path.join(CSRC, 'multi_tensor.hip'),
+    os.path.join(CSRC, 'muon_ns.hip'),
 ]
 
 setup(

@@ -42,6 +42,12 @@ at::Tensor dwconv_bwd_data(at::Tensor dy, at::Tensor w_t, long stride, long pad,
 std::vector<at::Tensor> dwconv_bwd_weight(at::Tensor dy, at::Tensor x, long stride, long pad,
                                           long K, bool need_bias);
 
+// muon_ns.hip
+at::Tensor ns_gemm_nt(at::Tensor l, at::Tensor r, c10::optional<at::Tensor> s,
+                      double alpha, double beta);
+at::Tensor ns_gemm_nn(at::Tensor l, at::Tensor r, c10::optional<at::Tensor> s,
+                      double alpha, double beta);
+
 // multi_tensor.hip
 void multi_tensor_adamw(
     std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
@@ -67,6 +73,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_fwd", &dwconv_fwd, "NHWC depthwise conv fwd (gfx950)");
   m.def("dwconv_bwd_data", &dwconv_bwd_data, "NHWC depthwise conv bwd-data");
   m.def("dwconv_bwd_weight", &dwconv_bwd_weight, "NHWC depthwise conv bwd-weight");
+  m.def("ns_gemm_nt", &ns_gemm_nt, "batched bf16 MFMA GEMM (L R^T + S) for Muon NS");
+  m.def("ns_gemm_nn", &ns_gemm_nn, "batched bf16 MFMA GEMM (L R + S) for Muon NS");
   m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
   m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");

@@ -42,10 +42,23 @@ def zeropower_via_newtonschulz(
     # normalize so top singular value <= 1
     X = X / (X.norm(dim=(-2, -1), keepdim=True) + eps)
 
-    for _ in range(steps):
-        A = X @ X.mT
-        B = b * A + c * (A @ A)
-        X = a * X + B @ X
+    ext = ops._load_extension() if X.is_cuda else None
+    if ext is not None:
+        # in-house batched MFMA kernels (muon_ns.hip): A and B are symmetric,
+        # so every product runs as row-major NT except the final BX (NN with
+        # transposed staging).  fp32 accumulate, bf16 IO.
+        shape = X.shape
+        Xb = X.reshape(-1, shape[-2], shape[-1]).contiguous()
+        for _ in range(steps):
+            A = ext.ns_gemm_nt(Xb, Xb, None, 1.0, 0.0)
+            B = ext.ns_gemm_nt(A, A, A, c, b)
+            Xb = ext.ns_gemm_nn(B, Xb, Xb, 1.0, a)
+        X = Xb.reshape(shape)
+    else:
+        for _ in range(steps):
+            A = X @ X.mT
+            B = b * A + c * (A @ A)
+            X = a * X + B @ X
 
     if transposed:
         X = X.mT
@@ -123,6 +136,10 @@ class Muon(Optimizer):
 
             adamw_params, adamw_grads, adamw_m, adamw_v = [], [], [], []
             adamw_step = None
+            # muon-path work items batched by 2d shape so the NS kernels run
+            # one [G,M,N] launch per shape (fills all 256 CUs instead of
+            # per-matrix grids of a few dozen workgroups)
+            ns_queue = {}  # (M, N) -> list of (param, grad, u2, orig_shape)
 
             for p in group['params']:
                 if p.grad is None:
@@ -146,17 +163,7 @@ class Muon(Optimizer):
                             u2 = u.reshape(-1, shape[-1])
                     else:
                         u2 = u
-                    u2 = zeropower_via_newtonschulz(u2, steps=ns_steps)
-                    scale = _lr_scale(u2.shape, group['lr_scale_mode'])
-                    if wd:
-                        p.mul_(1 - lr * wd)
-                    if group['caution']:
-                        upd = u2.reshape(shape)
-                        mask = (upd * g > 0).to(g.dtype)
-                        mask.div_(mask.mean().clamp_(min=1e-3))
-                        p.add_(upd * mask, alpha=-lr * scale)
-                    else:
-                        p.add_(u2.reshape(shape), alpha=-lr * scale)
+                    ns_queue.setdefault(tuple(u2.shape), []).append((p, g, u2, shape))
                 else:
                     if 'exp_avg' not in state:
                         state['exp_avg'] = torch.zeros_like(p, dtype=torch.float32)
@@ -168,6 +175,23 @@ class Muon(Optimizer):
                     adamw_grads.append(g)
                     adamw_m.append(state['exp_avg'])
                     adamw_v.append(state['exp_avg_sq'])
+
+            # run NS batched per shape, then apply updates
+            for (M, N), items in ns_queue.items():
+                stack = torch.stack([u2 for _p, _g, u2, _s in items]) if len(items) > 1 \
+                    else items[0][2].unsqueeze(0)
+                ortho = zeropower_via_newtonschulz(stack, steps=ns_steps)
+                scale = _lr_scale((M, N), group['lr_scale_mode'])
+                for (p, g, _u2, shape), o in zip(items, ortho.unbind(0)):
+                    if wd:
+                        p.mul_(1 - lr * wd)
+                    upd = o.to(p.dtype).reshape(shape)
+                    if group['caution']:
+                        mask = (upd * g > 0).to(g.dtype)
+                        mask.div_(mask.mean().clamp_(min=1e-3))
+                        p.add_(upd * mask, alpha=-lr * scale)
+                    else:
+                        p.add_(upd, alpha=-lr * scale)
 
             if adamw_params:
                 b1, b2 = group['adamw_betas']
