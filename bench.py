@@ -42,6 +42,8 @@ def parse_args():
                    help="'both' measures infer then train and emits one JSON line "
                         "with train as the headline metric + infer_* keys")
     p.add_argument('--amp-dtype', type=str, default='bfloat16')
+    p.add_argument('--opt', type=str, default='adamw', choices=['adamw', 'muon'],
+                   help='optimizer for the train phase (muon = NS-kernel path, BASELINE config #4)')
     p.add_argument('--channels-last', action='store_true', default=None,
                    help='NHWC memory format (default: auto-on for conv nets)')
     return p.parse_args()
@@ -126,11 +128,15 @@ def main():
 
     if args.mode in ('both', 'train'):
         model.train()
-        from timm_amd.optim import AdamW
         if distributed:
             from timm_amd.parallel import BucketedDataParallel
             model = BucketedDataParallel(model, bucket_cap_mb=50.)
-        optimizer = AdamW(model.parameters(), lr=1e-4, weight_decay=0.05)
+        if args.opt == 'muon':
+            from timm_amd.optim.muon import Muon
+            optimizer = Muon(model.parameters(), lr=1e-3, weight_decay=0.05)
+        else:
+            from timm_amd.optim import AdamW
+            optimizer = AdamW(model.parameters(), lr=1e-4, weight_decay=0.05)
 
         B = args.batch_size
         x = torch.randn(B, 3, args.img_size, args.img_size, device=device, dtype=dtype)
