@@ -9,6 +9,8 @@ from .dataset_factory import create_dataset
 from .distributed_sampler import OrderedDistributedSampler, RepeatAugSampler
 from .loader import create_loader, fast_collate, PrefetchLoader, MultiEpochsDataLoader
 from .mixup import Mixup, FastCollateMixup, mixup_target, rand_bbox, rand_bbox_minmax
+from .naflex_mixup import NaFlexMixup, mix_batch_variable_size, pairwise_mixup_target
+from .naflex_random_erasing import PatchRandomErasing
 from .random_erasing import RandomErasing
 from .real_labels import RealLabelsImagenet
 from .transforms import (

@@ -49,7 +49,11 @@ class NaFlexCollator:
     def __call__(self, batch) -> Tuple[Dict[str, torch.Tensor], torch.Tensor]:
         assert isinstance(batch[0], tuple)
         samples = [b[0] for b in batch]
-        targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
+        if isinstance(batch[0][1], torch.Tensor) and batch[0][1].ndim > 0:
+            # soft targets from NaFlexMixup
+            targets = torch.stack([b[1] for b in batch])
+        else:
+            targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
 
         seq_len = self.max_seq_len or max(s['patches'].shape[0] for s in samples)
         patch_dim = samples[0]['patches'].shape[-1]
@@ -165,13 +169,18 @@ class NaFlexMapDatasetWrapper(IterableDataset):
 
     def _make_batch(self, seq_len: int, sample_indices: List[int]):
         transform = self.transforms.get(seq_len)
-        samples = []
+        imgs, targets = [], []
         for idx in sample_indices:
             img, target = self.base_dataset[idx]
             if transform is not None:
                 img = transform(img)
-            sample = self.patchifier(img)
-            samples.append((sample, target))
+            imgs.append(img)
+            targets.append(target)
+        if self.mixup_fn is not None:
+            # variable-size mixup runs on the pre-patchify image tensors and
+            # returns provenance-matched soft targets (naflex_mixup.py)
+            imgs, targets = self.mixup_fn(imgs, torch.tensor(targets))
+        samples = [(self.patchifier(im), tg) for im, tg in zip(imgs, targets)]
         return self.collators[seq_len](samples)
 
     def __iter__(self) -> Iterator:

@@ -93,6 +93,7 @@ def create_naflex_loader(
         patch_size: Union[Tuple[int, int], int] = 16,
         train_seq_lens: Tuple[int, ...] = (128, 256, 576, 784, 1024),
         max_seq_len: int = 576,
+        mixup_fn: Optional[Callable] = None,
         batch_size: int = 32,  # used for eval & max for train
         max_tokens_per_batch: int = 4096 * 4,
         is_training: bool = False,
@@ -128,6 +129,7 @@ def create_naflex_loader(
             seq_lens=train_seq_lens,
             max_tokens_per_batch=max_tokens_per_batch,
             transform_factory=transform_factory,
+            mixup_fn=mixup_fn,
             seed=seed,
             distributed=distributed,
             rank=rank,

@@ -27,8 +27,10 @@ import yaml
 import timm_amd
 from timm_amd import utils
 from timm_amd.data import (
-    AugMixDataset, FastCollateMixup, Mixup, create_dataset, create_loader, resolve_data_config,
+    AugMixDataset, FastCollateMixup, Mixup, NaFlexMixup, create_dataset, create_loader,
+    resolve_data_config,
 )
+from timm_amd.data.naflex_loader import create_naflex_loader
 from timm_amd.layers import convert_sync_batchnorm
 from timm_amd.loss import BinaryCrossEntropy, JsdCrossEntropy, LabelSmoothingCrossEntropy, SoftTargetCrossEntropy
 from timm_amd.models import create_model, safe_model_name
@@ -168,6 +170,19 @@ group.add_argument('--mixup-switch-prob', type=float, default=0.5)
 group.add_argument('--mixup-mode', type=str, default='batch')
 group.add_argument('--mixup-off-epoch', default=0, type=int, metavar='N')
 group.add_argument('--smoothing', type=float, default=0.1)
+# NaFlex variable-resolution training (reference train.py:423-435)
+group.add_argument('--naflex-loader', action='store_true', default=False,
+                   help='enable the NaFlex variable-seq-len loader')
+group.add_argument('--naflex-train-seq-lens', type=int, nargs='+',
+                   default=[128, 256, 576, 784, 1024],
+                   help='sequence-length buckets for NaFlex training')
+group.add_argument('--naflex-max-seq-len', type=int, default=576,
+                   help='fixed sequence length for NaFlex validation')
+group.add_argument('--naflex-max-tokens-per-batch', type=int, default=4096 * 4,
+                   help='token budget per train batch (batch size varies by bucket)')
+group.add_argument('--naflex-loss-scale', default='linear', type=str,
+                   choices=('none', 'sqrt', 'linear'),
+                   help='variable-batch loss rescale mode')
 group.add_argument('--train-interpolation', type=str, default='random')
 group.add_argument('--drop', type=float, default=0.0, metavar='PCT')
 group.add_argument('--drop-path', type=float, default=None, metavar='PCT')
@@ -458,59 +473,120 @@ def main():
     if num_aug_splits > 1:
         dataset_train = AugMixDataset(dataset_train, num_splits=num_aug_splits)
 
+    naflex_mode = False
+    if args.naflex_loader:
+        # NaFlex variable-res pipeline (reference train.py:739-805): token-budget
+        # batches over seq-len buckets; mixup runs inside the dataset wrapper
+        assert num_aug_splits <= 1, 'Augmentation splits not supported in NaFlex mode'
+        naflex_mixup_fn = None
+        if mixup_active:
+            nf_args = dict(mixup_args)
+            nf_args.pop('mode', None)
+            nf_args.pop('cutmix_minmax', None)
+            naflex_mixup_fn = NaFlexMixup(**nf_args)
+            collate_fn = None
+            mixup_fn = None
+        patch_size = 16
+        if hasattr(model, 'patch_embed') and hasattr(model.patch_embed, 'patch_size'):
+            patch_size = model.patch_embed.patch_size
+            if isinstance(patch_size, (tuple, list)):
+                patch_size = patch_size[0]
+        elif hasattr(model, 'embeds') and hasattr(model.embeds, 'patch_size'):
+            patch_size = model.embeds.patch_size
+            if isinstance(patch_size, (tuple, list)):
+                patch_size = patch_size[0]
+        naflex_mode = True
+        loader_train = create_naflex_loader(
+            dataset_train,
+            patch_size=patch_size,
+            train_seq_lens=args.naflex_train_seq_lens,
+            max_tokens_per_batch=args.naflex_max_tokens_per_batch,
+            mixup_fn=naflex_mixup_fn,
+            is_training=True,
+            mean=data_config['mean'],
+            std=data_config['std'],
+            num_workers=args.workers,
+            distributed=args.distributed,
+            rank=args.rank,
+            world_size=args.world_size,
+            seed=args.seed,
+            pin_memory=args.pin_mem,
+            device=device,
+            use_prefetcher=args.prefetcher,
+        )
+        loader_eval = create_naflex_loader(
+            dataset_eval,
+            patch_size=patch_size,
+            max_seq_len=args.naflex_max_seq_len,
+            batch_size=args.validation_batch_size or args.batch_size,
+            is_training=False,
+            mean=data_config['mean'],
+            std=data_config['std'],
+            num_workers=args.workers,
+            distributed=args.distributed,
+            rank=args.rank,
+            world_size=args.world_size,
+            pin_memory=args.pin_mem,
+            device=device,
+            use_prefetcher=args.prefetcher,
+        )
+
     # create data loaders w/ augmentation pipeline
     train_interpolation = args.train_interpolation
     if args.no_aug or not train_interpolation:
         train_interpolation = data_config['interpolation']
-    loader_train = create_loader(
-        dataset_train,
-        input_size=data_config['input_size'],
-        batch_size=args.batch_size,
-        is_training=True,
-        no_aug=args.no_aug,
-        re_prob=args.reprob,
-        re_mode=args.remode,
-        re_count=args.recount,
-        re_split=args.resplit,
-        train_crop_mode=args.train_crop_mode,
-        scale=args.scale,
-        ratio=args.ratio,
-        hflip=args.hflip,
-        vflip=args.vflip,
-        color_jitter=args.color_jitter,
-        color_jitter_prob=args.color_jitter_prob,
-        grayscale_prob=args.grayscale_prob,
-        gaussian_blur_prob=args.gaussian_blur_prob,
-        auto_augment=args.aa,
-        num_aug_repeats=args.aug_repeats,
-        num_aug_splits=num_aug_splits,
-        interpolation=train_interpolation,
-        mean=data_config['mean'],
-        std=data_config['std'],
-        num_workers=args.workers,
-        distributed=args.distributed,
-        collate_fn=collate_fn,
-        pin_memory=args.pin_mem,
-        device=device,
-        use_prefetcher=args.prefetcher,
-        worker_seeding=args.worker_seeding,
-    )
+    if naflex_mode:
+        pass  # loaders built above
+    else:
+        loader_train = create_loader(
+            dataset_train,
+            input_size=data_config['input_size'],
+            batch_size=args.batch_size,
+            is_training=True,
+            no_aug=args.no_aug,
+            re_prob=args.reprob,
+            re_mode=args.remode,
+            re_count=args.recount,
+            re_split=args.resplit,
+            train_crop_mode=args.train_crop_mode,
+            scale=args.scale,
+            ratio=args.ratio,
+            hflip=args.hflip,
+            vflip=args.vflip,
+            color_jitter=args.color_jitter,
+            color_jitter_prob=args.color_jitter_prob,
+            grayscale_prob=args.grayscale_prob,
+            gaussian_blur_prob=args.gaussian_blur_prob,
+            auto_augment=args.aa,
+            num_aug_repeats=args.aug_repeats,
+            num_aug_splits=num_aug_splits,
+            interpolation=train_interpolation,
+            mean=data_config['mean'],
+            std=data_config['std'],
+            num_workers=args.workers,
+            distributed=args.distributed,
+            collate_fn=collate_fn,
+            pin_memory=args.pin_mem,
+            device=device,
+            use_prefetcher=args.prefetcher,
+            worker_seeding=args.worker_seeding,
+        )
 
-    loader_eval = create_loader(
-        dataset_eval,
-        input_size=data_config['input_size'],
-        batch_size=args.validation_batch_size or args.batch_size,
-        is_training=False,
-        interpolation=data_config['interpolation'],
-        mean=data_config['mean'],
-        std=data_config['std'],
-        num_workers=args.workers,
-        distributed=args.distributed,
-        crop_pct=data_config['crop_pct'],
-        pin_memory=args.pin_mem,
-        device=device,
-        use_prefetcher=args.prefetcher,
-    )
+        loader_eval = create_loader(
+            dataset_eval,
+            input_size=data_config['input_size'],
+            batch_size=args.validation_batch_size or args.batch_size,
+            is_training=False,
+            interpolation=data_config['interpolation'],
+            mean=data_config['mean'],
+            std=data_config['std'],
+            num_workers=args.workers,
+            distributed=args.distributed,
+            crop_pct=data_config['crop_pct'],
+            pin_memory=args.pin_mem,
+            device=device,
+            use_prefetcher=args.prefetcher,
+        )
 
     # setup checkpoint saver and eval metric tracking
     eval_metric = args.eval_metric
@@ -732,11 +808,17 @@ def train_one_epoch(
         if batch_idx >= last_batch_idx_to_accum:
             accum_steps = last_accum_steps
 
+        naflex_batch = isinstance(input, dict)
         if not args.prefetcher:
-            input, target = input.to(device), target.to(device)
-            if mixup_fn is not None:
-                input, target = mixup_fn(input, target)
-        if args.channels_last:
+            if naflex_batch:
+                input = {k: v.to(device) if isinstance(v, torch.Tensor) else v
+                         for k, v in input.items()}
+                target = target.to(device)
+            else:
+                input, target = input.to(device), target.to(device)
+                if mixup_fn is not None:
+                    input, target = mixup_fn(input, target)
+        if args.channels_last and not naflex_batch:
             input = input.contiguous(memory_format=torch.channels_last)
 
         # multiply by accum steps to get equivalent for full update
@@ -774,18 +856,34 @@ def train_one_epoch(
                         )
                     optimizer.step()
 
+        batch_size = input['patches'].shape[0] if naflex_batch else input.size(0)
+
+        # variable-batch loss rescale (reference train.py:1334-1370): NaFlex
+        # batch sizes differ per bucket AND per rank, so the local loss is
+        # scaled to the reference batch size and re-balanced across ranks by
+        # an all-reduced global batch size.
+        loss_mult = 1.0
+        if naflex_batch:
+            scale_mode = args.naflex_loss_scale
+            if scale_mode and scale_mode != 'none':
+                loss_mult = batch_size / float(args.batch_size)
+                if scale_mode == 'sqrt':
+                    loss_mult = loss_mult ** 0.5
+            if args.distributed:
+                global_bs = utils.reduce_tensor(
+                    torch.tensor(batch_size, device=device, dtype=torch.float32), 1)
+                loss_mult = loss_mult * args.world_size * batch_size / global_bs.item()
+
         if has_no_sync and not need_update:
             with task.no_sync():
                 loss = _forward()
-                _backward(loss)
+                _backward(loss * loss_mult if loss_mult != 1.0 else loss)
         else:
             loss = _forward()
-            if need_update and loss_scaler is None and task._distributed_model is None:
-                pass
-            _backward(loss)
+            _backward(loss * loss_mult if loss_mult != 1.0 else loss)
 
-        losses_m.update(loss.item() * accum_steps, input.size(0))
-        update_sample_count += input.size(0)
+        losses_m.update(loss.item() * accum_steps, batch_size)
+        update_sample_count += batch_size
 
         if not need_update:
             data_start_time = time.time()
@@ -809,7 +907,7 @@ def train_one_epoch(
 
             if args.distributed:
                 reduced_loss = utils.reduce_tensor(loss.data, args.world_size)
-                losses_m.update(reduced_loss.item() * accum_steps, input.size(0))
+                losses_m.update(reduced_loss.item() * accum_steps, batch_size)
                 update_sample_count *= args.world_size
 
             if utils.is_primary(args):
@@ -860,10 +958,15 @@ def validate(
     with torch.no_grad():
         for batch_idx, (input, target) in enumerate(loader):
             last_batch = batch_idx == last_idx
+            naflex_batch = isinstance(input, dict)
             if not args.prefetcher:
-                input = input.to(device)
+                if naflex_batch:
+                    input = {k: v.to(device) if isinstance(v, torch.Tensor) else v
+                             for k, v in input.items()}
+                else:
+                    input = input.to(device)
                 target = target.to(device)
-            if args.channels_last:
+            if args.channels_last and not naflex_batch:
                 input = input.contiguous(memory_format=torch.channels_last)
 
             with amp_autocast():
@@ -884,7 +987,9 @@ def validate(
             if device.type == 'cuda':
                 torch.cuda.synchronize()
 
-            losses_m.update(reduced_loss.item(), input.size(0))
+            losses_m.update(
+                reduced_loss.item(),
+                input['patches'].shape[0] if naflex_batch else input.size(0))
             top1_m.update(acc1.item(), output.size(0))
             top5_m.update(acc5.item(), output.size(0))
 
