@@ -1,29 +1,23 @@
-"""Random Erasing — device-side, runs inside the PrefetchLoader on the side
-HIP stream (reference `timm/data/random_erasing.py:26`)."""
+"""Random Erasing (arxiv 1708.04896) — runs on-device inside the prefetcher,
+after normalization, on the side HIP stream.
+
+Behavioral parity: /root/reference/timm/data/random_erasing.py:26
+(const/rand/pixel fill modes, count splitting, AugMix split skipping).
+"""
 import math
 import random
 
 import torch
 
-
-def _get_pixels(per_pixel, rand_color, patch_size, dtype=torch.float32, device='cuda'):
-    # NOTE I've seen CUDA illegal memory access errors being caused by the normal_()
-    # paths, flip the order so normal is run on CPU if this becomes a problem
-    if per_pixel:
-        return torch.empty(patch_size, dtype=dtype, device=device).normal_()
-    elif rand_color:
-        return torch.empty((patch_size[0], 1, 1), dtype=dtype, device=device).normal_()
-    else:
-        return torch.zeros((patch_size[0], 1, 1), dtype=dtype, device=device)
+__all__ = ['RandomErasing']
 
 
 class RandomErasing:
-    """Randomly selects a rectangle region in an image and erases its pixels.
+    """Erase random rectangles from (normalized) image tensors.
 
-    'Random Erasing Data Augmentation' by Zhong et al. https://arxiv.org/pdf/1708.04896.pdf
-
-    This variant of RandomErasing is intended to be applied to either a batch
-    or single image tensor after it has been normalized by dataset mean and std.
+    Fill modes: 'const' zeros · 'rand' one normal draw per block channel ·
+    'pixel' per-pixel normal noise.  With num_splits > 1 the first (clean)
+    split of an AugMix batch is left untouched.
     """
 
     def __init__(
@@ -48,49 +42,49 @@ class RandomErasing:
         self.max_count = max_count or min_count
         self.num_splits = num_splits
         self.mode = mode.lower()
-        self.rand_color = False
-        self.per_pixel = False
-        if self.mode == 'rand':
-            self.rand_color = True  # per block random normal
-        elif self.mode == 'pixel':
-            self.per_pixel = True  # per pixel random normal
-        else:
-            assert not self.mode or self.mode == 'const'
+        assert self.mode in ('', 'const', 'rand', 'pixel')
         self.device = device
+
+    def _fill(self, shape, dtype):
+        chan, h, w = shape
+        if self.mode == 'pixel':
+            return torch.empty(shape, dtype=dtype, device=self.device).normal_()
+        if self.mode == 'rand':
+            return torch.empty((chan, 1, 1), dtype=dtype, device=self.device).normal_()
+        return torch.zeros((chan, 1, 1), dtype=dtype, device=self.device)
+
+    def _sample_box(self, img_h, img_w, count):
+        """One rejection-sampled (top, left, h, w) box, or None."""
+        for _ in range(10):
+            area = random.uniform(self.min_area, self.max_area) * img_h * img_w / count
+            aspect = math.exp(random.uniform(*self.log_aspect_ratio))
+            h = int(round(math.sqrt(area * aspect)))
+            w = int(round(math.sqrt(area / aspect)))
+            if w < img_w and h < img_h:
+                return random.randint(0, img_h - h), random.randint(0, img_w - w), h, w
+        return None
 
     def _erase(self, img, chan, img_h, img_w, dtype):
         if random.random() > self.probability:
             return
-        area = img_h * img_w
         count = self.min_count if self.min_count == self.max_count else \
             random.randint(self.min_count, self.max_count)
         for _ in range(count):
-            for attempt in range(10):
-                target_area = random.uniform(self.min_area, self.max_area) * area / count
-                aspect_ratio = math.exp(random.uniform(*self.log_aspect_ratio))
-                h = int(round(math.sqrt(target_area * aspect_ratio)))
-                w = int(round(math.sqrt(target_area / aspect_ratio)))
-                if w < img_w and h < img_h:
-                    top = random.randint(0, img_h - h)
-                    left = random.randint(0, img_w - w)
-                    img[:, top:top + h, left:left + w] = _get_pixels(
-                        self.per_pixel, self.rand_color, (chan, h, w),
-                        dtype=dtype, device=self.device)
-                    break
+            box = self._sample_box(img_h, img_w, count)
+            if box is not None:
+                top, left, h, w = box
+                img[:, top:top + h, left:left + w] = self._fill((chan, h, w), dtype)
 
-    def __call__(self, input):
-        if len(input.size()) == 3:
-            self._erase(input, *input.size(), input.dtype)
-        else:
-            batch_size, chan, img_h, img_w = input.size()
-            # skip first slice of batch if num_splits is set (for clean portion of samples)
-            batch_start = batch_size // self.num_splits if self.num_splits > 1 else 0
-            for i in range(batch_start, batch_size):
-                self._erase(input[i], chan, img_h, img_w, input.dtype)
-        return input
+    def __call__(self, x):
+        if x.ndim == 3:
+            self._erase(x, *x.shape, x.dtype)
+            return x
+        batch_size = x.shape[0]
+        start = batch_size // self.num_splits if self.num_splits > 1 else 0
+        for i in range(start, batch_size):
+            self._erase(x[i], *x.shape[1:], x.dtype)
+        return x
 
     def __repr__(self):
-        # NOTE simplified state for repr
-        fs = self.__class__.__name__ + f'(p={self.probability}, mode={self.mode}'
-        fs += f', count=({self.min_count}, {self.max_count}))'
-        return fs
+        return (f'{self.__class__.__name__}(p={self.probability}, mode={self.mode}'
+                f', count=({self.min_count}, {self.max_count}))')

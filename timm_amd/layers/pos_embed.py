@@ -1,12 +1,27 @@
-"""Absolute position embedding resample (reference `timm/layers/pos_embed.py:19`)."""
+"""Learned absolute position-embedding resampling (variable input sizes).
+
+Behavioral parity: /root/reference/timm/layers/pos_embed.py:19 (prefix-token
+split, fp32 bicubic+antialias interpolation, dtype round-trip).
+"""
 import logging
 import math
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.nn.functional as F
 
 _logger = logging.getLogger(__name__)
+
+__all__ = ['resample_abs_pos_embed', 'resample_abs_pos_embed_nhwc']
+
+
+def _interp_grid(flat: torch.Tensor, old_size, new_size, interpolation, antialias):
+    """[1, H*W, C] -> [1, H'*W', C] via NCHW interpolation in fp32."""
+    dtype = flat.dtype
+    dim = flat.shape[-1]
+    grid = flat.float().reshape(1, old_size[0], old_size[1], dim).permute(0, 3, 1, 2)
+    grid = F.interpolate(grid, size=new_size, mode=interpolation, antialias=antialias)
+    return grid.permute(0, 2, 3, 1).reshape(1, -1, dim).to(dtype)
 
 
 def resample_abs_pos_embed(
@@ -18,37 +33,27 @@ def resample_abs_pos_embed(
         antialias: bool = True,
         verbose: bool = False,
 ):
-    # sort out sizes, assume square if old size not provided
-    num_pos_tokens = posemb.shape[1]
-    num_new_tokens = new_size[0] * new_size[1] + num_prefix_tokens
-    if num_new_tokens == num_pos_tokens and new_size[0] == new_size[1]:
+    """Resample a [1, prefix + H*W, C] pos-embed table to a new grid.
+
+    Prefix (cls/reg) tokens pass through untouched; the grid is assumed
+    square when old_size is not given.
+    """
+    total = posemb.shape[1]
+    if total == new_size[0] * new_size[1] + num_prefix_tokens and new_size[0] == new_size[1]:
         return posemb
-
     if old_size is None:
-        hw = int(math.sqrt(num_pos_tokens - num_prefix_tokens))
-        old_size = hw, hw
+        side = int(math.sqrt(total - num_prefix_tokens))
+        old_size = side, side
 
-    # remove class token / prefix
-    if num_prefix_tokens:
-        posemb_prefix, posemb = posemb[:, :num_prefix_tokens], posemb[:, num_prefix_tokens:]
-    else:
-        posemb_prefix = None
-
-    embed_dim = posemb.shape[-1]
-    orig_dtype = posemb.dtype
-    posemb = posemb.float()
-    posemb = posemb.reshape(1, old_size[0], old_size[1], -1).permute(0, 3, 1, 2)
-    posemb = F.interpolate(posemb, size=new_size, mode=interpolation, antialias=antialias)
-    posemb = posemb.permute(0, 2, 3, 1).reshape(1, -1, embed_dim)
-    posemb = posemb.to(orig_dtype)
-
-    if posemb_prefix is not None:
-        posemb = torch.cat([posemb_prefix, posemb], dim=1)
+    prefix = posemb[:, :num_prefix_tokens] if num_prefix_tokens else None
+    grid = posemb[:, num_prefix_tokens:] if num_prefix_tokens else posemb
+    grid = _interp_grid(grid, old_size, new_size, interpolation, antialias)
+    if prefix is not None:
+        grid = torch.cat([prefix, grid], dim=1)
 
     if not torch.jit.is_scripting() and verbose:
         _logger.info(f'Resized position embedding: {old_size} to {new_size}.')
-
-    return posemb
+    return grid
 
 
 def resample_abs_pos_embed_nhwc(
@@ -58,16 +63,14 @@ def resample_abs_pos_embed_nhwc(
         antialias: bool = True,
         verbose: bool = False,
 ):
-    if new_size[0] == posemb.shape[-3] and new_size[1] == posemb.shape[-2]:
+    """Resample an NHWC-laid-out pos-embed [..., H, W, C] to a new grid."""
+    if tuple(new_size) == (posemb.shape[-3], posemb.shape[-2]):
         return posemb
-
-    orig_dtype = posemb.dtype
-    posemb = posemb.float()
-    posemb = posemb.reshape(1, posemb.shape[-3], posemb.shape[-2], posemb.shape[-1]).permute(0, 3, 1, 2)
-    posemb = F.interpolate(posemb, size=new_size, mode=interpolation, antialias=antialias)
-    posemb = posemb.permute(0, 2, 3, 1).to(orig_dtype)
-
+    dtype = posemb.dtype
+    grid = posemb.float().reshape(
+        1, posemb.shape[-3], posemb.shape[-2], posemb.shape[-1]).permute(0, 3, 1, 2)
+    grid = F.interpolate(grid, size=new_size, mode=interpolation, antialias=antialias)
+    out = grid.permute(0, 2, 3, 1).to(dtype)
     if not torch.jit.is_scripting() and verbose:
         _logger.info(f'Resized position embedding to {new_size}.')
-
-    return posemb
+    return out
