@@ -1,8 +1,8 @@
-"""ConvMixer — MI355X-native implementation.
+"""ConvMixer ("Patches Are All You Need?", arxiv 2201.09792).
 
-Capability parity with reference `timm/models/convmixer.py`: patch-embed conv
-stem, depth x (residual depthwise conv + pointwise conv) blocks (:27),
-convmixer_1536_20 / 768_32 / 1024_20_ks9_p14 variants.
+Behavioral parity: /root/reference/timm/models/convmixer.py (patch-conv stem,
+depth x [residual depthwise + pointwise] mixer blocks, checkpoint key layout
+via nn.Sequential indices).
 """
 from typing import Optional, Type
 
@@ -27,8 +27,21 @@ class Residual(nn.Module):
         return self.fn(x) + x
 
 
+def _conv_act_bn(conv: nn.Conv2d, act_layer: Type[nn.Module]) -> list:
+    """The repeating conv -> act -> BN triple (BN after act, ConvMixer style)."""
+    return [conv, act_layer(), nn.BatchNorm2d(conv.out_channels)]
+
+
+def _mixer_block(dim: int, kernel_size: int, act_layer: Type[nn.Module]) -> nn.Sequential:
+    """One block: residual 'same'-padded depthwise mix + pointwise channel mix."""
+    spatial_mix = Residual(nn.Sequential(
+        *_conv_act_bn(nn.Conv2d(dim, dim, kernel_size, groups=dim, padding='same'), act_layer)))
+    channel_mix = _conv_act_bn(nn.Conv2d(dim, dim, kernel_size=1), act_layer)
+    return nn.Sequential(spatial_mix, *channel_mix)
+
+
 class ConvMixer(nn.Module):
-    """ConvMixer (reference `convmixer.py:27`; paper: Patches Are All You Need?)."""
+    """Isotropic conv mixer over patch embeddings."""
 
     def __init__(
             self,
@@ -49,22 +62,10 @@ class ConvMixer(nn.Module):
         self.grad_checkpointing = False
 
         self.stem = nn.Sequential(
-            nn.Conv2d(in_chans, dim, kernel_size=patch_size, stride=patch_size),
-            act_layer(),
-            nn.BatchNorm2d(dim),
-        )
+            *_conv_act_bn(
+                nn.Conv2d(in_chans, dim, kernel_size=patch_size, stride=patch_size), act_layer))
         self.blocks = nn.Sequential(
-            *[nn.Sequential(
-                Residual(nn.Sequential(
-                    nn.Conv2d(dim, dim, kernel_size, groups=dim, padding='same'),
-                    act_layer(),
-                    nn.BatchNorm2d(dim),
-                )),
-                nn.Conv2d(dim, dim, kernel_size=1),
-                act_layer(),
-                nn.BatchNorm2d(dim),
-            ) for _ in range(depth)]
-        )
+            *[_mixer_block(dim, kernel_size, act_layer) for _ in range(depth)])
         self.pooling = SelectAdaptivePool2d(pool_type=global_pool, flatten=True)
         self.head_drop = nn.Dropout(drop_rate)
         self.head = nn.Linear(dim, num_classes) if num_classes > 0 else nn.Identity()
@@ -90,25 +91,19 @@ class ConvMixer(nn.Module):
     def forward_features(self, x: torch.Tensor) -> torch.Tensor:
         x = self.stem(x)
         if self.grad_checkpointing and not torch.jit.is_scripting():
-            x = checkpoint_seq(self.blocks, x)
-        else:
-            x = self.blocks(x)
-        return x
+            return checkpoint_seq(self.blocks, x)
+        return self.blocks(x)
 
     def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
-        x = self.pooling(x)
-        x = self.head_drop(x)
+        x = self.head_drop(self.pooling(x))
         return x if pre_logits else self.head(x)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.forward_features(x)
-        x = self.forward_head(x)
-        return x
+        return self.forward_head(self.forward_features(x))
 
 
 def _create_convmixer(variant, pretrained=False, **kwargs):
-    model = build_model_with_cfg(ConvMixer, variant, pretrained, **kwargs)
-    return model
+    return build_model_with_cfg(ConvMixer, variant, pretrained, **kwargs)
 
 
 def _cfg(url='', **kwargs):
@@ -130,17 +125,17 @@ default_cfgs = generate_default_cfgs({
 
 @register_model
 def convmixer_1536_20(pretrained=False, **kwargs) -> ConvMixer:
-    model_args = dict(dim=1536, depth=20, kernel_size=9, patch_size=7)
-    return _create_convmixer('convmixer_1536_20', pretrained, **dict(model_args, **kwargs))
+    args = dict(dim=1536, depth=20, kernel_size=9, patch_size=7)
+    return _create_convmixer('convmixer_1536_20', pretrained, **dict(args, **kwargs))
 
 
 @register_model
 def convmixer_768_32(pretrained=False, **kwargs) -> ConvMixer:
-    model_args = dict(dim=768, depth=32, kernel_size=7, patch_size=7, act_layer=nn.ReLU)
-    return _create_convmixer('convmixer_768_32', pretrained, **dict(model_args, **kwargs))
+    args = dict(dim=768, depth=32, kernel_size=7, patch_size=7, act_layer=nn.ReLU)
+    return _create_convmixer('convmixer_768_32', pretrained, **dict(args, **kwargs))
 
 
 @register_model
 def convmixer_1024_20_ks9_p14(pretrained=False, **kwargs) -> ConvMixer:
-    model_args = dict(dim=1024, depth=20, kernel_size=9, patch_size=14)
-    return _create_convmixer('convmixer_1024_20_ks9_p14', pretrained, **dict(model_args, **kwargs))
+    args = dict(dim=1024, depth=20, kernel_size=9, patch_size=14)
+    return _create_convmixer('convmixer_1024_20_ks9_p14', pretrained, **dict(args, **kwargs))
