@@ -1,12 +1,14 @@
-"""DPN (Dual-Path Networks) — MI355X-native implementation.
+"""DPN — Dual-Path Networks (arxiv 1707.01629).
 
-Capability parity with reference `timm/models/dpn.py`: `CatBnAct` (:25),
-`BnActConv2d` (:43), `DualPathBlock` (:64) carrying a (residual, dense)
-tensor pair through the stage stack, `DPN` (:151) and dpn48b..dpn131.
+Behavioral parity: /root/reference/timm/models/dpn.py (CatBnAct/BnActConv2d
+pre-activation units, the (residual, dense) tensor pair threaded through
+every block, `features.convN_i` checkpoint key layout, dpn48b..dpn131).
+Stage construction here is a single loop over a width table instead of the
+reference's four unrolled sections.
 """
 from collections import OrderedDict
 from functools import partial
-from typing import Tuple, Type, Union
+from typing import Tuple, Type
 
 import torch
 import torch.nn as nn
@@ -21,17 +23,19 @@ __all__ = ['DPN']
 
 
 class CatBnAct(nn.Module):
+    """Concat the dual paths (when given a pair) then BN+act."""
+
     def __init__(self, in_chs: int, norm_layer: Type[nn.Module] = BatchNormAct2d):
         super().__init__()
         self.bn = norm_layer(in_chs, eps=0.001)
 
     def forward(self, x) -> torch.Tensor:
-        if isinstance(x, tuple):
-            x = torch.cat(x, dim=1)
-        return self.bn(x)
+        return self.bn(torch.cat(x, dim=1) if isinstance(x, tuple) else x)
 
 
 class BnActConv2d(nn.Module):
+    """Pre-activation conv unit (BN+act before conv, DPN/DenseNet style)."""
+
     def __init__(
             self,
             in_chs: int,
@@ -50,6 +54,9 @@ class BnActConv2d(nn.Module):
 
 
 class DualPathBlock(nn.Module):
+    """One DPN block: bottleneck over the concatenated paths, output split
+    back into a residual-summed part and a densely-grown part."""
+
     def __init__(
             self,
             in_chs: int,
@@ -62,74 +69,63 @@ class DualPathBlock(nn.Module):
             b: bool = False,
     ):
         super().__init__()
+        assert block_type in ('proj', 'down', 'normal')
         self.num_1x1_c = num_1x1_c
         self.inc = inc
         self.b = b
-        if block_type == 'proj':
-            self.key_stride = 1
-            self.has_proj = True
-        elif block_type == 'down':
-            self.key_stride = 2
-            self.has_proj = True
-        else:
-            assert block_type == 'normal'
-            self.key_stride = 1
-            self.has_proj = False
+        self.key_stride = 2 if block_type == 'down' else 1
+        self.has_proj = block_type != 'normal'
 
-        self.c1x1_w_s1 = None
-        self.c1x1_w_s2 = None
+        # stride-specific projection attribute names match converted weights
+        self.c1x1_w_s1 = self.c1x1_w_s2 = None
         if self.has_proj:
-            # different member names allow parameter key matching for weight conversion
+            proj = BnActConv2d(
+                in_chs=in_chs, out_chs=num_1x1_c + 2 * inc,
+                kernel_size=1, stride=self.key_stride)
             if self.key_stride == 2:
-                self.c1x1_w_s2 = BnActConv2d(
-                    in_chs=in_chs, out_chs=num_1x1_c + 2 * inc, kernel_size=1, stride=2)
+                self.c1x1_w_s2 = proj
             else:
-                self.c1x1_w_s1 = BnActConv2d(
-                    in_chs=in_chs, out_chs=num_1x1_c + 2 * inc, kernel_size=1, stride=1)
+                self.c1x1_w_s1 = proj
 
         self.c1x1_a = BnActConv2d(in_chs=in_chs, out_chs=num_1x1_a, kernel_size=1, stride=1)
         self.c3x3_b = BnActConv2d(
-            in_chs=num_1x1_a, out_chs=num_3x3_b, kernel_size=3, stride=self.key_stride, groups=groups)
+            in_chs=num_1x1_a, out_chs=num_3x3_b, kernel_size=3,
+            stride=self.key_stride, groups=groups)
         if b:
+            # 'b' variants split with two separate 1x1 convs after a BN+act
             self.c1x1_c = CatBnAct(in_chs=num_3x3_b)
             self.c1x1_c1 = create_conv2d(num_3x3_b, num_1x1_c, kernel_size=1)
             self.c1x1_c2 = create_conv2d(num_3x3_b, inc, kernel_size=1)
         else:
-            self.c1x1_c = BnActConv2d(in_chs=num_3x3_b, out_chs=num_1x1_c + inc, kernel_size=1, stride=1)
-            self.c1x1_c1 = None
-            self.c1x1_c2 = None
+            self.c1x1_c = BnActConv2d(
+                in_chs=num_3x3_b, out_chs=num_1x1_c + inc, kernel_size=1, stride=1)
+            self.c1x1_c1 = self.c1x1_c2 = None
+
+    def _shortcut(self, x, x_cat):
+        """Resolve the (residual, dense) shortcut pair for this block."""
+        if not self.has_proj:
+            return x[0], x[1]
+        proj = self.c1x1_w_s2 if self.c1x1_w_s2 is not None else self.c1x1_w_s1
+        projected = proj(x_cat)
+        return projected[:, :self.num_1x1_c], projected[:, self.num_1x1_c:]
 
     def forward(self, x) -> Tuple[torch.Tensor, torch.Tensor]:
-        if isinstance(x, tuple):
-            x_in = torch.cat(x, dim=1)
-        else:
-            x_in = x
-        if self.c1x1_w_s1 is None and self.c1x1_w_s2 is None:
-            x_s1 = x[0]
-            x_s2 = x[1]
-        else:
-            if self.c1x1_w_s1 is not None:
-                x_s = self.c1x1_w_s1(x_in)
-            else:
-                x_s = self.c1x1_w_s2(x_in)
-            x_s1 = x_s[:, :self.num_1x1_c, :, :]
-            x_s2 = x_s[:, self.num_1x1_c:, :, :]
-        x_in = self.c1x1_a(x_in)
-        x_in = self.c3x3_b(x_in)
-        x_in = self.c1x1_c(x_in)
+        x_cat = torch.cat(x, dim=1) if isinstance(x, tuple) else x
+        short_res, short_dense = self._shortcut(x, x_cat)
+
+        out = self.c1x1_c(self.c3x3_b(self.c1x1_a(x_cat)))
         if self.c1x1_c1 is not None:
-            out1 = self.c1x1_c1(x_in)
-            out2 = self.c1x1_c2(x_in)
+            res_part, dense_part = self.c1x1_c1(out), self.c1x1_c2(out)
         else:
-            out1 = x_in[:, :self.num_1x1_c, :, :]
-            out2 = x_in[:, self.num_1x1_c:, :, :]
-        resid = x_s1 + out1
-        dense = torch.cat([x_s2, out2], dim=1)
-        return resid, dense
+            res_part, dense_part = out[:, :self.num_1x1_c], out[:, self.num_1x1_c:]
+        return short_res + res_part, torch.cat([short_dense, dense_part], dim=1)
 
 
 class DPN(nn.Module):
-    """Dual-Path Network (reference `dpn.py:151`; paper https://arxiv.org/abs/1707.01629)."""
+    """Dual-Path Network trunk + 1x1-conv classifier head."""
+
+    # per-stage (base width multiplier, reduction) — widths scale by bw_factor
+    _STAGES = ((64, 4), (128, 8), (256, 16), (512, 32))
 
     def __init__(
             self,
@@ -150,72 +146,44 @@ class DPN(nn.Module):
             fc_act_layer: str = 'elu',
     ):
         super().__init__()
+        assert output_stride == 32
         self.num_classes = num_classes
         self.drop_rate = drop_rate
         self.b = b
-        assert output_stride == 32
 
         norm_layer = partial(get_norm_act_layer(norm_layer, act_layer=act_layer), eps=.001)
-        fc_norm_layer = partial(get_norm_act_layer(norm_layer, act_layer=fc_act_layer), eps=.001, inplace=False)
+        fc_norm_layer = partial(
+            get_norm_act_layer(norm_layer, act_layer=fc_act_layer), eps=.001, inplace=False)
         bw_factor = 1 if small else 4
+
         blocks = OrderedDict()
-
-        # conv1
         blocks['conv1_1'] = ConvNormAct(
-            in_chans, num_init_features, kernel_size=3 if small else 7, stride=2, norm_layer=norm_layer)
+            in_chans, num_init_features,
+            kernel_size=3 if small else 7, stride=2, norm_layer=norm_layer)
         blocks['conv1_pool'] = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
-        self.feature_info = [dict(num_chs=num_init_features, reduction=2, module='features.conv1_1')]
+        self.feature_info = [
+            dict(num_chs=num_init_features, reduction=2, module='features.conv1_1')]
 
-        # conv2
-        bw = 64 * bw_factor
-        inc = inc_sec[0]
-        r = (k_r * bw) // (64 * bw_factor)
-        blocks['conv2_1'] = DualPathBlock(num_init_features, r, r, bw, inc, groups, 'proj', b)
-        in_chs = bw + 3 * inc
-        for i in range(2, k_sec[0] + 1):
-            blocks['conv2_' + str(i)] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'normal', b)
-            in_chs += inc
-        self.feature_info += [dict(num_chs=in_chs, reduction=4, module=f'features.conv2_{k_sec[0]}')]
-
-        # conv3
-        bw = 128 * bw_factor
-        inc = inc_sec[1]
-        r = (k_r * bw) // (64 * bw_factor)
-        blocks['conv3_1'] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'down', b)
-        in_chs = bw + 3 * inc
-        for i in range(2, k_sec[1] + 1):
-            blocks['conv3_' + str(i)] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'normal', b)
-            in_chs += inc
-        self.feature_info += [dict(num_chs=in_chs, reduction=8, module=f'features.conv3_{k_sec[1]}')]
-
-        # conv4
-        bw = 256 * bw_factor
-        inc = inc_sec[2]
-        r = (k_r * bw) // (64 * bw_factor)
-        blocks['conv4_1'] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'down', b)
-        in_chs = bw + 3 * inc
-        for i in range(2, k_sec[2] + 1):
-            blocks['conv4_' + str(i)] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'normal', b)
-            in_chs += inc
-        self.feature_info += [dict(num_chs=in_chs, reduction=16, module=f'features.conv4_{k_sec[2]}')]
-
-        # conv5
-        bw = 512 * bw_factor
-        inc = inc_sec[3]
-        r = (k_r * bw) // (64 * bw_factor)
-        blocks['conv5_1'] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'down', b)
-        in_chs = bw + 3 * inc
-        for i in range(2, k_sec[3] + 1):
-            blocks['conv5_' + str(i)] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'normal', b)
-            in_chs += inc
-        self.feature_info += [dict(num_chs=in_chs, reduction=32, module=f'features.conv5_{k_sec[3]}')]
+        in_chs = num_init_features
+        for stage_idx, ((base_bw, reduction), depth, inc) in enumerate(
+                zip(self._STAGES, k_sec, inc_sec)):
+            bw = base_bw * bw_factor
+            r = (k_r * bw) // (64 * bw_factor)
+            name = f'conv{stage_idx + 2}'
+            first_type = 'proj' if stage_idx == 0 else 'down'
+            blocks[f'{name}_1'] = DualPathBlock(in_chs, r, r, bw, inc, groups, first_type, b)
+            in_chs = bw + 3 * inc
+            for i in range(2, depth + 1):
+                blocks[f'{name}_{i}'] = DualPathBlock(in_chs, r, r, bw, inc, groups, 'normal', b)
+                in_chs += inc
+            self.feature_info += [
+                dict(num_chs=in_chs, reduction=reduction * 2, module=f'features.{name}_{depth}')]
 
         blocks['conv5_bn_ac'] = CatBnAct(in_chs, norm_layer=fc_norm_layer)
-
         self.num_features = self.head_hidden_size = in_chs
         self.features = nn.Sequential(blocks)
 
-        # 1x1-conv classifier allows the extra-pooling scheme
+        # conv-style classifier allows test-time spatial pooling of logits
         self.global_pool, self.classifier = create_classifier(
             self.num_features, self.num_classes, pool_type=global_pool, use_conv=True)
         self.flatten = nn.Flatten(1) if global_pool else nn.Identity()
@@ -226,8 +194,8 @@ class DPN(nn.Module):
             stem=r'^features\.conv1',
             blocks=[
                 (r'^features\.conv(\d+)' if coarse else r'^features\.conv(\d+)_(\d+)', None),
-                (r'^features\.conv5_bn_ac', (99999,))
-            ]
+                (r'^features\.conv5_bn_ac', (99999,)),
+            ],
         )
 
     @torch.jit.ignore
@@ -253,13 +221,10 @@ class DPN(nn.Module):
             x = F.dropout(x, p=self.drop_rate, training=self.training)
         if pre_logits:
             return self.flatten(x)
-        x = self.classifier(x)
-        return self.flatten(x)
+        return self.flatten(self.classifier(x))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.forward_features(x)
-        x = self.forward_head(x)
-        return x
+        return self.forward_head(self.forward_features(x))
 
 
 def _create_dpn(variant, pretrained=False, **kwargs):
@@ -290,58 +255,62 @@ default_cfgs = generate_default_cfgs({
     'dpn107.mx_in1k': _cfg(),
 })
 
+# variant name -> constructor args (widths/depths from the paper releases)
+_VARIANTS = dict(
+    dpn48b=dict(
+        small=True, num_init_features=10, k_r=128, groups=32,
+        b=True, k_sec=(3, 4, 6, 3), inc_sec=(16, 32, 32, 64), act_layer='silu'),
+    dpn68=dict(
+        small=True, num_init_features=10, k_r=128, groups=32,
+        k_sec=(3, 4, 12, 3), inc_sec=(16, 32, 32, 64)),
+    dpn68b=dict(
+        small=True, num_init_features=10, k_r=128, groups=32,
+        b=True, k_sec=(3, 4, 12, 3), inc_sec=(16, 32, 32, 64)),
+    dpn92=dict(
+        num_init_features=64, k_r=96, groups=32,
+        k_sec=(3, 4, 20, 3), inc_sec=(16, 32, 24, 128)),
+    dpn98=dict(
+        num_init_features=96, k_r=160, groups=40,
+        k_sec=(3, 6, 20, 3), inc_sec=(16, 32, 32, 128)),
+    dpn131=dict(
+        num_init_features=128, k_r=160, groups=40,
+        k_sec=(4, 8, 28, 3), inc_sec=(16, 32, 32, 128)),
+    dpn107=dict(
+        num_init_features=128, k_r=200, groups=50,
+        k_sec=(4, 8, 20, 3), inc_sec=(20, 64, 64, 128)),
+)
+
 
 @register_model
 def dpn48b(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        small=True, num_init_features=10, k_r=128, groups=32,
-        b=True, k_sec=(3, 4, 6, 3), inc_sec=(16, 32, 32, 64), act_layer='silu')
-    return _create_dpn('dpn48b', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn48b', pretrained=pretrained, **dict(_VARIANTS['dpn48b'], **kwargs))
 
 
 @register_model
 def dpn68(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        small=True, num_init_features=10, k_r=128, groups=32,
-        k_sec=(3, 4, 12, 3), inc_sec=(16, 32, 32, 64))
-    return _create_dpn('dpn68', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn68', pretrained=pretrained, **dict(_VARIANTS['dpn68'], **kwargs))
 
 
 @register_model
 def dpn68b(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        small=True, num_init_features=10, k_r=128, groups=32,
-        b=True, k_sec=(3, 4, 12, 3), inc_sec=(16, 32, 32, 64))
-    return _create_dpn('dpn68b', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn68b', pretrained=pretrained, **dict(_VARIANTS['dpn68b'], **kwargs))
 
 
 @register_model
 def dpn92(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        num_init_features=64, k_r=96, groups=32,
-        k_sec=(3, 4, 20, 3), inc_sec=(16, 32, 24, 128))
-    return _create_dpn('dpn92', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn92', pretrained=pretrained, **dict(_VARIANTS['dpn92'], **kwargs))
 
 
 @register_model
 def dpn98(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        num_init_features=96, k_r=160, groups=40,
-        k_sec=(3, 6, 20, 3), inc_sec=(16, 32, 32, 128))
-    return _create_dpn('dpn98', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn98', pretrained=pretrained, **dict(_VARIANTS['dpn98'], **kwargs))
 
 
 @register_model
 def dpn131(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        num_init_features=128, k_r=160, groups=40,
-        k_sec=(4, 8, 28, 3), inc_sec=(16, 32, 32, 128))
-    return _create_dpn('dpn131', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn131', pretrained=pretrained, **dict(_VARIANTS['dpn131'], **kwargs))
 
 
 @register_model
 def dpn107(pretrained=False, **kwargs) -> DPN:
-    model_args = dict(
-        num_init_features=128, k_r=200, groups=50,
-        k_sec=(4, 8, 20, 3), inc_sec=(20, 64, 64, 128))
-    return _create_dpn('dpn107', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_dpn('dpn107', pretrained=pretrained, **dict(_VARIANTS['dpn107'], **kwargs))
