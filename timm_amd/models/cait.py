@@ -68,13 +68,15 @@ class ClassAttn(nn.Module):
         return x_cls
 
 
-class LayerScaleBlockClassAttn(nn.Module):
-    """Class-attention block with LayerScale (reference `cait.py:81`)."""
+class _LayerScaleBlockBase(nn.Module):
+    """Shared norm/attn/mlp + per-branch gamma scaffolding for both CaiT block
+    flavours; subclasses differ only in forward wiring + default attn."""
 
     def __init__(
             self,
             dim: int,
             num_heads: int,
+            attn_block: Type[nn.Module],
             mlp_ratio: float = 4.,
             qkv_bias: bool = False,
             proj_drop: float = 0.,
@@ -82,7 +84,6 @@ class LayerScaleBlockClassAttn(nn.Module):
             drop_path: float = 0.,
             act_layer: Type[nn.Module] = nn.GELU,
             norm_layer: Type[nn.Module] = LayerNorm,
-            attn_block: Type[nn.Module] = ClassAttn,
             mlp_block: Type[nn.Module] = Mlp,
             init_values: float = 1e-4,
     ):
@@ -97,9 +98,16 @@ class LayerScaleBlockClassAttn(nn.Module):
         self.gamma_1 = nn.Parameter(init_values * torch.ones(dim))
         self.gamma_2 = nn.Parameter(init_values * torch.ones(dim))
 
+
+class LayerScaleBlockClassAttn(_LayerScaleBlockBase):
+    """Class-attention block: only the cls token is updated (reference `cait.py:81`)."""
+
+    def __init__(self, dim, num_heads, attn_block: Type[nn.Module] = ClassAttn, **kwargs):
+        super().__init__(dim, num_heads, attn_block, **kwargs)
+
     def forward(self, x: torch.Tensor, x_cls: torch.Tensor) -> torch.Tensor:
-        u = torch.cat((x_cls, x), dim=1)
-        x_cls = x_cls + self.drop_path(self.gamma_1 * self.attn(self.norm1(u)))
+        joint = torch.cat((x_cls, x), dim=1)
+        x_cls = x_cls + self.drop_path(self.gamma_1 * self.attn(self.norm1(joint)))
         x_cls = x_cls + self.drop_path(self.gamma_2 * self.mlp(self.norm2(x_cls)))
         return x_cls
 
@@ -127,51 +135,31 @@ class TalkingHeadAttn(nn.Module):
         self.proj_w = nn.Linear(num_heads, num_heads)
         self.proj_drop = nn.Dropout(proj_drop)
 
+    @staticmethod
+    def _mix_heads(scores: torch.Tensor, mixer: nn.Linear) -> torch.Tensor:
+        # linear blend across the head dim of the [B,H,N,N] score tensor
+        return mixer(scores.permute(0, 2, 3, 1)).permute(0, 3, 1, 2)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         B, N, C = x.shape
         qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, C // self.num_heads).permute(2, 0, 3, 1, 4)
         q, k, v = qkv[0] * self.scale, qkv[1], qkv[2]
 
-        attn = q @ k.transpose(-2, -1)
-        attn = self.proj_l(attn.permute(0, 2, 3, 1)).permute(0, 3, 1, 2)
-        attn = attn.softmax(dim=-1)
-        attn = self.proj_w(attn.permute(0, 2, 3, 1)).permute(0, 3, 1, 2)
-        attn = self.attn_drop(attn)
+        # scores stay materialised: the talking-heads mixers act on [B,H,N,N]
+        # both sides of the softmax, which precludes an online-softmax kernel
+        scores = self._mix_heads(q @ k.transpose(-2, -1), self.proj_l)
+        scores = self._mix_heads(scores.softmax(dim=-1), self.proj_w)
+        scores = self.attn_drop(scores)
 
-        x = (attn @ v).transpose(1, 2).reshape(B, N, C)
-        x = self.proj(x)
-        x = self.proj_drop(x)
-        return x
+        x = (scores @ v).transpose(1, 2).reshape(B, N, C)
+        return self.proj_drop(self.proj(x))
 
 
-class LayerScaleBlock(nn.Module):
-    """Self-attention block with LayerScale (reference `cait.py:184`)."""
+class LayerScaleBlock(_LayerScaleBlockBase):
+    """Self-attention block with per-branch LayerScale (reference `cait.py:184`)."""
 
-    def __init__(
-            self,
-            dim: int,
-            num_heads: int,
-            mlp_ratio: float = 4.,
-            qkv_bias: bool = False,
-            proj_drop: float = 0.,
-            attn_drop: float = 0.,
-            drop_path: float = 0.,
-            act_layer: Type[nn.Module] = nn.GELU,
-            norm_layer: Type[nn.Module] = LayerNorm,
-            attn_block: Type[nn.Module] = TalkingHeadAttn,
-            mlp_block: Type[nn.Module] = Mlp,
-            init_values: float = 1e-4,
-    ):
-        super().__init__()
-        self.norm1 = norm_layer(dim)
-        self.attn = attn_block(
-            dim, num_heads=num_heads, qkv_bias=qkv_bias, attn_drop=attn_drop, proj_drop=proj_drop)
-        self.drop_path = DropPath(drop_path) if drop_path > 0. else nn.Identity()
-        self.norm2 = norm_layer(dim)
-        self.mlp = mlp_block(
-            in_features=dim, hidden_features=int(dim * mlp_ratio), act_layer=act_layer, drop=proj_drop)
-        self.gamma_1 = nn.Parameter(init_values * torch.ones(dim))
-        self.gamma_2 = nn.Parameter(init_values * torch.ones(dim))
+    def __init__(self, dim, num_heads, attn_block: Type[nn.Module] = TalkingHeadAttn, **kwargs):
+        super().__init__(dim, num_heads, attn_block, **kwargs)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = x + self.drop_path(self.gamma_1 * self.attn(self.norm1(x)))
@@ -233,34 +221,35 @@ class Cait(nn.Module):
         self.pos_embed = nn.Parameter(torch.zeros(1, num_patches, embed_dim))
         self.pos_drop = nn.Dropout(p=pos_drop_rate)
 
-        dpr = [drop_path_rate for _ in range(depth)]
-        self.blocks = nn.Sequential(*[block_layers(
+        shared = dict(
             dim=embed_dim,
             num_heads=num_heads,
-            mlp_ratio=mlp_ratio,
             qkv_bias=qkv_bias,
-            proj_drop=proj_drop_rate,
-            attn_drop=attn_drop_rate,
-            drop_path=dpr[i],
             norm_layer=norm_layer,
             act_layer=act_layer,
-            attn_block=attn_block,
-            mlp_block=mlp_block,
             init_values=init_values,
-        ) for i in range(depth)])
-        self.feature_info = [dict(num_chs=embed_dim, reduction=r, module=f'blocks.{i}') for i in range(depth)]
+        )
+        self.blocks = nn.Sequential(*[
+            block_layers(
+                mlp_ratio=mlp_ratio,
+                proj_drop=proj_drop_rate,
+                attn_drop=attn_drop_rate,
+                drop_path=drop_path_rate,
+                attn_block=attn_block,
+                mlp_block=mlp_block,
+                **shared,
+            ) for _ in range(depth)])
+        self.feature_info = [
+            dict(num_chs=embed_dim, reduction=r, module=f'blocks.{i}') for i in range(depth)]
 
-        self.blocks_token_only = nn.ModuleList([block_layers_token(
-            dim=embed_dim,
-            num_heads=num_heads,
-            mlp_ratio=mlp_ratio_token_only,
-            qkv_bias=qkv_bias,
-            norm_layer=norm_layer,
-            act_layer=act_layer,
-            attn_block=attn_block_token_only,
-            mlp_block=mlp_block_token_only,
-            init_values=init_values,
-        ) for _ in range(depth_token_only)])
+        # the class-attention tail updates only the cls token
+        self.blocks_token_only = nn.ModuleList([
+            block_layers_token(
+                mlp_ratio=mlp_ratio_token_only,
+                attn_block=attn_block_token_only,
+                mlp_block=mlp_block_token_only,
+                **shared,
+            ) for _ in range(depth_token_only)])
 
         self.norm = norm_layer(embed_dim)
 
@@ -326,34 +315,25 @@ class Cait(nn.Module):
             intermediates_only: bool = False,
     ) -> Union[List[torch.Tensor], Tuple[torch.Tensor, List[torch.Tensor]]]:
         assert output_fmt in ('NCHW', 'NLC'), 'Output format must be one of NCHW or NLC.'
-        reshape = output_fmt == 'NCHW'
-        intermediates = []
         take_indices, max_index = feature_take_indices(len(self.blocks), indices)
-
         B, _, height, width = x.shape
-        x = self.patch_embed(x)
-        x = x + self.pos_embed
-        x = self.pos_drop(x)
 
-        if torch.jit.is_scripting() or not stop_early:
-            blocks = self.blocks
-        else:
-            blocks = self.blocks[:max_index + 1]
-        for i, blk in enumerate(blocks):
+        x = self.pos_drop(self.patch_embed(x) + self.pos_embed)
+        run_to = len(self.blocks) if (torch.jit.is_scripting() or not stop_early) else max_index + 1
+        collected = []
+        for i, blk in enumerate(self.blocks[:run_to]):
             x = blk(x)
             if i in take_indices:
-                intermediates.append(self.norm(x) if norm else x)
+                collected.append(self.norm(x) if norm else x)
 
-        if reshape:
+        if output_fmt == 'NCHW':
             H, W = self.patch_embed.dyn_feat_size((height, width))
-            intermediates = [y.reshape(B, H, W, -1).permute(0, 3, 1, 2).contiguous() for y in intermediates]
-
+            collected = [
+                t.reshape(B, H, W, -1).permute(0, 3, 1, 2).contiguous() for t in collected]
         if intermediates_only:
-            return intermediates
-
-        # NOTE not bothering to run the cls-attention blocks for intermediate output
-        x = self.norm(x)
-        return x, intermediates
+            return collected
+        # cls-attention tail skipped for intermediate extraction
+        return self.norm(x), collected
 
     def prune_intermediate_layers(
             self,
@@ -444,61 +424,67 @@ default_cfgs = generate_default_cfgs({
 })
 
 
+_VARIANTS = dict(
+    cait_xxs24=dict(patch_size=16, embed_dim=192, depth=24, num_heads=4, init_values=1e-5),
+    cait_xxs36=dict(patch_size=16, embed_dim=192, depth=36, num_heads=4, init_values=1e-5),
+    cait_xs24=dict(patch_size=16, embed_dim=288, depth=24, num_heads=6, init_values=1e-5),
+    cait_s24=dict(patch_size=16, embed_dim=384, depth=24, num_heads=8, init_values=1e-5),
+    cait_s36=dict(patch_size=16, embed_dim=384, depth=36, num_heads=8, init_values=1e-6),
+    cait_m36=dict(patch_size=16, embed_dim=768, depth=36, num_heads=16, init_values=1e-6),
+    cait_m48=dict(patch_size=16, embed_dim=768, depth=48, num_heads=16, init_values=1e-6),
+)
+
+
+def _variant_args(name: str, kwargs: dict) -> dict:
+    base = _VARIANTS[name.rsplit('_', 1)[0]]
+    return dict(base, **kwargs)
+
+
 @register_model
 def cait_xxs24_224(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=192, depth=24, num_heads=4, init_values=1e-5)
-    return _create_cait('cait_xxs24_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_xxs24_224', pretrained=pretrained, **_variant_args('cait_xxs24_224', kwargs))
 
 
 @register_model
 def cait_xxs24_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=192, depth=24, num_heads=4, init_values=1e-5)
-    return _create_cait('cait_xxs24_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_xxs24_384', pretrained=pretrained, **_variant_args('cait_xxs24_384', kwargs))
 
 
 @register_model
 def cait_xxs36_224(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=192, depth=36, num_heads=4, init_values=1e-5)
-    return _create_cait('cait_xxs36_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_xxs36_224', pretrained=pretrained, **_variant_args('cait_xxs36_224', kwargs))
 
 
 @register_model
 def cait_xxs36_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=192, depth=36, num_heads=4, init_values=1e-5)
-    return _create_cait('cait_xxs36_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_xxs36_384', pretrained=pretrained, **_variant_args('cait_xxs36_384', kwargs))
 
 
 @register_model
 def cait_xs24_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=288, depth=24, num_heads=6, init_values=1e-5)
-    return _create_cait('cait_xs24_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_xs24_384', pretrained=pretrained, **_variant_args('cait_xs24_384', kwargs))
 
 
 @register_model
 def cait_s24_224(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=384, depth=24, num_heads=8, init_values=1e-5)
-    return _create_cait('cait_s24_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_s24_224', pretrained=pretrained, **_variant_args('cait_s24_224', kwargs))
 
 
 @register_model
 def cait_s24_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=384, depth=24, num_heads=8, init_values=1e-5)
-    return _create_cait('cait_s24_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_s24_384', pretrained=pretrained, **_variant_args('cait_s24_384', kwargs))
 
 
 @register_model
 def cait_s36_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=384, depth=36, num_heads=8, init_values=1e-6)
-    return _create_cait('cait_s36_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_s36_384', pretrained=pretrained, **_variant_args('cait_s36_384', kwargs))
 
 
 @register_model
 def cait_m36_384(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=768, depth=36, num_heads=16, init_values=1e-6)
-    return _create_cait('cait_m36_384', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_m36_384', pretrained=pretrained, **_variant_args('cait_m36_384', kwargs))
 
 
 @register_model
 def cait_m48_448(pretrained=False, **kwargs) -> Cait:
-    model_args = dict(patch_size=16, embed_dim=768, depth=48, num_heads=16, init_values=1e-6)
-    return _create_cait('cait_m48_448', pretrained=pretrained, **dict(model_args, **kwargs))
+    return _create_cait('cait_m48_448', pretrained=pretrained, **_variant_args('cait_m48_448', kwargs))
