@@ -36,26 +36,29 @@ class PositionalEncodingFourier(nn.Module):
         self.hidden_dim = hidden_dim
         self.dim = dim
 
+    def _axis_phase(self, n: int, device) -> torch.Tensor:
+        """Normalized 1..n axis positions scaled to [~0, 2pi]."""
+        coords = torch.arange(1, n + 1, dtype=torch.float32, device=device)
+        return coords / (n + 1e-6) * self.scale
+
+    def _sincos(self, phase: torch.Tensor, dim_t: torch.Tensor) -> torch.Tensor:
+        # interleave sin/cos over frequency pairs -> [..., hidden_dim]
+        freq = phase[..., None] / dim_t
+        return torch.stack((freq[..., 0::2].sin(), freq[..., 1::2].cos()), dim=-1).flatten(-2)
+
     def forward(self, shape: Tuple[int, int, int]) -> torch.Tensor:
+        B, H, W = shape
         device = self.token_projection.weight.device
         dtype = self.token_projection.weight.dtype
-        inv_mask = ~torch.zeros(shape).to(device=device, dtype=torch.bool)
-        y_embed = inv_mask.cumsum(1, dtype=torch.float32)
-        x_embed = inv_mask.cumsum(2, dtype=torch.float32)
-        eps = 1e-6
-        y_embed = y_embed / (y_embed[:, -1:, :] + eps) * self.scale
-        x_embed = x_embed / (x_embed[:, :, -1:] + eps) * self.scale
-
         dim_t = torch.arange(self.hidden_dim, dtype=torch.int64, device=device).to(torch.float32)
         dim_t = self.temperature ** (2 * torch.div(dim_t, 2, rounding_mode='floor') / self.hidden_dim)
 
-        pos_x = x_embed[:, :, :, None] / dim_t
-        pos_y = y_embed[:, :, :, None] / dim_t
-        pos_x = torch.stack((pos_x[:, :, :, 0::2].sin(), pos_x[:, :, :, 1::2].cos()), dim=4).flatten(3)
-        pos_y = torch.stack((pos_y[:, :, :, 0::2].sin(), pos_y[:, :, :, 1::2].cos()), dim=4).flatten(3)
+        pos_y = self._sincos(
+            self._axis_phase(H, device).view(1, H, 1).expand(B, H, W), dim_t)
+        pos_x = self._sincos(
+            self._axis_phase(W, device).view(1, 1, W).expand(B, H, W), dim_t)
         pos = torch.cat((pos_y, pos_x), dim=3).permute(0, 3, 1, 2)
-        pos = self.token_projection(pos.to(dtype))
-        return pos
+        return self.token_projection(pos.to(dtype))
 
 
 class ConvBlock(nn.Module):
@@ -89,17 +92,16 @@ class ConvBlock(nn.Module):
         shortcut = x
         x = self.conv_dw(x)
         if self.shortcut_after_dw:
+            # stride/width change: residual taps the dw output instead
             shortcut = x
+        return shortcut + self.drop_path(self._mlp_nchw(x))
 
-        x = x.permute(0, 2, 3, 1)  # NCHW -> NHWC
-        x = self.norm(x)
-        x = self.mlp(x)
+    def _mlp_nchw(self, x: torch.Tensor) -> torch.Tensor:
+        """norm + MLP + layer-scale computed channels-last, IO channels-first."""
+        x = self.mlp(self.norm(x.permute(0, 2, 3, 1)))
         if self.gamma is not None:
             x = self.gamma * x
-        x = x.permute(0, 3, 1, 2)  # NHWC -> NCHW
-
-        x = shortcut + self.drop_path(x)
-        return x
+        return x.permute(0, 3, 1, 2)
 
 
 class CrossCovarianceAttn(nn.Module):
@@ -185,38 +187,36 @@ class SplitTransposeBlock(nn.Module):
         self.gamma = nn.Parameter(ls_init_value * torch.ones(dim)) if ls_init_value > 0 else None
         self.drop_path = DropPath(drop_path) if drop_path > 0. else nn.Identity()
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
-        shortcut = x
-
-        spx = x.chunk(len(self.convs) + 1, dim=1)
-        spo = []
-        sp = spx[0]
+    def _multi_scale_mix(self, x: torch.Tensor) -> torch.Tensor:
+        """Res2Net-style cascaded depthwise mixing over channel splits."""
+        splits = x.chunk(len(self.convs) + 1, dim=1)
+        outs = []
+        acc = splits[0]
         for i, conv in enumerate(self.convs):
             if i > 0:
-                sp = sp + spx[i]
-            sp = conv(sp)
-            spo.append(sp)
-        spo.append(spx[-1])
-        x = torch.cat(spo, 1)
+                acc = acc + splits[i]
+            acc = conv(acc)
+            outs.append(acc)
+        outs.append(splits[-1])
+        return torch.cat(outs, 1)
 
-        # XCA
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        shortcut = x
+        x = self._multi_scale_mix(x)
+
+        # cross-covariance attention over flattened tokens
         B, C, H, W = x.shape
-        x = x.reshape(B, C, H * W).permute(0, 2, 1)
+        tokens = x.reshape(B, C, H * W).permute(0, 2, 1)
         if self.pos_embd is not None:
-            pos_encoding = self.pos_embd((B, H, W)).reshape(B, -1, x.shape[1]).permute(0, 2, 1)
-            x = x + pos_encoding
-        x = x + self.drop_path(self.gamma_xca * self.xca(self.norm_xca(x)))
-        x = x.reshape(B, H, W, C)
+            pos = self.pos_embd((B, H, W)).reshape(B, -1, tokens.shape[1]).permute(0, 2, 1)
+            tokens = tokens + pos
+        tokens = tokens + self.drop_path(self.gamma_xca * self.xca(self.norm_xca(tokens)))
 
-        # Inverted bottleneck
-        x = self.norm(x)
-        x = self.mlp(x)
+        # inverted-bottleneck MLP, channels-last
+        y = self.mlp(self.norm(tokens.reshape(B, H, W, C)))
         if self.gamma is not None:
-            x = self.gamma * x
-        x = x.permute(0, 3, 1, 2)
-
-        x = shortcut + self.drop_path(x)
-        return x
+            y = self.gamma * y
+        return shortcut + self.drop_path(y.permute(0, 3, 1, 2))
 
 
 class EdgeNeXtStage(nn.Module):
@@ -253,38 +253,35 @@ class EdgeNeXtStage(nn.Module):
             )
             in_chs = out_chs
 
+        # local ConvNeXt-style blocks first, global (XCA) blocks at the tail
+        common = dict(
+            expand_ratio=expand_ratio,
+            conv_bias=conv_bias,
+            ls_init_value=ls_init_value,
+            norm_layer=norm_layer_cl,
+            act_layer=act_layer,
+        )
         stage_blocks = []
         for i in range(depth):
             if i < depth - num_global_blocks:
-                stage_blocks.append(
-                    ConvBlock(
-                        dim=in_chs,
-                        dim_out=out_chs,
-                        stride=stride if downsample_block and i == 0 else 1,
-                        conv_bias=conv_bias,
-                        kernel_size=kernel_size,
-                        expand_ratio=expand_ratio,
-                        ls_init_value=ls_init_value,
-                        drop_path=drop_path_rates[i],
-                        norm_layer=norm_layer_cl,
-                        act_layer=act_layer,
-                    )
+                block = ConvBlock(
+                    dim=in_chs,
+                    dim_out=out_chs,
+                    stride=stride if downsample_block and i == 0 else 1,
+                    kernel_size=kernel_size,
+                    drop_path=drop_path_rates[i],
+                    **common,
                 )
             else:
-                stage_blocks.append(
-                    SplitTransposeBlock(
-                        dim=in_chs,
-                        num_scales=scales,
-                        num_heads=num_heads,
-                        expand_ratio=expand_ratio,
-                        use_pos_emb=use_pos_emb,
-                        conv_bias=conv_bias,
-                        ls_init_value=ls_init_value,
-                        drop_path=drop_path_rates[i],
-                        norm_layer=norm_layer_cl,
-                        act_layer=act_layer,
-                    )
+                block = SplitTransposeBlock(
+                    dim=in_chs,
+                    num_scales=scales,
+                    num_heads=num_heads,
+                    use_pos_emb=use_pos_emb,
+                    drop_path=drop_path_rates[i],
+                    **common,
                 )
+            stage_blocks.append(block)
             in_chs = out_chs
         self.blocks = nn.Sequential(*stage_blocks)
 
