@@ -499,19 +499,28 @@ def test_ns_gemm_kernels(shape):
 
 
 def test_muon_ns_kernel_matches_matmul_path():
-    """Full 5-step NS via HIP kernels vs the torch.matmul composition."""
+    """Full 5-step NS via HIP kernels vs the torch.matmul path on the same
+    device.  bf16 rounding compounds over 5 quintic iterations, so the
+    check is on the orthogonalization QUALITY (singular-value spread), which
+    must match the library-matmul path, plus a loose elementwise bound."""
+    import os
     _ext()
     from timm_amd.optim.muon import zeropower_via_newtonschulz
     torch.manual_seed(22)
     for M, N in [(768, 2304), (256, 100), (1024, 1024)]:
         g = torch.randn(M, N, device='cuda', dtype=torch.float32)
-        out_gpu = zeropower_via_newtonschulz(g)
-        out_cpu = zeropower_via_newtonschulz(g.cpu())
-        # both run in bf16; small drift from different accumulation orders
-        assert rel_err(out_gpu.cpu(), out_cpu) < 5e-2, (M, N)
-        # orthogonality: singular values nearly 1
-        sv = torch.linalg.svdvals(out_gpu.float())
-        assert 0.5 < sv.min().item() and sv.max().item() < 1.5
+        out_hip = zeropower_via_newtonschulz(g)
+        os.environ['TIMM_AMD_MUON_NS'] = 'torch'
+        try:
+            out_ref = zeropower_via_newtonschulz(g)
+        finally:
+            del os.environ['TIMM_AMD_MUON_NS']
+        sv_hip = torch.linalg.svdvals(out_hip.float())
+        sv_ref = torch.linalg.svdvals(out_ref.float())
+        # same convergence quality as the hipBLASLt path
+        assert sv_hip.min() > 0.8 * sv_ref.min().clamp(max=1.0), (M, N, sv_hip.min(), sv_ref.min())
+        assert sv_hip.max() < 1.25 * sv_ref.max(), (M, N)
+        assert rel_err(out_hip, out_ref) < 0.25, (M, N)
 
 
 def test_muon_optimizer_gpu_step():

@@ -7,6 +7,7 @@ unsuitable tensors fall back to AdamW-style updates inside the same
 optimizer (reference `muon.py:650`).
 """
 import math
+import os
 from typing import List, Optional, Tuple
 
 import torch
@@ -42,7 +43,8 @@ def zeropower_via_newtonschulz(
     # normalize so top singular value <= 1
     X = X / (X.norm(dim=(-2, -1), keepdim=True) + eps)
 
-    ext = ops._load_extension() if X.is_cuda else None
+    use_kernels = X.is_cuda and os.environ.get('TIMM_AMD_MUON_NS', 'hip') != 'torch'
+    ext = ops._load_extension() if use_kernels else None
     if ext is not None:
         # in-house batched MFMA kernels (muon_ns.hip): A and B are symmetric,
         # so every product runs as row-major NT except the final BX (NN with
