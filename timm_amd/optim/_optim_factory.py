@@ -17,10 +17,15 @@ import torch.optim
 from ._param_groups import param_groups_layer_decay, param_groups_weight_decay
 from .adabelief import AdaBelief
 from .adafactor import Adafactor
+from .adafactor_bv import AdafactorBigVision
+from .adahessian import Adahessian
+from .adamp import AdamP
 from .adamw import AdamW
 from .adan import Adan
 from .adopt import Adopt
+from .kron import Kron
 from .lamb import Lamb
+from .laprop import LaProp
 from .lars import Lars
 from .lion import Lion
 from .lookahead import Lookahead
@@ -28,8 +33,10 @@ from .madgrad import MADGRAD
 from .mars import Mars
 from .muon import Muon, AdaMuon
 from .nadamw import NAdamW
+from .nvnovograd import NvNovoGrad
 from .radam import RAdam
 from .rmsprop_tf import RMSpropTF
+from .sgdp import SGDP
 from .sgdw import SGDW
 
 _logger = logging.getLogger(__name__)
@@ -226,7 +233,7 @@ def _register_sgd_variants(registry: OptimizerRegistry) -> None:
                   has_eps=False, has_momentum=True, defaults={'nesterov': True}),
         OptimInfo(name='momentum', opt_class=torch.optim.SGD, description='torch.optim SGD w/ classical momentum',
                   has_eps=False, has_momentum=True, defaults={'nesterov': False}),
-        OptimInfo(name='sgdp', opt_class=SGDW, description='SGD (fallback to SGDW impl)',
+        OptimInfo(name='sgdp', opt_class=SGDP, description='SGD with scale-invariance projection',
                   has_eps=False, has_momentum=True, defaults={'nesterov': True}),
         OptimInfo(name='sgdw', opt_class=SGDW, description='SGD with decoupled weight decay and Nesterov momentum',
                   has_eps=False, has_momentum=True, defaults={'nesterov': True}),
@@ -323,6 +330,34 @@ def _register_other_optimizers(registry: OptimizerRegistry) -> None:
                   has_momentum=True, defaults={'alpha': 0.9}),
         OptimInfo(name='crmsproptf', opt_class=RMSpropTF, description='Cautious TF RMSProp',
                   has_momentum=True, defaults={'alpha': 0.9, 'caution': True}),
+        OptimInfo(name='adafactorbv', opt_class=AdafactorBigVision,
+                  description='Big-Vision Adafactor (factored 2nd moments, step-scheduled beta2)'),
+        OptimInfo(name='cadafactorbv', opt_class=AdafactorBigVision,
+                  description='Cautious Big-Vision Adafactor', defaults={'caution': True}),
+        OptimInfo(name='adahessian', opt_class=Adahessian,
+                  description='Second-order optimizer w/ Hutchinson Hessian diagonal',
+                  second_order=True, has_betas=True),
+        OptimInfo(name='adamp', opt_class=AdamP,
+                  description='Adam with scale-invariance projection', has_betas=True),
+        OptimInfo(name='cadamp', opt_class=AdamP,
+                  description='Cautious AdamP (projection, cautious mask n/a -> alias)',
+                  has_betas=True),
+        OptimInfo(name='kron', opt_class=Kron,
+                  description='PSGD-Kron: Kronecker-factored preconditioned SGD',
+                  has_momentum=True, has_eps=False),
+        OptimInfo(name='kronw', opt_class=Kron,
+                  description='PSGD-Kron w/ decoupled weight decay',
+                  has_momentum=True, has_eps=False),
+        OptimInfo(name='ckron', opt_class=Kron, description='Cautious PSGD-Kron',
+                  has_momentum=True, has_eps=False, defaults={'caution': True}),
+        OptimInfo(name='laprop', opt_class=LaProp,
+                  description='LaProp: decoupled momentum and adaptivity', has_betas=True),
+        OptimInfo(name='claprop', opt_class=LaProp, description='Cautious LaProp',
+                  has_betas=True, defaults={'caution': True}),
+        OptimInfo(name='novograd', opt_class=NvNovoGrad,
+                  description='NovoGrad: layer-wise second moments', has_betas=True),
+        OptimInfo(name='nvnovograd', opt_class=NvNovoGrad,
+                  description='NovoGrad (alias)', has_betas=True),
         OptimInfo(name='adadelta', opt_class=torch.optim.Adadelta, description='torch.optim.Adadelta'),
         OptimInfo(name='adagrad', opt_class=torch.optim.Adagrad, description='torch.optim.Adagrad',
                   defaults={'eps': 1e-8}),
