@@ -121,6 +121,7 @@ class AttentionRope(nn.Module):
             norm_layer: Optional[Type[nn.Module]] = None,
             qk_norm: bool = False,
             scale_norm: bool = False,
+            rotate_half: bool = False,
     ):
         super().__init__()
         if scale_norm or qk_norm:
@@ -134,6 +135,7 @@ class AttentionRope(nn.Module):
         self.scale = head_dim ** -0.5
         self.num_prefix_tokens = num_prefix_tokens
         self.fused_attn = use_fused_attn()
+        self.rotate_half = rotate_half
 
         if qkv_fused:
             self.qkv = nn.Linear(dim, attn_dim * 3, bias=qkv_bias)
@@ -172,8 +174,11 @@ class AttentionRope(nn.Module):
 
         if rope is not None:
             npt = self.num_prefix_tokens
-            q = torch.cat([q[:, :, :npt, :], apply_rot_embed_cat(q[:, :, npt:, :], rope)], dim=2).type_as(v)
-            k = torch.cat([k[:, :, :npt, :], apply_rot_embed_cat(k[:, :, npt:, :], rope)], dim=2).type_as(v)
+            half = self.rotate_half
+            q = torch.cat(
+                [q[:, :, :npt, :], apply_rot_embed_cat(q[:, :, npt:, :], rope, half=half)], dim=2).type_as(v)
+            k = torch.cat(
+                [k[:, :, :npt, :], apply_rot_embed_cat(k[:, :, npt:, :], rope, half=half)], dim=2).type_as(v)
 
         if self.fused_attn:
             x = ops.flash_attention(

@@ -11,7 +11,7 @@ Muon: dim 1024, depth 24, SwiGLU MLP, RoPE (RotaryEmbeddingCat), head_dim 64
 """
 import math
 from functools import partial
-from typing import Callable, Dict, Final, List, Optional, Set, Tuple, Type, Union
+from typing import Any, Callable, Dict, Final, List, Optional, Set, Tuple, Type, Union
 
 import torch
 import torch.nn as nn
@@ -23,7 +23,7 @@ from ..layers import (
     AttentionPoolLatent, DropPath, GluMlp, LayerNorm, Mlp, PatchDropout, PatchEmbed, RmsNorm,
     RotaryEmbeddingCat, SwiGLU, apply_keep_indices_nlc, apply_rot_embed_cat, calculate_drop_path_rates,
     get_act_layer, get_norm_layer, resample_abs_pos_embed, resample_patch_embed, to_2tuple, trunc_normal_,
-    use_fused_attn,
+    use_fused_attn, create_rope_embed, AttentionRope,
 )
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
@@ -52,8 +52,10 @@ class EvaAttention(nn.Module):
             norm_layer: Optional[Callable] = None,
             qk_norm: bool = False,
             scale_norm: bool = True,
+            rotate_half: bool = False,
     ):
         super().__init__()
+        self.rotate_half = rotate_half
         self.num_heads = num_heads
         head_dim = dim // num_heads
         if attn_head_dim is not None:
@@ -116,8 +118,9 @@ class EvaAttention(nn.Module):
 
         if rope is not None:
             npt = self.num_prefix_tokens
-            q = torch.cat([q[:, :, :npt, :], apply_rot_embed_cat(q[:, :, npt:, :], rope)], 2).type_as(v)
-            k = torch.cat([k[:, :, :npt, :], apply_rot_embed_cat(k[:, :, npt:, :], rope)], 2).type_as(v)
+            half = self.rotate_half
+            q = torch.cat([q[:, :, :npt, :], apply_rot_embed_cat(q[:, :, npt:, :], rope, half=half)], 2).type_as(v)
+            k = torch.cat([k[:, :, :npt, :], apply_rot_embed_cat(k[:, :, npt:, :], rope, half=half)], 2).type_as(v)
 
         if self.fused_attn:
             x = ops.flash_attention(
@@ -153,8 +156,11 @@ class EvaBlock(nn.Module):
             qkv_fused: bool = True,
             mlp_ratio: float = 4.,
             swiglu_mlp: bool = False,
+            swiglu_align_to: int = 0,
             scale_mlp: bool = False,
             scale_attn_inner: bool = False,
+            attn_type: str = 'eva',
+            rotate_half: bool = False,
             num_prefix_tokens: int = 1,
             attn_head_dim: Optional[int] = None,
             proj_drop: float = 0.,
@@ -166,7 +172,8 @@ class EvaBlock(nn.Module):
     ):
         super().__init__()
         self.norm1 = norm_layer(dim)
-        self.attn = EvaAttention(
+        attn_cls = AttentionRope if attn_type == 'rope' else EvaAttention
+        self.attn = attn_cls(
             dim,
             num_heads=num_heads,
             qkv_bias=qkv_bias,
@@ -177,6 +184,7 @@ class EvaBlock(nn.Module):
             attn_head_dim=attn_head_dim,
             norm_layer=norm_layer if scale_attn_inner else None,
             scale_norm=scale_attn_inner,
+            rotate_half=rotate_half,
         )
         self.gamma_1 = nn.Parameter(init_values * torch.ones(dim)) if init_values is not None else None
         self.drop_path1 = DropPath(drop_path) if drop_path > 0. else nn.Identity()
@@ -184,13 +192,14 @@ class EvaBlock(nn.Module):
         self.norm2 = norm_layer(dim)
         hidden_features = int(dim * mlp_ratio)
         if swiglu_mlp:
-            if scale_mlp:
-                # when norm in SwiGLU used, an impl with separate fc for gate & x is used
+            if scale_mlp or swiglu_align_to:
+                # norm inside SwiGLU or alignment: separate gate/x projections
                 self.mlp = SwiGLU(
                     in_features=dim,
                     hidden_features=hidden_features,
                     norm_layer=norm_layer if scale_mlp else None,
                     drop=proj_drop,
+                    align_to=swiglu_align_to,
                 )
             else:
                 # w/o any extra norm, an impl with packed weights is used
@@ -238,8 +247,11 @@ class EvaBlockPostNorm(nn.Module):
             qkv_fused: bool = True,
             mlp_ratio: float = 4.,
             swiglu_mlp: bool = False,
+            swiglu_align_to: int = 0,
             scale_mlp: bool = False,
             scale_attn_inner: bool = False,
+            attn_type: str = 'eva',
+            rotate_half: bool = False,
             num_prefix_tokens: int = 1,
             attn_head_dim: Optional[int] = None,
             proj_drop: float = 0.,
@@ -250,7 +262,8 @@ class EvaBlockPostNorm(nn.Module):
             norm_layer: Callable = nn.LayerNorm,
     ):
         super().__init__()
-        self.attn = EvaAttention(
+        attn_cls = AttentionRope if attn_type == 'rope' else EvaAttention
+        self.attn = attn_cls(
             dim,
             num_heads=num_heads,
             qkv_bias=qkv_bias,
@@ -261,18 +274,20 @@ class EvaBlockPostNorm(nn.Module):
             attn_head_dim=attn_head_dim,
             norm_layer=norm_layer if scale_attn_inner else None,
             scale_norm=scale_attn_inner,
+            rotate_half=rotate_half,
         )
         self.norm1 = norm_layer(dim)
         self.drop_path1 = DropPath(drop_path) if drop_path > 0. else nn.Identity()
 
         hidden_features = int(dim * mlp_ratio)
         if swiglu_mlp:
-            if scale_mlp:
+            if scale_mlp or swiglu_align_to:
                 self.mlp = SwiGLU(
                     in_features=dim,
                     hidden_features=hidden_features,
                     norm_layer=norm_layer if scale_mlp else None,
                     drop=proj_drop,
+                    align_to=swiglu_align_to,
                 )
             else:
                 self.mlp = GluMlp(
@@ -325,8 +340,10 @@ class Eva(nn.Module):
             qkv_fused: bool = True,
             mlp_ratio: float = 4.,
             swiglu_mlp: bool = False,
+            swiglu_align_to: int = 0,
             scale_mlp: bool = False,
             scale_attn_inner: bool = False,
+            attn_type: str = 'eva',
             attn_head_dim: Optional[int] = None,
             drop_rate: float = 0.,
             pos_drop_rate: float = 0.,
@@ -338,9 +355,20 @@ class Eva(nn.Module):
             init_values: Optional[float] = None,
             class_token: bool = True,
             num_reg_tokens: int = 0,
+            no_embed_class: bool = False,
             use_abs_pos_emb: bool = True,
             use_rot_pos_emb: bool = False,
+            rope_type: Optional[str] = 'cat',
+            rope_grid_offset: float = 0.,
+            rope_grid_indexing: str = 'ij',
+            rope_temperature: float = 10000.,
+            rope_rotate_half: bool = False,
             use_post_norm: bool = False,
+            use_pre_transformer_norm: bool = False,
+            use_post_transformer_norm: Optional[bool] = None,
+            use_fc_norm: Optional[bool] = None,
+            attn_pool_num_heads: Optional[int] = None,
+            attn_pool_mlp_ratio: Optional[float] = None,
             dynamic_img_size: bool = False,
             dynamic_img_pad: bool = False,
             ref_feat_shape: Optional[Union[Tuple[int, int], int]] = None,
@@ -348,12 +376,22 @@ class Eva(nn.Module):
     ):
         super().__init__()
         norm_layer = norm_layer or partial(LayerNorm, eps=1e-6)
+        assert global_pool in ('', 'avg', 'avgmax', 'max', 'token', 'map')
         self.num_classes = num_classes
         self.global_pool = global_pool
         self.num_features = self.head_hidden_size = self.embed_dim = embed_dim
         self.num_prefix_tokens = (1 if class_token else 0) + num_reg_tokens
+        self.no_embed_class = no_embed_class
         self.dynamic_img_size = dynamic_img_size
         self.grad_checkpointing = False
+
+        # norm placement resolution (reference eva.py:643-651): fc-norm
+        # defaults on for avg pooling; trunk post-norm defaults to its inverse
+        activate_pre_norm = use_pre_transformer_norm
+        activate_fc_norm = use_fc_norm if use_fc_norm is not None else global_pool == 'avg'
+        activate_post_norm = (
+            use_post_transformer_norm if use_post_transformer_norm is not None
+            else not activate_fc_norm)
 
         embed_args = {}
         if dynamic_img_size:
@@ -365,6 +403,7 @@ class Eva(nn.Module):
             in_chans=in_chans,
             embed_dim=embed_dim,
             dynamic_img_pad=dynamic_img_pad,
+            bias=not use_pre_transformer_norm,
             **embed_args,
         )
         num_patches = self.patch_embed.num_patches
@@ -374,7 +413,10 @@ class Eva(nn.Module):
         self.reg_token = nn.Parameter(torch.zeros(1, num_reg_tokens, embed_dim)) if num_reg_tokens else None
         self.cls_embed = class_token and self.reg_token is None
 
-        num_pos_tokens = num_patches + self.num_prefix_tokens if self.cls_embed else num_patches
+        if no_embed_class:
+            num_pos_tokens = num_patches
+        else:
+            num_pos_tokens = num_patches + self.num_prefix_tokens if self.cls_embed else num_patches
         self.pos_embed = nn.Parameter(
             torch.zeros(1, num_pos_tokens, embed_dim)) if use_abs_pos_emb else None
         self.pos_drop = nn.Dropout(p=pos_drop_rate)
@@ -387,14 +429,31 @@ class Eva(nn.Module):
         else:
             self.patch_drop = None
 
+        self.rope_mixed = rope_type == 'mixed'
         if use_rot_pos_emb:
             ref_feat_shape = to_2tuple(ref_feat_shape) if ref_feat_shape is not None else None
-            self.rope = RotaryEmbeddingCat(
-                embed_dim // num_heads,
-                in_pixels=False,
+            rope_kwargs = dict(
+                dim=embed_dim,
+                num_heads=num_heads,
                 feat_shape=None if dynamic_img_size else self.patch_embed.grid_size,
-                ref_feat_shape=ref_feat_shape,
+                temperature=rope_temperature,
+                grid_indexing=rope_grid_indexing,
             )
+            if self.rope_mixed:
+                rope_kwargs.update(dict(depth=depth))
+            elif rope_type in ('cat', None, '', 'regular'):
+                rope_type = 'cat'
+                rope_kwargs.update(dict(
+                    in_pixels=False,
+                    grid_offset=rope_grid_offset,
+                    ref_feat_shape=ref_feat_shape,
+                ))
+            elif rope_type == 'dinov3':
+                rope_kwargs.update(dict(
+                    grid_offset=rope_grid_offset,
+                    rotate_half=rope_rotate_half,
+                ))
+            self.rope = create_rope_embed(rope_type=rope_type, **rope_kwargs)
         else:
             self.rope = None
 
@@ -408,8 +467,11 @@ class Eva(nn.Module):
                 qkv_fused=qkv_fused,
                 mlp_ratio=mlp_ratio,
                 swiglu_mlp=swiglu_mlp,
+                swiglu_align_to=swiglu_align_to,
                 scale_mlp=scale_mlp,
                 scale_attn_inner=scale_attn_inner,
+                attn_type=attn_type,
+                rotate_half=rope_rotate_half,
                 num_prefix_tokens=self.num_prefix_tokens,
                 attn_head_dim=attn_head_dim,
                 proj_drop=proj_drop_rate,
@@ -422,9 +484,19 @@ class Eva(nn.Module):
         self.feature_info = [
             dict(module=f'blocks.{i}', num_chs=embed_dim, reduction=r) for i in range(depth)]
 
-        use_fc_norm = self.global_pool == 'avg'
-        self.norm = nn.Identity() if use_fc_norm else norm_layer(embed_dim)
-        self.fc_norm = norm_layer(embed_dim) if use_fc_norm else nn.Identity()
+        self.norm_pre = norm_layer(embed_dim) if activate_pre_norm else nn.Identity()
+        self.norm = norm_layer(embed_dim) if activate_post_norm else nn.Identity()
+        if global_pool == 'map':
+            self.attn_pool = AttentionPoolLatent(
+                self.embed_dim,
+                num_heads=attn_pool_num_heads or num_heads,
+                mlp_ratio=attn_pool_mlp_ratio or mlp_ratio,
+                norm_layer=norm_layer,
+                act_layer=nn.GELU,
+            )
+        else:
+            self.attn_pool = None
+        self.fc_norm = norm_layer(embed_dim) if activate_fc_norm else nn.Identity()
         self.head_drop = nn.Dropout(drop_rate)
         self.head = nn.Linear(embed_dim, num_classes) if num_classes > 0 else nn.Identity()
 
@@ -492,7 +564,8 @@ class Eva(nn.Module):
                     self.pos_embed,
                     new_size=(H, W),
                     old_size=prev_grid_size,
-                    num_prefix_tokens=self.num_prefix_tokens if self.cls_embed else 0,
+                    num_prefix_tokens=0 if self.no_embed_class
+                    else (self.num_prefix_tokens if self.cls_embed else 0),
                 )
             else:
                 pos_embed = None
@@ -502,11 +575,17 @@ class Eva(nn.Module):
             pos_embed = self.pos_embed
             rot_pos_embed = self.rope.get_embed() if self.rope is not None else None
 
-        if self.cls_token is not None:
-            x = torch.cat((self.cls_token.expand(x.shape[0], -1, -1), x), dim=1)
-
-        if pos_embed is not None:
-            x = x + pos_embed
+        if self.no_embed_class:
+            # pos embed covers patch tokens only; prefix tokens ride along
+            if pos_embed is not None:
+                x = x + pos_embed
+            if self.cls_token is not None:
+                x = torch.cat((self.cls_token.expand(x.shape[0], -1, -1), x), dim=1)
+        else:
+            if self.cls_token is not None:
+                x = torch.cat((self.cls_token.expand(x.shape[0], -1, -1), x), dim=1)
+            if pos_embed is not None:
+                x = x + pos_embed
 
         if self.reg_token is not None:
             to_cat = []
@@ -544,15 +623,17 @@ class Eva(nn.Module):
         B, _, height, width = x.shape
         x = self.patch_embed(x)
         x, rot_pos_embed = self._pos_embed(x)
+        x = self.norm_pre(x)
         if torch.jit.is_scripting() or not stop_early:  # can't slice blocks in torchscript
             blocks = self.blocks
         else:
             blocks = self.blocks[:max_index + 1]
         for i, blk in enumerate(blocks):
+            rope_i = rot_pos_embed[i] if self.rope_mixed and rot_pos_embed is not None else rot_pos_embed
             if self.grad_checkpointing and not torch.jit.is_scripting():
-                x = checkpoint(blk, x, rope=rot_pos_embed)
+                x = checkpoint(blk, x, rope=rope_i)
             else:
-                x = blk(x, rope=rot_pos_embed)
+                x = blk(x, rope=rope_i)
             if i in take_indices:
                 intermediates.append(self.norm(x) if norm else x)
 
@@ -596,16 +677,20 @@ class Eva(nn.Module):
     def forward_features(self, x: torch.Tensor) -> torch.Tensor:
         x = self.patch_embed(x)
         x, rot_pos_embed = self._pos_embed(x)
-        for blk in self.blocks:
+        x = self.norm_pre(x)
+        for i, blk in enumerate(self.blocks):
+            rope = rot_pos_embed[i] if self.rope_mixed and rot_pos_embed is not None else rot_pos_embed
             if self.grad_checkpointing and not torch.jit.is_scripting():
-                x = checkpoint(blk, x, rope=rot_pos_embed)
+                x = checkpoint(blk, x, rope=rope)
             else:
-                x = blk(x, rope=rot_pos_embed)
+                x = blk(x, rope=rope)
         x = self.norm(x)
         return x
 
     def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
-        if self.global_pool:
+        if self.attn_pool is not None:
+            x = self.attn_pool(x)
+        elif self.global_pool:
             x = x[:, self.num_prefix_tokens:].mean(dim=1) if self.global_pool == 'avg' else x[:, 0]
         x = self.fc_norm(x)
         x = self.head_drop(x)
@@ -717,21 +802,454 @@ def _cfg(url='', **kwargs):
     }
 
 
+def _pe_cfg(url: str = '', **kwargs) -> Dict[str, Any]:
+    """Cfg preset for Perception Encoder weights."""
+    return {
+        'url': url,
+        'num_classes': 0, 'input_size': (3, 224, 224), 'pool_size': None,
+        'crop_pct': 1.0, 'interpolation': 'bicubic', 'fixed_input_size': True,
+        'mean': (0.5, 0.5, 0.5), 'std': (0.5, 0.5, 0.5),
+        'first_conv': 'patch_embed.proj', 'classifier': 'head',
+        'license': 'apache-2.0', **kwargs
+    }
+
+
+def _dinov3_cfg(url: str = '', **kwargs) -> Dict[str, Any]:
+    """Cfg preset for DINOv3 weights (cls-token pooling upstream; pass
+    global_pool='token' at create time to match)."""
+    return {
+        'url': url,
+        'num_classes': 0, 'input_size': (3, 256, 256), 'pool_size': None,
+        'crop_pct': 1.0, 'interpolation': 'bicubic', 'fixed_input_size': True,
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'patch_embed.proj', 'classifier': 'head',
+        'license': 'dinov3-license', **kwargs
+    }
+
+
+def _eupe_cfg(url: str = '', **kwargs) -> Dict[str, Any]:
+    """Cfg preset for EUPE weights."""
+    return {
+        'url': url,
+        'num_classes': 0, 'input_size': (3, 256, 256), 'pool_size': None,
+        'crop_pct': 1.0, 'interpolation': 'bicubic', 'fixed_input_size': True,
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'patch_embed.proj', 'classifier': 'head',
+        'license': 'fair-noncommercial-research-license', **kwargs
+    }
+
+
 default_cfgs = generate_default_cfgs({
-    'eva_giant_patch14_224.clip_ft_in1k': _cfg(),
-    'eva_giant_patch14_336.clip_ft_in1k': _cfg(input_size=(3, 336, 336), crop_pct=1.0),
-    'eva02_tiny_patch14_224.mim_in22k': _cfg(num_classes=0),
-    'eva02_small_patch14_224.mim_in22k': _cfg(num_classes=0),
-    'eva02_base_patch14_224.mim_in22k': _cfg(num_classes=0),
-    'eva02_large_patch14_224.mim_in22k': _cfg(num_classes=0),
-    'eva02_tiny_patch14_336.mim_in22k_ft_in1k': _cfg(input_size=(3, 336, 336), crop_pct=1.0),
-    'eva02_small_patch14_336.mim_in22k_ft_in1k': _cfg(input_size=(3, 336, 336), crop_pct=1.0),
-    'eva02_base_patch14_448.mim_in22k_ft_in1k': _cfg(input_size=(3, 448, 448), crop_pct=1.0),
-    'eva02_large_patch14_336.mim_in22k_ft_in1k': _cfg(
+
+    # EVA 01 CLIP fine-tuned on imagenet-1k
+    'eva_giant_patch14_224.clip_ft_in1k': _cfg(
+        # hf_hub_id='BAAI/EVA', hf_hub_filename='eva_clip_vis_enc_sz224_ftcls_89p1.pt',
+        hf_hub_id='timm/',
+    ),
+    'eva_giant_patch14_336.clip_ft_in1k': _cfg(
+        # hf_hub_id='BAAI/EVA', hf_hub_filename='eva_clip_vis_enc_sz336_ftcls_89p4.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 336, 336), crop_pct=1.0, crop_mode='squash'),
+
+    # MIM EVA 01 pretrain, ft on in22k -> in1k
+    'eva_giant_patch14_336.m30m_ft_in22k_in1k': _cfg(
+        # hf_hub_id='BAAI/EVA', hf_hub_filename='eva_21k_1k_336px_psz14_ema_89p6.pt',
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        input_size=(3, 336, 336), crop_pct=1.0, crop_mode='squash'),
+    'eva_giant_patch14_560.m30m_ft_in22k_in1k': _cfg(
+        # hf_hub_id='BAAI/EVA', hf_hub_filename='eva_21k_1k_560px_psz14_ema_89p7.pt',
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        input_size=(3, 560, 560), crop_pct=1.0, crop_mode='squash'),
+
+    # in22k or m38m MIM pretrain w/ intermediate in22k fine-tune and final in1k fine-tune
+    'eva02_base_patch14_448.mim_in22k_ft_in22k_in1k': _cfg(
+        # hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k_to_in1k/eva02_B_pt_in21k_medft_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash',
+    ),
+    'eva02_large_patch14_448.mim_in22k_ft_in22k_in1k': _cfg(
+        # hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k_to_in1k/eva02_L_pt_in21k_medft_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash',
+    ),
+    'eva02_large_patch14_448.mim_m38m_ft_in22k_in1k': _cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k_to_in1k/eva02_L_pt_m38m_medft_in21k_ft_in1k_p14.pt',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash',
+    ),
+
+    # in22k or m3m MIM pretrain w/ in1k fine-tune
+    'eva02_tiny_patch14_336.mim_in22k_ft_in1k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in1k/eva02_Ti_pt_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
         input_size=(3, 336, 336), crop_pct=1.0,
-        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD),
-    'eva02_large_patch14_448.mim_in22k_ft_in1k': _cfg(input_size=(3, 448, 448), crop_pct=1.0),
-    'eva02_large_patch14_clip_336.merged2b': _cfg(input_size=(3, 336, 336), crop_pct=1.0, num_classes=768),
+    ),
+    'eva02_small_patch14_336.mim_in22k_ft_in1k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in1k/eva02_S_pt_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 336, 336), crop_pct=1.0,
+    ),
+    'eva02_base_patch14_448.mim_in22k_ft_in1k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in1k/eva02_B_pt_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0,
+    ),
+    'eva02_large_patch14_448.mim_in22k_ft_in1k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in1k/eva02_L_pt_in21k_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0,
+    ),
+    'eva02_large_patch14_448.mim_m38m_ft_in1k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in1k/eva02_L_pt_m38m_ft_in1k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0,
+    ),
+
+    # in22k or m3m MIM pretrain w/ in22k fine-tune
+    'eva02_base_patch14_448.mim_in22k_ft_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k/eva02_B_pt_in21k_medft_in21k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash', num_classes=21841,
+    ),
+    'eva02_large_patch14_448.mim_in22k_ft_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k/eva02_L_pt_in21k_medft_in21k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash', num_classes=21841,
+    ),
+    'eva02_large_patch14_448.mim_m38m_ft_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/cls/in21k/eva02_L_pt_m38m_medft_in21k_p14.pt',
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), crop_pct=1.0, crop_mode='squash', num_classes=21841,
+    ),
+
+    # in22k or m38m MIM pretrain
+    'eva02_tiny_patch14_224.mim_in22k': _cfg(
+        # hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/pt/eva02_Ti_pt_in21k_p14.pt',
+        hf_hub_id='timm/',
+        num_classes=0,
+    ),
+    'eva02_small_patch14_224.mim_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/pt/eva02_S_pt_in21k_p14.pt',
+        hf_hub_id='timm/',
+        num_classes=0,
+    ),
+    'eva02_base_patch14_224.mim_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/pt/eva02_B_pt_in21k_p14.pt',
+        hf_hub_id='timm/',
+        num_classes=0,
+    ),
+    'eva02_large_patch14_224.mim_in22k': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/pt/eva02_L_pt_in21k_p14.pt',
+        hf_hub_id='timm/',
+        num_classes=0,
+    ),
+    'eva02_large_patch14_224.mim_m38m': _cfg(
+        #hf_hub_id='Yuxin-CV/EVA-02', hf_hub_filename='eva02/pt/eva02_L_pt_m38m_p14.pt',
+        hf_hub_id='timm/',
+        num_classes=0,
+    ),
+
+    # EVA01 and EVA02 CLIP image towers
+    'eva_giant_patch14_clip_224.laion400m': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA01_CLIP_g_14_plus_psz14_s11B.pt',
+        # hf_hub_id='timm/eva_giant_patch14_clip_224.laion400m_s11b_b41k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=1024,
+    ),
+    'eva_giant_patch14_clip_224.merged2b': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA01_CLIP_g_14_plus_psz14_s11B.pt',
+        # hf_hub_id='timm/eva_giant_patch14_plus_clip_224.merged2b_s11b_b114k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=1024,
+    ),
+    'eva02_base_patch16_clip_224.merged2b': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_CLIP_L_psz14_s4B.pt',
+        # hf_hub_id='timm/eva02_base_patch16_clip_224.merged2b_s8b_b131k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=512,
+    ),
+    'eva02_large_patch14_clip_224.merged2b': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_CLIP_L_psz14_s4B.pt',
+        # hf_hub_id='timm/eva02_large_patch14_clip_224.merged2b_s4b_b131k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=768,
+    ),
+    'eva02_large_patch14_clip_336.merged2b': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_CLIP_L_psz14_s4B.pt',
+        # hf_hub_id='timm/eva02_large_patch14_clip_336.merged2b_s6b_b61k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        input_size=(3, 336, 336), crop_pct=1.0,
+        num_classes=768,
+    ),
+    'eva02_enormous_patch14_clip_224.laion2b': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_CLIP_E_psz14_plus_s9B.pt',
+        # hf_hub_id='timm/eva02_enormous_patch14_clip_224.laion2b_s4b_b115k',  # float16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=1024,
+    ),
+    'eva02_enormous_patch14_clip_224.laion2b_plus': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_CLIP_E_psz14_plus_s9B.pt',
+        # hf_hub_id='timm/eva02_enormous_patch14_plus_clip_224.laion2b_s9b_b144k',  # bfloat16 weights
+        # hf_hub_filename='open_clip_pytorch_model.bin',
+        hf_hub_id='timm/',
+        num_classes=1024,
+    ),
+    'eva02_enormous_patch14_clip_224.pretrain': _cfg(
+        # hf_hub_id='QuanSun/EVA-CLIP', hf_hub_filename='EVA02_E_psz14.pt',
+        num_classes=0,
+    ),
+
+    'vit_medium_patch16_rope_reg1_gap_256.sbb_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5)
+    ),
+    'vit_mediumd_patch16_rope_reg1_gap_256.sbb_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5)
+    ),
+    'vit_betwixt_patch16_rope_reg4_gap_256.sbb_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
+    'vit_base_patch16_rope_reg1_gap_256.sbb_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5)
+    ),
+
+    # Perception Encoder weights
+    'vit_pe_core_tiny_patch16_384.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Core-T16-384',
+        #hf_hub_filename='PE-Core-T16-384.pt',
+        input_size=(3, 384, 384),
+        num_classes=512,  # output proj dim
+    ),
+    'vit_pe_core_small_patch16_384.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Core-S16-384',
+        #hf_hub_filename='PE-Core-S16-384.pt',
+        input_size=(3, 384, 384),
+        num_classes=512,  # output proj dim
+    ),
+    'vit_pe_core_base_patch16_224.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Core-B16-224',
+        #hf_hub_filename='PE-Core-B16-224.pt',
+        input_size=(3, 224, 224),
+        num_classes=1024,  # output proj dim
+    ),
+    'vit_pe_core_large_patch14_336.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Core-L14-336',
+        #hf_hub_filename='PE-Core-L14-336.pt',
+        input_size=(3, 336, 336),
+        num_classes=1024,  # output proj dim
+    ),
+    'vit_pe_core_gigantic_patch14_448.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Core-G14-448',
+        #hf_hub_filename='PE-Core-G14-448.pt',
+        input_size=(3, 448, 448),
+        num_classes=1280,  # output proj dim
+    ),
+
+    'vit_pe_lang_large_patch14_448.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Lang-L14-448',
+        #hf_hub_filename='PE-Lang-L14-448.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+    'vit_pe_lang_large_patch14_448.fb_tiling': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Lang-L14-448-Tiling',
+        #hf_hub_filename='PE-Lang-L14-448-Tiling.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+    'vit_pe_lang_gigantic_patch14_448.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Lang-G14-448',
+        #hf_hub_filename='PE-Lang-G14-448.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+    'vit_pe_lang_gigantic_patch14_448.fb_tiling': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Lang-G14-448-Tiling',
+        #hf_hub_filename='PE-Lang-G14-448-Tiling.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+
+    'vit_pe_spatial_tiny_patch16_512.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Spatial-T16-512',
+        #hf_hub_filename='PE-Spatial-T16-512.pt',
+        input_size=(3, 512, 512),
+        num_classes=0,
+    ),
+    'vit_pe_spatial_small_patch16_512.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Spatial-S16-512',
+        #hf_hub_filename='PE-Spatial-S16-512.pt',
+        input_size=(3, 512, 512),
+        num_classes=0,
+    ),
+    'vit_pe_spatial_base_patch16_512.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Spatial-B16-512',
+        #hf_hub_filename='PE-Spatial-B16-512.pt',
+        input_size=(3, 512, 512),
+        num_classes=0,
+    ),
+    'vit_pe_spatial_large_patch14_448.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Spatial-L14-448',
+        #hf_hub_filename='PE-Spatial-L14-448.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+    'vit_pe_spatial_gigantic_patch14_448.fb': _pe_cfg(
+        hf_hub_id='timm/',
+        #hf_hub_id='facebook/PE-Spatial-G14-448',
+        #hf_hub_filename='PE-Spatial-G14-448.pt',
+        input_size=(3, 448, 448),
+        num_classes=0,
+    ),
+
+    # RoPE-ViT models from Naver
+    'vit_small_patch16_rope_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_base_patch16_rope_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_large_patch16_rope_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_small_patch16_rope_mixed_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_base_patch16_rope_mixed_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_large_patch16_rope_mixed_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_small_patch16_rope_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_base_patch16_rope_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_large_patch16_rope_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_small_patch16_rope_mixed_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_base_patch16_rope_mixed_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+    'vit_large_patch16_rope_mixed_ape_224.naver_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        license='apache-2.0',
+    ),
+
+    # DINOv3 weights are under a specific license with redistribution terms, please see
+    # https://github.com/facebookresearch/dinov3/blob/main/LICENSE.md
+    # NOTE: Original DINOv3 uses CLS-token pooling (global_pool='token') which may be better
+    # for some tasks. Default here is avg pooling inherited from the Eva base class.
+    'vit_small_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_small_patch16_dinov3_qkvb.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_small_plus_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_small_plus_patch16_dinov3_qkvb.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_base_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_base_patch16_dinov3_qkvb.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_tiny_patch16_dinov3_qkvb.eupe_lvd1689m': _eupe_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_small_patch16_dinov3_qkvb.eupe_lvd1689m': _eupe_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_base_patch16_dinov3_qkvb.eupe_lvd1689m': _eupe_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_large_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_large_patch16_dinov3_qkvb.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_large_patch16_dinov3.sat493m': _dinov3_cfg(
+        hf_hub_id='timm/',
+        mean=(0.430, 0.411, 0.296), std=(0.213, 0.156, 0.143),
+    ),
+    'vit_large_patch16_dinov3_qkvb.sat493m': _dinov3_cfg(
+        hf_hub_id='timm/',
+        mean=(0.430, 0.411, 0.296), std=(0.213, 0.156, 0.143),
+    ),
+    'vit_huge_plus_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_huge_plus_patch16_dinov3_qkvb.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_7b_patch16_dinov3.lvd1689m': _dinov3_cfg(
+        hf_hub_id='timm/',
+    ),
+    'vit_7b_patch16_dinov3.sat493m': _dinov3_cfg(
+        hf_hub_id='timm/',
+        mean=(0.430, 0.411, 0.296), std=(0.213, 0.156, 0.143),
+    ),
+
 })
 
 
@@ -934,3 +1452,1077 @@ def eva02_large_patch14_clip_336(pretrained: bool = False, **kwargs) -> Eva:
     )
     model = _create_eva('eva02_large_patch14_clip_336', pretrained=pretrained, **dict(model_args, **kwargs))
     return model
+
+
+@register_model
+def eva_giant_patch14_560(pretrained: bool = False, **kwargs) -> Eva:
+    """EVA-g model https://arxiv.org/abs/2211.07636"""
+    model_args = dict(patch_size=14, embed_dim=1408, depth=40, num_heads=16, mlp_ratio=6144 / 1408)
+    model = _create_eva('eva_giant_patch14_560', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def eva_giant_patch14_clip_224(pretrained: bool = False, **kwargs) -> Eva:
+    """EVA-g CLIP model (only difference from non-CLIP is the pooling)"""
+    model_args = dict(
+        patch_size=14, embed_dim=1408, depth=40, num_heads=16, mlp_ratio=6144 / 1408,
+        global_pool=kwargs.pop('global_pool', 'token'))
+    model = _create_eva('eva_giant_patch14_clip_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def eva02_base_patch16_clip_224(pretrained: bool = False, **kwargs) -> Eva:
+    """An EVA-CLIP specific variant that adds additional attn scale layer-norm to eva02_base"""
+    model_args = dict(
+        img_size=224,
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        qkv_fused=False,
+        mlp_ratio=4 * 2 / 3,
+        swiglu_mlp=True,
+        scale_mlp=True,
+        scale_attn_inner=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(16, 16),  # 224/14
+        global_pool=kwargs.pop('global_pool', 'token'),
+    )
+    model = _create_eva('eva02_base_patch16_clip_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def eva02_large_patch14_clip_224(pretrained: bool = False, **kwargs) -> Eva:
+    """An EVA-CLIP specific variant that adds additional attn scale layer-norm to eva02_large"""
+    model_args = dict(
+        img_size=224,
+        patch_size=14,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4 * 2 / 3,
+        qkv_fused=False,
+        swiglu_mlp=True,
+        scale_mlp=True,
+        scale_attn_inner=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(16, 16),  # 224/14
+        global_pool=kwargs.pop('global_pool', 'token'),
+    )
+    model = _create_eva('eva02_large_patch14_clip_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def eva02_enormous_patch14_clip_224(pretrained: bool = False, **kwargs) -> Eva:
+    """An EVA-CLIP specific variant that uses residual post-norm in blocks"""
+    model_args = dict(
+        img_size=224,
+        patch_size=14,
+        embed_dim=1792,
+        depth=64,
+        num_heads=16,
+        mlp_ratio=15360 / 1792,
+        use_post_norm=True,
+        global_pool=kwargs.pop('global_pool', 'token'),
+    )
+    model = _create_eva('eva02_enormous_patch14_clip_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_medium_patch16_rope_reg1_gap_256(pretrained: bool = False, **kwargs) -> Eva:
+    """timm SBB ViT with ROPE"""
+    model_args = dict(
+        img_size=256,
+        patch_size=16,
+        embed_dim=512,
+        depth=12,
+        num_heads=8,
+        qkv_fused=True,
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=False,
+        num_reg_tokens=1,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        ref_feat_shape=(16, 16),  # 224/14
+    )
+    model = _create_eva('vit_medium_patch16_rope_reg1_gap_256', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_mediumd_patch16_rope_reg1_gap_256(pretrained: bool = False, **kwargs) -> Eva:
+    """timm SBB ViT with ROPE"""
+    model_args = dict(
+        img_size=256,
+        patch_size=16,
+        embed_dim=512,
+        depth=20,
+        num_heads=8,
+        qkv_fused=True,
+        qkv_bias=False,
+        init_values=1e-5,
+        class_token=False,
+        num_reg_tokens=1,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        ref_feat_shape=(16, 16),  # 224/14
+    )
+    model = _create_eva('vit_mediumd_patch16_rope_reg1_gap_256', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_betwixt_patch16_rope_reg4_gap_256(pretrained: bool = False, **kwargs) -> Eva:
+    """timm SBB ViT with ROPE"""
+    model_args = dict(
+        img_size=256,
+        patch_size=16,
+        embed_dim=640,
+        depth=12,
+        num_heads=10,
+        qkv_fused=True,
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=False,
+        num_reg_tokens=4,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        ref_feat_shape=(16, 16),  # 224/14
+    )
+    model = _create_eva('vit_betwixt_patch16_rope_reg4_gap_256', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_rope_reg1_gap_256(pretrained: bool = False, **kwargs) -> Eva:
+    """timm SBB ViT with ROPE"""
+    model_args = dict(
+        img_size=256,
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        qkv_fused=True,
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=False,
+        num_reg_tokens=1,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        ref_feat_shape=(16, 16),  # 224/14
+    )
+    model = _create_eva('vit_base_patch16_rope_reg1_gap_256', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_pe_core_tiny_patch16_384(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=192,
+        depth=12,
+        num_heads=3,
+        mlp_ratio=4.0,
+        global_pool='map',
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(24, 24),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        attn_pool_num_heads=8,
+        attn_pool_mlp_ratio=4.,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_core_tiny_patch16_384', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_core_small_patch16_384(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4.0,
+        global_pool='map',
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(24, 24),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        attn_pool_num_heads=8,
+        attn_pool_mlp_ratio=4.,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_core_small_patch16_384', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_core_base_patch16_224(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4.0,
+        global_pool='map',
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(14, 14),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        attn_pool_num_heads=8,
+        attn_pool_mlp_ratio=4.,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_core_base_patch16_224', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_core_large_patch14_336(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4.0,
+        global_pool='map',
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(24, 24),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        attn_pool_num_heads=8,
+        attn_pool_mlp_ratio=4.,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_core_large_patch14_336', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_core_gigantic_patch14_448(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1536,
+        depth=50,
+        num_heads=16,
+        mlp_ratio=8960 / 1536,
+        global_pool='map',
+        attn_type='rope',
+        class_token=False,
+        use_pre_transformer_norm=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_indexing='xy',
+        attn_pool_num_heads=8,
+        attn_pool_mlp_ratio=4.,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_core_gigantic_patch14_448', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_lang_large_patch14_448(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1024,
+        depth=23,
+        num_heads=16,
+        mlp_ratio=4.0,
+        attn_type='rope',
+        class_token=True,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        init_values=0.1,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_lang_large_patch14_448', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_lang_gigantic_patch14_448(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1536,
+        depth=47,
+        num_heads=16,
+        mlp_ratio=8960 / 1536,
+        attn_type='rope',
+        class_token=False,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_indexing='xy',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        init_values=0.1,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_lang_gigantic_patch14_448', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_spatial_tiny_patch16_512(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=192,
+        depth=12,
+        num_heads=3,
+        mlp_ratio=4.0,
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_spatial_tiny_patch16_512', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_spatial_small_patch16_512(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4.0,
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_spatial_small_patch16_512', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_spatial_base_patch16_512(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4.0,
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True
+    )
+    return _create_eva('vit_pe_spatial_base_patch16_512', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_spatial_large_patch14_448(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4.0,
+        attn_type='rope',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_offset=1.,
+        rope_grid_indexing='xy',
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_spatial_large_patch14_448', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def vit_pe_spatial_gigantic_patch14_448(pretrained: bool = False, **kwargs) -> Eva:
+    """Perception Encoder (PE) ViT from Meta (https://arxiv.org/abs/2504.13181)"""
+    model_args = dict(
+        patch_size=14,
+        embed_dim=1536,
+        depth=50,
+        num_heads=16,
+        mlp_ratio=8960 / 1536,
+        attn_type='rope',
+        class_token=False,
+        use_rot_pos_emb=True,
+        ref_feat_shape=(32, 32),
+        rope_grid_indexing='xy',
+        use_pre_transformer_norm=True,
+        use_post_transformer_norm=False,
+        use_fc_norm=False,  # explicitly disable
+        init_values=0.1,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+        #dynamic_img_size=True,
+    )
+    return _create_eva('vit_pe_spatial_gigantic_patch14_448', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+# RoPE-ViT models from https://github.com/naver-ai/rope-vit
+@register_model
+def vit_small_patch16_rope_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial ViT-S/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+    model = _create_eva('vit_small_patch16_rope_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_rope_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial ViT-B/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4,
+        attn_type='rope',
+        use_fc_norm=False,
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+    model = _create_eva('vit_base_patch16_rope_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_rope_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial ViT-L/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+    model = _create_eva('vit_large_patch16_rope_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_patch16_rope_mixed_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed ViT-S/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+    model = _create_eva('vit_small_patch16_rope_mixed_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_rope_mixed_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed ViT-B/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4,
+        qkv_bias=True,
+        attn_type='rope',
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+    model = _create_eva('vit_base_patch16_rope_mixed_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_rope_mixed_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed ViT-L/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        use_abs_pos_emb=False,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+    model = _create_eva('vit_large_patch16_rope_mixed_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+# APE variants (with absolute position embeddings)
+@register_model
+def vit_small_patch16_rope_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial + APE ViT-S/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+    model = _create_eva('vit_small_patch16_rope_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_rope_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial + APE ViT-B/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+
+    model = _create_eva('vit_base_patch16_rope_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_rope_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Axial + APE ViT-L/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=100.0,
+    )
+
+    model = _create_eva('vit_large_patch16_rope_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_patch16_rope_mixed_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed + APE ViT-S/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+
+    model = _create_eva('vit_small_patch16_rope_mixed_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_rope_mixed_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed + APE ViT-B/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+    model = _create_eva('vit_base_patch16_rope_mixed_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_rope_mixed_ape_224(pretrained: bool = False, **kwargs) -> Eva:
+    """RoPE-Mixed + APE ViT-L/16 from https://github.com/naver-ai/rope-vit"""
+    model_args = dict(
+        patch_size=16,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        mlp_ratio=4,
+        attn_type='rope',
+        qkv_bias=True,
+        init_values=1e-5,
+        class_token=True,
+        global_pool='token',
+        no_embed_class=True,
+        use_abs_pos_emb=True,
+        use_rot_pos_emb=True,
+        rope_grid_indexing='xy',
+        rope_temperature=10.0,
+        rope_type='mixed'
+    )
+    model = _create_eva('vit_large_patch16_rope_mixed_ape_224', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_tiny_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3-style T/16 w/ QKV bias enabled."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=192,
+        depth=12,
+        num_heads=3,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_tiny_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 S/16 https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_small_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 S/16 w/ QKV bias enabled (but zero) https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_small_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_plus_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 S/16 Plus https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        swiglu_mlp=True,
+        swiglu_align_to=8,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_small_plus_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_small_plus_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 S/16 Plus w/ QKV bias enabled (but 0) https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=384,
+        depth=12,
+        num_heads=6,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        swiglu_mlp=True,
+        swiglu_align_to=8,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_small_plus_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 B/16 https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_base_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_base_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 B/16 w/ QKV bias enabled (but zero) https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-05, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        #rope_rescale_coords=2,  # haven't added to interface
+        rope_rotate_half=True,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_base_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 L/16 https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-5, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        rope_rotate_half=True,
+        #rope_rescale_coords=2,  # haven't added to interface
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_large_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_large_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 w/ QKV bias enabled (but zero) https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=1024,
+        depth=24,
+        num_heads=16,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-5, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        rope_rotate_half=True,
+        #rope_rescale_coords=2,  # haven't added to interface
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+    model = _create_eva('vit_large_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_huge_plus_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 H/16 Plus https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=1280,
+        depth=32,
+        num_heads=20,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-5, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        rope_rotate_half=True,
+        swiglu_mlp=True,
+        swiglu_align_to=8,
+        #rope_rescale_coords=2,  # haven't added to interface
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+
+    model = _create_eva('vit_huge_plus_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+
+@register_model
+def vit_huge_plus_patch16_dinov3_qkvb(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 H/16 Plus w/ QKV bias enabled (but zero) https://arxiv.org/abs/2508.10104 NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3)."""
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=1280,
+        depth=32,
+        num_heads=20,
+        qkv_bias=True,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        init_values=1.0e-5, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        rope_rotate_half=True,
+        swiglu_mlp=True,
+        swiglu_align_to=8,
+        #rope_rescale_coords=2,  # haven't added to interface
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+
+    model = _create_eva('vit_huge_plus_patch16_dinov3_qkvb', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
+@register_model
+def vit_7b_patch16_dinov3(pretrained: bool = False, **kwargs) -> Eva:
+    """DINOv3 7B/16 https://arxiv.org/abs/2508.10104
+    NOTE: Pass global_pool='token' to use CLS-token pooling (matches upstream DINOv3).
+    """
+    model_args = dict(
+        patch_size=16,
+        dynamic_img_size=True,
+        embed_dim=4096,
+        depth=40,
+        num_heads=32,
+        qkv_bias=False,
+        # global_pool='token',  # upstream uses CLS token; default here is 'avg', pass via kwargs or --gp
+        mlp_ratio=2,
+        init_values=1.0e-5, # layer-scale
+        rope_type='dinov3',
+        rope_temperature=100,
+        use_rot_pos_emb=True,
+        use_abs_pos_emb=False,
+        rope_rotate_half=True,
+        swiglu_mlp=True,
+        swiglu_align_to=64,
+        #rope_rescale_coords=2,  # haven't added to interface
+        num_reg_tokens=4,
+        use_fc_norm=False,
+        norm_layer=partial(LayerNorm, eps=1e-5),
+    )
+
+    model = _create_eva('vit_7b_patch16_dinov3', pretrained=pretrained, **dict(model_args, **kwargs))
+    return model
+
