@@ -258,11 +258,23 @@ def get_norm_act_layer(norm_layer, act_layer=None):
     return norm_act_layer
 
 
+def _resolve_num_groups(num_channels, num_groups, group_size):
+    """group_size (channels per group) takes precedence over num_groups."""
+    if group_size:
+        assert num_channels % group_size == 0
+        return num_channels // group_size
+    return num_groups
+
+
 class GroupNormAct(GroupNorm):
     def __init__(
             self, num_channels, num_groups=32, eps=1e-5, affine=True,
+            group_size=None,
             apply_act=True, act_layer=nn.ReLU, act_kwargs=None, inplace=True, drop_layer=None):
-        super().__init__(num_channels, num_groups=num_groups, eps=eps, affine=affine)
+        super().__init__(
+            num_channels,
+            num_groups=_resolve_num_groups(num_channels, num_groups, group_size),
+            eps=eps, affine=affine)
         self.drop = drop_layer() if drop_layer is not None else nn.Identity()
         self.act = _create_act(act_layer, act_kwargs=act_kwargs, inplace=inplace, apply_act=apply_act)
 
