@@ -20,10 +20,11 @@ from typing import Any, Callable, Dict, List, Optional, Sequence, Tuple, Type, U
 import torch
 import torch.nn as nn
 
-from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
+from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD, IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD, OPENAI_CLIP_MEAN, OPENAI_CLIP_STD
 from ..layers import (
-    AvgPool2dSame, BatchNormAct2d, ClassifierHead, ConvNormAct, DropBlock2d, DropPath, create_conv2d,
-    get_act_layer, get_attn, get_norm_act_layer, make_divisible, to_2tuple,
+    AttentionPool2d, AvgPool2dSame, BatchNormAct2d, ClassifierHead, ConvNormAct, DropBlock2d, DropPath,
+    EvoNorm2dS0a, NormMlpClassifierHead, RotAttentionPool2d, create_conv2d, get_act_layer, get_attn,
+    get_norm_act_layer, make_divisible, to_2tuple,
 )
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
@@ -67,11 +68,17 @@ class ByoModelCfg:
     act_layer: str = 'relu'
     norm_layer: str = 'batchnorm'
 
+    aa_layer: str = ''
+
     attn_layer: Optional[str] = None
     attn_kwargs: dict = field(default_factory=lambda: dict())
     self_attn_layer: Optional[str] = None
     self_attn_kwargs: dict = field(default_factory=lambda: dict())
     block_kwargs: Dict[str, Any] = field(default_factory=lambda: dict())
+
+    # head: 'classifier' (pool+fc), 'mlp' (norm+mlp), 'attn_abs'/'attn_rot' (CLIP attention pooling)
+    head_hidden_size: Optional[int] = None
+    head_type: str = 'classifier'
 
 
 def _rep_vgg_bcfg(
@@ -621,6 +628,125 @@ class SelfAttnBlock(nn.Module):
         return self.act(x)
 
 
+
+
+class MobileOneBlock(nn.Module):
+    """MobileOne over-parameterized block (arxiv 2206.04040): N parallel kxk
+    conv branches + a 1x1 scale branch + BN identity at train time, folded
+    into a single conv by `reparameterize()` (reference `byobnet.py:848`)."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            kernel_size: int = 3,
+            stride: int = 1,
+            dilation: Tuple[int, int] = (1, 1),
+            bottle_ratio: float = 1.0,  # unused, byob interface
+            group_size: Optional[int] = None,
+            downsample: str = '',  # unused, byob interface
+            inference_mode: bool = False,
+            num_conv_branches: int = 1,
+            layers: LayerFn = None,
+            drop_block: Callable = None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        self.num_conv_branches = num_conv_branches
+        self.groups = groups = num_groups(group_size, in_chs)
+        layers = layers or LayerFn()
+
+        if inference_mode:
+            self.reparam_conv = nn.Conv2d(
+                in_channels=in_chs, out_channels=out_chs, kernel_size=kernel_size,
+                stride=stride, padding=kernel_size // 2, dilation=dilation[0],
+                groups=groups, bias=True)
+        else:
+            self.reparam_conv = None
+            use_ident = in_chs == out_chs and stride == 1 and dilation[0] == dilation[1]
+            self.identity = layers.norm_act(out_chs, apply_act=False) if use_ident else None
+            self.conv_kxk = nn.ModuleList([
+                layers.conv_norm_act(
+                    in_chs, out_chs, kernel_size=kernel_size, stride=stride,
+                    groups=groups, apply_act=False)
+                for _ in range(num_conv_branches)
+            ])
+            self.conv_scale = None
+            if kernel_size > 1:
+                self.conv_scale = layers.conv_norm_act(
+                    in_chs, out_chs, kernel_size=1, stride=stride, groups=groups, apply_act=False)
+            self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0. and use_ident else nn.Identity()
+
+        self.attn = nn.Identity() if layers.attn is None else layers.attn(out_chs)
+        self.act = layers.act(inplace=True)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.reparam_conv is not None:
+            return self.act(self.attn(self.reparam_conv(x)))
+        identity_out = self.identity(x) if self.identity is not None else 0
+        out = self.conv_scale(x) if self.conv_scale is not None else 0
+        for branch in self.conv_kxk:
+            out = out + branch(x)
+        out = self.drop_path(out) + identity_out
+        return self.act(self.attn(out))
+
+    def reparameterize(self):
+        """Fold all branches into one dense conv (RepVGG-style BN fusion)."""
+        if self.reparam_conv is not None:
+            return
+        first = self.conv_kxk[0].conv
+        ksize = first.kernel_size[0]
+        pad = ksize // 2
+
+        kernel, bias = 0, 0
+        if self.conv_scale is not None:
+            k1, b1 = _fuse_conv_bn_branch(self.conv_scale, self.groups)
+            kernel = torch.nn.functional.pad(k1, [pad, pad, pad, pad])
+            bias = b1
+        if self.identity is not None:
+            ki, bi = _fuse_conv_bn_branch(self.identity, self.groups, kernel_size=ksize)
+            kernel = kernel + ki
+            bias = bias + bi
+        for branch in self.conv_kxk:
+            kb, bb = _fuse_conv_bn_branch(branch, self.groups)
+            kernel = kernel + kb
+            bias = bias + bb
+
+        self.reparam_conv = nn.Conv2d(
+            in_channels=first.in_channels, out_channels=first.out_channels,
+            kernel_size=first.kernel_size, stride=first.stride,
+            padding=first.padding, dilation=first.dilation,
+            groups=first.groups, bias=True)
+        self.reparam_conv.weight.data = kernel
+        self.reparam_conv.bias.data = bias
+        for name, para in self.named_parameters():
+            if 'reparam_conv' in name:
+                continue
+            para.detach_()
+        for attr in ('conv_kxk', 'conv_scale', 'identity', 'drop_path'):
+            self.__delattr__(attr)
+
+
+def _fuse_conv_bn_branch(branch, groups: int, kernel_size: int = 3) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Fold a ConvNormAct (or bare BN identity) branch into (kernel, bias)."""
+    if isinstance(branch, ConvNormAct):
+        kernel = branch.conv.weight
+        bn = branch.bn
+    else:
+        # identity BN: synthesize a center-tap identity kernel
+        bn = branch
+        in_chs = bn.running_mean.shape[0]
+        input_dim = in_chs // groups
+        kernel = torch.zeros(
+            (in_chs, input_dim, kernel_size, kernel_size),
+            dtype=bn.weight.dtype, device=bn.weight.device)
+        for i in range(in_chs):
+            kernel[i, i % input_dim, kernel_size // 2, kernel_size // 2] = 1
+    std = (bn.running_var + bn.eps).sqrt()
+    scale = (bn.weight / std).reshape(-1, 1, 1, 1)
+    return kernel * scale, bn.bias - bn.running_mean * bn.weight / std
+
+
 _block_registry = dict(
     basic=BasicBlock,
     bottle=BottleneckBlock,
@@ -628,6 +754,7 @@ _block_registry = dict(
     edge=EdgeBlock,
     rep=RepVggBlock,
     self_attn=SelfAttnBlock,
+    one=MobileOneBlock,
 )
 
 
@@ -720,7 +847,7 @@ def create_byob_stem(
         layers: LayerFn = None,
 ):
     layers = layers or LayerFn()
-    assert stem_type in ('', 'quad', 'quad2', 'tiered', 'deep', 'rep', '7x7', '3x3')
+    assert stem_type in ('', 'quad', 'quad2', 'tiered', 'deep', 'rep', 'one', '7x7', '3x3')
     if 'quad' in stem_type:
         num_act = 2 if 'quad2' in stem_type else None
         stem = Stem(in_chs, out_chs, num_rep=4, num_act=num_act, pool=pool_type, layers=layers)
@@ -730,6 +857,8 @@ def create_byob_stem(
         stem = Stem(in_chs, out_chs, num_rep=3, chs_decay=1.0, pool=pool_type, layers=layers)
     elif 'rep' in stem_type:
         stem = RepVggBlock(in_chs, out_chs, stride=2, layers=layers)
+    elif 'one' in stem_type:
+        stem = MobileOneBlock(in_chs, out_chs, kernel_size=3, stride=2, layers=layers)
     elif '7x7' in stem_type:
         if pool_type:
             stem = Stem(in_chs, out_chs, 7, num_rep=1, pool=pool_type, layers=layers)
@@ -865,10 +994,13 @@ def create_byob_stages(
     return nn.Sequential(*stages), feature_info
 
 
-def get_layer_fns(cfg: ByoModelCfg) -> LayerFn:
+def get_layer_fns(cfg: ByoModelCfg, allow_aa: bool = True) -> LayerFn:
     act = get_act_layer(cfg.act_layer)
     norm_act = get_norm_act_layer(norm_layer=cfg.norm_layer, act_layer=act)
-    conv_norm_act = partial(ConvNormAct, norm_layer=cfg.norm_layer, act_layer=act)
+    if cfg.aa_layer and allow_aa:
+        conv_norm_act = partial(ConvNormAct, norm_layer=cfg.norm_layer, act_layer=act, aa_layer=cfg.aa_layer)
+    else:
+        conv_norm_act = partial(ConvNormAct, norm_layer=cfg.norm_layer, act_layer=act)
     attn = partial(get_attn(cfg.attn_layer), **cfg.attn_kwargs) if cfg.attn_layer else None
     self_attn = partial(get_attn(cfg.self_attn_layer), **cfg.self_attn_kwargs) if cfg.self_attn_layer else None
     return LayerFn(conv_norm_act=conv_norm_act, norm_act=norm_act, act=act, attn=attn, self_attn=self_attn)
@@ -898,6 +1030,7 @@ class ByobNet(nn.Module):
         self.grad_checkpointing = False
 
         cfg = replace(cfg, **kwargs)
+        stem_layers = get_layer_fns(cfg, allow_aa=False)
         layers = get_layer_fns(cfg)
         if cfg.fixed_input_size:
             assert img_size is not None, 'img_size argument is required for fixed input size model'
@@ -913,7 +1046,7 @@ class ByobNet(nn.Module):
             out_chs=stem_chs,
             stem_type=cfg.stem_type,
             pool_type=cfg.stem_pool,
-            layers=layers,
+            layers=stem_layers,
         )
         self.feature_info.extend(stem_feat[:-1])
 
@@ -930,6 +1063,7 @@ class ByobNet(nn.Module):
         )
         self.feature_info.extend(stage_feat[:-1])
         reduction = stage_feat[-1]['reduction']
+        head_feat_size = reduce_feat_size(to_2tuple(img_size), reduction) if img_size is not None else None
 
         prev_chs = stage_feat[-1]['num_chs']
         if cfg.num_features:
@@ -943,14 +1077,58 @@ class ByobNet(nn.Module):
         self.stage_ends = [f['stage'] for f in self.feature_info]
 
         self.head_hidden_size = self.num_features
-        if global_pool is None:
-            global_pool = 'avg'
-        self.head = ClassifierHead(
-            self.num_features,
-            num_classes,
-            pool_type=global_pool,
-            drop_rate=self.drop_rate,
-        )
+        assert cfg.head_type in ('', 'classifier', 'mlp', 'attn_abs', 'attn_rot')
+        if cfg.head_type == 'mlp':
+            if global_pool is None:
+                global_pool = 'avg'
+            self.head = NormMlpClassifierHead(
+                self.num_features,
+                num_classes,
+                hidden_size=cfg.head_hidden_size,
+                pool_type=global_pool,
+                norm_layer=cfg.norm_layer,
+                act_layer=cfg.act_layer,
+                drop_rate=self.drop_rate,
+            )
+            self.head_hidden_size = self.head.hidden_size
+        elif cfg.head_type == 'attn_abs':
+            if global_pool is None:
+                global_pool = 'token'
+            assert global_pool in ('', 'token')
+            self.head = AttentionPool2d(
+                self.num_features,
+                embed_dim=cfg.head_hidden_size,
+                out_features=num_classes,
+                feat_size=head_feat_size if head_feat_size is not None else 7,
+                pool_type=global_pool,
+                drop_rate=self.drop_rate,
+                qkv_separate=True,
+            )
+            self.head_hidden_size = self.head.embed_dim
+        elif cfg.head_type == 'attn_rot':
+            if global_pool is None:
+                global_pool = 'token'
+            assert global_pool in ('', 'token')
+            self.head = RotAttentionPool2d(
+                self.num_features,
+                embed_dim=cfg.head_hidden_size,
+                out_features=num_classes,
+                ref_feat_size=head_feat_size if head_feat_size is not None else 7,
+                pool_type=global_pool,
+                drop_rate=self.drop_rate,
+                qkv_separate=True,
+            )
+            self.head_hidden_size = self.head.embed_dim
+        else:
+            if global_pool is None:
+                global_pool = 'avg'
+            assert cfg.head_hidden_size is None
+            self.head = ClassifierHead(
+                self.num_features,
+                num_classes,
+                pool_type=global_pool,
+                drop_rate=self.drop_rate,
+            )
         self.global_pool = global_pool
 
         named_apply(partial(_init_weights, zero_init_last=zero_init_last), self)
@@ -1061,6 +1239,43 @@ def _init_weights(module: nn.Module, name: str = '', zero_init_last: bool = Fals
     elif isinstance(module, nn.BatchNorm2d):
         nn.init.ones_(module.weight)
         nn.init.zeros_(module.bias)
+
+
+def _mobileone_bcfg(
+        d: Tuple[int, ...] = (2, 8, 10, 1),
+        wf: Tuple[float, ...] = (1., 1., 1., 1.),
+        se_blocks: Tuple[int, ...] = (),
+        num_conv_branches: int = 1
+) -> List[List[ByoBlockCfg]]:
+    """Create MobileOne block configuration.
+
+    Args:
+        d: Depth (number of blocks) per stage.
+        wf: Width factor per stage.
+        se_blocks: Number of SE blocks per stage.
+        num_conv_branches: Number of conv branches.
+
+    Returns:
+        List of block configurations per stage.
+    """
+    c = (64, 128, 256, 512)
+    prev_c = min(64, c[0] * wf[0])
+    se_blocks = se_blocks or (0,) * len(d)
+    bcfg = []
+    for d, c, w, se in zip(d, c, wf, se_blocks):
+        scfg = []
+        for i in range(d):
+            out_c = c * w
+            bk = dict(num_conv_branches=num_conv_branches)
+            ak = {}
+            if i >= d - se:
+                ak['attn_layer'] = 'se'
+            scfg += [ByoBlockCfg(type='one', d=1, c=prev_c, gs=1, block_kwargs=bk, **ak)]  # depthwise block
+            scfg += [ByoBlockCfg(
+                type='one', d=1, c=out_c, gs=0, block_kwargs=dict(kernel_size=1, **bk), **ak)]  # pointwise block
+            prev_c = out_c
+        bcfg += [scfg]
+    return bcfg
 
 
 model_cfgs = dict(
@@ -1307,7 +1522,330 @@ model_cfgs = dict(
         act_layer='silu',
         attn_layer='eca',
     ),
+
+    bat_resnext26ts=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=2, c=256, s=1, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=2, c=512, s=2, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=2, c=1024, s=2, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=2, c=2048, s=2, gs=32, br=0.25),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='maxpool',
+        act_layer='silu',
+        attn_layer='bat',
+        attn_kwargs=dict(block_size=8)
+    ),
+
+    gcresnet50t=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=4, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=3, c=2048, s=2, br=0.25),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='',
+        attn_layer='gca',
+    ),
+
+    gcresnext50ts=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=4, c=512, s=2, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=1024, s=2, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=3, c=2048, s=2, gs=32, br=0.25),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='maxpool',
+        act_layer='silu',
+        attn_layer='gca',
+    ),
+
+    regnetz_b16=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=2, c=48, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=6, c=96, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=12, c=192, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=2, c=288, s=2, gs=16, br=3),
+        ),
+        stem_chs=32,
+        stem_pool='',
+        downsample='',
+        num_features=1536,
+        act_layer='silu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_b16_evos=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=2, c=48, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=6, c=96, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=12, c=192, s=2, gs=16, br=3),
+            ByoBlockCfg(type='bottle', d=2, c=288, s=2, gs=16, br=3),
+        ),
+        stem_chs=32,
+        stem_pool='',
+        downsample='',
+        num_features=1536,
+        act_layer='silu',
+        norm_layer=partial(EvoNorm2dS0a, group_size=16),
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_c16=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=2, c=48, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=6, c=96, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=12, c=192, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=2, c=288, s=2, gs=16, br=4),
+        ),
+        stem_chs=32,
+        stem_pool='',
+        downsample='',
+        num_features=1536,
+        act_layer='silu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_c16_evos=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=2, c=48, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=6, c=96, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=12, c=192, s=2, gs=16, br=4),
+            ByoBlockCfg(type='bottle', d=2, c=288, s=2, gs=16, br=4),
+        ),
+        stem_chs=32,
+        stem_pool='',
+        downsample='',
+        num_features=1536,
+        act_layer='silu',
+        norm_layer=partial(EvoNorm2dS0a, group_size=16),
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_d32=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=64, s=1, gs=32, br=4),
+            ByoBlockCfg(type='bottle', d=6, c=128, s=2, gs=32, br=4),
+            ByoBlockCfg(type='bottle', d=12, c=256, s=2, gs=32, br=4),
+            ByoBlockCfg(type='bottle', d=3, c=384, s=2, gs=32, br=4),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='',
+        downsample='',
+        num_features=1792,
+        act_layer='silu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_d8=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=64, s=1, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=6, c=128, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=12, c=256, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=3, c=384, s=2, gs=8, br=4),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='',
+        downsample='',
+        num_features=1792,
+        act_layer='silu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_d8_evos=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=64, s=1, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=6, c=128, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=12, c=256, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=3, c=384, s=2, gs=8, br=4),
+        ),
+        stem_chs=64,
+        stem_type='deep',
+        stem_pool='',
+        downsample='',
+        num_features=1792,
+        act_layer='silu',
+        norm_layer=partial(EvoNorm2dS0a, group_size=16),
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    regnetz_e8=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=96, s=1, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=8, c=192, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=16, c=384, s=2, gs=8, br=4),
+            ByoBlockCfg(type='bottle', d=3, c=512, s=2, gs=8, br=4),
+        ),
+        stem_chs=64,
+        stem_type='tiered',
+        stem_pool='',
+        downsample='',
+        num_features=2048,
+        act_layer='silu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+        block_kwargs=dict(bottle_in=True, linear_out=True),
+    ),
+
+    resnet50_clip=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=4, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=3, c=2048, s=2, br=0.25),
+        ),
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_type='attn_abs',
+    ),
+
+    resnet101_clip=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=4, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=23, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=3, c=2048, s=2, br=0.25),
+        ),
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_type='attn_abs',
+    ),
+
+    resnet50x4_clip=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=4, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=10, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=2048, s=2, br=0.25),
+        ),
+        width_factor=1.25,
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_type='attn_abs',
+    ),
+
+    resnet50x16_clip=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=6, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=8, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=18, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=8, c=2048, s=2, br=0.25),
+        ),
+        width_factor=1.5,
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_type='attn_abs',
+    ),
+
+    resnet50x64_clip=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=15, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=36, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=10, c=2048, s=2, br=0.25),
+        ),
+        width_factor=2.0,
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_type='attn_abs',
+    ),
+
+    resnet50_mlp=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='bottle', d=3, c=256, s=1, br=0.25),
+            ByoBlockCfg(type='bottle', d=4, c=512, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=6, c=1024, s=2, br=0.25),
+            ByoBlockCfg(type='bottle', d=3, c=2048, s=2, br=0.25),
+        ),
+        stem_chs=(32, 32, 64),
+        stem_type='',
+        stem_pool='avg2',
+        downsample='avg',
+        aa_layer='avg',
+        head_hidden_size=1024,
+        head_type='mlp',
+    ),
+
+    test_byobnet=ByoModelCfg(
+        blocks=(
+            ByoBlockCfg(type='edge', d=1, c=32, s=2, gs=0, br=0.5),
+            ByoBlockCfg(type='dark', d=1, c=64, s=2, gs=0, br=0.5),
+            ByoBlockCfg(type='basic', d=1, c=128, s=2, gs=32, br=0.25),
+            ByoBlockCfg(type='bottle', d=1, c=256, s=2, gs=64, br=0.25),
+        ),
+        stem_chs=24,
+        downsample='avg',
+        stem_pool='',
+        act_layer='relu',
+        attn_layer='se',
+        attn_kwargs=dict(rd_ratio=0.25),
+    ),
+
+    mobileone_s0=ByoModelCfg(
+        blocks=_mobileone_bcfg(wf=(0.75, 1.0, 1.0, 2.), num_conv_branches=4),
+        stem_type='one',
+        stem_chs=48,
+    ),
+    mobileone_s1=ByoModelCfg(
+        blocks=_mobileone_bcfg(wf=(1.5, 1.5, 2.0, 2.5)),
+        stem_type='one',
+        stem_chs=64,
+    ),
+    mobileone_s2=ByoModelCfg(
+        blocks=_mobileone_bcfg(wf=(1.5, 2.0, 2.5, 4.0)),
+        stem_type='one',
+        stem_chs=64,
+    ),
+    mobileone_s3=ByoModelCfg(
+        blocks=_mobileone_bcfg(wf=(2.0, 2.5, 3.0, 4.0)),
+        stem_type='one',
+        stem_chs=64,
+    ),
+    mobileone_s4=ByoModelCfg(
+        blocks=_mobileone_bcfg(wf=(3.0, 3.5, 3.5, 4.0), se_blocks=(0, 0, 5, 1)),
+        stem_type='one',
+        stem_chs=64,
+    ),
 )
+for _k in ('resnet50_clip', 'resnet101_clip', 'resnet50x4_clip', 'resnet50x16_clip', 'resnet50x64_clip'):
+    model_cfgs[_k + '_gap'] = replace(model_cfgs[_k], head_type='classifier')
+
 
 
 def _create_byobnet(variant: str, pretrained: bool = False, **kwargs) -> ByobNet:
@@ -1340,42 +1878,573 @@ def _cfgr(url: str = '', **kwargs) -> Dict[str, Any]:
 
 
 default_cfgs = generate_default_cfgs({
-    'gernet_s.idstcv_in1k': _cfg(),
-    'gernet_m.idstcv_in1k': _cfg(),
-    'gernet_l.idstcv_in1k': _cfg(input_size=(3, 256, 256), pool_size=(8, 8)),
-    'repvgg_a0.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_a1.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_a2.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b0.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b1.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b1g4.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b2.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b2g4.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b3.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
-    'repvgg_b3g4.rvgg_in1k': _cfg(first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv')),
+    # GPU-Efficient (ResNet) weights
+    'gernet_s.idstcv_in1k': _cfg(hf_hub_id='timm/'),
+    'gernet_m.idstcv_in1k': _cfg(hf_hub_id='timm/'),
+    'gernet_l.idstcv_in1k': _cfg(hf_hub_id='timm/', input_size=(3, 256, 256), pool_size=(8, 8)),
+
+    # RepVGG weights
+    'repvgg_a0.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_a1.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_a2.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b0.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b1.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b1g4.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b2.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b2g4.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b3.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
+    'repvgg_b3g4.rvgg_in1k': _cfg(
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit'),
     'repvgg_d2se.rvgg_in1k': _cfg(
-        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'),
-        input_size=(3, 320, 320), pool_size=(10, 10), crop_pct=1.0),
-    'resnet51q.ra2_in1k': _cfgr(first_conv='stem.conv1', test_input_size=(3, 288, 288)),
-    'resnet61q.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'resnext26ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'seresnext26ts.ch_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'gcresnext26ts.ch_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'eca_resnext26ts.ch_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'resnet32ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'resnet33ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'gcresnet33ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'seresnet33ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
-    'eca_resnet33ts.ra2_in1k': _cfgr(test_input_size=(3, 288, 288)),
+        hf_hub_id='timm/',
+        first_conv=('stem.conv_kxk.conv', 'stem.conv_1x1.conv'), license='mit',
+        input_size=(3, 320, 320), pool_size=(10, 10), crop_pct=1.0,
+    ),
+
+    # experimental ResNet configs
+    'resnet51q.ra2_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/resnet51q_ra2-d47dcc76.pth',
+        first_conv='stem.conv1', input_size=(3, 256, 256), pool_size=(8, 8),
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnet61q.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/resnet61q_ra2-6afc536c.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+
+    # ResNeXt-26 models with different attention in Bottleneck blocks
+    'resnext26ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/resnext26ts_256_ra2-8bbd9106.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'seresnext26ts.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/seresnext26ts_256-6f0d74a3.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'gcresnext26ts.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/gcresnext26ts_256-e414378b.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'eca_resnext26ts.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/eca_resnext26ts_256-5a1d030f.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'bat_resnext26ts.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/bat_resnext26ts_256-fa6fd595.pth',
+        min_input_size=(3, 256, 256)),
+
+    # ResNet-32 / 33 models with different attention in Bottleneck blocks
+    'resnet32ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/resnet32ts_256-aacf5250.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnet33ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/resnet33ts_256-e91b09a4.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'gcresnet33ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/gcresnet33ts_256-0e0cd345.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'seresnet33ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/seresnet33ts_256-f8ad44d9.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'eca_resnet33ts.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/eca_resnet33ts_256-8f98face.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+
+    'gcresnet50t.ra2_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/gcresnet50t_256-96374d1c.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+
+    'gcresnext50ts.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/gcresnext50ts_256-3e0f515e.pth',
+        test_input_size=(3, 288, 288), test_crop_pct=1.0),
+
+    # custom `timm` specific RegNetZ inspired models w/ different sizing from paper
+    'regnetz_b16.ra3_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/regnetz_b_raa-677d9606.pth',
+        first_conv='stem.conv', mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        input_size=(3, 224, 224), pool_size=(7, 7), crop_pct=0.94, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'regnetz_c16.ra3_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/regnetz_c_rab2_256-a54bf36a.pth',
+        first_conv='stem.conv', mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        crop_pct=0.94, test_input_size=(3, 320, 320), test_crop_pct=1.0),
+    'regnetz_d32.ra3_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/regnetz_d_rab_256-b8073a89.pth',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5), crop_pct=0.95, test_input_size=(3, 320, 320)),
+    'regnetz_d8.ra3_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/regnetz_d8_bh-afc03c55.pth',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5), crop_pct=0.94, test_input_size=(3, 320, 320), test_crop_pct=1.0),
+    'regnetz_e8.ra3_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-attn-weights/regnetz_e8_bh-aace8e6e.pth',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5), crop_pct=0.94, test_input_size=(3, 320, 320), test_crop_pct=1.0),
+
+    'regnetz_b16_evos.untrained': _cfgr(
+        first_conv='stem.conv', mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        input_size=(3, 224, 224), pool_size=(7, 7), crop_pct=0.95, test_input_size=(3, 288, 288)),
+    'regnetz_c16_evos.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetz_c16_evos_ch-d8311942.pth',
+        first_conv='stem.conv', mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        crop_pct=0.95, test_input_size=(3, 320, 320)),
+    'regnetz_d8_evos.ch_in1k': _cfgr(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetz_d8_evos_ch-2bc12646.pth',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5), crop_pct=0.95, test_input_size=(3, 320, 320), test_crop_pct=1.0),
+
+    'mobileone_s0.apple_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=0.875,
+        first_conv=('stem.conv_kxk.0.conv', 'stem.conv_scale.conv'),
+        license='mobileone-license',
+    ),
+    'mobileone_s1.apple_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=0.9,
+        first_conv=('stem.conv_kxk.0.conv', 'stem.conv_scale.conv'),
+        license='mobileone-license',
+    ),
+    'mobileone_s2.apple_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=0.9,
+        first_conv=('stem.conv_kxk.0.conv', 'stem.conv_scale.conv'),
+        license='mobileone-license',
+    ),
+    'mobileone_s3.apple_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=0.9,
+        first_conv=('stem.conv_kxk.0.conv', 'stem.conv_scale.conv'),
+        license='mobileone-license',
+    ),
+    'mobileone_s4.apple_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=0.9,
+        first_conv=('stem.conv_kxk.0.conv', 'stem.conv_scale.conv'),
+        license='mobileone-license',
+    ),
+
+    # original attention pool head variants
+    'resnet50_clip.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=1024, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 224, 224), pool_size=(7, 7),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet101_clip.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=512, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 224, 224), pool_size=(7, 7),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet50x4_clip.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=640, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 288, 288), pool_size=(9, 9),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet50x16_clip.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=768, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 384, 384), pool_size=(12, 12),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet50x64_clip.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=1024, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 448, 448), pool_size=(14, 14),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet50_clip.cc12m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=1024, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 224, 224), pool_size=(7, 7),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet50_clip.yfcc15m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=1024, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 224, 224), pool_size=(7, 7),
+        classifier='head.proj',
+        license='mit',
+    ),
+    'resnet101_clip.yfcc15m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=512, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        fixed_input_size=True, input_size=(3, 224, 224), pool_size=(7, 7),
+        classifier='head.proj',
+        license='mit',
+    ),
+
+    # avg-pool w/ optional standard classifier head variants
+    'resnet50_clip_gap.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 224, 224), pool_size=(7, 7),
+        license='mit',
+    ),
+    'resnet101_clip_gap.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 224, 224), pool_size=(7, 7),
+        license='mit',
+    ),
+    'resnet50x4_clip_gap.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 288, 288), pool_size=(9, 9),
+        license='mit',
+    ),
+    'resnet50x16_clip_gap.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 384, 384), pool_size=(12, 12),
+        license='mit',
+    ),
+    'resnet50x64_clip_gap.openai': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 448, 448), pool_size=(14, 14),
+        license='mit',
+    ),
+    'resnet50_clip_gap.cc12m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 224, 224), pool_size=(7, 7),
+        license='mit',
+    ),
+    'resnet50_clip_gap.yfcc15m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 224, 224), pool_size=(7, 7),
+        license='mit',
+    ),
+    'resnet101_clip_gap.yfcc15m': _cfgr(
+        hf_hub_id='timm/',
+        num_classes=0, mean=OPENAI_CLIP_MEAN, std=OPENAI_CLIP_STD,
+        input_size=(3, 224, 224), pool_size=(7, 7),
+        license='mit',
+    ),
+
+    'resnet50_mlp.untrained': _cfgr(
+        input_size=(3, 256, 256), pool_size=(8, 8),
+    ),
+
+    'test_byobnet.r160_in1k': _cfgr(
+        hf_hub_id='timm/',
+        first_conv='stem.conv',
+        input_size=(3, 160, 160), crop_pct=0.95, pool_size=(5, 5),
+    ),
 })
 
 
-def _register(name):
-    def fn(pretrained: bool = False, **kwargs) -> ByobNet:
-        return _create_byobnet(name, pretrained=pretrained, **kwargs)
-    fn.__name__ = name
-    return register_model(fn)
+@register_model
+def gernet_l(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gernet_l', pretrained=pretrained, **kwargs)
 
 
-for _name in list(model_cfgs):
-    globals()[_name] = _register(_name)
+@register_model
+def gernet_m(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gernet_m', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def gernet_s(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gernet_s', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_a0(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_a0', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_a1(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_a1', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_a2(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_a2', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b0(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b0', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b1(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b1', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b1g4(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b1g4', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b2(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b2', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b2g4(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b2g4', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b3(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b3', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_b3g4(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_b3g4', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def repvgg_d2se(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('repvgg_d2se', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet51q(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnet51q', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet61q(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnet61q', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnext26ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnext26ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def gcresnext26ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gcresnext26ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def seresnext26ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('seresnext26ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def eca_resnext26ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('eca_resnext26ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet32ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnet32ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet33ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnet33ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def gcresnet33ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gcresnet33ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def seresnet33ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('seresnet33ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def eca_resnet33ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('eca_resnet33ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def bat_resnext26ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('bat_resnext26ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def gcresnet50t(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gcresnet50t', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def gcresnext50ts(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('gcresnext50ts', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_b16(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_b16', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_c16(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_c16', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_d32(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_d32', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_d8(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_d8', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_e8(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_e8', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_b16_evos(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_b16_evos', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_c16_evos(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_c16_evos', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def regnetz_d8_evos(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('regnetz_d8_evos', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def mobileone_s0(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('mobileone_s0', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def mobileone_s1(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('mobileone_s1', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def mobileone_s2(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('mobileone_s2', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def mobileone_s3(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('mobileone_s3', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def mobileone_s4(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('mobileone_s4', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50_clip(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50 CLIP image tower"""
+    return _create_byobnet('resnet50_clip', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet101_clip(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-101 CLIP image tower"""
+    return _create_byobnet('resnet101_clip', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x4_clip(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x4 CLIP image tower"""
+    return _create_byobnet('resnet50x4_clip', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x16_clip(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x16 CLIP image tower"""
+    return _create_byobnet('resnet50x16_clip', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x64_clip(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x64 CLIP image tower"""
+    return _create_byobnet('resnet50x64_clip', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50_clip_gap(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50 CLIP image tower w/ avg pool (no attention pool)"""
+    return _create_byobnet('resnet50_clip_gap', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet101_clip_gap(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-101 CLIP image tower w/ avg pool (no attention pool)"""
+    return _create_byobnet('resnet101_clip_gap', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x4_clip_gap(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x4 CLIP image tower w/ avg pool (no attention pool)"""
+    return _create_byobnet('resnet50x4_clip_gap', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x16_clip_gap(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x16 CLIP image tower w/ avg pool (no attention pool)"""
+    return _create_byobnet('resnet50x16_clip_gap', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50x64_clip_gap(pretrained=False, **kwargs) -> ByobNet:
+    """OpenAI Modified ResNet-50x64 CLIP image tower w/ avg pool (no attention pool)"""
+    return _create_byobnet('resnet50x64_clip_gap', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def resnet50_mlp(pretrained=False, **kwargs) -> ByobNet:
+    return _create_byobnet('resnet50_mlp', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def test_byobnet(pretrained=False, **kwargs) -> ByobNet:
+    """Minimal test ResNet (BYOB based) model."""
+    return _create_byobnet('test_byobnet', pretrained=pretrained, **kwargs)
+

@@ -469,16 +469,37 @@ def _cfg(url='', **kwargs):
 
 
 default_cfgs = generate_default_cfgs({
-    'focalnet_tiny_srf.ms_in1k': _cfg(),
-    'focalnet_small_srf.ms_in1k': _cfg(),
-    'focalnet_base_srf.ms_in1k': _cfg(),
-    'focalnet_tiny_lrf.ms_in1k': _cfg(),
-    'focalnet_small_lrf.ms_in1k': _cfg(),
-    'focalnet_base_lrf.ms_in1k': _cfg(),
-    'focalnet_large_fl3.ms_in22k': _cfg(input_size=(3, 384, 384), crop_pct=1.0, num_classes=21842),
-    'focalnet_large_fl4.ms_in22k': _cfg(input_size=(3, 384, 384), crop_pct=1.0, num_classes=21842),
-    'focalnet_xlarge_fl3.ms_in22k': _cfg(input_size=(3, 384, 384), crop_pct=1.0, num_classes=21842),
-    'focalnet_xlarge_fl4.ms_in22k': _cfg(input_size=(3, 384, 384), crop_pct=1.0, num_classes=21842),
+    "focalnet_tiny_srf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+    "focalnet_small_srf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+    "focalnet_base_srf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+    "focalnet_tiny_lrf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+    "focalnet_small_lrf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+    "focalnet_base_lrf.ms_in1k": _cfg(
+        hf_hub_id='timm/'),
+
+    "focalnet_large_fl3.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0, num_classes=21842),
+    "focalnet_large_fl4.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0, num_classes=21842),
+    "focalnet_xlarge_fl3.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0, num_classes=21842),
+    "focalnet_xlarge_fl4.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0, num_classes=21842),
+    "focalnet_huge_fl3.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        num_classes=21842),
+    "focalnet_huge_fl4.ms_in22k": _cfg(
+        hf_hub_id='timm/',
+        num_classes=0),
 })
 
 
@@ -548,3 +569,20 @@ def focalnet_xlarge_fl4(pretrained=False, **kwargs) -> FocalNet:
         depths=[2, 2, 18, 2], embed_dim=256, focal_levels=[4, 4, 4, 4],
         use_post_norm=True, use_overlap_down=True, layerscale_value=1e-4, **kwargs)
     return _create_focalnet('focalnet_xlarge_fl4', pretrained=pretrained, **model_kwargs)
+
+
+@register_model
+def focalnet_huge_fl3(pretrained=False, **kwargs) -> FocalNet:
+    model_kwargs = dict(
+        depths=[2, 2, 18, 2], embed_dim=352, focal_levels=[3, 3, 3, 3], focal_windows=[3] * 4,
+        use_post_norm=True, use_post_norm_in_modulation=True, use_overlap_down=True, layerscale_value=1e-4, **kwargs)
+    return _create_focalnet('focalnet_huge_fl3', pretrained=pretrained, **model_kwargs)
+
+
+@register_model
+def focalnet_huge_fl4(pretrained=False, **kwargs) -> FocalNet:
+    model_kwargs = dict(
+        depths=[2, 2, 18, 2], embed_dim=352, focal_levels=[4, 4, 4, 4],
+        use_post_norm=True, use_post_norm_in_modulation=True, use_overlap_down=True, layerscale_value=1e-4, **kwargs)
+    return _create_focalnet('focalnet_huge_fl4', pretrained=pretrained, **model_kwargs)
+

@@ -356,11 +356,30 @@ def _cfg(url='', **kwargs):
 
 default_cfgs = generate_default_cfgs({
     'ghostnet_050.untrained': _cfg(),
-    'ghostnet_100.in1k': _cfg(),
+    'ghostnet_100.in1k': _cfg(
+        hf_hub_id='timm/',
+        # url='https://github.com/huawei-noah/CV-backbones/releases/download/ghostnet_pth/ghostnet_1x.pth'
+    ),
     'ghostnet_130.untrained': _cfg(),
-    'ghostnetv2_100.in1k': _cfg(),
-    'ghostnetv2_130.in1k': _cfg(),
-    'ghostnetv2_160.in1k': _cfg(),
+    'ghostnetv2_100.in1k': _cfg(
+        hf_hub_id='timm/',
+        # url='https://github.com/huawei-noah/Efficient-AI-Backbones/releases/download/GhostNetV2/ck_ghostnetv2_10.pth.tar'
+    ),
+    'ghostnetv2_130.in1k': _cfg(
+        hf_hub_id='timm/',
+        # url='https://github.com/huawei-noah/Efficient-AI-Backbones/releases/download/GhostNetV2/ck_ghostnetv2_13.pth.tar'
+    ),
+    'ghostnetv2_160.in1k': _cfg(
+        hf_hub_id='timm/',
+        # url='https://github.com/huawei-noah/Efficient-AI-Backbones/releases/download/GhostNetV2/ck_ghostnetv2_16.pth.tar'
+    ),
+    'ghostnetv3_050.untrained': _cfg(),
+    'ghostnetv3_100.in1k': _cfg(
+        hf_hub_id='timm/',
+        #url='https://github.com/huawei-noah/Efficient-AI-Backbones/releases/download/GhostNetV3/ghostnetv3-1.0.pth.tar'
+    ),
+    'ghostnetv3_130.untrained': _cfg(),
+    'ghostnetv3_160.untrained': _cfg(),
 })
 
 
@@ -392,3 +411,32 @@ def ghostnetv2_130(pretrained=False, **kwargs) -> GhostNet:
 @register_model
 def ghostnetv2_160(pretrained=False, **kwargs) -> GhostNet:
     return _create_ghostnet('ghostnetv2_160', width=1.6, pretrained=pretrained, version='v2', **kwargs)
+
+
+@register_model
+def ghostnetv3_050(pretrained: bool = False, **kwargs: Any) -> GhostNet:
+    """GhostNetV3-0.5x"""
+    model = _create_ghostnet('ghostnetv3_050', width=0.5, pretrained=pretrained, version='v3', **kwargs)
+    return model
+
+
+@register_model
+def ghostnetv3_100(pretrained: bool = False, **kwargs: Any) -> GhostNet:
+    """GhostNetV3-1.0x"""
+    model = _create_ghostnet('ghostnetv3_100', width=1.0, pretrained=pretrained, version='v3', **kwargs)
+    return model
+
+
+@register_model
+def ghostnetv3_130(pretrained: bool = False, **kwargs: Any) -> GhostNet:
+    """GhostNetV3-1.3x"""
+    model = _create_ghostnet('ghostnetv3_130', width=1.3, pretrained=pretrained, version='v3', **kwargs)
+    return model
+
+
+@register_model
+def ghostnetv3_160(pretrained: bool = False, **kwargs: Any) -> GhostNet:
+    """GhostNetV3-1.6x"""
+    model = _create_ghostnet('ghostnetv3_160', width=1.6, pretrained=pretrained, version='v3', **kwargs)
+    return model
+

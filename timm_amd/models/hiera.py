@@ -746,13 +746,88 @@ def _cfg(url='', **kwargs):
 
 
 default_cfgs = generate_default_cfgs({
-    'hiera_tiny_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_small_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_base_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_base_plus_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_large_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_huge_224.mae_in1k_ft_in1k': _cfg(),
-    'hiera_small_abswin_256.sbb2_e200_in12k_ft_in1k': _cfg(input_size=(3, 256, 256), crop_pct=0.95),
+    "hiera_tiny_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_tiny_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_small_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_small_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_base_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_base_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_base_plus_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_base_plus_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_large_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_large_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_huge_224.mae_in1k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+    ),
+    "hiera_huge_224.mae": _cfg(
+        hf_hub_id='timm/',
+        license='cc-by-nc-4.0',
+        num_classes=0,
+    ),
+
+    "hiera_small_abswin_256.sbb2_e200_in12k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
+    "hiera_small_abswin_256.sbb2_pd_e200_in12k_ft_in1k": _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
+    "hiera_small_abswin_256.sbb2_e200_in12k": _cfg(
+        hf_hub_id='timm/',
+        num_classes=11821,
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
+    "hiera_small_abswin_256.sbb2_pd_e200_in12k": _cfg(
+        hf_hub_id='timm/',
+        num_classes=11821,
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
+    "hiera_base_abswin_256.untrained": _cfg(
+        # hf_hub_id='timm/',
+        input_size=(3, 256, 256), crop_pct=0.95,
+    ),
 })
 
 
@@ -831,3 +906,11 @@ def hiera_small_abswin_256(pretrained=False, **kwargs):
         init_values=1e-5, weight_init='jax', use_expand_proj=False,
     )
     return _create_hiera('hiera_small_abswin_256', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def hiera_base_abswin_256(pretrained=False, **kwargs):
+    model_args = dict(
+        embed_dim=96, num_heads=1, stages=(2, 3, 16, 3), abs_win_pos_embed=True, init_values=1e-5, weight_init='jax')
+    return _create_hiera('hiera_base_abswin_256', pretrained=pretrained, **dict(model_args, **kwargs))
+

@@ -51,6 +51,7 @@ class NaFlexVitCfg:
     pos_embed: str = 'learned'          # 'learned' | 'factorized' | 'rope' | 'learned_rope' | 'none'
     pos_embed_grid_size: Tuple[int, int] = (16, 16)
     pos_embed_interp_mode: str = 'bicubic'
+    pos_embed_ar_preserving: bool = False  # interpolate to a square grid, then crop, so AR is kept
     rope_ref_feat_shape: Optional[Tuple[int, int]] = (16, 16)
     final_norm: bool = True
     fc_norm: Optional[bool] = None
@@ -165,6 +166,7 @@ class NaFlexEmbeds(nn.Module):
             pos_embed: str = 'learned',
             pos_embed_grid_size: Tuple[int, int] = (16, 16),
             pos_embed_interp_mode: str = 'bicubic',
+            pos_embed_ar_preserving: bool = False,
             proj_norm_layer: Optional[Type[nn.Module]] = None,
             pos_drop_rate: float = 0.,
             class_token: bool = False,
@@ -178,6 +180,7 @@ class NaFlexEmbeds(nn.Module):
         self.embed_dim = embed_dim
         self.pos_embed_type = pos_embed
         self.pos_embed_interp_mode = pos_embed_interp_mode
+        self.pos_embed_ar_preserving = pos_embed_ar_preserving
         self.pos_embed_grid_size = pos_embed_grid_size
         self.num_prefix_tokens = (1 if class_token else 0) + reg_tokens
 
@@ -212,9 +215,13 @@ class NaFlexEmbeds(nn.Module):
         pe = self.pos_embed.permute(0, 3, 1, 2)  # [1, C, gh, gw]
         gh, gw = grid_size
         if (gh, gw) != tuple(pe.shape[-2:]):
+            # ar-preserving: scale the square table uniformly to cover the larger
+            # side, then crop — both axes keep the same stretch factor
+            interp_size = (max(gh, gw),) * 2 if self.pos_embed_ar_preserving else (gh, gw)
             pe = F.interpolate(
-                pe.float(), size=(gh, gw), mode=self.pos_embed_interp_mode,
+                pe.float(), size=interp_size, mode=self.pos_embed_interp_mode,
                 antialias=True, align_corners=False).to(pe.dtype)
+            pe = pe[:, :, :gh, :gw]
         pe = pe.permute(0, 2, 3, 1)  # [1, gh, gw, C]
         flat = pe.reshape(1, gh * gw, self.embed_dim).expand(B, -1, -1)
         idx = (patch_coord[..., 0].clamp(max=gh - 1) * gw + patch_coord[..., 1].clamp(max=gw - 1)).long()
@@ -298,6 +305,7 @@ class NaFlexVit(nn.Module):
             embed_dim=cfg.embed_dim,
             pos_embed=cfg.pos_embed,
             pos_embed_grid_size=tuple(cfg.pos_embed_grid_size),
+            pos_embed_ar_preserving=cfg.pos_embed_ar_preserving,
             pos_embed_interp_mode=cfg.pos_embed_interp_mode,
             proj_norm_layer=embed_norm_layer,
             pos_drop_rate=cfg.pos_drop_rate,
@@ -522,10 +530,26 @@ def _cfg(url: str = '', **kwargs):
 
 
 default_cfgs = generate_default_cfgs({
-    'naflexvit_base_patch16_gap.untrained': _cfg(),
+    'naflexvit_base_patch16_gap.e300_s576_in1k': _cfg(
+        hf_hub_id='timm/',
+    ),
+    'naflexvit_base_patch16_par_gap.e300_s576_in1k': _cfg(
+        hf_hub_id='timm/',
+    ),
+    'naflexvit_base_patch16_parfac_gap.e300_s576_in1k': _cfg(
+        hf_hub_id='timm/',
+    ),
     'naflexvit_base_patch16_map.untrained': _cfg(),
-    'naflexvit_base_patch16_siglip.untrained': _cfg(),
     'naflexvit_so150m2_patch16_reg1_gap.untrained': _cfg(),
+    'naflexvit_so150m2_patch16_reg1_map.untrained': _cfg(),
+
+    # SigLIP-2 NaFlex vit encoder weights
+    'naflexvit_base_patch16_siglip.v2_webli': _cfg(
+        hf_hub_id='timm/',
+        num_classes=0),
+    'naflexvit_so400m_patch16_siglip.v2_webli': _cfg(
+        hf_hub_id='timm/',
+        num_classes=0),
 })
 
 
@@ -577,3 +601,75 @@ def naflexvit_so150m2_patch16_reg1_gap(pretrained: bool = False, **kwargs) -> Na
         global_pool='avg', reg_tokens=1, fc_norm=True,
     )
     return _create_naflexvit('naflexvit_so150m2_patch16_reg1_gap', pretrained=pretrained, cfg=cfg, **kwargs)
+
+
+@register_model
+def naflexvit_base_patch16_par_gap(pretrained: bool = False, **kwargs) -> NaFlexVit:
+    """ViT-Base with NaFlex functionality, aspect preserving pos embed, global average pooling."""
+    cfg = NaFlexVitCfg(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        init_values=1e-5,
+        pos_embed_ar_preserving=True,
+        global_pool='avg',
+        reg_tokens=4,
+        fc_norm=True,
+    )
+    model = _create_naflexvit('naflexvit_base_patch16_par_gap', pretrained=pretrained, cfg=cfg, **kwargs)
+    return model
+
+
+@register_model
+def naflexvit_base_patch16_parfac_gap(pretrained: bool = False, **kwargs) -> NaFlexVit:
+    """ViT-Base with NaFlex functionality, aspect preserving & factorized pos embed, global average pooling."""
+    cfg = NaFlexVitCfg(
+        patch_size=16,
+        embed_dim=768,
+        depth=12,
+        num_heads=12,
+        init_values=1e-5,
+        pos_embed_ar_preserving=True,
+        pos_embed='factorized',
+        global_pool='avg',
+        reg_tokens=4,
+        fc_norm=True,
+    )
+    model = _create_naflexvit('naflexvit_base_patch16_parfac_gap', pretrained=pretrained, cfg=cfg, **kwargs)
+    return model
+
+
+@register_model
+def naflexvit_so150m2_patch16_reg1_map(pretrained: bool = False, **kwargs) -> NaFlexVit:
+    """ViT-SO150M2 with NaFlex functionality for variable aspect ratios and resolutions. This model supports: 1. Variable aspect ratios and resolutions via patch coordinates 2. Position embedding interpolation for arbitrary grid sizes 3. Explicit patch coordinates and valid token masking"""
+    cfg = NaFlexVitCfg(
+        patch_size=16,
+        embed_dim=832,
+        depth=21,
+        num_heads=13,
+        mlp_ratio=34/13,
+        init_values=1e-5,
+        qkv_bias=False,
+        reg_tokens=1,
+        global_pool='map',
+    )
+    model = _create_naflexvit('naflexvit_so150m2_patch16_reg1_map', pretrained=pretrained, cfg=cfg, **kwargs)
+    return model
+
+
+@register_model
+def naflexvit_so400m_patch16_siglip(pretrained: bool = False, **kwargs) -> NaFlexVit:
+    """ViT-SO400M with NaFlex functionality for variable aspect ratios and resolutions."""
+    cfg = NaFlexVitCfg(
+        patch_size=16,
+        embed_dim=1152,
+        depth=27,
+        num_heads=16,
+        mlp_ratio=3.7362,
+        act_layer='gelu_tanh',
+        global_pool='map',
+    )
+    model = _create_naflexvit('naflexvit_so400m_patch16_siglip', pretrained=pretrained, cfg=cfg, **kwargs)
+    return model
+

@@ -462,6 +462,9 @@ model_cfgs = dict(
     nfnet_f2=_nfnet_cfg(depths=(3, 6, 18, 9)),
     nfnet_f3=_nfnet_cfg(depths=(4, 8, 24, 12)),
     nfnet_f4=_nfnet_cfg(depths=(5, 10, 30, 15)),
+    nfnet_f5=_nfnet_cfg(depths=(6, 12, 36, 18)),
+    nfnet_f6=_nfnet_cfg(depths=(7, 14, 42, 21)),
+    nfnet_f7=_nfnet_cfg(depths=(8, 16, 48, 24)),
 
     # NFNet-L lightweight variants
     nfnet_l0=_nfnet_cfg(
@@ -476,17 +479,33 @@ model_cfgs = dict(
     eca_nfnet_l2=_nfnet_cfg(
         depths=(3, 6, 18, 9), feat_mult=2, group_size=64, bottle_ratio=0.25,
         attn_layer='eca', attn_kwargs=dict(), act_layer='silu'),
+    eca_nfnet_l3=_nfnet_cfg(
+        depths=(4, 8, 24, 12), feat_mult=2, group_size=64, bottle_ratio=0.25,
+        attn_layer='eca', attn_kwargs=dict(), act_layer='silu'),
 
     # NF-RegNet-B
     nf_regnet_b0=_nfreg_cfg(depths=(1, 3, 6, 6)),
     nf_regnet_b1=_nfreg_cfg(depths=(2, 4, 7, 7)),
     nf_regnet_b2=_nfreg_cfg(depths=(2, 4, 8, 8), channels=(56, 112, 232, 488)),
     nf_regnet_b3=_nfreg_cfg(depths=(2, 5, 9, 9), channels=(56, 128, 248, 528)),
+    nf_regnet_b4=_nfreg_cfg(depths=(2, 6, 11, 11), channels=(64, 144, 288, 616)),
+    nf_regnet_b5=_nfreg_cfg(depths=(3, 7, 14, 14), channels=(80, 168, 336, 704)),
 
     # NF-ResNet
     nf_resnet26=_nfres_cfg(depths=(2, 2, 2, 2)),
     nf_resnet50=_nfres_cfg(depths=(3, 4, 6, 3)),
     nf_resnet101=_nfres_cfg(depths=(3, 4, 23, 3)),
+
+    nf_seresnet26=_nfres_cfg(depths=(2, 2, 2, 2), attn_layer='se', attn_kwargs=dict(rd_ratio=1/16)),
+    nf_seresnet50=_nfres_cfg(depths=(3, 4, 6, 3), attn_layer='se', attn_kwargs=dict(rd_ratio=1/16)),
+    nf_seresnet101=_nfres_cfg(depths=(3, 4, 23, 3), attn_layer='se', attn_kwargs=dict(rd_ratio=1/16)),
+    nf_ecaresnet26=_nfres_cfg(depths=(2, 2, 2, 2), attn_layer='eca', attn_kwargs=dict()),
+    nf_ecaresnet50=_nfres_cfg(depths=(3, 4, 6, 3), attn_layer='eca', attn_kwargs=dict()),
+    nf_ecaresnet101=_nfres_cfg(depths=(3, 4, 23, 3), attn_layer='eca', attn_kwargs=dict()),
+
+    test_nfnet=_nfnet_cfg(
+        depths=(1, 1, 1, 1), channels=(32, 64, 96, 128), feat_mult=1.5, group_size=8, bottle_ratio=0.25,
+        attn_kwargs=dict(rd_ratio=0.25, rd_divisor=8), act_layer='silu'),
 )
 
 
@@ -511,38 +530,300 @@ def _dcfg(url: str = '', **kwargs: Any) -> Dict[str, Any]:
 
 
 default_cfgs = generate_default_cfgs({
-    'dm_nfnet_f0.dm_in1k': _dcfg(input_size=(3, 192, 192), test_input_size=(3, 256, 256), pool_size=(6, 6)),
-    'dm_nfnet_f1.dm_in1k': _dcfg(input_size=(3, 224, 224), test_input_size=(3, 320, 320), pool_size=(7, 7)),
-    'dm_nfnet_f2.dm_in1k': _dcfg(input_size=(3, 256, 256), test_input_size=(3, 352, 352), pool_size=(8, 8)),
-    'dm_nfnet_f3.dm_in1k': _dcfg(input_size=(3, 320, 320), test_input_size=(3, 416, 416), pool_size=(10, 10)),
-    'dm_nfnet_f4.dm_in1k': _dcfg(input_size=(3, 384, 384), test_input_size=(3, 512, 512), pool_size=(12, 12)),
-    'dm_nfnet_f5.dm_in1k': _dcfg(input_size=(3, 416, 416), test_input_size=(3, 544, 544), pool_size=(13, 13)),
-    'dm_nfnet_f6.dm_in1k': _dcfg(input_size=(3, 448, 448), test_input_size=(3, 576, 576), pool_size=(14, 14)),
-    'nfnet_f0.untrained': _dcfg(),
-    'nfnet_f1.untrained': _dcfg(),
-    'nfnet_f2.untrained': _dcfg(),
-    'nfnet_f3.untrained': _dcfg(),
-    'nfnet_f4.untrained': _dcfg(),
-    'nfnet_l0.ra2_in1k': _dcfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'eca_nfnet_l0.ra2_in1k': _dcfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'eca_nfnet_l1.ra2_in1k': _dcfg(input_size=(3, 256, 256), test_input_size=(3, 320, 320), pool_size=(8, 8)),
-    'eca_nfnet_l2.ra3_in1k': _dcfg(input_size=(3, 320, 320), test_input_size=(3, 384, 384), pool_size=(10, 10)),
-    'nf_regnet_b0.untrained': _dcfg(first_conv='stem.conv'),
-    'nf_regnet_b1.ra2_in1k': _dcfg(first_conv='stem.conv', test_input_size=(3, 288, 288)),
-    'nf_regnet_b2.untrained': _dcfg(first_conv='stem.conv'),
-    'nf_regnet_b3.untrained': _dcfg(first_conv='stem.conv'),
-    'nf_resnet26.untrained': _dcfg(first_conv='stem.conv'),
-    'nf_resnet50.ra2_in1k': _dcfg(first_conv='stem.conv', test_input_size=(3, 288, 288), crop_pct=0.94),
-    'nf_resnet101.untrained': _dcfg(first_conv='stem.conv'),
+    'dm_nfnet_f0.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f0-604f9c3a.pth',
+        pool_size=(6, 6), input_size=(3, 192, 192), test_input_size=(3, 256, 256), crop_pct=.9, crop_mode='squash'),
+    'dm_nfnet_f1.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f1-fc540f82.pth',
+        pool_size=(7, 7), input_size=(3, 224, 224), test_input_size=(3, 320, 320), crop_pct=0.91, crop_mode='squash'),
+    'dm_nfnet_f2.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f2-89875923.pth',
+        pool_size=(8, 8), input_size=(3, 256, 256), test_input_size=(3, 352, 352), crop_pct=0.92, crop_mode='squash'),
+    'dm_nfnet_f3.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f3-d74ab3aa.pth',
+        pool_size=(10, 10), input_size=(3, 320, 320), test_input_size=(3, 416, 416), crop_pct=0.94, crop_mode='squash'),
+    'dm_nfnet_f4.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f4-0ac5b10b.pth',
+        pool_size=(12, 12), input_size=(3, 384, 384), test_input_size=(3, 512, 512), crop_pct=0.951, crop_mode='squash'),
+    'dm_nfnet_f5.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f5-ecb20ab1.pth',
+        pool_size=(13, 13), input_size=(3, 416, 416), test_input_size=(3, 544, 544), crop_pct=0.954, crop_mode='squash'),
+    'dm_nfnet_f6.dm_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-dnf-weights/dm_nfnet_f6-e0f12116.pth',
+        pool_size=(14, 14), input_size=(3, 448, 448), test_input_size=(3, 576, 576), crop_pct=0.956, crop_mode='squash'),
+
+    'nfnet_f0': _dcfg(
+        url='', pool_size=(6, 6), input_size=(3, 192, 192), test_input_size=(3, 256, 256)),
+    'nfnet_f1': _dcfg(
+        url='', pool_size=(7, 7), input_size=(3, 224, 224), test_input_size=(3, 320, 320)),
+    'nfnet_f2': _dcfg(
+        url='', pool_size=(8, 8), input_size=(3, 256, 256), test_input_size=(3, 352, 352)),
+    'nfnet_f3': _dcfg(
+        url='', pool_size=(10, 10), input_size=(3, 320, 320), test_input_size=(3, 416, 416)),
+    'nfnet_f4': _dcfg(
+        url='', pool_size=(12, 12), input_size=(3, 384, 384), test_input_size=(3, 512, 512)),
+    'nfnet_f5': _dcfg(
+        url='', pool_size=(13, 13), input_size=(3, 416, 416), test_input_size=(3, 544, 544)),
+    'nfnet_f6': _dcfg(
+        url='', pool_size=(14, 14), input_size=(3, 448, 448), test_input_size=(3, 576, 576)),
+    'nfnet_f7': _dcfg(
+        url='', pool_size=(15, 15), input_size=(3, 480, 480), test_input_size=(3, 608, 608)),
+
+    'nfnet_l0.ra2_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/nfnet_l0_ra2-45c6688d.pth',
+        pool_size=(7, 7), input_size=(3, 224, 224), test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'eca_nfnet_l0.ra2_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/ecanfnet_l0_ra2-e3e9ac50.pth',
+        pool_size=(7, 7), input_size=(3, 224, 224), test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'eca_nfnet_l1.ra2_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/ecanfnet_l1_ra2-7dce93cd.pth',
+        pool_size=(8, 8), input_size=(3, 256, 256), test_input_size=(3, 320, 320), test_crop_pct=1.0),
+    'eca_nfnet_l2.ra3_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/ecanfnet_l2_ra3-da781a61.pth',
+        pool_size=(10, 10), input_size=(3, 320, 320), test_input_size=(3, 384, 384), test_crop_pct=1.0),
+    'eca_nfnet_l3': _dcfg(
+        url='',
+        pool_size=(11, 11), input_size=(3, 352, 352), test_input_size=(3, 448, 448), test_crop_pct=1.0),
+
+    'nf_regnet_b0': _dcfg(
+        url='', pool_size=(6, 6), input_size=(3, 192, 192), test_input_size=(3, 256, 256), first_conv='stem.conv'),
+    'nf_regnet_b1.ra2_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/nf_regnet_b1_256_ra2-ad85cfef.pth',
+        pool_size=(8, 8), input_size=(3, 256, 256), test_input_size=(3, 288, 288), first_conv='stem.conv'),  # NOT to paper spec
+    'nf_regnet_b2': _dcfg(
+        url='', pool_size=(8, 8), input_size=(3, 240, 240), test_input_size=(3, 272, 272), first_conv='stem.conv'),
+    'nf_regnet_b3': _dcfg(
+        url='', pool_size=(9, 9), input_size=(3, 288, 288), test_input_size=(3, 320, 320), first_conv='stem.conv'),
+    'nf_regnet_b4': _dcfg(
+        url='', pool_size=(10, 10), input_size=(3, 320, 320), test_input_size=(3, 384, 384), first_conv='stem.conv'),
+    'nf_regnet_b5': _dcfg(
+        url='', pool_size=(12, 12), input_size=(3, 384, 384), test_input_size=(3, 456, 456), first_conv='stem.conv'),
+
+    'nf_resnet26': _dcfg(url='', first_conv='stem.conv'),
+    'nf_resnet50.ra2_in1k': _dcfg(
+        hf_hub_id='timm/',
+        url='https://github.com/rwightman/pytorch-image-models/releases/download/v0.1-weights/nf_resnet50_ra2-9f236009.pth',
+        pool_size=(8, 8), input_size=(3, 256, 256), test_input_size=(3, 288, 288), crop_pct=0.94, first_conv='stem.conv'),
+    'nf_resnet101': _dcfg(url='', first_conv='stem.conv'),
+
+    'nf_seresnet26': _dcfg(url='', first_conv='stem.conv'),
+    'nf_seresnet50': _dcfg(url='', first_conv='stem.conv'),
+    'nf_seresnet101': _dcfg(url='', first_conv='stem.conv'),
+
+    'nf_ecaresnet26': _dcfg(url='', first_conv='stem.conv'),
+    'nf_ecaresnet50': _dcfg(url='', first_conv='stem.conv'),
+    'nf_ecaresnet101': _dcfg(url='', first_conv='stem.conv'),
+
+    'test_nfnet.r160_in1k': _dcfg(
+        hf_hub_id='timm/',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        crop_pct=0.95, input_size=(3, 160, 160), pool_size=(5, 5)),
 })
 
 
-def _register(name):
-    def fn(pretrained: bool = False, **kwargs) -> NormFreeNet:
-        return _create_normfreenet(name, pretrained, **kwargs)
-    fn.__name__ = name
-    return register_model(fn)
 
 
-for _name in list(model_cfgs):
-    globals()[_name] = _register(_name)
+@register_model
+def nfnet_f5(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """NFNet-F5."""
+    return _create_normfreenet('nfnet_f5', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f6(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """NFNet-F6."""
+    return _create_normfreenet('nfnet_f6', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f7(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """NFNet-F7."""
+    return _create_normfreenet('nfnet_f7', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def eca_nfnet_l3(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """ECA-NFNet-L3 w/ SiLU. My experimental 'light' model w/ F3 repeats, 2.0x final_conv mult, 64 group_size, .25 bottleneck & ECA attn"""
+    return _create_normfreenet('eca_nfnet_l3', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b4(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free RegNet-B4."""
+    return _create_normfreenet('nf_regnet_b4', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b5(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free RegNet-B5."""
+    return _create_normfreenet('nf_regnet_b5', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_seresnet26(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free SE-ResNet26."""
+    return _create_normfreenet('nf_seresnet26', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_seresnet50(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free SE-ResNet50."""
+    return _create_normfreenet('nf_seresnet50', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_seresnet101(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free SE-ResNet101."""
+    return _create_normfreenet('nf_seresnet101', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_ecaresnet26(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free ECA-ResNet26."""
+    return _create_normfreenet('nf_ecaresnet26', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_ecaresnet50(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free ECA-ResNet50."""
+    return _create_normfreenet('nf_ecaresnet50', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def nf_ecaresnet101(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Normalization-Free ECA-ResNet101."""
+    return _create_normfreenet('nf_ecaresnet101', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def test_nfnet(pretrained: bool = False, **kwargs: Any) -> NormFreeNet:
+    """Test NFNet model for experimentation."""
+    return _create_normfreenet('test_nfnet', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f0(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f0', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f1(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f1', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f2(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f2', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f3(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f3', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f4(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f4', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f5(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f5', pretrained, **kwargs)
+
+
+@register_model
+def dm_nfnet_f6(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('dm_nfnet_f6', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f0(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_f0', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f1(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_f1', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f2(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_f2', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f3(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_f3', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_f4(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_f4', pretrained, **kwargs)
+
+
+@register_model
+def nfnet_l0(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nfnet_l0', pretrained, **kwargs)
+
+
+@register_model
+def eca_nfnet_l0(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('eca_nfnet_l0', pretrained, **kwargs)
+
+
+@register_model
+def eca_nfnet_l1(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('eca_nfnet_l1', pretrained, **kwargs)
+
+
+@register_model
+def eca_nfnet_l2(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('eca_nfnet_l2', pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b0(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_regnet_b0', pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b1(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_regnet_b1', pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b2(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_regnet_b2', pretrained, **kwargs)
+
+
+@register_model
+def nf_regnet_b3(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_regnet_b3', pretrained, **kwargs)
+
+
+@register_model
+def nf_resnet26(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_resnet26', pretrained, **kwargs)
+
+
+@register_model
+def nf_resnet50(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_resnet50', pretrained, **kwargs)
+
+
+@register_model
+def nf_resnet101(pretrained=False, **kwargs) -> NormFreeNet:
+    return _create_normfreenet('nf_resnet101', pretrained, **kwargs)
+

@@ -17,7 +17,7 @@ import torch
 import torch.nn as nn
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
-from ..layers import ClassifierHead, ConvNormAct, DropPath, SEModule, get_act_layer, get_norm_act_layer
+from ..layers import ClassifierHead, ConvNormAct, DropPath, GroupNormAct, SEModule, get_act_layer, get_norm_act_layer
 from ._builder import build_model_with_cfg
 from ._manipulate import checkpoint_seq, named_apply
 from ._registry import generate_default_cfgs, register_model
@@ -106,9 +106,11 @@ def downsample_avg(in_chs, out_chs, kernel_size=1, stride=1, dilation=1, norm_la
     return nn.Sequential(*[pool, conv])
 
 
-def create_conv2d_basic(in_chs, out_chs, kernel_size, stride=1, dilation=1):
+def create_conv2d_basic(in_chs, out_chs, kernel_size, stride=1, dilation=1, groups=1, bias=False):
     padding = ((stride - 1) + dilation * (kernel_size - 1)) // 2
-    return nn.Conv2d(in_chs, out_chs, kernel_size, stride=stride, padding=padding, dilation=dilation, bias=True)
+    return nn.Conv2d(
+        in_chs, out_chs, kernel_size, stride=stride, padding=padding, dilation=dilation,
+        groups=groups, bias=bias)
 
 
 def create_shortcut(downsample_type, in_chs, out_chs, kernel_size, stride, dilation=(1, 1), norm_layer=None, preact=False):
@@ -187,6 +189,66 @@ class Bottleneck(nn.Module):
         return x
 
 
+
+class PreBottleneck(nn.Module):
+    """Pre-activation RegNet bottleneck — norm+act precede each conv
+    (reference `regnet.py:378`, used by the regnetv_* 'preact regnety' cfgs)."""
+
+    def __init__(
+            self,
+            in_chs: int,
+            out_chs: int,
+            stride: int = 1,
+            dilation: Tuple[int, int] = (1, 1),
+            bottle_ratio: float = 1,
+            group_size: int = 1,
+            se_ratio: float = 0.25,
+            downsample: str = 'conv1x1',
+            linear_out: bool = False,
+            act_layer: Union[str, Callable] = 'relu',
+            norm_layer: Union[str, Callable] = 'batchnorm',
+            drop_block=None,
+            drop_path_rate: float = 0.,
+    ):
+        super().__init__()
+        norm_act_layer = get_norm_act_layer(norm_layer, act_layer)
+        bottleneck_chs = int(round(out_chs * bottle_ratio))
+        groups = bottleneck_chs // group_size
+
+        self.norm1 = norm_act_layer(in_chs)
+        self.conv1 = create_conv2d_basic(in_chs, bottleneck_chs, 1)
+        self.norm2 = norm_act_layer(bottleneck_chs)
+        self.conv2 = create_conv2d_basic(
+            bottleneck_chs, bottleneck_chs, 3, stride=stride, dilation=dilation[0], groups=groups)
+        if se_ratio:
+            se_channels = int(round(in_chs * se_ratio))
+            self.se = SEModule(bottleneck_chs, rd_channels=se_channels, act_layer=act_layer)
+        else:
+            self.se = nn.Identity()
+        self.norm3 = norm_act_layer(bottleneck_chs)
+        self.conv3 = create_conv2d_basic(bottleneck_chs, out_chs, 1)
+        self.downsample = create_shortcut(
+            downsample, in_chs, out_chs, kernel_size=1, stride=stride, dilation=dilation, preact=True)
+        self.drop_path = DropPath(drop_path_rate) if drop_path_rate > 0 else nn.Identity()
+
+    def zero_init_last(self):
+        pass
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.norm1(x)
+        shortcut = x
+        x = self.conv1(x)
+        x = self.norm2(x)
+        x = self.conv2(x)
+        x = self.se(x)
+        x = self.norm3(x)
+        x = self.conv3(x)
+        if self.downsample is not None:
+            # NOTE stuck with downsample as the attr name due to weight compatibility
+            x = self.drop_path(x) + self.downsample(shortcut)
+        return x
+
+
 class RegStage(nn.Module):
     """Stage (sequence of blocks w/ the same output shape, reference `regnet.py:484`)."""
 
@@ -253,8 +315,10 @@ class RegNet(nn.Module):
         # Construct the stem
         stem_width = cfg.stem_width
         na_args = dict(act_layer=cfg.act_layer, norm_layer=cfg.norm_layer)
-        assert not cfg.preact, 'preact RegNet (Z) not yet supported'
-        self.stem = ConvNormAct(in_chans, stem_width, 3, stride=2, **na_args)
+        if cfg.preact:
+            self.stem = create_conv2d_basic(in_chans, stem_width, 3, stride=2)
+        else:
+            self.stem = ConvNormAct(in_chans, stem_width, 3, stride=2, **na_args)
         self.feature_info = [dict(num_chs=stem_width, reduction=2, module='stem')]
 
         # Construct the stages
@@ -266,7 +330,7 @@ class RegNet(nn.Module):
             drop_path_rate=drop_path_rate,
         )
         assert len(per_stage_args) == 4
-        block_fn = Bottleneck
+        block_fn = PreBottleneck if cfg.preact else Bottleneck
         for i, stage_args in enumerate(per_stage_args):
             stage_name = "s{}".format(i + 1)
             self.add_module(
@@ -282,7 +346,7 @@ class RegNet(nn.Module):
             self.final_conv = ConvNormAct(prev_width, cfg.num_features, kernel_size=1, **na_args)
             self.num_features = cfg.num_features
         else:
-            final_act = cfg.linear_out
+            final_act = cfg.linear_out or cfg.preact
             self.final_conv = get_act_layer(cfg.act_layer)() if final_act else nn.Identity()
             self.num_features = prev_width
         self.head_hidden_size = self.num_features
@@ -394,6 +458,7 @@ model_cfgs = dict(
     # RegNet-X
     regnetx_002=RegNetCfg(w0=24, wa=36.44, wm=2.49, group_size=8, depth=13),
     regnetx_004=RegNetCfg(w0=24, wa=24.48, wm=2.54, group_size=16, depth=22),
+    regnetx_004_tv=RegNetCfg(w0=24, wa=24.48, wm=2.54, group_size=16, depth=22, group_min_ratio=0.9),
     regnetx_006=RegNetCfg(w0=48, wa=36.97, wm=2.24, group_size=24, depth=16),
     regnetx_008=RegNetCfg(w0=56, wa=35.73, wm=2.28, group_size=16, depth=16),
     regnetx_016=RegNetCfg(w0=80, wa=34.01, wm=2.25, group_size=24, depth=18),
@@ -409,14 +474,44 @@ model_cfgs = dict(
     regnety_004=RegNetCfg(w0=48, wa=27.89, wm=2.09, group_size=8, depth=16, se_ratio=0.25),
     regnety_006=RegNetCfg(w0=48, wa=32.54, wm=2.32, group_size=16, depth=15, se_ratio=0.25),
     regnety_008=RegNetCfg(w0=56, wa=38.84, wm=2.4, group_size=16, depth=14, se_ratio=0.25),
+    regnety_008_tv=RegNetCfg(w0=56, wa=38.84, wm=2.4, group_size=16, depth=14, se_ratio=0.25, group_min_ratio=0.9),
     regnety_016=RegNetCfg(w0=48, wa=20.71, wm=2.65, group_size=24, depth=27, se_ratio=0.25),
     regnety_032=RegNetCfg(w0=80, wa=42.63, wm=2.66, group_size=24, depth=21, se_ratio=0.25),
     regnety_040=RegNetCfg(w0=96, wa=31.41, wm=2.24, group_size=64, depth=22, se_ratio=0.25),
     regnety_064=RegNetCfg(w0=112, wa=33.22, wm=2.27, group_size=72, depth=25, se_ratio=0.25),
     regnety_080=RegNetCfg(w0=192, wa=76.82, wm=2.19, group_size=56, depth=17, se_ratio=0.25),
+    regnety_080_tv=RegNetCfg(w0=192, wa=76.82, wm=2.19, group_size=56, depth=17, se_ratio=0.25, group_min_ratio=0.9),
     regnety_120=RegNetCfg(w0=168, wa=73.36, wm=2.37, group_size=112, depth=19, se_ratio=0.25),
     regnety_160=RegNetCfg(w0=200, wa=106.23, wm=2.48, group_size=112, depth=18, se_ratio=0.25),
     regnety_320=RegNetCfg(w0=232, wa=115.89, wm=2.53, group_size=232, depth=20, se_ratio=0.25),
+    regnety_640=RegNetCfg(w0=352, wa=147.48, wm=2.4, group_size=328, depth=20, se_ratio=0.25),
+    regnety_1280=RegNetCfg(w0=456, wa=160.83, wm=2.52, group_size=264, depth=27, se_ratio=0.25),
+    regnety_2560=RegNetCfg(w0=640, wa=230.83, wm=2.53, group_size=373, depth=27, se_ratio=0.25),
+
+    regnety_040_sgn=RegNetCfg(
+        w0=96, wa=31.41, wm=2.24, group_size=64, depth=22, se_ratio=0.25,
+        act_layer='silu', norm_layer=partial(GroupNormAct, group_size=16)),
+
+    # regnetv = pre-activation regnety
+    regnetv_040=RegNetCfg(
+        depth=22, w0=96, wa=31.41, wm=2.24, group_size=64, se_ratio=0.25, preact=True, act_layer='silu'),
+    regnetv_064=RegNetCfg(
+        depth=25, w0=112, wa=33.22, wm=2.27, group_size=72, se_ratio=0.25, preact=True, act_layer='silu',
+        downsample='avg'),
+
+    # RegNet-Z
+    regnetz_005=RegNetCfg(
+        depth=21, w0=16, wa=10.7, wm=2.51, group_size=4, bottle_ratio=4.0, se_ratio=0.25,
+        downsample=None, linear_out=True, num_features=1024, act_layer='silu',
+    ),
+    regnetz_040=RegNetCfg(
+        depth=28, w0=48, wa=14.5, wm=2.226, group_size=8, bottle_ratio=4.0, se_ratio=0.25,
+        downsample=None, linear_out=True, num_features=0, act_layer='silu',
+    ),
+    regnetz_040_h=RegNetCfg(
+        depth=28, w0=48, wa=14.5, wm=2.226, group_size=8, bottle_ratio=4.0, se_ratio=0.25,
+        downsample=None, linear_out=True, num_features=1536, act_layer='silu',
+    ),
 )
 
 
@@ -439,22 +534,400 @@ def _cfg(url: str = '', **kwargs) -> Dict[str, Any]:
     }
 
 
+def _cfgpyc(url: str = '', **kwargs) -> Dict[str, Any]:
+    """pycls-recipe cfg preset."""
+    return {
+        'url': url, 'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': (7, 7),
+        'crop_pct': 0.875, 'interpolation': 'bicubic',
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'stem.conv', 'classifier': 'head.fc',
+        'license': 'mit', 'origin_url': 'https://github.com/facebookresearch/pycls', **kwargs
+    }
+
+
+def _cfgtv2(url: str = '', **kwargs) -> Dict[str, Any]:
+    """torchvision-v2-recipe cfg preset."""
+    return {
+        'url': url, 'num_classes': 1000, 'input_size': (3, 224, 224), 'pool_size': (7, 7),
+        'crop_pct': 0.965, 'interpolation': 'bicubic',
+        'mean': IMAGENET_DEFAULT_MEAN, 'std': IMAGENET_DEFAULT_STD,
+        'first_conv': 'stem.conv', 'classifier': 'head.fc',
+        'license': 'bsd-3-clause', 'origin_url': 'https://github.com/pytorch/vision', **kwargs
+    }
+
+
 default_cfgs = generate_default_cfgs({
-    **{f'{name}.pycls_in1k': _cfg() for name in model_cfgs},
-    'regnety_032.ra_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'regnety_040.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'regnety_064.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'regnety_080.ra3_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=1.0),
-    'regnety_160.deit_in1k': _cfg(),
+    # timm trained models
+    'regnety_032.ra_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-weights/regnety_032_ra-7f2439f9.pth'),
+    'regnety_040.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnety_040_ra3-670e1166.pth'),
+    'regnety_064.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnety_064_ra3-aa26dc7d.pth'),
+    'regnety_080.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnety_080_ra3-1fdc4344.pth'),
+    'regnety_120.sw_in12k_ft_in1k': _cfg(hf_hub_id='timm/'),
+    'regnety_160.sw_in12k_ft_in1k': _cfg(hf_hub_id='timm/'),
+    'regnety_160.lion_in12k_ft_in1k': _cfg(hf_hub_id='timm/'),
+
+    # timm in12k pretrain
+    'regnety_120.sw_in12k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=11821),
+    'regnety_160.sw_in12k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=11821),
+
+    # timm custom arch (v and z guess) + trained models
+    'regnety_040_sgn.untrained': _cfg(url=''),
+    'regnetv_040.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetv_040_ra3-c248f51f.pth',
+        first_conv='stem'),
+    'regnetv_064.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetv_064_ra3-530616c2.pth',
+        first_conv='stem'),
+
+    'regnetz_005.untrained': _cfg(url=''),
+    'regnetz_040.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetz_040_ra3-9007edf5.pth',
+        input_size=(3, 256, 256), pool_size=(8, 8), crop_pct=1.0, test_input_size=(3, 320, 320)),
+    'regnetz_040_h.ra3_in1k': _cfg(
+        hf_hub_id='timm/',
+        url='https://github.com/huggingface/pytorch-image-models/releases/download/v0.1-tpu-weights/regnetz_040h_ra3-f594343b.pth',
+        input_size=(3, 256, 256), pool_size=(8, 8), crop_pct=1.0, test_input_size=(3, 320, 320)),
+
+    # used in DeiT for distillation (from Facebook DeiT GitHub repository)
+    'regnety_160.deit_in1k': _cfg(
+        hf_hub_id='timm/', url='https://dl.fbaipublicfiles.com/deit/regnety_160-a5fe301d.pth'),
+
+    'regnetx_004_tv.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_400mf-62229a5f.pth'),
+    'regnetx_008.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_800mf-94a99ebd.pth'),
+    'regnetx_016.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_1_6gf-a12f2b72.pth'),
+    'regnetx_032.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_3_2gf-7071aa85.pth'),
+    'regnetx_080.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_8gf-2b70d774.pth'),
+    'regnetx_160.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_16gf-ba3796d7.pth'),
+    'regnetx_320.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_x_32gf-6eb8fdc6.pth'),
+
+    'regnety_004.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_400mf-e6988f5f.pth'),
+    'regnety_008_tv.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_800mf-58fc7688.pth'),
+    'regnety_016.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_1_6gf-0d7bc02a.pth'),
+    'regnety_032.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_3_2gf-9180c971.pth'),
+    'regnety_080_tv.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_8gf-dc2b1b54.pth'),
+    'regnety_160.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_16gf-3e4a00f9.pth'),
+    'regnety_320.tv2_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_32gf-8db6d4b5.pth'),
+
+    'regnety_160.swag_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_16gf_swag-43afe44d.pth', license='cc-by-nc-4.0',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+    'regnety_320.swag_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_32gf_swag-04fdfa75.pth', license='cc-by-nc-4.0',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+    'regnety_1280.swag_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_128gf_swag-c8ce3e52.pth', license='cc-by-nc-4.0',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+
+    'regnety_160.swag_lc_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_16gf_lc_swag-f3ec0043.pth', license='cc-by-nc-4.0'),
+    'regnety_320.swag_lc_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_32gf_lc_swag-e1583746.pth', license='cc-by-nc-4.0'),
+    'regnety_1280.swag_lc_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://download.pytorch.org/models/regnet_y_128gf_lc_swag-cbe8ce12.pth', license='cc-by-nc-4.0'),
+
+    'regnety_320.seer_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        license='seer-license', origin_url='https://github.com/facebookresearch/vissl',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_finetuned/seer_regnet32_finetuned_in1k_model_final_checkpoint_phase78.torch',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+    'regnety_640.seer_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        license='seer-license', origin_url='https://github.com/facebookresearch/vissl',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_finetuned/seer_regnet64_finetuned_in1k_model_final_checkpoint_phase78.torch',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+    'regnety_1280.seer_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        license='seer-license', origin_url='https://github.com/facebookresearch/vissl',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_finetuned/seer_regnet128_finetuned_in1k_model_final_checkpoint_phase78.torch',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+    'regnety_2560.seer_ft_in1k': _cfgtv2(
+        hf_hub_id='timm/',
+        license='seer-license', origin_url='https://github.com/facebookresearch/vissl',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_finetuned/seer_regnet256_finetuned_in1k_model_final_checkpoint_phase38.torch',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0),
+
+    'regnety_320.seer': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_regnet32d/seer_regnet32gf_model_iteration244000.torch',
+        num_classes=0, license='seer-license', origin_url='https://github.com/facebookresearch/vissl'),
+    'regnety_640.seer': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/seer_regnet64/seer_regnet64gf_model_final_checkpoint_phase0.torch',
+        num_classes=0, license='seer-license', origin_url='https://github.com/facebookresearch/vissl'),
+    'regnety_1280.seer': _cfgtv2(
+        hf_hub_id='timm/',
+        url='https://dl.fbaipublicfiles.com/vissl/model_zoo/swav_ig1b_regnet128Gf_cnstant_bs32_node16_sinkhorn10_proto16k_syncBN64_warmup8k/model_final_checkpoint_phase0.torch',
+        num_classes=0, license='seer-license', origin_url='https://github.com/facebookresearch/vissl'),
+    # FIXME invalid weight <-> model match, mistake on their end
+    #'regnety_2560.seer': _cfgtv2(
+    #    url='https://dl.fbaipublicfiles.com/vissl/model_zoo/swav_ig1b_cosine_rg256gf_noBNhead_wd1e5_fairstore_bs16_node64_sinkhorn10_proto16k_apex_syncBN64_warmup8k/model_final_checkpoint_phase0.torch',
+    #    num_classes=0, license='other', origin_url='https://github.com/facebookresearch/vissl'),
+
+    'regnetx_002.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_004.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_006.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_008.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_016.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_032.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_040.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_064.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_080.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_120.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_160.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnetx_320.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+
+    'regnety_002.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_004.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_006.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_008.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_016.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_032.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_040.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_064.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_080.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_120.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_160.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
+    'regnety_320.pycls_in1k': _cfgpyc(hf_hub_id='timm/'),
 })
 
 
-def _register(name):
-    def fn(pretrained: bool = False, **kwargs) -> RegNet:
-        return _create_regnet(name, pretrained, **kwargs)
-    fn.__name__ = name
-    return register_model(fn)
+@register_model
+def regnetx_002(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_002', pretrained, **kwargs)
 
 
-for _name in list(model_cfgs):
-    globals()[_name] = _register(_name)
+@register_model
+def regnetx_004(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_004', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_006(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_006', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_008(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_008', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_016(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_016', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_032(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_032', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_040(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_040', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_064(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_064', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_080(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_080', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_120(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_120', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_160(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_160', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_320(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnetx_320', pretrained, **kwargs)
+
+
+@register_model
+def regnety_002(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_002', pretrained, **kwargs)
+
+
+@register_model
+def regnety_004(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_004', pretrained, **kwargs)
+
+
+@register_model
+def regnety_006(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_006', pretrained, **kwargs)
+
+
+@register_model
+def regnety_008(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_008', pretrained, **kwargs)
+
+
+@register_model
+def regnety_016(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_016', pretrained, **kwargs)
+
+
+@register_model
+def regnety_032(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_032', pretrained, **kwargs)
+
+
+@register_model
+def regnety_040(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_040', pretrained, **kwargs)
+
+
+@register_model
+def regnety_064(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_064', pretrained, **kwargs)
+
+
+@register_model
+def regnety_080(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_080', pretrained, **kwargs)
+
+
+@register_model
+def regnety_120(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_120', pretrained, **kwargs)
+
+
+@register_model
+def regnety_160(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_160', pretrained, **kwargs)
+
+
+@register_model
+def regnety_320(pretrained=False, **kwargs) -> RegNet:
+    return _create_regnet('regnety_320', pretrained, **kwargs)
+
+
+@register_model
+def regnetx_004_tv(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetX-400MF w/ torchvision group rounding"""
+    return _create_regnet('regnetx_004_tv', pretrained, **kwargs)
+
+
+@register_model
+def regnety_008_tv(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-800MF w/ torchvision group rounding"""
+    return _create_regnet('regnety_008_tv', pretrained, **kwargs)
+
+
+@register_model
+def regnety_080_tv(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-8.0GF w/ torchvision group rounding"""
+    return _create_regnet('regnety_080_tv', pretrained, **kwargs)
+
+
+@register_model
+def regnety_640(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-64GF"""
+    return _create_regnet('regnety_640', pretrained, **kwargs)
+
+
+@register_model
+def regnety_1280(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-128GF"""
+    return _create_regnet('regnety_1280', pretrained, **kwargs)
+
+
+@register_model
+def regnety_2560(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-256GF"""
+    return _create_regnet('regnety_2560', pretrained, **kwargs)
+
+
+@register_model
+def regnety_040_sgn(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetY-4.0GF w/ GroupNorm"""
+    return _create_regnet('regnety_040_sgn', pretrained, **kwargs)
+
+
+@register_model
+def regnetv_040(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetV-4.0GF (pre-activation)"""
+    return _create_regnet('regnetv_040', pretrained, **kwargs)
+
+
+@register_model
+def regnetv_064(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetV-6.4GF (pre-activation)"""
+    return _create_regnet('regnetv_064', pretrained, **kwargs)
+
+
+@register_model
+def regnetz_005(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetZ-500MF NOTE: config found in https://github.com/facebookresearch/ClassyVision/blob/main/classy_vision/models/regnet.py but it's not clear it is equivalent to paper model as not detailed in the paper."""
+    return _create_regnet('regnetz_005', pretrained, zero_init_last=False, **kwargs)
+
+
+@register_model
+def regnetz_040(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetZ-4.0GF NOTE: config found in https://github.com/facebookresearch/ClassyVision/blob/main/classy_vision/models/regnet.py but it's not clear it is equivalent to paper model as not detailed in the paper."""
+    return _create_regnet('regnetz_040', pretrained, zero_init_last=False, **kwargs)
+
+
+@register_model
+def regnetz_040_h(pretrained: bool = False, **kwargs) -> RegNet:
+    """RegNetZ-4.0GF NOTE: config found in https://github.com/facebookresearch/ClassyVision/blob/main/classy_vision/models/regnet.py but it's not clear it is equivalent to paper model as not detailed in the paper."""
+    return _create_regnet('regnetz_040_h', pretrained, zero_init_last=False, **kwargs)
+

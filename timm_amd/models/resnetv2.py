@@ -574,24 +574,102 @@ def _cfg(url: str = '', **kwargs: Any) -> Dict[str, Any]:
 
 
 default_cfgs = generate_default_cfgs({
-    'resnetv2_50x1_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
-    'resnetv2_50x1_bit.goog_in21k': _cfg(num_classes=21843),
-    'resnetv2_50x3_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
-    'resnetv2_101x1_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
-    'resnetv2_101x3_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
-    'resnetv2_152x2_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0),
-    'resnetv2_152x4_bit.goog_in21k_ft_in1k': _cfg(input_size=(3, 480, 480), pool_size=(15, 15), crop_pct=1.0),
-    'resnetv2_50.a1h_in1k': _cfg(interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288)),
-    'resnetv2_50d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
-    'resnetv2_50t.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
-    'resnetv2_101.a1h_in1k': _cfg(interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288)),
-    'resnetv2_101d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
-    'resnetv2_152.untrained': _cfg(interpolation='bicubic'),
-    'resnetv2_152d.untrained': _cfg(interpolation='bicubic', first_conv='stem.conv1'),
-    'resnetv2_50d_gn.ah_in1k': _cfg(interpolation='bicubic', first_conv='stem.conv1',
-                                    crop_pct=0.95, test_input_size=(3, 288, 288)),
-    'resnetv2_50d_evos.ah_in1k': _cfg(interpolation='bicubic', first_conv='stem.conv1',
-                                      crop_pct=0.95, test_input_size=(3, 288, 288)),
+    #  Paper: Knowledge distillation: A good teacher is patient and consistent - https://arxiv.org/abs/2106.05237
+    'resnetv2_50x1_bit.goog_distilled_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', custom_load=True),
+    'resnetv2_152x2_bit.goog_teacher_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', custom_load=True),
+    'resnetv2_152x2_bit.goog_teacher_in21k_ft_in1k_384': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 384, 384), pool_size=(12, 12), crop_pct=1.0, interpolation='bicubic', custom_load=True),
+
+    # pretrained on imagenet21k, finetuned on imagenet1k
+    'resnetv2_50x1_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0, custom_load=True),
+    'resnetv2_50x3_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0, custom_load=True),
+    'resnetv2_101x1_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0, custom_load=True),
+    'resnetv2_101x3_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0, custom_load=True),
+    'resnetv2_152x2_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 448, 448), pool_size=(14, 14), crop_pct=1.0, custom_load=True),
+    'resnetv2_152x4_bit.goog_in21k_ft_in1k': _cfg(
+        hf_hub_id='timm/',
+        input_size=(3, 480, 480), pool_size=(15, 15), crop_pct=1.0, custom_load=True),  # only one at 480x480?
+
+    # trained on imagenet-21k
+    'resnetv2_50x1_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+    'resnetv2_50x3_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+    'resnetv2_101x1_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+    'resnetv2_101x3_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+    'resnetv2_152x2_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+    'resnetv2_152x4_bit.goog_in21k': _cfg(
+        hf_hub_id='timm/',
+        num_classes=21843, custom_load=True),
+
+    'resnetv2_18.ra4_e3600_r224_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.9, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_18d.ra4_e3600_r224_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.9, test_input_size=(3, 288, 288), test_crop_pct=1.0,
+        first_conv='stem.conv1'),
+    'resnetv2_34.ra4_e3600_r224_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.9, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_34d.ra4_e3600_r224_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.9, test_input_size=(3, 288, 288), test_crop_pct=1.0,
+        first_conv='stem.conv1'),
+    'resnetv2_34d.ra4_e3600_r384_in1k': _cfg(
+        hf_hub_id='timm/',
+        crop_pct=1.0, input_size=(3, 384, 384), pool_size=(12, 12), test_input_size=(3, 448, 448),
+        interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_50.a1h_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_50d.untrained': _cfg(
+        interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_50t.untrained': _cfg(
+        interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_101.a1h_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', crop_pct=0.95, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_101d.untrained': _cfg(
+        interpolation='bicubic', first_conv='stem.conv1'),
+    'resnetv2_152.untrained': _cfg(
+        interpolation='bicubic'),
+    'resnetv2_152d.untrained': _cfg(
+        interpolation='bicubic', first_conv='stem.conv1'),
+
+    'resnetv2_50d_gn.ah_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', first_conv='stem.conv1',
+        crop_pct=0.95, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_50d_evos.ah_in1k': _cfg(
+        hf_hub_id='timm/',
+        interpolation='bicubic', first_conv='stem.conv1',
+        crop_pct=0.95, test_input_size=(3, 288, 288), test_crop_pct=1.0),
+    'resnetv2_50d_frn.untrained': _cfg(
+        interpolation='bicubic', first_conv='stem.conv1'),
 })
 
 
@@ -696,3 +774,53 @@ def resnetv2_50d_evos(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
         layers=[3, 4, 6, 3], conv_layer=StdConv2d, norm_layer=EvoNorm2dS0,
         stem_type='deep', avg_down=True)
     return _create_resnetv2('resnetv2_50d_evos', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_18(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    """ResNetV2-18 model."""
+    model_args = dict(
+        layers=[2, 2, 2, 2], channels=(64, 128, 256, 512), basic=True, bottle_ratio=1.0,
+        conv_layer=create_conv2d, norm_layer=BatchNormAct2d
+    )
+    return _create_resnetv2('resnetv2_18', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_18d(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    """ResNetV2-18d model (deep stem variant)."""
+    model_args = dict(
+        layers=[2, 2, 2, 2], channels=(64, 128, 256, 512), basic=True, bottle_ratio=1.0,
+        conv_layer=create_conv2d, norm_layer=BatchNormAct2d, stem_type='deep', avg_down=True
+    )
+    return _create_resnetv2('resnetv2_18d', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_34(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    """ResNetV2-34 model."""
+    model_args = dict(
+        layers=(3, 4, 6, 3), channels=(64, 128, 256, 512), basic=True, bottle_ratio=1.0,
+        conv_layer=create_conv2d, norm_layer=BatchNormAct2d
+    )
+    return _create_resnetv2('resnetv2_34', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_34d(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    """ResNetV2-34d model (deep stem variant)."""
+    model_args = dict(
+        layers=(3, 4, 6, 3), channels=(64, 128, 256, 512), basic=True, bottle_ratio=1.0,
+        conv_layer=create_conv2d, norm_layer=BatchNormAct2d, stem_type='deep', avg_down=True
+    )
+    return _create_resnetv2('resnetv2_34d', pretrained=pretrained, **dict(model_args, **kwargs))
+
+
+@register_model
+def resnetv2_50d_frn(pretrained: bool = False, **kwargs: Any) -> ResNetV2:
+    """ResNetV2-50d model with Filter Response Normalization."""
+    model_args = dict(
+        layers=[3, 4, 6, 3], conv_layer=create_conv2d, norm_layer=FilterResponseNormTlu2d,
+        stem_type='deep', avg_down=True)
+    return _create_resnetv2('resnetv2_50d_frn', pretrained=pretrained, **dict(model_args, **kwargs))
+

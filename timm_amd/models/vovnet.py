@@ -373,24 +373,71 @@ def _cfg(url='', **kwargs):
 
 
 default_cfgs = generate_default_cfgs({
-    'vovnet39a.untrained': _cfg(),
-    'vovnet57a.untrained': _cfg(),
-    'ese_vovnet19b_slim_dw.untrained': _cfg(),
-    'ese_vovnet19b_dw.ra_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=0.95),
-    'ese_vovnet19b_slim.untrained': _cfg(),
-    'ese_vovnet39b.ra_in1k': _cfg(test_input_size=(3, 288, 288), crop_pct=0.95),
-    'ese_vovnet57b.untrained': _cfg(),
-    'ese_vovnet99b.untrained': _cfg(),
-    'eca_vovnet39b.untrained': _cfg(),
+    'vovnet39a.untrained': _cfg(url=''),
+    'vovnet57a.untrained': _cfg(url=''),
+    'ese_vovnet19b_slim_dw.untrained': _cfg(url=''),
+    'ese_vovnet19b_dw.ra_in1k': _cfg(
+        hf_hub_id='timm/',
+        test_input_size=(3, 288, 288), test_crop_pct=0.95),
+    'ese_vovnet19b_slim.untrained': _cfg(url=''),
+    'ese_vovnet39b.ra_in1k': _cfg(
+        hf_hub_id='timm/',
+        test_input_size=(3, 288, 288), test_crop_pct=0.95),
+    'ese_vovnet57b.ra4_e3600_r256_in1k': _cfg(
+        hf_hub_id='timm/',
+        mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5),
+        crop_pct=0.95, input_size=(3, 256, 256), pool_size=(8, 8),
+        test_input_size=(3, 320, 320), test_crop_pct=1.0
+    ),
+    'ese_vovnet99b.untrained': _cfg(url=''),
+    'eca_vovnet39b.untrained': _cfg(url=''),
+    'ese_vovnet39b_evos.untrained': _cfg(url=''),
 })
 
 
-def _register(name):
-    def fn(pretrained: bool = False, **kwargs) -> VovNet:
-        return _create_vovnet(name, pretrained=pretrained, **kwargs)
-    fn.__name__ = name
-    return register_model(fn)
+@register_model
+def vovnet39a(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('vovnet39a', pretrained=pretrained, **kwargs)
 
 
-for _name in list(model_cfgs):
-    globals()[_name] = _register(_name)
+@register_model
+def vovnet57a(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('vovnet57a', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet19b_slim_dw(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet19b_slim_dw', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet19b_dw(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet19b_dw', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet19b_slim(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet19b_slim', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet39b(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet39b', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet57b(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet57b', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet99b(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('ese_vovnet99b', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def eca_vovnet39b(pretrained=False, **kwargs) -> VovNet:
+    return _create_vovnet('eca_vovnet39b', pretrained=pretrained, **kwargs)
+
+
+
