@@ -43,7 +43,17 @@ INTEROP_MODELS = [
     ('convnextv2_atto', 128),
     ('swin_tiny_patch4_window7_224', 224),
     ('swinv2_tiny_window8_256', 256),
-    # maxvit reference-style entrypoint names land with the zoo-depth pass
+    ('coatnet_nano_rw_224', 224),
+    ('coatnet_0_rw_224', 224),
+    ('maxvit_rmlp_nano_rw_256', 256),
+    ('maxvit_tiny_pm_256', 256),
+    ('coatnext_nano_rw_224', 224),
+    ('maxxvitv2_nano_rw_256', 256),
+    ('mobileone_s0', 128),
+    ('regnetz_005', 128),
+    ('regnetz_b16', 128),
+    ('resnet50_clip_gap', 224),
+    ('nf_regnet_b0', 128),
     ('efficientnet_b0', 128),
     ('mobilenetv3_small_100', 128),
     ('mobilenetv4_conv_small', 128),

@@ -326,21 +326,43 @@ class NaFlexVit(nn.Module):
             self.rope = None
 
         dpr = calculate_drop_path_rates(cfg.drop_path_rate, cfg.depth)
-        self.blocks = nn.ModuleList([
-            Block(
-                dim=cfg.embed_dim,
-                num_heads=cfg.num_heads,
-                mlp_ratio=cfg.mlp_ratio,
-                qkv_bias=cfg.qkv_bias,
-                qk_norm=cfg.qk_norm,
-                init_values=cfg.init_values,
-                proj_drop=cfg.proj_drop_rate,
-                attn_drop=cfg.attn_drop_rate,
-                drop_path=dpr[i],
-                norm_layer=norm_layer,
-                act_layer=act_layer,
-            )
-            for i in range(cfg.depth)])
+        if self.rope is not None:
+            # rope needs a block whose attention applies the rotary embed past
+            # the prefix tokens (reference `naflexvit.py:291-335` uses the EVA
+            # block family for all rope attn types)
+            from .eva import EvaBlock
+            self.blocks = nn.ModuleList([
+                EvaBlock(
+                    dim=cfg.embed_dim,
+                    num_heads=cfg.num_heads,
+                    mlp_ratio=cfg.mlp_ratio,
+                    qkv_bias=cfg.qkv_bias,
+                    attn_type='rope',
+                    num_prefix_tokens=self.num_prefix_tokens,
+                    init_values=cfg.init_values,
+                    proj_drop=cfg.proj_drop_rate,
+                    attn_drop=cfg.attn_drop_rate,
+                    drop_path=dpr[i],
+                    norm_layer=norm_layer,
+                    act_layer=act_layer,
+                )
+                for i in range(cfg.depth)])
+        else:
+            self.blocks = nn.ModuleList([
+                Block(
+                    dim=cfg.embed_dim,
+                    num_heads=cfg.num_heads,
+                    mlp_ratio=cfg.mlp_ratio,
+                    qkv_bias=cfg.qkv_bias,
+                    qk_norm=cfg.qk_norm,
+                    init_values=cfg.init_values,
+                    proj_drop=cfg.proj_drop_rate,
+                    attn_drop=cfg.attn_drop_rate,
+                    drop_path=dpr[i],
+                    norm_layer=norm_layer,
+                    act_layer=act_layer,
+                )
+                for i in range(cfg.depth)])
         self.feature_info = [
             dict(module=f'blocks.{i}', num_chs=cfg.embed_dim, reduction=cfg.patch_size)
             for i in range(cfg.depth)]
@@ -434,10 +456,13 @@ class NaFlexVit(nn.Module):
                 rope_embed = self.rope.get_embed((g, g))
 
         for blk in self.blocks:
+            blk_kwargs = dict(attn_mask=attn_mask)
+            if rope_embed is not None:
+                blk_kwargs['rope'] = rope_embed
             if self.grad_checkpointing and not torch.jit.is_scripting():
-                h = checkpoint(blk, h, attn_mask=attn_mask)
+                h = checkpoint(blk, h, **blk_kwargs)
             else:
-                h = blk(h, attn_mask=attn_mask)
+                h = blk(h, **blk_kwargs)
         h = self.norm(h)
         return h, patch_valid
 
