@@ -22,6 +22,8 @@ sources = [
     os.path.join(CSRC, 'depthwise_conv.hip'),
     os.path.join(CSRC, 'multi_tensor.hip'),
     os.path.join(CSRC, 'muon_ns.hip'),
+    os.path.join(CSRC, 'ce_loss.hip'),
+    os.path.join(CSRC, 'data_ops.hip'),
 ]
 
 setup(

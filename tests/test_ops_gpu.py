@@ -538,3 +538,117 @@ def test_muon_optimizer_gpu_step():
         Muon(m.parameters(), lr=0.05, weight_decay=0.01).step()
     assert rel_err(lin_gpu.weight.cpu(), lin_cpu.weight) < 1e-2
     assert rel_err(lin_gpu.bias.cpu(), lin_cpu.bias) < 1e-2
+
+
+# ---------------- fused cross-entropy ----------------
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+@pytest.mark.parametrize('smoothing', [0.0, 0.1])
+def test_fused_ce_hard_labels(dtype, smoothing):
+    torch.manual_seed(0)
+    B, C = 64, 1000
+    logits = torch.randn(B, C, device='cuda', dtype=dtype, requires_grad=True)
+    target = torch.randint(0, C, (B,), device='cuda')
+
+    loss = ops.fused_cross_entropy(logits, target, smoothing=smoothing)
+    loss.backward()
+    got_grad = logits.grad.clone()
+
+    ref_logits = logits.detach().float().cpu().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(ref_logits, target.cpu(), label_smoothing=smoothing)
+    ref.backward()
+
+    tol = 1e-5 if dtype == torch.float32 else 2e-3
+    assert abs(loss.item() - ref.item()) < tol * 10
+    err = (got_grad.float().cpu() - ref_logits.grad).abs().max().item()
+    assert err < tol, f'grad err {err}'
+
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_fused_ce_soft_target(dtype):
+    torch.manual_seed(0)
+    B, C = 32, 1000
+    logits = torch.randn(B, C, device='cuda', dtype=dtype, requires_grad=True)
+    target = torch.softmax(torch.randn(B, C, device='cuda'), -1).to(dtype)
+
+    loss = ops.fused_cross_entropy(logits, target)
+    loss.backward()
+    got_grad = logits.grad.clone()
+
+    ref_logits = logits.detach().float().cpu().requires_grad_(True)
+    ref = torch.sum(-target.float().cpu() * torch.log_softmax(ref_logits, -1), -1).mean()
+    ref.backward()
+
+    tol = 1e-5 if dtype == torch.float32 else 2e-3
+    assert abs(loss.item() - ref.item()) < tol * 10
+    err = (got_grad.float().cpu() - ref_logits.grad).abs().max().item()
+    assert err < tol, f'grad err {err}'
+
+
+def test_fused_ce_module_paths():
+    from timm_amd.loss import LabelSmoothingCrossEntropy, SoftTargetCrossEntropy
+    torch.manual_seed(0)
+    x = torch.randn(16, 200, device='cuda', requires_grad=True)
+    t = torch.randint(0, 200, (16,), device='cuda')
+    l1 = LabelSmoothingCrossEntropy(0.1)(x, t)
+    l1_ref = LabelSmoothingCrossEntropy(0.1)(x.detach().cpu().requires_grad_(True), t.cpu())
+    assert abs(l1.item() - l1_ref.item()) < 1e-5
+    soft = torch.softmax(torch.randn(16, 200, device='cuda'), -1)
+    l2 = SoftTargetCrossEntropy()(x, soft)
+    l2_ref = SoftTargetCrossEntropy()(x.detach().cpu(), soft.cpu())
+    assert abs(l2.item() - l2_ref.item()) < 1e-5
+
+
+# ---------------- fused uint8 normalize ----------------
+
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16, torch.float16])
+def test_u8_normalize(dtype):
+    torch.manual_seed(0)
+    x = torch.randint(0, 256, (4, 3, 32, 32), device='cuda', dtype=torch.uint8)
+    mean = torch.tensor([0.485, 0.456, 0.406], device='cuda') * 255
+    std = torch.tensor([0.229, 0.224, 0.225], device='cuda') * 255
+    got = ops.u8_normalize(x, mean.view(1, 3, 1, 1), std.view(1, 3, 1, 1), dtype)
+    ref = (x.float() - mean.view(1, 3, 1, 1)) / std.view(1, 3, 1, 1)
+    assert got.dtype == dtype
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    err = (got.float() - ref).abs().max().item()
+    assert err < tol, f'err {err}'
+
+
+def test_u8_normalize_odd_plane():
+    # H*W not a multiple of 4 exercises the channel-boundary slow path
+    x = torch.randint(0, 256, (2, 3, 5, 10), device='cuda', dtype=torch.uint8)
+    mean = torch.tensor([10., 20., 30.], device='cuda')
+    std = torch.tensor([2., 3., 4.], device='cuda')
+    got = ops.u8_normalize(x, mean, std, torch.float32)
+    ref = (x.float() - mean.view(1, 3, 1, 1)) / std.view(1, 3, 1, 1)
+    assert (got - ref).abs().max().item() < 1e-5
+
+
+# ---------------- masked global pooling ----------------
+
+@pytest.mark.parametrize('pool_type', ['avg', 'max', 'avgmax'])
+@pytest.mark.parametrize('dtype', [torch.float32, torch.bfloat16])
+def test_masked_global_pool(pool_type, dtype):
+    torch.manual_seed(0)
+    B, N, C = 4, 37, 192
+    x = torch.randn(B, N, C, device='cuda', dtype=dtype, requires_grad=True)
+    valid = torch.rand(B, N, device='cuda') > 0.3
+    valid[:, 0] = True  # at least one valid token per row
+
+    out = ops.masked_global_pool(x, valid, pool_type)
+    out.sum().backward()
+    got_grad = x.grad.clone()
+
+    xr = x.detach().float().cpu().requires_grad_(True)
+    vr = valid.cpu()
+    vm = vr.float()
+    denom = vm.sum(1, keepdim=True).clamp(min=1)
+    avg = (xr * vm.unsqueeze(-1)).sum(1) / denom
+    mx = xr.masked_fill(~vr.unsqueeze(-1), float('-inf')).amax(1)
+    ref = {'avg': avg, 'max': mx, 'avgmax': 0.5 * (avg + mx)}[pool_type]
+    ref.sum().backward()
+
+    tol = 1e-5 if dtype == torch.float32 else 2e-2
+    assert (out.float().cpu() - ref).abs().max().item() < tol
+    assert (got_grad.float().cpu() - xr.grad).abs().max().item() < tol

@@ -15,6 +15,8 @@ from itertools import repeat
 from typing import Any, Callable, Iterator, List, Optional, Tuple, Union
 
 import torch
+
+from .. import ops
 import torch.utils.data
 import numpy as np
 
@@ -135,7 +137,11 @@ class PrefetchLoader:
             with stream_context():
                 next_input = next_input.to(device=self.device, non_blocking=True)
                 next_target = next_target.to(device=self.device, non_blocking=True)
-                next_input = next_input.to(self.img_dtype).sub_(self.mean).div_(self.std)
+                if next_input.dtype == torch.uint8 and next_input.dim() == 4:
+                    # fused uint8 -> normalized tensor (one HIP kernel, csrc/data_ops.hip)
+                    next_input = ops.u8_normalize(next_input, self.mean, self.std, self.img_dtype)
+                else:
+                    next_input = next_input.to(self.img_dtype).sub_(self.mean).div_(self.std)
                 if self.random_erasing is not None:
                     next_input = self.random_erasing(next_input)
 

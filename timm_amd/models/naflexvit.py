@@ -142,6 +142,10 @@ def global_pool_naflex(
 
     if num_prefix_tokens:
         x = x[:, num_prefix_tokens:]
+    if x.is_cuda:
+        # fused mask+reduce kernel (csrc/data_ops.hip); bwd from count/argmax
+        from .. import ops
+        return ops.masked_global_pool(x, patch_valid, pool_type)
     patch_valid = patch_valid.to(x.dtype)
     denom = patch_valid.sum(dim=1, keepdim=True).clamp(min=1)
     if pool_type == 'avg':

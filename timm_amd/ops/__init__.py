@@ -72,6 +72,8 @@ from .layer_norm import layer_norm_act, rms_norm_act, layer_norm, rms_norm  # no
 from .attention import flash_attention, flash_attention_qkv, attention_available  # noqa: E402
 from .elementwise import bias_act, residual_scale_add  # noqa: E402
 from .fused_optim import fused_adamw_step, fused_lerp_, fused_l2norm  # noqa: E402
+from .loss import fused_cross_entropy  # noqa: E402
+from .data import u8_normalize, masked_global_pool  # noqa: E402
 
 __all__ = [
     'has_ext', 'require_ext', 'use_hip',
@@ -80,4 +82,5 @@ __all__ = [
     'flash_attention', 'flash_attention_qkv', 'attention_available',
     'bias_act', 'residual_scale_add',
     'fused_adamw_step', 'fused_lerp_', 'fused_l2norm',
+    'fused_cross_entropy', 'u8_normalize', 'masked_global_pool',
 ]

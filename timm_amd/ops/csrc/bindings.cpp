@@ -48,6 +48,17 @@ at::Tensor ns_gemm_nt(at::Tensor l, at::Tensor r, c10::optional<at::Tensor> s,
 at::Tensor ns_gemm_nn(at::Tensor l, at::Tensor r, c10::optional<at::Tensor> s,
                       double alpha, double beta);
 
+
+// ce_loss.hip
+std::vector<at::Tensor> ce_loss_fwd(at::Tensor logits, at::Tensor target, double smoothing);
+at::Tensor ce_loss_bwd(at::Tensor logits, at::Tensor target, at::Tensor lse,
+                       at::Tensor dloss, double smoothing);
+
+// data_ops.hip
+at::Tensor u8_normalize(at::Tensor x, at::Tensor mean, at::Tensor inv_std, at::ScalarType out_dtype);
+std::vector<at::Tensor> masked_pool_fwd(at::Tensor x, at::Tensor valid, bool is_max);
+at::Tensor masked_pool_bwd(at::Tensor dy, at::Tensor valid, at::Tensor aux, long n_tokens, bool is_max);
+
 // multi_tensor.hip
 void multi_tensor_adamw(
     std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
@@ -75,6 +86,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dwconv_bwd_weight", &dwconv_bwd_weight, "NHWC depthwise conv bwd-weight");
   m.def("ns_gemm_nt", &ns_gemm_nt, "batched bf16 MFMA GEMM (L R^T + S) for Muon NS");
   m.def("ns_gemm_nn", &ns_gemm_nn, "batched bf16 MFMA GEMM (L R + S) for Muon NS");
+  m.def("ce_loss_fwd", &ce_loss_fwd, "fused cross-entropy fwd (loss + lse)");
+  m.def("ce_loss_bwd", &ce_loss_bwd, "fused cross-entropy bwd (dlogits from lse)");
+  m.def("u8_normalize", &u8_normalize, "fused uint8 -> normalized tensor (loader prefetch)");
+  m.def("masked_pool_fwd", &masked_pool_fwd, "masked global pool fwd (NaFlex)");
+  m.def("masked_pool_bwd", &masked_pool_bwd, "masked global pool bwd (NaFlex)");
   m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
   m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");
