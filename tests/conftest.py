@@ -1,8 +1,14 @@
 import os
 import sys
 
+# xdist runs 4 workers on an 8-core box: cap intra-op threads per worker so
+# the model matrix doesn't thrash on oversubscribed OpenMP pools
+os.environ.setdefault('OMP_NUM_THREADS', '2')
+
 import pytest
 import torch
+
+torch.set_num_threads(max(1, (os.cpu_count() or 8) // 4))
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
