@@ -187,3 +187,25 @@ def test_train_script_naflex_distributed(tmp_path):
     r = subprocess.run(cmd, capture_output=True, text=True, timeout=600, cwd=repo)
     assert r.returncode == 0, f'naflex train failed:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}'
     assert (tmp_path / 'nfsmoke' / 'last.pth.tar').exists()
+
+
+def test_train_script_kd(tmp_path):
+    """train.py end-to-end with each KD task type on CPU synthetic data."""
+    import subprocess, sys, torch, timm_amd
+    teacher_path = str(tmp_path / 'teacher.pth')
+    torch.save(timm_amd.create_model('resnet18', num_classes=10).state_dict(), teacher_path)
+    base = [
+        sys.executable, 'train.py', '--synthetic', '--synthetic-len', '16', '-b', '8',
+        '--num-classes', '10', '--epochs', '1', '--device', 'cpu', '--no-prefetcher',
+        '--workers', '0', '--kd-model-name', 'resnet18', '--kd-teacher-path', teacher_path,
+        '--task-loss-weight', '0.5', '--output', str(tmp_path),
+    ]
+    for model, kd_type, extra in (
+            ('resnet18', 'logit', ['--kd-temperature', '4']),
+            ('resnet18', 'feature', []),
+            ('deit_tiny_distilled_patch16_224', 'token', ['--kd-token-distill-type', 'hard']),
+    ):
+        r = subprocess.run(
+            base + ['--model', model, '--kd-distill-type', kd_type, '--experiment', f'kd_{kd_type}'] + extra,
+            capture_output=True, text=True, timeout=900, cwd=os.path.dirname(os.path.dirname(__file__)))
+        assert r.returncode == 0, f'{kd_type}: {r.stdout[-1500:]}\n{r.stderr[-1500:]}'

@@ -651,4 +651,17 @@ def test_masked_global_pool(pool_type, dtype):
 
     tol = 1e-5 if dtype == torch.float32 else 2e-2
     assert (out.float().cpu() - ref).abs().max().item() < tol
-    assert (got_grad.float().cpu() - xr.grad).abs().max().item() < tol
+    if pool_type == 'avg' or dtype == torch.float32:
+        assert (got_grad.float().cpu() - xr.grad).abs().max().item() < tol
+    else:
+        # bf16 max has frequent ties: torch splits the gradient among tied
+        # maxima, the kernel routes it to one of them. Check the tie-agnostic
+        # invariants instead: per-(b, c) grad mass is right and the max part
+        # lands only on valid max positions.
+        g = got_grad.float().cpu()
+        expect_sum = {'max': 1.0, 'avgmax': 1.0}[pool_type]  # d(sum out)/d contribution per (b,c)
+        assert torch.allclose(g.sum(dim=1), torch.full((B, C), expect_sum), atol=2e-2)
+        if pool_type == 'max':
+            xm = x.detach().float().cpu().masked_fill(~vr.unsqueeze(-1), float('-inf'))
+            is_max = xm == xm.amax(dim=1, keepdim=True)
+            assert bool(((g != 0) <= is_max).all())

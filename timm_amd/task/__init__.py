@@ -1,4 +1,5 @@
 from ._helpers import resume_task_checkpoint
 from .classification import ClassificationTask
 from .distillation import DistillationTeacher, LogitDistillationTask, FeatureDistillationTask
+from .token_distillation import TokenDistillationTeacher, TokenDistillationTask
 from .task import TrainingTask

@@ -36,7 +36,10 @@ from timm_amd.loss import BinaryCrossEntropy, JsdCrossEntropy, LabelSmoothingCro
 from timm_amd.models import create_model, safe_model_name
 from timm_amd.optim import create_optimizer_v2, optimizer_kwargs
 from timm_amd.scheduler import create_scheduler_v2, scheduler_kwargs
-from timm_amd.task import ClassificationTask, resume_task_checkpoint
+from timm_amd.task import (
+    ClassificationTask, FeatureDistillationTask, LogitDistillationTask, TokenDistillationTask,
+    resume_task_checkpoint,
+)
 from timm_amd.utils import ModelEmaV3, NativeScaler, dispatch_clip_grad
 
 _logger = logging.getLogger('train')
@@ -88,6 +91,31 @@ group.add_argument('--model-kwargs', nargs='*', default={}, action=utils.ParseKw
 group.add_argument('--torchcompile', nargs='?', type=str, default=None, const='inductor')
 
 # Device & distributed
+group = parser.add_argument_group('Knowledge-distillation parameters')
+group.add_argument('--kd-model-name', default=None, type=str,
+                   help='Name of teacher model for knowledge distillation')
+group.add_argument('--kd-distill-type', default='logit', type=str, choices=['logit', 'feature', 'token'],
+                   help='Type of distillation: "logit" (output KL), "feature" (intermediate features), '
+                        '"token" (models with distillation heads, e.g. deit distilled) (default: logit)')
+group.add_argument('--kd-loss-type', default='kl', type=str,
+                   help='Loss function for logit distillation (default: kl)')
+group.add_argument('--distill-loss-weight', default=None, type=float,
+                   help='Weight for distillation loss. Both weights given -> independent mix; '
+                        'only --task-loss-weight -> complementary (distill = 1 - task).')
+group.add_argument('--task-loss-weight', default=None, type=float,
+                   help='Weight for task (classification) loss; see --distill-loss-weight')
+group.add_argument('--kd-temperature', default=4.0, type=float,
+                   help='Softmax temperature for distillation (default: 4.0)')
+group.add_argument('--kd-teacher-path', default=None, type=str,
+                   help='Local checkpoint path for the teacher weights')
+group.add_argument('--kd-student-feature-dim', default=None, type=int,
+                   help='Student feature dim for feature distillation (auto-detected if unset)')
+group.add_argument('--kd-teacher-feature-dim', default=None, type=int,
+                   help='Teacher feature dim for feature distillation (auto-detected if unset)')
+group.add_argument('--kd-token-distill-type', default='soft', type=str, choices=['soft', 'hard'],
+                   help='Token distillation: "soft" KL w/ temperature, "hard" CE vs teacher argmax')
+
+# Device parameters
 group = parser.add_argument_group('Device parameters')
 group.add_argument('--device', default='cuda', type=str)
 group.add_argument('--amp', action='store_true', default=False, help='use mixed precision training')
@@ -343,12 +371,6 @@ def main():
     if use_amp == 'native':
         amp_autocast = partial(torch.autocast, device_type=device.type, dtype=amp_dtype)
 
-    optimizer = create_optimizer_v2(
-        model,
-        **optimizer_kwargs(cfg=args),
-        **args.opt_kwargs,
-    )
-
     # setup learning rate schedule and starting epoch
     resume_epoch = None
 
@@ -382,7 +404,59 @@ def main():
     train_loss_fn = train_loss_fn.to(device=device)
     validate_loss_fn = nn.CrossEntropyLoss().to(device=device)
 
-    task = ClassificationTask(model, criterion=train_loss_fn)
+    if args.kd_model_name is not None:
+        if args.kd_distill_type == 'logit':
+            task = LogitDistillationTask(
+                student_model=model,
+                teacher_model=args.kd_model_name,
+                criterion=train_loss_fn,
+                teacher_pretrained_path=args.kd_teacher_path,
+                loss_type=args.kd_loss_type,
+                distill_loss_weight=args.distill_loss_weight,
+                task_loss_weight=args.task_loss_weight,
+                temperature=args.kd_temperature,
+                device=device,
+                verbose=utils.is_primary(args),
+            )
+        elif args.kd_distill_type == 'feature':
+            task = FeatureDistillationTask(
+                student_model=model,
+                teacher_model=args.kd_model_name,
+                criterion=train_loss_fn,
+                teacher_pretrained_path=args.kd_teacher_path,
+                distill_loss_weight=args.distill_loss_weight,
+                task_loss_weight=args.task_loss_weight,
+                student_feature_dim=args.kd_student_feature_dim,
+                teacher_feature_dim=args.kd_teacher_feature_dim,
+                device=device,
+                verbose=utils.is_primary(args),
+            )
+        elif args.kd_distill_type == 'token':
+            task = TokenDistillationTask(
+                student_model=model,
+                teacher_model=args.kd_model_name,
+                criterion=train_loss_fn,
+                teacher_pretrained_path=args.kd_teacher_path,
+                distill_type=args.kd_token_distill_type,
+                distill_loss_weight=args.distill_loss_weight,
+                task_loss_weight=args.task_loss_weight,
+                temperature=args.kd_temperature,
+                device=device,
+                verbose=utils.is_primary(args),
+            )
+        else:
+            raise ValueError(f'Unknown distillation type: {args.kd_distill_type}')
+        task.teacher.to(device=device)
+    else:
+        task = ClassificationTask(model, criterion=train_loss_fn)
+
+    # optimizer over the task's trainable module (includes e.g. the feature
+    # distillation projection, not just the bare model)
+    optimizer = create_optimizer_v2(
+        task.model,
+        **optimizer_kwargs(cfg=args),
+        **args.opt_kwargs,
+    )
 
     if args.resume:
         resume_epoch = resume_task_checkpoint(
