@@ -209,3 +209,63 @@ def test_train_script_kd(tmp_path):
             base + ['--model', model, '--kd-distill-type', kd_type, '--experiment', f'kd_{kd_type}'] + extra,
             capture_output=True, text=True, timeout=900, cwd=os.path.dirname(os.path.dirname(__file__)))
         assert r.returncode == 0, f'{kd_type}: {r.stdout[-1500:]}\n{r.stderr[-1500:]}'
+
+
+def test_scheduled_batch_sampler_determinism():
+    """Same (seed, epoch, length) -> identical schedule on every 'rank'."""
+    import torch
+    from timm_amd.data import ScheduledBatchSampler
+
+    class _FixedSampler:
+        def __init__(self, n):
+            self.n = n
+        def __len__(self):
+            return self.n
+        def __iter__(self):
+            return iter(range(self.n))
+
+    def shapes(seed, epoch):
+        s = ScheduledBatchSampler(_FixedSampler(64), batch_sizes=[8, 4, 16], seed=seed)
+        s.set_epoch(epoch)
+        return [len(b) for b in s]
+
+    assert shapes(0, 1) == shapes(0, 1)
+    assert sum(shapes(0, 1)) <= 64
+    # epochs reshuffle but keep the same composition
+    a, b = sorted(shapes(0, 0)), sorted(shapes(0, 1))
+    assert a == b
+
+    # progressive schedule moves probability mass from first to last choice
+    s = ScheduledBatchSampler(
+        _FixedSampler(640), batch_sizes=[8, 4, 16], choice_schedule='progressive',
+        schedule_epochs=10, schedule_random_mix=0.0)
+    w0 = s.choice_weights_for_epoch(0)
+    w9 = s.choice_weights_for_epoch(9)
+    assert w0[0] > w0[2] and w9[2] > w9[0]
+
+
+def test_scheduled_transform_dataset():
+    from timm_amd.data import ScheduledTransformDataset
+
+    class _DS:
+        def __getitem__(self, i):
+            return (i, 'target')
+        def __len__(self):
+            return 4
+
+    ds = ScheduledTransformDataset(_DS(), [lambda x: x * 10, lambda x: x * 100])
+    assert ds[(3, 0)] == (30, 'target')
+    assert ds[(3, 1)] == (300, 'target')
+
+
+def test_train_script_scheduled_res(tmp_path):
+    import subprocess, sys
+    r = subprocess.run([
+        sys.executable, 'train.py', '--model', 'resnet18', '--synthetic',
+        '--synthetic-len', '48', '-b', '8', '--num-classes', '10', '--epochs', '1',
+        '--device', 'cpu', '--no-prefetcher', '--workers', '0',
+        '--train-img-sizes', '96', '128', '--train-batch-sizes', '10', '6',
+        '--variable-batch-loss-scale', 'linear', '--output', str(tmp_path),
+    ], capture_output=True, text=True, timeout=600,
+        cwd=os.path.dirname(os.path.dirname(__file__)))
+    assert r.returncode == 0, r.stdout[-1200:] + r.stderr[-1200:]

@@ -7,6 +7,7 @@ from .constants import *
 from .dataset import ImageDataset, IterableImageDataset, AugMixDataset
 from .dataset_factory import create_dataset
 from .distributed_sampler import OrderedDistributedSampler, RepeatAugSampler
+from .scheduled_sampler import ScheduledBatchSampler, ScheduledTransformDataset
 from .loader import create_loader, fast_collate, PrefetchLoader, MultiEpochsDataLoader
 from .mixup import Mixup, FastCollateMixup, mixup_target, rand_bbox, rand_bbox_minmax
 from .naflex_mixup import NaFlexMixup, mix_batch_variable_size, pairwise_mixup_target

@@ -12,7 +12,7 @@ import random
 from contextlib import suppress
 from functools import partial
 from itertools import repeat
-from typing import Any, Callable, Iterator, List, Optional, Tuple, Union
+from typing import Any, Callable, Iterator, List, Optional, Sequence, Tuple, Union
 
 import torch
 
@@ -23,6 +23,7 @@ import numpy as np
 from .constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
 from .dataset import IterableImageDataset, ImageDataset
 from .distributed_sampler import OrderedDistributedSampler, RepeatAugSampler
+from .scheduled_sampler import ScheduledBatchSampler, ScheduledTransformDataset
 from .random_erasing import RandomErasing
 from .mixup import FastCollateMixup
 from .transforms_factory import create_transform
@@ -238,6 +239,15 @@ def create_loader(
         persistent_workers: bool = True,
         worker_seeding: str = 'all',
         tf_preprocessing: bool = False,
+        input_size_choices: Optional[Sequence] = None,
+        batch_size_choices: Optional[Sequence[int]] = None,
+        batch_choice_weights: Optional[Sequence[float]] = None,
+        batch_choice_seed: int = 0,
+        batch_choice_schedule: str = 'constant',
+        batch_schedule_epochs: Optional[int] = None,
+        batch_schedule_spread: float = 0.65,
+        batch_schedule_random_mix: float = 0.1,
+        num_batches: Optional[int] = None,
 ):
     """Create the train/eval loader (reference `loader.py:205-469`)."""
     re_num_splits = 0
@@ -250,9 +260,45 @@ def create_loader(
         # are correct before worker processes are launched
         dataset.set_loader_cfg(num_workers=num_workers)
 
+    scheduled_batching = input_size_choices is not None
+    if scheduled_batching:
+        if not is_training:
+            raise ValueError('Scheduled input sizes are only supported for training loaders.')
+        if use_multi_epochs_loader:
+            raise ValueError('MultiEpochsDataLoader is not supported with scheduled input sizes.')
+        if isinstance(dataset, torch.utils.data.IterableDataset):
+            raise TypeError('Scheduled input sizes require a map-style dataset.')
+        if num_aug_splits > 0:
+            raise ValueError('Augmentation splits are not supported with scheduled input sizes.')
+        if not input_size_choices:
+            raise ValueError('input_size_choices must contain at least one size.')
+        channels = input_size[0] if isinstance(input_size, (tuple, list)) and len(input_size) == 3 else len(mean)
+        resolved_input_sizes = []
+        for size in input_size_choices:
+            if isinstance(size, int):
+                size = (channels, size, size)
+            elif len(size) == 2:
+                size = (channels, *size)
+            elif len(size) == 3:
+                size = tuple(size)
+                if size[0] != channels:
+                    raise ValueError('All scheduled input sizes must use the same number of channels.')
+            else:
+                raise ValueError('Scheduled input sizes must be scalars, HW tuples, or CHW tuples.')
+            if any(d <= 0 for d in size):
+                raise ValueError('All scheduled input size dimensions must be positive.')
+            resolved_input_sizes.append(size)
+        if batch_size_choices is None:
+            batch_size_choices = [batch_size] * len(resolved_input_sizes)
+        elif len(batch_size_choices) != len(resolved_input_sizes):
+            raise ValueError('batch_size_choices and input_size_choices must have the same length.')
+        if batch_choice_weights is not None and len(batch_choice_weights) != len(resolved_input_sizes):
+            raise ValueError('batch_choice_weights and input_size_choices must have the same length.')
+
     if getattr(dataset, 'transform', None) is None or not tf_preprocessing:
-        dataset.transform = create_transform(
-            input_size,
+        def _make_transform(_size):
+            return create_transform(
+            _size,
             is_training=is_training,
             no_aug=no_aug,
             train_crop_mode=train_crop_mode,
@@ -280,6 +326,13 @@ def create_loader(
             separate=num_aug_splits > 0,
         )
 
+        if scheduled_batching:
+            transforms = [_make_transform(size) for size in resolved_input_sizes]
+            dataset.transform = None
+            dataset = ScheduledTransformDataset(dataset, transforms)
+        else:
+            dataset.transform = _make_transform(input_size)
+
     if isinstance(dataset, IterableImageDataset):
         # wrap dataset in AugMix helper
         if num_aug_splits > 1:
@@ -299,6 +352,9 @@ def create_loader(
     else:
         assert num_aug_repeats == 0, "RepeatAugment not currently supported in non-distributed or IterableDataset use"
 
+    if scheduled_batching and sampler is None:
+        sampler = torch.utils.data.RandomSampler(dataset)
+
     if collate_fn is None:
         collate_fn = fast_collate if use_prefetcher else torch.utils.data.dataloader.default_collate
 
@@ -306,17 +362,38 @@ def create_loader(
     if use_multi_epochs_loader:
         loader_class = MultiEpochsDataLoader
 
-    loader_args = dict(
-        batch_size=batch_size,
-        shuffle=not isinstance(dataset, torch.utils.data.IterableDataset) and sampler is None and is_training,
-        num_workers=num_workers,
-        sampler=sampler,
-        collate_fn=collate_fn,
-        pin_memory=pin_memory,
-        drop_last=is_training,
-        worker_init_fn=partial(_worker_init, worker_seeding=worker_seeding),
-        persistent_workers=persistent_workers and num_workers > 0,
-    )
+    if scheduled_batching:
+        loader_args = dict(
+            batch_sampler=ScheduledBatchSampler(
+                sampler,
+                batch_sizes=batch_size_choices,
+                choice_weights=batch_choice_weights,
+                seed=batch_choice_seed,
+                drop_last=is_training,
+                num_batches=num_batches,
+                choice_schedule=batch_choice_schedule,
+                schedule_epochs=batch_schedule_epochs,
+                schedule_spread=batch_schedule_spread,
+                schedule_random_mix=batch_schedule_random_mix,
+            ),
+            num_workers=num_workers,
+            collate_fn=collate_fn,
+            pin_memory=pin_memory,
+            worker_init_fn=partial(_worker_init, worker_seeding=worker_seeding),
+            persistent_workers=persistent_workers and num_workers > 0,
+        )
+    else:
+        loader_args = dict(
+            batch_size=batch_size,
+            shuffle=not isinstance(dataset, torch.utils.data.IterableDataset) and sampler is None and is_training,
+            num_workers=num_workers,
+            sampler=sampler,
+            collate_fn=collate_fn,
+            pin_memory=pin_memory,
+            drop_last=is_training,
+            worker_init_fn=partial(_worker_init, worker_seeding=worker_seeding),
+            persistent_workers=persistent_workers and num_workers > 0,
+        )
     try:
         loader = loader_class(dataset, **loader_args)
     except TypeError:

@@ -199,6 +199,22 @@ group.add_argument('--mixup-mode', type=str, default='batch')
 group.add_argument('--mixup-off-epoch', default=0, type=int, metavar='N')
 group.add_argument('--smoothing', type=float, default=0.1)
 # NaFlex variable-resolution training (reference train.py:423-435)
+group.add_argument('--train-img-sizes', type=int, nargs='+', default=None,
+                   help='Per-batch square image sizes for scheduled resolution training')
+group.add_argument('--train-batch-sizes', type=int, nargs='+', default=None,
+                   help='Batch size for each --train-img-sizes choice (default: --batch-size for every choice)')
+group.add_argument('--train-size-probs', type=float, nargs='+', default=None,
+                   help='Base sampling weights for --train-img-sizes (default: uniform)')
+group.add_argument('--train-size-schedule', type=str, default='constant', choices=('constant', 'progressive'),
+                   help='Resolution choice schedule; progressive moves from the first size to the last')
+group.add_argument('--train-size-schedule-spread', type=float, default=0.65,
+                   help='Progressive schedule spread in resolution-choice index units (default: 0.65)')
+group.add_argument('--train-size-random-mix', type=float, default=0.1,
+                   help='Fraction of uniform random active choices mixed into a progressive schedule')
+group.add_argument('--train-batches-per-epoch', '--train-steps-per-epoch', type=int, default=None,
+                   help='Fixed loader batches (before grad accumulation) per epoch; inferred if unspecified')
+group.add_argument('--variable-batch-loss-scale', default='none', type=str, choices=('none', 'sqrt', 'linear'),
+                   help='Scale gradients relative to the policy-average scheduled batch size')
 group.add_argument('--naflex-loader', action='store_true', default=False,
                    help='enable the NaFlex variable-seq-len loader')
 group.add_argument('--naflex-train-seq-lens', type=int, nargs='+',
@@ -644,6 +660,15 @@ def main():
             device=device,
             use_prefetcher=args.prefetcher,
             worker_seeding=args.worker_seeding,
+            input_size_choices=args.train_img_sizes,
+            batch_size_choices=args.train_batch_sizes,
+            batch_choice_weights=args.train_size_probs,
+            batch_choice_seed=args.seed,
+            batch_choice_schedule=args.train_size_schedule,
+            batch_schedule_epochs=args.epochs if args.train_size_schedule == 'progressive' else None,
+            batch_schedule_spread=args.train_size_schedule_spread,
+            batch_schedule_random_mix=args.train_size_random_mix,
+            num_batches=args.train_batches_per_epoch,
         )
 
         loader_eval = create_loader(
@@ -737,11 +762,26 @@ def main():
 
     results = []
     try:
+        scheduled_batch_mode = args.train_img_sizes is not None
+        batch_size_reference = float(args.batch_size)
+        def _sched_sampler(loader):
+            inner = getattr(loader, 'loader', loader)  # unwrap PrefetchLoader
+            return inner.batch_sampler
+        if scheduled_batch_mode:
+            batch_size_reference = _sched_sampler(loader_train).average_batch_size
+            if utils.is_primary(args):
+                _logger.info(
+                    f'Scheduled training resolutions {args.train_img_sizes} '
+                    f'batch sizes {_sched_sampler(loader_train).batch_sizes} '
+                    f'({len(_sched_sampler(loader_train))} batches/epoch).')
+
         for epoch in range(start_epoch, num_epochs):
             if hasattr(dataset_train, 'set_epoch'):
                 dataset_train.set_epoch(epoch)
             elif args.distributed and hasattr(loader_train.sampler, 'set_epoch'):
                 loader_train.sampler.set_epoch(epoch)
+            if scheduled_batch_mode:
+                _sched_sampler(loader_train).set_epoch(epoch)
             if mixup_fn is not None and args.mixup_off_epoch and epoch >= args.mixup_off_epoch:
                 mixup_fn.mixup_enabled = False
 
@@ -758,6 +798,8 @@ def main():
                 loss_scaler=loss_scaler,
                 mixup_fn=mixup_fn,
                 num_updates_total=num_epochs * updates_per_epoch,
+                scheduled_batch_mode=scheduled_batch_mode,
+                batch_size_reference=batch_size_reference,
             )
 
             if args.distributed and args.dist_bn in ('broadcast', 'reduce'):
@@ -847,6 +889,8 @@ def train_one_epoch(
         amp_autocast=suppress,
         loss_scaler=None,
         mixup_fn=None,
+        scheduled_batch_mode=False,
+        batch_size_reference=None,
         num_updates_total=None,
 ):
     device = device or torch.device(args.device)
@@ -947,6 +991,15 @@ def train_one_epoch(
                 global_bs = utils.reduce_tensor(
                     torch.tensor(batch_size, device=device, dtype=torch.float32), 1)
                 loss_mult = loss_mult * args.world_size * batch_size / global_bs.item()
+        elif scheduled_batch_mode:
+            # schedule is identical on every rank (seed-derived), so a local
+            # rescale to the policy-average batch size is enough
+            scale_mode = args.variable_batch_loss_scale
+            if scale_mode and scale_mode != 'none':
+                ref = batch_size_reference or float(args.batch_size)
+                loss_mult = batch_size / ref
+                if scale_mode == 'sqrt':
+                    loss_mult = loss_mult ** 0.5
 
         if has_no_sync and not need_update:
             with task.no_sync():
