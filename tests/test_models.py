@@ -29,6 +29,7 @@ EXCLUDE_FILTERS = [
     '*_enc', 'mobilenetv5_300m*', 'gemma4_vit_570m*', 'fastvit_mci4',
     'nfnet_f5*', 'nfnet_f6*', 'nfnet_f7*', 'eca_nfnet_l3', 'regnety_640', 'regnety_1280',
     'regnety_2560', 'resnet50x16_clip*', 'resnet50x64_clip*', 'resnext101_32x16d', 'resnext101_32x32d',
+    'vit_7b*', 'eva02_enormous*', 'vit_giantopt*', 'sam2_hiera_large*',
 ]
 
 

@@ -24,6 +24,7 @@ from .eca import CecaModule, EcaModule, EfficientChannelAttn
 from .gather_excite import GatherExcite
 from .global_context import GlobalContext
 from .create_norm import create_norm_layer, get_norm_layer
+from .diff_attention import DiffAttention
 from .drop import DropBlock2d, DropPath, calculate_drop_path_rates, drop_block_2d, drop_path
 from .format import Format, FormatT, get_channel_dim, get_spatial_dim, nchw_to, nhwc_to
 from .grn import GlobalResponseNorm

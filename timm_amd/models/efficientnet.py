@@ -13,7 +13,9 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD, IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD
-from ..layers import GroupNormAct, create_conv2d, create_classifier, get_norm_act_layer
+from ..layers import (
+    EvoNorm2dS0, GroupNormAct, LayerNormAct2d, create_conv2d, create_classifier, get_norm_act_layer,
+)
 from ._builder import build_model_with_cfg, pretrained_cfg_for_features
 from ._efficientnet_blocks import SqueezeExcite
 from ._efficientnet_builder import (

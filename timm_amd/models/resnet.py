@@ -21,6 +21,7 @@ from ..data.constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
 from ..layers import (
     AvgPool2dSame, BlurPool2d, DropBlock2d, DropPath, GroupNorm, SEModule, calculate_drop_path_rates,
     create_classifier, get_act_layer, get_attn, get_norm_layer, create_attn_layer,
+    to_ntuple,
 )
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
@@ -262,7 +263,7 @@ def drop_blocks(drop_prob: float = 0.):
 
 
 def make_blocks(
-        block_fn: Union[Type[BasicBlock], Type[Bottleneck]],
+        block_fns: Union[Type[BasicBlock], Type[Bottleneck], Tuple],
         channels: Tuple[int, ...],
         block_repeats: Tuple[int, ...],
         inplanes: int,
@@ -280,7 +281,9 @@ def make_blocks(
     net_block_idx = 0
     net_stride = 4
     dilation = prev_dilation = 1
-    for stage_idx, (planes, num_blocks, db) in enumerate(zip(channels, block_repeats, drop_blocks(drop_block_rate))):
+    block_fns = to_ntuple(len(channels))(block_fns)
+    for stage_idx, (block_fn, planes, num_blocks, db) in enumerate(
+            zip(block_fns, channels, block_repeats, drop_blocks(drop_block_rate))):
         stage_name = f'layer{stage_idx + 1}'  # never liked this name, but weight compat requires it
         stride = 1 if stage_idx == 0 else 2
         if net_stride >= output_stride:
@@ -430,7 +433,8 @@ class ResNet(nn.Module):
         self.feature_info.extend(stage_feature_info)
 
         # Head (Pooling and Classifier)
-        self.num_features = self.head_hidden_size = channels[-1] * block.expansion
+        last_block = block[-1] if isinstance(block, (tuple, list)) else block
+        self.num_features = self.head_hidden_size = channels[-1] * last_block.expansion
         self.global_pool, self.fc = create_classifier(self.num_features, self.num_classes, pool_type=global_pool)
 
         self.init_weights(zero_init_last=zero_init_last)

@@ -18,6 +18,7 @@ from ..data.constants import IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD
 from ..layers import (
     AvgPool2dSame, BatchNormAct2d, ClassifierHead, DropPath, GroupNormAct, StdConv2d, create_conv2d,
     create_pool2d, get_act_layer, get_norm_act_layer, make_divisible,
+    FilterResponseNormTlu2d,
 )
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices

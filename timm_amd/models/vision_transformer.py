@@ -35,6 +35,7 @@ from ..layers import (
     Attention, AttentionPoolLatent, DropPath, LayerNorm, LayerScale, Mlp, PatchDropout, PatchEmbed, RmsNorm,
     SwiGLUPacked, SwiGLU, get_act_layer, get_norm_layer, init_weight_jax, init_weight_vit, lecun_normal_,
     resample_abs_pos_embed, resample_patch_embed, trunc_normal_, use_fused_attn,
+    DiffAttention,
 )
 from ._builder import build_model_with_cfg
 from ._features import feature_take_indices
@@ -72,6 +73,49 @@ def global_pool_nlc(
     return x
 
 
+
+ATTN_LAYERS = {
+    '': Attention,
+    'attn': Attention,
+    'diff': DiffAttention,
+}
+
+
+def _create_attn(
+        attn_layer,
+        dim: int,
+        num_heads: int,
+        qkv_bias: bool = False,
+        qk_norm: bool = False,
+        scale_norm: bool = False,
+        proj_bias: bool = True,
+        attn_drop: float = 0.,
+        proj_drop: float = 0.,
+        norm_layer: Optional[Type[nn.Module]] = None,
+        depth: int = 0,
+        **kwargs,
+) -> nn.Module:
+    """Resolve an attention layer by name/class (reference `vision_transformer.py:92`)."""
+    if isinstance(attn_layer, str):
+        resolved = ATTN_LAYERS.get(attn_layer, None)
+        assert resolved is not None, f'Unknown attn_layer: {attn_layer}'
+        attn_layer = resolved
+    if isinstance(attn_layer, type) and issubclass(attn_layer, DiffAttention):
+        kwargs['depth'] = depth
+    return attn_layer(
+        dim,
+        num_heads=num_heads,
+        qkv_bias=qkv_bias,
+        qk_norm=qk_norm,
+        scale_norm=scale_norm,
+        proj_bias=proj_bias,
+        attn_drop=attn_drop,
+        proj_drop=proj_drop,
+        norm_layer=norm_layer,
+        **kwargs,
+    )
+
+
 class Block(nn.Module):
     """Pre-norm transformer block (reference `vision_transformer.py:128`).
 
@@ -95,10 +139,13 @@ class Block(nn.Module):
             act_layer: Type[nn.Module] = nn.GELU,
             norm_layer: Type[nn.Module] = LayerNorm,
             mlp_layer: Type[nn.Module] = Mlp,
+            attn_layer = Attention,
+            depth: int = 0,
     ) -> None:
         super().__init__()
         self.norm1 = norm_layer(dim)
-        self.attn = Attention(
+        self.attn = _create_attn(
+            attn_layer,
             dim,
             num_heads=num_heads,
             qkv_bias=qkv_bias,
@@ -108,6 +155,7 @@ class Block(nn.Module):
             attn_drop=attn_drop,
             proj_drop=proj_drop,
             norm_layer=norm_layer,
+            depth=depth,
         )
         self.ls1 = LayerScale(dim, init_values=init_values) if init_values else nn.Identity()
         self.drop_path1 = DropPath(drop_path) if drop_path > 0. else nn.Identity()
@@ -160,11 +208,14 @@ class ResPostBlock(nn.Module):
             act_layer: Type[nn.Module] = nn.GELU,
             norm_layer: Type[nn.Module] = LayerNorm,
             mlp_layer: Type[nn.Module] = Mlp,
+            attn_layer = Attention,
+            depth: int = 0,
     ) -> None:
         super().__init__()
         self.init_values = init_values
 
-        self.attn = Attention(
+        self.attn = _create_attn(
+            attn_layer,
             dim,
             num_heads=num_heads,
             qkv_bias=qkv_bias,
@@ -174,6 +225,7 @@ class ResPostBlock(nn.Module):
             attn_drop=attn_drop,
             proj_drop=proj_drop,
             norm_layer=norm_layer,
+            depth=depth,
         )
         self.norm1 = norm_layer(dim)
         self.drop_path1 = DropPath(drop_path) if drop_path > 0. else nn.Identity()
@@ -321,6 +373,10 @@ class ParallelThingsBlock(nn.Module):
             act_layer: Type[nn.Module] = nn.GELU,
             norm_layer: Type[nn.Module] = LayerNorm,
             mlp_layer: Type[nn.Module] = Mlp,
+            attn_layer = Attention,
+            scale_attn_norm: bool = False,
+            scale_mlp_norm: bool = False,
+            depth: int = 0,
     ) -> None:
         super().__init__()
         self.num_parallel = num_parallel
@@ -329,15 +385,18 @@ class ParallelThingsBlock(nn.Module):
         for _ in range(num_parallel):
             self.attns.append(nn.Sequential(OrderedDict([
                 ('norm', norm_layer(dim)),
-                ('attn', Attention(
+                ('attn', _create_attn(
+                    attn_layer,
                     dim,
                     num_heads=num_heads,
                     qkv_bias=qkv_bias,
                     qk_norm=qk_norm,
+                    scale_norm=scale_attn_norm,
                     proj_bias=proj_bias,
                     attn_drop=attn_drop,
                     proj_drop=proj_drop,
                     norm_layer=norm_layer,
+                    depth=depth,
                 )),
                 ('ls', LayerScale(dim, init_values=init_values) if init_values else nn.Identity()),
                 ('drop_path', DropPath(drop_path) if drop_path > 0. else nn.Identity())
@@ -348,6 +407,7 @@ class ParallelThingsBlock(nn.Module):
                     dim,
                     hidden_features=int(dim * mlp_ratio),
                     act_layer=act_layer,
+                    norm_layer=norm_layer if scale_mlp_norm else None,
                     bias=proj_bias,
                     drop=proj_drop,
                 )),
@@ -416,6 +476,7 @@ class VisionTransformer(nn.Module):
             norm_layer: Optional[Union[str, Callable, Type[nn.Module]]] = None,
             act_layer: Optional[Union[str, Callable, Type[nn.Module]]] = None,
             block_fn: Type[nn.Module] = Block,
+            attn_layer: Union[str, Type[nn.Module], None] = None,
             mlp_layer: Type[nn.Module] = Mlp,
     ) -> None:
         super().__init__()
@@ -491,6 +552,7 @@ class VisionTransformer(nn.Module):
                 norm_layer=norm_layer,
                 act_layer=act_layer,
                 mlp_layer=mlp_layer,
+                **(dict(attn_layer=attn_layer, depth=i) if attn_layer is not None else {}),
             )
             for i in range(depth)])
         self.feature_info = [
