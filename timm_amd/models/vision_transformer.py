@@ -31,6 +31,7 @@ from ..data.constants import (
     IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD, IMAGENET_INCEPTION_MEAN, IMAGENET_INCEPTION_STD,
     OPENAI_CLIP_MEAN, OPENAI_CLIP_STD,
 )
+from ..layers.attention import maybe_add_mask
 from ..layers import (
     Attention, AttentionPoolLatent, DropPath, LayerNorm, LayerScale, Mlp, PatchDropout, PatchEmbed, RmsNorm,
     SwiGLUPacked, SwiGLU, get_act_layer, get_norm_layer, init_weight_jax, init_weight_vit, lecun_normal_,
@@ -350,6 +351,146 @@ class ParallelScalingBlock(nn.Module):
         # Add residual w/ drop path & layer scale applied
         y = self.drop_path(self.ls(x_attn + x_mlp))
         x = x + y
+        return x
+
+
+
+class DiffParallelScalingBlock(nn.Module):
+    """Parallel MLP+attention block (ViT-22B style fused in/out projections)
+    with differential attention (reference `vision_transformer.py:423`)."""
+    fused_attn: torch.jit.Final[bool]
+
+    def __init__(
+            self,
+            dim: int,
+            num_heads: int,
+            mlp_ratio: float = 4.,
+            qkv_bias: bool = False,
+            qk_norm: bool = False,
+            scale_attn_norm: bool = False,
+            scale_mlp_norm: bool = False,
+            proj_bias: bool = True,
+            proj_drop: float = 0.,
+            attn_drop: float = 0.,
+            init_values: Optional[float] = None,
+            drop_path: float = 0.,
+            act_layer: Type[nn.Module] = nn.GELU,
+            norm_layer: Type[nn.Module] = LayerNorm,
+            mlp_layer: Optional[Type[nn.Module]] = None,
+            attn_layer=None,
+            depth: int = 0,
+            dual_lambda: bool = False,
+    ) -> None:
+        super().__init__()
+        assert dim % num_heads == 0, 'dim should be divisible by num_heads'
+        assert not scale_attn_norm and not scale_mlp_norm, 'Scale norms not supported'
+        self.num_heads = num_heads
+        self.head_dim = dim // num_heads // 2  # halved for differential attention
+        self.scale = self.head_dim ** -0.5
+        self.fused_attn = use_fused_attn()
+        mlp_hidden_dim = int(mlp_ratio * dim)
+        in_proj_out_dim = mlp_hidden_dim + 3 * dim
+
+        self.in_norm = norm_layer(dim)
+        self.in_proj = nn.Linear(dim, in_proj_out_dim, bias=qkv_bias)
+        self.in_split = [mlp_hidden_dim] + [dim] * 3
+        if qkv_bias:
+            self.register_parameter('mlp_bias', None)
+        else:
+            self.mlp_bias = nn.Parameter(torch.zeros(mlp_hidden_dim))
+
+        self.q_norm = norm_layer(self.head_dim) if qk_norm else nn.Identity()
+        self.k_norm = norm_layer(self.head_dim) if qk_norm else nn.Identity()
+        self.attn_drop = nn.Dropout(attn_drop)
+        self.attn_drop_p = attn_drop
+
+        self.sub_norm = RmsNorm(2 * self.head_dim, eps=1e-5)
+        self.dual_lambda = dual_lambda
+        if dual_lambda:
+            self.lambda_a = nn.Parameter(torch.zeros((), dtype=torch.float32))
+            self.lambda_b = nn.Parameter(torch.zeros((), dtype=torch.float32))
+            self.lambda_q1 = self.lambda_k1 = self.lambda_q2 = self.lambda_k2 = None
+        else:
+            self.lambda_a = self.lambda_b = None
+            self.lambda_q1 = nn.Parameter(torch.empty(self.head_dim, dtype=torch.float32))
+            self.lambda_k1 = nn.Parameter(torch.empty(self.head_dim, dtype=torch.float32))
+            self.lambda_q2 = nn.Parameter(torch.empty(self.head_dim, dtype=torch.float32))
+            self.lambda_k2 = nn.Parameter(torch.empty(self.head_dim, dtype=torch.float32))
+
+        self.mlp_drop = nn.Dropout(proj_drop)
+        self.mlp_act = act_layer()
+        self.out_proj = nn.Linear(dim + mlp_hidden_dim, dim, bias=proj_bias)
+
+        self.ls = LayerScale(dim, init_values=init_values) if init_values is not None else nn.Identity()
+        self.drop_path = DropPath(drop_path) if drop_path > 0. else nn.Identity()
+
+        self.lambda_init = 0.8
+        self.set_lambda_init(depth)
+        self.reset_parameters()
+
+    def set_lambda_init(self, depth: int):
+        self.lambda_init = 0.8 - 0.6 * math.exp(-0.3 * depth)
+
+    def reset_parameters(self) -> None:
+        if self.mlp_bias is not None:
+            nn.init.zeros_(self.mlp_bias)
+        if not self.dual_lambda:
+            nn.init.normal_(self.lambda_q1, mean=0, std=0.1)
+            nn.init.normal_(self.lambda_k1, mean=0, std=0.1)
+            nn.init.normal_(self.lambda_q2, mean=0, std=0.1)
+            nn.init.normal_(self.lambda_k2, mean=0, std=0.1)
+
+    def _compute_lambda(self) -> torch.Tensor:
+        if self.lambda_a is not None:
+            lambda_1 = torch.exp(self.lambda_a)
+            lambda_2 = torch.exp(self.lambda_b)
+        else:
+            lambda_1 = torch.exp(torch.sum(self.lambda_q1 * self.lambda_k1, dim=-1).float())
+            lambda_2 = torch.exp(torch.sum(self.lambda_q2 * self.lambda_k2, dim=-1).float())
+        return lambda_1 - lambda_2 + self.lambda_init
+
+    def forward(self, x: torch.Tensor, attn_mask: Optional[torch.Tensor] = None) -> torch.Tensor:
+        B, N, C = x.shape
+
+        y = self.in_proj(self.in_norm(x))
+        x_mlp, q, k, v = torch.split(y, self.in_split, dim=-1)
+        if self.mlp_bias is not None:
+            x_mlp = x_mlp + self.mlp_bias
+
+        q = q.reshape(B, N, 2 * self.num_heads, self.head_dim).transpose(1, 2)
+        k = k.reshape(B, N, 2 * self.num_heads, self.head_dim).transpose(1, 2)
+        v = v.reshape(B, N, self.num_heads, 2 * self.head_dim).transpose(1, 2)
+        q, k = self.q_norm(q), self.k_norm(k)
+        lambda_full = self._compute_lambda().type_as(q)
+
+        if self.fused_attn:
+            q = q.reshape(B, self.num_heads, 2, N, self.head_dim)
+            k = k.reshape(B, self.num_heads, 2, N, self.head_dim)
+            q1, q2 = q.unbind(2)
+            k1, k2 = k.unbind(2)
+            dropout_p = self.attn_drop_p if self.training else 0.0
+            attn1 = ops.flash_attention(q1, k1, v, attn_mask=attn_mask, dropout_p=dropout_p)
+            attn2 = ops.flash_attention(q2, k2, v, attn_mask=attn_mask, dropout_p=dropout_p)
+            x_attn = attn1 - lambda_full * attn2
+        else:
+            q = q * self.scale
+            attn = q @ k.transpose(-2, -1)
+            attn = maybe_add_mask(attn, attn_mask)
+            attn = attn.softmax(dim=-1)
+            attn = self.attn_drop(attn)
+            attn = attn.view(B, self.num_heads, 2, N, N)
+            attn = attn[:, :, 0] - lambda_full * attn[:, :, 1]
+            x_attn = attn @ v
+
+        x_attn = self.sub_norm(x_attn)
+        x_attn = x_attn * (1 - self.lambda_init)
+        x_attn = x_attn.transpose(1, 2).reshape(B, N, C)
+
+        x_mlp = self.mlp_act(x_mlp)
+        x_mlp = self.mlp_drop(x_mlp)
+
+        y = self.out_proj(torch.cat((x_attn, x_mlp), dim=-1))
+        x = x + self.drop_path(self.ls(y))
         return x
 
 
