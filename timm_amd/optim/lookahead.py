@@ -1,45 +1,49 @@
-"""Lookahead optimizer wrapper (reference `timm/optim/lookahead.py`;
-paper https://arxiv.org/abs/1907.08610)."""
-from collections import OrderedDict
+"""Lookahead wrapper: k fast steps, then pull slow weights toward the fast
+ones (reference `timm/optim/lookahead.py`; paper arxiv 1907.08610).
+
+Slow-weight buffers live inside the wrapped optimizer's per-param state so a
+single state_dict round-trips both optimizers.
+"""
+from collections import OrderedDict, defaultdict
 from typing import Callable, Dict
 
 import torch
 from torch.optim.optimizer import Optimizer
-from collections import defaultdict
 
 
 class Lookahead(Optimizer):
     def __init__(self, base_optimizer, alpha=0.5, k=6):
-        # NOTE super().__init__() not called on purpose
-        self._optimizer_step_pre_hooks: Dict[int, Callable] = OrderedDict()
-        self._optimizer_step_post_hooks: Dict[int, Callable] = OrderedDict()
         if not 0.0 <= alpha <= 1.0:
             raise ValueError(f'Invalid slow update rate: {alpha}')
-        if not 1 <= k:
+        if k < 1:
             raise ValueError(f'Invalid lookahead steps: {k}')
-        defaults = dict(lookahead_alpha=alpha, lookahead_k=k, lookahead_step=0)
+        # Deliberately skip Optimizer.__init__: this wrapper aliases the base
+        # optimizer's groups/defaults instead of owning its own.
+        self._optimizer_step_pre_hooks: Dict[int, Callable] = OrderedDict()
+        self._optimizer_step_post_hooks: Dict[int, Callable] = OrderedDict()
+        extra = dict(lookahead_alpha=alpha, lookahead_k=k, lookahead_step=0)
         self._base_optimizer = base_optimizer
         self.param_groups = base_optimizer.param_groups
         self.defaults = base_optimizer.defaults
-        self.defaults.update(defaults)
+        self.defaults.update(extra)
         self.state = defaultdict(dict)
-        # manually add our defaults to the param groups
-        for name, default in defaults.items():
-            for group in self._base_optimizer.param_groups:
-                group.setdefault(name, default)
+        for group in self._base_optimizer.param_groups:
+            for name, value in extra.items():
+                group.setdefault(name, value)
 
     @torch.no_grad()
     def update_slow(self, group):
-        for fast_p in group["params"]:
-            if fast_p.grad is None:
+        alpha = group['lookahead_alpha']
+        for fast in group['params']:
+            if fast.grad is None:
                 continue
-            param_state = self._base_optimizer.state[fast_p]
-            if 'lookahead_slow_buff' not in param_state:
-                param_state['lookahead_slow_buff'] = torch.empty_like(fast_p)
-                param_state['lookahead_slow_buff'].copy_(fast_p)
-            slow = param_state['lookahead_slow_buff']
-            slow.add_(fast_p - slow, alpha=group['lookahead_alpha'])
-            fast_p.copy_(slow)
+            state = self._base_optimizer.state[fast]
+            slow = state.get('lookahead_slow_buff')
+            if slow is None:
+                slow = state['lookahead_slow_buff'] = torch.empty_like(fast)
+                slow.copy_(fast)
+            slow.add_(fast - slow, alpha=alpha)
+            fast.copy_(slow)
 
     def sync_lookahead(self):
         for group in self._base_optimizer.param_groups:
