@@ -1,4 +1,4 @@
-"""Tensor memory-format helpers (reference `timm/layers/format.py`)."""
+"""Tensor layout tags and conversions (reference `timm/layers/format.py`)."""
 from enum import Enum
 from typing import Union
 
@@ -14,46 +14,44 @@ class Format(str, Enum):
 
 FormatT = Union[str, Format]
 
+_SPATIAL_DIMS = {
+    Format.NLC: (1,),
+    Format.NCL: (2,),
+    Format.NHWC: (1, 2),
+    Format.NCHW: (2, 3),
+}
+
+_CHANNEL_DIM = {
+    Format.NHWC: 3,
+    Format.NLC: 2,
+    Format.NCL: 1,
+    Format.NCHW: 1,
+}
+
 
 def get_spatial_dim(fmt: FormatT):
-    fmt = Format(fmt)
-    if fmt is Format.NLC:
-        dim = (1,)
-    elif fmt is Format.NCL:
-        dim = (2,)
-    elif fmt is Format.NHWC:
-        dim = (1, 2)
-    else:
-        dim = (2, 3)
-    return dim
+    return _SPATIAL_DIMS[Format(fmt)]
 
 
 def get_channel_dim(fmt: FormatT):
-    fmt = Format(fmt)
-    if fmt is Format.NHWC:
-        dim = 3
-    elif fmt is Format.NLC:
-        dim = 2
-    else:
-        dim = 1
-    return dim
+    return _CHANNEL_DIM[Format(fmt)]
 
 
 def nchw_to(x: torch.Tensor, fmt: Format):
     if fmt == Format.NHWC:
-        x = x.permute(0, 2, 3, 1)
-    elif fmt == Format.NLC:
-        x = x.flatten(2).transpose(1, 2)
-    elif fmt == Format.NCL:
-        x = x.flatten(2)
+        return x.permute(0, 2, 3, 1)
+    if fmt == Format.NLC:
+        return x.flatten(2).transpose(1, 2)
+    if fmt == Format.NCL:
+        return x.flatten(2)
     return x
 
 
 def nhwc_to(x: torch.Tensor, fmt: Format):
     if fmt == Format.NCHW:
-        x = x.permute(0, 3, 1, 2)
-    elif fmt == Format.NLC:
-        x = x.flatten(1, 2)
-    elif fmt == Format.NCL:
-        x = x.flatten(1, 2).transpose(1, 2)
+        return x.permute(0, 3, 1, 2)
+    if fmt == Format.NLC:
+        return x.flatten(1, 2)
+    if fmt == Format.NCL:
+        return x.flatten(1, 2).transpose(1, 2)
     return x

@@ -1,4 +1,5 @@
-"""Model manipulation utils (reference `timm/models/_manipulate.py`, 346 LoC)."""
+"""Module-tree traversal, parameter grouping and checkpointing helpers
+(reference `timm/models/_manipulate.py`, 346 LoC)."""
 import collections.abc
 import math
 import re
@@ -8,7 +9,7 @@ from typing import Any, Callable, Dict, Iterator, Optional, Tuple, Type, Union
 
 import torch
 import torch.utils.checkpoint
-from torch import nn as nn
+from torch import nn
 
 __all__ = [
     'model_parameters', 'named_apply', 'named_modules', 'named_modules_with_params',
@@ -19,25 +20,31 @@ __all__ = [
 
 def model_parameters(model: nn.Module, exclude_head: bool = False):
     if exclude_head:
-        # FIXME this a bit of a quick and dirty hack to skip classifier head params based on ordering
-        return [p for p in model.parameters()][:-2]
+        # positional heuristic: the classifier's weight+bias are the last two params
+        return list(model.parameters())[:-2]
     return model.parameters()
+
+
+def _walk(module: nn.Module, name: str, depth_first: bool, include_root: bool):
+    """Shared pre/post-order traversal yielding (qualified name, module)."""
+    if include_root and not depth_first:
+        yield name, module
+    for child_name, child in module.named_children():
+        qualified = f'{name}.{child_name}' if name else child_name
+        yield from _walk(child, qualified, depth_first, True)
+    if include_root and depth_first:
+        yield name, module
 
 
 def named_apply(
         fn: Callable,
         module: nn.Module,
-        name='',
+        name: str = '',
         depth_first: bool = True,
         include_root: bool = False,
 ) -> nn.Module:
-    if not depth_first and include_root:
-        fn(module=module, name=name)
-    for child_name, child_module in module.named_children():
-        child_name = '.'.join((name, child_name)) if name else child_name
-        named_apply(fn=fn, module=child_module, name=child_name, depth_first=depth_first, include_root=True)
-    if depth_first and include_root:
-        fn(module=module, name=name)
+    for mod_name, mod in _walk(module, name, depth_first, include_root):
+        fn(module=mod, name=mod_name)
     return module
 
 
@@ -47,14 +54,7 @@ def named_modules(
         depth_first: bool = True,
         include_root: bool = False,
 ):
-    if not depth_first and include_root:
-        yield name, module
-    for child_name, child_module in module.named_children():
-        child_name = '.'.join((name, child_name)) if name else child_name
-        yield from named_modules(
-            module=child_module, name=child_name, depth_first=depth_first, include_root=True)
-    if depth_first and include_root:
-        yield name, module
+    yield from _walk(module, name, depth_first, include_root)
 
 
 def named_modules_with_params(
@@ -63,14 +63,9 @@ def named_modules_with_params(
         depth_first: bool = True,
         include_root: bool = False,
 ):
-    if module._parameters and not depth_first and include_root:
-        yield name, module
-    for child_name, child_module in module.named_children():
-        child_name = '.'.join((name, child_name)) if name else child_name
-        yield from named_modules_with_params(
-            module=child_module, name=child_name, depth_first=depth_first, include_root=True)
-    if module._parameters and depth_first and include_root:
-        yield name, module
+    for mod_name, mod in _walk(module, name, depth_first, include_root):
+        if mod._parameters:
+            yield mod_name, mod
 
 
 MATCH_PREV_GROUP = (99999,)
@@ -82,80 +77,65 @@ def group_with_matcher(
         return_values: bool = False,
         reverse: bool = False,
 ):
-    """Group parameters/modules by a regex matcher spec (reference `_manipulate.py:80`).
+    """Bucket named params/modules into ordered layer groups (reference
+    `_manipulate.py:80`); drives layer-wise LR decay in
+    `optim/_param_groups.py`.
 
-    Drives layer-wise LR decay grouping (`optim/_param_groups.py`)."""
+    A dict matcher maps group name -> regex (or list of (regex, suffix)
+    pairs); regex capture groups become sort ordinals. A callable matcher
+    returns the ordinal(s) directly. Names nothing matches sort last
+    (typically neck/head). A suffix of MATCH_PREV_GROUP merges the bucket
+    into the preceding layer id.
+    """
     if isinstance(group_matcher, dict):
-        # dictionary matcher contains a dict of raw-string regex expr that must be compiled
         compiled = []
-        for group_ordinal, (group_name, mspec) in enumerate(group_matcher.items()):
-            if mspec is None:
+        for ordinal, (_, spec) in enumerate(group_matcher.items()):
+            if spec is None:
                 continue
-            # map all matching specifications into 3-tuple (compiled re, prefix, suffix)
-            if isinstance(mspec, (tuple, list)):
-                # multi-entry match specifications require each sub-spec to be a 2-tuple (re, suffix)
-                for sspec in mspec:
-                    compiled += [(re.compile(sspec[0]), (group_ordinal,), sspec[1])]
+            if isinstance(spec, (tuple, list)):
+                # each sub-spec is (regex, suffix ordinal tuple)
+                compiled += [(re.compile(p), (ordinal,), s) for p, s in spec]
             else:
-                compiled += [(re.compile(mspec), (group_ordinal,), None)]
+                compiled += [(re.compile(spec), (ordinal,), None)]
         group_matcher = compiled
 
-    def _get_grouping(name):
+    def ordinal_of(name):
         if isinstance(group_matcher, (list, tuple)):
-            for match_fn, prefix, suffix in group_matcher:
-                r = match_fn.match(name)
-                if r:
-                    parts = (prefix, r.groups(), suffix)
-                    # map all tuple elem to int for numeric sort, filter out None entries
+            for regex, prefix, suffix in group_matcher:
+                m = regex.match(name)
+                if m:
+                    parts = (prefix, m.groups(), suffix)
                     return tuple(map(float, chain.from_iterable(filter(None, parts))))
-            return float('inf'),  # un-matched layers (neck, head) mapped to largest ordinal
-        else:
-            ord = group_matcher(name)
-            if not isinstance(ord, collections.abc.Iterable):
-                return ord,
-            return tuple(ord)
+            return (float('inf'),)  # unmatched -> last group
+        ordinal = group_matcher(name)
+        if not isinstance(ordinal, collections.abc.Iterable):
+            return (ordinal,)
+        return tuple(ordinal)
 
-    # map layers into groups via ordinals (ints or tuples of ints) from matcher
-    grouping = defaultdict(list)
-    for k, v in named_objects:
-        grouping[_get_grouping(k)].append(v if return_values else k)
+    buckets = defaultdict(list)
+    for name, value in named_objects:
+        buckets[ordinal_of(name)].append(value if return_values else name)
 
-    # remap to integers
-    layer_id_to_param = defaultdict(list)
-    lid = -1
-    for k in sorted(filter(lambda x: x is not None, grouping.keys())):
-        if lid < 0 or k[-1] != MATCH_PREV_GROUP[0]:
-            lid += 1
-        layer_id_to_param[lid].extend(grouping[k])
+    # collapse ordinal tuples into consecutive integer layer ids
+    layer_to_members = defaultdict(list)
+    layer_id = -1
+    for key in sorted(k for k in buckets.keys() if k is not None):
+        if layer_id < 0 or key[-1] != MATCH_PREV_GROUP[0]:
+            layer_id += 1
+        layer_to_members[layer_id].extend(buckets[key])
 
     if reverse:
-        assert not return_values, "reverse mapping only sensible for name output"
-        # output reverse mapping
-        param_to_layer_id = {}
-        for lid, lm in layer_id_to_param.items():
-            for n in lm:
-                param_to_layer_id[n] = lid
-        return param_to_layer_id
-
-    return layer_id_to_param
+        assert not return_values, 'reverse mapping only sensible for name output'
+        return {name: lid for lid, names in layer_to_members.items() for name in names}
+    return layer_to_members
 
 
-def group_parameters(
-        module: nn.Module,
-        group_matcher,
-        return_values: bool = False,
-        reverse: bool = False,
-):
+def group_parameters(module: nn.Module, group_matcher, return_values: bool = False, reverse: bool = False):
     return group_with_matcher(
         module.named_parameters(), group_matcher, return_values=return_values, reverse=reverse)
 
 
-def group_modules(
-        module: nn.Module,
-        group_matcher,
-        return_values: bool = False,
-        reverse: bool = False,
-):
+def group_modules(module: nn.Module, group_matcher, return_values: bool = False, reverse: bool = False):
     return group_with_matcher(
         named_modules_with_params(module), group_matcher, return_values=return_values, reverse=reverse)
 
@@ -166,7 +146,8 @@ def flatten_modules(
         prefix: Union[str, Tuple[str, ...]] = '',
         module_types: Union[str, Tuple[Type[nn.Module]]] = 'sequential',
 ):
-    prefix_is_tuple = isinstance(prefix, tuple)
+    """Expand container modules up to `depth` levels, yielding leaf entries."""
+    tuple_prefix = isinstance(prefix, tuple)
     if isinstance(module_types, str):
         if module_types == 'container':
             module_types = (nn.Sequential, nn.ModuleList, nn.ModuleDict)
@@ -177,34 +158,21 @@ def flatten_modules(
             yield from flatten_modules(
                 module.named_children(),
                 depth - 1,
-                prefix=(name,) if prefix_is_tuple else name,
+                prefix=(name,) if tuple_prefix else name,
                 module_types=module_types,
             )
+        elif tuple_prefix:
+            yield prefix + (name,), module
         else:
-            if prefix_is_tuple:
-                name = prefix + (name,)
-                yield name, module
-            else:
-                if prefix:
-                    name = '.'.join([prefix, name])
-                yield name, module
+            yield (f'{prefix}.{name}' if prefix else name), module
 
 
-def checkpoint(
-        function,
-        *args,
-        use_reentrant: Optional[bool] = None,
-        **kwargs,
-):
-    """checkpoint wrapper fn providing default for `use_reentrant` (reference `:191`)."""
+def checkpoint(function, *args, use_reentrant: Optional[bool] = None, **kwargs):
+    """torch checkpoint with the session-wide reentrant default (reference `:191`)."""
     if use_reentrant is None:
         from ..layers.config import use_reentrant_ckpt
         use_reentrant = use_reentrant_ckpt()
-    return torch.utils.checkpoint.checkpoint(
-        function, *args,
-        use_reentrant=use_reentrant,
-        **kwargs,
-    )
+    return torch.utils.checkpoint.checkpoint(function, *args, use_reentrant=use_reentrant, **kwargs)
 
 
 def checkpoint_seq(
@@ -215,15 +183,16 @@ def checkpoint_seq(
         skip_last: bool = False,
         use_reentrant: Optional[bool] = None,
 ):
-    """Checkpoint a sequential model in segments (reference `_manipulate.py:213-287`)."""
+    """Activation-checkpoint a module sequence in segments of `every`
+    (reference `_manipulate.py:213-287`)."""
     if use_reentrant is None:
         from ..layers.config import use_reentrant_ckpt
         use_reentrant = use_reentrant_ckpt()
 
-    def run_function(start, end, functions):
+    def segment(start, end):
         def forward(_x):
-            for j in range(start, end + 1):
-                _x = functions[j](_x)
+            for idx in range(start, end + 1):
+                _x = functions[idx](_x)
             return _x
         return forward
 
@@ -234,43 +203,34 @@ def checkpoint_seq(
     if not isinstance(functions, (tuple, list)):
         functions = tuple(functions)
 
-    num_checkpointed = len(functions)
-    if skip_last:
-        num_checkpointed -= 1
+    num_checkpointed = len(functions) - 1 if skip_last else len(functions)
     end = -1
     for start in range(0, num_checkpointed, every):
         end = min(start + every - 1, num_checkpointed - 1)
-        x = torch.utils.checkpoint.checkpoint(
-            run_function(start, end, functions), x,
-            use_reentrant=use_reentrant,
-        )
+        x = torch.utils.checkpoint.checkpoint(segment(start, end), x, use_reentrant=use_reentrant)
     if skip_last:
-        return run_function(end + 1, len(functions) - 1, functions)(x)
+        return segment(end + 1, len(functions) - 1)(x)
     return x
 
 
 def adapt_input_conv(in_chans: int, conv_weight: torch.Tensor) -> torch.Tensor:
-    """Adapt first-conv weights for a different number of input channels
-    (reference `_manipulate.py:289`)."""
+    """Reshape a pretrained stem conv for a different input channel count
+    (reference `_manipulate.py:289`): grayscale sums the RGB filters,
+    in_chans > 3 tiles and rescales them."""
     conv_type = conv_weight.dtype
-    conv_weight = conv_weight.float()  # Some weights are in torch.half, ensure it's float for sum on CPU
+    conv_weight = conv_weight.float()  # fp16 weights can't .sum() on CPU
     O, I, J, K = conv_weight.shape
     if in_chans == 1:
         if I > 3:
+            # space2depth stems carry I = 3 * s2d_factor
             assert conv_weight.shape[1] % 3 == 0
-            # For models with space2depth stems
-            conv_weight = conv_weight.reshape(O, I // 3, 3, J, K)
-            conv_weight = conv_weight.sum(dim=2, keepdim=False)
+            conv_weight = conv_weight.reshape(O, I // 3, 3, J, K).sum(dim=2)
         else:
             conv_weight = conv_weight.sum(dim=1, keepdim=True)
     elif in_chans != 3:
         if I != 3:
             raise NotImplementedError('Weight format not supported by conversion.')
-        else:
-            # NOTE this strategy should be better than random init, but there could be other combinations of
-            # the original RGB input layer weights that'd work better for specific cases.
-            repeat = int(math.ceil(in_chans / 3))
-            conv_weight = conv_weight.repeat(1, repeat, 1, 1)[:, :in_chans, :, :]
-            conv_weight *= (3 / float(in_chans))
-    conv_weight = conv_weight.to(conv_type)
-    return conv_weight
+        repeat = int(math.ceil(in_chans / 3))
+        conv_weight = conv_weight.repeat(1, repeat, 1, 1)[:, :in_chans, :, :]
+        conv_weight *= 3 / float(in_chans)
+    return conv_weight.to(conv_type)

@@ -1,11 +1,16 @@
-"""LARS / LARC optimizer (reference `timm/optim/lars.py:17`)."""
+"""LARS / LARC (reference `timm/optim/lars.py:17`; papers arxiv 1708.03888 /
+LARC from NVIDIA apex).
+
+SGD with a per-tensor trust ratio: the effective step is scaled by
+`trust_coeff * ||w|| / (||g|| + wd * ||w||)`, optionally clipped at 1 relative
+to the base LR (LARC). Large-batch training keeps layer updates proportional
+to layer magnitude.
+"""
 import torch
 from torch.optim.optimizer import Optimizer
 
 
 class Lars(Optimizer):
-    """LARS for SGD w/ layer-wise adaptive rate scaling + optional clipping (LARC)."""
-
     def __init__(
             self,
             params,
@@ -20,15 +25,14 @@ class Lars(Optimizer):
             always_adapt=False,
     ):
         if lr < 0.0:
-            raise ValueError(f"Invalid learning rate: {lr}")
+            raise ValueError(f'Invalid learning rate: {lr}')
         if momentum < 0.0:
-            raise ValueError(f"Invalid momentum value: {momentum}")
+            raise ValueError(f'Invalid momentum value: {momentum}')
         if weight_decay < 0.0:
-            raise ValueError(f"Invalid weight_decay value: {weight_decay}")
+            raise ValueError(f'Invalid weight_decay value: {weight_decay}')
         if nesterov and (momentum <= 0 or dampening != 0):
-            raise ValueError("Nesterov momentum requires a momentum and zero dampening")
-
-        defaults = dict(
+            raise ValueError('Nesterov momentum requires a momentum and zero dampening')
+        super().__init__(params, dict(
             lr=lr,
             momentum=momentum,
             dampening=dampening,
@@ -38,13 +42,12 @@ class Lars(Optimizer):
             eps=eps,
             trust_clip=trust_clip,
             always_adapt=always_adapt,
-        )
-        super().__init__(params, defaults)
+        ))
 
     def __setstate__(self, state):
         super().__setstate__(state)
         for group in self.param_groups:
-            group.setdefault("nesterov", False)
+            group.setdefault('nesterov', False)
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -54,48 +57,39 @@ class Lars(Optimizer):
                 loss = closure()
 
         for group in self.param_groups:
-            weight_decay = group['weight_decay']
+            wd = group['weight_decay']
             momentum = group['momentum']
             dampening = group['dampening']
-            nesterov = group['nesterov']
-            trust_coeff = group['trust_coeff']
-            eps = group['eps']
 
             for p in group['params']:
                 if p.grad is None:
                     continue
                 grad = p.grad
 
-                # apply LARS LR adaptation, LARC clipping, weight decay
-                # ref: https://github.com/NVIDIA/apex/blob/master/apex/parallel/LARC.py
-                if weight_decay != 0 or group['always_adapt']:
+                if wd != 0 or group['always_adapt']:
                     w_norm = p.norm(2.0)
                     g_norm = grad.norm(2.0)
-                    trust_ratio = trust_coeff * w_norm / (g_norm + w_norm * weight_decay + eps)
-                    # FIXME nested where required since logical and/or not working in PT XLA
-                    trust_ratio = torch.where(
+                    ratio = group['trust_coeff'] * w_norm / (g_norm + w_norm * wd + group['eps'])
+                    # nested where: either norm being zero disables adaptation
+                    ratio = torch.where(
                         w_norm > 0,
-                        torch.where(g_norm > 0, trust_ratio, 1.0),
+                        torch.where(g_norm > 0, ratio, 1.0),
                         1.0,
                     )
                     if group['trust_clip']:
-                        trust_ratio = torch.clamp(trust_ratio / group['lr'], max=1.0)
-                    grad.add_(p, alpha=weight_decay)
-                    grad.mul_(trust_ratio)
+                        # LARC: cap the adapted LR at the base LR
+                        ratio = torch.clamp(ratio / group['lr'], max=1.0)
+                    grad.add_(p, alpha=wd)
+                    grad.mul_(ratio)
 
-                # apply SGD update https://github.com/pytorch/pytorch/blob/1.7/torch/optim/sgd.py#L100
                 if momentum != 0:
                     state = self.state[p]
-                    buf = state.get('momentum_buffer', None)
+                    buf = state.get('momentum_buffer')
                     if buf is None:
-                        buf = torch.clone(grad).detach()
-                        state['momentum_buffer'] = buf
+                        state['momentum_buffer'] = buf = torch.clone(grad).detach()
                     else:
                         buf.mul_(momentum).add_(grad, alpha=1. - dampening)
-                    if nesterov:
-                        grad = grad.add(buf, alpha=momentum)
-                    else:
-                        grad = buf
+                    grad = grad.add(buf, alpha=momentum) if group['nesterov'] else buf
 
                 p.add_(grad, alpha=-group['lr'])
 

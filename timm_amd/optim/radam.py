@@ -1,5 +1,11 @@
-"""RAdam optimizer (reference `timm/optim/radam.py`;
-paper https://arxiv.org/abs/1908.03265)."""
+"""RAdam — Adam with rectified adaptive learning rate warmup (reference
+`timm/optim/radam.py`; paper arxiv 1908.03265).
+
+While the variance estimate is still unreliable (SMA length < 5) the update
+falls back to plain momentum SGD; once it stabilizes, a rectification factor
+scales the adaptive step. Per-step coefficients are cached in a 10-slot ring
+buffer shared by all params at the same step count.
+"""
 import math
 
 import torch
@@ -16,15 +22,35 @@ class RAdam(Optimizer):
             weight_decay=0,
             caution=False,
     ):
-        defaults = dict(
+        super().__init__(params, dict(
             lr=lr, betas=betas, eps=eps, weight_decay=weight_decay, caution=caution,
-            buffer=[[None, None, None] for _ in range(10)])
-        super().__init__(params, defaults)
+            buffer=[[None, None, None] for _ in range(10)]))
 
     def __setstate__(self, state):
         super().__setstate__(state)
         for group in self.param_groups:
             group.setdefault('caution', False)
+
+    def _step_coeffs(self, group, step):
+        """(num_sma, step_size) for this step, memoized in the ring buffer."""
+        slot = group['buffer'][step % 10]
+        if slot[0] == step:
+            return slot[1], slot[2]
+        beta1, beta2 = group['betas']
+        beta2_t = beta2 ** step
+        sma_max = 2 / (1 - beta2) - 1
+        num_sma = sma_max - 2 * step * beta2_t / (1 - beta2_t)
+        if num_sma >= 5:
+            rect = math.sqrt(
+                (1 - beta2_t)
+                * (num_sma - 4) / (sma_max - 4)
+                * (num_sma - 2) / num_sma
+                * sma_max / (sma_max - 2))
+            step_size = group['lr'] * rect / (1 - beta1 ** step)
+        else:
+            step_size = group['lr'] / (1 - beta1 ** step)
+        slot[0], slot[1], slot[2] = step, num_sma, step_size
+        return num_sma, step_size
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -34,13 +60,13 @@ class RAdam(Optimizer):
                 loss = closure()
 
         for group in self.param_groups:
+            beta1, beta2 = group['betas']
             for p in group['params']:
                 if p.grad is None:
                     continue
                 grad = p.grad.float()
                 if grad.is_sparse:
                     raise RuntimeError('RAdam does not support sparse gradients')
-
                 p_fp32 = p.float()
 
                 state = self.state[p]
@@ -51,42 +77,21 @@ class RAdam(Optimizer):
                 else:
                     state['exp_avg'] = state['exp_avg'].type_as(p_fp32)
                     state['exp_avg_sq'] = state['exp_avg_sq'].type_as(p_fp32)
-
                 exp_avg, exp_avg_sq = state['exp_avg'], state['exp_avg_sq']
-                beta1, beta2 = group['betas']
 
                 exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
                 exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
-
                 state['step'] += 1
-                buffered = group['buffer'][int(state['step'] % 10)]
-                if state['step'] == buffered[0]:
-                    num_sma, step_size = buffered[1], buffered[2]
-                else:
-                    buffered[0] = state['step']
-                    beta2_t = beta2 ** state['step']
-                    num_sma_max = 2 / (1 - beta2) - 1
-                    num_sma = num_sma_max - 2 * state['step'] * beta2_t / (1 - beta2_t)
-                    buffered[1] = num_sma
 
-                    # more conservative since it's an approximated value
-                    if num_sma >= 5:
-                        step_size = group['lr'] * math.sqrt(
-                            (1 - beta2_t) *
-                            (num_sma - 4) / (num_sma_max - 4) *
-                            (num_sma - 2) / num_sma *
-                            num_sma_max / (num_sma_max - 2)) / (1 - beta1 ** state['step'])
-                    else:
-                        step_size = group['lr'] / (1 - beta1 ** state['step'])
-                    buffered[2] = step_size
+                num_sma, step_size = self._step_coeffs(group, state['step'])
 
                 if group['weight_decay'] != 0:
                     p_fp32.add_(p_fp32, alpha=-group['weight_decay'] * group['lr'])
 
-                # more conservative since it's an approximated value
                 if num_sma >= 5:
                     denom = exp_avg_sq.sqrt().add_(group['eps'])
                     if group['caution']:
+                        # zero components whose sign disagrees with the raw grad
                         update = exp_avg / denom
                         mask = (update * grad > 0).to(grad.dtype)
                         mask.div_(mask.mean().clamp_(min=1e-3))

@@ -1,13 +1,14 @@
-"""Adaptive Gradient Clipping (NFNet) — reference `timm/utils/agc.py:30`."""
+"""Adaptive gradient clipping — clip each unit's gradient relative to its
+parameter norm (NFNets, arxiv 2102.06171; reference `timm/utils/agc.py:30`)."""
 import torch
 
 
 def unitwise_norm(x, norm_type=2.0):
+    """Per-output-unit norm: scalar for vectors/biases, per-row (first dim)
+    for Conv/Linear kernels."""
     if x.ndim <= 1:
         return x.norm(norm_type)
-    else:
-        # works for nn.ConvNd and nn.Linear where output dim is first in the kernel/weight tensor
-        return x.norm(norm_type, dim=tuple(range(1, x.ndim)), keepdim=True)
+    return x.norm(norm_type, dim=tuple(range(1, x.ndim)), keepdim=True)
 
 
 def adaptive_clip_grad(parameters, clip_factor=0.01, eps=1e-3, norm_type=2.0):
@@ -16,10 +17,9 @@ def adaptive_clip_grad(parameters, clip_factor=0.01, eps=1e-3, norm_type=2.0):
     for p in parameters:
         if p.grad is None:
             continue
-        p_data = p.detach()
-        g_data = p.grad.detach()
-        max_norm = unitwise_norm(p_data, norm_type=norm_type).clamp_(min=eps).mul_(clip_factor)
-        grad_norm = unitwise_norm(g_data, norm_type=norm_type)
-        clipped_grad = g_data * (max_norm / grad_norm.clamp(min=1e-6))
-        new_grads = torch.where(grad_norm < max_norm, g_data, clipped_grad)
-        p.grad.detach().copy_(new_grads)
+        w = p.detach()
+        g = p.grad.detach()
+        limit = unitwise_norm(w, norm_type=norm_type).clamp_(min=eps).mul_(clip_factor)
+        g_norm = unitwise_norm(g, norm_type=norm_type)
+        scaled = g * (limit / g_norm.clamp(min=1e-6))
+        p.grad.detach().copy_(torch.where(g_norm < limit, g, scaled))

@@ -1,11 +1,20 @@
-"""JIT fuser selection (reference `timm/utils/jit.py:23`)."""
-import os
+"""torch.jit fuser selection (reference `timm/utils/jit.py:23`).
 
+On ROCm the TE fuser is the only relevant scripted-graph fuser; nvfuser is a
+CUDA-only backend and is always disabled.
+"""
 import torch
 
 
+def _disable_nvfuser():
+    try:
+        torch._C._jit_set_nvfuser_enabled(False)
+    except Exception:
+        pass
+
+
 def set_jit_legacy():
-    """Set JIT executor to legacy w/ support for op fusion."""
+    """Legacy JIT executor with GPU op fusion."""
     assert hasattr(torch._C, '_jit_set_profiling_executor'), "Old JIT behavior doesn't exist!"
     torch._C._jit_set_profiling_executor(False)
     torch._C._jit_set_profiling_mode(False)
@@ -13,35 +22,25 @@ def set_jit_legacy():
 
 
 def set_jit_fuser(fuser):
-    if fuser == "te":
-        # default fuser should be == 'te'
+    if fuser == 'te':
         torch._C._jit_set_profiling_executor(True)
         torch._C._jit_set_profiling_mode(True)
         torch._C._jit_override_can_fuse_on_cpu(False)
         torch._C._jit_override_can_fuse_on_gpu(True)
         torch._C._jit_set_texpr_fuser_enabled(True)
-        try:
-            torch._C._jit_set_nvfuser_enabled(False)
-        except Exception:
-            pass
-    elif fuser == "old" or fuser == "legacy":
+        _disable_nvfuser()
+    elif fuser in ('old', 'legacy'):
         torch._C._jit_set_profiling_executor(False)
         torch._C._jit_set_profiling_mode(False)
         torch._C._jit_override_can_fuse_on_gpu(True)
         torch._C._jit_set_texpr_fuser_enabled(False)
-        try:
-            torch._C._jit_set_nvfuser_enabled(False)
-        except Exception:
-            pass
-    elif fuser == "none" or not fuser:
+        _disable_nvfuser()
+    elif fuser in ('none', '', None):
         torch._C._jit_set_profiling_executor(True)
         torch._C._jit_set_profiling_mode(True)
         torch._C._jit_override_can_fuse_on_cpu(False)
         torch._C._jit_override_can_fuse_on_gpu(False)
         torch._C._jit_set_texpr_fuser_enabled(False)
-        try:
-            torch._C._jit_set_nvfuser_enabled(False)
-        except Exception:
-            pass
+        _disable_nvfuser()
     else:
-        assert False, f"Invalid jit fuser ({fuser})"
+        raise AssertionError(f'Invalid jit fuser ({fuser})')
