@@ -269,3 +269,46 @@ def test_train_script_scheduled_res(tmp_path):
     ], capture_output=True, text=True, timeout=600,
         cwd=os.path.dirname(os.path.dirname(__file__)))
     assert r.returncode == 0, r.stdout[-1200:] + r.stderr[-1200:]
+
+
+def _spawn_n(fn, port, world):
+    ctx = mp.get_context('spawn')
+    q = ctx.Queue()
+    procs = [ctx.Process(target=fn, args=(r, world, port, q)) for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=300) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+    assert all(ok for _, ok in results), results
+
+
+def test_bucketed_ddp_grad_allreduce_world4():
+    """Same grad-averaging check at world_size=4 (multi-bucket, gloo)."""
+    _spawn_n(_run_ddp_grads, 29617, 4)
+
+
+def _run_distribute_bn(rank, world_size, port, q):
+    _init(rank, world_size, port)
+    from timm_amd.utils.distributed import distribute_bn
+    model = torch.nn.Sequential(torch.nn.Conv2d(3, 4, 3), torch.nn.BatchNorm2d(4))
+    with torch.no_grad():
+        model[1].running_mean.fill_(float(rank))
+        model[1].running_var.fill_(float(rank + 1))
+    distribute_bn(model, world_size, reduce=True)
+    expect_mean = sum(range(world_size)) / world_size
+    expect_var = sum(range(1, world_size + 1)) / world_size
+    ok = torch.allclose(model[1].running_mean, torch.full((4,), expect_mean)) and \
+        torch.allclose(model[1].running_var, torch.full((4,), expect_var))
+    # broadcast mode: everyone gets rank 0 stats
+    with torch.no_grad():
+        model[1].running_mean.fill_(float(rank))
+    distribute_bn(model, world_size, reduce=False)
+    ok = ok and torch.allclose(model[1].running_mean, torch.zeros(4))
+    q.put((rank, ok))
+    dist.destroy_process_group()
+
+
+def test_distribute_bn_world4():
+    """distribute_bn reduce + broadcast across 4 ranks."""
+    _spawn_n(_run_distribute_bn, 29619, 4)

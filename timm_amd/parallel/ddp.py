@@ -92,6 +92,16 @@ class BucketedDataParallel(nn.Module):
             buckets.append(cur)
 
         self._buckets = [_Bucket(b) for b in buckets]
+
+        # Dedicated comm stream (GPU backends): each ready bucket's all-reduce
+        # AND its 1/world scale run on this stream, overlapping with the rest
+        # of backward; finish_gradient_sync() then only inserts a device-side
+        # wait on the default stream instead of serializing the scales there.
+        self._comm_stream = None
+        self._bucket_events = None
+        if torch.cuda.is_available() and self._buckets and self._buckets[0].flat.is_cuda:
+            self._comm_stream = torch.cuda.Stream()
+            self._bucket_events = [torch.cuda.Event() for _ in self._buckets]
         self._param_bucket: Dict[nn.Parameter, tuple] = {}
         for bi, bucket in enumerate(self._buckets):
             for pi, p in enumerate(bucket.params):
@@ -109,8 +119,20 @@ class BucketedDataParallel(nn.Module):
             bucket.ready_count += 1
             if bucket.ready_count == len(bucket.params):
                 bucket.ready_count = 0
-                bucket.work = dist.all_reduce(
-                    bucket.flat, op=dist.ReduceOp.SUM, group=self.process_group, async_op=True)
+                if self._comm_stream is not None:
+                    # hand the bucket to the comm stream: reduce + scale there
+                    ready = torch.cuda.Event()
+                    ready.record()  # grads for this bucket are complete on the current stream
+                    with torch.cuda.stream(self._comm_stream):
+                        ready.wait()
+                        dist.all_reduce(
+                            bucket.flat, op=dist.ReduceOp.SUM, group=self.process_group)
+                        bucket.flat.mul_(1.0 / self.world_size)
+                        self._bucket_events[bucket_idx].record()
+                    bucket.work = True
+                else:
+                    bucket.work = dist.all_reduce(
+                        bucket.flat, op=dist.ReduceOp.SUM, group=self.process_group, async_op=True)
         return hook
 
     def forward(self, *args, **kwargs):
@@ -134,11 +156,16 @@ class BucketedDataParallel(nn.Module):
         if not self._require_sync:
             return
         inv = 1.0 / self.world_size
-        for bucket in self._buckets:
+        for bi, bucket in enumerate(self._buckets):
             if bucket.work is not None:
-                bucket.work.wait()
+                if self._comm_stream is not None:
+                    # device-side wait: optimizer kernels on the default stream
+                    # order after the comm stream's reduce+scale, host never blocks
+                    self._bucket_events[bi].wait()
+                else:
+                    bucket.work.wait()
+                    bucket.flat.mul_(inv)
                 bucket.work = None
-                bucket.flat.mul_(inv)
             elif bucket.ready_count:
                 # A partially-ready bucket means some params received no grad
                 # this backward (unused branch / aux head).  Silently skipping
