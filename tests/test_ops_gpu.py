@@ -729,3 +729,49 @@ def test_model_train_bn_stats_gpu():
     rm0 = m.bn1.running_mean.clone()
     m(torch.randn(4, 3, 160, 160, device='cuda'))
     assert not torch.allclose(m.bn1.running_mean, rm0)
+
+
+def test_fused_lamb_matches_reference():
+    """Fused two-launch LAMB step vs the composable fp32 loop."""
+    _ext()
+    import copy, math
+    from timm_amd.optim.lamb import Lamb
+    torch.manual_seed(3)
+    model = torch.nn.Sequential(
+        torch.nn.Linear(64, 128), torch.nn.GELU(), torch.nn.Linear(128, 32),
+        torch.nn.LayerNorm(32))
+    model_ref = copy.deepcopy(model)
+
+    model.cuda()
+    x = torch.randn(16, 64)
+    for mdl, dev in ((model, 'cuda'), (model_ref, 'cpu')):
+        opt = Lamb(mdl.parameters(), lr=1e-2, weight_decay=0.02, trust_clip=True)
+        for _ in range(3):
+            opt.zero_grad()
+            mdl(x.to(dev)).pow(2).mean().backward()
+            opt.step()
+
+    for (n, p_gpu), (_, p_ref) in zip(model.named_parameters(), model_ref.named_parameters()):
+        err = (p_gpu.cpu() - p_ref).abs().max().item()
+        assert err < 1e-4, f'{n}: {err}'
+
+
+def test_fused_lamb_bf16_and_no_decay_group():
+    """bf16 params + a zero-decay group (no trust adaptation)."""
+    _ext()
+    from timm_amd.optim.lamb import Lamb
+    torch.manual_seed(4)
+    w_decay = torch.nn.Parameter(torch.randn(256, 64, device='cuda', dtype=torch.bfloat16))
+    w_plain = torch.nn.Parameter(torch.randn(64, device='cuda', dtype=torch.bfloat16))
+    opt = Lamb([
+        {'params': [w_decay], 'weight_decay': 0.05},
+        {'params': [w_plain], 'weight_decay': 0.0},
+    ], lr=1e-2)
+    before = (w_decay.detach().clone(), w_plain.detach().clone())
+    for _ in range(2):
+        opt.zero_grad()
+        (w_decay.float().pow(2).sum() + w_plain.float().pow(2).sum()).backward()
+        opt.step()
+    assert not torch.equal(w_decay.detach(), before[0])
+    assert not torch.equal(w_plain.detach(), before[1])
+    assert torch.isfinite(w_decay.float()).all() and torch.isfinite(w_plain.float()).all()

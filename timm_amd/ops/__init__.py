@@ -71,7 +71,7 @@ from .conv import depthwise_conv2d  # noqa: E402
 from .layer_norm import layer_norm_act, rms_norm_act, layer_norm, rms_norm  # noqa: E402
 from .attention import flash_attention, flash_attention_qkv, attention_available  # noqa: E402
 from .elementwise import bias_act, residual_scale_add  # noqa: E402
-from .fused_optim import fused_adamw_step, fused_lerp_, fused_l2norm  # noqa: E402
+from .fused_optim import fused_adamw_step, fused_lamb_step, fused_lerp_, fused_l2norm  # noqa: E402
 from .loss import fused_cross_entropy  # noqa: E402
 from .data import u8_normalize, masked_global_pool  # noqa: E402
 
@@ -81,6 +81,6 @@ __all__ = [
     'layer_norm', 'layer_norm_act', 'rms_norm', 'rms_norm_act',
     'flash_attention', 'flash_attention_qkv', 'attention_available',
     'bias_act', 'residual_scale_add',
-    'fused_adamw_step', 'fused_lerp_', 'fused_l2norm',
+    'fused_adamw_step', 'fused_lamb_step', 'fused_lerp_', 'fused_l2norm',
     'fused_cross_entropy', 'u8_normalize', 'masked_global_pool',
 ]

@@ -65,6 +65,11 @@ void multi_tensor_adamw(
     std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
     double lr, double beta1, double beta2, double eps, double wd, double bc1, double bc2);
 void multi_tensor_lerp(std::vector<at::Tensor> dsts, std::vector<at::Tensor> srcs, double weight);
+void multi_tensor_lamb(
+    std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+    double lr, double beta1, double beta2, double beta3, double eps, double wd,
+    double bc1, double bc2, double clip_norm, bool adapt, bool trust_clip);
 at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -93,5 +98,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("masked_pool_bwd", &masked_pool_bwd, "masked global pool bwd (NaFlex)");
   m.def("multi_tensor_adamw", &multi_tensor_adamw, "fused multi-tensor AdamW step");
   m.def("multi_tensor_lerp", &multi_tensor_lerp, "fused multi-tensor lerp (EMA)");
+  m.def("multi_tensor_lamb", &multi_tensor_lamb, "fused multi-tensor LAMB step (trust-ratio, gfx950)");
   m.def("multi_tensor_l2norm", &multi_tensor_l2norm, "fused multi-tensor global L2 norm");
 }

@@ -133,6 +133,95 @@ __global__ void l2norm_kernel(
   }
 }
 
+
+// ---- LAMB ----
+// Two phases (per param group):
+//  1. update m/v, form the raw Adam update u (+ attached weight decay),
+//     OVERWRITING the grad buffer with u, while accumulating per-tensor
+//     sum(p^2) and sum(u^2) (wave-segmented atomics, ~1 atomic/wave).
+//  2. per-tensor trust ratio ||p|| / ||u|| (zero-guarded, optional clip at 1)
+//     scales the step: p -= lr * trust * u.
+
+template <typename T>
+__global__ void lamb_phase1_kernel(
+    const long* __restrict__ ptrs, const long* __restrict__ prefix, int n, long total,
+    float beta1, float beta2, float beta3, float eps, float wd,
+    float bc1, float bc2, float inv_clip,
+    float* __restrict__ sq_acc) {            // [n, 2]: sum p^2, sum u^2
+  const long stride = (long)gridDim.x * blockDim.x;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const float inv_bc2_sqrt = rsqrtf(bc2);
+  // converged-wave loop: every lane of a wave takes the same number of
+  // iterations (inactive tail lanes contribute zeros), so the wave-level
+  // reductions below stay well-defined
+  const long wave_first = (long)blockIdx.x * blockDim.x + (threadIdx.x - lane);
+  for (long base = wave_first; base < total; base += stride) {
+    const long i = base + lane;
+    const bool active = i < total;
+    const long ci = active ? i : total - 1;
+    const int t = find_tensor(prefix, n, ci);
+
+    float pf = 0.f, uf = 0.f;
+    if (active) {
+      const long off = i - prefix[t];
+      const T* p = reinterpret_cast<const T*>(ptrs[0 * n + t]);
+      T* g = reinterpret_cast<T*>(ptrs[1 * n + t]);
+      float* m = reinterpret_cast<float*>(ptrs[2 * n + t]);
+      float* v = reinterpret_cast<float*>(ptrs[3 * n + t]);
+
+      const float gf = Elem<T>::to_f32(g[off]) * inv_clip;
+      pf = Elem<T>::to_f32(p[off]);
+      const float mf = m[off] * beta1 + beta3 * gf;
+      const float vf = v[off] * beta2 + (1.f - beta2) * gf * gf;
+      m[off] = mf;
+      v[off] = vf;
+      const float denom = sqrtf(vf) * inv_bc2_sqrt + eps;
+      uf = (mf / bc1) / denom + wd * pf;
+      g[off] = Elem<T>::from_f32(uf);
+    }
+
+    // per-tensor squared-sum accumulation: a wave almost always lies inside
+    // one tensor -> one atomic per wave; mixed waves fall back to per-lane
+    const int t0 = __shfl(t, 0, 64);
+    const bool uniform = __all(!active || t == t0);
+    if (uniform) {
+      float p2 = wave_reduce_sum(pf * pf);
+      float u2 = wave_reduce_sum(uf * uf);
+      if (lane == 0) {
+        atomicAdd(&sq_acc[2 * t0 + 0], p2);
+        atomicAdd(&sq_acc[2 * t0 + 1], u2);
+      }
+    } else if (active) {
+      atomicAdd(&sq_acc[2 * t + 0], pf * pf);
+      atomicAdd(&sq_acc[2 * t + 1], uf * uf);
+    }
+  }
+}
+
+template <typename T>
+__global__ void lamb_phase2_kernel(
+    const long* __restrict__ ptrs, const long* __restrict__ prefix, int n, long total,
+    float lr, int adapt, int trust_clip,
+    const float* __restrict__ sq_acc) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < total; i += stride) {
+    int t = find_tensor(prefix, n, i);
+    long off = i - prefix[t];
+    T* p = reinterpret_cast<T*>(ptrs[0 * n + t]);
+    const T* g = reinterpret_cast<const T*>(ptrs[1 * n + t]);
+    float trust = 1.f;
+    if (adapt) {
+      float w_norm = sqrtf(sq_acc[2 * t + 0]);
+      float u_norm = sqrtf(sq_acc[2 * t + 1]);
+      if (w_norm > 0.f && u_norm > 0.f) trust = w_norm / u_norm;
+      if (trust_clip && trust > 1.f) trust = 1.f;
+    }
+    float pf = Elem<T>::to_f32(p[off]) - lr * trust * Elem<T>::to_f32(g[off]);
+    p[off] = Elem<T>::from_f32(pf);
+  }
+}
+
 template <typename scalar_t> struct ToHip2 { using type = float; };
 template <> struct ToHip2<at::BFloat16> { using type = __hip_bfloat16; };
 template <> struct ToHip2<at::Half> { using type = __half; };
@@ -184,4 +273,29 @@ at::Tensor multi_tensor_l2norm(std::vector<at::Tensor> tensors) {
   });
   HIP_CHECK_LAST();
   return out.sqrt();
+}
+
+
+void multi_tensor_lamb(
+    std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+    std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+    double lr, double beta1, double beta2, double beta3, double eps, double wd,
+    double bc1, double bc2, double clip_norm, bool adapt, bool trust_clip) {
+  TORCH_CHECK(!params.empty());
+  auto table = build_table({params, grads, exp_avgs, exp_avg_sqs});
+  auto sq_acc = at::zeros({table.n, 2}, params[0].options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  long blocks = std::min((long)4096, (table.total + kBlock - 1) / kBlock);
+  float inv_clip = clip_norm > 0 ? (float)(1.0 / clip_norm) : 1.f;
+  AT_DISPATCH_FLOATING_TYPES_AND2(at::kBFloat16, at::kHalf, params[0].scalar_type(), "mt_lamb", [&] {
+    using T = typename ToHip2<scalar_t>::type;
+    hipLaunchKernelGGL((lamb_phase1_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream,
+        table.ptrs.data_ptr<long>(), table.prefix.data_ptr<long>(), table.n, table.total,
+        (float)beta1, (float)beta2, (float)beta3, (float)eps, (float)wd,
+        (float)bc1, (float)bc2, inv_clip, sq_acc.data_ptr<float>());
+    hipLaunchKernelGGL((lamb_phase2_kernel<T>), dim3(blocks), dim3(kBlock), 0, stream,
+        table.ptrs.data_ptr<long>(), table.prefix.data_ptr<long>(), table.n, table.total,
+        (float)lr, adapt ? 1 : 0, trust_clip ? 1 : 0, sq_acc.data_ptr<float>());
+  });
+  HIP_CHECK_LAST();
 }

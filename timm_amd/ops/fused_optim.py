@@ -86,3 +86,39 @@ def fused_l2norm(tensors: List[torch.Tensor]) -> torch.Tensor:
         return ext.multi_tensor_l2norm(tensors)
     norms = torch._foreach_norm(tensors)
     return torch.linalg.vector_norm(torch.stack([n.float() for n in norms]))
+
+
+def fused_lamb_step(
+        params: List[torch.Tensor],
+        grads: List[torch.Tensor],
+        exp_avgs: List[torch.Tensor],
+        exp_avg_sqs: List[torch.Tensor],
+        lr: float,
+        beta1: float,
+        beta2: float,
+        beta3: float,
+        eps: float,
+        weight_decay: float,
+        bias_correction1: float,
+        bias_correction2: float,
+        clip_norm: Optional[float] = None,
+        always_adapt: bool = False,
+        trust_clip: bool = False,
+) -> bool:
+    """LAMB step over a flat list of same-device tensors: two kernel
+    launches for the whole group (m/v update + per-tensor trust-ratio
+    scaling; `csrc/multi_tensor.hip`). Returns False when the fused path is
+    unavailable and the caller must run the composable step."""
+    if not params:
+        return True
+    ext = _load_extension()
+    if not params[0].is_cuda or ext is None:
+        return False
+    adapt = weight_decay != 0 or always_adapt
+    ext.multi_tensor_lamb(
+        params, grads, exp_avgs, exp_avg_sqs,
+        lr, beta1, beta2, beta3, eps, weight_decay,
+        bias_correction1, bias_correction2,
+        clip_norm if clip_norm is not None else 0.0,
+        adapt, trust_clip)
+    return True

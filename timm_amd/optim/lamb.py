@@ -87,6 +87,37 @@ class Lamb(Optimizer):
             else:
                 bias_correction1, bias_correction2 = 1.0, 1.0
 
+            # fused multi-tensor path: whole group in two HIP launches
+            # (csrc/multi_tensor.hip); the composable loop below stays the
+            # reference for CPU and for caution/decoupled-decay variants
+            if not group['caution'] and not group.get('decoupled_decay', False):
+                group_params, group_grads, group_m, group_v = [], [], [], []
+                for p in group['params']:
+                    if p.grad is None:
+                        continue
+                    state = self.state[p]
+                    if len(state) == 0:
+                        state['exp_avg'] = torch.zeros_like(p, dtype=torch.float32)
+                        state['exp_avg_sq'] = torch.zeros_like(p, dtype=torch.float32)
+                    if state['exp_avg'].dtype != torch.float32:
+                        group_params = None  # pre-existing non-fp32 state: composable path
+                        break
+                    group_params.append(p)
+                    group_grads.append(p.grad)
+                    group_m.append(state['exp_avg'])
+                    group_v.append(state['exp_avg_sq'])
+                from .. import ops
+                clip = None
+                if clip_grad_norm is not None:
+                    clip = float(clip_grad_norm)
+                if group_params is not None and group_params and ops.fused_lamb_step(
+                        group_params, group_grads, group_m, group_v,
+                        group['lr'], beta1, beta2, beta3, group['eps'],
+                        group['weight_decay'], bias_correction1, bias_correction2,
+                        clip_norm=clip, always_adapt=group['always_adapt'],
+                        trust_clip=group['trust_clip']):
+                    continue
+
             for p in group['params']:
                 if p.grad is None:
                     continue
