@@ -1,8 +1,15 @@
-"""Feature extraction wrappers (reference `timm/models/_features.py`, 483 LoC).
+"""Feature-extraction wrappers (reference `timm/models/_features.py`, 483 LoC).
 
-Turns any classification model into a multi-scale feature backbone:
-`FeatureListNet`/`FeatureDictNet` (re-built sequential), `FeatureHookNet`
-(forward hooks), `FeatureGetterNet` (`forward_intermediates` getter).
+Four strategies turn a classifier into a multi-scale feature backbone:
+
+* `FeatureDictNet` / `FeatureListNet` — rebuild the model as a flat module
+  dict, tapping outputs at the registered feature points
+* `FeatureHookNet` — leave the model intact and capture taps with forward /
+  forward-pre hooks
+* `FeatureGetterNet` — delegate to the model's own `forward_intermediates()`
+
+`FeatureInfo` carries the per-tap metadata (channels, reduction, module
+path); `feature_take_indices` normalizes user index specs.
 """
 from collections import OrderedDict, defaultdict
 from copy import deepcopy
@@ -26,15 +33,13 @@ def feature_take_indices(
         indices: Optional[Union[int, List[int]]] = None,
         as_set: bool = False,
 ) -> Tuple[List[int], int]:
-    """Determine the absolute feature indices to 'take' from (reference `_features.py:28`).
-
-    `indices` may be None (all), an int (last n), or a list of ints (absolute
-    or negative indices)."""
+    """Normalize an index spec to absolute indices + the max index
+    (reference `_features.py:28`). `indices`: None = all, int = last n,
+    list = absolute/negative positions."""
     if indices is None:
-        indices = num_features  # all features if None
+        indices = num_features
 
     if isinstance(indices, int):
-        # convert int -> last n indices
         _assert(0 < indices <= num_features, f'last-n ({indices}) is out of range (1 to {num_features})')
         take_indices = [num_features - indices + i for i in range(indices)]
     else:
@@ -46,14 +51,12 @@ def feature_take_indices(
 
     if not torch.jit.is_scripting() and as_set:
         return set(take_indices), max(take_indices)
-
     return take_indices, max(take_indices)
 
 
 def _out_indices_as_tuple(x: Union[int, Tuple[int, ...]]) -> Tuple[int, ...]:
     if isinstance(x, int):
-        # if indices is an int, take last N features
-        return tuple(range(-x, 0))
+        return tuple(range(-x, 0))  # int spec = last N taps
     return tuple(x)
 
 
@@ -61,48 +64,40 @@ OutIndicesT = Union[int, Tuple[int, ...]]
 
 
 class FeatureInfo:
+    """Validated list of feature-tap dicts + the selected out indices."""
 
-    def __init__(
-            self,
-            feature_info: List[Dict],
-            out_indices: OutIndicesT,
-    ):
-        out_indices = _out_indices_as_tuple(out_indices)
+    def __init__(self, feature_info: List[Dict], out_indices: OutIndicesT):
         prev_reduction = 1
         for i, fi in enumerate(feature_info):
-            # sanity check the mandatory fields, there may be additional fields depending on the model
+            # required keys; models may attach extras (e.g. 'stage')
             assert 'num_chs' in fi and fi['num_chs'] > 0
             assert 'reduction' in fi and fi['reduction'] >= prev_reduction
-            prev_reduction = fi['reduction']
             assert 'module' in fi
+            prev_reduction = fi['reduction']
             fi.setdefault('index', i)
-        self.out_indices = out_indices
+        self.out_indices = _out_indices_as_tuple(out_indices)
         self.info = feature_info
 
     def from_other(self, out_indices: OutIndicesT):
-        out_indices = _out_indices_as_tuple(out_indices)
         return FeatureInfo(deepcopy(self.info), out_indices)
 
     def get(self, key: str, idx: Optional[Union[int, List[int]]] = None):
-        """Get value for feature modules at idx, can be a single or multiple select."""
+        """Value(s) of `key` at idx (default: the selected out indices)."""
         if idx is None:
-            return [self.info[i][key] for i in self.out_indices]
+            idx = self.out_indices
         if isinstance(idx, (tuple, list)):
             return [self.info[i][key] for i in idx]
-        else:
-            return self.info[idx][key]
+        return self.info[idx][key]
 
     def get_dicts(self, keys: Optional[List[str]] = None, idx: Optional[Union[int, List[int]]] = None):
-        """return info dicts for specified keys (or all if None) at specified indices (or out_indices if None)."""
+        """Info dict(s), optionally projected onto `keys`."""
+        def project(i):
+            return self.info[i] if keys is None else {k: self.info[i][k] for k in keys}
         if idx is None:
-            if keys is None:
-                return [self.info[i] for i in self.out_indices]
-            else:
-                return [{k: self.info[i][k] for k in keys} for i in self.out_indices]
+            idx = self.out_indices
         if isinstance(idx, (tuple, list)):
-            return [self.info[i] if keys is None else {k: self.info[i][k] for k in keys} for i in idx]
-        else:
-            return self.info[idx] if keys is None else {k: self.info[idx][k] for k in keys}
+            return [project(i) for i in idx]
+        return project(idx)
 
     def channels(self, idx: Optional[Union[int, List[int]]] = None):
         return self.get('num_chs', idx)
@@ -121,7 +116,8 @@ class FeatureInfo:
 
 
 class FeatureHooks:
-    """Feature Hook Helper (reference `_features.py:150`)."""
+    """Registers forward/forward-pre hooks and collects their outputs per
+    device (reference `_features.py:150`)."""
 
     def __init__(
             self,
@@ -130,76 +126,70 @@ class FeatureHooks:
             out_map: Sequence[Union[int, str]] = None,
             default_hook_type: str = 'forward',
     ):
-        # setup feature hooks
         self._feature_outputs = defaultdict(OrderedDict)
         self._handles = []
-        modules = {k: v for k, v in named_modules}
-        for i, h in enumerate(hooks):
-            hook_name = h if isinstance(h, str) else h['module']
-            m = modules[hook_name]
-            hook_id = out_map[i] if out_map else hook_name
-            hook_fn = partial(self._collect_output_hook, hook_id)
+        modules = dict(named_modules)
+        for i, spec in enumerate(hooks):
+            target = spec if isinstance(spec, str) else spec['module']
             hook_type = default_hook_type
-            if isinstance(h, dict):
-                hook_type = h.get('hook_type', default_hook_type)
+            if isinstance(spec, dict):
+                hook_type = spec.get('hook_type', default_hook_type)
+            hook_id = out_map[i] if out_map else target
+            fn = partial(self._collect, hook_id)
+            module = modules[target]
             if hook_type == 'forward_pre':
-                handle = m.register_forward_pre_hook(hook_fn)
+                self._handles.append(module.register_forward_pre_hook(fn))
             elif hook_type == 'forward':
-                handle = m.register_forward_hook(hook_fn)
+                self._handles.append(module.register_forward_hook(fn))
             else:
-                assert False, "Unsupported hook type"
-            self._handles.append(handle)
+                raise AssertionError('Unsupported hook type')
 
-    def _collect_output_hook(self, hook_id, *args):
-        x = args[-1]  # tensor we want is last argument, output for fwd, input for fwd_pre
+    def _collect(self, hook_id, *args):
+        # last positional arg is the interesting tensor: output for forward
+        # hooks, the input tuple for forward-pre hooks
+        x = args[-1]
         if isinstance(x, tuple):
-            x = x[0]  # unwrap input tuple
+            x = x[0]
         self._feature_outputs[x.device][hook_id] = x
 
     def get_output(self, device) -> Dict[str, torch.tensor]:
-        output = self._feature_outputs[device]
-        self._feature_outputs[device] = OrderedDict()  # clear after reading
-        return output
+        out = self._feature_outputs[device]
+        self._feature_outputs[device] = OrderedDict()  # drain
+        return out
 
 
 def _module_list(module, flatten_sequential=False):
-    # a yield/iter would be better for this but wouldn't be compatible with torchscript
-    ml = []
-    for name, module in module.named_children():
-        if flatten_sequential and isinstance(module, nn.Sequential):
-            # first level of Sequential containers is flattened into containing model
-            for child_name, child_module in module.named_children():
-                combined = [name, child_name]
-                ml.append(('_'.join(combined), '.'.join(combined), child_module))
+    # list (not generator) for torchscript compatibility
+    entries = []
+    for name, child in module.named_children():
+        if flatten_sequential and isinstance(child, nn.Sequential):
+            # flatten ONE level of Sequential containers into the parent
+            for sub_name, sub_child in child.named_children():
+                entries.append((f'{name}_{sub_name}', f'{name}.{sub_name}', sub_child))
         else:
-            ml.append((name, name, module))
-    return ml
+            entries.append((name, name, child))
+    return entries
 
 
 def _get_feature_info(net, out_indices: OutIndicesT):
     feature_info = getattr(net, 'feature_info')
     if isinstance(feature_info, FeatureInfo):
         return feature_info.from_other(out_indices)
-    elif isinstance(feature_info, (list, tuple)):
+    if isinstance(feature_info, (list, tuple)):
         return FeatureInfo(net.feature_info, out_indices)
-    else:
-        assert False, "Provided feature_info is not valid"
+    raise AssertionError('Provided feature_info is not valid')
 
 
 def _get_return_layers(feature_info, out_map):
-    module_names = feature_info.module_name()
-    return_layers = {}
-    for i, name in enumerate(module_names):
-        return_layers[name] = out_map[i] if out_map is not None else feature_info.out_indices[i]
-    return return_layers
+    return {
+        name: (out_map[i] if out_map is not None else feature_info.out_indices[i])
+        for i, name in enumerate(feature_info.module_name())
+    }
 
 
 class FeatureDictNet(nn.ModuleDict):
-    """Feature extractor with OrderedDict return (reference `_features.py:230`).
-
-    Wrap a model and extract features as specified by the out indices, the
-    network is partially re-built from contained modules.
-    """
+    """Rebuilt-model extractor returning {tap_id: tensor}
+    (reference `_features.py:230`). Children past the deepest tap are dropped."""
 
     def __init__(
             self,
@@ -218,14 +208,12 @@ class FeatureDictNet(nn.ModuleDict):
         self.return_layers = {}
 
         return_layers = _get_return_layers(self.feature_info, out_map)
-        modules = _module_list(model, flatten_sequential=flatten_sequential)
         remaining = set(return_layers.keys())
         layers = OrderedDict()
-        for new_name, old_name, module in modules:
+        for new_name, old_name, module in _module_list(model, flatten_sequential=flatten_sequential):
             layers[new_name] = module
             if old_name in remaining:
-                # return id has to be consistently str type for torchscript
-                self.return_layers[new_name] = str(return_layers[old_name])
+                self.return_layers[new_name] = str(return_layers[old_name])  # str ids for torchscript
                 remaining.remove(old_name)
             if not remaining:
                 break
@@ -240,19 +228,17 @@ class FeatureDictNet(nn.ModuleDict):
         out = OrderedDict()
         for i, (name, module) in enumerate(self.items()):
             if self.grad_checkpointing and not torch.jit.is_scripting():
-                # Skipping checkpoint of first module because need a gradient at input
-                # Skipping last because networks with in-place ops might fail w/ checkpointing enabled
-                # NOTE: first_or_last module could be static, but recalc in is_scripting guard to avoid jit issues
-                first_or_last_module = i == 0 or i == max(len(self) - 1, 0)
-                x = module(x) if first_or_last_module else checkpoint(module, x)
+                # keep the first module (input needs grad) and the last
+                # (in-place ops can break under checkpointing) un-checkpointed
+                boundary = i == 0 or i == max(len(self) - 1, 0)
+                x = module(x) if boundary else checkpoint(module, x)
             else:
                 x = module(x)
 
             if name in self.return_layers:
                 out_id = self.return_layers[name]
                 if isinstance(x, (tuple, list)) and not self.concat:
-                    # If model tap is a tuple or list, concat or select first element
-                    # FIXME this may need to be more generic / flexible for some nets
+                    # multi-output tap: concat on request, else take the first
                     out[out_id] = torch.cat(x, 1) if self.concat else x[0]
                 else:
                     out[out_id] = x
@@ -263,7 +249,8 @@ class FeatureDictNet(nn.ModuleDict):
 
 
 class FeatureListNet(FeatureDictNet):
-    """Feature extractor with list return (reference `_features.py:315`)."""
+    """Same as FeatureDictNet but returns the taps as a list
+    (reference `_features.py:315`)."""
 
     def __init__(
             self,
@@ -286,11 +273,8 @@ class FeatureListNet(FeatureDictNet):
 
 
 class FeatureHookNet(nn.ModuleDict):
-    """FeatureHookNet (reference `_features.py:348`).
-
-    Wrap a model and extract features specified by the out indices using
-    forward/forward-pre hooks.
-    """
+    """Hook-based extractor: the wrapped model runs unmodified (or lightly
+    truncated) and taps are captured by hooks (reference `_features.py:348`)."""
 
     def __init__(
             self,
@@ -311,25 +295,24 @@ class FeatureHookNet(nn.ModuleDict):
         self.grad_checkpointing = False
         if no_rewrite is None:
             no_rewrite = not flatten_sequential
+
         layers = OrderedDict()
         hooks = []
         if no_rewrite:
-            if hasattr(model, 'reset_classifier'):  # make sure classifier is removed?
-                model.reset_classifier(0)
+            if hasattr(model, 'reset_classifier'):
+                model.reset_classifier(0)  # head output is never a tap
             layers['body'] = model
             hooks.extend(self.feature_info.get_dicts())
         else:
-            modules = _module_list(model, flatten_sequential=flatten_sequential)
             remaining = {
-                f['module']: f['hook_type'] if 'hook_type' in f else default_hook_type
+                f['module']: f.get('hook_type', default_hook_type)
                 for f in self.feature_info.get_dicts()
             }
-            for new_name, old_name, module in modules:
+            for new_name, old_name, module in _module_list(model, flatten_sequential=flatten_sequential):
                 layers[new_name] = module
-                for fn, fm in module.named_modules(prefix=old_name):
+                for fn, _ in module.named_modules(prefix=old_name):
                     if fn in remaining:
-                        hooks.append(dict(module=fn, hook_type=remaining[fn]))
-                        del remaining[fn]
+                        hooks.append(dict(module=fn, hook_type=remaining.pop(fn)))
                 if not remaining:
                     break
             assert not remaining, f'Return layers ({remaining}) are not present in model'
@@ -342,8 +325,8 @@ class FeatureHookNet(nn.ModuleDict):
     def forward(self, x):
         for i, (name, module) in enumerate(self.items()):
             if self.grad_checkpointing and not torch.jit.is_scripting():
-                first_or_last_module = i == 0 or i == max(len(self) - 1, 0)
-                x = module(x) if first_or_last_module else checkpoint(module, x)
+                boundary = i == 0 or i == max(len(self) - 1, 0)
+                x = module(x) if boundary else checkpoint(module, x)
             else:
                 x = module(x)
         out = self.hooks.get_output(x.device)
@@ -351,8 +334,8 @@ class FeatureHookNet(nn.ModuleDict):
 
 
 class FeatureGetterNet(nn.ModuleDict):
-    """FeatureGetterNet (reference `_features.py:435`) — uses model's
-    `forward_intermediates()` API."""
+    """Extractor over the model's own `forward_intermediates()` API
+    (reference `_features.py:435`)."""
 
     def __init__(
             self,
@@ -366,7 +349,7 @@ class FeatureGetterNet(nn.ModuleDict):
     ):
         super().__init__()
         if prune and hasattr(model, 'prune_intermediate_layers'):
-            # replace out_indices after they've been normalized in the model pruning fn
+            # pruning normalizes the indices, so adopt the returned spec
             out_indices = model.prune_intermediate_layers(
                 out_indices,
                 prune_norm=not norm,
@@ -383,11 +366,10 @@ class FeatureGetterNet(nn.ModuleDict):
         self.model.set_grad_checkpointing(enable)
 
     def forward(self, x):
-        features = self.model.forward_intermediates(
+        return self.model.forward_intermediates(
             x,
             indices=self.out_indices,
             norm=self.norm,
             output_fmt=self.output_fmt,
             intermediates_only=True,
         )
-        return features
