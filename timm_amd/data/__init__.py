@@ -4,6 +4,8 @@ from .auto_augment import (
 )
 from .config import resolve_data_config, resolve_model_data_config
 from .constants import *
+from .dataset_info import CustomDatasetInfo, DatasetInfo, DatasetInfoLabelMapper
+from .imagenet_info import ImageNetInfo, infer_imagenet_subset
 from .dataset import ImageDataset, IterableImageDataset, AugMixDataset
 from .dataset_factory import create_dataset
 from .distributed_sampler import OrderedDistributedSampler, RepeatAugSampler

@@ -138,6 +138,8 @@ def init_distributed_device_so(
     if distributed and device_type == 'cuda' and torch.cuda.is_available():
         device = f'cuda:{local_rank}'
     if device_type == 'cuda' and torch.cuda.is_available():
+        if ':' not in device:
+            device = f'{device}:0'  # set_device needs an explicit index
         torch.cuda.set_device(device)
 
     return dict(
