@@ -465,6 +465,13 @@ def test_model_gpu_vs_cpu(model_name):
     # gemma4: 16 sandwich-RMSNorm blocks with scale=1.0 attention accumulate more
     # bf16 drift than pre-norm ViTs; still well-correlated with the fp32 reference
     if model_name.startswith('gemma4'):
+        # bf16 drift through 16 sandwich-RMSNorm blocks is large; don't let a
+        # loose end-to-end bound mask kernel bugs — verify kernel correctness
+        # against an fp32 GPU run (tight) and bf16 drift separately (loose)
+        with torch.no_grad():
+            y_gpu32 = m_gpu.float()(xg.float())
+        err32 = rel_err(y_gpu32.cpu(), y_cpu)
+        assert err32 < 5e-3, f'{model_name} fp32 GPU output err {err32}'
         tol = 0.25
     elif model_name.startswith('visformer'):
         tol = 0.15
