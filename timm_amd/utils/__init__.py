@@ -1,3 +1,4 @@
+from .attention_extract import AttentionExtract
 from .agc import adaptive_clip_grad
 from .checkpoint_saver import CheckpointSaver
 from .clip_grad import dispatch_clip_grad
