@@ -1,4 +1,6 @@
-"""Attention-module factory for CNN blocks (reference `timm/layers/create_attn.py`)."""
+"""CNN attention-module factory (reference `timm/layers/create_attn.py`):
+resolve a short name / bool / class to an attention-module constructor.
+All constructors take the channel count as their first positional arg."""
 import functools
 
 import torch
@@ -14,75 +16,53 @@ from .selective_kernel import SelectiveKernel
 from .split_attn import SplitAttn
 from .squeeze_excite import SEModule, EffectiveSEModule
 
+_ATTN_MAP = {
+    # lightweight channel / coarse-spatial gates
+    'se': SEModule,
+    'ese': EffectiveSEModule,
+    'eca': EcaModule,
+    'ecam': functools.partial(EcaModule, use_mlp=True),
+    'ceca': CecaModule,
+    'ge': GatherExcite,
+    'gc': GlobalContext,
+    'gca': functools.partial(GlobalContext, fuse_add=True, fuse_scale=False),
+    'cbam': CbamModule,
+    'lcbam': LightCbamModule,
+    'coord': CoordAttn,
+    'scoord': SimpleCoordAttn,
+    'ela': EfficientLocalAttn,
+    'strip': StripAttn,
+    # heavier attention-like blocks
+    'sk': SelectiveKernel,
+    'splat': SplitAttn,
+    'nl': NonLocalAttn,
+    'bat': BatNonLocalAttn,
+    # spatial self-attention
+    'lambda': LambdaLayer,
+    'bottleneck': BottleneckAttn,
+    'halo': HaloAttn,
+}
+
 
 def get_attn(attn_type):
     if isinstance(attn_type, torch.nn.Module):
         return attn_type
-    module_cls = None
-    if attn_type:
-        if isinstance(attn_type, str):
-            attn_type = attn_type.lower()
-            # Lightweight attention modules (channel and/or coarse spatial).
-            if attn_type == 'se':
-                module_cls = SEModule
-            elif attn_type == 'ese':
-                module_cls = EffectiveSEModule
-            elif attn_type == 'eca':
-                module_cls = EcaModule
-            elif attn_type == 'ecam':
-                module_cls = functools.partial(EcaModule, use_mlp=True)
-            elif attn_type == 'ceca':
-                module_cls = CecaModule
-            elif attn_type == 'ge':
-                module_cls = GatherExcite
-            elif attn_type == 'gc':
-                module_cls = GlobalContext
-            elif attn_type == 'gca':
-                module_cls = functools.partial(GlobalContext, fuse_add=True, fuse_scale=False)
-            elif attn_type == 'cbam':
-                module_cls = CbamModule
-            elif attn_type == 'lcbam':
-                module_cls = LightCbamModule
-            elif attn_type == 'coord':
-                module_cls = CoordAttn
-            elif attn_type == 'scoord':
-                module_cls = SimpleCoordAttn
-            elif attn_type == 'ela':
-                module_cls = EfficientLocalAttn
-            elif attn_type == 'strip':
-                module_cls = StripAttn
-            # Attention / attention-like modules w/ significant compute
-            elif attn_type == 'sk':
-                module_cls = SelectiveKernel
-            elif attn_type == 'splat':
-                module_cls = SplitAttn
-            elif attn_type == 'nl':
-                module_cls = NonLocalAttn
-            elif attn_type == 'bat':
-                module_cls = BatNonLocalAttn
-            # Self-attention / attention-like modules w/ significant compute
-            elif attn_type == 'lambda':
-                module_cls = LambdaLayer
-            elif attn_type == 'bottleneck':
-                module_cls = BottleneckAttn
-            elif attn_type == 'halo':
-                module_cls = HaloAttn
-            else:
-                assert False, "Invalid attn module (%s)" % attn_type
-        elif isinstance(attn_type, bool):
-            if attn_type:
-                module_cls = SEModule
-        else:
-            module_cls = attn_type
-    return module_cls
+    if not attn_type:
+        return None
+    if isinstance(attn_type, str):
+        key = attn_type.lower()
+        assert key in _ATTN_MAP, f'Invalid attn module ({attn_type})'
+        return _ATTN_MAP[key]
+    if isinstance(attn_type, bool):
+        return SEModule if attn_type else None
+    return attn_type  # already a class / partial
 
 
 def create_attn(attn_type, channels, **kwargs):
     module_cls = get_attn(attn_type)
-    if module_cls is not None:
-        # NOTE: it's expected the first (positional) argument of all attention layers is the # input channels
-        return module_cls(channels, **kwargs)
-    return None
+    if module_cls is None:
+        return None
+    return module_cls(channels, **kwargs)
 
 
 # explicit-name alias used by model files

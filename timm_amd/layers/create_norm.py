@@ -1,7 +1,7 @@
-"""Norm layer factory (reference `timm/layers/create_norm.py`)."""
+"""Norm-layer factory (reference `timm/layers/create_norm.py`): name or
+callable -> norm constructor, preserving partial-bound kwargs."""
 import functools
 import types
-from typing import Type
 
 import torch.nn as nn
 
@@ -22,34 +22,29 @@ _NORM_MAP = dict(
     simplenorm=SimpleNorm,
     simplenorm2d=SimpleNorm2d,
 )
-_NORM_TYPES = {m for n, m in _NORM_MAP.items()}
+_NORM_TYPES = set(_NORM_MAP.values())
 
 
 def create_norm_layer(layer_name, num_features, **kwargs):
-    layer = get_norm_layer(layer_name)
-    layer_instance = layer(num_features, **kwargs)
-    return layer_instance
+    return get_norm_layer(layer_name)(num_features, **kwargs)
 
 
 def get_norm_layer(norm_layer):
     if norm_layer is None:
         return None
     assert isinstance(norm_layer, (type, str, types.FunctionType, functools.partial))
-    norm_kwargs = {}
 
-    # unbind partial fn, so args can be rebound later
+    bound_kwargs = {}
     if isinstance(norm_layer, functools.partial):
-        norm_kwargs.update(norm_layer.keywords)
+        # unwrap so any bound kwargs survive the name lookup and re-bind below
+        bound_kwargs.update(norm_layer.keywords)
         norm_layer = norm_layer.func
 
     if isinstance(norm_layer, str):
         if not norm_layer:
             return None
-        layer_name = norm_layer.replace('_', '').lower()
-        norm_layer = _NORM_MAP[layer_name]
-    else:
-        norm_layer = norm_layer
+        norm_layer = _NORM_MAP[norm_layer.replace('_', '').lower()]
 
-    if norm_kwargs:
-        norm_layer = functools.partial(norm_layer, **norm_kwargs)  # bind/rebind args
+    if bound_kwargs:
+        norm_layer = functools.partial(norm_layer, **bound_kwargs)
     return norm_layer

@@ -1,6 +1,6 @@
-"""Assertion helper usable under torch.fx tracing (reference `timm/layers/trace_utils.py`)."""
+"""Tracing-compatible helpers (reference `timm/layers/trace_utils.py`):
+`_assert` stays symbolic under torch.fx instead of branching."""
 import torch
-
 
 try:
     from torch import _assert

@@ -42,6 +42,12 @@ __device__ __forceinline__ int swz(int row, int byte_off) {
   return kSwz ? (byte_off ^ ((row & 7) << 4)) : byte_off;
 }
 
+// swizzle for 64-byte LDS rows (transposed tiles over a 32-row dimension)
+template <bool kSwz>
+__device__ __forceinline__ int swz64(int row, int byte_off) {
+  return kSwz ? (byte_off ^ ((row & 3) << 4)) : byte_off;
+}
+
 // Cooperative stage of a [rows x D] bf16 tile into (optionally swizzled) LDS,
 // zero-filling rows past `n_limit`.  16B chunks, whole block participates.
 template <bool kSwz>
@@ -70,6 +76,7 @@ __device__ __forceinline__ bf16x8_t frag_row(const __bf16* lds, int row, int d0,
 
 // Gather one MFMA B column fragment: lane holds elements
 // T[k = g4*8+j][col] for j=0..7 (scalar reads, swizzle-aware addressing).
+// Fallback path for kMaxD > 64 where the transposed tiles don't fit in LDS.
 template <bool kSwz>
 __device__ __forceinline__ bf16x8_t frag_col(const __bf16* lds, int g4, int col, int D) {
   bf16x8_t out;
@@ -80,6 +87,38 @@ __device__ __forceinline__ bf16x8_t frag_col(const __bf16* lds, int g4, int col,
         reinterpret_cast<const char*>(lds) + swz<kSwz>(row, row * D * 2 + col * 2));
   }
   return out;
+}
+
+// Cooperative transposed stage: scatter a [rows x D] global tile into
+// LDS as T[D][rows] (64-byte rows, swz64). One b128 global read per chunk,
+// eight u16 scattered LDS writes — paid once per tile so the MFMA loop can
+// read column fragments as vector ds_read_b128 instead of 8 scalar reads.
+template <bool kSwz>
+__device__ __forceinline__ void stage_tile_t(
+    __bf16* lds_t, const __bf16* src, long src_sn, int base, int rows, int n_limit,
+    int D, int d8) {
+  const int chunks = rows * d8;
+  for (int c = threadIdx.x; c < chunks; c += kBlockThreads) {
+    int row = c / d8, col8 = c % d8;
+    bf16x8_t val = {};
+    if (base + row < n_limit) {
+      val = *reinterpret_cast<const bf16x8_t*>(src + (long)(base + row) * src_sn + col8 * 8);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int d = col8 * 8 + j;
+      *reinterpret_cast<__bf16*>(
+          reinterpret_cast<char*>(lds_t) + swz64<kSwz>(d, d * rows * 2 + row * 2)) = val[j];
+    }
+  }
+}
+
+// Vector read of a transposed-tile B fragment: lane (l16, g4) takes
+// T[drow][g4*8 .. +8) — contiguous, one ds_read_b128.
+template <bool kSwz>
+__device__ __forceinline__ bf16x8_t frag_row_t(const __bf16* lds_t, int drow, int g4, int rows) {
+  return *reinterpret_cast<const bf16x8_t*>(
+      reinterpret_cast<const char*>(lds_t) + swz64<kSwz>(drow, drow * rows * 2 + g4 * 8 * 2));
 }
 
 // ---------------------------------------------------------------------------
@@ -103,11 +142,13 @@ void attn_bwd_dq_kernel(
     long v_sb, long v_sh, long v_sn,
     long dq_sb, long dq_sh, long dq_sn) {
   constexpr int kQB = 64, kKvB = 32;
+  constexpr bool kUseT = kMaxD <= 64;  // transposed K tile fits in LDS
   __shared__ __bf16 q_lds[kQB * kMaxD];
   __shared__ __bf16 do_lds[kQB * kMaxD];
   __shared__ __bf16 k_lds[kKvB * kMaxD];
   __shared__ __bf16 v_lds[kKvB * kMaxD];
   __shared__ __bf16 ds_lds[4 * 16 * kKvB];
+  __shared__ __bf16 kt_lds[kUseT ? kMaxD * kKvB : 1];
 
   const int bh = blockIdx.y;
   const int b = bh / H;
@@ -144,6 +185,9 @@ void attn_bwd_dq_kernel(
     __syncthreads();  // previous iteration's reads done before restage
     stage_tile<kSwizzle>(k_lds, k_bh, k_sn, kv0, kKvB, Nk, D, d8);
     stage_tile<kSwizzle>(v_lds, v_bh, v_sn, kv0, kKvB, Nk, D, d8);
+    if (kUseT) {
+      stage_tile_t<kSwizzle>(kt_lds, k_bh, k_sn, kv0, kKvB, Nk, D, d8);
+    }
     __syncthreads();
 
     // S and dP fragments: [n16][reg], rows = q (wave*16 + g4*4 + r), col = kv
@@ -193,7 +237,9 @@ void attn_bwd_dq_kernel(
 #pragma unroll
     for (int f = 0; f < kMaxD / 16; ++f) {
       if (f * 16 >= D) break;
-      bf16x8_t kcol = frag_col<kSwizzle>(k_lds, g4, f * 16 + l16, D);
+      bf16x8_t kcol = kUseT
+          ? frag_row_t<kSwizzle>(kt_lds, f * 16 + l16, g4, kKvB)
+          : frag_col<kSwizzle>(k_lds, g4, f * 16 + l16, D);
       acc_dq[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, kcol, acc_dq[f], 0, 0, 0);
     }
   }
@@ -234,11 +280,14 @@ void attn_bwd_dkdv_kernel(
     long dk_sb, long dk_sh, long dk_sn,
     long dv_sb, long dv_sh, long dv_sn) {
   constexpr int kKvB = 64, kQB = 32;
+  constexpr bool kUseT = kMaxD <= 64;  // transposed Q/dO tiles fit in LDS
   __shared__ __bf16 k_lds[kKvB * kMaxD];
   __shared__ __bf16 v_lds[kKvB * kMaxD];
   __shared__ __bf16 q_lds[kQB * kMaxD];
   __shared__ __bf16 do_lds[kQB * kMaxD];
   __shared__ __bf16 pt_lds[4 * 16 * kQB];   // reused for P^T then dS^T
+  __shared__ __bf16 qt_lds[kUseT ? kMaxD * kQB : 1];
+  __shared__ __bf16 dot_lds[kUseT ? kMaxD * kQB : 1];
 
   const int bh = blockIdx.y;
   const int b = bh / H;
@@ -270,6 +319,10 @@ void attn_bwd_dkdv_kernel(
     __syncthreads();
     stage_tile<kSwizzle>(q_lds, q_bh, q_sn, q0, kQB, Nq, D, d8);
     stage_tile<kSwizzle>(do_lds, do_bh, D, q0, kQB, Nq, D, d8);
+    if (kUseT) {
+      stage_tile_t<kSwizzle>(qt_lds, q_bh, q_sn, q0, kQB, Nq, D, d8);
+      stage_tile_t<kSwizzle>(dot_lds, do_bh, D, q0, kQB, Nq, D, d8);
+    }
     __syncthreads();
 
     // per-q-column stats: col = q0 + n16*16 + l16
@@ -322,7 +375,9 @@ void attn_bwd_dkdv_kernel(
 #pragma unroll
       for (int f = 0; f < kMaxD / 16; ++f) {
         if (f * 16 >= D) break;
-        bf16x8_t dcol = frag_col<kSwizzle>(do_lds, g4, f * 16 + l16, D);
+        bf16x8_t dcol = kUseT
+            ? frag_row_t<kSwizzle>(dot_lds, f * 16 + l16, g4, kQB)
+            : frag_col<kSwizzle>(do_lds, g4, f * 16 + l16, D);
         acc_dv[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, dcol, acc_dv[f], 0, 0, 0);
       }
     }
@@ -344,7 +399,9 @@ void attn_bwd_dkdv_kernel(
 #pragma unroll
       for (int f = 0; f < kMaxD / 16; ++f) {
         if (f * 16 >= D) break;
-        bf16x8_t qcolf = frag_col<kSwizzle>(q_lds, g4, f * 16 + l16, D);
+        bf16x8_t qcolf = kUseT
+            ? frag_row_t<kSwizzle>(qt_lds, f * 16 + l16, g4, kQB)
+            : frag_col<kSwizzle>(q_lds, g4, f * 16 + l16, D);
         acc_dk[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsa, qcolf, acc_dk[f], 0, 0, 0);
       }
     }
