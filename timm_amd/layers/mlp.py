@@ -49,6 +49,20 @@ class Mlp(nn.Module):
 
     def forward(self, x):
         if (
+                self._fused_act in ('gelu', 'gelu_tanh', 'relu')
+                and x.is_cuda
+                and not torch.is_grad_enabled()
+                and self.fc1.bias is not None
+                and isinstance(self.norm, nn.Identity)
+        ):
+            # inference: bias+act fused into the hipBLASLt GEMM epilogue
+            # (GELU here is the tanh approximation — within bf16 noise)
+            x2 = x.reshape(-1, x.shape[-1])
+            h = torch._addmm_activation(
+                self.fc1.bias, x2, self.fc1.weight.t(),
+                use_gelu=self._fused_act != 'relu')
+            h = h.reshape(*x.shape[:-1], h.shape[-1])
+        elif (
                 self._fused_act is not None
                 and x.is_cuda
                 and self.fc1.bias is not None
