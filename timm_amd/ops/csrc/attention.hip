@@ -34,6 +34,12 @@ __device__ __forceinline__ int swz(int row, int byte_off) {
   return kSwz ? (byte_off ^ ((row & 7) << 4)) : byte_off;
 }
 
+// swizzle for 64-byte LDS rows (transposed V tile over the 32-row kv dim)
+template <bool kSwz>
+__device__ __forceinline__ int swz64(int row, int byte_off) {
+  return kSwz ? (byte_off ^ ((row & 3) << 4)) : byte_off;
+}
+
 template <int kMaxD, bool kHasMask, bool kSwizzle>
 __global__ __launch_bounds__(kBlockThreads)
 void attn_fwd_kernel(
@@ -53,7 +59,7 @@ void attn_fwd_kernel(
   // LDS: Q[64][D] | K[32][D] | V[32][D] | P[4][16][32]
   __shared__ __bf16 q_lds[kQTile * kMaxD];
   __shared__ __bf16 k_lds[kKvTile * kMaxD];
-  __shared__ __bf16 v_lds[kKvTile * kMaxD];
+  __shared__ __bf16 vt_lds[kMaxD * kKvTile];  // V transposed: [d][kv], 64B rows
   __shared__ __bf16 p_lds[4 * 16 * kKvTile];
 
   const int bh = blockIdx.y;
@@ -108,8 +114,15 @@ void attn_fwd_kernel(
         }
         *reinterpret_cast<bf16x8_t*>(
             reinterpret_cast<char*>(k_lds) + swz<kSwizzle>(row, row * D * 2 + col8 * 16)) = kval;
-        // V is read column-wise in PV (no 16B-row reads) -> keep linear
-        *reinterpret_cast<bf16x8_t*>(reinterpret_cast<char*>(v_lds) + row * D * 2 + col8 * 16) = vval;
+        // V is read column-wise in PV: stage TRANSPOSED so the B-fragment
+        // loads below are single ds_read_b128 instead of 8 scalar reads
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int d = col8 * 8 + j;
+          *reinterpret_cast<__bf16*>(
+              reinterpret_cast<char*>(vt_lds) +
+              swz64<kSwizzle>(d, d * kKvTile * 2 + row * 2)) = vval[j];
+        }
       }
     }
     __syncthreads();
@@ -188,12 +201,11 @@ void attn_fwd_kernel(
       if (f * 16 >= D) break;
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc_o[f][r] *= factor[r];
-      // B operand: V[kv=(g4*8+j)][dcol=f*16+l16] — strided column gather
-      bf16x8_t bv;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        bv[j] = v_lds[(g4 * 8 + j) * D + f * 16 + l16];
-      }
+      // B operand: V^T[dcol=f*16+l16][kv=g4*8 .. +8) — one vector LDS read
+      const int drow = f * 16 + l16;
+      bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(
+          reinterpret_cast<const char*>(vt_lds) +
+          swz64<kSwizzle>(drow, drow * kKvTile * 2 + g4 * 8 * 2));
       acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, acc_o[f], 0, 0, 0);
     }
     __syncthreads();  // before next tile overwrites K/V
