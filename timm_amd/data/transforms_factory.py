@@ -1,18 +1,38 @@
-"""Transform factory (reference `timm/data/transforms_factory.py:20,65,273,387`)."""
+"""Transform-pipeline factory (reference
+`timm/data/transforms_factory.py:20,65,273,387`).
+
+Three pipeline builders (no-aug train, full ImageNet train, eval) share one
+output-stage rule: with the GPU prefetcher the pipeline ends at a uint8
+numpy array (tensor conversion + normalize happen on-device in the fused
+prefetch kernel); otherwise it ends in ToTensor+Normalize (or a raw dtype
+tensor when normalize=False).
+"""
 import math
 from typing import Optional, Tuple, Union
 
 import torch
 
 from . import image_ops as transforms
-
-from .auto_augment import rand_augment_transform, augment_and_mix_transform, auto_augment_transform
-from .constants import IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD, DEFAULT_CROP_PCT, DEFAULT_CROP_MODE
+from .auto_augment import augment_and_mix_transform, auto_augment_transform, rand_augment_transform
+from .constants import DEFAULT_CROP_MODE, DEFAULT_CROP_PCT, IMAGENET_DEFAULT_MEAN, IMAGENET_DEFAULT_STD
 from .random_erasing import RandomErasing
 from .transforms import (
-    str_to_interp_mode, str_to_pil_interp, RandomResizedCropAndInterpolation, ResizeKeepRatio,
-    CenterCropOrPad, RandomCropOrPad, TrimBorder, ToNumpy, MaybeToTensor, MaybePILToTensor,
+    CenterCropOrPad, MaybePILToTensor, MaybeToTensor, RandomCropOrPad,
+    RandomResizedCropAndInterpolation, ResizeKeepRatio, ToNumpy, TrimBorder,
+    str_to_interp_mode, str_to_pil_interp,
 )
+
+
+def _output_stage(mean, std, use_prefetcher, normalize):
+    """Pipeline tail shared by all three builders (see module docstring)."""
+    if use_prefetcher:
+        return [ToNumpy()]
+    if not normalize:
+        return [MaybePILToTensor()]
+    return [
+        MaybeToTensor(),
+        transforms.Normalize(mean=torch.tensor(mean), std=torch.tensor(std)),
+    ]
 
 
 def transforms_noaug_train(
@@ -23,29 +43,55 @@ def transforms_noaug_train(
         use_prefetcher: bool = False,
         normalize: bool = True,
 ):
-    """No-augmentation image transforms for training (reference `:20`)."""
+    """Deterministic resize+crop training pipeline (reference `:20`)."""
     if interpolation == 'random':
-        # random interpolation not supported with no-aug
-        interpolation = 'bilinear'
+        interpolation = 'bilinear'  # no random interpolation without aug
     tfl = [
         transforms.Resize(img_size, interpolation=str_to_interp_mode(interpolation)),
-        transforms.CenterCrop(img_size)
+        transforms.CenterCrop(img_size),
     ]
-    if use_prefetcher:
-        # prefetcher and collate will handle tensor conversion and norm
-        tfl += [ToNumpy()]
-    elif not normalize:
-        # when normalize disabled, converted to tensor without scaling, keeps original dtype
-        tfl += [MaybePILToTensor()]
-    else:
-        tfl += [
-            MaybeToTensor(),
-            transforms.Normalize(
-                mean=torch.tensor(mean),
-                std=torch.tensor(std),
-            ),
-        ]
+    tfl += _output_stage(mean, std, use_prefetcher, normalize)
     return transforms.Compose(tfl)
+
+
+def _crop_stage(img_size, train_crop_mode, scale, ratio, interpolation):
+    """Primary train geometry: RRC default, or resize-keep-ratio + crop/pad."""
+    if train_crop_mode in ('rkrc', 'rkrr'):
+        scale = tuple(scale or (0.8, 1.00))
+        ratio = tuple(ratio or (0.9, 1 / .9))
+        crop_cls = CenterCropOrPad if train_crop_mode == 'rkrc' else RandomCropOrPad
+        return [
+            ResizeKeepRatio(
+                img_size,
+                interpolation=interpolation,
+                random_scale_prob=0.5,
+                random_scale_range=scale,
+                random_scale_area=True,  # scale semantics compatible with RRC
+                random_aspect_prob=0.5,
+                random_aspect_range=ratio,
+            ),
+            crop_cls(img_size, padding_mode='reflect'),
+        ]
+    scale = tuple(scale or (0.08, 1.0))       # ImageNet defaults
+    ratio = tuple(ratio or (3. / 4., 4. / 3.))
+    return [RandomResizedCropAndInterpolation(img_size, scale=scale, ratio=ratio, interpolation=interpolation)]
+
+
+def _auto_augment_stage(auto_augment, img_size, mean, interpolation):
+    """Build the AA/RA/AugMix op from a config string."""
+    img_size_min = min(img_size) if isinstance(img_size, (tuple, list)) else img_size
+    aa_params = dict(
+        translate_const=int(img_size_min * 0.45),
+        img_mean=tuple(min(255, round(255 * x)) for x in mean),
+    )
+    if interpolation and interpolation != 'random':
+        aa_params['interpolation'] = str_to_pil_interp(interpolation)
+    if auto_augment.startswith('rand'):
+        return rand_augment_transform(auto_augment, aa_params)
+    if auto_augment.startswith('augmix'):
+        aa_params['translate_pct'] = 0.3
+        return augment_and_mix_transform(auto_augment, aa_params)
+    return auto_augment_transform(auto_augment, aa_params)
 
 
 def transforms_imagenet_train(
@@ -72,37 +118,14 @@ def transforms_imagenet_train(
         normalize: bool = True,
         separate: bool = False,
 ):
-    """ImageNet-oriented image transforms for training (reference `:65`).
+    """Full ImageNet train pipeline (reference `:65`).
 
-    Returns either a single Compose or, when `separate`, a tuple of 3 stages
-    (primary crop/flip, secondary color/aug, final tensor+norm+erase) for
-    AugMix split handling.
+    With `separate` the three stages (geometry+flip, color/aug, output+erase)
+    come back as a tuple for AugMix split handling.
     """
     train_crop_mode = train_crop_mode or 'rrc'
     assert train_crop_mode in {'rrc', 'rkrc', 'rkrr'}
-    if train_crop_mode in ('rkrc', 'rkrr'):
-        # FIXME integration of RKR is a WIP
-        scale = tuple(scale or (0.8, 1.00))
-        ratio = tuple(ratio or (0.9, 1/.9))
-        primary_tfl = [
-            ResizeKeepRatio(
-                img_size,
-                interpolation=interpolation,
-                random_scale_prob=0.5,
-                random_scale_range=scale,
-                random_scale_area=True,  # scale compatible with RRC
-                random_aspect_prob=0.5,
-                random_aspect_range=ratio,
-            ),
-            CenterCropOrPad(img_size, padding_mode='reflect')
-            if train_crop_mode == 'rkrc' else
-            RandomCropOrPad(img_size, padding_mode='reflect')
-        ]
-    else:
-        scale = tuple(scale or (0.08, 1.0))  # default imagenet scale range
-        ratio = tuple(ratio or (3. / 4., 4. / 3.))  # default imagenet ratio range
-        primary_tfl = [
-            RandomResizedCropAndInterpolation(img_size, scale=scale, ratio=ratio, interpolation=interpolation)]
+    primary_tfl = _crop_stage(img_size, train_crop_mode, scale, ratio, interpolation)
     if hflip > 0.:
         primary_tfl += [transforms.RandomHorizontalFlip(p=hflip)]
     if vflip > 0.:
@@ -112,89 +135,40 @@ def transforms_imagenet_train(
     disable_color_jitter = False
     if auto_augment:
         assert isinstance(auto_augment, str)
-        # color jitter is typically disabled if AA/RA on,
-        # this allows override without breaking old hparm cfgs
+        # AA/RA traditionally replaces color jitter; '3a' policies and the
+        # force flag keep both
         disable_color_jitter = not (force_color_jitter or '3a' in auto_augment)
-        if isinstance(img_size, (tuple, list)):
-            img_size_min = min(img_size)
-        else:
-            img_size_min = img_size
-        aa_params = dict(
-            translate_const=int(img_size_min * 0.45),
-            img_mean=tuple([min(255, round(255 * x)) for x in mean]),
-        )
-        if interpolation and interpolation != 'random':
-            aa_params['interpolation'] = str_to_pil_interp(interpolation)
-        if auto_augment.startswith('rand'):
-            secondary_tfl += [rand_augment_transform(auto_augment, aa_params)]
-        elif auto_augment.startswith('augmix'):
-            aa_params['translate_pct'] = 0.3
-            secondary_tfl += [augment_and_mix_transform(auto_augment, aa_params)]
-        else:
-            secondary_tfl += [auto_augment_transform(auto_augment, aa_params)]
+        secondary_tfl += [_auto_augment_stage(auto_augment, img_size, mean, interpolation)]
 
     if color_jitter is not None and not disable_color_jitter:
-        # color jitter is enabled when not using AA or when forced
         if isinstance(color_jitter, (list, tuple)):
-            # color jitter should be a 3-tuple/list if spec brightness/contrast/saturation
-            # or 4 if also augmenting hue
+            # 3-tuple = brightness/contrast/saturation, 4th adds hue
             assert len(color_jitter) in (3, 4)
         else:
-            # if it's a scalar, duplicate for brightness, contrast, and saturation, no hue
             color_jitter = (float(color_jitter),) * 3
+        jitter = transforms.ColorJitter(*color_jitter)
         if color_jitter_prob is not None:
-            secondary_tfl += [
-                transforms.RandomApply([
-                    transforms.ColorJitter(*color_jitter),
-                ],
-                    p=color_jitter_prob,
-                )
-            ]
-        else:
-            secondary_tfl += [transforms.ColorJitter(*color_jitter)]
+            jitter = transforms.RandomApply([jitter], p=color_jitter_prob)
+        secondary_tfl += [jitter]
 
     if grayscale_prob:
         secondary_tfl += [transforms.RandomGrayscale(p=grayscale_prob)]
-
     if gaussian_blur_prob:
         secondary_tfl += [
-            transforms.RandomApply([
-                transforms.GaussianBlur(kernel_size=23),  # hardcoded for now
-            ],
-                p=gaussian_blur_prob,
-            )
-        ]
+            transforms.RandomApply([transforms.GaussianBlur(kernel_size=23)], p=gaussian_blur_prob)]
 
-    final_tfl = []
-    if use_prefetcher:
-        # prefetcher and collate will handle tensor conversion and norm
-        final_tfl += [ToNumpy()]
-    elif not normalize:
-        # when normalize disable, converted to tensor without scaling, keep original dtype
-        final_tfl += [MaybePILToTensor()]
-    else:
+    final_tfl = _output_stage(mean, std, use_prefetcher, normalize)
+    if not use_prefetcher and normalize and re_prob > 0.:
         final_tfl += [
-            MaybeToTensor(),
-            transforms.Normalize(
-                mean=torch.tensor(mean),
-                std=torch.tensor(std),
-            ),
-        ]
-        if re_prob > 0.:
-            final_tfl += [
-                RandomErasing(
-                    re_prob,
-                    mode=re_mode,
-                    max_count=re_count,
-                    num_splits=re_num_splits,
-                    device='cpu',
-                )
-            ]
+            RandomErasing(re_prob, mode=re_mode, max_count=re_count, num_splits=re_num_splits, device='cpu')]
 
     if separate:
-        return transforms.Compose(primary_tfl), transforms.Compose(secondary_tfl), transforms.Compose(final_tfl)
-    else:
-        return transforms.Compose(primary_tfl + secondary_tfl + final_tfl)
+        return (
+            transforms.Compose(primary_tfl),
+            transforms.Compose(secondary_tfl),
+            transforms.Compose(final_tfl),
+        )
+    return transforms.Compose(primary_tfl + secondary_tfl + final_tfl)
 
 
 def transforms_imagenet_eval(
@@ -208,63 +182,41 @@ def transforms_imagenet_eval(
         use_prefetcher: bool = False,
         normalize: bool = True,
 ):
-    """ImageNet-oriented image transform for evaluation and inference (reference `:273`)."""
+    """Eval pipeline (reference `:273`): resize to img_size/crop_pct then
+    crop by mode — 'squash' (no aspect preserve), 'border' (pad to keep all
+    pixels) or default center crop."""
     crop_pct = crop_pct or DEFAULT_CROP_PCT
 
     if isinstance(img_size, (tuple, list)):
         assert len(img_size) == 2
-        scale_size = tuple([math.floor(x / crop_pct) for x in img_size])
+        scale_size = tuple(math.floor(x / crop_pct) for x in img_size)
     else:
-        scale_size = math.floor(img_size / crop_pct)
-        scale_size = (scale_size, scale_size)
+        scale_size = (math.floor(img_size / crop_pct),) * 2
 
     tfl = []
-
     if crop_border_pixels:
         tfl += [TrimBorder(crop_border_pixels)]
 
     if crop_mode == 'squash':
-        # squash mode scales each edge to 1/pct of target, then crops
-        # aspect ratio is not preserved, no img lost if crop_pct == 1.0
         tfl += [
             transforms.Resize(scale_size, interpolation=str_to_interp_mode(interpolation)),
             transforms.CenterCrop(img_size),
         ]
     elif crop_mode == 'border':
-        # scale the longest edge of image to 1/pct of target edge, add borders to pad, then crop
-        # no image lost if crop_pct == 1.0
         fill = [round(255 * v) for v in mean]
         tfl += [
             ResizeKeepRatio(scale_size, interpolation=interpolation, longest=1.0),
             CenterCropOrPad(img_size, fill=fill),
         ]
     else:
-        # default crop model is center
-        # aspect ratio is preserved, crops center within image, no borders are added, image is lost
+        # center crop: shortest-edge resize keeps aspect, crop discards rest
         if scale_size[0] == scale_size[1]:
-            # simple case, use torchvision built-in Resize w/ shortest edge mode (scalar size arg)
-            tfl += [
-                transforms.Resize(scale_size[0], interpolation=str_to_interp_mode(interpolation))
-            ]
+            tfl += [transforms.Resize(scale_size[0], interpolation=str_to_interp_mode(interpolation))]
         else:
-            # resize the shortest edge to matching target dim for non-square target
             tfl += [ResizeKeepRatio(scale_size)]
         tfl += [transforms.CenterCrop(img_size)]
 
-    if use_prefetcher:
-        # prefetcher and collate will handle tensor conversion and norm
-        tfl += [ToNumpy()]
-    elif not normalize:
-        # when normalize disabled, converted to tensor without scaling, keeps original dtype
-        tfl += [MaybePILToTensor()]
-    else:
-        tfl += [
-            MaybeToTensor(),
-            transforms.Normalize(
-                mean=torch.tensor(mean),
-                std=torch.tensor(std),
-            ),
-        ]
+    tfl += _output_stage(mean, std, use_prefetcher, normalize)
     return transforms.Compose(tfl)
 
 
@@ -297,23 +249,17 @@ def create_transform(
         normalize: bool = True,
         separate: bool = False,
 ):
-    """Factory for image transforms (reference `:387`)."""
-    if isinstance(input_size, (tuple, list)):
-        img_size = input_size[-2:]
-    else:
-        img_size = input_size
+    """Transform dispatcher (reference `:387`)."""
+    img_size = input_size[-2:] if isinstance(input_size, (tuple, list)) else input_size
 
     if tf_preprocessing and use_prefetcher:
-        assert not separate, "Separate transforms not supported for TF preprocessing"
+        assert not separate, 'Separate transforms not supported for TF preprocessing'
         from .tf_preprocessing import TfPreprocessTransform
-        transform = TfPreprocessTransform(
-            is_training=is_training,
-            size=img_size,
-            interpolation=interpolation,
-        )
-    elif is_training and no_aug:
-        assert not separate, "Cannot perform split augmentation with no_aug"
-        transform = transforms_noaug_train(
+        return TfPreprocessTransform(is_training=is_training, size=img_size, interpolation=interpolation)
+
+    if is_training and no_aug:
+        assert not separate, 'Cannot perform split augmentation with no_aug'
+        return transforms_noaug_train(
             img_size,
             interpolation=interpolation,
             mean=mean,
@@ -321,8 +267,9 @@ def create_transform(
             use_prefetcher=use_prefetcher,
             normalize=normalize,
         )
-    elif is_training:
-        transform = transforms_imagenet_train(
+
+    if is_training:
+        return transforms_imagenet_train(
             img_size,
             train_crop_mode=train_crop_mode,
             scale=scale,
@@ -345,18 +292,16 @@ def create_transform(
             normalize=normalize,
             separate=separate,
         )
-    else:
-        assert not separate, "Separate transforms not supported for validation preprocessing"
-        transform = transforms_imagenet_eval(
-            img_size,
-            interpolation=interpolation,
-            mean=mean,
-            std=std,
-            crop_pct=crop_pct,
-            crop_mode=crop_mode,
-            crop_border_pixels=crop_border_pixels,
-            use_prefetcher=use_prefetcher,
-            normalize=normalize,
-        )
 
-    return transform
+    assert not separate, 'Separate transforms not supported for validation preprocessing'
+    return transforms_imagenet_eval(
+        img_size,
+        interpolation=interpolation,
+        mean=mean,
+        std=std,
+        crop_pct=crop_pct,
+        crop_mode=crop_mode,
+        crop_border_pixels=crop_border_pixels,
+        use_prefetcher=use_prefetcher,
+        normalize=normalize,
+    )
