@@ -32,51 +32,48 @@ _logger = logging.getLogger(__name__)
 
 
 def fast_collate(batch):
-    """A fast collation function optimized for uint8 images (np array or torch)
-    and int64 targets (labels)."""
+    """uint8 batch assembly without the default-collate overhead
+    (reference `loader.py:30`). Handles numpy arrays, torch tensors, and
+    aug-split tuples of arrays (flattened split-major so torch.split by
+    batch_size recovers each split)."""
     assert isinstance(batch[0], tuple)
     batch_size = len(batch)
-    if isinstance(batch[0][0], tuple):
-        # This branch 'deinterleaves' and flattens tuples of input tensors into
-        # one tensor ordered by position such that all tuple of position n will end up
-        # in a torch.split(tensor, batch_size) in nth position
-        inner_tuple_size = len(batch[0][0])
-        flattened_batch_size = batch_size * inner_tuple_size
-        targets = torch.zeros(flattened_batch_size, dtype=torch.int64)
-        tensor = torch.zeros((flattened_batch_size, *batch[0][0][0].shape), dtype=torch.uint8)
-        for i in range(batch_size):
-            assert len(batch[i][0]) == inner_tuple_size  # all input tensor tuples must be same length
-            for j in range(inner_tuple_size):
-                targets[i + j * batch_size] = batch[i][1]
-                tensor[i + j * batch_size] += torch.from_numpy(batch[i][0][j])
+    sample0 = batch[0][0]
+
+    if isinstance(sample0, tuple):
+        # aug-split samples: position j of every sample lands in slot
+        # [j * batch_size + i] so splits stay contiguous
+        inner = len(sample0)
+        targets = torch.zeros(batch_size * inner, dtype=torch.int64)
+        tensor = torch.zeros((batch_size * inner, *sample0[0].shape), dtype=torch.uint8)
+        for i, (imgs, target) in enumerate(batch):
+            assert len(imgs) == inner
+            for j, img in enumerate(imgs):
+                targets[i + j * batch_size] = target
+                tensor[i + j * batch_size] += torch.from_numpy(img)
         return tensor, targets
-    elif isinstance(batch[0][0], np.ndarray):
-        targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
-        assert len(targets) == batch_size
-        tensor = torch.zeros((batch_size, *batch[0][0].shape), dtype=torch.uint8)
-        for i in range(batch_size):
-            tensor[i] += torch.from_numpy(batch[i][0])
-        return tensor, targets
-    elif isinstance(batch[0][0], torch.Tensor):
-        targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
-        assert len(targets) == batch_size
-        tensor = torch.zeros((batch_size, *batch[0][0].shape), dtype=torch.uint8)
-        for i in range(batch_size):
-            tensor[i].copy_(batch[i][0])
-        return tensor, targets
+
+    targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
+    tensor = torch.zeros((batch_size, *sample0.shape), dtype=torch.uint8)
+    if isinstance(sample0, np.ndarray):
+        for i, (img, _) in enumerate(batch):
+            tensor[i] += torch.from_numpy(img)
+    elif isinstance(sample0, torch.Tensor):
+        for i, (img, _) in enumerate(batch):
+            tensor[i].copy_(img)
     else:
-        assert False
+        raise AssertionError(f'Unexpected sample type {type(sample0)}')
+    return tensor, targets
 
 
 def adapt_to_chs(x, n):
+    """Broadcast / reconcile per-channel normalization stats to n channels."""
     if not isinstance(x, (tuple, list)):
-        x = tuple(repeat(x, n))
-    elif len(x) != n:
-        x_mean = np.mean(x).item()
-        x = (x_mean,) * n
-        _logger.warning(f'Pretrained mean/std different shape than model, using avg value {x}.')
-    else:
-        assert len(x) == n, 'normalization stats must match image channels'
+        return tuple(repeat(x, n))
+    if len(x) != n:
+        mean_val = np.mean(x).item()
+        _logger.warning(f'Pretrained mean/std different shape than model, using avg value {(mean_val,) * n}.')
+        return (mean_val,) * n
     return x
 
 
