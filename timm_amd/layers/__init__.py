@@ -77,3 +77,13 @@ from .trace_utils import _assert
 from .weight_init import (
     init_weight_jax, init_weight_vit, lecun_normal_, trunc_normal_, trunc_normal_tf_, variance_scaling_,
 )
+from .activations_me import HardMishMe, HardSigmoidMe, HardSwishMe, MishMe, SwishMe
+from .fast_norm import (
+    fast_group_norm, fast_layer_norm, fast_rms_norm, is_fast_norm, set_fast_norm,
+)
+from .interpolate import RegularGridInterpolator
+from .median_pool import MedianPool2d
+from .ml_decoder import MLDecoder, add_ml_decoder_head
+from .pool1d import global_pool_nlc
+from .inplace_abn import InplaceAbn
+from .typing import LayerType, PadType, disable_compiler, nullwrap
