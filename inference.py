@@ -143,10 +143,17 @@ def main():
 
     to_label = None
     if args.label_type in ('name', 'description', 'detail'):
-        class_to_idx = getattr(dataset, 'class_to_idx', None) or getattr(dataset.reader, 'class_to_idx', {})
-        if class_to_idx:
-            idx_to_class = {v: k for k, v in class_to_idx.items()}
-            to_label = lambda x: idx_to_class.get(x, str(x))  # noqa: E731
+        # prefer the shipped ImageNet synset metadata when the classifier
+        # width matches a known subset; fall back to the dataset's class map
+        from timm_amd.data import DatasetInfoLabelMapper, ImageNetInfo, infer_imagenet_subset
+        subset = infer_imagenet_subset(model)
+        if subset is not None:
+            to_label = DatasetInfoLabelMapper(ImageNetInfo(subset), label_type=args.label_type)
+        else:
+            class_to_idx = getattr(dataset, 'class_to_idx', None) or getattr(dataset.reader, 'class_to_idx', {})
+            if class_to_idx:
+                idx_to_class = {v: k for k, v in class_to_idx.items()}
+                to_label = lambda x: idx_to_class.get(x, str(x))  # noqa: E731
     top_k = min(args.topk, args.num_classes)
     batch_time = AverageMeter()
     end = time.time()
