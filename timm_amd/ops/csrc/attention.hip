@@ -23,7 +23,7 @@ namespace {
 typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
 
 constexpr int kQTile = 64;     // q rows per block
-constexpr int kKvTile = 32;    // kv rows per LDS stage
+constexpr int kKvTile = 64;    // kv rows per LDS stage (4 fragment columns)
 constexpr int kBlockThreads = 256;
 constexpr float kNegInf = -1e30f;
 
@@ -127,15 +127,16 @@ void attn_fwd_kernel(
     }
     __syncthreads();
 
-    // ---- S = scale * Q K^T (+mask), two 16x16 fragments along kv ----
-    float p_frag[2][4];  // [n16][reg] exp'd probabilities
+    // ---- S = scale * Q K^T (+mask), kKvTile/16 fragments along kv ----
+    constexpr int kNF = kKvTile / 16;
+    float p_frag[kNF][4];  // [n16][reg] exp'd probabilities
     float m_tile[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) m_tile[r] = kNegInf;
 
-    float s_frag[2][4];
+    float s_frag[kNF][4];
 #pragma unroll
-    for (int n16 = 0; n16 < 2; ++n16) {
+    for (int n16 = 0; n16 < kNF; ++n16) {
       f32x4 acc = {0.f, 0.f, 0.f, 0.f};
       for (int d0 = 0; d0 < D; d0 += 32) {
         bf16x8_t a = *reinterpret_cast<const bf16x8_t*>(
@@ -174,7 +175,7 @@ void attn_fwd_kernel(
       m_run[r] = m_new;
       float psum = 0.f;
 #pragma unroll
-      for (int n16 = 0; n16 < 2; ++n16) {
+      for (int n16 = 0; n16 < kNF; ++n16) {
         float p = (s_frag[n16][r] <= kNegInf) ? 0.f : __expf(s_frag[n16][r] - m_new);
         p_frag[n16][r] = p;
         psum += p;
@@ -182,9 +183,9 @@ void attn_fwd_kernel(
       l_run[r] = l_run[r] * factor[r] + group16_reduce_sum(psum);
     }
 
-    // ---- write P to this wave's LDS region (16 rows x 32 cols) ----
+    // ---- write P to this wave's LDS region (16 rows x kKvTile cols) ----
 #pragma unroll
-    for (int n16 = 0; n16 < 2; ++n16) {
+    for (int n16 = 0; n16 < kNF; ++n16) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         p_lds[wave * (16 * kKvTile) + (g4 * 4 + r) * kKvTile + n16 * 16 + l16] =
@@ -193,20 +194,28 @@ void attn_fwd_kernel(
     }
     __asm__ volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
-    // ---- O += P V, rescaling old accumulator ----
-    bf16x8_t pa = *reinterpret_cast<const bf16x8_t*>(
-        p_lds + wave * (16 * kKvTile) + l16 * kKvTile + g4 * 8);
+    // ---- O += P V, rescaling old accumulator; kv extent covered in
+    //      kKvTile/32 MFMA k-steps ----
+    bf16x8_t pa[kKvTile / 32];
+#pragma unroll
+    for (int ks = 0; ks < kKvTile / 32; ++ks) {
+      pa[ks] = *reinterpret_cast<const bf16x8_t*>(
+          p_lds + wave * (16 * kKvTile) + l16 * kKvTile + ks * 32 + g4 * 8);
+    }
 #pragma unroll
     for (int f = 0; f < kMaxD / 16; ++f) {
       if (f * 16 >= D) break;
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc_o[f][r] *= factor[r];
-      // B operand: V^T[dcol=f*16+l16][kv=g4*8 .. +8) — one vector LDS read
       const int drow = f * 16 + l16;
-      bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(
-          reinterpret_cast<const char*>(vt_lds) +
-          swz64<kSwizzle>(drow, drow * kKvTile * 2 + g4 * 8 * 2));
-      acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, bv, acc_o[f], 0, 0, 0);
+#pragma unroll
+      for (int ks = 0; ks < kKvTile / 32; ++ks) {
+        // B operand: V^T[dcol][kv=ks*32+g4*8 .. +8) — one vector LDS read
+        bf16x8_t bv = *reinterpret_cast<const bf16x8_t*>(
+            reinterpret_cast<const char*>(vt_lds) +
+            swz64<kSwizzle>(drow, drow * kKvTile * 2 + (ks * 32 + g4 * 8) * 2));
+        acc_o[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa[ks], bv, acc_o[f], 0, 0, 0);
+      }
     }
     __syncthreads();  // before next tile overwrites K/V
   }
