@@ -713,10 +713,15 @@ def test_model_train_step_gpu(model_name):
     import copy
     import timm_amd
     torch.manual_seed(7)
-    model_cpu = timm_amd.create_model(model_name, num_classes=10, drop_rate=0., drop_path_rate=0.)
+    try:
+        model_cpu = timm_amd.create_model(model_name, num_classes=10, drop_rate=0., drop_path_rate=0.)
+    except TypeError:
+        model_cpu = timm_amd.create_model(model_name, num_classes=10, drop_rate=0.)  # no drop_path arg
     model_gpu = copy.deepcopy(model_cpu).to('cuda', torch.bfloat16)
-    in_sz = model_cpu.pretrained_cfg.get('input_size', (3, 224, 224))[-1]
-    in_sz = min(in_sz, 224)
+    cfg = model_cpu.pretrained_cfg
+    in_sz = cfg.get('input_size', (3, 224, 224))[-1]
+    if not cfg.get('fixed_input_size', False):
+        in_sz = min(in_sz, 224)
     x = torch.randn(2, 3, in_sz, in_sz)
     t = torch.randint(0, 10, (2,))
 
