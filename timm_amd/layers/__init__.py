@@ -39,7 +39,8 @@ from .norm import (
 )
 from .norm_act import (
     BatchNormAct2d, FrozenBatchNormAct2d, GroupNormAct, LayerNormAct, LayerNormAct2d, RmsNormAct, RmsNormAct2d, SyncBatchNormAct,
-    convert_sync_batchnorm, freeze_batch_norm_2d, get_norm_act_layer, unfreeze_batch_norm_2d,
+    convert_sync_batchnorm, create_norm_act_layer, freeze_batch_norm_2d, get_norm_act_layer,
+    unfreeze_batch_norm_2d,
 )
 from .padding import get_padding, get_same_padding, pad_same
 from .patch_dropout import PatchDropout, patch_dropout_forward

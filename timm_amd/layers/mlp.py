@@ -162,7 +162,9 @@ class SwiGLU(nn.Module):
         self.drop2 = nn.Dropout(drop_probs[1])
 
     def init_weights(self):
-        nn.init.ones_(self.fc1_g.bias)
+        # gate starts near-closed: weight ~0, bias 1 (when present)
+        if self.fc1_g.bias is not None:
+            nn.init.ones_(self.fc1_g.bias)
         nn.init.normal_(self.fc1_g.weight, std=1e-6)
 
     def forward(self, x):

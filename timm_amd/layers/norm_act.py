@@ -203,7 +203,24 @@ def _norm_act_map():
         layernorm2d=LayerNormAct2d,
         rmsnorm=RmsNormAct,
         rmsnorm2d=RmsNormAct2d,
+        # evonorms fuse norm+act by construction
+        evonormb0=_evo('EvoNorm2dB0'),
+        evonormb1=_evo('EvoNorm2dB1'),
+        evonormb2=_evo('EvoNorm2dB2'),
+        evonorms0=_evo('EvoNorm2dS0'),
+        evonorms0a=_evo('EvoNorm2dS0a'),
+        evonorms1=_evo('EvoNorm2dS1'),
+        evonorms1a=_evo('EvoNorm2dS1a'),
+        evonorms2=_evo('EvoNorm2dS2'),
+        evonorms2a=_evo('EvoNorm2dS2a'),
+        frn=_evo('FilterResponseNormAct2d'),
+        frntlu=_evo('FilterResponseNormTlu2d'),
     )
+
+
+def _evo(name):
+    from . import evo_norm, filter_response_norm
+    return getattr(evo_norm, name, None) or getattr(filter_response_norm, name)
 _NORM_ACT_TYPES = {BatchNormAct2d, SyncBatchNormAct, FrozenBatchNormAct2d}
 # has act_layer arg to define act type
 _NORM_ACT_REQUIRES_ARG = {BatchNormAct2d}
@@ -350,3 +367,14 @@ class RmsNormAct2d(RmsNorm2d):
         x = self.drop(x)
         x = self.act(x)
         return x
+
+
+def create_norm_act_layer(layer_name, num_features, act_layer=None, apply_act=True, jit=False, **kwargs):
+    """Instantiate a fused norm+act layer by name (reference
+    `timm/layers/create_norm_act.py:create_norm_act_layer`)."""
+    layer = get_norm_act_layer(layer_name, act_layer=act_layer)
+    layer_instance = layer(num_features, apply_act=apply_act, **kwargs)
+    if jit:
+        import torch
+        layer_instance = torch.jit.script(layer_instance)
+    return layer_instance

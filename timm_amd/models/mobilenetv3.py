@@ -971,6 +971,18 @@ def mobilenetv3_large_100(pretrained=False, **kwargs) -> MobileNetV3:
 
 
 @register_model
+def mobilenetv3_large_150d(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_large_150d', 1.5, depth_multiplier=1.2, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
+def mobilenetv3_small_050(pretrained=False, **kwargs) -> MobileNetV3:
+    model = _gen_mobilenet_v3('mobilenetv3_small_050', 0.50, pretrained=pretrained, **kwargs)
+    return model
+
+
+@register_model
 def mobilenetv3_large_075(pretrained=False, **kwargs) -> MobileNetV3:
     model = _gen_mobilenet_v3('mobilenetv3_large_075', 0.75, pretrained=pretrained, **kwargs)
     return model

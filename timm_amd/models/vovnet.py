@@ -372,7 +372,10 @@ def _cfg(url='', **kwargs):
     }
 
 
+model_cfgs['ese_vovnet39b_evos'] = model_cfgs['ese_vovnet39b']
+
 default_cfgs = generate_default_cfgs({
+    'ese_vovnet39b_evos.untrained': _cfg(url=''),
     'vovnet39a.untrained': _cfg(url=''),
     'vovnet57a.untrained': _cfg(url=''),
     'ese_vovnet19b_slim_dw.untrained': _cfg(url=''),
@@ -423,6 +426,14 @@ def ese_vovnet19b_slim(pretrained=False, **kwargs) -> VovNet:
 @register_model
 def ese_vovnet39b(pretrained=False, **kwargs) -> VovNet:
     return _create_vovnet('ese_vovnet39b', pretrained=pretrained, **kwargs)
+
+
+@register_model
+def ese_vovnet39b_evos(pretrained=False, **kwargs) -> VovNet:
+    def norm_act_fn(num_features, **nkwargs):
+        from ..layers import create_norm_act_layer
+        return create_norm_act_layer('evonorms0', num_features, jit=False, **nkwargs)
+    return _create_vovnet('ese_vovnet39b_evos', pretrained=pretrained, norm_layer=norm_act_fn, **kwargs)
 
 
 @register_model
