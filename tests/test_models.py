@@ -30,6 +30,7 @@ EXCLUDE_FILTERS = [
     'nfnet_f5*', 'nfnet_f6*', 'nfnet_f7*', 'eca_nfnet_l3', 'regnety_640', 'regnety_1280',
     'regnety_2560', 'resnet50x16_clip*', 'resnet50x64_clip*', 'resnext101_32x16d', 'resnext101_32x32d',
     'vit_7b*', 'eva02_enormous*', 'vit_giantopt*', 'sam2_hiera_large*',
+    'aimv2_1b*', 'aimv2_3b*', 'beit3_giant*', 'mobilenetv3_large_150d',
 ]
 
 

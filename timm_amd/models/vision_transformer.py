@@ -617,6 +617,7 @@ class VisionTransformer(nn.Module):
             norm_layer: Optional[Union[str, Callable, Type[nn.Module]]] = None,
             act_layer: Optional[Union[str, Callable, Type[nn.Module]]] = None,
             block_fn: Type[nn.Module] = Block,
+            pool_include_prefix: bool = False,
             attn_layer: Union[str, Type[nn.Module], None] = None,
             mlp_layer: Type[nn.Module] = Mlp,
     ) -> None:
@@ -709,8 +710,10 @@ class VisionTransformer(nn.Module):
                 norm_layer=norm_layer,
                 act_layer=act_layer,
             )
+            pool_include_prefix = True
         else:
             self.attn_pool = None
+        self.pool_include_prefix = pool_include_prefix
         self.fc_norm = norm_layer(embed_dim) if final_norm and use_fc_norm else nn.Identity()
         self.head_drop = nn.Dropout(drop_rate)
         self.head = nn.Linear(self.embed_dim, num_classes) if num_classes > 0 else nn.Identity()
@@ -957,10 +960,14 @@ class VisionTransformer(nn.Module):
 
     def pool(self, x: torch.Tensor, pool_type: Optional[str] = None) -> torch.Tensor:
         if self.attn_pool is not None:
+            if not self.pool_include_prefix:
+                x = x[:, self.num_prefix_tokens:]
             x = self.attn_pool(x)
             return x
         pool_type = self.global_pool if pool_type is None else pool_type
-        x = global_pool_nlc(x, pool_type=pool_type, num_prefix_tokens=self.num_prefix_tokens)
+        x = global_pool_nlc(
+            x, pool_type=pool_type, num_prefix_tokens=self.num_prefix_tokens,
+            reduce_include_prefix=self.pool_include_prefix)
         return x
 
     def forward_head(self, x: torch.Tensor, pre_logits: bool = False) -> torch.Tensor:
