@@ -43,6 +43,11 @@ INTEROP_MODELS = [
     ('convnextv2_atto', 128),
     ('swin_tiny_patch4_window7_224', 224),
     ('swinv2_tiny_window8_256', 256),
+    ('aimv2_large_patch14_224', 224),
+    ('flexivit_small', 240),
+    ('beit3_base_patch16_224', 224),
+    ('test_vit3', 160),
+    ('mobilenetv3_small_050', 128),
     ('coatnet_nano_rw_224', 224),
     ('coatnet_0_rw_224', 224),
     ('maxvit_rmlp_nano_rw_256', 256),
